@@ -1,0 +1,51 @@
+"""Data source registry (parquet/csv/json/memory).
+
+ref: crates/sail-data-source/src/formats/ for the reference's format set.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+from ..engine import types as T
+
+
+def infer_source_schema(fmt: str, paths: List[str], options: Dict[str, str]):
+    from . import parquet_io, csv_io
+
+    if fmt == "parquet":
+        return parquet_io.infer_schema(paths)
+    if fmt == "csv":
+        return csv_io.infer_schema(paths, options)
+    if fmt == "json":
+        from . import json_io
+
+        return json_io.infer_schema(paths, options)
+    raise ValueError(f"unsupported format {fmt}")
+
+
+def read_source(fmt: str, paths: List[str], options: Dict[str, str], schema, device):
+    from . import parquet_io, csv_io
+
+    if fmt == "parquet":
+        return parquet_io.read(paths, schema, device, options)
+    if fmt == "csv":
+        return csv_io.read(paths, schema, device, options)
+    if fmt == "json":
+        from . import json_io
+
+        return json_io.read(paths, schema, device, options)
+    raise ValueError(f"unsupported format {fmt}")
+
+
+def write_source(fmt: str, path: str, chunk, mode: str, options, partition_by):
+    from . import parquet_io, csv_io
+
+    if fmt == "parquet":
+        return parquet_io.write(path, chunk, mode, options)
+    if fmt == "csv":
+        return csv_io.write(path, chunk, mode, options)
+    if fmt == "delta":
+        from . import delta
+
+        return delta.write(path, chunk, mode, options)
+    raise ValueError(f"unsupported write format {fmt}")

@@ -1,0 +1,359 @@
+"""Columnar device data model.
+
+A Column is a torch tensor living in HBM (or host RAM for the CPU path) plus
+optional validity and string payloads. Unlike the reference — which streams
+8192-row Arrow batches through pull-based operator trees
+(ref: crates/sail-common/src/config/application.yaml execution.batch_size) —
+this engine holds *whole table partitions* resident in the 288 GB HBM of each
+GPU and runs operators over entire columns in one kernel launch.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from . import types as T
+
+
+def _dev(device) -> torch.device:
+    return torch.device(device) if not isinstance(device, torch.device) else device
+
+
+class Column:
+    """A typed column of values.
+
+    Fixed-width types: `data` holds the values (decimal -> scaled int64,
+    date -> int32 days, timestamp -> int64 micros).
+    Strings: `data` holds int64 offsets? No — see StringColumn.
+
+    `validity`: optional uint8 mask tensor, 1 = valid. None means all-valid.
+    """
+
+    __slots__ = ("dtype", "data", "validity")
+
+    def __init__(self, dtype: T.DataType, data: torch.Tensor, validity: Optional[torch.Tensor] = None):
+        self.dtype = dtype
+        self.data = data
+        self.validity = validity
+
+    # -- construction ------------------------------------------------------
+    @staticmethod
+    def from_values(values: Sequence, dtype: T.DataType, device="cpu") -> "Column":
+        device = _dev(device)
+        if isinstance(dtype, T.StringType):
+            return StringColumn.from_pylist(list(values), device=device, dtype=dtype)
+        validity = None
+        if any(v is None for v in values):
+            validity = torch.tensor([0 if v is None else 1 for v in values], dtype=torch.uint8, device=device)
+        if isinstance(dtype, T.DecimalType):
+            scale = 10 ** dtype.scale
+            vals = [0 if v is None else int(round(float(v) * scale)) for v in values]
+            data = torch.tensor(vals, dtype=torch.int64, device=device)
+        elif isinstance(dtype, T.DateType):
+            vals = [0 if v is None else _to_days(v) for v in values]
+            data = torch.tensor(vals, dtype=torch.int32, device=device)
+        elif isinstance(dtype, T.BooleanType):
+            data = torch.tensor([bool(v) for v in [False if v is None else v for v in values]], dtype=torch.bool, device=device)
+        else:
+            zero = 0 if dtype.is_integer or dtype.is_temporal else 0.0
+            vals = [zero if v is None else v for v in values]
+            data = torch.tensor(vals, dtype=dtype.storage, device=device)
+        return Column(dtype, data, validity)
+
+    # -- basics ------------------------------------------------------------
+    def __len__(self) -> int:
+        return int(self.data.shape[0])
+
+    @property
+    def device(self) -> torch.device:
+        return self.data.device
+
+    @property
+    def is_cuda(self) -> bool:
+        return self.data.is_cuda
+
+    def to(self, device) -> "Column":
+        device = _dev(device)
+        if self.device == device:
+            return self
+        return Column(self.dtype, self.data.to(device),
+                      self.validity.to(device) if self.validity is not None else None)
+
+    def gather(self, indices: torch.Tensor) -> "Column":
+        v = self.validity.index_select(0, indices) if self.validity is not None else None
+        return Column(self.dtype, self.data.index_select(0, indices), v)
+
+    def filter(self, mask: torch.Tensor) -> "Column":
+        v = self.validity[mask] if self.validity is not None else None
+        return Column(self.dtype, self.data[mask], v)
+
+    def slice(self, start: int, length: int) -> "Column":
+        v = self.validity[start : start + length] if self.validity is not None else None
+        return Column(self.dtype, self.data[start : start + length], v)
+
+    def null_count(self) -> int:
+        if self.validity is None:
+            return 0
+        return int(len(self) - int(self.validity.sum().item()))
+
+    def valid_mask(self) -> torch.Tensor:
+        if self.validity is None:
+            return torch.ones(len(self), dtype=torch.bool, device=self.device)
+        return self.validity.to(torch.bool)
+
+    # -- host conversion ---------------------------------------------------
+    def to_pylist(self) -> List:
+        vals = self.data.cpu()
+        out: List = []
+        scale = 10 ** self.dtype.scale if isinstance(self.dtype, T.DecimalType) else None
+        vmask = self.validity.cpu().tolist() if self.validity is not None else None
+        lst = vals.tolist()
+        for i, v in enumerate(lst):
+            if vmask is not None and not vmask[i]:
+                out.append(None)
+            elif scale is not None:
+                out.append(v / scale)
+            elif isinstance(self.dtype, T.DateType):
+                out.append(_from_days(v))
+            else:
+                out.append(v)
+        return out
+
+    def __repr__(self):
+        return f"Column({self.dtype!r}, n={len(self)}, dev={self.device}, nulls={self.null_count()})"
+
+
+class StringColumn(Column):
+    """Arrow-style UTF-8 column: int64 offsets (n+1) + uint8 bytes, with an
+    optional dictionary encoding (codes int32 into a unique-values column).
+
+    When `codes` is not None the column is dictionary-encoded: `offsets`/
+    `bytes_` describe the *dictionary* values and `codes[i]` picks row i's
+    value (-1 = null). Low-cardinality TPC-H columns (flags, status,
+    segments, ...) stay dict-encoded end-to-end so that predicates and joins
+    on them are integer ops on device.
+    """
+
+    __slots__ = ("offsets", "bytes_", "codes")
+
+    def __init__(self, offsets: torch.Tensor, bytes_: torch.Tensor,
+                 validity: Optional[torch.Tensor] = None,
+                 codes: Optional[torch.Tensor] = None,
+                 dtype: Optional[T.DataType] = None):
+        n = (codes.shape[0] if codes is not None else offsets.shape[0] - 1)
+        # `data` for a string column is a row-count-sized placeholder view used
+        # only for len()/device; real payloads are offsets/bytes_/codes.
+        anchor = codes if codes is not None else offsets[:-1] if offsets.numel() else offsets
+        super().__init__(dtype or T.STRING, anchor, validity)
+        self.offsets = offsets
+        self.bytes_ = bytes_
+        self.codes = codes
+
+    # -- construction ------------------------------------------------------
+    @staticmethod
+    def from_pylist(values: List[Optional[str]], device="cpu", dict_encode: Optional[bool] = None,
+                    dtype: Optional[T.DataType] = None) -> "StringColumn":
+        device = _dev(device)
+        n = len(values)
+        validity = None
+        if any(v is None for v in values):
+            validity = torch.tensor([0 if v is None else 1 for v in values], dtype=torch.uint8, device=device)
+        uniq = set(v for v in values if v is not None)
+        if dict_encode is None:
+            dict_encode = n > 64 and len(uniq) * 16 < n
+        if dict_encode:
+            udict = sorted(uniq)
+            idx = {s: i for i, s in enumerate(udict)}
+            codes = torch.tensor([(-1 if v is None else idx[v]) for v in values], dtype=torch.int32, device=device)
+            offsets, bytes_ = _pack_strings(udict, device)
+            return StringColumn(offsets, bytes_, validity, codes, dtype=dtype)
+        offsets, bytes_ = _pack_strings(["" if v is None else v for v in values], device)
+        return StringColumn(offsets, bytes_, validity, None, dtype=dtype)
+
+    @staticmethod
+    def from_buffers(offsets: torch.Tensor, bytes_: torch.Tensor, validity=None, codes=None) -> "StringColumn":
+        return StringColumn(offsets, bytes_, validity, codes)
+
+    # -- basics ------------------------------------------------------------
+    def __len__(self) -> int:
+        if self.codes is not None:
+            return int(self.codes.shape[0])
+        return int(self.offsets.shape[0]) - 1
+
+    @property
+    def is_dict(self) -> bool:
+        return self.codes is not None
+
+    @property
+    def dict_size(self) -> int:
+        return int(self.offsets.shape[0]) - 1
+
+    def to(self, device) -> "StringColumn":
+        device = _dev(device)
+        if self.device == device:
+            return self
+        return StringColumn(self.offsets.to(device), self.bytes_.to(device),
+                            self.validity.to(device) if self.validity is not None else None,
+                            self.codes.to(device) if self.codes is not None else None,
+                            dtype=self.dtype)
+
+    @property
+    def device(self) -> torch.device:
+        return self.offsets.device if self.codes is None else self.codes.device
+
+    def gather(self, indices: torch.Tensor) -> "StringColumn":
+        v = self.validity.index_select(0, indices) if self.validity is not None else None
+        if self.codes is not None:
+            return StringColumn(self.offsets, self.bytes_, v, self.codes.index_select(0, indices), dtype=self.dtype)
+        # materialized gather of raw strings
+        offs, byts = _gather_strings(self.offsets, self.bytes_, indices)
+        return StringColumn(offs, byts, v, None, dtype=self.dtype)
+
+    def filter(self, mask: torch.Tensor) -> "StringColumn":
+        return self.gather(torch.nonzero(mask, as_tuple=False).squeeze(1))
+
+    def slice(self, start: int, length: int) -> "StringColumn":
+        idx = torch.arange(start, start + length, device=self.device)
+        return self.gather(idx)
+
+    def decode_dict(self) -> "StringColumn":
+        """Materialize a dict-encoded column into raw offsets/bytes."""
+        if self.codes is None:
+            return self
+        offs, byts = _gather_strings(self.offsets, self.bytes_, self.codes.clamp_min(0).to(torch.int64))
+        return StringColumn(offs, byts, self.validity, None, dtype=self.dtype)
+
+    def dict_values(self) -> List[str]:
+        """Host copy of the dictionary (or all values if not dict-encoded)."""
+        offs = self.offsets.cpu().numpy()
+        byts = self.bytes_.cpu().numpy().tobytes()
+        return [byts[offs[i]:offs[i + 1]].decode("utf-8", "replace") for i in range(len(offs) - 1)]
+
+    def to_pylist(self) -> List[Optional[str]]:
+        vals = self.dict_values()
+        vmask = self.validity.cpu().tolist() if self.validity is not None else None
+        if self.codes is not None:
+            codes = self.codes.cpu().tolist()
+            out = [vals[c] if c >= 0 else None for c in codes]
+        else:
+            out = list(vals)
+        if vmask is not None:
+            out = [None if not vmask[i] else v for i, v in enumerate(out)]
+        return out
+
+    def __repr__(self):
+        kind = f"dict[{self.dict_size}]" if self.is_dict else "utf8"
+        return f"StringColumn({kind}, n={len(self)}, dev={self.device})"
+
+
+def _pack_strings(strs: List[str], device) -> tuple:
+    enc = [s.encode("utf-8") for s in strs]
+    lens = np.fromiter((len(e) for e in enc), dtype=np.int64, count=len(enc))
+    offsets = np.zeros(len(enc) + 1, dtype=np.int64)
+    np.cumsum(lens, out=offsets[1:])
+    data = b"".join(enc)
+    byts = torch.frombuffer(bytearray(data), dtype=torch.uint8) if data else torch.zeros(0, dtype=torch.uint8)
+    return torch.from_numpy(offsets).to(device), byts.to(device)
+
+
+def _gather_strings(offsets: torch.Tensor, bytes_: torch.Tensor, indices: torch.Tensor):
+    """Gather rows of a raw string column by index (torch fallback path)."""
+    indices = indices.to(torch.int64)
+    starts = offsets.index_select(0, indices)
+    ends = offsets.index_select(0, indices + 1)
+    lens = ends - starts
+    out_offsets = torch.zeros(indices.shape[0] + 1, dtype=torch.int64, device=offsets.device)
+    torch.cumsum(lens, 0, out=out_offsets[1:])
+    total = int(out_offsets[-1].item())
+    out_bytes = torch.empty(total, dtype=torch.uint8, device=bytes_.device)
+    # ranges -> flat gather indices without a python loop
+    if total > 0:
+        pos = torch.arange(total, device=offsets.device)
+        row = torch.searchsorted(out_offsets[1:], pos, right=True)
+        src = starts.index_select(0, row) + (pos - out_offsets.index_select(0, row))
+        out_bytes = bytes_.index_select(0, src)
+    return out_offsets, out_bytes
+
+
+def _to_days(v) -> int:
+    import datetime as _dt
+
+    if isinstance(v, int):
+        return v
+    if isinstance(v, str):
+        y, m, d = v.split("-")
+        v = _dt.date(int(y), int(m), int(d))
+    if isinstance(v, _dt.datetime):
+        v = v.date()
+    return (v - _dt.date(1970, 1, 1)).days
+
+
+def _from_days(days: int):
+    import datetime as _dt
+
+    return _dt.date(1970, 1, 1) + _dt.timedelta(days=int(days))
+
+
+@dataclass
+class Table:
+    """A named collection of equal-length columns (one partition)."""
+
+    columns: Dict[str, Column] = field(default_factory=dict)
+
+    @property
+    def num_rows(self) -> int:
+        for c in self.columns.values():
+            return len(c)
+        return 0
+
+    @property
+    def names(self) -> List[str]:
+        return list(self.columns.keys())
+
+    @property
+    def device(self) -> torch.device:
+        for c in self.columns.values():
+            return c.device
+        return torch.device("cpu")
+
+    def column(self, name: str) -> Column:
+        return self.columns[name]
+
+    def to(self, device) -> "Table":
+        return Table({k: v.to(device) for k, v in self.columns.items()})
+
+    def gather(self, indices: torch.Tensor) -> "Table":
+        return Table({k: v.gather(indices) for k, v in self.columns.items()})
+
+    def filter(self, mask: torch.Tensor) -> "Table":
+        idx = torch.nonzero(mask, as_tuple=False).squeeze(1)
+        return self.gather(idx)
+
+    def select(self, names: Sequence[str]) -> "Table":
+        return Table({n: self.columns[n] for n in names})
+
+    def with_column(self, name: str, col: Column) -> "Table":
+        cols = dict(self.columns)
+        cols[name] = col
+        return Table(cols)
+
+    def rename(self, mapping: Dict[str, str]) -> "Table":
+        return Table({mapping.get(k, k): v for k, v in self.columns.items()})
+
+    def to_pydict(self) -> Dict[str, List]:
+        return {k: v.to_pylist() for k, v in self.columns.items()}
+
+    def to_rows(self) -> List[tuple]:
+        d = self.to_pydict()
+        names = list(d.keys())
+        return [tuple(d[n][i] for n in names) for i in range(self.num_rows)]
+
+    @staticmethod
+    def from_pydict(data: Dict[str, Sequence], schema: Dict[str, T.DataType], device="cpu") -> "Table":
+        return Table({k: Column.from_values(data[k], schema[k], device=device) for k in schema})
+
+    def __repr__(self):
+        return f"Table(rows={self.num_rows}, cols={self.names})"

@@ -1,0 +1,509 @@
+"""Plan executor.
+
+Executes resolved (and optimized) logical plans over whole-partition Chunks.
+Operator implementations here are the engine's reference semantics; on GPU
+the hot operators (filter/project fusion, hash join, hash aggregate) are
+routed through HIP kernels in ops/ with identical results.
+
+The reference executes pull-based DataFusion streams per partition
+(ref: crates/sail-execution/src/job_runner.rs:53 LocalJobRunner); here a
+partition is the whole per-device table and operators run bottom-up.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+
+from ..plan import spec as S
+from . import types as T
+from .aggregates import agg_eval, global_ids, group_ids
+from .chunk import Chunk
+from .column import Column, StringColumn, Table
+from .eval import Evaluator, Scalar, broadcast, cast_column, cast_value
+from .joins import equi_join
+
+
+class ExecError(Exception):
+    pass
+
+
+class ExecutionContext:
+    """Per-query execution context: session, device, subquery cache."""
+
+    def __init__(self, session, device="cpu"):
+        self.session = session
+        self.device = torch.device(device)
+        self._subquery_cache: Dict[int, Scalar] = {}
+
+    def execute_scalar_subquery(self, e: S.ScalarSubquery) -> Scalar:
+        if id(e) not in self._subquery_cache:
+            chunk = Executor(self).execute(e.plan)
+            if chunk.num_rows > 1:
+                raise ExecError("scalar subquery returned more than one row")
+            if chunk.num_rows == 0:
+                val = Scalar(None, e.dtype)
+            else:
+                col = chunk.columns[0]
+                v = col.to_pylist()[0]
+                val = Scalar(v, e.dtype)
+            self._subquery_cache[id(e)] = val
+        return self._subquery_cache[id(e)]
+
+
+class Executor:
+    def __init__(self, ctx: ExecutionContext):
+        self.ctx = ctx
+        self.ev = Evaluator(ctx)
+
+    def execute(self, plan: S.Plan) -> Chunk:
+        m = getattr(self, "_x_" + type(plan).__name__, None)
+        if m is None:
+            raise ExecError(f"cannot execute {type(plan).__name__}")
+        return m(plan)
+
+    # -- leaves ------------------------------------------------------------
+    def _x_Read(self, p: S.Read) -> Chunk:
+        t = self.ctx.session.catalog.get_table_data(p.table, self.ctx.device)
+        if t is None:
+            raise ExecError(f"no data for table {p.table}")
+        return Chunk.from_table(t)
+
+    def _x_DataSourceRead(self, p: S.DataSourceRead) -> Chunk:
+        from ..datasource.registry import read_source
+
+        t = read_source(p.format, p.paths, p.options, p.schema, self.ctx.device)
+        return Chunk.from_table(t)
+
+    def _x_LocalRelation(self, p: S.LocalRelation) -> Chunk:
+        cols = []
+        names = []
+        for (name, dtype) in p.schema:
+            cols.append(Column.from_values(p.data[name], dtype, device=self.ctx.device))
+            names.append(name)
+        return Chunk(cols, names)
+
+    def _x_Range(self, p: S.Range) -> Chunk:
+        data = torch.arange(p.start, p.end, p.step, dtype=torch.int64, device=self.ctx.device)
+        return Chunk([Column(T.I64, data)], ["id"])
+
+    def _x_SubqueryAlias(self, p: S.SubqueryAlias) -> Chunk:
+        c = self.execute(p.input)
+        return Chunk(c.columns, [n for n, _ in p.schema])
+
+    # -- row ops -----------------------------------------------------------
+    def _x_Filter(self, p: S.Filter) -> Chunk:
+        child = self.execute(p.input)
+        if child.num_rows == 0:
+            return child
+        mask = self.ev.eval_mask(p.condition, child)
+        return child.filter_mask(mask)
+
+    def _x_Project(self, p: S.Project) -> Chunk:
+        child = self.execute(p.input)
+        cols = []
+        for e in p.exprs:
+            v = self.ev.eval(e, child)
+            cols.append(broadcast(v, child.num_rows, child.device))
+        return Chunk(cols, [n for n, _ in p.schema])
+
+    def _x_Limit(self, p: S.Limit) -> Chunk:
+        child = self.execute(p.input)
+        start = p.offset
+        n = p.n if p.n is not None else child.num_rows - start
+        return child.slice(start, max(0, min(n, child.num_rows - start)))
+
+    def _x_Distinct(self, p: S.Distinct) -> Chunk:
+        child = self.execute(p.input)
+        if child.num_rows == 0:
+            return child
+        gid, rep, ng = group_ids(child.columns)
+        return child.gather(rep)
+
+    # -- sort --------------------------------------------------------------
+    def _x_Sort(self, p: S.Sort) -> Chunk:
+        child = self.execute(p.input)
+        if child.num_rows <= 1:
+            return child
+        idx = sort_indices(self.ev, p.keys, child)
+        return child.gather(idx)
+
+    # -- aggregate ---------------------------------------------------------
+    def _x_Aggregate(self, p: S.Aggregate) -> Chunk:
+        child = self.execute(p.input)
+        n = child.num_rows
+        dev = child.device
+        if p.group_by:
+            key_cols = [broadcast(self.ev.eval(g, child), n, dev) for g in p.group_by]
+            if n == 0:
+                return Chunk(key_cols + [_empty_agg_col(a, dev) for a in p.aggs],
+                             [nm for nm, _ in p.schema])
+            gid, rep, ng = group_ids(key_cols)
+            out_keys = [c.gather(rep) for c in key_cols]
+        else:
+            if n == 0:
+                # global aggregate over empty input still yields one row
+                cols = [_empty_global_agg(a, dev) for a in p.aggs]
+                return Chunk(cols, [nm for nm, _ in p.schema])
+            gid, ng = global_ids(n, dev)
+            out_keys = []
+        agg_cols = []
+        for a in p.aggs:
+            args = [broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
+            fmask = self.ev.eval_mask(a.filter, child) if a.filter is not None else None
+            agg_cols.append(agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype))
+        return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
+
+    # -- joins -------------------------------------------------------------
+    def _x_Join(self, p: S.Join) -> Chunk:
+        left = self.execute(p.left)
+        right = self.execute(p.right)
+        return join_chunks(self.ev, p, left, right)
+
+    # -- set ops -----------------------------------------------------------
+    def _x_SetOp(self, p: S.SetOp) -> Chunk:
+        left = self.execute(p.left)
+        right = self.execute(p.right)
+        # align right columns to left schema types
+        rcols = []
+        for i, (n_, t_) in enumerate(p.schema):
+            c = right.columns[i]
+            if c.dtype != t_:
+                c = cast_column(c, t_)
+            rcols.append(c)
+        lcols = []
+        for i, (n_, t_) in enumerate(p.schema):
+            c = left.columns[i]
+            if c.dtype != t_:
+                c = cast_column(c, t_)
+            lcols.append(c)
+        names = [n_ for n_, _ in p.schema]
+        if p.op == "union":
+            cols = [concat_columns([a, b]) for a, b in zip(lcols, rcols)]
+            out = Chunk(cols, names)
+            if not p.is_all:
+                gid, rep, ng = group_ids(out.columns)
+                out = out.gather(rep)
+            return out
+        lchunk = Chunk(lcols, names)
+        rchunk = Chunk(rcols, names)
+        _, _, counts = equi_join(rchunk.columns, lchunk.columns, "semi")
+        if p.op == "intersect":
+            mask = counts > 0
+        else:  # except
+            mask = counts == 0
+        out = lchunk.filter_mask(mask)
+        if not p.is_all:
+            gid, rep, ng = group_ids(out.columns)
+            out = out.gather(rep)
+        return out
+
+    # -- window ------------------------------------------------------------
+    def _x_WindowPlan(self, p: S.WindowPlan) -> Chunk:
+        from .window import eval_window
+
+        child = self.execute(p.input)
+        cols = list(child.columns)
+        for we in p.window_exprs:
+            e = we.child if isinstance(we, S.Alias) else we
+            cols.append(eval_window(self.ev, e, child))
+        return Chunk(cols, [n for n, _ in p.schema])
+
+    # -- commands ----------------------------------------------------------
+    def _x_CreateView(self, p: S.CreateView) -> Chunk:
+        self.ctx.session.catalog.create_view(p.name, p.input, replace=p.replace)
+        return Chunk([], [])
+
+    def _x_CreateTable(self, p: S.CreateTable) -> Chunk:
+        sess = self.ctx.session
+        if p.input is not None:
+            data = self.execute(p.input)
+            sess.catalog.register_table_chunk(p.name, data, [(n, t) for n, t in p.input.schema])
+        else:
+            sess.catalog.create_empty_table(p.name, p.columns)
+        return Chunk([], [])
+
+    def _x_DropTable(self, p: S.DropTable) -> Chunk:
+        self.ctx.session.catalog.drop(p.name, if_exists=p.if_exists)
+        return Chunk([], [])
+
+    def _x_InsertInto(self, p: S.InsertInto) -> Chunk:
+        data = self.execute(p.input)
+        self.ctx.session.catalog.insert_into(p.table, data, overwrite=p.overwrite)
+        return Chunk([], [])
+
+    def _x_Write(self, p: S.Write) -> Chunk:
+        from ..datasource.registry import write_source
+
+        data = self.execute(p.input)
+        write_source(p.format, p.path, data, p.mode, p.options, p.partition_by)
+        return Chunk([], [])
+
+    def _x_Explain(self, p: S.Explain) -> Chunk:
+        text = S.plan_tree_string(p.input)
+        return Chunk([StringColumn.from_pylist([text])], ["plan"])
+
+    def _x_SetConfig(self, p: S.SetConfig) -> Chunk:
+        if p.value is not None:
+            self.ctx.session.conf[p.key] = p.value
+        val = self.ctx.session.conf.get(p.key)
+        return Chunk([StringColumn.from_pylist([p.key]), StringColumn.from_pylist([val])],
+                     ["key", "value"])
+
+    def _x_ShowTables(self, p: S.ShowTables) -> Chunk:
+        names = self.ctx.session.catalog.list_tables()
+        return Chunk([StringColumn.from_pylist([""] * len(names), dict_encode=False),
+                      StringColumn.from_pylist(names, dict_encode=False),
+                      Column.from_values([True] * len(names), T.BOOL)],
+                     ["namespace", "tableName", "isTemporary"])
+
+    def _x_DescribeTable(self, p: S.DescribeTable) -> Chunk:
+        schema = self.ctx.session.catalog.table_schema(p.name)
+        if schema is None:
+            raise ExecError(f"table not found: {p.name}")
+        return Chunk([StringColumn.from_pylist([n for n, _ in schema], dict_encode=False),
+                      StringColumn.from_pylist([repr(t) for _, t in schema], dict_encode=False),
+                      StringColumn.from_pylist([""] * len(schema), dict_encode=False)],
+                     ["col_name", "data_type", "comment"])
+
+
+# ---------------------------------------------------------------------------
+# helpers
+# ---------------------------------------------------------------------------
+
+def sort_indices(ev: Evaluator, keys: List[S.SortKey], chunk: Chunk) -> torch.Tensor:
+    """Stable multi-key argsort: iterate keys last-to-first with stable sorts.
+    Null ordering per Spark: NULLS FIRST for ASC, NULLS LAST for DESC
+    unless overridden."""
+    n = chunk.num_rows
+    dev = chunk.device
+    idx = torch.arange(n, device=dev)
+    for k in reversed(keys):
+        col = broadcast(ev.eval(k.child, chunk), n, dev)
+        keyvals = _sortable(col)
+        keyvals = keyvals[idx]
+        nulls_first = k.nulls_first if k.nulls_first is not None else k.ascending
+        if col.validity is not None:
+            vm = col.valid_mask()[idx]
+            big = _null_sentinel(keyvals, nulls_first == k.ascending)
+            keyvals = torch.where(vm, keyvals, big)
+        order = torch.argsort(keyvals, stable=True, descending=not k.ascending)
+        idx = idx[order]
+    return idx
+
+
+def _sortable(col: Column) -> torch.Tensor:
+    if isinstance(col, StringColumn):
+        from .joins import normalize_key
+
+        if col.is_dict:
+            return col.codes.to(torch.int64)  # dict is sorted => codes ordered
+        # raw strings: host rank
+        vals = col.to_pylist()
+        order = sorted(range(len(vals)), key=lambda i: (vals[i] is None, vals[i] or ""))
+        rank = [0] * len(vals)
+        for r, i in enumerate(order):
+            rank[i] = r
+        return torch.tensor(rank, dtype=torch.int64, device=col.device)
+    d = col.data
+    if d.dtype == torch.bool:
+        return d.to(torch.int8)
+    return d
+
+
+def _null_sentinel(vals: torch.Tensor, lo: bool):
+    if vals.dtype.is_floating_point:
+        v = float("-inf") if lo else float("inf")
+    else:
+        info = torch.iinfo(vals.dtype)
+        v = info.min if lo else info.max
+    return torch.full_like(vals, v)
+
+
+def concat_columns(cols: List[Column]) -> Column:
+    c0 = cols[0]
+    if isinstance(c0, StringColumn):
+        vals = []
+        for c in cols:
+            vals.extend(c.to_pylist())
+        return StringColumn.from_pylist(vals, device=c0.device)
+    data = torch.cat([c.data for c in cols])
+    if any(c.validity is not None for c in cols):
+        validity = torch.cat([c.valid_mask() for c in cols]).to(torch.uint8)
+    else:
+        validity = None
+    return Column(c0.dtype, data, validity)
+
+
+def join_chunks(ev: Evaluator, p: S.Join, left: Chunk, right: Chunk) -> Chunk:
+    """Execute a join given both sides. Equi-conditions go through the
+    hash/sort join; residual non-equi predicates are applied to the matched
+    pairs (and fixed up for outer joins)."""
+    how = p.how
+    names = [n for n, _ in p.schema]
+    if how == "cross" or p.on is None:
+        if how not in ("cross", "inner"):
+            raise ExecError(f"{how} join requires a condition")
+        nl, nr = left.num_rows, right.num_rows
+        li = torch.arange(nl, device=left.device).repeat_interleave(nr)
+        ri = torch.arange(nr, device=left.device).repeat(nl)
+        lcols = [c.gather(li) for c in left.columns]
+        rcols = [c.gather(ri) for c in right.columns]
+        return Chunk(lcols + rcols, names)
+
+    equi, residual = split_join_condition(p.on, len(left.columns))
+
+    if not equi:
+        return _nested_loop_join(ev, p, left, right, names)
+
+    lkeys = [left.columns[i] for i, _ in equi]
+    rkeys = [right.columns[j - len(left.columns)] for _, j in equi]
+
+    # orient: build on right, probe on left
+    probe_idx, build_idx, counts = equi_join(rkeys, lkeys, how)
+
+    if residual is not None and probe_idx.shape[0] > 0:
+        pairs = Chunk([c.gather(probe_idx) for c in left.columns]
+                      + [c.gather(build_idx) for c in right.columns],
+                      [f"c{i}" for i in range(len(left.columns) + len(right.columns))])
+        rmask = ev.eval_mask(residual, pairs)
+        probe_idx = probe_idx[rmask]
+        build_idx = build_idx[rmask]
+        if how in ("semi", "anti", "left", "full"):
+            counts = torch.zeros(left.num_rows, dtype=torch.int64, device=left.device)
+            counts.index_add_(0, probe_idx, torch.ones(probe_idx.shape[0], dtype=torch.int64, device=left.device))
+
+    dev = left.device
+    if how == "inner":
+        return Chunk([c.gather(probe_idx) for c in left.columns]
+                     + [c.gather(build_idx) for c in right.columns], names)
+    if how == "semi":
+        mask = counts > 0
+        return left.filter_mask(mask)
+    if how == "anti":
+        mask = counts == 0
+        return left.filter_mask(mask)
+    if how in ("left", "full", "right"):
+        unmatched_l = torch.nonzero(counts == 0, as_tuple=False).squeeze(1)
+        li = torch.cat([probe_idx, unmatched_l])
+        ri = torch.cat([build_idx, torch.full((unmatched_l.shape[0],), -1, dtype=torch.int64, device=dev)])
+        if how == "right" or how == "full":
+            matched_r = torch.zeros(right.num_rows, dtype=torch.bool, device=dev)
+            if build_idx.shape[0]:
+                matched_r[build_idx] = True
+            unmatched_r = torch.nonzero(~matched_r, as_tuple=False).squeeze(1)
+            if how == "right":
+                li = torch.cat([probe_idx, torch.full((unmatched_r.shape[0],), -1, dtype=torch.int64, device=dev)])
+                ri = torch.cat([build_idx, unmatched_r])
+            else:
+                li = torch.cat([li, torch.full((unmatched_r.shape[0],), -1, dtype=torch.int64, device=dev)])
+                ri = torch.cat([ri, unmatched_r])
+        lcols = [_gather_nullable(c, li) for c in left.columns]
+        rcols = [_gather_nullable(c, ri) for c in right.columns]
+        return Chunk(lcols + rcols, names)
+    if how in ("rightsemi", "rightanti"):
+        matched_r = torch.zeros(right.num_rows, dtype=torch.bool, device=dev)
+        if build_idx.shape[0]:
+            matched_r[build_idx] = True
+        mask = matched_r if how == "rightsemi" else ~matched_r
+        return right.filter_mask(mask)
+    raise ExecError(f"join type {how}")
+
+
+def _gather_nullable(c: Column, idx: torch.Tensor) -> Column:
+    """Gather where idx == -1 produces NULL."""
+    null = idx < 0
+    safe = torch.where(null, torch.zeros_like(idx), idx)
+    out = c.gather(safe)
+    if bool(null.any()):
+        v = out.valid_mask() & ~null
+        if isinstance(out, StringColumn):
+            out.validity = v.to(torch.uint8)
+            return out
+        return Column(out.dtype, out.data, v.to(torch.uint8))
+    return out
+
+
+def _nested_loop_join(ev: Evaluator, p: S.Join, left: Chunk, right: Chunk, names):
+    nl, nr = left.num_rows, right.num_rows
+    dev = left.device
+    li = torch.arange(nl, device=dev).repeat_interleave(nr)
+    ri = torch.arange(nr, device=dev).repeat(nl)
+    pairs = Chunk([c.gather(li) for c in left.columns] + [c.gather(ri) for c in right.columns],
+                  [f"c{i}" for i in range(len(left.columns) + len(right.columns))])
+    mask = ev.eval_mask(p.on, pairs)
+    how = p.how
+    if how == "inner":
+        return Chunk([c for c in pairs.filter_mask(mask).columns], names)
+    counts = torch.zeros(nl, dtype=torch.int64, device=dev)
+    counts.index_add_(0, li[mask], torch.ones(int(mask.sum()), dtype=torch.int64, device=dev))
+    if how == "semi":
+        return left.filter_mask(counts > 0)
+    if how == "anti":
+        return left.filter_mask(counts == 0)
+    if how == "left":
+        keep_li = li[mask]
+        keep_ri = ri[mask]
+        unmatched = torch.nonzero(counts == 0, as_tuple=False).squeeze(1)
+        lidx = torch.cat([keep_li, unmatched])
+        ridx = torch.cat([keep_ri, torch.full((unmatched.shape[0],), -1, dtype=torch.int64, device=dev)])
+        lcols = [c.gather(lidx) for c in left.columns]
+        rcols = [_gather_nullable(c, ridx) for c in right.columns]
+        return Chunk(lcols + rcols, names)
+    raise ExecError(f"nested loop join type {how} TODO")
+
+
+def split_join_condition(on: S.Expr, n_left: int):
+    """Split an ON condition into equi-key pairs [(left_idx, right_idx)] and a
+    residual expression (or None). A conjunct qualifies as an equi-key when it
+    is BoundRef == BoundRef with sides on opposite inputs."""
+    conjuncts = []
+
+    def flatten(e):
+        if isinstance(e, S.BinaryOp) and e.op == "and":
+            flatten(e.left)
+            flatten(e.right)
+        else:
+            conjuncts.append(e)
+
+    flatten(on)
+    equi = []
+    residual = []
+    for c in conjuncts:
+        if isinstance(c, S.BinaryOp) and c.op == "=":
+            l, r = c.left, c.right
+            # strip widening int casts (int32->int64): normalize_key makes the
+            # raw columns join-compatible; decimal rescales must stay.
+            l = l.child if (isinstance(l, S.Cast) and isinstance(l.child, S.BoundRef)
+                            and l.child.dtype is not None and l.child.dtype.is_integer
+                            and l.dtype is not None and l.dtype.is_integer) else l
+            r = r.child if (isinstance(r, S.Cast) and isinstance(r.child, S.BoundRef)
+                            and r.child.dtype is not None and r.child.dtype.is_integer
+                            and r.dtype is not None and r.dtype.is_integer) else r
+            if isinstance(l, S.BoundRef) and isinstance(r, S.BoundRef):
+                if l.index < n_left <= r.index:
+                    equi.append((l.index, r.index))
+                    continue
+                if r.index < n_left <= l.index:
+                    equi.append((r.index, l.index))
+                    continue
+        residual.append(c)
+    res = None
+    for c in residual:
+        res = c if res is None else S.BinaryOp("and", res, c, T.BOOL)
+    return equi, res
+
+
+def _empty_agg_col(a: S.AggFunc, dev) -> Column:
+    if isinstance(a.dtype, T.StringType):
+        return StringColumn.from_pylist([], device=dev)
+    return Column(a.dtype, torch.zeros(0, dtype=a.dtype.storage or torch.int64, device=dev))
+
+
+def _empty_global_agg(a: S.AggFunc, dev) -> Column:
+    if a.name in ("count", "count_if"):
+        return Column(T.I64, torch.zeros(1, dtype=torch.int64, device=dev))
+    if isinstance(a.dtype, T.StringType):
+        return StringColumn.from_pylist([None], device=dev)
+    return Column(a.dtype, torch.zeros(1, dtype=a.dtype.storage or torch.int64, device=dev),
+                  torch.zeros(1, dtype=torch.uint8, device=dev))

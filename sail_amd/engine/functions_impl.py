@@ -1,0 +1,655 @@
+"""Scalar function implementations (torch path).
+
+The dispatch table of per-function kernels used by the Evaluator; Spark
+semantics (ref: crates/sail-function/src/scalar/* for behavior parity).
+Functions receive `Val`s (Column or Scalar) and the enclosing chunk.
+"""
+from __future__ import annotations
+
+import datetime as _dt
+import math
+from typing import List
+
+import torch
+
+from . import types as T
+from .chunk import Chunk
+from .column import Column, StringColumn
+
+_EPOCH = _dt.date(1970, 1, 1)
+
+
+def dispatch_function(name: str, args: List, out_type, chunk: Chunk, ev):
+    fn = _IMPLS.get(name)
+    if fn is None:
+        raise NotImplementedError(f"function {name} has no implementation yet")
+    return fn(args, out_type, chunk, ev)
+
+
+def _col(v, chunk: Chunk) -> Column:
+    from .eval import broadcast
+
+    return broadcast(v, chunk.num_rows, chunk.device)
+
+
+def _scalarize(v):
+    from .eval import Scalar
+
+    return v if isinstance(v, Scalar) else None
+
+
+# ---------------------------------------------------------------------------
+# datetime — civil-calendar math on int32 day counts, vectorized.
+# Uses the standard days-from-civil algorithm (Howard Hinnant's) so that
+# year/month extraction runs as pure tensor arithmetic on device.
+# ---------------------------------------------------------------------------
+
+def _civil_from_days(z: torch.Tensor):
+    z = z.to(torch.int64) + 719468
+    era = torch.div(torch.where(z >= 0, z, z - 146096), 146097, rounding_mode="floor")
+    doe = z - era * 146097
+    yoe = torch.div(doe - torch.div(doe, 1460, rounding_mode="floor")
+                    + torch.div(doe, 36524, rounding_mode="floor")
+                    - torch.div(doe, 146096, rounding_mode="floor"), 365, rounding_mode="floor")
+    y = yoe + era * 400
+    doy = doe - (365 * yoe + torch.div(yoe, 4, rounding_mode="floor") - torch.div(yoe, 100, rounding_mode="floor"))
+    mp = torch.div(5 * doy + 2, 153, rounding_mode="floor")
+    d = doy - torch.div(153 * mp + 2, 5, rounding_mode="floor") + 1
+    m = mp + torch.where(mp < 10, torch.full_like(mp, 3), torch.full_like(mp, -9))
+    y = y + (m <= 2).to(torch.int64)
+    return y, m, d
+
+
+def _days_from_civil(y: torch.Tensor, m: torch.Tensor, d: torch.Tensor) -> torch.Tensor:
+    y = y - (m <= 2).to(torch.int64)
+    era = torch.div(torch.where(y >= 0, y, y - 399), 400, rounding_mode="floor")
+    yoe = y - era * 400
+    mp = torch.where(m > 2, m - 3, m + 9)
+    doy = torch.div(153 * mp + 2, 5, rounding_mode="floor") + d - 1
+    doe = yoe * 365 + torch.div(yoe, 4, rounding_mode="floor") - torch.div(yoe, 100, rounding_mode="floor") + doy
+    return era * 146097 + doe - 719468
+
+
+def _as_days(c: Column) -> torch.Tensor:
+    if isinstance(c.dtype, T.TimestampType):
+        return torch.div(c.data, 86_400_000_000, rounding_mode="floor")
+    return c.data.to(torch.int64)
+
+
+def _f_year(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    y, m, d = _civil_from_days(_as_days(c))
+    return Column(T.I32, y.to(torch.int32), c.validity)
+
+
+def _f_month(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    y, m, d = _civil_from_days(_as_days(c))
+    return Column(T.I32, m.to(torch.int32), c.validity)
+
+
+def _f_day(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    y, m, d = _civil_from_days(_as_days(c))
+    return Column(T.I32, d.to(torch.int32), c.validity)
+
+
+def _f_quarter(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    y, m, d = _civil_from_days(_as_days(c))
+    return Column(T.I32, (torch.div(m - 1, 3, rounding_mode="floor") + 1).to(torch.int32), c.validity)
+
+
+def _f_dayofweek(args, out, chunk, ev):
+    # Spark: 1 = Sunday ... 7 = Saturday
+    c = _col(args[0], chunk)
+    days = _as_days(c)
+    dow = torch.remainder(days + 4, 7) + 1  # 1970-01-01 was Thursday
+    return Column(T.I32, dow.to(torch.int32), c.validity)
+
+
+def _f_weekday(args, out, chunk, ev):
+    # Spark weekday: 0 = Monday ... 6 = Sunday
+    c = _col(args[0], chunk)
+    days = _as_days(c)
+    wd = torch.remainder(days + 3, 7)
+    return Column(T.I32, wd.to(torch.int32), c.validity)
+
+
+def _f_dayofyear(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    days = _as_days(c)
+    y, m, d = _civil_from_days(days)
+    jan1 = _days_from_civil(y, torch.ones_like(y), torch.ones_like(y))
+    return Column(T.I32, (days - jan1 + 1).to(torch.int32), c.validity)
+
+
+def _f_date_add(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    n = _col(args[1], chunk)
+    return Column(T.DATE, (_as_days(c) + n.data.to(torch.int64)).to(torch.int32),
+                  _merge(c, n))
+
+
+def _f_date_sub(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    n = _col(args[1], chunk)
+    return Column(T.DATE, (_as_days(c) - n.data.to(torch.int64)).to(torch.int32),
+                  _merge(c, n))
+
+
+def _f_datediff(args, out, chunk, ev):
+    a = _col(args[0], chunk)
+    b = _col(args[1], chunk)
+    return Column(T.I32, (_as_days(a) - _as_days(b)).to(torch.int32), _merge(a, b))
+
+
+def _f_add_months(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    n = _col(args[1], chunk)
+    y, m, d = _civil_from_days(_as_days(c))
+    total = y * 12 + (m - 1) + n.data.to(torch.int64)
+    ny = torch.div(total, 12, rounding_mode="floor")
+    nm = total - ny * 12 + 1
+    # clamp day to end of month
+    dim = _days_in_month(ny, nm)
+    nd = torch.minimum(d, dim)
+    return Column(T.DATE, _days_from_civil(ny, nm, nd).to(torch.int32), _merge(c, n))
+
+
+def _days_in_month(y: torch.Tensor, m: torch.Tensor) -> torch.Tensor:
+    base = torch.tensor([31, 28, 31, 30, 31, 30, 31, 31, 30, 31, 30, 31],
+                        dtype=torch.int64, device=y.device)
+    dim = base[(m - 1).clamp(0, 11)]
+    leap = ((y % 4 == 0) & ((y % 100 != 0) | (y % 400 == 0))) & (m == 2)
+    return dim + leap.to(torch.int64)
+
+
+def _f_last_day(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    y, m, d = _civil_from_days(_as_days(c))
+    dim = _days_in_month(y, m)
+    return Column(T.DATE, _days_from_civil(y, m, dim).to(torch.int32), c.validity)
+
+
+def _f_trunc(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    fmt = _scalarize(args[1])
+    unit = (fmt.value if fmt else "month").lower()
+    y, m, d = _civil_from_days(_as_days(c))
+    one = torch.ones_like(y)
+    if unit in ("year", "yyyy", "yy"):
+        days = _days_from_civil(y, one, one)
+    elif unit in ("quarter",):
+        qm = (torch.div(m - 1, 3, rounding_mode="floor")) * 3 + 1
+        days = _days_from_civil(y, qm, one)
+    elif unit in ("month", "mm", "mon"):
+        days = _days_from_civil(y, m, one)
+    elif unit in ("week",):
+        wd = torch.remainder(_as_days(c) + 3, 7)
+        days = _as_days(c) - wd
+    else:
+        days = _as_days(c)
+    return Column(T.DATE, days.to(torch.int32), c.validity)
+
+
+def _f_date_trunc(args, out, chunk, ev):
+    # date_trunc(fmt, timestamp) -> timestamp
+    fmt = _scalarize(args[0])
+    c = _col(args[1], chunk)
+    unit = (fmt.value if fmt else "day").lower()
+    us = c.data.to(torch.int64)
+    day_us = 86_400_000_000
+    if unit in ("year", "yyyy", "yy", "quarter", "month", "mm", "mon", "week"):
+        days = torch.div(us, day_us, rounding_mode="floor")
+        dcol = Column(T.DATE, days.to(torch.int32), c.validity)
+        tr = _f_trunc([dcol, args[0]], T.DATE, chunk, ev)
+        return Column(T.TIMESTAMP, tr.data.to(torch.int64) * day_us, c.validity)
+    step = {"day": day_us, "dd": day_us, "hour": 3_600_000_000,
+            "minute": 60_000_000, "second": 1_000_000}.get(unit, day_us)
+    return Column(T.TIMESTAMP, torch.div(us, step, rounding_mode="floor") * step, c.validity)
+
+
+def _f_to_date(args, out, chunk, ev):
+    from .eval import cast_value
+
+    return cast_value(args[0], T.DATE, chunk)
+
+
+def _f_make_date(args, out, chunk, ev):
+    y = _col(args[0], chunk).data.to(torch.int64)
+    m = _col(args[1], chunk).data.to(torch.int64)
+    d = _col(args[2], chunk).data.to(torch.int64)
+    return Column(T.DATE, _days_from_civil(y, m, d).to(torch.int32), None)
+
+
+def _f_hour(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    us = torch.remainder(c.data.to(torch.int64), 86_400_000_000)
+    return Column(T.I32, torch.div(us, 3_600_000_000, rounding_mode="floor").to(torch.int32), c.validity)
+
+
+def _f_minute(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    us = torch.remainder(c.data.to(torch.int64), 3_600_000_000)
+    return Column(T.I32, torch.div(us, 60_000_000, rounding_mode="floor").to(torch.int32), c.validity)
+
+
+def _f_second(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    us = torch.remainder(c.data.to(torch.int64), 60_000_000)
+    return Column(T.I32, torch.div(us, 1_000_000, rounding_mode="floor").to(torch.int32), c.validity)
+
+
+# ---------------------------------------------------------------------------
+# math
+# ---------------------------------------------------------------------------
+
+def _merge(a: Column, b: Column):
+    from .eval import _merge_validity
+
+    return _merge_validity(a, b)
+
+
+def _f_abs(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    return Column(c.dtype, torch.abs(c.data), c.validity)
+
+
+def _unary_float(fn):
+    def impl(args, out, chunk, ev):
+        c = _col(args[0], chunk)
+        x = c.data.to(torch.float64)
+        if isinstance(c.dtype, T.DecimalType):
+            x = x / (10.0 ** c.dtype.scale)
+        return Column(T.F64, fn(x), c.validity)
+
+    return impl
+
+
+def _f_round(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    nd = _scalarize(args[1]).value if len(args) > 1 else 0
+    if isinstance(c.dtype, T.DecimalType):
+        from .eval import _rescale_int
+
+        s = c.dtype.scale
+        if nd >= s:
+            return c
+        data = _rescale_int(c.data, s, nd)
+        return Column(T.DecimalType(c.dtype.precision, nd), data, c.validity)
+    if c.dtype.is_integer:
+        return c
+    f = 10.0 ** nd
+    x = c.data.to(torch.float64) * f
+    # HALF_UP (away from zero), like Spark round
+    data = torch.where(x >= 0, torch.floor(x + 0.5), torch.ceil(x - 0.5)) / f
+    return Column(T.F64, data, c.validity)
+
+
+def _f_floor(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    if isinstance(c.dtype, T.DecimalType):
+        s = 10 ** c.dtype.scale
+        data = torch.div(torch.where(c.data >= 0, c.data, c.data - (s - 1)), s, rounding_mode="trunc")
+        return Column(out, data, c.validity)
+    if c.dtype.is_integer:
+        return Column(T.I64, c.data.to(torch.int64), c.validity)
+    return Column(T.I64, torch.floor(c.data.to(torch.float64)).to(torch.int64), c.validity)
+
+
+def _f_ceil(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    if isinstance(c.dtype, T.DecimalType):
+        s = 10 ** c.dtype.scale
+        data = torch.div(torch.where(c.data >= 0, c.data + (s - 1), c.data), s, rounding_mode="trunc")
+        return Column(out, data, c.validity)
+    if c.dtype.is_integer:
+        return Column(T.I64, c.data.to(torch.int64), c.validity)
+    return Column(T.I64, torch.ceil(c.data.to(torch.float64)).to(torch.int64), c.validity)
+
+
+def _f_power(args, out, chunk, ev):
+    a = _col(args[0], chunk)
+    b = _col(args[1], chunk)
+    return Column(T.F64, torch.pow(a.data.to(torch.float64), b.data.to(torch.float64)), _merge(a, b))
+
+
+def _f_greatest(args, out, chunk, ev):
+    cols = [_col(a, chunk) for a in args]
+    data = cols[0].data.clone()
+    for c in cols[1:]:
+        data = torch.maximum(data, c.data.to(data.dtype))
+    return Column(out, data, None)
+
+
+def _f_least(args, out, chunk, ev):
+    cols = [_col(a, chunk) for a in args]
+    data = cols[0].data.clone()
+    for c in cols[1:]:
+        data = torch.minimum(data, c.data.to(data.dtype))
+    return Column(out, data, None)
+
+
+def _f_sign(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    return Column(T.F64, torch.sign(c.data.to(torch.float64)), c.validity)
+
+
+def _f_isnan(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    if c.dtype.is_float:
+        return Column(T.BOOL, torch.isnan(c.data), None)
+    return Column(T.BOOL, torch.zeros(len(c), dtype=torch.bool, device=c.device), None)
+
+
+# ---------------------------------------------------------------------------
+# conditional
+# ---------------------------------------------------------------------------
+
+def _f_coalesce(args, out, chunk, ev):
+    from .eval import Scalar, cast_value
+
+    cols = []
+    for a in args:
+        if isinstance(a, Scalar) and a.is_null:
+            continue
+        cols.append(cast_value(a, out, chunk))
+    if not cols:
+        return Scalar(None, out)
+    first = _col(cols[0], chunk)
+    data = first.data.clone()
+    valid = first.valid_mask().clone()
+    for a in cols[1:]:
+        c = _col(a, chunk)
+        need = ~valid
+        data = torch.where(need, c.data.to(data.dtype), data)
+        valid = valid | (need & c.valid_mask())
+    return Column(out, data, None if bool(valid.all()) else valid.to(torch.uint8))
+
+
+def _f_if(args, out, chunk, ev):
+    from .eval import Scalar, cast_value
+
+    cond = _col(args[0], chunk)
+    a = _col(cast_value(args[1], out, chunk), chunk)
+    b = _col(cast_value(args[2], out, chunk), chunk)
+    cmask = cond.data.to(torch.bool) & cond.valid_mask()
+    data = torch.where(cmask, a.data, b.data.to(a.data.dtype))
+    valid = torch.where(cmask, a.valid_mask(), b.valid_mask())
+    return Column(out, data, None if bool(valid.all()) else valid.to(torch.uint8))
+
+
+def _f_nullif(args, out, chunk, ev):
+    a = _col(args[0], chunk)
+    b = _col(args[1], chunk)
+    eq = a.data == b.data.to(a.data.dtype)
+    valid = a.valid_mask() & ~eq
+    return Column(out, a.data, valid.to(torch.uint8))
+
+
+# ---------------------------------------------------------------------------
+# strings — CPU reference implementations round-trip through host python;
+# the GPU path uses dict-encoding (O(|dict|) host work) or HIP string kernels.
+# ---------------------------------------------------------------------------
+
+def _str_map(c: StringColumn, fn, out_is_string=True):
+    vals = c.to_pylist()
+    res = [None if v is None else fn(v) for v in vals]
+    if out_is_string:
+        return StringColumn.from_pylist(res, device=c.device)
+    return res
+
+
+def _f_upper(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    if isinstance(c, StringColumn) and c.is_dict:
+        vals = [v.upper() for v in c.dict_values()]
+        from .column import _pack_strings
+
+        offs, byts = _pack_strings(vals, c.device)
+        return StringColumn(offs, byts, c.validity, c.codes)
+    return _str_map(c, str.upper)
+
+
+def _f_lower(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    if isinstance(c, StringColumn) and c.is_dict:
+        vals = [v.lower() for v in c.dict_values()]
+        from .column import _pack_strings
+
+        offs, byts = _pack_strings(vals, c.device)
+        return StringColumn(offs, byts, c.validity, c.codes)
+    return _str_map(c, str.lower)
+
+
+def _f_length(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    if not isinstance(c, StringColumn):
+        raise NotImplementedError("length on non-string")
+    if c.is_dict:
+        lens = torch.tensor([len(v) for v in c.dict_values()], dtype=torch.int32, device=c.device)
+        return Column(T.I32, lens[c.codes.long().clamp_min(0)], c.validity)
+    lens = (c.offsets[1:] - c.offsets[:-1]).to(torch.int32)
+    return Column(T.I32, lens, c.validity)
+
+
+def _f_substring(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    start = _scalarize(args[1])
+    length = _scalarize(args[2]) if len(args) > 2 else None
+    if start is None:
+        raise NotImplementedError("substring with non-literal start")
+    s = start.value
+    ln = length.value if length is not None else None
+
+    def sub(v: str) -> str:
+        if s > 0:
+            i = s - 1
+        elif s == 0:
+            i = 0
+        else:
+            i = max(len(v) + s, 0)
+        return v[i : i + ln] if ln is not None else v[i:]
+
+    if isinstance(c, StringColumn) and c.is_dict:
+        vals = [sub(v) for v in c.dict_values()]
+        from .column import _pack_strings
+
+        offs, byts = _pack_strings(vals, c.device)
+        return StringColumn(offs, byts, c.validity, c.codes)
+    return _str_map(c, sub)
+
+
+def _f_concat(args, out, chunk, ev):
+    from .eval import Scalar
+
+    n = chunk.num_rows
+    parts = []
+    for a in args:
+        if isinstance(a, Scalar):
+            parts.append([a.value] * n)
+        else:
+            parts.append(_col(a, chunk).to_pylist())
+    res = []
+    for i in range(n):
+        vs = [p[i] for p in parts]
+        res.append(None if any(v is None for v in vs) else "".join(str(v) for v in vs))
+    return StringColumn.from_pylist(res, device=chunk.device)
+
+
+def _str_pred(fn):
+    def impl(args, out, chunk, ev):
+        c = _col(args[0], chunk)
+        pat = _scalarize(args[1])
+        if pat is None:
+            raise NotImplementedError("string predicate with column pattern")
+        p = pat.value
+        if isinstance(c, StringColumn) and c.is_dict:
+            hit = torch.tensor([fn(v, p) for v in c.dict_values()], dtype=torch.bool, device=c.device)
+            return Column(T.BOOL, hit[c.codes.long().clamp_min(0)] & (c.codes >= 0), c.validity)
+        if c.is_cuda:
+            from ..ops import kernels as K
+
+            m = K.string_predicate(c, fn.__name__, p)
+            if m is not None:
+                return Column(T.BOOL, m, c.validity)
+        vals = c.to_pylist()
+        return Column(T.BOOL, torch.tensor([fn(v, p) if v is not None else False for v in vals],
+                                           dtype=torch.bool, device=c.device), c.validity)
+
+    return impl
+
+
+def startswith(v, p):
+    return v.startswith(p)
+
+
+def endswith(v, p):
+    return v.endswith(p)
+
+
+def contains(v, p):
+    return p in v
+
+
+def _f_trim(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    return _str_map(c, str.strip)
+
+
+def _f_replace(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    old = _scalarize(args[1]).value
+    new = _scalarize(args[2]).value if len(args) > 2 else ""
+    return _str_map(c, lambda v: v.replace(old, new))
+
+
+def _f_split_part(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    delim = _scalarize(args[1]).value
+    part = _scalarize(args[2]).value
+
+    def sp(v):
+        parts = v.split(delim)
+        i = part - 1 if part > 0 else len(parts) + part
+        return parts[i] if 0 <= i < len(parts) else ""
+
+    return _str_map(c, sp)
+
+
+def _f_instr(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    sub = _scalarize(args[1]).value
+    res = _str_map(c, lambda v: v.find(sub) + 1, out_is_string=False)
+    return Column.from_values(res, T.I32, device=chunk.device)
+
+
+# ---------------------------------------------------------------------------
+# hashing / misc
+# ---------------------------------------------------------------------------
+
+def _f_xxhash64(args, out, chunk, ev):
+    # engine-internal hash (shuffle partitioning); not Spark-bit-exact yet
+    cols = [_col(a, chunk) for a in args]
+    acc = torch.zeros(chunk.num_rows, dtype=torch.int64, device=chunk.device)
+    for c in cols:
+        if isinstance(c, StringColumn):
+            data = c.codes.to(torch.int64) if c.is_dict else _cheap_string_hash(c)
+        else:
+            data = c.data.to(torch.int64) if c.data.dtype != torch.float64 else c.data.view(torch.int64)
+        acc = acc * 31 + data
+        mixed = acc ^ (acc >> 33)
+        mixed = mixed * -49064778989728563  # 0xFF51AFD7ED558CCD as signed
+        acc = mixed ^ (mixed >> 33)
+    return Column(T.I64, acc, None)
+
+
+def _cheap_string_hash(c: StringColumn) -> torch.Tensor:
+    vals = c.to_pylist()
+    return torch.tensor([hash(v) if v is not None else 0 for v in vals],
+                        dtype=torch.int64, device=c.device)
+
+
+def _f_monotonic_id(args, out, chunk, ev):
+    return Column(T.I64, torch.arange(chunk.num_rows, dtype=torch.int64, device=chunk.device), None)
+
+
+def _f_current_date(args, out, chunk, ev):
+    from .eval import Scalar
+
+    return Scalar((_dt.date.today() - _EPOCH).days, T.DATE)
+
+
+def _f_current_timestamp(args, out, chunk, ev):
+    from .eval import Scalar
+    import time
+
+    return Scalar(int(time.time() * 1e6), T.TIMESTAMP)
+
+
+def _f_typeof(args, out, chunk, ev):
+    from .eval import Scalar
+
+    a = args[0]
+    return Scalar(repr(a.dtype), T.STRING)
+
+
+def _cast_fn(target):
+    def impl(args, out, chunk, ev):
+        from .eval import cast_value
+
+        return cast_value(args[0], target, chunk)
+
+    return impl
+
+
+_IMPLS = {
+    # datetime
+    "year": _f_year, "month": _f_month, "day": _f_day, "dayofmonth": _f_day,
+    "quarter": _f_quarter, "dayofweek": _f_dayofweek, "weekday": _f_weekday,
+    "dayofyear": _f_dayofyear, "hour": _f_hour, "minute": _f_minute, "second": _f_second,
+    "date_add": _f_date_add, "dateadd": _f_date_add, "date_sub": _f_date_sub,
+    "datediff": _f_datediff, "date_diff": _f_datediff, "add_months": _f_add_months,
+    "last_day": _f_last_day, "trunc": _f_trunc, "date_trunc": _f_date_trunc,
+    "to_date": _f_to_date, "make_date": _f_make_date,
+    "current_date": _f_current_date, "curdate": _f_current_date,
+    "current_timestamp": _f_current_timestamp, "now": _f_current_timestamp,
+    # math
+    "abs": _f_abs, "round": _f_round, "floor": _f_floor, "ceil": _f_ceil,
+    "ceiling": _f_ceil, "power": _f_power, "pow": _f_power,
+    "greatest": _f_greatest, "least": _f_least, "sign": _f_sign, "signum": _f_sign,
+    "isnan": _f_isnan,
+    "sqrt": _unary_float(torch.sqrt), "exp": _unary_float(torch.exp),
+    "ln": _unary_float(torch.log), "log": _unary_float(torch.log),
+    "log10": _unary_float(torch.log10), "log2": _unary_float(torch.log2),
+    "log1p": _unary_float(torch.log1p), "expm1": _unary_float(torch.expm1),
+    "sin": _unary_float(torch.sin), "cos": _unary_float(torch.cos),
+    "tan": _unary_float(torch.tan), "asin": _unary_float(torch.asin),
+    "acos": _unary_float(torch.acos), "atan": _unary_float(torch.atan),
+    "sinh": _unary_float(torch.sinh), "cosh": _unary_float(torch.cosh),
+    "tanh": _unary_float(torch.tanh), "cbrt": _unary_float(lambda x: torch.sign(x) * torch.abs(x) ** (1 / 3)),
+    "degrees": _unary_float(torch.rad2deg), "radians": _unary_float(torch.deg2rad),
+    "rint": _unary_float(torch.round),
+    # conditional
+    "coalesce": _f_coalesce, "nvl": _f_coalesce, "ifnull": _f_coalesce,
+    "if": _f_if, "iff": _f_if, "nullif": _f_nullif,
+    # strings
+    "upper": _f_upper, "ucase": _f_upper, "lower": _f_lower, "lcase": _f_lower,
+    "length": _f_length, "len": _f_length, "char_length": _f_length,
+    "character_length": _f_length, "octet_length": _f_length,
+    "substring": _f_substring, "substr": _f_substring, "concat": _f_concat,
+    "startswith": _str_pred(startswith), "endswith": _str_pred(endswith),
+    "contains": _str_pred(contains), "trim": _f_trim,
+    "replace": _f_replace, "split_part": _f_split_part, "instr": _f_instr,
+    "locate": _f_instr,
+    # misc
+    "xxhash64": _f_xxhash64, "hash": _f_xxhash64,
+    "monotonically_increasing_id": _f_monotonic_id,
+    "typeof": _f_typeof,
+    # conversion helpers
+    "double": _cast_fn(T.F64), "float": _cast_fn(T.F32), "int": _cast_fn(T.I32),
+    "integer": _cast_fn(T.I32), "bigint": _cast_fn(T.I64), "long": _cast_fn(T.I64),
+    "string": _cast_fn(T.STRING), "date": _cast_fn(T.DATE),
+    "timestamp": _cast_fn(T.TIMESTAMP), "boolean": _cast_fn(T.BOOL),
+}

@@ -1,0 +1,299 @@
+"""Session layer: catalog + SessionContext + DataFrame surface.
+
+The analogue of the reference's session factory and catalog manager
+(ref: crates/sail-session/src/session_factory/server.rs:84,
+crates/sail-catalog/src/manager/). One SessionContext per client session;
+tables live as device-resident Tables (or lazy providers) in the catalog.
+"""
+from __future__ import annotations
+
+import threading
+from typing import Callable, Dict, List, Optional, Tuple
+
+import torch
+
+from ..plan import spec as S
+from . import types as T
+from .chunk import Chunk
+from .column import Column, Table
+
+
+class Catalog:
+    """In-memory catalog: tables (materialized or lazy providers) and views.
+    ref: crates/sail-catalog-memory/ for the reference's memory catalog."""
+
+    def __init__(self):
+        self._tables: Dict[str, Table] = {}
+        self._schemas: Dict[str, List[Tuple[str, T.DataType]]] = {}
+        self._providers: Dict[str, Callable[[torch.device], Table]] = {}
+        self._views: Dict[str, S.Plan] = {}
+        self._lock = threading.RLock()
+
+    def _key(self, name: str) -> str:
+        return name.lower().split(".")[-1]
+
+    # -- registration ------------------------------------------------------
+    def register_table(self, name: str, table: Table,
+                       schema: Optional[List[Tuple[str, T.DataType]]] = None):
+        with self._lock:
+            k = self._key(name)
+            self._tables[k] = table
+            if schema is None:
+                schema = [(n, c.dtype) for n, c in table.columns.items()]
+            self._schemas[k] = schema
+
+    def register_table_chunk(self, name: str, chunk: Chunk, schema):
+        self.register_table(name, chunk.to_table(), schema)
+
+    def register_provider(self, name: str, schema: List[Tuple[str, T.DataType]],
+                          provider: Callable[[torch.device], Table]):
+        """Lazy table: materialized per device on first access."""
+        with self._lock:
+            k = self._key(name)
+            self._providers[k] = provider
+            self._schemas[k] = schema
+
+    def create_view(self, name: str, plan: S.Plan, replace: bool = False):
+        with self._lock:
+            k = self._key(name)
+            if k in self._views and not replace:
+                raise ValueError(f"view {name} already exists")
+            self._views[k] = plan
+
+    def create_empty_table(self, name: str, columns: List[Tuple[str, T.DataType]]):
+        cols = {}
+        for n, t in columns:
+            cols[n] = Column.from_values([], t)
+        self.register_table(name, Table(cols), columns)
+
+    def insert_into(self, name: str, chunk: Chunk, overwrite: bool = False):
+        from .executor import concat_columns
+
+        with self._lock:
+            k = self._key(name)
+            existing = self._tables.get(k)
+            new = chunk.to_table()
+            if existing is None or overwrite or existing.num_rows == 0:
+                # keep declared schema names
+                names = [n for n, _ in self._schemas.get(k, [])] or list(new.columns)
+                self._tables[k] = Table({nm: c for nm, c in zip(names, new.columns.values())})
+            else:
+                cols = {}
+                for (nm, old), newc in zip(existing.columns.items(), new.columns.values()):
+                    cols[nm] = concat_columns([old, newc])
+                self._tables[k] = Table(cols)
+
+    def drop(self, name: str, if_exists: bool = False):
+        with self._lock:
+            k = self._key(name)
+            found = False
+            for d in (self._tables, self._schemas, self._views, self._providers):
+                if k in d:
+                    del d[k]
+                    found = True
+            if not found and not if_exists:
+                raise ValueError(f"table or view not found: {name}")
+
+    # -- lookup ------------------------------------------------------------
+    def view_plan(self, name: str) -> Optional[S.Plan]:
+        return self._views.get(self._key(name))
+
+    def table_schema(self, name: str) -> Optional[List[Tuple[str, T.DataType]]]:
+        k = self._key(name)
+        if k in self._views:
+            return None  # resolved via view_plan
+        return self._schemas.get(k)
+
+    def get_table_data(self, name: str, device) -> Optional[Table]:
+        k = self._key(name)
+        with self._lock:
+            t = self._tables.get(k)
+            if t is not None:
+                return t.to(device)
+            prov = self._providers.get(k)
+        if prov is not None:
+            t = prov(torch.device(device))
+            return t
+        return None
+
+    def list_tables(self) -> List[str]:
+        return sorted(set(self._tables) | set(self._views) | set(self._providers))
+
+
+class SessionContext:
+    """Per-session context. `sql()` is the whole pipeline:
+    parse -> resolve -> optimize -> execute (ref: the reference's
+    resolve_and_execute_plan, crates/sail-plan/src/lib.rs:34)."""
+
+    def __init__(self, device: Optional[str] = None, catalog: Optional[Catalog] = None):
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        self.catalog = catalog or Catalog()
+        self.conf: Dict[str, str] = {
+            "spark.sql.session.timeZone": "UTC",
+            "spark.sql.ansi.enabled": "false",
+        }
+
+    # -- pipeline ----------------------------------------------------------
+    def parse(self, sql: str) -> S.Plan:
+        from ..sql.parser import parse_sql
+
+        return parse_sql(sql)
+
+    def resolve(self, plan: S.Plan) -> S.Plan:
+        from ..plan.resolver import Resolver
+
+        return _CatalogAdapter(self).resolve(plan)
+
+    def optimize(self, plan: S.Plan) -> S.Plan:
+        from ..plan.optimizer import optimize
+
+        return optimize(plan)
+
+    def plan_sql(self, sql: str) -> S.Plan:
+        return self.optimize(self.resolve(self.parse(sql)))
+
+    def sql(self, sql: str) -> "DataFrame":
+        plan = self.plan_sql(sql)
+        if isinstance(plan, S.Command):
+            # commands (DDL/config/writes) execute eagerly, like Spark sql()
+            chunk = self.execute_plan(plan)
+            return _MaterializedDataFrame(self, plan, chunk)
+        return DataFrame(self, plan)
+
+    def table(self, name: str) -> "DataFrame":
+        return DataFrame(self, self.optimize(self.resolve(S.Read(table=name))))
+
+    def execute_plan(self, plan: S.Plan) -> Chunk:
+        from .executor import ExecutionContext, Executor
+
+        ctx = ExecutionContext(self, self.device)
+        return Executor(ctx).execute(plan)
+
+    # -- convenience -------------------------------------------------------
+    def create_dataframe(self, data: Dict[str, list],
+                         schema: Optional[Dict[str, T.DataType]] = None,
+                         name: Optional[str] = None) -> "DataFrame":
+        if schema is None:
+            schema = {}
+            from ..plan.resolver import _infer_pytype
+
+            for k, v in data.items():
+                schema[k] = _infer_pytype(v)
+        t = Table.from_pydict(data, schema, device="cpu")
+        if name:
+            self.catalog.register_table(name, t)
+        plan = S.LocalRelation(data=data)
+        plan.schema = [(k, schema[k]) for k in data]
+        return DataFrame(self, plan)
+
+
+class _CatalogAdapter:
+    """Bridges the resolver's catalog protocol to Catalog + view expansion."""
+
+    def __init__(self, session: SessionContext):
+        self.session = session
+
+    def resolve(self, plan: S.Plan) -> S.Plan:
+        from ..plan.resolver import Resolver
+
+        return Resolver(self).resolve(plan)
+
+    def table_schema(self, name: str):
+        cat = self.session.catalog
+        v = cat.view_plan(name)
+        if v is not None:
+            return None
+        return cat.table_schema(name)
+
+    def view_plan(self, name: str):
+        return self.session.catalog.view_plan(name)
+
+
+# patch Resolver to consult views: Read resolution checks views first
+def _read_with_views(resolver, p, outer):
+    cat = resolver.catalog
+    vp = cat.view_plan(p.table) if hasattr(cat, "view_plan") else None
+    if vp is not None:
+        import copy
+
+        sub = copy.deepcopy(vp)
+        resolved = resolver._plan(sub, None)
+        out = S.SubqueryAlias(input=resolved, alias=p.table.split(".")[-1])
+        out.schema = resolved.schema
+        return out
+    return resolver._p_Read_orig(p, outer)
+
+
+def _install_view_hook():
+    from ..plan.resolver import Resolver
+
+    if not hasattr(Resolver, "_p_Read_orig"):
+        Resolver._p_Read_orig = Resolver._p_Read
+        Resolver._p_Read = _read_with_views
+
+
+_install_view_hook()
+
+
+class DataFrame:
+    """Minimal DataFrame facade over a resolved plan (result surface for the
+    Connect server and Python API)."""
+
+    def __init__(self, session: SessionContext, plan: S.Plan):
+        self.session = session
+        self.plan = plan
+
+    @property
+    def schema(self):
+        return self.plan.schema
+
+    def collect_chunk(self) -> Chunk:
+        return self.session.execute_plan(self.plan)
+
+    def to_pydict(self) -> Dict[str, list]:
+        c = self.collect_chunk()
+        out = {}
+        for n, col in zip(c.names, c.columns):
+            name, k = n, 1
+            while name in out:
+                k += 1
+                name = f"{n}_{k}"
+            out[name] = col.to_pylist()
+        return out
+
+    def collect(self) -> List[tuple]:
+        d = self.to_pydict()
+        names = list(d.keys())
+        nrows = len(d[names[0]]) if names else 0
+        return [tuple(d[n][i] for n in names) for i in range(nrows)]
+
+    def to_arrow(self):
+        import pyarrow as pa
+
+        from ..datasource.arrow_io import chunk_to_arrow
+
+        return chunk_to_arrow(self.collect_chunk(), self.plan.schema)
+
+    def count(self) -> int:
+        return self.collect_chunk().num_rows
+
+    def show(self, n: int = 20):
+        rows = self.collect()[:n]
+        names = [nm for nm, _ in self.plan.schema]
+        print(" | ".join(names))
+        for r in rows:
+            print(" | ".join(str(v) for v in r))
+
+    def explain(self) -> str:
+        return S.plan_tree_string(self.plan)
+
+
+class _MaterializedDataFrame(DataFrame):
+    def __init__(self, session, plan, chunk):
+        super().__init__(session, plan)
+        self._chunk = chunk
+
+    def collect_chunk(self) -> Chunk:
+        return self._chunk

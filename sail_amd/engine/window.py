@@ -1,0 +1,154 @@
+"""Window function evaluation (torch path).
+
+Partition/order by sort, then per-function segment ops
+(ref: crates/sail-function/src/window/, DataFusion BoundedWindowAggExec role).
+Supported: row_number, rank, dense_rank, percent_rank, cume_dist, ntile,
+lag/lead, sum/count/min/max/avg over the default or running frame.
+"""
+from __future__ import annotations
+
+from typing import List
+
+import torch
+
+from ..plan import spec as S
+from . import types as T
+from .chunk import Chunk
+from .column import Column, StringColumn
+from .eval import Evaluator, broadcast
+from .joins import normalize_key
+
+
+def eval_window(ev: Evaluator, e: S.WindowExpr, chunk: Chunk) -> Column:
+    n = chunk.num_rows
+    dev = chunk.device
+    if n == 0:
+        return Column(e.dtype, torch.zeros(0, dtype=(e.dtype.storage or torch.int64), device=dev))
+
+    # 1) sort by (partition, order)
+    from .executor import sort_indices
+
+    keys = [S.SortKey(p, True, None) for p in e.partition_by] + list(e.order_by)
+    idx = sort_indices(ev, keys, chunk) if keys else torch.arange(n, device=dev)
+    sorted_chunk = chunk.gather(idx)
+
+    # 2) partition boundaries in sorted order
+    if e.partition_by:
+        pcols = [broadcast(ev.eval(p, sorted_chunk), n, dev) for p in e.partition_by]
+        pk = [normalize_key(c) for c in pcols]
+        change = torch.zeros(n, dtype=torch.bool, device=dev)
+        for k in pk:
+            change[1:] |= k[1:] != k[:-1]
+        change[0] = True
+    else:
+        change = torch.zeros(n, dtype=torch.bool, device=dev)
+        change[0] = True
+    part_id = torch.cumsum(change.to(torch.int64), 0) - 1
+    part_start_pos = torch.nonzero(change, as_tuple=False).squeeze(1)
+    pos_in_part = torch.arange(n, device=dev) - part_start_pos[part_id]
+
+    # order-key ties (for rank)
+    if e.order_by:
+        ocols = [broadcast(ev.eval(k.child, sorted_chunk), n, dev) for k in e.order_by]
+        ok = [normalize_key(c) for c in ocols]
+        newval = change.clone()
+        for k in ok:
+            newval[1:] |= k[1:] != k[:-1]
+        newval[0] = True
+    else:
+        newval = change
+
+    f = e.func
+    fname = f.name.lower() if isinstance(f, (S.Func, S.AggFunc)) else None
+    out = None
+
+    if fname == "row_number":
+        out = Column(T.I32, (pos_in_part + 1).to(torch.int32))
+    elif fname in ("rank", "dense_rank"):
+        if fname == "rank":
+            rank_at = torch.where(newval, pos_in_part + 1, torch.zeros_like(pos_in_part))
+            run = torch.cummax(torch.where(newval, pos_in_part + 1, torch.zeros_like(pos_in_part))
+                               + part_id * (2 * n), 0).values - part_id * (2 * n)
+            out = Column(T.I32, run.to(torch.int32))
+        else:
+            dr = torch.cumsum(newval.to(torch.int64), 0)
+            base = dr[part_start_pos[part_id]]
+            out = Column(T.I32, (dr - base + 1).to(torch.int32))
+    elif fname == "percent_rank":
+        part_sizes = torch.zeros(int(part_id.max().item()) + 1, dtype=torch.int64, device=dev)
+        part_sizes.index_add_(0, part_id, torch.ones(n, dtype=torch.int64, device=dev))
+        run = torch.cummax(torch.where(newval, pos_in_part + 1, torch.zeros_like(pos_in_part))
+                           + part_id * (2 * n), 0).values - part_id * (2 * n)
+        denom = (part_sizes[part_id] - 1).clamp_min(1).to(torch.float64)
+        out = Column(T.F64, (run - 1).to(torch.float64) / denom)
+    elif fname in ("lag", "lead"):
+        src = broadcast(ev.eval(f.args[0], sorted_chunk), n, dev)
+        off = int(f.args[1].value) if len(f.args) > 1 and isinstance(f.args[1], S.Literal) else 1
+        if fname == "lead":
+            off = -off
+        tgt = torch.arange(n, device=dev) - off
+        valid = (tgt >= 0) & (tgt < n)
+        tgt_part = torch.where(valid, part_id[tgt.clamp(0, n - 1)], torch.full_like(tgt, -1))
+        valid &= tgt_part == part_id
+        safe = tgt.clamp(0, n - 1)
+        g = src.gather(safe)
+        vmask = g.valid_mask() & valid
+        if isinstance(g, StringColumn):
+            g.validity = vmask.to(torch.uint8)
+            out = g
+        else:
+            out = Column(src.dtype, g.data, vmask.to(torch.uint8))
+    elif fname == "ntile":
+        buckets = int(f.args[0].value)
+        part_sizes = torch.zeros(int(part_id.max().item()) + 1, dtype=torch.int64, device=dev)
+        part_sizes.index_add_(0, part_id, torch.ones(n, dtype=torch.int64, device=dev))
+        sz = part_sizes[part_id]
+        out = Column(T.I32, (pos_in_part * buckets // sz + 1).to(torch.int32))
+    elif isinstance(f, S.AggFunc):
+        out = _window_agg(ev, f, sorted_chunk, part_id, pos_in_part, e, n, dev)
+    if out is None:
+        raise NotImplementedError(f"window function {fname}")
+
+    # 3) scatter back to original row order
+    inv = torch.empty_like(idx)
+    inv[idx] = torch.arange(n, device=dev)
+    return out.gather(inv)
+
+
+def _window_agg(ev, f: S.AggFunc, sorted_chunk, part_id, pos_in_part, e, n, dev):
+    """sum/count/avg/min/max over whole partition (default frame with no
+    ORDER BY) or running frame (with ORDER BY)."""
+    from .aggregates import agg_eval
+
+    running = bool(e.order_by) and (e.frame is None or e.frame[1][0] == "unbounded_preceding")
+    whole = not e.order_by or (e.frame is not None and e.frame[1][0] == "unbounded_preceding"
+                               and e.frame[2][0] == "unbounded_following")
+    ng = int(part_id.max().item()) + 1
+    args = [broadcast(ev.eval(a, sorted_chunk), n, dev) for a in f.args] if f.args else []
+    if whole:
+        per_group = agg_eval(f.name, args, part_id, ng, f.distinct, None, f.dtype)
+        return per_group.gather(part_id)
+    if running:
+        # cumulative within partition
+        if f.name == "count":
+            return Column(T.I64, pos_in_part + 1)
+        c = args[0]
+        x = c.data.to(torch.float64 if c.dtype.is_float else torch.int64)
+        cum = torch.cumsum(x, 0)
+        part_start = torch.nonzero(torch.cat([torch.ones(1, dtype=torch.bool, device=dev),
+                                              part_id[1:] != part_id[:-1]]), as_tuple=False).squeeze(1)
+        base = torch.where(part_start[part_id] > 0, cum[(part_start[part_id] - 1).clamp(0)],
+                           torch.zeros_like(cum[0]).expand(n) if cum.dim() else torch.zeros(n, dtype=cum.dtype, device=dev))
+        run = cum - base
+        if f.name == "sum":
+            return Column(f.dtype, run if not isinstance(f.dtype, T.DecimalType) else run.to(torch.int64))
+        if f.name == "avg":
+            return Column(T.F64, run.to(torch.float64) / (pos_in_part + 1).to(torch.float64))
+        if f.name in ("min", "max"):
+            opped = torch.cummin if f.name == "min" else torch.cummax
+            # segment cumulative min/max: reset at partition starts via offset trick
+            big = x.max() - x.min() + 1 if n else 1
+            biased = x + part_id * (big if f.name == "max" else -big)
+            res = opped(biased, 0).values - part_id * (big if f.name == "max" else -big)
+            return Column(f.dtype, res.to(args[0].data.dtype))
+    raise NotImplementedError(f"window frame for {f.name}")

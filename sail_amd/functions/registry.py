@@ -1,0 +1,223 @@
+"""Spark function registry.
+
+Mirrors the reference's function name registry
+(ref: crates/sail-plan/src/function/scalar/*, aggregate.rs, window.rs) —
+names and semantics from Spark, implementations our own. Each scalar function
+declares a return-type rule; implementations live in engine/eval.py (torch
+path, runs on CPU and on ROCm) and, for the hot subset, as ExprVM opcodes in
+ops/csrc/exprvm.hip.
+"""
+from __future__ import annotations
+
+from typing import Callable, Dict, List, Optional
+
+from ..engine import types as T
+
+# ---------------------------------------------------------------------------
+# Aggregate functions (ref: sail-plan function/aggregate.rs ~85 names)
+# ---------------------------------------------------------------------------
+
+AGG_FUNCTIONS = {
+    "sum", "avg", "mean", "count", "min", "max", "first", "first_value",
+    "last", "last_value", "stddev", "stddev_samp", "stddev_pop", "variance",
+    "var_samp", "var_pop", "count_if", "any", "some", "bool_or", "every",
+    "bool_and", "collect_list", "array_agg", "collect_set", "approx_count_distinct",
+    "corr", "covar_samp", "covar_pop", "skewness", "kurtosis", "median",
+    "percentile", "percentile_approx", "approx_percentile", "mode", "product",
+    "sum_distinct", "bit_and", "bit_or", "bit_xor", "max_by", "min_by",
+    "any_value", "try_sum", "try_avg", "regr_count", "regr_avgx", "regr_avgy",
+    "regr_slope", "regr_intercept", "regr_r2", "regr_sxx", "regr_syy", "regr_sxy",
+    "grouping", "grouping_id", "histogram_numeric", "string_agg", "listagg",
+}
+
+WINDOW_FUNCTIONS = {
+    "row_number", "rank", "dense_rank", "percent_rank", "cume_dist", "ntile",
+    "lag", "lead", "nth_value",
+}
+
+
+def agg_return_type(name: str, arg_types: List[T.DataType], distinct: bool = False) -> T.DataType:
+    a = arg_types[0] if arg_types else T.NULL
+    if name in ("count", "count_if", "approx_count_distinct", "regr_count"):
+        return T.I64
+    if name in ("sum", "try_sum", "sum_distinct"):
+        if isinstance(a, T.DecimalType):
+            return T.DecimalType(min(38, a.precision + 10), a.scale)
+        if a.is_integer:
+            return T.I64
+        return T.F64
+    if name in ("avg", "mean", "try_avg"):
+        if isinstance(a, T.DecimalType):
+            return T.DecimalType(min(38, a.precision + 4), min(a.scale + 4, 10))
+        return T.F64
+    if name in ("min", "max", "first", "first_value", "last", "last_value",
+                "any_value", "mode", "median", "max_by", "min_by"):
+        return a
+    if name in ("stddev", "stddev_samp", "stddev_pop", "variance", "var_samp",
+                "var_pop", "corr", "covar_samp", "covar_pop", "skewness",
+                "kurtosis", "percentile", "percentile_approx", "approx_percentile",
+                "product") or name.startswith("regr_"):
+        return T.F64
+    if name in ("any", "some", "bool_or", "every", "bool_and"):
+        return T.BOOL
+    if name in ("collect_list", "array_agg", "collect_set"):
+        return T.ArrayType(a)
+    if name in ("bit_and", "bit_or", "bit_xor"):
+        return a if a.is_integer else T.I64
+    if name in ("grouping", "grouping_id"):
+        return T.I32 if name == "grouping" else T.I64
+    if name in ("string_agg", "listagg"):
+        return T.STRING
+    raise KeyError(f"unknown aggregate {name}")
+
+
+# ---------------------------------------------------------------------------
+# Scalar function type inference
+# ---------------------------------------------------------------------------
+
+def _same(args):
+    return args[0]
+
+
+def _f64(args):
+    return T.F64
+
+
+def _i32(args):
+    return T.I32
+
+
+def _i64(args):
+    return T.I64
+
+
+def _bool(args):
+    return T.BOOL
+
+
+def _string(args):
+    return T.STRING
+
+
+def _date(args):
+    return T.DATE
+
+
+def _ts(args):
+    return T.TIMESTAMP
+
+
+def _numeric_common(args):
+    t = args[0]
+    for a in args[1:]:
+        t = T.common_type(t, a)
+    return t
+
+
+# name -> return-type rule.  Grouped like the reference's scalar modules
+# (ref: crates/sail-function/src/scalar/).
+SCALAR_RETURN: Dict[str, Callable[[List[T.DataType]], T.DataType]] = {}
+
+
+def _reg(names, rule):
+    for n in names.split():
+        SCALAR_RETURN[n] = rule
+
+
+# math (ref: sail-function/src/scalar/math)
+_reg("abs", _same)
+_reg("ceil ceiling floor", lambda a: T.I64 if not isinstance(a[0], T.DecimalType) else T.DecimalType(a[0].precision - a[0].scale + 1, 0))
+_reg("round bround", lambda a: a[0])
+_reg("sqrt cbrt exp expm1 ln log log10 log2 log1p sin cos tan asin acos atan "
+     "sinh cosh tanh asinh acosh atanh atan2 degrees radians pi e rand randn "
+     "power pow hypot", _f64)
+_reg("sign signum", _f64)
+_reg("mod pmod", _numeric_common)
+_reg("greatest least", _numeric_common)
+_reg("factorial", _i64)
+_reg("bin hex unhex conv", _string)
+_reg("bitwise_not shiftleft shiftright shiftrightunsigned bit_count", lambda a: a[0] if a and a[0].is_integer else T.I32)
+_reg("positive negative", _same)
+_reg("width_bucket", _i64)
+_reg("try_add try_subtract try_multiply try_divide", _numeric_common)
+_reg("rint", _f64)
+
+# string (ref: sail-function/src/scalar/string)
+_reg("concat concat_ws upper ucase lower lcase trim ltrim rtrim btrim initcap "
+     "reverse repeat lpad rpad substring substr left right replace translate "
+     "regexp_replace regexp_extract regexp_extract_all split_part soundex "
+     "chr char space format_string printf format_number substring_index "
+     "overlay sentences elt base64 unbase64 decode encode to_char", _string)
+_reg("length len char_length character_length octet_length bit_length "
+     "instr locate position levenshtein crc32 ascii find_in_set", _i32)
+_reg("startswith endswith contains like ilike rlike regexp regexp_like", _bool)
+_reg("split", lambda a: T.ArrayType(T.STRING))
+_reg("md5 sha sha1 sha2 uuid", _string)
+_reg("hash", _i32)
+_reg("xxhash64", _i64)
+
+# datetime (ref: sail-function/src/scalar/datetime)
+_reg("year month day dayofmonth dayofweek dayofyear weekday weekofyear quarter "
+     "hour minute second", _i32)
+_reg("date_part_doy date_part_dow", _i32)
+_reg("datediff date_diff", _i32)
+_reg("date_add dateadd date_sub add_months last_day next_day to_date trunc", _date)
+_reg("date_trunc to_timestamp timestamp_seconds timestamp_millis timestamp_micros "
+     "from_unixtime from_utc_timestamp to_utc_timestamp make_timestamp", _ts)
+_reg("current_date curdate", _date)
+_reg("current_timestamp now localtimestamp", _ts)
+_reg("unix_timestamp to_unix_timestamp unix_seconds unix_millis unix_micros "
+     "unix_date", _i64)
+_reg("months_between", _f64)
+_reg("date_format", _string)
+_reg("make_date", _date)
+_reg("extract", _i32)
+
+# conditional / misc (ref: sail-function/src/scalar/misc, predicate)
+_reg("coalesce nvl ifnull", _numeric_common)
+_reg("nvl2", lambda a: T.common_type(a[1], a[2]))
+_reg("nullif", _same)
+_reg("if iff", lambda a: T.common_type(a[1], a[2]))
+_reg("isnull isnotnull isnan", _bool)
+_reg("nanvl", _f64)
+_reg("assert_true raise_error", lambda a: T.NULL)
+_reg("monotonically_increasing_id spark_partition_id input_file_block_start "
+     "input_file_block_length", _i64)
+_reg("input_file_name current_user current_database current_catalog version", _string)
+_reg("typeof", _string)
+_reg("get_field", lambda a: T.NULL)  # resolved structurally
+_reg("struct named_struct", lambda a: T.StructType(tuple(T.StructField(f"col{i+1}", t) for i, t in enumerate(a))))
+
+# array/collection (subset; ref: sail-function/src/scalar/array, collection)
+_reg("array", lambda a: T.ArrayType(a[0] if a else T.NULL))
+_reg("array_contains", _bool)
+_reg("size cardinality array_size", _i32)
+_reg("array_max array_min", lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
+_reg("element_at element_at_sql get", lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
+_reg("sort_array array_sort array_distinct array_remove array_compact flatten "
+     "slice array_repeat shuffle", _same)
+_reg("array_join", _string)
+_reg("array_position", _i64)
+_reg("arrays_overlap", _bool)
+_reg("sequence", lambda a: T.ArrayType(a[0]))
+_reg("explode explode_outer posexplode", lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
+
+# type conversion helpers
+_reg("double", _f64)
+_reg("float", lambda a: T.F32)
+_reg("int integer", _i32)
+_reg("bigint long", _i64)
+_reg("smallint", lambda a: T.I16)
+_reg("tinyint", lambda a: T.I8)
+_reg("boolean", _bool)
+_reg("string", _string)
+_reg("decimal", lambda a: T.DecimalType(10, 0))
+_reg("date", _date)
+_reg("timestamp", _ts)
+
+
+def scalar_return_type(name: str, arg_types: List[T.DataType]) -> Optional[T.DataType]:
+    rule = SCALAR_RETURN.get(name)
+    if rule is None:
+        return None
+    return rule(arg_types)

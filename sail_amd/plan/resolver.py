@@ -1,0 +1,828 @@
+"""Plan resolver: spec IR -> typed, bound logical plan.
+
+The analogue of the reference's PlanResolver
+(ref: crates/sail-plan/src/resolver/mod.rs:20, resolver/plan.rs:18): binds
+column names to input ordinals, infers types, inserts casts, normalizes
+aggregates (projection split), inlines CTEs, expands USING/NATURAL joins and
+`*`, and marks correlated references as OuterRef for the decorrelator.
+
+The resolved tree uses the same spec node classes with `schema`/`dtype`
+filled in and all Col nodes replaced by BoundRef/OuterRef.
+"""
+from __future__ import annotations
+
+import copy
+from dataclasses import dataclass, field, replace
+from typing import Dict, List, Optional, Tuple
+
+from ..engine import types as T
+from ..functions.registry import AGG_FUNCTIONS, WINDOW_FUNCTIONS, agg_return_type, scalar_return_type
+from . import spec as S
+
+
+class ResolutionError(Exception):
+    pass
+
+
+@dataclass
+class Field:
+    name: str
+    dtype: T.DataType
+    qualifier: Optional[str] = None
+
+
+@dataclass
+class Scope:
+    fields: List[Field] = field(default_factory=list)
+    outer: Optional["Scope"] = None
+
+    def find(self, name: str, qualifier: Optional[str]) -> List[int]:
+        lname = name.lower()
+        lq = qualifier.lower() if qualifier else None
+        hits = []
+        for i, f in enumerate(self.fields):
+            if f.name.lower() != lname:
+                continue
+            if lq is not None and (f.qualifier or "").lower() != lq:
+                continue
+            hits.append(i)
+        return hits
+
+
+class Resolver:
+    """Resolves a spec plan against a catalog (dict of table -> schema)."""
+
+    def __init__(self, catalog):
+        self.catalog = catalog  # engine/session.Catalog
+        self.cte_scope: List[Dict[str, S.Plan]] = []
+
+    # =====================================================================
+    def resolve(self, plan: S.Plan) -> S.Plan:
+        return self._plan(plan, outer=None)
+
+    # -- plans -------------------------------------------------------------
+    def _plan(self, p: S.Plan, outer: Optional[Scope]) -> S.Plan:
+        m = getattr(self, "_p_" + type(p).__name__, None)
+        if m is None:
+            raise ResolutionError(f"cannot resolve plan node {type(p).__name__}")
+        return m(p, outer)
+
+    def _scope(self, p: S.Plan, qualifier: Optional[str] = None, outer=None) -> Scope:
+        return Scope([Field(n, t, qualifier) for n, t in p.schema], outer)
+
+    def _p_Read(self, p: S.Read, outer):
+        # CTE reference?
+        for scope in reversed(self.cte_scope):
+            if p.table.lower() in scope:
+                sub = copy.deepcopy(scope[p.table.lower()])
+                resolved = self._plan(sub, None) if sub.schema is None else sub
+                aliased = S.SubqueryAlias(input=resolved, alias=p.table)
+                aliased.schema = resolved.schema
+                return self._qualify(aliased, p.table)
+        schema = self.catalog.table_schema(p.table)
+        if schema is None:
+            raise ResolutionError(f"table not found: {p.table}")
+        out = S.Read(table=p.table, options=p.options)
+        out.schema = list(schema)
+        base = p.table.split(".")[-1]
+        return self._qualify(out, base)
+
+    def _qualify(self, p: S.Plan, name: str) -> S.Plan:
+        """Attach a qualifier by wrapping in SubqueryAlias (resolved)."""
+        out = S.SubqueryAlias(input=p, alias=name)
+        out.schema = p.schema
+        return out
+
+    def _p_DataSourceRead(self, p: S.DataSourceRead, outer):
+        from ..datasource.registry import infer_source_schema
+
+        out = S.DataSourceRead(format=p.format, paths=p.paths, options=p.options,
+                               user_schema=p.user_schema)
+        out.schema = p.user_schema or infer_source_schema(p.format, p.paths, p.options)
+        return out
+
+    def _p_LocalRelation(self, p: S.LocalRelation, outer):
+        out = S.LocalRelation(data=p.data)
+        if p.schema:
+            out.schema = p.schema
+        else:
+            out.schema = [(k, _infer_pytype(v)) for k, v in p.data.items()]
+        return out
+
+    def _p_Range(self, p: S.Range, outer):
+        out = S.Range(p.start, p.end, p.step)
+        out.schema = [("id", T.I64)]
+        return out
+
+    def _p_WithCte(self, p: S.WithCte, outer):
+        scope: Dict[str, S.Plan] = {}
+        self.cte_scope.append(scope)
+        try:
+            for name, sub in p.ctes:
+                resolved = self._plan(sub, outer)
+                scope[name.lower()] = resolved
+            return self._plan(p.input, outer)
+        finally:
+            self.cte_scope.pop()
+
+    def _p_SubqueryAlias(self, p: S.SubqueryAlias, outer):
+        child = self._plan(p.input, outer)
+        names = p.column_aliases or [n for n, _ in child.schema]
+        if len(names) != len(child.schema):
+            raise ResolutionError(f"alias {p.alias}: {len(names)} aliases for {len(child.schema)} columns")
+        out = S.SubqueryAlias(input=child, alias=p.alias, column_aliases=p.column_aliases)
+        out.schema = [(names[i], child.schema[i][1]) for i in range(len(names))]
+        return out
+
+    def _p_Filter(self, p: S.Filter, outer):
+        child = self._plan(p.input, outer)
+        scope = self._child_scope(child, outer)
+        cond = self._expr(p.condition, scope)
+        cond = _coerce_to_bool(cond)
+        out = S.Filter(input=child, condition=cond)
+        out.schema = child.schema
+        return out
+
+    def _p_Project(self, p: S.Project, outer):
+        child = self._plan(p.input, outer)
+        scope = self._child_scope(child, outer)
+        exprs: List[S.Expr] = []
+        for e in p.exprs:
+            exprs.extend(self._expand_star(e, scope))
+        bound = [self._expr(e, scope) for e in exprs]
+
+        # extract window expressions into a WindowPlan below the projection
+        windows: List[S.Expr] = []
+
+        def extract_windows(e: S.Expr) -> S.Expr:
+            if isinstance(e, S.WindowExpr):
+                idx = len(child.schema) + len(windows)
+                windows.append(e)
+                return S.BoundRef(idx, f"__w{len(windows)-1}", e.dtype)
+            if isinstance(e, S.Alias):
+                return S.Alias(extract_windows(e.child), e.name, e.dtype)
+            ch = e.children()
+            if not ch:
+                return e
+            out = e.with_children([extract_windows(c) for c in ch])
+            out.dtype = e.dtype
+            return out
+
+        bound2 = [extract_windows(e) for e in bound]
+        if windows:
+            wp = S.WindowPlan(input=child, window_exprs=windows)
+            wp.schema = list(child.schema) + [(f"__w{i}", w.dtype) for i, w in enumerate(windows)]
+            out = S.Project(input=wp, exprs=bound2)
+            out.schema = [(_expr_name(e, i), e.dtype) for i, e in enumerate(bound2)]
+            return out
+        out = S.Project(input=child, exprs=bound)
+        out.schema = [(_expr_name(e, i), e.dtype) for i, e in enumerate(bound)]
+        return out
+
+    def _p_Distinct(self, p: S.Distinct, outer):
+        child = self._plan(p.input, outer)
+        out = S.Distinct(input=child)
+        out.schema = child.schema
+        return out
+
+    def _p_Limit(self, p: S.Limit, outer):
+        child = self._plan(p.input, outer)
+        out = S.Limit(input=child, n=p.n, offset=p.offset)
+        out.schema = child.schema
+        return out
+
+    def _p_Sort(self, p: S.Sort, outer):
+        child = self._plan(p.input, outer)
+        scope = self._child_scope(child, outer)
+        keys: List[S.SortKey] = []
+        extra_exprs: List[S.Expr] = []
+        for k in p.keys:
+            ke = self._resolve_sort_expr(k.child, child, scope, extra_exprs)
+            keys.append(S.SortKey(ke, k.ascending, k.nulls_first))
+        if extra_exprs and isinstance(child, S.Project):
+            # widen the projection with hidden sort columns; Sort then projects back
+            inner = child
+            new_exprs = list(inner.exprs) + extra_exprs
+            wide = S.Project(input=inner.input, exprs=new_exprs)
+            wide.schema = [(_expr_name(e, i), e.dtype) for i, e in enumerate(new_exprs)]
+            srt = S.Sort(input=wide, keys=keys)
+            srt.schema = wide.schema
+            trim = S.Project(input=srt, exprs=[
+                S.BoundRef(i, n, t) for i, (n, t) in enumerate(inner.schema)])
+            trim.schema = inner.schema
+            return trim
+        out = S.Sort(input=child, keys=keys)
+        out.schema = child.schema
+        return out
+
+    def _resolve_sort_expr(self, e: S.Expr, child: S.Plan, scope: Scope, extra: List[S.Expr]) -> S.Expr:
+        # ordinal sort key: ORDER BY 1
+        if isinstance(e, S.Literal) and isinstance(e.value, int) and not isinstance(e.value, bool):
+            idx = e.value - 1
+            if 0 <= idx < len(child.schema):
+                n, t = child.schema[idx]
+                return S.BoundRef(idx, n, t)
+        try:
+            return self._expr(e, scope)
+        except ResolutionError:
+            if isinstance(child, S.Project):
+                inner_scope = self._child_scope(child.input, scope.outer)
+                bound = self._expr(e, inner_scope)
+                idx = len(child.schema) + len(extra)
+                extra.append(bound)
+                return S.BoundRef(idx, f"__sort{idx}", bound.dtype)
+            raise
+
+    def _p_Aggregate(self, p: S.Aggregate, outer):
+        child = self._plan(p.input, outer)
+        scope = self._child_scope(child, outer)
+
+        # expand * in projections (rare in aggregates: count(*) handled below)
+        projections: List[S.Expr] = []
+        for e in p.aggs:
+            projections.extend(self._expand_star(e, scope) if isinstance(e, S.Star) else [e])
+
+        # group-by keys: ordinals and alias references resolve against projections
+        group_exprs: List[S.Expr] = []
+        for g in p.group_by:
+            if isinstance(g, S.Literal) and isinstance(g.value, int) and not isinstance(g.value, bool):
+                idx = g.value - 1
+                if not (0 <= idx < len(projections)):
+                    raise ResolutionError(f"GROUP BY ordinal {g.value} out of range")
+                tgt = projections[idx]
+                group_exprs.append(tgt.child if isinstance(tgt, S.Alias) else tgt)
+            elif isinstance(g, S.Col) and g.qualifier is None and not scope.find(g.name, None):
+                # alias reference to a projection
+                matched = None
+                for pr in projections:
+                    if isinstance(pr, S.Alias) and pr.name.lower() == g.name.lower():
+                        matched = pr.child
+                        break
+                if matched is None:
+                    raise ResolutionError(f"cannot resolve group key {g.name}")
+                group_exprs.append(matched)
+            else:
+                group_exprs.append(g)
+        bound_groups = [self._expr(g, scope) for g in group_exprs]
+
+        # collect aggregate functions from projections
+        agg_funcs: List[S.AggFunc] = []
+
+        def bind_agg(e: S.Expr) -> S.Expr:
+            """Bind a projection expr: AggFuncs -> refs into agg output;
+            group exprs -> refs to group keys."""
+            if isinstance(e, S.Alias):
+                return S.Alias(bind_agg(e.child), e.name, None)
+            # whole-expression matches a group key?
+            for gi, (ge, be) in enumerate(zip(group_exprs, bound_groups)):
+                if _expr_equal_unbound(e, ge):
+                    return S.BoundRef(gi, _expr_name(e, gi), be.dtype)
+            if isinstance(e, S.AggFunc):
+                if len(e.args) == 1 and isinstance(e.args[0], S.Star):
+                    bargs = []
+                else:
+                    bargs = [self._expr(a, scope) for a in e.args]
+                bfilter = _coerce_to_bool(self._expr(e.filter, scope)) if e.filter is not None else None
+                name = _normalize_agg_name(e.name)
+                rtype = agg_return_type(name, [a.dtype for a in bargs], e.distinct)
+                bound = S.AggFunc(name, bargs, e.distinct, rtype, bfilter)
+                pretty = _expr_name(e, 0)
+                # dedup identical aggregates
+                for ai, existing in enumerate(agg_funcs):
+                    if repr(existing) == repr(bound):
+                        return S.BoundRef(len(bound_groups) + ai, pretty, existing.dtype)
+                agg_funcs.append(bound)
+                ai = len(agg_funcs) - 1
+                return S.BoundRef(len(bound_groups) + ai, pretty, rtype)
+            ch = e.children()
+            if not ch:
+                if isinstance(e, S.Col):
+                    # alias reference to a projection (HAVING n > 1)
+                    if e.qualifier is None:
+                        for pr in projections:
+                            if isinstance(pr, S.Alias) and pr.name.lower() == e.name.lower() \
+                                    and pr.child is not e:
+                                return bind_agg(pr.child)
+                    # column not in group keys — Spark errors; match that
+                    hits = scope.find(e.name, e.qualifier)
+                    if hits:
+                        raise ResolutionError(
+                            f"column {e.name} must appear in GROUP BY or inside an aggregate")
+                    raise ResolutionError(f"cannot resolve column {e.name}")
+                return e
+            return e.with_children([bind_agg(c) for c in ch])
+
+        final_exprs = [self._type_expr(bind_agg(e)) for e in projections]
+        having_bound = None
+        if p.having is not None:
+            having_bound = _coerce_to_bool(self._type_expr(bind_agg(p.having)))
+
+        agg = S.Aggregate(input=child, group_by=bound_groups, aggs=agg_funcs,
+                          grouping_sets=p.grouping_sets)
+        agg.schema = ([(_expr_name(g, i), g.dtype) for i, g in enumerate(bound_groups)]
+                      + [(f"__agg{i}", a.dtype) for i, a in enumerate(agg_funcs)])
+        top: S.Plan = agg
+        if having_bound is not None:
+            filt = S.Filter(input=agg, condition=having_bound)
+            filt.schema = agg.schema
+            top = filt
+        proj = S.Project(input=top, exprs=final_exprs)
+        proj.schema = [(_expr_name(e, i), e.dtype) for i, e in enumerate(final_exprs)]
+        return proj
+
+    def _p_Join(self, p: S.Join, outer):
+        left = self._plan(p.left, outer)
+        right = self._plan(p.right, outer)
+        lq = _plan_qualifier(left)
+        rq = _plan_qualifier(right)
+        lfields = [Field(n, t, lq) for n, t in left.schema]
+        rfields = [Field(n, t, rq) for n, t in right.schema]
+        scope = Scope(lfields + rfields, outer)
+
+        using = p.using
+        if using and using == ["__natural__"]:
+            lnames = {n.lower() for n, _ in left.schema}
+            using = [n for n, _ in right.schema if n.lower() in lnames]
+        on = None
+        if using:
+            conds = []
+            for c in using:
+                li = Scope(lfields).find(c, None)
+                ri = Scope(rfields).find(c, None)
+                if not li or not ri:
+                    raise ResolutionError(f"USING column {c} not found on both sides")
+                lref = S.BoundRef(li[0], c, lfields[li[0]].dtype)
+                rref = S.BoundRef(len(lfields) + ri[0], c, rfields[ri[0]].dtype)
+                cond = S.BinaryOp("=", lref, rref, T.BOOL)
+                conds.append(cond)
+            on = conds[0]
+            for c in conds[1:]:
+                on = S.BinaryOp("and", on, c, T.BOOL)
+        elif p.on is not None:
+            on = _coerce_to_bool(self._expr(p.on, scope))
+
+        out = S.Join(left=left, right=right, how=p.how, on=on, using=using)
+        if p.how in ("semi", "anti"):
+            out.schema = list(left.schema)
+        elif p.how in ("rightsemi", "rightanti"):
+            out.schema = list(right.schema)
+        elif using:
+            # USING join: shared columns appear once (from the left side)
+            used = {c.lower() for c in using}
+            out.schema = list(left.schema) + [(n, t) for n, t in right.schema if n.lower() not in used]
+        else:
+            out.schema = list(left.schema) + list(right.schema)
+        return out
+
+    def _p_SetOp(self, p: S.SetOp, outer):
+        left = self._plan(p.left, outer)
+        right = self._plan(p.right, outer)
+        if len(left.schema) != len(right.schema):
+            raise ResolutionError("set operation inputs have different column counts")
+        schema = []
+        for (ln, lt), (rn, rt) in zip(left.schema, right.schema):
+            schema.append((ln, T.common_type(lt, rt)))
+        out = S.SetOp(op=p.op, left=left, right=right, is_all=p.is_all, by_name=p.by_name)
+        out.schema = schema
+        return out
+
+    def _p_WindowPlan(self, p: S.WindowPlan, outer):
+        child = self._plan(p.input, outer)
+        scope = self._child_scope(child, outer)
+        bound = [self._expr(e, scope) for e in p.window_exprs]
+        out = S.WindowPlan(input=child, window_exprs=bound)
+        out.schema = list(child.schema) + [(_expr_name(e, len(child.schema) + i), e.dtype)
+                                           for i, e in enumerate(bound)]
+        return out
+
+    # commands ------------------------------------------------------------
+    def _p_CreateView(self, p: S.CreateView, outer):
+        body = self._plan(p.input, outer)
+        out = S.CreateView(name=p.name, input=body, replace=p.replace, temporary=p.temporary)
+        out.schema = []
+        return out
+
+    def _p_CreateTable(self, p: S.CreateTable, outer):
+        inp = self._plan(p.input, outer) if p.input is not None else None
+        out = S.CreateTable(name=p.name, columns=p.columns, input=inp, format=p.format,
+                            location=p.location, replace=p.replace,
+                            if_not_exists=p.if_not_exists, options=p.options)
+        out.schema = []
+        return out
+
+    def _p_DropTable(self, p: S.DropTable, outer):
+        p.schema = []
+        return p
+
+    def _p_InsertInto(self, p: S.InsertInto, outer):
+        inp = self._plan(p.input, outer)
+        out = S.InsertInto(table=p.table, input=inp, overwrite=p.overwrite)
+        out.schema = []
+        return out
+
+    def _p_Write(self, p: S.Write, outer):
+        inp = self._plan(p.input, outer)
+        out = S.Write(input=inp, format=p.format, path=p.path, table=p.table,
+                      mode=p.mode, partition_by=p.partition_by, options=p.options)
+        out.schema = []
+        return out
+
+    def _p_Explain(self, p: S.Explain, outer):
+        inp = self._plan(p.input, outer)
+        out = S.Explain(input=inp, mode=p.mode)
+        out.schema = [("plan", T.STRING)]
+        return out
+
+    def _p_SetConfig(self, p: S.SetConfig, outer):
+        p.schema = [("key", T.STRING), ("value", T.STRING)]
+        return p
+
+    def _p_ShowTables(self, p: S.ShowTables, outer):
+        p.schema = [("namespace", T.STRING), ("tableName", T.STRING), ("isTemporary", T.BOOL)]
+        return p
+
+    def _p_DescribeTable(self, p: S.DescribeTable, outer):
+        p.schema = [("col_name", T.STRING), ("data_type", T.STRING), ("comment", T.STRING)]
+        return p
+
+    # =====================================================================
+    # expressions
+    # =====================================================================
+    def _child_scope(self, child: S.Plan, outer) -> Scope:
+        q = _plan_qualifier(child)
+        if isinstance(child, S.Join):
+            lq = _plan_qualifier(child.left)
+            rq = _plan_qualifier(child.right)
+            nleft = len(child.left.schema)
+            fields = []
+            for i, (n, t) in enumerate(child.schema):
+                if child.how in ("semi", "anti"):
+                    fields.append(Field(n, t, lq))
+                elif child.how in ("rightsemi", "rightanti"):
+                    fields.append(Field(n, t, rq))
+                elif i < nleft:
+                    fields.append(Field(n, t, lq))
+                else:
+                    fields.append(Field(n, t, rq))
+            return Scope(fields, outer)
+        return Scope([Field(n, t, q) for n, t in child.schema], outer)
+
+    def _expand_star(self, e: S.Expr, scope: Scope) -> List[S.Expr]:
+        if isinstance(e, S.Star):
+            out = []
+            for i, f in enumerate(scope.fields):
+                if e.qualifier and (f.qualifier or "").lower() != e.qualifier.lower():
+                    continue
+                out.append(S.BoundRef(i, f.name, f.dtype))
+            if not out:
+                raise ResolutionError(f"star expansion found no columns for {e.qualifier}")
+            return out
+        return [e]
+
+    def _expr(self, e: S.Expr, scope: Scope) -> S.Expr:
+        if e is None:
+            return None
+        if isinstance(e, S.Literal):
+            out = S.Literal(e.value, e.dtype or _infer_literal_type(e.value))
+            return out
+        if isinstance(e, S.BoundRef):
+            return e
+        if isinstance(e, S.Col):
+            hits = scope.find(e.name, e.qualifier)
+            if len(hits) == 1:
+                f = scope.fields[hits[0]]
+                return S.BoundRef(hits[0], f.name, f.dtype)
+            if len(hits) > 1:
+                # identical duplicates (e.g. USING join remnants): take first
+                f0 = scope.fields[hits[0]]
+                if all(scope.fields[h].dtype == f0.dtype for h in hits):
+                    return S.BoundRef(hits[0], f0.name, f0.dtype)
+                raise ResolutionError(f"ambiguous column {e.name}")
+            # outer scope (correlated subquery)
+            s = scope.outer
+            depth = 0
+            while s is not None:
+                hits = s.find(e.name, e.qualifier)
+                if hits:
+                    if depth > 0:
+                        raise ResolutionError("correlation beyond one level not supported")
+                    f = s.fields[hits[0]]
+                    return S.OuterRef(hits[0], f.name, f.dtype)
+                s = s.outer
+                depth += 1
+            raise ResolutionError(f"cannot resolve column "
+                                  f"{(e.qualifier + '.') if e.qualifier else ''}{e.name}")
+        if isinstance(e, S.Alias):
+            c = self._expr(e.child, scope)
+            return S.Alias(c, e.name, c.dtype)
+        if isinstance(e, S.ScalarSubquery):
+            sub = self._plan(e.plan, scope)
+            if len(sub.schema) != 1:
+                raise ResolutionError("scalar subquery must return one column")
+            return S.ScalarSubquery(plan=sub, dtype=sub.schema[0][1])
+        if isinstance(e, S.InSubquery):
+            c = self._expr(e.child, scope)
+            sub = self._plan(e.plan, scope)
+            return S.InSubquery(c, sub, e.negated, T.BOOL)
+        if isinstance(e, S.Exists):
+            sub = self._plan(e.plan, scope)
+            return S.Exists(plan=sub, negated=e.negated, dtype=T.BOOL)
+        if isinstance(e, S.AggFunc):
+            bargs = [self._expr(a, scope) for a in e.args if not isinstance(a, S.Star)]
+            name = _normalize_agg_name(e.name)
+            rtype = agg_return_type(name, [a.dtype for a in bargs], e.distinct)
+            bfilter = _coerce_to_bool(self._expr(e.filter, scope)) if e.filter is not None else None
+            return S.AggFunc(name, bargs, e.distinct, rtype, bfilter)
+        if isinstance(e, S.WindowExpr):
+            f = self._expr_window_func(e.func, scope)
+            part = [self._expr(x, scope) for x in e.partition_by]
+            order = [S.SortKey(self._expr(k.child, scope), k.ascending, k.nulls_first)
+                     for k in e.order_by]
+            return S.WindowExpr(func=f, partition_by=part, order_by=order,
+                                frame=e.frame, dtype=f.dtype)
+        # generic: resolve children then type
+        ch = [self._expr(c, scope) for c in e.children()]
+        out = e.with_children(ch) if ch else copy.copy(e)
+        return self._type_expr(out)
+
+    def _expr_window_func(self, f: S.Expr, scope: Scope) -> S.Expr:
+        if isinstance(f, S.AggFunc):
+            return self._expr(f, scope)
+        if isinstance(f, S.Func):
+            args = [self._expr(a, scope) for a in f.args]
+            name = f.name.lower()
+            if name in ("row_number", "rank", "dense_rank", "ntile"):
+                t = T.I32
+            elif name in ("percent_rank", "cume_dist"):
+                t = T.F64
+            elif name in ("lag", "lead", "nth_value"):
+                t = args[0].dtype if args else T.NULL
+            else:
+                t = scalar_return_type(name, [a.dtype for a in args]) or T.NULL
+            return S.Func(name, args, t)
+        raise ResolutionError(f"unsupported window function {f!r}")
+
+    # -- typing rules ------------------------------------------------------
+    def _type_expr(self, e: S.Expr) -> S.Expr:
+        if e.dtype is not None and not isinstance(e, (S.BinaryOp, S.UnaryOp, S.Func, S.Cast,
+                                                      S.CaseWhen, S.InList, S.Between, S.Like)):
+            return e
+        if isinstance(e, S.BinaryOp):
+            return _type_binary(e)
+        if isinstance(e, S.UnaryOp):
+            if e.op == "not":
+                e.child = _coerce_to_bool(e.child)
+                e.dtype = T.BOOL
+            elif e.op in ("isnull", "isnotnull"):
+                e.dtype = T.BOOL
+            elif e.op == "neg":
+                e.dtype = e.child.dtype
+            return e
+        if isinstance(e, S.Cast):
+            e.dtype = e.to
+            return e
+        if isinstance(e, S.CaseWhen):
+            e.branches = [(_coerce_to_bool(c), v) for c, v in e.branches]
+            ts = [v.dtype for _, v in e.branches] + ([e.else_.dtype] if e.else_ is not None else [])
+            t = ts[0]
+            for x in ts[1:]:
+                t = T.common_type(t, x)
+            e.dtype = t
+            return e
+        if isinstance(e, (S.InList, S.Between, S.Like, S.Exists, S.InSubquery)):
+            e.dtype = T.BOOL
+            return e
+        if isinstance(e, S.SortKey):
+            e.dtype = e.child.dtype
+            return e
+        if isinstance(e, S.Func):
+            t = scalar_return_type(e.name, [a.dtype for a in e.args])
+            if t is None:
+                raise ResolutionError(f"unknown function {e.name}")
+            # structural return types
+            if e.name == "coalesce" or e.name in ("nvl", "ifnull"):
+                tt = e.args[0].dtype
+                for a in e.args[1:]:
+                    tt = T.common_type(tt, a.dtype)
+                t = tt
+            e.dtype = t
+            return e
+        if isinstance(e, S.Alias):
+            e.dtype = e.child.dtype
+            return e
+        if e.dtype is None:
+            raise ResolutionError(f"cannot type expression {e!r}")
+        return e
+
+
+# ---------------------------------------------------------------------------
+
+_INTERVAL = "__interval__"
+
+
+def _is_interval_lit(e: S.Expr) -> bool:
+    return isinstance(e, S.Literal) and isinstance(e.value, tuple) and len(e.value) == 3 \
+        and e.value[0] == _INTERVAL
+
+
+def _type_binary(e: S.BinaryOp) -> S.Expr:
+    lt, rt = e.left.dtype, e.right.dtype
+    if e.op in ("and", "or"):
+        e.left = _coerce_to_bool(e.left)
+        e.right = _coerce_to_bool(e.right)
+        e.dtype = T.BOOL
+        return e
+    # date/timestamp +- interval
+    if e.op in ("+", "-") and (_is_interval_lit(e.right) or _is_interval_lit(e.left)):
+        ivl = e.right if _is_interval_lit(e.right) else e.left
+        other = e.left if ivl is e.right else e.right
+        _, months, micros = ivl.value
+        sign = -1 if e.op == "-" else 1
+        if isinstance(other.dtype, T.DateType):
+            if months:
+                out = S.Func("add_months", [other, S.Literal(sign * months, T.I32)], T.DATE)
+            else:
+                out = S.Func("date_add", [other, S.Literal(sign * (micros // 86_400_000_000), T.I32)], T.DATE)
+            return out
+        if isinstance(other.dtype, T.TimestampType):
+            total = sign * (micros + months * 2_592_000_000_000)  # months≈30d only if ts; Spark uses calendar — handled in add_months path
+            if months:
+                out = S.Func("ts_add_months", [other, S.Literal(sign * months, T.I32)], T.TIMESTAMP)
+            else:
+                out = S.BinaryOp("+", other, S.Literal(sign * micros, T.I64), T.TIMESTAMP)
+            return out
+        raise ResolutionError("interval arithmetic requires date/timestamp operand")
+    if e.op in ("=", "!=", "<", "<=", ">", ">=", "<=>"):
+        _coerce_pair(e)
+        e.dtype = T.BOOL
+        return e
+    if e.op in ("+", "-", "*", "/", "%", "div"):
+        # date +- int => date_add
+        if e.op in ("+", "-") and isinstance(lt, T.DateType) and rt is not None and rt.is_integer:
+            name = "date_add" if e.op == "+" else "date_sub"
+            return S.Func(name, [e.left, e.right], T.DATE)
+        if e.op == "-" and isinstance(lt, T.DateType) and isinstance(rt, T.DateType):
+            return S.Func("datediff", [e.left, e.right], T.I32)
+        if isinstance(lt, T.DecimalType) or isinstance(rt, T.DecimalType):
+            return _type_decimal_arith(e)
+        ct = T.common_type(lt, rt)
+        if e.op == "/":
+            # Spark: integer / integer -> double
+            ct = T.F64 if not isinstance(ct, T.DecimalType) else ct
+        if e.op == "div":
+            ct = T.I64
+        _cast_operands(e, ct if e.op != "/" else T.F64 if not isinstance(ct, T.DecimalType) else ct)
+        e.dtype = ct
+        return e
+    raise ResolutionError(f"unknown binary op {e.op}")
+
+
+def _type_decimal_arith(e: S.BinaryOp) -> S.BinaryOp:
+    lt = _as_decimal(e.left)
+    rt = _as_decimal(e.right)
+    if e.op in ("+", "-"):
+        scale = max(lt.scale, rt.scale)
+        prec = max(lt.precision - lt.scale, rt.precision - rt.scale) + scale + 1
+        e.dtype = T.DecimalType(min(38, prec), scale)
+    elif e.op == "*":
+        e.dtype = T.decimal_mul_type(lt, rt)
+    elif e.op in ("/",):
+        e.dtype = T.decimal_div_type(lt, rt)
+    elif e.op in ("%",):
+        scale = max(lt.scale, rt.scale)
+        e.dtype = T.DecimalType(min(38, max(lt.precision, rt.precision)), scale)
+    elif e.op == "div":
+        e.dtype = T.I64
+    # record operand decimal types via Cast insertion when int literals involved
+    if not isinstance(e.left.dtype, T.DecimalType):
+        e.left = S.Cast(e.left, lt, dtype=lt)
+    if not isinstance(e.right.dtype, T.DecimalType):
+        e.right = S.Cast(e.right, rt, dtype=rt)
+    return e
+
+
+def _as_decimal(x: S.Expr) -> T.DecimalType:
+    t = x.dtype
+    if isinstance(t, T.DecimalType):
+        return t
+    if t.is_integer:
+        return T.DecimalType(19, 0)
+    if t.is_float:
+        # float wins: Spark converts decimal to double; approximate by scale-6 decimal
+        return T.DecimalType(30, 6)
+    raise ResolutionError(f"cannot use {t!r} in decimal arithmetic")
+
+
+def _coerce_pair(e: S.BinaryOp):
+    lt, rt = e.left.dtype, e.right.dtype
+    if lt == rt:
+        return
+    # string literal vs date/timestamp: parse literal
+    if isinstance(lt, T.DateType) and isinstance(rt, T.StringType) and isinstance(e.right, S.Literal):
+        from ..sql.parser import _parse_date
+        e.right = S.Literal(_parse_date(e.right.value), T.DATE)
+        return
+    if isinstance(rt, T.DateType) and isinstance(lt, T.StringType) and isinstance(e.left, S.Literal):
+        from ..sql.parser import _parse_date
+        e.left = S.Literal(_parse_date(e.left.value), T.DATE)
+        return
+    ct = T.common_type(lt, rt)
+    _cast_operands(e, ct)
+
+
+def _cast_operands(e: S.BinaryOp, ct: T.DataType):
+    if e.left.dtype != ct:
+        e.left = S.Cast(e.left, ct, dtype=ct)
+    if e.right.dtype != ct:
+        e.right = S.Cast(e.right, ct, dtype=ct)
+
+
+def _coerce_to_bool(e: S.Expr) -> S.Expr:
+    if e is None:
+        return None
+    if isinstance(e.dtype, T.BooleanType) or e.dtype is None:
+        return e
+    return S.Cast(e, T.BOOL, dtype=T.BOOL)
+
+
+def _infer_literal_type(v) -> T.DataType:
+    if v is None:
+        return T.NULL
+    if isinstance(v, bool):
+        return T.BOOL
+    if isinstance(v, int):
+        return T.I32 if -(2 ** 31) <= v < 2 ** 31 else T.I64
+    if isinstance(v, float):
+        return T.F64
+    if isinstance(v, str):
+        return T.STRING
+    return T.NULL
+
+
+def _infer_pytype(values) -> T.DataType:
+    for v in values:
+        if v is None:
+            continue
+        return _infer_literal_type(v)
+    return T.NULL
+
+
+def _expr_name(e: S.Expr, i: int) -> str:
+    if isinstance(e, S.Alias):
+        return e.name
+    if isinstance(e, (S.BoundRef, S.Col)):
+        return e.name
+    if isinstance(e, S.Cast):
+        return _expr_name(e.child, i)
+    if isinstance(e, S.AggFunc):
+        return f"{e.name}({', '.join(_expr_name(a, i) for a in e.args) or '1'})"
+    if isinstance(e, S.Func):
+        return f"{e.name}({', '.join(_expr_name(a, i) for a in e.args)})"
+    if isinstance(e, S.Literal):
+        return str(e.value)
+    return f"col{i}"
+
+
+def _normalize_agg_name(name: str) -> str:
+    name = name.lower()
+    return {"mean": "avg", "first_value": "first", "last_value": "last",
+            "some": "any", "bool_or": "any", "every": "bool_and",
+            "array_agg": "collect_list", "stddev": "stddev_samp",
+            "variance": "var_samp", "approx_percentile": "percentile_approx"}.get(name, name)
+
+
+def _expr_equal_unbound(a: S.Expr, b: S.Expr) -> bool:
+    """Structural equality between two *unbound* expressions (pre-resolution),
+    used to match projection exprs against group-by exprs."""
+    if isinstance(a, S.Alias):
+        return _expr_equal_unbound(a.child, b)
+    if isinstance(b, S.Alias):
+        return _expr_equal_unbound(a, b.child)
+    if type(a) is not type(b):
+        return False
+    if isinstance(a, S.Col):
+        return a.name.lower() == b.name.lower() and (
+            a.qualifier is None or b.qualifier is None
+            or a.qualifier.lower() == b.qualifier.lower())
+    if isinstance(a, S.Literal):
+        return a.value == b.value
+    if isinstance(a, S.BinaryOp) and a.op != b.op:
+        return False
+    if isinstance(a, S.UnaryOp) and a.op != b.op:
+        return False
+    if isinstance(a, S.Func) and a.name.lower() != b.name.lower():
+        return False
+    ca, cb = a.children(), b.children()
+    if len(ca) != len(cb):
+        return False
+    return all(_expr_equal_unbound(x, y) for x, y in zip(ca, cb))
+
+
+def _plan_qualifier(p: S.Plan) -> Optional[str]:
+    if isinstance(p, S.SubqueryAlias):
+        return p.alias
+    if isinstance(p, S.Read):
+        return p.table.split(".")[-1]
+    if isinstance(p, (S.Filter, S.Limit, S.Sort, S.Distinct)):
+        return _plan_qualifier(p.input)
+    return None

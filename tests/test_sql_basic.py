@@ -1,0 +1,217 @@
+"""End-to-end SQL tests on the CPU path (parser -> resolver -> executor)."""
+import math
+
+import pytest
+
+import sail_amd
+from sail_amd.engine import types as T
+
+
+@pytest.fixture()
+def s():
+    s = sail_amd.SessionContext(device="cpu")
+    s.create_dataframe(
+        {"a": [1, 2, 3, 4, 5], "b": [10.0, 20.0, 30.0, 40.0, 50.0],
+         "c": ["x", "y", "x", "z", None], "d": [None, 2, None, 4, 5]},
+        schema={"a": T.I32, "b": T.F64, "c": T.STRING, "d": T.I32},
+        name="t")
+    return s
+
+
+def test_select_where(s):
+    assert s.sql("SELECT a FROM t WHERE a > 2").collect() == [(3,), (4,), (5,)]
+
+
+def test_arith_and_alias(s):
+    rows = s.sql("SELECT a * 2 + 1 AS x FROM t WHERE a <= 2").collect()
+    assert rows == [(3,), (5,)]
+
+
+def test_division_is_double(s):
+    rows = s.sql("SELECT a / 2 FROM t WHERE a = 3").collect()
+    assert rows == [(1.5,)]
+
+
+def test_group_by(s):
+    rows = s.sql("SELECT c, sum(a) AS sa, count(*) AS n FROM t GROUP BY c ORDER BY c NULLS LAST").collect()
+    assert rows == [("x", 4, 2), ("y", 2, 1), ("z", 4, 1), (None, 5, 1)]
+
+
+def test_global_agg(s):
+    rows = s.sql("SELECT sum(a), min(b), max(b), avg(a), count(d) FROM t").collect()
+    assert rows == [(15, 10.0, 50.0, 3.0, 3)]
+
+
+def test_count_distinct(s):
+    rows = s.sql("SELECT count(DISTINCT c) FROM t").collect()
+    assert rows == [(3,)]
+
+
+def test_order_limit_offset(s):
+    rows = s.sql("SELECT a FROM t ORDER BY a DESC LIMIT 2").collect()
+    assert rows == [(5,), (4,)]
+    rows = s.sql("SELECT a FROM t ORDER BY a LIMIT 2 OFFSET 1").collect()
+    assert rows == [(2,), (3,)]
+
+
+def test_order_by_hidden_column(s):
+    rows = s.sql("SELECT a FROM t ORDER BY b DESC LIMIT 1").collect()
+    assert rows == [(5,)]
+
+
+def test_nulls_in_filter(s):
+    # NULL comparisons are not true
+    rows = s.sql("SELECT a FROM t WHERE d > 0").collect()
+    assert rows == [(2,), (4,), (5,)]
+
+
+def test_is_null(s):
+    assert s.sql("SELECT a FROM t WHERE d IS NULL").collect() == [(1,), (3,)]
+    assert s.sql("SELECT count(*) FROM t WHERE c IS NOT NULL").collect() == [(4,)]
+
+
+def test_case_when(s):
+    rows = s.sql("SELECT CASE WHEN a < 3 THEN 'lo' ELSE 'hi' END AS k FROM t ORDER BY a").collect()
+    assert rows == [("lo",), ("lo",), ("hi",), ("hi",), ("hi",)]
+
+
+def test_in_list(s):
+    assert s.sql("SELECT a FROM t WHERE a IN (1, 3, 9)").collect() == [(1,), (3,)]
+    assert s.sql("SELECT a FROM t WHERE c IN ('x')").collect() == [(1,), (3,)]
+
+
+def test_between(s):
+    assert s.sql("SELECT a FROM t WHERE a BETWEEN 2 AND 4").collect() == [(2,), (3,), (4,)]
+    assert s.sql("SELECT a FROM t WHERE a NOT BETWEEN 2 AND 4").collect() == [(1,), (5,)]
+
+
+def test_like(s):
+    s2 = sail_amd.SessionContext(device="cpu")
+    s2.create_dataframe({"v": ["apple", "banana", "grape", "pineapple"]}, name="f")
+    assert s2.sql("SELECT v FROM f WHERE v LIKE '%apple%'").collect() == [("apple",), ("pineapple",)]
+    assert s2.sql("SELECT v FROM f WHERE v LIKE 'gra_e'").collect() == [("grape",)]
+    assert s2.sql("SELECT v FROM f WHERE v NOT LIKE '%a%'").collect() == []
+
+
+def test_joins_inner_left(s):
+    s.create_dataframe({"a": [1, 2, 6], "z": ["p", "q", "r"]}, name="u")
+    rows = s.sql("SELECT t.a, u.z FROM t JOIN u ON t.a = u.a ORDER BY t.a").collect()
+    assert rows == [(1, "p"), (2, "q")]
+    rows = s.sql("SELECT t.a, u.z FROM t LEFT JOIN u ON t.a = u.a ORDER BY t.a").collect()
+    assert rows == [(1, "p"), (2, "q"), (3, None), (4, None), (5, None)]
+
+
+def test_join_semi_anti(s):
+    s.create_dataframe({"a": [1, 2, 6]}, name="u")
+    assert s.sql("SELECT a FROM t LEFT SEMI JOIN u USING (a) ORDER BY a").collect() == [(1,), (2,)]
+    assert s.sql("SELECT a FROM t LEFT ANTI JOIN u USING (a) ORDER BY a").collect() == [(3,), (4,), (5,)]
+
+
+def test_cross_join_count(s):
+    assert s.sql("SELECT count(*) FROM t, t t2").collect() == [(25,)]
+
+
+def test_union(s):
+    rows = s.sql("SELECT a FROM t WHERE a <= 2 UNION ALL SELECT a FROM t WHERE a >= 4 ORDER BY a").collect()
+    assert rows == [(1,), (2,), (4,), (5,)]
+    rows = s.sql("SELECT 1 AS x UNION SELECT 1 UNION SELECT 2 ORDER BY x").collect()
+    assert rows == [(1,), (2,)]
+
+
+def test_distinct(s):
+    assert s.sql("SELECT DISTINCT c FROM t WHERE c IS NOT NULL ORDER BY c").collect() == [
+        ("x",), ("y",), ("z",)]
+
+
+def test_cte(s):
+    rows = s.sql("WITH big AS (SELECT a FROM t WHERE a > 3) SELECT count(*) FROM big").collect()
+    assert rows == [(2,)]
+
+
+def test_subquery_in_from(s):
+    rows = s.sql("SELECT x.a2 FROM (SELECT a * 2 AS a2 FROM t) x WHERE x.a2 > 6 ORDER BY 1").collect()
+    assert rows == [(8,), (10,)]
+
+
+def test_scalar_subquery_uncorrelated(s):
+    rows = s.sql("SELECT a FROM t WHERE a > (SELECT avg(a) FROM t) ORDER BY a").collect()
+    assert rows == [(4,), (5,)]
+
+
+def test_having(s):
+    rows = s.sql("SELECT c, count(*) AS n FROM t GROUP BY c HAVING count(*) > 1").collect()
+    assert rows == [("x", 2)]
+
+
+def test_group_by_ordinal_and_alias(s):
+    rows = s.sql("SELECT c AS k, sum(a) FROM t WHERE c IS NOT NULL GROUP BY 1 ORDER BY 1").collect()
+    assert rows == [("x", 4), ("y", 2), ("z", 4)]
+    rows = s.sql("SELECT c AS k, sum(a) FROM t WHERE c IS NOT NULL GROUP BY k ORDER BY k").collect()
+    assert rows == [("x", 4), ("y", 2), ("z", 4)]
+
+
+def test_agg_expression_over_groups(s):
+    rows = s.sql("SELECT c, sum(a) / count(*) AS r FROM t WHERE c = 'x' GROUP BY c").collect()
+    assert rows == [("x", 2.0)]
+
+
+def test_dates():
+    s = sail_amd.SessionContext(device="cpu")
+    s.create_dataframe({"d": ["2024-01-15", "2023-06-30", "2024-12-31"]},
+                       schema={"d": T.DATE}, name="dt")
+    assert s.sql("SELECT year(d) FROM dt ORDER BY d").collect() == [(2023,), (2024,), (2024,)]
+    assert s.sql("SELECT month(d), day(d) FROM dt WHERE year(d) = 2023").collect() == [(6, 30)]
+    rows = s.sql("SELECT count(*) FROM dt WHERE d >= DATE '2024-01-01'").collect()
+    assert rows == [(2,)]
+    rows = s.sql("SELECT count(*) FROM dt WHERE d < DATE '1994-01-01' + INTERVAL '1' YEAR").collect()
+    assert rows == [(0,)]
+
+
+def test_decimal_literals_and_agg():
+    s = sail_amd.SessionContext(device="cpu")
+    s.create_dataframe({"p": [10.25, 3.75, 1.00]},
+                       schema={"p": T.DecimalType(12, 2)}, name="d")
+    assert s.sql("SELECT sum(p) FROM d").collect() == [(15.0,)]
+    assert s.sql("SELECT count(*) FROM d WHERE p > 3.50").collect() == [(2,)]
+    # decimal * decimal keeps exactness
+    rows = s.sql("SELECT sum(p * 2.00) FROM d").collect()
+    assert rows == [(30.0,)]
+
+
+def test_values_clause(s):
+    rows = s.sql("SELECT col1, col2 FROM (VALUES (1, 'a'), (2, 'b')) v ORDER BY col1").collect()
+    assert rows == [(1, "a"), (2, "b")]
+
+
+def test_select_without_from(s):
+    assert s.sql("SELECT 1 + 1").collect() == [(2,)]
+
+
+def test_functions(s):
+    assert s.sql("SELECT abs(-5), round(2.567, 2), floor(2.9), ceil(2.1)").collect() == [
+        (5, 2.57, 2, 3)]
+    assert s.sql("SELECT upper('ab'), length('abc'), substring('hello', 2, 3)").collect() == [
+        ("AB", 3, "ell")]
+    assert s.sql("SELECT coalesce(NULL, 5)").collect() == [(5,)]
+
+
+def test_range_table(s):
+    assert s.sql("SELECT count(*), sum(id) FROM range(10)").collect() == [(10, 45)]
+
+
+def test_create_view(s):
+    s.sql("CREATE OR REPLACE TEMP VIEW v AS SELECT a FROM t WHERE a > 3")
+    # view resolution happens at query time
+    assert s.sql("SELECT count(*) FROM v").collect() == [(2,)]
+
+
+def test_explain(s):
+    out = s.sql("EXPLAIN SELECT a FROM t WHERE a > 1").collect()
+    assert "Filter" in out[0][0]
+
+
+def test_window_row_number(s):
+    rows = s.sql(
+        "SELECT a, row_number() OVER (PARTITION BY c ORDER BY a) AS rn FROM t WHERE c = 'x' ORDER BY a"
+    ).collect()
+    assert rows == [(1, 1), (3, 2)]

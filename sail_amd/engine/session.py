@@ -116,6 +116,13 @@ class Catalog:
             return t
         return None
 
+    def table_rows(self, name: str) -> Optional[int]:
+        k = self._key(name)
+        t = self._tables.get(k)
+        if t is not None:
+            return t.num_rows
+        return None
+
     def list_tables(self) -> List[str]:
         return sorted(set(self._tables) | set(self._views) | set(self._providers))
 
@@ -149,7 +156,9 @@ class SessionContext:
     def optimize(self, plan: S.Plan) -> S.Plan:
         from ..plan.optimizer import optimize
 
-        return optimize(plan)
+        reorder = self.conf.get("sail.optimizer.enable_join_reorder", "true") == "true"
+        return optimize(plan, enable_join_reorder=reorder,
+                        stats=self.catalog.table_rows)
 
     def plan_sql(self, sql: str) -> S.Plan:
         return self.optimize(self.resolve(self.parse(sql)))

@@ -27,11 +27,10 @@ from .rules.prune import prune_columns
 from .rules.join_order import reorder_joins
 
 
-def optimize(plan: S.Plan, enable_join_reorder: bool = True) -> S.Plan:
+def optimize(plan: S.Plan, enable_join_reorder: bool = True, stats=None) -> S.Plan:
     plan = decorrelate(plan)
     plan = pushdown_filters(plan)
     if enable_join_reorder:
-        plan = reorder_joins(plan)
-        plan = pushdown_filters(plan)
+        plan = reorder_joins(plan, stats)
     plan = prune_columns(plan)
     return plan

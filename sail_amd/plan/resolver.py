@@ -273,6 +273,11 @@ class Resolver:
             group exprs -> refs to group keys."""
             if isinstance(e, S.Alias):
                 return S.Alias(bind_agg(e.child), e.name, None)
+            if isinstance(e, (S.ScalarSubquery, S.Exists)):
+                return self._expr(e, scope)
+            if isinstance(e, S.InSubquery):
+                sub = self._plan(e.plan, scope)
+                return S.InSubquery(bind_agg(e.child), sub, e.negated, T.BOOL)
             # whole-expression matches a group key?
             for gi, (ge, be) in enumerate(zip(group_exprs, bound_groups)):
                 if _expr_equal_unbound(e, ge):
@@ -310,7 +315,7 @@ class Resolver:
                             f"column {e.name} must appear in GROUP BY or inside an aggregate")
                     raise ResolutionError(f"cannot resolve column {e.name}")
                 return e
-            return e.with_children([bind_agg(c) for c in ch])
+            return self._type_expr(e.with_children([bind_agg(c) for c in ch]))
 
         final_exprs = [self._type_expr(bind_agg(e)) for e in projections]
         having_bound = None
@@ -333,10 +338,8 @@ class Resolver:
     def _p_Join(self, p: S.Join, outer):
         left = self._plan(p.left, outer)
         right = self._plan(p.right, outer)
-        lq = _plan_qualifier(left)
-        rq = _plan_qualifier(right)
-        lfields = [Field(n, t, lq) for n, t in left.schema]
-        rfields = [Field(n, t, rq) for n, t in right.schema]
+        lfields = _scope_fields(left)
+        rfields = _scope_fields(right)
         scope = Scope(lfields + rfields, outer)
 
         using = p.using
@@ -449,23 +452,7 @@ class Resolver:
     # expressions
     # =====================================================================
     def _child_scope(self, child: S.Plan, outer) -> Scope:
-        q = _plan_qualifier(child)
-        if isinstance(child, S.Join):
-            lq = _plan_qualifier(child.left)
-            rq = _plan_qualifier(child.right)
-            nleft = len(child.left.schema)
-            fields = []
-            for i, (n, t) in enumerate(child.schema):
-                if child.how in ("semi", "anti"):
-                    fields.append(Field(n, t, lq))
-                elif child.how in ("rightsemi", "rightanti"):
-                    fields.append(Field(n, t, rq))
-                elif i < nleft:
-                    fields.append(Field(n, t, lq))
-                else:
-                    fields.append(Field(n, t, rq))
-            return Scope(fields, outer)
-        return Scope([Field(n, t, q) for n, t in child.schema], outer)
+        return Scope(_scope_fields(child), outer)
 
     def _expand_star(self, e: S.Expr, scope: Scope) -> List[S.Expr]:
         if isinstance(e, S.Star):
@@ -816,6 +803,30 @@ def _expr_equal_unbound(a: S.Expr, b: S.Expr) -> bool:
     if len(ca) != len(cb):
         return False
     return all(_expr_equal_unbound(x, y) for x, y in zip(ca, cb))
+
+
+def _scope_fields(p: S.Plan) -> List[Field]:
+    """Per-column qualifiers, preserved through nested join trees so that
+    `n1.n_nationkey` resolves inside a 6-way comma join (q7/q8/q21)."""
+    if isinstance(p, S.Join):
+        if p.how in ("semi", "anti"):
+            return _scope_fields(p.left)
+        if p.how in ("rightsemi", "rightanti"):
+            return _scope_fields(p.right)
+        lf = _scope_fields(p.left)
+        rf = _scope_fields(p.right)
+        if p.using:
+            used = {c.lower() for c in p.using}
+            rf = [f for f in rf if f.name.lower() not in used]
+        return lf + rf
+    if isinstance(p, S.SubqueryAlias):
+        return [Field(n, t, p.alias) for n, t in p.schema]
+    if isinstance(p, (S.Filter, S.Limit, S.Sort, S.Distinct)):
+        inner = _scope_fields(p.input)
+        if len(inner) == len(p.schema):
+            return inner
+    q = _plan_qualifier(p)
+    return [Field(n, t, q) for n, t in p.schema]
 
 
 def _plan_qualifier(p: S.Plan) -> Optional[str]:

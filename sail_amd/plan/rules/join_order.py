@@ -1,6 +1,253 @@
-"""Join reordering (stub — implemented in a later pass)."""
+"""Join reordering.
+
+Flattens maximal inner/cross-join trees into (leaves, conjuncts), estimates
+leaf cardinalities with selectivity heuristics, and rebuilds a left-deep tree
+greedily: start from the smallest filtered leaf, repeatedly join the
+edge-connected leaf minimizing the estimated intermediate size. The greedy
+counterpart of the reference's DP join reorderer
+(ref: crates/sail-physical-optimizer/src/join_reorder/mod.rs:31 — graph
+builder + cardinality estimator + cost model; DP upgrade is planned).
+
+Build-side note: the executor builds the hash table on the RIGHT input, so
+the accumulated (large) side stays on the left as probe.
+"""
+from __future__ import annotations
+
+from typing import Callable, Dict, List, Optional, Set, Tuple
+
+from ...engine import types as T
 from .. import spec as S
+from .util import conjoin, expr_refs, remap_expr, split_conjuncts
 
 
-def reorder_joins(plan: S.Plan) -> S.Plan:
+def reorder_joins(plan: S.Plan, stats: Optional[Callable[[str], Optional[int]]] = None) -> S.Plan:
+    return _walk(plan, stats)
+
+
+def _walk(plan: S.Plan, stats) -> S.Plan:
+    # rewrite subquery plans too
+    for e in _exprs(plan):
+        _walk_expr(e, stats)
+    if isinstance(plan, S.Join) and plan.how in ("inner", "cross"):
+        return _reorder_tree(plan, stats)
+    for attr in ("input", "left", "right"):
+        child = getattr(plan, attr, None)
+        if isinstance(child, S.Plan):
+            setattr(plan, attr, _walk(child, stats))
     return plan
+
+
+def _exprs(p: S.Plan):
+    if isinstance(p, S.Project):
+        return p.exprs
+    if isinstance(p, S.Filter):
+        return [p.condition]
+    if isinstance(p, S.Join) and p.on is not None:
+        return [p.on]
+    if isinstance(p, S.Aggregate):
+        return list(p.group_by) + list(p.aggs)
+    return []
+
+
+def _walk_expr(e: S.Expr, stats):
+    if isinstance(e, (S.ScalarSubquery, S.Exists)):
+        e.plan = _walk(e.plan, stats)
+        return
+    if isinstance(e, S.InSubquery):
+        e.plan = _walk(e.plan, stats)
+    for c in e.children():
+        _walk_expr(c, stats)
+
+
+# ---------------------------------------------------------------------------
+
+def _flatten(plan: S.Plan, leaves: List[S.Plan], conds: List[S.Expr], offset: int) -> int:
+    """Collect inner-join leaves in order; returns total width. Conjunct
+    indices are valid in the concatenation of leaf schemas (schema concat is
+    associative for inner joins)."""
+    if isinstance(plan, S.Join) and plan.how in ("inner", "cross"):
+        lw = _flatten(plan.left, leaves, conds, offset)
+        rw = _flatten(plan.right, leaves, conds, offset + lw)
+        if plan.on is not None:
+            for c in split_conjuncts(plan.on):
+                conds.append(remap_expr(c, {i: i + offset for i in expr_refs(c)}))
+        return lw + rw
+    leaves.append(plan)
+    return len(plan.schema)
+
+
+def _estimate(plan: S.Plan, stats) -> Tuple[float, float]:
+    """(estimated rows, base rows) for a leaf subtree."""
+    if isinstance(plan, S.Read):
+        n = None
+        if stats is not None:
+            n = stats(plan.table)
+        base = float(n) if n else 1000.0
+        return base, base
+    if isinstance(plan, (S.SubqueryAlias, S.Project, S.Limit, S.Sort)):
+        child = plan.input
+        est, base = _estimate(child, stats)
+        if isinstance(plan, S.Limit) and plan.n is not None:
+            est = min(est, float(plan.n))
+        return est, base
+    if isinstance(plan, S.Filter):
+        est, base = _estimate(plan.input, stats)
+        sel = 1.0
+        for c in split_conjuncts(plan.condition):
+            sel *= _selectivity(c)
+        return max(est * sel, 1.0), base
+    if isinstance(plan, S.Aggregate):
+        est, base = _estimate(plan.input, stats)
+        if not plan.group_by:
+            return 1.0, 1.0
+        return max(est / 10.0, 1.0), base
+    if isinstance(plan, S.Join):
+        le, lb = _estimate(plan.left, stats)
+        re_, rb = _estimate(plan.right, stats)
+        if plan.how in ("semi", "anti"):
+            return max(le * 0.5, 1.0), lb
+        denom = max(lb, rb, 1.0)
+        return max(le * re_ / denom, 1.0), max(lb, rb)
+    if isinstance(plan, (S.LocalRelation, S.Range)):
+        n = len(next(iter(plan.data.values()))) if isinstance(plan, S.LocalRelation) and plan.data else 10
+        if isinstance(plan, S.Range):
+            n = max((plan.end - plan.start) // max(plan.step, 1), 1)
+        return float(n), float(n)
+    if isinstance(plan, S.SetOp):
+        le, lb = _estimate(plan.left, stats)
+        re_, rb = _estimate(plan.right, stats)
+        return le + re_, lb + rb
+    if isinstance(plan, S.Distinct):
+        est, base = _estimate(plan.input, stats)
+        return max(est / 2.0, 1.0), base
+    return 1000.0, 1000.0
+
+
+def _selectivity(c: S.Expr) -> float:
+    if isinstance(c, S.BinaryOp):
+        if c.op == "=":
+            return 0.05
+        if c.op in ("<", "<=", ">", ">="):
+            return 0.33
+        if c.op == "!=":
+            return 0.9
+        if c.op == "or":
+            return min(0.9, _selectivity(c.left) + _selectivity(c.right))
+        if c.op == "and":
+            return _selectivity(c.left) * _selectivity(c.right)
+    if isinstance(c, S.Between):
+        return 0.15
+    if isinstance(c, S.InList):
+        return min(0.9, 0.05 * max(len(c.values), 1))
+    if isinstance(c, S.Like):
+        return 0.1 if not c.negated else 0.9
+    if isinstance(c, S.UnaryOp) and c.op in ("isnull",):
+        return 0.05
+    return 0.5
+
+
+def _reorder_tree(root: S.Join, stats) -> S.Plan:
+    leaves: List[S.Plan] = []
+    conds: List[S.Expr] = []
+    _flatten(root, leaves, conds, 0)
+    n = len(leaves)
+    if n > 12:
+        # beyond reorder budget: keep original shape but recurse into leaves
+        for i, lf in enumerate(leaves):
+            leaves[i] = _walk(lf, stats)
+        return root
+    leaves = [_walk(lf, stats) for lf in leaves]
+
+    # old concatenated offsets
+    offs = [0]
+    for lf in leaves:
+        offs.append(offs[-1] + len(lf.schema))
+
+    def leaf_of(idx: int) -> int:
+        for li in range(n):
+            if offs[li] <= idx < offs[li + 1]:
+                return li
+        raise IndexError(idx)
+
+    ests = [_estimate(lf, stats) for lf in leaves]
+
+    # edges: conjuncts joining exactly two leaves with an equality
+    edges: Dict[int, Set[int]] = {i: set() for i in range(n)}
+    for c in conds:
+        ls = {leaf_of(i) for i in expr_refs(c)}
+        if len(ls) == 2:
+            a, b = sorted(ls)
+            edges[a].add(b)
+            edges[b].add(a)
+
+    # greedy order
+    order = [min(range(n), key=lambda i: ests[i][0])]
+    placed = {order[0]}
+    cur_est = ests[order[0]][0]
+    while len(placed) < n:
+        candidates = [i for i in range(n) if i not in placed
+                      and any(j in placed for j in edges[i])]
+        disconnected = False
+        if not candidates:
+            candidates = [i for i in range(n) if i not in placed]
+            disconnected = True
+        best, best_cost = None, None
+        for i in candidates:
+            est_i, base_i = ests[i]
+            if disconnected:
+                cost = cur_est * est_i
+            else:
+                cost = cur_est * est_i / max(base_i, 1.0)
+            if best_cost is None or cost < best_cost:
+                best, best_cost = i, cost
+        order.append(best)
+        placed.add(best)
+        cur_est = max(best_cost, 1.0)
+
+    # rebuild left-deep tree in `order`, remapping conjunct indices
+    new_off: Dict[int, int] = {}
+    pos = 0
+    for li in order:
+        new_off[li] = pos
+        pos += len(leaves[li].schema)
+
+    def remap_cond(c: S.Expr) -> S.Expr:
+        mapping = {}
+        for i in expr_refs(c):
+            li = leaf_of(i)
+            mapping[i] = new_off[li] + (i - offs[li])
+        return remap_expr(c, mapping)
+
+    remaining = [(c, {leaf_of(i) for i in expr_refs(c)}, remap_cond(c)) for c in conds]
+    cur = leaves[order[0]]
+    avail = {order[0]}
+    avail_width = len(cur.schema)
+    for li in order[1:]:
+        right = leaves[li]
+        avail.add(li)
+        place_now = [rc for c, ls, rc in remaining if ls and ls.issubset(avail)]
+        remaining = [(c, ls, rc) for c, ls, rc in remaining if not (ls and ls.issubset(avail))]
+        on = conjoin(place_now)
+        j = S.Join(left=cur, right=right, how="inner" if on is not None else "cross",
+                   on=on, using=None)
+        j.schema = list(cur.schema) + list(right.schema)
+        cur = j
+        avail_width += len(right.schema)
+    # degenerate conjuncts (no refs / single-leaf leftovers) become a filter
+    leftover = [rc for c, ls, rc in remaining]
+    if leftover:
+        f = S.Filter(input=cur, condition=conjoin(leftover))
+        f.schema = cur.schema
+        cur = f
+
+    # restore original column order with a projection
+    perm = []
+    for li in range(n):
+        for k in range(len(leaves[li].schema)):
+            perm.append(new_off[li] + k)
+    if perm != list(range(len(perm))):
+        pr = S.Project(input=cur, exprs=[
+            S.BoundRef(p, cur.schema[p][0], cur.schema[p][1]) for p in perm])
+        pr.schema = [cur.schema[p] for p in perm]
+        cur = pr
+    return cur

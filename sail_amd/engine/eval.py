@@ -372,9 +372,37 @@ def _cmp_data(a: Column, b: Column, op: str) -> torch.Tensor:
     raise EvalError(op)
 
 
+def _literal_string_keys(col: StringColumn, lits) -> torch.Tensor:
+    """Literal keys consistent with joins.raw_string_key for `col`."""
+    from .eval_keys import literal_keys_like
+
+    lens = col.offsets[1:] - col.offsets[:-1]
+    force_hash = (int(lens.max().item()) if len(col) else 0) > 7
+    return literal_keys_like(list(lits), force_hash, col.device)
+
+
 def _string_cmp(a: Column, b: Column, op: str) -> torch.Tensor:
     # dict-encoded vs scalar-dict single-value fast path
     if isinstance(a, StringColumn) and isinstance(b, StringColumn):
+        a_short = (len(a) == 0 or int((a.offsets[1:] - a.offsets[:-1]).max().item()) <= 7) \
+            if not a.is_dict else False
+        if op in ("=", "!=") and not a.is_dict and (a.is_cuda or a_short) \
+                and b.is_dict and b.dict_size == 1:
+            from .joins import raw_string_key
+
+            target = b.dict_values()[0]
+            keys = raw_string_key(a)
+            litk = _literal_string_keys(a, [target])
+            m = keys == litk[0]
+            return m if op == "=" else ~m
+        if op in ("=", "!=") and not a.is_dict and not b.is_dict and (a.is_cuda or a_short):
+            from .joins import raw_string_key
+
+            # column-vs-column equality via keys (exact <=7B, hashed beyond)
+            ka, kb = raw_string_key(a), raw_string_key(b)
+            if len(a) == len(b):
+                m = ka == kb
+                return m if op == "=" else ~m
         if a.is_dict and b.is_dict and len(b) > 0:
             if b.dict_size == 1:
                 # b is a broadcast literal
@@ -411,6 +439,14 @@ def _string_isin(c: StringColumn, lits: List[str]) -> torch.Tensor:
         vals = c.dict_values()
         hit = torch.tensor([v in lits for v in vals], dtype=torch.bool, device=c.device)
         return hit[c.codes.long().clamp_min(0)] & (c.codes >= 0)
+    from .joins import raw_string_key
+
+    lens = c.offsets[1:] - c.offsets[:-1]
+    max_len = int(lens.max().item()) if len(c) else 0
+    if c.is_cuda or max_len <= 7:
+        keys = raw_string_key(c)
+        litk = _literal_string_keys(c, lits)
+        return torch.isin(keys, litk)
     vals = c.to_pylist()
     return torch.tensor([v in lits if v is not None else False for v in vals],
                         dtype=torch.bool, device=c.device)

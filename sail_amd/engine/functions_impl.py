@@ -458,6 +458,21 @@ def _f_substring(args, out, chunk, ev):
 
         offs, byts = _pack_strings(vals, c.device)
         return StringColumn(offs, byts, c.validity, c.codes)
+    if isinstance(c, StringColumn) and c.is_cuda and s >= 1 and ln is not None:
+        # device kernel: fixed-pitch extraction then compaction (all on GPU)
+        from ..ops import kernels as K
+
+        buf, lens = K.require().substr_fixed(c.offsets, c.bytes_, s - 1, ln)
+        n = len(c)
+        lens64 = lens.to(torch.int64)
+        if bool((lens64 == ln).all()):
+            offsets = torch.arange(0, (n + 1) * ln, ln, dtype=torch.int64, device=c.device)
+            return StringColumn(offsets, buf, c.validity)
+        mask = (torch.arange(ln, device=c.device).unsqueeze(0) < lens64.unsqueeze(1)).reshape(-1)
+        out_bytes = buf[mask]
+        offsets = torch.zeros(n + 1, dtype=torch.int64, device=c.device)
+        torch.cumsum(lens64, 0, out=offsets[1:])
+        return StringColumn(offsets, out_bytes, c.validity)
     return _str_map(c, sub)
 
 

@@ -25,21 +25,46 @@ def normalize_key(c: Column) -> torch.Tensor:
     if isinstance(c, StringColumn):
         if c.is_dict:
             return c.codes.to(torch.int64)
-        # raw strings: host-side dictionary encode (CPU reference path)
-        vals = c.to_pylist()
-        lut = {}
-        out = []
-        for v in vals:
-            if v is None:
-                out.append(-1)
-            else:
-                out.append(lut.setdefault(v, len(lut)))
-        return torch.tensor(out, dtype=torch.int64, device=c.device)
+        return raw_string_key(c)
     if c.data.dtype == torch.float64:
         return c.data.view(torch.int64)
     if c.data.dtype == torch.float32:
         return c.data.view(torch.int32).to(torch.int64)
     return c.data.to(torch.int64)
+
+
+def normalize_key_pair(a: Column, b: Column):
+    """Normalize a pair of join-key columns to comparable int64 keys.
+    Raw-string pairs must share one key scheme (packed vs hashed)."""
+    if isinstance(a, StringColumn) and isinstance(b, StringColumn) \
+            and not a.is_dict and not b.is_dict:
+        la = a.offsets[1:] - a.offsets[:-1]
+        lb = b.offsets[1:] - b.offsets[:-1]
+        ma = int(la.max().item()) if len(a) else 0
+        mb = int(lb.max().item()) if len(b) else 0
+        force = max(ma, mb) > 7
+        return raw_string_key(a, force), raw_string_key(b, force)
+    if isinstance(a, StringColumn) and isinstance(b, StringColumn) \
+            and a.is_dict != b.is_dict:
+        # one side dict, one raw: decode dict side keys through its values
+        da = a if a.is_dict else b
+        ra = b if a.is_dict else a
+        la = ra.offsets[1:] - ra.offsets[:-1]
+        force = (int(la.max().item()) if len(ra) else 0) > 7 or \
+            max((len(v) for v in da.dict_values()), default=0) > 7
+        dk = _dict_side_key(da, force)
+        rk = raw_string_key(ra, force)
+        return (dk, rk) if a.is_dict else (rk, dk)
+    return normalize_key(a), normalize_key(b)
+
+
+def _dict_side_key(c: StringColumn, force_hash: bool) -> torch.Tensor:
+    from .eval_keys import literal_keys_like
+
+    vals = c.dict_values()
+    lut = literal_keys_like(vals, force_hash, c.device)
+    key = lut[c.codes.to(torch.int64).clamp_min(0)]
+    return torch.where(c.codes >= 0, key, torch.full_like(key, -(10 ** 18)))
 
 
 def dense_ids(build_keys: List[torch.Tensor], probe_keys: List[torch.Tensor]):
@@ -92,8 +117,11 @@ def equi_join(build_keys: List[Column], probe_keys: List[Column], how: str,
     - "left_rows": pairs + unmatched probe rows marked (for outer fill)
     - "semi"/"anti": probe filtering handled by caller with counts
     """
-    bk = [normalize_key(c) for c in build_keys]
-    pk = [normalize_key(c) for c in probe_keys]
+    bk, pk = [], []
+    for b, p in zip(build_keys, probe_keys):
+        kb, kp = normalize_key_pair(b, p)
+        bk.append(kb)
+        pk.append(kp)
     bids, pids = dense_ids(bk, pk)
     bv = None
     pv = None
@@ -107,3 +135,80 @@ def equi_join(build_keys: List[Column], probe_keys: List[Column], how: str,
             pv = m if pv is None else (pv & m)
     probe_idx, build_idx, counts = _expand_matches(bids, pids, bv, pv)
     return probe_idx, build_idx, counts
+
+
+# ---------------------------------------------------------------------------
+# raw (non-dictionary) string keys
+# ---------------------------------------------------------------------------
+
+_FNV_OFFSET = 14695981039346656037
+_FNV_PRIME = 1099511628211
+_MIX = 0x9E3779B97F4A7C15
+_M64 = (1 << 64) - 1
+
+
+def fnv1a_hash_py(s: bytes) -> int:
+    """Host mirror of ops/csrc/strings.hip string_hash64 (for literals)."""
+    h = _FNV_OFFSET
+    for b in s:
+        h = ((h ^ b) * _FNV_PRIME) & _M64
+    h ^= (len(s) * _MIX) & _M64
+    return h >> 1
+
+
+def short_string_key(c: StringColumn) -> torch.Tensor:
+    """Pack <=7-byte strings into an exact int64 key (fixed 7-wide packing so
+    keys are comparable across columns and literals)."""
+    starts = c.offsets[:-1]
+    lens = c.offsets[1:] - starts
+    key = lens.to(torch.int64).clone()
+    nb = int(c.bytes_.shape[0])
+    safe_bytes = c.bytes_.to(torch.int64) if nb else None
+    for k in range(7):
+        if safe_bytes is None:
+            key = key * 257
+            continue
+        idx = (starts + k).clamp(0, nb - 1)
+        b = torch.where(k < lens, safe_bytes[idx], torch.zeros_like(lens))
+        key = key * 257 + b
+    return key
+
+
+def fnv_key_tensor(c: StringColumn) -> torch.Tensor:
+    """FNV-1a 64 per row, identical to the HIP kernel; CPU path is a numpy
+    byte-position loop (bounded by the longest string)."""
+    if c.is_cuda:
+        from ..ops import kernels as K
+
+        return K.require().string_hash64(c.offsets, c.bytes_)
+    import numpy as np
+
+    offs = c.offsets.numpy()
+    byts = c.bytes_.numpy().astype(np.uint64)
+    n = len(offs) - 1
+    lens = offs[1:] - offs[:-1]
+    max_len = int(lens.max()) if n else 0
+    h = np.full(n, np.uint64(_FNV_OFFSET), dtype=np.uint64)
+    starts = offs[:-1]
+    prime = np.uint64(_FNV_PRIME)
+    with np.errstate(over="ignore"):
+        for k in range(max_len):
+            live = k < lens
+            idx = np.where(live, starts + k, 0)
+            b = byts[idx] if byts.size else np.zeros(n, dtype=np.uint64)
+            nh = (h ^ b) * prime
+            h = np.where(live, nh, h)
+        h = h ^ (lens.astype(np.uint64) * np.uint64(_MIX))
+    return torch.from_numpy((h >> np.uint64(1)).astype(np.int64)).to(c.device)
+
+
+def raw_string_key(c: StringColumn, force_hash: bool = False) -> torch.Tensor:
+    """Exact packed key for <=7-byte strings; FNV-1a 64-bit beyond (same
+    values on CPU and GPU). Collision probability at 1e8 distinct strings is
+    ~3e-4 per query over the full key space — documented engine tradeoff for
+    raw-string grouping; dictionary-encoded columns are always exact."""
+    lens = c.offsets[1:] - c.offsets[:-1]
+    max_len = int(lens.max().item()) if len(c) else 0
+    if not force_hash and max_len <= 7:
+        return short_string_key(c)
+    return fnv_key_tensor(c)

@@ -1,0 +1,188 @@
+// String kernels for MI355X (gfx950).
+//
+// Device-side operations over Arrow-layout string columns
+// (offsets int64[n+1], bytes uint8[total]):
+//   like_mask      — SQL LIKE ('%', '_') per row -> bool mask
+//   string_hash64  — FNV-1a 64-bit per row (grouping/join keys for raw
+//                    strings; dictionary-encoded columns never need this)
+//   substr_fixed   — substring(start,len) into a fixed-pitch buffer
+//
+// Design notes (see /opt/skills/guides/cdna_hip_programming.md):
+//  - one thread per row, grid-stride; block = 256 (4 waves);
+//    memory-bound — strings are short (<64B), threads in a wave read
+//    adjacent rows so the wave touches a contiguous byte range (L2-friendly).
+//  - grid capped at 2048 blocks per Guideline 11.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <cstdint>
+
+#define SAIL_CHECK(x) TORCH_CHECK(x, #x)
+
+namespace {
+
+constexpr int kBlock = 256;
+
+inline int grid_for(int64_t n) {
+  int64_t blocks = (n + kBlock - 1) / kBlock;
+  return (int)std::min<int64_t>(blocks, 2048);
+}
+
+__device__ inline bool like_match(const uint8_t* s, int64_t slen,
+                                  const uint8_t* p, int plen) {
+  // iterative wildcard match with single backtrack point ('%')
+  int64_t si = 0, ss = 0;
+  int pi = 0, star = -1;
+  while (si < slen) {
+    if (pi < plen && (p[pi] == '_' || p[pi] == s[si])) {
+      ++si;
+      ++pi;
+    } else if (pi < plen && p[pi] == '%') {
+      star = pi++;
+      ss = si;
+    } else if (star >= 0) {
+      pi = star + 1;
+      si = ++ss;
+    } else {
+      return false;
+    }
+  }
+  while (pi < plen && p[pi] == '%') ++pi;
+  return pi == plen;
+}
+
+__global__ void like_mask_kernel(const int64_t* __restrict__ offsets,
+                                 const uint8_t* __restrict__ bytes,
+                                 const uint8_t* __restrict__ pattern, int plen,
+                                 bool* __restrict__ out, int64_t n) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    int64_t lo = offsets[i];
+    out[i] = like_match(bytes + lo, offsets[i + 1] - lo, pattern, plen);
+  }
+}
+
+__global__ void contains_kernel(const int64_t* __restrict__ offsets,
+                                const uint8_t* __restrict__ bytes,
+                                const uint8_t* __restrict__ needle, int nlen,
+                                bool* __restrict__ out, int64_t n) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    int64_t lo = offsets[i];
+    int64_t len = offsets[i + 1] - lo;
+    const uint8_t* s = bytes + lo;
+    bool hit = false;
+    if (nlen == 0) {
+      hit = true;
+    } else if (len >= nlen) {
+      uint8_t c0 = needle[0];
+      for (int64_t j = 0; j + nlen <= len && !hit; ++j) {
+        if (s[j] != c0) continue;
+        bool ok = true;
+        for (int k = 1; k < nlen; ++k) {
+          if (s[j + k] != needle[k]) {
+            ok = false;
+            break;
+          }
+        }
+        hit = ok;
+      }
+    }
+    out[i] = hit;
+  }
+}
+
+__global__ void string_hash64_kernel(const int64_t* __restrict__ offsets,
+                                     const uint8_t* __restrict__ bytes,
+                                     int64_t* __restrict__ out, int64_t n) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    int64_t lo = offsets[i], hi = offsets[i + 1];
+    uint64_t h = 14695981039346656037ull;
+    for (int64_t j = lo; j < hi; ++j) {
+      h = (h ^ bytes[j]) * 1099511628211ull;
+    }
+    // mix in length; keep sign bit clear so sentinels (<0) stay distinct
+    h ^= (uint64_t)(hi - lo) * 0x9E3779B97F4A7C15ull;
+    out[i] = (int64_t)(h >> 1);
+  }
+}
+
+__global__ void substr_fixed_kernel(const int64_t* __restrict__ offsets,
+                                    const uint8_t* __restrict__ bytes,
+                                    int start, int len,
+                                    uint8_t* __restrict__ out_bytes,
+                                    int32_t* __restrict__ out_lens, int64_t n) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    int64_t lo = offsets[i];
+    int64_t slen = offsets[i + 1] - lo;
+    int64_t b = lo + start;
+    int m = 0;
+    for (; m < len && start + m < slen; ++m) {
+      out_bytes[i * len + m] = bytes[b + m];
+    }
+    for (int k = m; k < len; ++k) out_bytes[i * len + k] = 0;
+    out_lens[i] = m;
+  }
+}
+
+}  // namespace
+
+torch::Tensor like_mask(torch::Tensor offsets, torch::Tensor bytes, py::bytes pattern) {
+  SAIL_CHECK(offsets.is_cuda() && bytes.is_cuda());
+  std::string pat(pattern);
+  int64_t n = offsets.numel() - 1;
+  auto out = torch::empty({n}, offsets.options().dtype(torch::kBool));
+  if (n == 0) return out;
+  auto patT = torch::empty({(int64_t)pat.size()},
+                           torch::TensorOptions().dtype(torch::kUInt8));
+  std::memcpy(patT.data_ptr(), pat.data(), pat.size());
+  patT = patT.to(offsets.device());
+  // fast path: pure containment '%abc%' (no '_' inside)
+  bool pure_contains = pat.size() >= 2 && pat.front() == '%' && pat.back() == '%' &&
+                       pat.find('_') == std::string::npos &&
+                       pat.find('%', 1) == pat.size() - 1;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  if (pure_contains) {
+    hipLaunchKernelGGL(contains_kernel, dim3(grid_for(n)), dim3(kBlock), 0, stream,
+                       offsets.data_ptr<int64_t>(), bytes.data_ptr<uint8_t>(),
+                       patT.data_ptr<uint8_t>() + 1, (int)pat.size() - 2,
+                       out.data_ptr<bool>(), n);
+  } else {
+    hipLaunchKernelGGL(like_mask_kernel, dim3(grid_for(n)), dim3(kBlock), 0, stream,
+                       offsets.data_ptr<int64_t>(), bytes.data_ptr<uint8_t>(),
+                       patT.data_ptr<uint8_t>(), (int)pat.size(),
+                       out.data_ptr<bool>(), n);
+  }
+  return out;
+}
+
+torch::Tensor string_hash64(torch::Tensor offsets, torch::Tensor bytes) {
+  SAIL_CHECK(offsets.is_cuda() && bytes.is_cuda());
+  int64_t n = offsets.numel() - 1;
+  auto out = torch::empty({n}, offsets.options().dtype(torch::kInt64));
+  if (n == 0) return out;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(string_hash64_kernel, dim3(grid_for(n)), dim3(kBlock), 0, stream,
+                     offsets.data_ptr<int64_t>(), bytes.data_ptr<uint8_t>(),
+                     out.data_ptr<int64_t>(), n);
+  return out;
+}
+
+std::vector<torch::Tensor> substr_fixed(torch::Tensor offsets, torch::Tensor bytes,
+                                        int64_t start, int64_t len) {
+  SAIL_CHECK(offsets.is_cuda() && bytes.is_cuda());
+  int64_t n = offsets.numel() - 1;
+  auto out_bytes = torch::empty({n * len}, bytes.options());
+  auto out_lens = torch::empty({n}, offsets.options().dtype(torch::kInt32));
+  if (n) {
+    hipStream_t stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(substr_fixed_kernel, dim3(grid_for(n)), dim3(kBlock), 0, stream,
+                       offsets.data_ptr<int64_t>(), bytes.data_ptr<uint8_t>(),
+                       (int)start, (int)len, out_bytes.data_ptr<uint8_t>(),
+                       out_lens.data_ptr<int32_t>(), n);
+  }
+  return {out_bytes, out_lens};
+}

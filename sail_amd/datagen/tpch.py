@@ -180,7 +180,7 @@ class TpchGenerator:
         dev = self.device
         return Table({
             "r_regionkey": Column(T.I64, torch.arange(5, dtype=torch.int64, device=dev)),
-            "r_name": StringColumn.from_pylist(REGIONS, device=dev, dict_encode=False),
+            "r_name": StringColumn.from_pylist(REGIONS, device=dev, dict_encode=True),
             "r_comment": StringColumn.from_pylist(["" for _ in REGIONS], device=dev, dict_encode=False),
         })
 
@@ -188,7 +188,7 @@ class TpchGenerator:
         dev = self.device
         return Table({
             "n_nationkey": Column(T.I64, torch.arange(25, dtype=torch.int64, device=dev)),
-            "n_name": StringColumn.from_pylist([n for n, _ in NATIONS], device=dev, dict_encode=False),
+            "n_name": StringColumn.from_pylist([n for n, _ in NATIONS], device=dev, dict_encode=True),
             "n_regionkey": Column(T.I64, torch.tensor([r for _, r in NATIONS], dtype=torch.int64, device=dev)),
             "n_comment": StringColumn.from_pylist(["" for _ in NATIONS], device=dev, dict_encode=False),
         })

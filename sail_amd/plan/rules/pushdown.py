@@ -186,14 +186,21 @@ def _push_join(plan: S.Join, conds: List[S.Expr]) -> S.Plan:
             else:
                 post_conds.append(c)
 
-    # ON conjuncts that reference only one side can also sink (inner joins)
-    if how in ("inner", "cross"):
+    # single-side ON conjuncts sink into their input where semantics allow:
+    #  - inner/cross: both sides
+    #  - left/semi/anti: right-side-only conjuncts only gate matching, so
+    #    they can filter the right input (left rows survive as unmatched);
+    #    left-side-only conjuncts must stay in ON for outer joins
+    #  - right: mirror
+    if how in ("inner", "cross", "left", "right", "semi", "anti"):
         sunk_on: List[S.Expr] = []
         for c in on_conds:
             refs = expr_refs(c)
-            if not _has_subquery(c) and refs and all(i < nleft for i in refs):
+            can_left = how in ("inner", "cross", "semi", "anti")
+            can_right = how in ("inner", "cross", "left", "semi", "anti")
+            if not _has_subquery(c) and refs and all(i < nleft for i in refs) and can_left:
                 left_conds.append(c)
-            elif not _has_subquery(c) and refs and all(i >= nleft for i in refs):
+            elif not _has_subquery(c) and refs and all(i >= nleft for i in refs) and can_right:
                 right_conds.append(remap_expr(c, {i: i - nleft for i in refs}))
             else:
                 sunk_on.append(c)

@@ -420,6 +420,14 @@ def register_tpch(session, sf: float = 0.01, device=None, rank: int = 0, world: 
     dev = device or session.device
     gen = TpchGenerator(sf=sf, device=dev, seed=seed, rank=rank, world=world, full=full)
     tables = gen.generate_all()
+    # planning statistics must match on every rank: use the deterministic
+    # whole-table cardinalities, not the local shard size
+    globals_ = {"region": 5, "nation": 25, "supplier": gen.n_supplier,
+                "customer": gen.n_customer, "part": gen.n_part,
+                "partsupp": gen.n_part * 4, "orders": gen.n_orders,
+                "lineitem": gen.n_orders * 4}
     for name, tbl in tables.items():
-        session.catalog.register_table(name, tbl)
+        session.catalog.register_table(
+            name, tbl, replicated=(world == 1 or name in ("region", "nation")),
+            global_rows=globals_[name])
     return tables

@@ -92,6 +92,22 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
         data.index_add_(0, gidm, torch.ones(n_used, dtype=torch.int64, device=dev))
         return Column(T.I64, data, None)
 
+    if name == "sumf":
+        vals = c.data[mask].to(torch.float64)
+        if isinstance(c.dtype, T.DecimalType):
+            vals = vals / (10.0 ** c.dtype.scale)
+        data = torch.zeros(ng, dtype=torch.float64, device=dev)
+        data.index_add_(0, gidm, vals)
+        return Column(T.F64, data, None)
+
+    if name == "sumsq":
+        vals = c.data[mask].to(torch.float64)
+        if isinstance(c.dtype, T.DecimalType):
+            vals = vals / (10.0 ** c.dtype.scale)
+        data = torch.zeros(ng, dtype=torch.float64, device=dev)
+        data.index_add_(0, gidm, vals * vals)
+        return Column(T.F64, data, None)
+
     if name == "count_if":
         vals = c.data[mask].to(torch.int64)
         data = torch.zeros(ng, dtype=torch.int64, device=dev)

@@ -17,6 +17,10 @@ from .column import Column, StringColumn, Table
 class Chunk:
     columns: List[Column] = field(default_factory=list)
     names: List[str] = field(default_factory=list)
+    #: distribution across ranks in SPMD mode: "sharded" (each rank holds a
+    #: horizontal slice) or "replicated" (identical on every rank). Ignored
+    #: in single-process execution.
+    partitioning: str = "replicated"
 
     @property
     def num_rows(self) -> int:
@@ -27,14 +31,16 @@ class Chunk:
         return self.columns[0].device if self.columns else torch.device("cpu")
 
     def gather(self, indices: torch.Tensor) -> "Chunk":
-        return Chunk([c.gather(indices) for c in self.columns], list(self.names))
+        return Chunk([c.gather(indices) for c in self.columns], list(self.names),
+                     self.partitioning)
 
     def filter_mask(self, mask: torch.Tensor) -> "Chunk":
         idx = torch.nonzero(mask, as_tuple=False).squeeze(1)
         return self.gather(idx)
 
     def slice(self, start: int, length: int) -> "Chunk":
-        return Chunk([c.slice(start, length) for c in self.columns], list(self.names))
+        return Chunk([c.slice(start, length) for c in self.columns], list(self.names),
+                     self.partitioning)
 
     def to_table(self) -> Table:
         cols = {}
@@ -47,7 +53,8 @@ class Chunk:
         return Table(cols)
 
     def to(self, device) -> "Chunk":
-        return Chunk([c.to(device) for c in self.columns], list(self.names))
+        return Chunk([c.to(device) for c in self.columns], list(self.names),
+                     self.partitioning)
 
     @staticmethod
     def from_table(t: Table) -> "Chunk":

@@ -55,6 +55,18 @@ class Executor:
     def __init__(self, ctx: ExecutionContext):
         self.ctx = ctx
         self.ev = Evaluator(ctx)
+        d = getattr(ctx.session, "dist", None)
+        self.dctx = d if (d is not None and d.world > 1) else None
+
+    def _gather(self, chunk: Chunk) -> Chunk:
+        """Replicate a sharded chunk on every rank (all_gather over RCCL)."""
+        if self.dctx is None or chunk.partitioning != "sharded":
+            return chunk
+        from ..exec.distributed import gather_chunk
+
+        out = gather_chunk(chunk, self.dctx)
+        out.partitioning = "replicated"
+        return out
 
     def execute(self, plan: S.Plan) -> Chunk:
         m = getattr(self, "_x_" + type(plan).__name__, None)
@@ -64,10 +76,14 @@ class Executor:
 
     # -- leaves ------------------------------------------------------------
     def _x_Read(self, p: S.Read) -> Chunk:
-        t = self.ctx.session.catalog.get_table_data(p.table, self.ctx.device)
+        cat = self.ctx.session.catalog
+        t = cat.get_table_data(p.table, self.ctx.device)
         if t is None:
             raise ExecError(f"no data for table {p.table}")
-        return Chunk.from_table(t)
+        out = Chunk.from_table(t)
+        if self.dctx is not None and not cat.is_replicated(p.table):
+            out.partitioning = "sharded"
+        return out
 
     def _x_DataSourceRead(self, p: S.DataSourceRead) -> Chunk:
         from ..datasource.registry import read_source
@@ -89,7 +105,7 @@ class Executor:
 
     def _x_SubqueryAlias(self, p: S.SubqueryAlias) -> Chunk:
         c = self.execute(p.input)
-        return Chunk(c.columns, [n for n, _ in p.schema])
+        return Chunk(c.columns, [n for n, _ in p.schema], c.partitioning)
 
     # -- row ops -----------------------------------------------------------
     def _x_Filter(self, p: S.Filter) -> Chunk:
@@ -105,16 +121,22 @@ class Executor:
         for e in p.exprs:
             v = self.ev.eval(e, child)
             cols.append(broadcast(v, child.num_rows, child.device))
-        return Chunk(cols, [n for n, _ in p.schema])
+        return Chunk(cols, [n for n, _ in p.schema], child.partitioning)
 
     def _x_Limit(self, p: S.Limit) -> Chunk:
-        child = self.execute(p.input)
+        child = self._gather(self.execute(p.input))
         start = p.offset
         n = p.n if p.n is not None else child.num_rows - start
         return child.slice(start, max(0, min(n, child.num_rows - start)))
 
     def _x_Distinct(self, p: S.Distinct) -> Chunk:
         child = self.execute(p.input)
+        if self.dctx is not None and child.partitioning == "sharded":
+            # local distinct first (shrinks the exchange), then global
+            if child.num_rows:
+                gid, rep, ng = group_ids(child.columns)
+                child = child.gather(rep)
+            child = self._gather(child)
         if child.num_rows == 0:
             return child
         gid, rep, ng = group_ids(child.columns)
@@ -122,7 +144,7 @@ class Executor:
 
     # -- sort --------------------------------------------------------------
     def _x_Sort(self, p: S.Sort) -> Chunk:
-        child = self.execute(p.input)
+        child = self._gather(self.execute(p.input))
         if child.num_rows <= 1:
             return child
         idx = sort_indices(self.ev, p.keys, child)
@@ -131,6 +153,8 @@ class Executor:
     # -- aggregate ---------------------------------------------------------
     def _x_Aggregate(self, p: S.Aggregate) -> Chunk:
         child = self.execute(p.input)
+        if self.dctx is not None and child.partitioning == "sharded":
+            return self._dist_aggregate(p, child)
         n = child.num_rows
         dev = child.device
         if p.group_by:
@@ -154,10 +178,118 @@ class Executor:
             agg_cols.append(agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype))
         return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
 
+    def _dist_aggregate(self, p: S.Aggregate, child: Chunk) -> Chunk:
+        """Two-phase distributed aggregate: local partials -> all_gather the
+        (tiny) partial tables -> merge. Falls back to gathering the input for
+        non-decomposable aggregates (DISTINCT, percentiles, ...)."""
+        from ..exec.distributed import decompose_agg, gather_chunk
+
+        decomps = [decompose_agg(a) for a in p.aggs]
+        if any(d is None for d in decomps):
+            child = self._gather(child)
+            p2 = S.Aggregate(input=None, group_by=p.group_by, aggs=p.aggs)
+            p2.schema = p.schema
+            return self._local_aggregate(p2, child)
+
+        n, dev = child.num_rows, child.device
+        # 1) local partials
+        if p.group_by:
+            key_cols = [broadcast(self.ev.eval(g, child), n, dev) for g in p.group_by]
+            if n:
+                gid, rep, ng = group_ids(key_cols)
+                local_keys = [c.gather(rep) for c in key_cols]
+            else:
+                gid, ng = torch.zeros(0, dtype=torch.int64, device=dev), 0
+                local_keys = key_cols
+        else:
+            if n:
+                gid, ng = global_ids(n, dev)
+            else:
+                gid, ng = torch.zeros(0, dtype=torch.int64, device=dev), 0
+            local_keys = []
+        partial_cols = []
+        partial_merges = []
+        for a, d in zip(p.aggs, decomps):
+            args = [broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
+            fmask = self.ev.eval_mask(a.filter, child) if a.filter is not None else None
+            for pname, mname in zip(d.partials, d.merges):
+                use_args = args if (args or pname != "count") else []
+                col = agg_eval(pname, use_args, gid, ng, False, fmask,
+                               None) if ng else _empty_partial(pname, dev)
+                partial_cols.append(col)
+                partial_merges.append(mname)
+        # 2) exchange partial tables
+        partial = Chunk(local_keys + partial_cols,
+                        [f"k{i}" for i in range(len(local_keys))]
+                        + [f"p{i}" for i in range(len(partial_cols))], "sharded")
+        gathered = gather_chunk(partial, self.dctx)
+        # 3) merge
+        gn = gathered.num_rows
+        nk = len(local_keys)
+        if p.group_by:
+            if gn == 0:
+                return Chunk(gathered.columns[:nk]
+                             + [_empty_agg_col(a, dev) for a in p.aggs],
+                             [nm for nm, _ in p.schema])
+            mgid, mrep, mng = group_ids(gathered.columns[:nk])
+            out_keys = [c.gather(mrep) for c in gathered.columns[:nk]]
+        else:
+            mgid, mng = global_ids(gn, dev) if gn else (torch.zeros(0, dtype=torch.int64, device=dev), 1)
+            out_keys = []
+        merged = []
+        for i, mname in enumerate(partial_merges):
+            col = gathered.columns[nk + i]
+            merged.append(agg_eval(mname, [col], mgid, mng, False, None, None))
+        out_cols = []
+        ci = 0
+        for a, d in zip(p.aggs, decomps):
+            k = len(d.partials)
+            out_cols.append(d.finalize(merged[ci:ci + k], a.dtype))
+            ci += k
+        return Chunk(out_keys + out_cols, [nm for nm, _ in p.schema])
+
+    def _local_aggregate(self, p: S.Aggregate, child: Chunk) -> Chunk:
+        n = child.num_rows
+        dev = child.device
+        if p.group_by:
+            key_cols = [broadcast(self.ev.eval(g, child), n, dev) for g in p.group_by]
+            if n == 0:
+                return Chunk(key_cols + [_empty_agg_col(a, dev) for a in p.aggs],
+                             [nm for nm, _ in p.schema])
+            gid, rep, ng = group_ids(key_cols)
+            out_keys = [c.gather(rep) for c in key_cols]
+        else:
+            if n == 0:
+                cols = [_empty_global_agg(a, dev) for a in p.aggs]
+                return Chunk(cols, [nm for nm, _ in p.schema])
+            gid, ng = global_ids(n, dev)
+            out_keys = []
+        agg_cols = []
+        for a in p.aggs:
+            args = [broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
+            fmask = self.ev.eval_mask(a.filter, child) if a.filter is not None else None
+            agg_cols.append(agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype))
+        return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
+
     # -- joins -------------------------------------------------------------
     def _x_Join(self, p: S.Join) -> Chunk:
         left = self.execute(p.left)
         right = self.execute(p.right)
+        if self.dctx is not None:
+            lp, rp = left.partitioning, right.partitioning
+            if lp == "sharded" or rp == "sharded":
+                out_part = "sharded"
+                if rp == "sharded" and lp == "sharded":
+                    # broadcast join: replicate the (smaller) build side
+                    right = self._gather(right)
+                elif lp == "replicated" and rp == "sharded":
+                    if p.how not in ("inner", "cross"):
+                        # left-side semantics need every left row exactly once
+                        right = self._gather(right)
+                        out_part = "replicated"
+                out = join_chunks(self.ev, p, left, right)
+                out.partitioning = out_part
+                return out
         return join_chunks(self.ev, p, left, right)
 
     # -- set ops -----------------------------------------------------------
@@ -202,7 +334,7 @@ class Executor:
     def _x_WindowPlan(self, p: S.WindowPlan) -> Chunk:
         from .window import eval_window
 
-        child = self.execute(p.input)
+        child = self._gather(self.execute(p.input))
         cols = list(child.columns)
         for we in p.window_exprs:
             e = we.child if isinstance(we, S.Alias) else we
@@ -298,11 +430,17 @@ def _sortable(col: Column) -> torch.Tensor:
 
         if col.is_dict:
             return col.codes.to(torch.int64)  # dict is sorted => codes ordered
-        # raw strings: host rank
+        # raw strings: host DENSE rank (equal values must get equal ranks or
+        # stable multi-key sorting breaks on the later keys)
         vals = col.to_pylist()
         order = sorted(range(len(vals)), key=lambda i: (vals[i] is None, vals[i] or ""))
         rank = [0] * len(vals)
-        for r, i in enumerate(order):
+        r = -1
+        prev = object()
+        for i in order:
+            if vals[i] != prev:
+                r += 1
+                prev = vals[i]
             rank[i] = r
         return torch.tensor(rank, dtype=torch.int64, device=col.device)
     d = col.data
@@ -507,3 +645,9 @@ def _empty_global_agg(a: S.AggFunc, dev) -> Column:
         return StringColumn.from_pylist([None], device=dev)
     return Column(a.dtype, torch.zeros(1, dtype=a.dtype.storage or torch.int64, device=dev),
                   torch.zeros(1, dtype=torch.uint8, device=dev))
+
+
+def _empty_partial(pname: str, dev) -> Column:
+    if pname in ("count", "count_if", "sum"):
+        return Column(T.I64, torch.zeros(0, dtype=torch.int64, device=dev))
+    return Column(T.F64, torch.zeros(0, dtype=torch.float64, device=dev))

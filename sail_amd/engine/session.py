@@ -27,6 +27,8 @@ class Catalog:
         self._schemas: Dict[str, List[Tuple[str, T.DataType]]] = {}
         self._providers: Dict[str, Callable[[torch.device], Table]] = {}
         self._views: Dict[str, S.Plan] = {}
+        self._replicated = set()
+        self._global_rows: Dict[str, int] = {}
         self._lock = threading.RLock()
 
     def _key(self, name: str) -> str:
@@ -34,13 +36,23 @@ class Catalog:
 
     # -- registration ------------------------------------------------------
     def register_table(self, name: str, table: Table,
-                       schema: Optional[List[Tuple[str, T.DataType]]] = None):
+                       schema: Optional[List[Tuple[str, T.DataType]]] = None,
+                       replicated: bool = True, global_rows: Optional[int] = None):
+        """replicated=False marks a table as rank-sharded in SPMD mode;
+        global_rows is the whole-table count (planning statistics must be
+        identical on every rank so all ranks pick the same join order)."""
         with self._lock:
             k = self._key(name)
             self._tables[k] = table
             if schema is None:
                 schema = [(n, c.dtype) for n, c in table.columns.items()]
             self._schemas[k] = schema
+            if global_rows is not None:
+                self._global_rows[k] = global_rows
+            if replicated:
+                self._replicated.add(k)
+            else:
+                self._replicated.discard(k)
 
     def register_table_chunk(self, name: str, chunk: Chunk, schema):
         self.register_table(name, chunk.to_table(), schema)
@@ -116,8 +128,13 @@ class Catalog:
             return t
         return None
 
+    def is_replicated(self, name: str) -> bool:
+        return self._key(name) in self._replicated
+
     def table_rows(self, name: str) -> Optional[int]:
         k = self._key(name)
+        if k in self._global_rows:
+            return self._global_rows[k]
         t = self._tables.get(k)
         if t is not None:
             return t.num_rows

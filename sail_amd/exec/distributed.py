@@ -1,0 +1,126 @@
+"""Distributed operator strategies (SPMD, one process per GPU).
+
+Every rank holds a shard of each big table (datagen/tpch._shard) and runs
+the same plan; this module supplies the exchange points:
+
+  * gather_chunk — all_gather a sharded chunk -> replicated chunk
+    (broadcast joins, global sorts/limits)
+  * partial/final aggregate decomposition — local partial aggregates are
+    all_gathered (tiny) and merged, replacing the reference's shuffle-based
+    two-phase aggregation (ref: crates/sail-execution InputMode::Shuffle /
+    Merge) with collectives over xGMI.
+
+Exchange payloads are whole columns (few large buffers), matching xGMI's
+preference for large point-to-point transfers over many small messages.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+
+from ..engine import types as T
+from ..engine.chunk import Chunk
+from ..engine.column import Column, StringColumn
+from ..plan import spec as S
+from .context import DistContext
+
+
+def gather_column(c: Column, d: DistContext) -> Column:
+    from ..engine.executor import concat_columns
+
+    if isinstance(c, StringColumn):
+        raw = c.decode_dict()
+        offs_list = d.all_gather_tensors(raw.offsets[1:] - raw.offsets[:-1])
+        bytes_list = d.all_gather_tensors(raw.bytes_)
+        val_list = d.all_gather_tensors(raw.valid_mask().to(torch.uint8)) \
+            if raw.validity is not None else None
+        cols = []
+        for i in range(d.world):
+            lens = offs_list[i]
+            offs = torch.zeros(lens.shape[0] + 1, dtype=torch.int64, device=lens.device)
+            torch.cumsum(lens, 0, out=offs[1:])
+            v = val_list[i] if val_list is not None else None
+            cols.append(StringColumn(offs, bytes_list[i], v, None, dtype=c.dtype))
+        return concat_columns(cols)
+    data_list = d.all_gather_tensors(c.data)
+    if c.validity is not None:
+        val_list = d.all_gather_tensors(c.validity)
+        validity = torch.cat(val_list)
+    else:
+        validity = None
+    return Column(c.dtype, torch.cat(data_list), validity)
+
+
+def gather_chunk(chunk: Chunk, d: DistContext) -> Chunk:
+    return Chunk([gather_column(c, d) for c in chunk.columns], list(chunk.names))
+
+
+# ---------------------------------------------------------------------------
+# aggregate decomposition
+# ---------------------------------------------------------------------------
+
+class AggDecomposition:
+    """Partial aggregates + a merge recipe for one logical aggregate.
+
+    partials: list of (agg_name, use_original_args: bool) executed locally;
+    `finalize(cols)` combines the gathered partial columns (one Column per
+    partial, already re-aggregated by group with the merge agg) into the
+    final output column.
+    """
+
+    def __init__(self, partials, merges, finalize):
+        self.partials = partials  # names for local partial aggs
+        self.merges = merges      # agg names used to re-aggregate partials
+        self.finalize = finalize  # fn(list[Column], out_type) -> Column
+
+
+def decompose_agg(a: S.AggFunc) -> Optional[AggDecomposition]:
+    name = a.name
+    if a.distinct:
+        return None  # handled by input gather fallback
+    if name in ("sum", "try_sum"):
+        return AggDecomposition(["sum"], ["sum"], lambda cols, t: _retype(cols[0], t))
+    if name in ("count", "count_if"):
+        return AggDecomposition([name], ["sum"], lambda cols, t: _retype(cols[0], T.I64))
+    if name in ("min", "max", "any", "bool_and"):
+        return AggDecomposition([name], [name], lambda cols, t: _retype(cols[0], t))
+    if name == "avg":
+        def fin(cols, t):
+            from ..engine.aggregates import _avg_result
+
+            sums, cnt = cols[0], cols[1]
+            return _avg_result(sums.data, cnt.data, sums.dtype, t)
+
+        return AggDecomposition(["sum", "count"], ["sum", "sum"], fin)
+    if name in ("stddev_samp", "stddev_pop", "var_samp", "var_pop"):
+        def fin(cols, t, _name=name):
+            cnt = cols[0].data.to(torch.float64)
+            s1 = cols[1].data.to(torch.float64)
+            s2 = cols[2].data.to(torch.float64)
+            mean = s1 / cnt.clamp_min(1)
+            m2 = (s2 - cnt * mean * mean).clamp_min(0)
+            denom = cnt - (1.0 if _name.endswith("_samp") else 0.0)
+            var = m2 / denom.clamp_min(1e-300)
+            data = torch.sqrt(var) if _name.startswith("stddev") else var
+            valid = denom > 0
+            return Column(T.F64, data,
+                          None if bool(valid.all()) else valid.to(torch.uint8))
+
+        return AggDecomposition(["count", "sumf", "sumsq"], ["sum", "sum", "sum"], fin)
+    if name in ("first", "any_value"):
+        return AggDecomposition(["first"], ["first"], lambda cols, t: cols[0])
+    if name == "last":
+        return AggDecomposition(["last"], ["last"], lambda cols, t: cols[0])
+    return None
+
+
+def _retype(c: Column, t) -> Column:
+    if t is None or c.dtype == t:
+        return c
+    if isinstance(c, StringColumn):
+        return c
+    data = c.data
+    if t.storage is not None and data.dtype != t.storage:
+        data = data.to(t.storage)
+    return Column(t, data, c.validity)

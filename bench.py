@@ -16,6 +16,9 @@ import os
 import sys
 import time
 
+# reduce HBM fragmentation for the large transient intermediates
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 
 def main():
     ap = argparse.ArgumentParser()

@@ -21,12 +21,33 @@ def pack_vocab(words: List[str], device) -> Tuple[torch.Tensor, torch.Tensor, to
 
 def assemble_words(word_ids: torch.Tensor, nwords: torch.Tensor,
                    vocab: Tuple[torch.Tensor, torch.Tensor, torch.Tensor],
-                   sep: int = 32) -> StringColumn:
+                   sep: int = 32, row_chunk: int = 8_000_000) -> StringColumn:
     """Build one string per row by joining words from a vocabulary.
 
     word_ids: [n, kmax] int64 vocabulary indices
     nwords:   [n] number of words used per row (<= kmax)
+
+    Processed in row chunks: the per-byte index temporaries are ~24 B per
+    output byte, which at SF100 (16 GB of comment text) would otherwise need
+    ~400 GB transient HBM.
     """
+    n = word_ids.shape[0]
+    if n <= row_chunk:
+        return _assemble_words_chunk(word_ids, nwords, vocab, sep)
+    parts = [
+        _assemble_words_chunk(word_ids[i : i + row_chunk], nwords[i : i + row_chunk], vocab, sep)
+        for i in range(0, n, row_chunk)
+    ]
+    dev = word_ids.device
+    lens = torch.cat([p.offsets[1:] - p.offsets[:-1] for p in parts])
+    offsets = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(lens, 0, out=offsets[1:])
+    byts = torch.cat([p.bytes_ for p in parts])
+    return StringColumn(offsets, byts)
+
+
+def _assemble_words_chunk(word_ids: torch.Tensor, nwords: torch.Tensor,
+                          vocab, sep: int) -> StringColumn:
     voffs, vbytes, vlens = vocab
     n, kmax = word_ids.shape
     dev = word_ids.device
@@ -40,22 +61,22 @@ def assemble_words(word_ids: torch.Tensor, nwords: torch.Tensor,
     torch.cumsum(row_len, 0, out=offsets[1:])
     total = int(offsets[-1].item())
     # flatten valid words in row-major order
-    flat_span = span.reshape(-1)
-    flat_ids = word_ids.reshape(-1)
     keep = valid.reshape(-1)
-    spans = flat_span[keep]
-    ids = flat_ids[keep]
+    spans = span.reshape(-1)[keep]
+    ids = word_ids.reshape(-1)[keep]
+    del valid, wlen, is_last, span, keep
     # destination start of each word
-    dstart = torch.zeros(spans.shape[0], dtype=torch.int64, device=dev)
-    torch.cumsum(spans, 0, out=dstart)
+    dstart = torch.cumsum(spans, 0)
     dstart = dstart - spans
-    # per byte: which word, offset within word
-    byte_word = torch.repeat_interleave(torch.arange(spans.shape[0], device=dev), spans)
-    off_in = torch.arange(total, device=dev) - dstart[byte_word]
-    wl = vlens[ids][byte_word]
-    src = voffs[ids][byte_word] + torch.minimum(off_in, wl - 1).clamp_min(0)
-    out = torch.where(off_in < wl, vbytes[src],
-                      torch.full((total,), sep, dtype=torch.uint8, device=dev))
+    # per byte: which word, offset within word (int32 per-chunk indices)
+    byte_word = torch.repeat_interleave(
+        torch.arange(spans.shape[0], device=dev, dtype=torch.int64), spans)
+    off_in = torch.arange(total, device=dev) - dstart.index_select(0, byte_word)
+    wl = vlens[ids].index_select(0, byte_word)
+    src = voffs[ids].index_select(0, byte_word) + torch.minimum(off_in, wl - 1).clamp_min(0)
+    del dstart, byte_word
+    out = torch.where(off_in < wl, vbytes.index_select(0, src),
+                      torch.full((1,), sep, dtype=torch.uint8, device=dev))
     return StringColumn(offsets, out)
 
 

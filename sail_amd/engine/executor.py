@@ -501,10 +501,22 @@ def join_chunks(ev: Evaluator, p: S.Join, left: Chunk, right: Chunk) -> Chunk:
     probe_idx, build_idx, counts = equi_join(rkeys, lkeys, how)
 
     if residual is not None and probe_idx.shape[0] > 0:
-        pairs = Chunk([c.gather(probe_idx) for c in left.columns]
-                      + [c.gather(build_idx) for c in right.columns],
-                      [f"c{i}" for i in range(len(left.columns) + len(right.columns))])
-        rmask = ev.eval_mask(residual, pairs)
+        # gather only the columns the residual references (a q21-style
+        # self-join can have ~B matched pairs; full-width gathers OOM)
+        from ..plan.rules.util import expr_refs, remap_expr
+
+        nl = len(left.columns)
+        refs = sorted(expr_refs(residual))
+        remap = {old: new for new, old in enumerate(refs)}
+        pair_cols = []
+        for old in refs:
+            if old < nl:
+                pair_cols.append(left.columns[old].gather(probe_idx))
+            else:
+                pair_cols.append(right.columns[old - nl].gather(build_idx))
+        pairs = Chunk(pair_cols, [f"c{i}" for i in range(len(pair_cols))])
+        rmask = ev.eval_mask(remap_expr(residual, remap), pairs)
+        del pairs, pair_cols
         probe_idx = probe_idx[rmask]
         build_idx = build_idx[rmask]
         if how in ("semi", "anti", "left", "full"):

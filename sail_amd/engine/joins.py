@@ -68,14 +68,46 @@ def _dict_side_key(c: StringColumn, force_hash: bool) -> torch.Tensor:
 
 
 def dense_ids(build_keys: List[torch.Tensor], probe_keys: List[torch.Tensor]):
-    """Assign identical int64 ids to identical key tuples across both sides."""
+    """Comparable int64 ids for identical key tuples across both sides.
+
+    Single-column int64 keys are used directly (no dense remap needed — the
+    sort/searchsorted matcher handles arbitrary values). Multi-column keys
+    try exact range-packing (k = ((k0-min0)*span1 + (k1-min1))*span2 + ...)
+    and only fall back to torch.unique row-dedup when the packed domain
+    overflows int64."""
     nb = build_keys[0].shape[0]
     if len(build_keys) == 1:
-        allk = torch.cat([build_keys[0], probe_keys[0]])
-        _, inv = torch.unique(allk, return_inverse=True)
-    else:
-        allk = torch.stack([torch.cat([b, p]) for b, p in zip(build_keys, probe_keys)], dim=1)
-        _, inv = torch.unique(allk, dim=0, return_inverse=True)
+        return build_keys[0], probe_keys[0]
+    # exact range packing
+    spans = []
+    mins = []
+    ok = True
+    for b, p in zip(build_keys, probe_keys):
+        if b.numel() == 0 and p.numel() == 0:
+            mins.append(0)
+            spans.append(1)
+            continue
+        candidates = [t for t in (b, p) if t.numel()]
+        lo = min(int(t.min().item()) for t in candidates)
+        hi = max(int(t.max().item()) for t in candidates)
+        mins.append(lo)
+        spans.append(hi - lo + 1)
+    total = 1
+    for s in spans:
+        total *= s
+        if total > (1 << 62):
+            ok = False
+            break
+    if ok:
+        def pack(cols):
+            acc = (cols[0] - mins[0])
+            for i in range(1, len(cols)):
+                acc = acc * spans[i] + (cols[i] - mins[i])
+            return acc
+
+        return pack(build_keys), pack(probe_keys)
+    allk = torch.stack([torch.cat([b, p]) for b, p in zip(build_keys, probe_keys)], dim=1)
+    _, inv = torch.unique(allk, dim=0, return_inverse=True)
     return inv[:nb], inv[nb:]
 
 
@@ -85,10 +117,12 @@ def _expand_matches(bids: torch.Tensor, pids: torch.Tensor,
     plus per-probe match counts and build-side matched flags."""
     nb, np_ = bids.shape[0], pids.shape[0]
     dev = bids.device
+    # null sentinels at the extreme of the int64 domain so they can never
+    # collide with real keys (keys may be arbitrary int64 now) nor each other
     if b_valid is not None:
-        bids = torch.where(b_valid, bids, torch.full_like(bids, -1))
+        bids = torch.where(b_valid, bids, torch.full_like(bids, -(2 ** 63) + 1))
     if p_valid is not None:
-        pids = torch.where(p_valid, pids, torch.full_like(pids, -2))
+        pids = torch.where(p_valid, pids, torch.full_like(pids, -(2 ** 63) + 2))
     order = torch.argsort(bids)
     bsorted = bids[order]
     lo = torch.searchsorted(bsorted, pids, right=False)

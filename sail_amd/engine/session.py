@@ -29,6 +29,7 @@ class Catalog:
         self._views: Dict[str, S.Plan] = {}
         self._replicated = set()
         self._global_rows: Dict[str, int] = {}
+        self._col_stats: Dict[tuple, Optional[tuple]] = {}
         self._lock = threading.RLock()
 
     def _key(self, name: str) -> str:
@@ -128,6 +129,41 @@ class Catalog:
             return t
         return None
 
+    def column_stats(self, name: str, col: str):
+        """(rows, ndv_estimate) for a base-table column; ndv from min/max
+        span for integer-like storage, dict size for dictionary strings.
+        Cached; cheap device reductions on first use."""
+        k = self._key(name)
+        ck = (k, col.lower())
+        if ck in self._col_stats:
+            return self._col_stats[ck]
+        t = self._tables.get(k)
+        out = None
+        if t is not None:
+            for cn, c in t.columns.items():
+                if cn.lower() != col.lower():
+                    continue
+                rows = len(c)
+                from .column import StringColumn
+
+                if isinstance(c, StringColumn):
+                    ndv = c.dict_size if c.is_dict else None
+                elif rows == 0 or c.data.dtype == torch.bool:
+                    ndv = 2 if rows else 1
+                elif c.data.dtype.is_floating_point:
+                    ndv = None
+                else:
+                    lo = int(c.data.min().item())
+                    hi = int(c.data.max().item())
+                    ndv = min(rows, hi - lo + 1)
+                grows = self._global_rows.get(k, rows)
+                if ndv is not None and rows:
+                    ndv = min(max(1, int(ndv * grows / max(rows, 1)) if ndv == rows else ndv), grows)
+                out = (grows, ndv)
+                break
+        self._col_stats[ck] = out
+        return out
+
     def is_replicated(self, name: str) -> bool:
         return self._key(name) in self._replicated
 
@@ -174,8 +210,7 @@ class SessionContext:
         from ..plan.optimizer import optimize
 
         reorder = self.conf.get("sail.optimizer.enable_join_reorder", "true") == "true"
-        return optimize(plan, enable_join_reorder=reorder,
-                        stats=self.catalog.table_rows)
+        return optimize(plan, enable_join_reorder=reorder, stats=self.catalog)
 
     def plan_sql(self, sql: str) -> S.Plan:
         return self.optimize(self.resolve(self.parse(sql)))

@@ -81,7 +81,7 @@ def _estimate(plan: S.Plan, stats) -> Tuple[float, float]:
     if isinstance(plan, S.Read):
         n = None
         if stats is not None:
-            n = stats(plan.table)
+            n = stats.table_rows(plan.table) if hasattr(stats, "table_rows") else stats(plan.table)
         base = float(n) if n else 1000.0
         return base, base
     if isinstance(plan, (S.SubqueryAlias, S.Project, S.Limit, S.Sort)):
@@ -146,6 +146,53 @@ def _selectivity(c: S.Expr) -> float:
     return 0.5
 
 
+def _leaf_column_source(leaf: S.Plan, k: int):
+    """Trace output column k of a leaf subtree to (table_name, column_name)
+    if it is a direct base-table column."""
+    p = leaf
+    while True:
+        if isinstance(p, (S.SubqueryAlias, S.Filter, S.Limit, S.Sort, S.Distinct)):
+            p = p.input
+            continue
+        if isinstance(p, S.Project):
+            e = p.exprs[k]
+            if isinstance(e, S.Alias):
+                e = e.child
+            if isinstance(e, S.BoundRef):
+                k = e.index
+                p = p.input
+                continue
+            return None
+        if isinstance(p, S.Read):
+            if 0 <= k < len(p.schema):
+                return (p.table, p.schema[k][0])
+            return None
+        return None
+
+
+def _conjunct_ndv(c: S.Expr, leaves, offs, leaf_of, ests, stats) -> float:
+    """ndv denominator for one equality conjunct between two leaves."""
+    best = None
+    if isinstance(c, S.BinaryOp) and c.op == "=":
+        for side in (c.left, c.right):
+            e = side.child if isinstance(side, S.Cast) else side
+            if not isinstance(e, S.BoundRef):
+                continue
+            li = leaf_of(e.index)
+            col_k = e.index - offs[li]
+            ndv = None
+            if stats is not None and hasattr(stats, "column_stats"):
+                srcinfo = _leaf_column_source(leaves[li], col_k)
+                if srcinfo is not None:
+                    cs = stats.column_stats(srcinfo[0], srcinfo[1])
+                    if cs is not None and cs[1] is not None:
+                        ndv = float(cs[1])
+            if ndv is None:
+                ndv = ests[li][1]  # base rows fallback
+            best = ndv if best is None else max(best, ndv)
+    return best if best is not None else 1000.0
+
+
 def _reorder_tree(root: S.Join, stats) -> S.Plan:
     leaves: List[S.Plan] = []
     conds: List[S.Expr] = []
@@ -171,38 +218,59 @@ def _reorder_tree(root: S.Join, stats) -> S.Plan:
 
     ests = [_estimate(lf, stats) for lf in leaves]
 
-    # edges: conjuncts joining exactly two leaves with an equality
+    # edge conjuncts with per-key ndv estimates from base-table stats
     edges: Dict[int, Set[int]] = {i: set() for i in range(n)}
+    edge_ndv: List[Tuple[int, int, float]] = []  # (leaf_a, leaf_b, denom)
     for c in conds:
         ls = {leaf_of(i) for i in expr_refs(c)}
         if len(ls) == 2:
             a, b = sorted(ls)
             edges[a].add(b)
             edges[b].add(a)
+            denom = _conjunct_ndv(c, leaves, offs, leaf_of, ests, stats)
+            edge_ndv.append((a, b, denom))
 
-    # greedy order
-    order = [min(range(n), key=lambda i: ests[i][0])]
-    placed = {order[0]}
-    cur_est = ests[order[0]][0]
-    while len(placed) < n:
-        candidates = [i for i in range(n) if i not in placed
-                      and any(j in placed for j in edges[i])]
-        disconnected = False
-        if not candidates:
-            candidates = [i for i in range(n) if i not in placed]
-            disconnected = True
-        best, best_cost = None, None
-        for i in candidates:
-            est_i, base_i = ests[i]
-            if disconnected:
-                cost = cur_est * est_i
-            else:
-                cost = cur_est * est_i / max(base_i, 1.0)
-            if best_cost is None or cost < best_cost:
-                best, best_cost = i, cost
-        order.append(best)
-        placed.add(best)
-        cur_est = max(best_cost, 1.0)
+    def join_denom(j: int, placed: Set[int]) -> float:
+        """Combined ndv for all equality conjuncts connecting leaf j to the
+        placed set (composite keys multiply, capped)."""
+        d = 1.0
+        for a, b, nd in edge_ndv:
+            if (a == j and b in placed) or (b == j and a in placed):
+                d = min(d * nd, 1e15)
+        return d
+
+    # greedy from EVERY seed, objective = sum of intermediate sizes; a single
+    # smallest-leaf seed can dead-end into a low-ndv edge (q5: region ->
+    # nation -> supplier forces customer x supplier on nationkey)
+    def run_greedy(seed: int):
+        order = [seed]
+        placed = {seed}
+        cur_est = ests[seed][0]
+        total = cur_est
+        while len(placed) < n:
+            best, best_cost = None, None
+            for i in range(n):
+                if i in placed:
+                    continue
+                est_i, _ = ests[i]
+                if any(j in placed for j in edges[i]):
+                    cost = cur_est * est_i / max(join_denom(i, placed), 1.0)
+                else:
+                    cost = cur_est * est_i  # cross product
+                if best_cost is None or cost < best_cost:
+                    best, best_cost = i, cost
+            order.append(best)
+            placed.add(best)
+            cur_est = max(best_cost, 1.0)
+            total += cur_est
+        return order, total
+
+    best_order, best_total = None, None
+    for seed in range(n):
+        order_s, total_s = run_greedy(seed)
+        if best_total is None or total_s < best_total:
+            best_order, best_total = order_s, total_s
+    order = best_order
 
     # rebuild left-deep tree in `order`, remapping conjunct indices
     new_off: Dict[int, int] = {}

@@ -21,14 +21,25 @@ class Chunk:
     #: horizontal slice) or "replicated" (identical on every rank). Ignored
     #: in single-process execution.
     partitioning: str = "replicated"
+    #: row-count override for sparse chunks (lazy Project∘Filter fusion may
+    #: leave unreferenced column slots as None)
+    forced_rows: "Optional[int]" = None
 
     @property
     def num_rows(self) -> int:
-        return len(self.columns[0]) if self.columns else 0
+        if self.forced_rows is not None:
+            return self.forced_rows
+        for c in self.columns:
+            if c is not None:
+                return len(c)
+        return 0
 
     @property
     def device(self) -> torch.device:
-        return self.columns[0].device if self.columns else torch.device("cpu")
+        for c in self.columns:
+            if c is not None:
+                return c.device
+        return torch.device("cpu")
 
     def gather(self, indices: torch.Tensor) -> "Chunk":
         return Chunk([c.gather(indices) for c in self.columns], list(self.names),

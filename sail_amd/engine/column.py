@@ -259,23 +259,46 @@ def _pack_strings(strs: List[str], device) -> tuple:
     return torch.from_numpy(offsets).to(device), byts.to(device)
 
 
-def _gather_strings(offsets: torch.Tensor, bytes_: torch.Tensor, indices: torch.Tensor):
-    """Gather rows of a raw string column by index (torch fallback path)."""
+def _gather_strings(offsets: torch.Tensor, bytes_: torch.Tensor, indices: torch.Tensor,
+                    byte_chunk: int = 1 << 28):
+    """Gather rows of a raw string column by index (torch path).
+
+    Output is assembled in ~256 MB byte slices: the per-byte int64 index
+    temporaries are 24 B per output byte and must stay bounded."""
     indices = indices.to(torch.int64)
     starts = offsets.index_select(0, indices)
     ends = offsets.index_select(0, indices + 1)
     lens = ends - starts
-    out_offsets = torch.zeros(indices.shape[0] + 1, dtype=torch.int64, device=offsets.device)
+    n = indices.shape[0]
+    dev = offsets.device
+    out_offsets = torch.zeros(n + 1, dtype=torch.int64, device=dev)
     torch.cumsum(lens, 0, out=out_offsets[1:])
     total = int(out_offsets[-1].item())
-    out_bytes = torch.empty(total, dtype=torch.uint8, device=bytes_.device)
-    # ranges -> flat gather indices without a python loop
-    if total > 0:
-        pos = torch.arange(total, device=offsets.device)
+    if total == 0:
+        return out_offsets, torch.zeros(0, dtype=torch.uint8, device=bytes_.device)
+    if total <= byte_chunk:
+        pos = torch.arange(total, device=dev)
         row = torch.searchsorted(out_offsets[1:], pos, right=True)
         src = starts.index_select(0, row) + (pos - out_offsets.index_select(0, row))
-        out_bytes = bytes_.index_select(0, src)
-    return out_offsets, out_bytes
+        return out_offsets, bytes_.index_select(0, src)
+    # row-sliced assembly
+    parts = []
+    row_start = 0
+    while row_start < n:
+        # find the row range covering ~byte_chunk bytes
+        target = int(out_offsets[row_start].item()) + byte_chunk
+        row_end = int(torch.searchsorted(out_offsets, torch.tensor(target, device=dev)).item())
+        row_end = max(row_start + 1, min(row_end, n))
+        seg_lo = int(out_offsets[row_start].item())
+        seg_hi = int(out_offsets[row_end].item())
+        seg = seg_hi - seg_lo
+        pos = torch.arange(seg, device=dev)
+        local_offs = out_offsets[row_start : row_end + 1] - seg_lo
+        row = torch.searchsorted(local_offs[1:], pos, right=True)
+        src = starts[row_start:row_end].index_select(0, row) + (pos - local_offs.index_select(0, row))
+        parts.append(bytes_.index_select(0, src))
+        row_start = row_end
+    return out_offsets, torch.cat(parts)
 
 
 def _to_days(v) -> int:

@@ -116,7 +116,29 @@ class Executor:
         return child.filter_mask(mask)
 
     def _x_Project(self, p: S.Project) -> Chunk:
-        child = self.execute(p.input)
+        # Project∘Filter fusion: evaluate the predicate on the unfiltered
+        # child, then gather ONLY the columns the projection references —
+        # predicate-only columns (e.g. o_comment in q13) are never gathered.
+        if isinstance(p.input, S.Filter):
+            f = p.input
+            base = self.execute(f.input)
+            if base.num_rows == 0:
+                child = base
+            else:
+                mask = self.ev.eval_mask(f.condition, base)
+                idx = torch.nonzero(mask, as_tuple=False).squeeze(1)
+                from ..plan.rules.util import expr_refs
+
+                refs = set()
+                for e in p.exprs:
+                    refs |= expr_refs(e)
+                cols: List[Optional[Column]] = [None] * len(base.columns)
+                for i in refs:
+                    cols[i] = base.columns[i].gather(idx)
+                child = Chunk(cols, list(base.names), base.partitioning)
+                child.forced_rows = int(idx.shape[0])
+        else:
+            child = self.execute(p.input)
         cols = []
         for e in p.exprs:
             v = self.ev.eval(e, child)

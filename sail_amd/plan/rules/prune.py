@@ -82,6 +82,16 @@ def _prune(plan: S.Plan, needed: Set[int]) -> Tuple[S.Plan, Dict[int, int]]:
         cond = remap_expr(_remap_subqueries(plan.condition), cmap)
         out = S.Filter(input=child, condition=cond)
         out.schema = child.schema
+        if child_needed - set(needed):
+            # predicate-only columns: trim IMMEDIATELY above the filter so
+            # they are never gathered through joins/aggregates above (the
+            # executor fuses Project∘Filter into a lazy selection)
+            keep = sorted(needed)
+            pr = S.Project(input=out, exprs=[
+                S.BoundRef(cmap[i], out.schema[cmap[i]][0], out.schema[cmap[i]][1])
+                for i in keep])
+            pr.schema = [plan.schema[i] for i in keep]
+            return pr, {old: new for new, old in enumerate(keep)}
         return out, cmap
 
     if isinstance(plan, S.SubqueryAlias):

@@ -18,9 +18,13 @@ from .joins import normalize_key
 
 def group_ids(keys: List[Column]) -> Tuple[torch.Tensor, torch.Tensor, int]:
     """Returns (gid per row, representative row index per group, n_groups).
-    Null keys form their own groups (SQL GROUP BY semantics)."""
+    Null keys form their own groups (SQL GROUP BY semantics).
+
+    Fast path: when the combined key domain is dense (dict codes, dense
+    surrogate keys like l_orderkey) the gids come from a presence-bitmap +
+    prefix-sum remap — no sort. The torch.unique sort path only runs for
+    genuinely sparse domains (e.g. hashed raw strings)."""
     if not keys:
-        n = 0
         raise ValueError("group_ids requires keys")
     dev = keys[0].device
     n = len(keys[0])
@@ -31,6 +35,34 @@ def group_ids(keys: List[Column]) -> Tuple[torch.Tensor, torch.Tensor, int]:
             # give nulls a dedicated code below the domain
             k = torch.where(c.valid_mask(), k, torch.full_like(k, k.min().item() - 1 if n else -1))
         norm.append(k)
+
+    # dense-domain fast path
+    if n > 0:
+        mins, spans = [], []
+        total = 1
+        for k in norm:
+            lo = int(k.min().item())
+            hi = int(k.max().item())
+            mins.append(lo)
+            spans.append(hi - lo + 1)
+            total *= spans[-1]
+            if total > max(4 * n, 1 << 22) or total > (1 << 31):
+                total = -1
+                break
+        if total > 0:
+            packed = norm[0] - mins[0]
+            for i in range(1, len(norm)):
+                packed = packed * spans[i] + (norm[i] - mins[i])
+            present = torch.zeros(total, dtype=torch.bool, device=dev)
+            present[packed] = True
+            lut = torch.cumsum(present.to(torch.int32), 0) - 1
+            gid = lut.index_select(0, packed).to(torch.int64)
+            ng = int(lut[-1].item()) + 1
+            rep = torch.full((ng,), n, dtype=torch.int64, device=dev)
+            rep.scatter_reduce_(0, gid, torch.arange(n, device=dev), reduce="amin",
+                                include_self=True)
+            return gid, rep, ng
+
     if len(norm) == 1:
         uniq, gid = torch.unique(norm[0], return_inverse=True)
         ng = int(uniq.shape[0])
@@ -270,3 +302,150 @@ def _avg_result(sums: torch.Tensor, cnt: torch.Tensor, in_type: T.DataType, out_
 def global_ids(n: int, device) -> Tuple[torch.Tensor, int]:
     """gid tensor for a global (no GROUP BY) aggregate: all rows in group 0."""
     return torch.zeros(n, dtype=torch.int64, device=device), 1
+
+
+# ---------------------------------------------------------------------------
+# fused GPU aggregation (ops/csrc/hash_agg.hip)
+# ---------------------------------------------------------------------------
+
+def fused_agg_batch(aggs, args_list, fmasks, gid: torch.Tensor, ng: int):
+    """Evaluate a whole aggregate list in one (or few) fused kernel passes.
+
+    aggs: list of S.AggFunc; args_list[i]: evaluated arg Columns;
+    fmasks[i]: optional bool mask (FILTER clause). Returns list[Column] or
+    None when ineligible (falls back to per-agg torch path).
+    """
+    if not gid.is_cuda or ng > 4096 or ng == 0:
+        return None
+    from ..ops import kernels as K
+
+    if not K.available():
+        return None
+    ext = K.require()
+
+    n = gid.shape[0]
+    # plan columns: per agg -> list of (value_tensor|None, op) + finalizer
+    launches = {}  # key: id of fmask tensor (None -> 0) -> list of col specs
+    plans = []
+    for a, args, fm in zip(aggs, args_list, fmasks):
+        name = a.name
+        if a.distinct or name not in ("sum", "try_sum", "count", "count_if",
+                                      "avg", "try_avg", "min", "max"):
+            return None
+        c = args[0] if args else None
+        if c is not None and isinstance(c, StringColumn) and not c.is_dict:
+            return None
+        if c is not None and c.dtype.is_float and name in ("min", "max"):
+            return None  # float min/max: bit-cast doesn't order negatives
+        mkey = id(fm) if fm is not None else 0
+        cols = launches.setdefault(mkey, (fm, []))[1]
+
+        def add(spec):
+            cols.append(spec)
+            return len(cols) - 1
+
+        if name in ("count", "count_if"):
+            if name == "count_if":
+                vt = (c.data.to(torch.uint8) if c.data.dtype == torch.bool else c.data)
+                if c.validity is not None:
+                    vt = vt * c.validity
+                idx = add((vt, 0))
+            elif c is None or c.validity is None:
+                idx = add((None, 2))
+            else:
+                idx = add((c.validity, 0))
+            plans.append(("count", mkey, idx, None, a))
+            continue
+        if name in ("sum", "try_sum", "avg", "try_avg"):
+            is_f = c.dtype.is_float
+            data = c.data
+            if c.validity is not None:
+                data = data * c.validity.to(data.dtype)
+            if is_f:
+                vidx = add((data.to(torch.float64), 1))
+            else:
+                # overflow check: int64 accumulation only when safe
+                amax = int(data.abs().max().item()) if n else 0
+                if amax and amax > (1 << 62) // max(n, 1):
+                    vidx = add((data.to(torch.float64), 1))
+                    is_f = True
+                else:
+                    vidx = add((data, 0))
+            cidx = None
+            if name in ("avg", "try_avg") or c.validity is not None or fm is not None:
+                cidx = add((c.validity, 0) if c.validity is not None else (None, 2))
+            plans.append(("sumavg", mkey, vidx, (cidx, is_f, c.dtype), a))
+            continue
+        # min/max
+        if isinstance(c, StringColumn):
+            data = c.codes
+        else:
+            data = c.data
+        if c.validity is not None:
+            return None  # rare; keep torch path for nullable min/max
+        op = 3 if name == "min" else 4
+        vidx = add((data, op))
+        plans.append(("minmax", mkey, vidx, (op, c), a))
+
+    gid32 = gid.to(torch.int32)
+    results = {}
+    for mkey, (fm, cols) in launches.items():
+        outs = []
+        for i in range(0, len(cols), 10):
+            chunk_cols = cols[i : i + 10]
+            vals = [t for t, _ in chunk_cols]
+            ops = [int(op) for _, op in chunk_cols]
+            mask_t = fm.to(torch.uint8) if fm is not None and fm.dtype == torch.bool else fm
+            out = ext.grouped_acc(gid32, mask_t, vals, ops, ng)
+            outs.extend(out[j] for j in range(out.shape[0]))
+        results[mkey] = outs
+
+    # finalize per agg
+    out_cols: List[Column] = []
+    for kind, mkey, vidx, extra, a in plans:
+        outs = results[mkey]
+        if kind == "count":
+            out_cols.append(Column(T.I64, outs[vidx]))
+        elif kind == "sumavg":
+            cidx, is_f, in_type = extra
+            acc = outs[vidx]
+            accv = acc.view(torch.float64) if is_f else acc
+            cnt = outs[cidx] if cidx is not None else None
+            if a.name in ("avg", "try_avg"):
+                if is_f and not isinstance(in_type, T.DecimalType):
+                    out_cols.append(_avg_result(accv, cnt, T.F64, a.dtype))
+                else:
+                    sums = accv
+                    if is_f and isinstance(in_type, T.DecimalType):
+                        sums = torch.round(accv).to(torch.int64)
+                    out_cols.append(_avg_result(sums, cnt, in_type, a.dtype))
+            else:
+                rt = a.dtype or (T.I64 if not isinstance(in_type, T.DecimalType) and not in_type.is_float else in_type)
+                if isinstance(in_type, T.DecimalType):
+                    data = torch.round(accv).to(torch.int64) if is_f else accv
+                    if isinstance(rt, T.DecimalType) and rt.scale != in_type.scale:
+                        from .eval import _rescale_int
+
+                        data = _rescale_int(data, in_type.scale, rt.scale)
+                    col = Column(rt, data)
+                elif in_type.is_float:
+                    col = Column(T.F64, accv.view(torch.float64) if not is_f else accv)
+                else:
+                    col = Column(T.I64, accv)
+                if cnt is not None:
+                    zero = cnt == 0
+                    if bool(zero.any()):
+                        col.validity = (~zero).to(torch.uint8)
+                out_cols.append(col)
+        else:  # minmax
+            op, c = extra
+            acc = outs[vidx]
+            sentinel = (1 << 63) - 1 if op == 3 else -(1 << 63)
+            valid = acc != sentinel
+            v = None if bool(valid.all()) else valid.to(torch.uint8)
+            if isinstance(c, StringColumn):
+                out_cols.append(StringColumn(c.offsets, c.bytes_, v, acc.to(torch.int32)))
+            else:
+                out_cols.append(Column(a.dtype or c.dtype,
+                                       acc.to(c.data.dtype) if c.data.dtype != torch.int64 else acc, v))
+    return out_cols

@@ -193,11 +193,16 @@ class Executor:
                 return Chunk(cols, [nm for nm, _ in p.schema])
             gid, ng = global_ids(n, dev)
             out_keys = []
-        agg_cols = []
-        for a in p.aggs:
-            args = [broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
-            fmask = self.ev.eval_mask(a.filter, child) if a.filter is not None else None
-            agg_cols.append(agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype))
+        args_list = [[broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
+                     for a in p.aggs]
+        fmasks = [self.ev.eval_mask(a.filter, child) if a.filter is not None else None
+                  for a in p.aggs]
+        from .aggregates import fused_agg_batch
+
+        agg_cols = fused_agg_batch(p.aggs, args_list, fmasks, gid, ng)
+        if agg_cols is None:
+            agg_cols = [agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype)
+                        for a, args, fmask in zip(p.aggs, args_list, fmasks)]
         return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
 
     def _dist_aggregate(self, p: S.Aggregate, child: Chunk) -> Chunk:
@@ -286,11 +291,16 @@ class Executor:
                 return Chunk(cols, [nm for nm, _ in p.schema])
             gid, ng = global_ids(n, dev)
             out_keys = []
-        agg_cols = []
-        for a in p.aggs:
-            args = [broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
-            fmask = self.ev.eval_mask(a.filter, child) if a.filter is not None else None
-            agg_cols.append(agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype))
+        args_list = [[broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
+                     for a in p.aggs]
+        fmasks = [self.ev.eval_mask(a.filter, child) if a.filter is not None else None
+                  for a in p.aggs]
+        from .aggregates import fused_agg_batch
+
+        agg_cols = fused_agg_batch(p.aggs, args_list, fmasks, gid, ng)
+        if agg_cols is None:
+            agg_cols = [agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype)
+                        for a, args, fmask in zip(p.aggs, args_list, fmasks)]
         return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
 
     # -- joins -------------------------------------------------------------

@@ -1,0 +1,295 @@
+// Grouped aggregation kernels for MI355X (gfx950).
+//
+// The hot path of TPC-H Q1-style aggregates: one pass over the value
+// columns, accumulating per group id. Two variants by group count G:
+//
+//  * grouped_acc_tiny  (G <= 8): per-thread REGISTER accumulators with
+//    predicated updates (`acc[g] += (gid==g) ? v : 0`, fully unrolled so
+//    every index is compile-time — runtime-indexed register arrays spill to
+//    scratch, guide rule #20), then a block reduction through LDS and one
+//    global atomic per (block, group, column).
+//  * grouped_acc_lds   (G <= 4096): LDS accumulator table with LDS atomics,
+//    flushed once per block. 160 KB LDS/CU covers 4096 groups x 4 columns.
+//
+// Larger G falls back to torch index_add_ (scattered global atomics are
+// fine once contention is low).
+//
+// Column ops: 0 = sum i64, 1 = sum f64, 2 = count (value ptr may be null),
+//             3 = min i64, 4 = max i64 (lds variant only).
+// All accumulators are 64-bit; i64 sums use two's-complement atomicAdd.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <cstdint>
+
+namespace {
+
+constexpr int kBlock = 256;
+
+inline int grid_for_n(int64_t n, int per_thread = 16) {
+  int64_t blocks = (n + (int64_t)kBlock * per_thread - 1) / ((int64_t)kBlock * per_thread);
+  return (int)std::max<int64_t>(1, std::min<int64_t>(blocks, 2048));
+}
+
+struct ColArg {
+  const void* ptr;   // value column (elem type per tag) or null for count
+  int op;            // 0 sum_i64, 1 sum_f64, 2 count, 3 min_i64, 4 max_i64
+  int elem;          // 0 i64, 1 f64, 2 i32, 3 f32, 4 u8/bool
+};
+
+__device__ inline long long load_as_i64(const void* p, int elem, int64_t i) {
+  switch (elem) {
+    case 0: return ((const long long*)p)[i];
+    case 2: return (long long)((const int*)p)[i];
+    case 4: return (long long)((const uint8_t*)p)[i];
+    default: return 0;
+  }
+}
+
+__device__ inline double load_as_f64(const void* p, int elem, int64_t i) {
+  switch (elem) {
+    case 1: return ((const double*)p)[i];
+    case 3: return (double)((const float*)p)[i];
+    case 0: return (double)((const long long*)p)[i];
+    case 2: return (double)((const int*)p)[i];
+    default: return 0.0;
+  }
+}
+
+// ---------------------------------------------------------------------
+// tiny-G: register accumulators, G <= TG (compile-time), NC columns
+// ---------------------------------------------------------------------
+template <int TG, int NC>
+__global__ void grouped_acc_tiny(const int32_t* __restrict__ gid,
+                                 const uint8_t* __restrict__ mask, int64_t n,
+                                 ColArg c0, ColArg c1, ColArg c2, ColArg c3,
+                                 ColArg c4, ColArg c5, ColArg c6, ColArg c7,
+                                 ColArg c8, ColArg c9,
+                                 int64_t* __restrict__ out /* [NC][G] */,
+                                 int G) {
+  ColArg cols[10] = {c0, c1, c2, c3, c4, c5, c6, c7, c8, c9};
+  // per-thread register accumulators: acc[c][g], all statically indexed
+  long long acc[NC][TG];
+#pragma unroll
+  for (int c = 0; c < NC; ++c)
+#pragma unroll
+    for (int g = 0; g < TG; ++g) acc[c][g] = 0;
+
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    if (mask && !mask[i]) continue;
+    int g = gid[i];
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      const ColArg& a = cols[c];
+      if (a.op == 0) {
+        long long v = load_as_i64(a.ptr, a.elem, i);
+#pragma unroll
+        for (int gg = 0; gg < TG; ++gg) acc[c][gg] += (g == gg) ? v : 0;
+      } else if (a.op == 1) {
+        double v = load_as_f64(a.ptr, a.elem, i);
+#pragma unroll
+        for (int gg = 0; gg < TG; ++gg) {
+          double cur = __longlong_as_double(acc[c][gg]);
+          cur += (g == gg) ? v : 0.0;
+          acc[c][gg] = __double_as_longlong(cur);
+        }
+      } else {  // count
+#pragma unroll
+        for (int gg = 0; gg < TG; ++gg) acc[c][gg] += (g == gg) ? 1 : 0;
+      }
+    }
+  }
+
+  // block reduction: LDS [NC][TG][waves]
+  __shared__ long long red[NC][TG][kBlock / 64];
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+#pragma unroll
+  for (int c = 0; c < NC; ++c) {
+#pragma unroll
+    for (int g = 0; g < TG; ++g) {
+      long long v = acc[c][g];
+      bool isf = (cols[c].op == 1);
+      // wave reduce
+      for (int off = 32; off > 0; off >>= 1) {
+        long long o = __shfl_down(v, off, 64);
+        if (isf)
+          v = __double_as_longlong(__longlong_as_double(v) + __longlong_as_double(o));
+        else
+          v += o;
+      }
+      if (lane == 0) red[c][g][wave] = v;
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x < TG * NC) {
+    int c = threadIdx.x / TG;
+    int g = threadIdx.x % TG;
+    if (g < G) {
+      bool isf = (cols[c].op == 1);
+      long long v = red[c][g][0];
+      for (int w = 1; w < kBlock / 64; ++w) {
+        if (isf)
+          v = __double_as_longlong(__longlong_as_double(v) + __longlong_as_double(red[c][g][w]));
+        else
+          v += red[c][g][w];
+      }
+      if (isf)
+        atomicAdd((double*)&out[(int64_t)c * G + g], __longlong_as_double(v));
+      else
+        atomicAdd((unsigned long long*)&out[(int64_t)c * G + g], (unsigned long long)v);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// LDS variant: G <= 4096, ncols runtime (LDS handles runtime indexing)
+// ---------------------------------------------------------------------
+__global__ void grouped_acc_lds(const int32_t* __restrict__ gid,
+                                const uint8_t* __restrict__ mask, int64_t n,
+                                ColArg c0, ColArg c1, ColArg c2, ColArg c3,
+                                ColArg c4, ColArg c5, ColArg c6, ColArg c7,
+                                ColArg c8, ColArg c9, int ncols,
+                                int64_t* __restrict__ out, int G) {
+  ColArg cols[10] = {c0, c1, c2, c3, c4, c5, c6, c7, c8, c9};
+  extern __shared__ long long lds[];  // [ncols][G]
+  for (int c = 0; c < ncols; ++c) {
+    long long init = 0;
+    if (cols[c].op == 3) init = INT64_MAX;
+    if (cols[c].op == 4) init = INT64_MIN;
+    for (int g = threadIdx.x; g < G; g += blockDim.x) lds[(int64_t)c * G + g] = init;
+  }
+  __syncthreads();
+
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    if (mask && !mask[i]) continue;
+    int g = gid[i];
+    for (int c = 0; c < ncols; ++c) {
+      const ColArg& a = cols[c];
+      long long* cell = &lds[(int64_t)c * G + g];
+      switch (a.op) {
+        case 0:
+          atomicAdd((unsigned long long*)cell,
+                    (unsigned long long)load_as_i64(a.ptr, a.elem, i));
+          break;
+        case 1:
+          atomicAdd((double*)cell, load_as_f64(a.ptr, a.elem, i));
+          break;
+        case 2:
+          atomicAdd((unsigned long long*)cell, 1ull);
+          break;
+        case 3:
+          atomicMin((long long*)cell, load_as_i64(a.ptr, a.elem, i));
+          break;
+        case 4:
+          atomicMax((long long*)cell, load_as_i64(a.ptr, a.elem, i));
+          break;
+      }
+    }
+  }
+  __syncthreads();
+  for (int c = 0; c < ncols; ++c) {
+    const ColArg& a = cols[c];
+    for (int g = threadIdx.x; g < G; g += blockDim.x) {
+      long long v = lds[(int64_t)c * G + g];
+      int64_t* cell = &out[(int64_t)c * G + g];
+      switch (a.op) {
+        case 0:
+        case 2:
+          if (v) atomicAdd((unsigned long long*)cell, (unsigned long long)v);
+          break;
+        case 1:
+          if (v) atomicAdd((double*)cell, __longlong_as_double(v));
+          break;
+        case 3:
+          if (v != INT64_MAX) atomicMin(reinterpret_cast<long long*>(cell), v);
+          break;
+        case 4:
+          if (v != INT64_MIN) atomicMax(reinterpret_cast<long long*>(cell), v);
+          break;
+      }
+    }
+  }
+}
+
+ColArg make_col(const c10::optional<torch::Tensor>& t, int op) {
+  ColArg a;
+  a.op = op;
+  a.ptr = nullptr;
+  a.elem = 0;
+  if (t.has_value() && t->defined()) {
+    auto& x = *t;
+    a.ptr = x.data_ptr();
+    switch (x.scalar_type()) {
+      case torch::kInt64: a.elem = 0; break;
+      case torch::kFloat64: a.elem = 1; break;
+      case torch::kInt32: a.elem = 2; break;
+      case torch::kFloat32: a.elem = 3; break;
+      case torch::kUInt8: a.elem = 4; break;
+      case torch::kBool: a.elem = 4; break;
+      default: TORCH_CHECK(false, "unsupported value dtype for grouped_acc");
+    }
+  }
+  return a;
+}
+
+}  // namespace
+
+// vals[i] may be undefined for count columns. Returns [ncols][G] int64 view
+// (f64 accumulators bit-cast).
+torch::Tensor grouped_acc(torch::Tensor gid, c10::optional<torch::Tensor> mask,
+                          std::vector<c10::optional<torch::Tensor>> vals,
+                          std::vector<int64_t> ops, int64_t G) {
+  TORCH_CHECK(gid.is_cuda() && gid.scalar_type() == torch::kInt32, "gid must be cuda int32");
+  int64_t n = gid.numel();
+  int ncols = (int)vals.size();
+  TORCH_CHECK(ncols >= 1 && ncols <= 10, "1..10 columns per launch");
+  auto out = torch::zeros({ncols, G}, gid.options().dtype(torch::kInt64));
+  // min/max initialization
+  for (int c = 0; c < ncols; ++c) {
+    if (ops[c] == 3) out[c].fill_(INT64_MAX);
+    if (ops[c] == 4) out[c].fill_(INT64_MIN);
+  }
+  if (n == 0) return out;
+  ColArg cols[10];
+  for (int c = 0; c < 10; ++c)
+    cols[c] = make_col(c < ncols ? vals[c] : c10::nullopt, c < ncols ? (int)ops[c] : 0);
+  const uint8_t* mptr = nullptr;
+  if (mask.has_value() && mask->defined()) {
+    TORCH_CHECK(mask->scalar_type() == torch::kBool || mask->scalar_type() == torch::kUInt8);
+    mptr = (const uint8_t*)mask->data_ptr();
+  }
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  bool tiny_ok = G <= 8;
+  for (int c = 0; c < ncols && tiny_ok; ++c)
+    if (ops[c] == 3 || ops[c] == 4) tiny_ok = false;
+  if (tiny_ok) {
+    int grid = grid_for_n(n);
+#define TINY_CASE(NCV)                                                          \
+  case NCV:                                                                     \
+    hipLaunchKernelGGL((grouped_acc_tiny<8, NCV>), dim3(grid), dim3(kBlock), 0, \
+                       stream, gid.data_ptr<int32_t>(), mptr, n, cols[0],       \
+                       cols[1], cols[2], cols[3], cols[4], cols[5], cols[6],    \
+                       cols[7], cols[8], cols[9],                               \
+                       out.data_ptr<int64_t>(), (int)G);                        \
+    break;
+    switch (ncols) {
+      TINY_CASE(1) TINY_CASE(2) TINY_CASE(3) TINY_CASE(4) TINY_CASE(5)
+      TINY_CASE(6) TINY_CASE(7) TINY_CASE(8) TINY_CASE(9) TINY_CASE(10)
+    }
+#undef TINY_CASE
+    return out;
+  }
+  TORCH_CHECK(G <= 4096, "grouped_acc: G too large for LDS variant");
+  size_t lds_bytes = (size_t)ncols * G * 8;
+  TORCH_CHECK(lds_bytes <= 160 * 1024, "grouped_acc: LDS budget exceeded");
+  int grid = grid_for_n(n, 8);
+  hipLaunchKernelGGL(grouped_acc_lds, dim3(grid), dim3(kBlock), lds_bytes, stream,
+                     gid.data_ptr<int32_t>(), mptr, n, cols[0], cols[1], cols[2],
+                     cols[3], cols[4], cols[5], cols[6], cols[7], cols[8], cols[9],
+                     ncols, out.data_ptr<int64_t>(), (int)G);
+  return out;
+}

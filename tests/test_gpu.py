@@ -103,3 +103,65 @@ def test_tpch_gpu_matches_cpu(tpch_pair, q):
                 assert gv == pytest.approx(wv, rel=1e-9, abs=1e-9), f"q{q} row {i}"
             else:
                 assert gv == wv, f"q{q} row {i}"
+
+
+def test_grouped_acc_kernel_matches_torch(ext):
+    torch.manual_seed(0)
+    n = 1_000_000
+    for G in (4, 7, 64, 1500):
+        gid = torch.randint(0, G, (n,), dtype=torch.int32, device="cuda")
+        vi = torch.randint(-1000, 1000, (n,), dtype=torch.int64, device="cuda")
+        vf = torch.rand(n, dtype=torch.float64, device="cuda")
+        mask = (torch.rand(n, device="cuda") < 0.7)
+        out = ext.grouped_acc(gid, mask, [vi, vf, None], [0, 1, 2], G)
+        g64 = gid.to(torch.int64)
+        m = mask
+        want_i = torch.zeros(G, dtype=torch.int64, device="cuda")
+        want_i.index_add_(0, g64[m], vi[m])
+        want_f = torch.zeros(G, dtype=torch.float64, device="cuda")
+        want_f.index_add_(0, g64[m], vf[m])
+        want_c = torch.zeros(G, dtype=torch.int64, device="cuda")
+        want_c.index_add_(0, g64[m], torch.ones(int(m.sum()), dtype=torch.int64, device="cuda"))
+        assert torch.equal(out[0], want_i), G
+        assert torch.allclose(out[1].view(torch.float64), want_f, rtol=1e-12), G
+        assert torch.equal(out[2], want_c), G
+        # min/max via lds variant
+        out2 = ext.grouped_acc(gid, None, [vi, vi], [3, 4], G)
+        wmin = torch.full((G,), 2**62, dtype=torch.int64, device="cuda")
+        wmin.scatter_reduce_(0, g64, vi, reduce="amin", include_self=True)
+        wmax = torch.full((G,), -2**62, dtype=torch.int64, device="cuda")
+        wmax.scatter_reduce_(0, g64, vi, reduce="amax", include_self=True)
+        assert torch.equal(out2[0], wmin), G
+        assert torch.equal(out2[1], wmax), G
+
+
+def test_fused_agg_end_to_end_gpu():
+    import sail_amd
+    from sail_amd.engine import types as T
+
+    s = sail_amd.SessionContext(device="cuda")
+    import random
+
+    rng = random.Random(3)
+    n = 200_000
+    ks = [rng.randint(0, 5) for _ in range(n)]
+    vs = [rng.randint(-100, 100) for _ in range(n)]
+    ds = [round(rng.uniform(0, 100), 2) for _ in range(n)]
+    s.create_dataframe({"k": ks, "v": vs, "d": ds},
+                       schema={"k": T.I32, "v": T.I64, "d": T.DecimalType(12, 2)}, name="t")
+    got = s.sql("SELECT k, sum(v), count(*), avg(d), min(v), max(v) FROM t GROUP BY k ORDER BY k").collect()
+    import collections
+
+    acc = collections.defaultdict(lambda: [0, 0, 0.0, 10**9, -10**9])
+    for k, v, d in zip(ks, vs, ds):
+        a = acc[k]
+        a[0] += v
+        a[1] += 1
+        a[2] += d
+        a[3] = min(a[3], v)
+        a[4] = max(a[4], v)
+    for row in got:
+        k, sv, cnt, avgd, mn, mx = row
+        a = acc[k]
+        assert sv == a[0] and cnt == a[1] and mn == a[3] and mx == a[4]
+        assert abs(avgd - a[2] / a[1]) < 1e-4

@@ -16,9 +16,13 @@ from .column import Column, StringColumn
 from .joins import normalize_key
 
 
-def group_ids(keys: List[Column]) -> Tuple[torch.Tensor, torch.Tensor, int]:
+def group_ids(keys: List[Column], mask: Optional[torch.Tensor] = None
+              ) -> Tuple[torch.Tensor, torch.Tensor, int]:
     """Returns (gid per row, representative row index per group, n_groups).
     Null keys form their own groups (SQL GROUP BY semantics).
+
+    `mask`: optional selection — groups are defined by masked-in rows only
+    (masked-out rows get an arbitrary gid the caller must also mask).
 
     Fast path: when the combined key domain is dense (dict codes, dense
     surrogate keys like l_orderkey) the gids come from a presence-bitmap +
@@ -54,17 +58,27 @@ def group_ids(keys: List[Column]) -> Tuple[torch.Tensor, torch.Tensor, int]:
             for i in range(1, len(norm)):
                 packed = packed * spans[i] + (norm[i] - mins[i])
             present = torch.zeros(total, dtype=torch.bool, device=dev)
-            present[packed] = True
+            sel = packed if mask is None else packed[mask]
+            present[sel] = True
             lut = torch.cumsum(present.to(torch.int32), 0) - 1
             gid = lut.index_select(0, packed).to(torch.int64)
-            ng = int(lut[-1].item()) + 1
+            ng = int(present.sum().item())
             rep = torch.full((ng,), n, dtype=torch.int64, device=dev)
-            rep.scatter_reduce_(0, gid, torch.arange(n, device=dev), reduce="amin",
-                                include_self=True)
+            rows = torch.arange(n, device=dev)
+            if mask is None:
+                rep.scatter_reduce_(0, gid, rows, reduce="amin", include_self=True)
+            else:
+                rep.scatter_reduce_(0, gid[mask], rows[mask], reduce="amin", include_self=True)
             return gid, rep, ng
 
+    if mask is not None and len(norm) > 1:
+        # sparse multi-key domains can't cheaply id all rows against the
+        # selected-row unique set; caller falls back to materialization
+        raise MaskedGroupsUnsupported()
+    sel_norm = norm if mask is None else [k[mask] for k in norm]
     if len(norm) == 1:
-        uniq, gid = torch.unique(norm[0], return_inverse=True)
+        uniq, _ = torch.unique(sel_norm[0], return_inverse=True)
+        gid = torch.searchsorted(uniq, norm[0]).clamp(0, uniq.shape[0] - 1)
         ng = int(uniq.shape[0])
     else:
         stacked = torch.stack(norm, dim=1)
@@ -72,8 +86,16 @@ def group_ids(keys: List[Column]) -> Tuple[torch.Tensor, torch.Tensor, int]:
         ng = int(uniq.shape[0])
     # representative row per group (first occurrence for determinism)
     rep = torch.full((ng,), n, dtype=torch.int64, device=dev)
-    rep.scatter_reduce_(0, gid, torch.arange(n, device=dev), reduce="amin", include_self=True)
+    rows = torch.arange(n, device=dev)
+    if mask is None:
+        rep.scatter_reduce_(0, gid, rows, reduce="amin", include_self=True)
+    else:
+        rep.scatter_reduce_(0, gid[mask], rows[mask], reduce="amin", include_self=True)
     return gid, rep, ng
+
+
+class MaskedGroupsUnsupported(Exception):
+    """group_ids(mask=...) fallback signal for sparse multi-key domains."""
 
 
 def _masked(values: Column, extra_mask: Optional[torch.Tensor]):

@@ -174,6 +174,13 @@ class Executor:
 
     # -- aggregate ---------------------------------------------------------
     def _x_Aggregate(self, p: S.Aggregate) -> Chunk:
+        # Aggregate∘[Project]∘Filter fusion: evaluate group keys and agg
+        # inputs on the UNFILTERED child and pass the selection mask into the
+        # aggregation — avoids materializing high-selectivity filters (Q1
+        # keeps 98.6% of lineitem; the gather costs more than the aggregate).
+        fused = self._try_masked_aggregate(p)
+        if fused is not None:
+            return fused
         child = self.execute(p.input)
         if self.dctx is not None and child.partitioning == "sharded":
             return self._dist_aggregate(p, child)
@@ -199,6 +206,92 @@ class Executor:
                   for a in p.aggs]
         from .aggregates import fused_agg_batch
 
+        agg_cols = fused_agg_batch(p.aggs, args_list, fmasks, gid, ng)
+        if agg_cols is None:
+            agg_cols = [agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype)
+                        for a, args, fmask in zip(p.aggs, args_list, fmasks)]
+        return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
+
+    def _try_masked_aggregate(self, p: S.Aggregate) -> Optional[Chunk]:
+        from ..plan.rules.util import substitute_refs
+        from .aggregates import MaskedGroupsUnsupported, fused_agg_batch
+
+        node = p.input
+        proj = None
+        if isinstance(node, S.Project) and not any(
+                isinstance(e, (S.ScalarSubquery, S.Exists, S.InSubquery))
+                for pe in node.exprs for e in pe.walk()):
+            f = node.input
+            if not isinstance(f, S.Filter):
+                return None
+            proj = node
+        elif isinstance(node, S.Filter):
+            f = node
+        else:
+            return None
+        if self.dctx is not None:
+            return None  # distributed path handles its own exchange first
+
+        base = self.execute(f.input)
+        n = base.num_rows
+        dev = base.device
+        if n == 0:
+            return None  # empty: use the standard path
+
+        mask = self.ev.eval_mask(f.condition, base)
+
+        def rebased(e: S.Expr) -> S.Expr:
+            return substitute_refs(e, proj.exprs) if proj is not None else e
+
+        # selectivity decides: high -> keep the mask (no materialization);
+        # low -> gather only the referenced columns once and drop the mask
+        n_pass = int(mask.sum().item())
+        if n_pass == 0:
+            return None
+        if n_pass < 0.2 * n:
+            from ..plan.rules.util import expr_refs
+
+            refs = set()
+            for g in p.group_by:
+                refs |= expr_refs(rebased(g))
+            for a in p.aggs:
+                for x in a.args:
+                    refs |= expr_refs(rebased(x))
+                if a.filter is not None:
+                    refs |= expr_refs(rebased(a.filter))
+            idx = torch.nonzero(mask, as_tuple=False).squeeze(1)
+            cols: List[Optional[Column]] = [None] * len(base.columns)
+            for i in refs:
+                cols[i] = base.columns[i].gather(idx)
+            base = Chunk(cols, list(base.names), base.partitioning)
+            base.forced_rows = n_pass
+            n = n_pass
+            mask = None
+
+        try:
+            if p.group_by:
+                key_exprs = [rebased(g) for g in p.group_by]
+                key_cols = [broadcast(self.ev.eval(g, base), n, dev) for g in key_exprs]
+                gid, rep, ng = group_ids(key_cols, mask=mask)
+                out_keys = [c.gather(rep) for c in key_cols]
+            else:
+                gid, ng = global_ids(n, dev)
+                out_keys = []
+        except MaskedGroupsUnsupported:
+            return None
+
+        args_list = []
+        fmasks = []
+        for a in p.aggs:
+            args_list.append([broadcast(self.ev.eval(rebased(x), base), n, dev)
+                              for x in a.args])
+            if a.filter is not None:
+                fm = self.ev.eval_mask(rebased(a.filter), base)
+                if mask is not None:
+                    fm = fm & mask
+            else:
+                fm = mask
+            fmasks.append(fm)
         agg_cols = fused_agg_batch(p.aggs, args_list, fmasks, gid, ng)
         if agg_cols is None:
             agg_cols = [agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype)

@@ -248,7 +248,9 @@ class Executor:
         n_pass = int(mask.sum().item())
         if n_pass == 0:
             return None
-        if n_pass < 0.2 * n:
+
+        def materialize():
+            nonlocal base, n, mask
             from ..plan.rules.util import expr_refs
 
             refs = set()
@@ -268,17 +270,26 @@ class Executor:
             n = n_pass
             mask = None
 
+        # CPU reference path always materializes (masked per-agg indexing is
+        # slower there); on GPU keep the mask unless selectivity is low
+        if n_pass < 0.2 * n or not base.device.type == "cuda":
+            materialize()
+
+        key_cols = None
+        if p.group_by:
+            key_exprs = [rebased(g) for g in p.group_by]
+            key_cols = [broadcast(self.ev.eval(g, base), n, dev) for g in key_exprs]
         try:
-            if p.group_by:
-                key_exprs = [rebased(g) for g in p.group_by]
-                key_cols = [broadcast(self.ev.eval(g, base), n, dev) for g in key_exprs]
+            if key_cols is not None:
                 gid, rep, ng = group_ids(key_cols, mask=mask)
-                out_keys = [c.gather(rep) for c in key_cols]
             else:
                 gid, ng = global_ids(n, dev)
-                out_keys = []
         except MaskedGroupsUnsupported:
-            return None
+            materialize()
+            key_cols = [broadcast(self.ev.eval(g, base), n, dev)
+                        for g in [rebased(g) for g in p.group_by]]
+            gid, rep, ng = group_ids(key_cols, mask=None)
+        out_keys = [c.gather(rep) for c in key_cols] if key_cols is not None else []
 
         args_list = []
         fmasks = []

@@ -119,8 +119,11 @@ class Executor:
         # Project∘Filter fusion: evaluate the predicate on the unfiltered
         # child, then gather ONLY the columns the projection references —
         # predicate-only columns (e.g. o_comment in q13) are never gathered.
-        if isinstance(p.input, S.Filter):
-            f = p.input
+        fin = p.input
+        while isinstance(fin, S.SubqueryAlias):
+            fin = fin.input
+        if isinstance(fin, S.Filter):
+            f = fin
             base = self.execute(f.input)
             if base.num_rows == 0:
                 child = base
@@ -217,11 +220,15 @@ class Executor:
         from .aggregates import MaskedGroupsUnsupported, fused_agg_batch
 
         node = p.input
+        while isinstance(node, S.SubqueryAlias):  # identity wrappers
+            node = node.input
         proj = None
         if isinstance(node, S.Project) and not any(
                 isinstance(e, (S.ScalarSubquery, S.Exists, S.InSubquery))
                 for pe in node.exprs for e in pe.walk()):
             f = node.input
+            while isinstance(f, S.SubqueryAlias):
+                f = f.input
             if not isinstance(f, S.Filter):
                 return None
             proj = node

@@ -35,6 +35,7 @@ class ExecutionContext:
         self.session = session
         self.device = torch.device(device)
         self._subquery_cache: Dict[int, Scalar] = {}
+        self._cte_cache: Dict[int, Chunk] = {}
 
     def execute_scalar_subquery(self, e: S.ScalarSubquery) -> Scalar:
         if id(e) not in self._subquery_cache:
@@ -69,10 +70,18 @@ class Executor:
         return out
 
     def execute(self, plan: S.Plan) -> Chunk:
+        ck = plan.__dict__.get("_cte_cache_key") if hasattr(plan, "__dict__") else None
+        if ck is not None:
+            cached = self.ctx._cte_cache.get(ck)
+            if cached is not None:
+                return cached
         m = getattr(self, "_x_" + type(plan).__name__, None)
         if m is None:
             raise ExecError(f"cannot execute {type(plan).__name__}")
-        return m(plan)
+        out = m(plan)
+        if ck is not None:
+            self.ctx._cte_cache[ck] = out
+        return out
 
     # -- leaves ------------------------------------------------------------
     def _x_Read(self, p: S.Read) -> Chunk:

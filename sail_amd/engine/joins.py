@@ -123,6 +123,11 @@ def _expand_matches(bids: torch.Tensor, pids: torch.Tensor,
         bids = torch.where(b_valid, bids, torch.full_like(bids, -(2 ** 63) + 1))
     if p_valid is not None:
         pids = torch.where(p_valid, pids, torch.full_like(pids, -(2 ** 63) + 2))
+    if bids.is_cuda and nb > 0:
+        from ..ops import kernels as K
+
+        if K.available():
+            return _expand_matches_gpu(bids, pids, K.require())
     order = torch.argsort(bids)
     bsorted = bids[order]
     lo = torch.searchsorted(bsorted, pids, right=False)
@@ -246,3 +251,15 @@ def raw_string_key(c: StringColumn, force_hash: bool = False) -> torch.Tensor:
     if not force_hash and max_len <= 7:
         return short_string_key(c)
     return fnv_key_tensor(c)
+
+
+def _expand_matches_gpu(bids: torch.Tensor, pids: torch.Tensor, ext):
+    """HIP open-addressing hash join (ops/csrc/hash_join.hip): chain build +
+    two-phase count/fill probe. Same contract as the sort-based matcher."""
+    tkeys, theads, nxt = ext.hj_build_chain(bids)
+    counts32 = ext.hj_probe_count(tkeys, theads, nxt, pids)
+    counts = counts32.to(torch.int64)
+    offsets = torch.cumsum(counts, 0) - counts
+    total = int((offsets[-1] + counts[-1]).item()) if counts.numel() else 0
+    probe_idx, build_idx = ext.hj_probe_fill(tkeys, theads, nxt, pids, offsets, total)
+    return probe_idx, build_idx, counts

@@ -10,6 +10,15 @@ std::vector<torch::Tensor> substr_fixed(torch::Tensor offsets, torch::Tensor byt
 torch::Tensor grouped_acc(torch::Tensor gid, c10::optional<torch::Tensor> mask,
                           std::vector<c10::optional<torch::Tensor>> vals,
                           std::vector<int64_t> ops, int64_t G);
+// hash_join.hip
+std::vector<torch::Tensor> hj_build(torch::Tensor keys);
+torch::Tensor hj_probe_unique(torch::Tensor tkeys, torch::Tensor tvals, torch::Tensor keys);
+std::vector<torch::Tensor> hj_build_chain(torch::Tensor keys);
+torch::Tensor hj_probe_count(torch::Tensor tkeys, torch::Tensor theads,
+                             torch::Tensor next, torch::Tensor keys);
+std::vector<torch::Tensor> hj_probe_fill(torch::Tensor tkeys, torch::Tensor theads,
+                                         torch::Tensor next, torch::Tensor keys,
+                                         torch::Tensor offsets, int64_t total);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sail_amd MI355X (gfx950) kernels";
@@ -18,4 +27,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("substr_fixed", &substr_fixed, "fixed-length substring extraction");
   m.def("grouped_acc", &grouped_acc,
         "fused grouped accumulation (sum/count/min/max) for small group counts");
+  m.def("hj_build", &hj_build, "hash join build (unique keys)");
+  m.def("hj_probe_unique", &hj_probe_unique, "hash join probe, first match");
+  m.def("hj_build_chain", &hj_build_chain, "hash join build with chains");
+  m.def("hj_probe_count", &hj_probe_count, "hash join probe, count matches");
+  m.def("hj_probe_fill", &hj_probe_fill, "hash join probe, emit match pairs");
 }

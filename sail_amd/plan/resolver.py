@@ -74,8 +74,11 @@ class Resolver:
         # CTE reference?
         for scope in reversed(self.cte_scope):
             if p.table.lower() in scope:
-                sub = copy.deepcopy(scope[p.table.lower()])
+                original = scope[p.table.lower()]
+                sub = copy.deepcopy(original)
                 resolved = self._plan(sub, None) if sub.schema is None else sub
+                # every use of this CTE shares one execution-cache token
+                resolved.__dict__["_cte_cache_key"] = id(original)
                 aliased = S.SubqueryAlias(input=resolved, alias=p.table)
                 aliased.schema = resolved.schema
                 return self._qualify(aliased, p.table)

@@ -65,7 +65,8 @@ def _flatten(plan: S.Plan, leaves: List[S.Plan], conds: List[S.Expr], offset: in
     """Collect inner-join leaves in order; returns total width. Conjunct
     indices are valid in the concatenation of leaf schemas (schema concat is
     associative for inner joins)."""
-    if isinstance(plan, S.Join) and plan.how in ("inner", "cross"):
+    if isinstance(plan, S.Join) and plan.how in ("inner", "cross") \
+            and plan.__dict__.get("_cte_cache_key") is None:
         lw = _flatten(plan.left, leaves, conds, offset)
         rw = _flatten(plan.right, leaves, conds, offset + lw)
         if plan.on is not None:

@@ -64,6 +64,22 @@ def _prune(plan: S.Plan, needed: Set[int]) -> Tuple[S.Plan, Dict[int, int]]:
     # (count(*)-only aggregates, literal-only projections)
     if not needed and plan.schema:
         needed = {0}
+    if plan.__dict__.get("_cte_cache_key") is not None:
+        # shared CTE body: every use must see the identical (full) schema
+        full = set(range(len(plan.schema)))
+        if needed != full:
+            key = plan.__dict__["_cte_cache_key"]
+            plan.__dict__["_cte_cache_key"] = None
+            out, mapping = _prune(plan, full)
+            out.__dict__["_cte_cache_key"] = key
+            plan.__dict__["_cte_cache_key"] = key
+            return out, mapping
+        key = plan.__dict__["_cte_cache_key"]
+        plan.__dict__["_cte_cache_key"] = None
+        out, mapping = _prune(plan, full)
+        out.__dict__["_cte_cache_key"] = key
+        plan.__dict__["_cte_cache_key"] = key
+        return out, mapping
     if isinstance(plan, S.Project):
         keep = sorted(needed)
         kept_exprs = [_remap_subqueries(plan.exprs[i]) for i in keep]

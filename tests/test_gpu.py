@@ -165,3 +165,28 @@ def test_fused_agg_end_to_end_gpu():
         a = acc[k]
         assert sv == a[0] and cnt == a[1] and mn == a[3] and mx == a[4]
         assert abs(avgd - a[2] / a[1]) < 1e-4
+
+
+def test_hash_join_kernels_match_sort_join(ext):
+    torch.manual_seed(1)
+    for nb, np_, kr in ((1000, 5000, 300), (100000, 400000, 50000), (7, 3, 4)):
+        bids = torch.randint(0, kr, (nb,), dtype=torch.int64, device="cuda")
+        pids = torch.randint(0, kr, (np_,), dtype=torch.int64, device="cuda")
+        from sail_amd.engine.joins import _expand_matches_gpu
+
+        p_idx, b_idx, counts = _expand_matches_gpu(bids, pids, ext)
+        # reference: sort-based matcher on CPU
+        bc, pc = bids.cpu(), pids.cpu()
+        order = torch.argsort(bc)
+        bs = bc[order]
+        lo = torch.searchsorted(bs, pc, right=False)
+        hi = torch.searchsorted(bs, pc, right=True)
+        want_counts = (hi - lo)
+        assert torch.equal(counts.cpu(), want_counts)
+        # pair set equality (order within a key is arbitrary)
+        got = set(zip(p_idx.cpu().tolist(), b_idx.cpu().tolist()))
+        want = set()
+        for i in range(np_):
+            for j in range(int(lo[i]), int(hi[i])):
+                want.add((i, int(order[j])))
+        assert got == want, (nb, np_, kr)

@@ -58,6 +58,18 @@ class Executor:
         self.ev = Evaluator(ctx)
         d = getattr(ctx.session, "dist", None)
         self.dctx = d if (d is not None and d.world > 1) else None
+        # tracer is shared across nested executors (subqueries) via the ctx
+        if not hasattr(ctx, "tracer"):
+            ctx.tracer = None
+            import os
+
+            if os.environ.get("SAIL_TRACE") == "1" or \
+                    ctx.session.conf.get("sail.trace") == "true":
+                from ..utils.trace import Tracer
+
+                ctx.tracer = Tracer(ctx.device)
+                ctx.session.last_trace = ctx.tracer.trace
+        self.tracer = ctx.tracer
 
     def _gather(self, chunk: Chunk) -> Chunk:
         """Replicate a sharded chunk on every rank (all_gather over RCCL)."""
@@ -78,7 +90,17 @@ class Executor:
         m = getattr(self, "_x_" + type(plan).__name__, None)
         if m is None:
             raise ExecError(f"cannot execute {type(plan).__name__}")
-        out = m(plan)
+        if self.tracer is not None:
+            detail = ""
+            if isinstance(plan, S.Read):
+                detail = plan.table
+            elif isinstance(plan, S.Join):
+                detail = plan.how
+            elif isinstance(plan, S.Aggregate):
+                detail = f"{len(plan.group_by)}keys/{len(plan.aggs)}aggs"
+            out = self.tracer.wrap(type(plan).__name__, detail, lambda: m(plan))
+        else:
+            out = m(plan)
         if ck is not None:
             self.ctx._cte_cache[ck] = out
         return out

@@ -43,7 +43,11 @@ def decorrelate(plan: S.Plan) -> S.Plan:
         _decorrelate_nested(e)
 
     if isinstance(plan, S.Filter):
-        return _rewrite_filter(plan)
+        key = plan.__dict__.get("_cte_cache_key")
+        out = _rewrite_filter(plan)
+        if key is not None:
+            out.__dict__["_cte_cache_key"] = key
+        return out
     return plan
 
 

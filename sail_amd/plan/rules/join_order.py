@@ -28,7 +28,13 @@ def _walk(plan: S.Plan, stats) -> S.Plan:
     # rewrite subquery plans too
     for e in _exprs(plan):
         _walk_expr(e, stats)
+    key = plan.__dict__.get("_cte_cache_key")
     if isinstance(plan, S.Join) and plan.how in ("inner", "cross"):
+        out = _reorder_tree(plan, stats)
+        if key is not None:
+            out.__dict__["_cte_cache_key"] = key
+        return out
+    if False:
         return _reorder_tree(plan, stats)
     for attr in ("input", "left", "right"):
         child = getattr(plan, attr, None)

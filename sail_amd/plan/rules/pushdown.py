@@ -33,10 +33,16 @@ def _is_volatile(e: S.Expr) -> bool:
 def _push(plan: S.Plan, conds: List[S.Expr]) -> S.Plan:
     """Push the list of predicates (bound against `plan`'s schema) into the
     subtree; apply any that can't move as a Filter on the result."""
-    if plan.__dict__.get("_cte_cache_key") is not None and conds:
+    key = plan.__dict__.get("_cte_cache_key")
+    if key is not None:
         # shared CTE body: rewrites inside must not depend on the use site
-        # (all uses share one executed result)
-        return _apply(_push(plan, []), conds)
+        # (all uses share one executed result) and the cache marker must
+        # survive node rebuilding
+        plan.__dict__["_cte_cache_key"] = None
+        out = _push(plan, [])
+        plan.__dict__["_cte_cache_key"] = key
+        out.__dict__["_cte_cache_key"] = key
+        return _apply(out, conds)
     if isinstance(plan, S.Filter):
         newconds = []
         for c in split_conjuncts(plan.condition):

@@ -66,21 +66,48 @@ def group_ids(keys: List[Column], mask: Optional[torch.Tensor] = None
             rep = _any_representative(gid, n, ng, mask, dev)
             return gid, rep, ng
 
-    if mask is not None and len(norm) > 1:
-        # sparse multi-key domains can't cheaply id all rows against the
-        # selected-row unique set; caller falls back to materialization
-        raise MaskedGroupsUnsupported()
-    sel_norm = norm if mask is None else [k[mask] for k in norm]
-    if len(norm) == 1:
-        uniq, _ = torch.unique(sel_norm[0], return_inverse=True)
-        gid = torch.searchsorted(uniq, norm[0]).clamp(0, uniq.shape[0] - 1)
-        ng = int(uniq.shape[0])
+    # sparse domain: reduce multi-key to ONE int64 — exact range packing when
+    # the combined span fits, otherwise a mixed 64-bit hash combine (collision
+    # odds ~n^2/2^64; documented engine tradeoff, exact for <=2 keys in range)
+    if len(norm) > 1:
+        packed1 = _pack_or_hash(norm)
     else:
-        stacked = torch.stack(norm, dim=1)
-        uniq, gid = torch.unique(stacked, dim=0, return_inverse=True)
-        ng = int(uniq.shape[0])
+        packed1 = norm[0]
+    sel = packed1 if mask is None else packed1[mask]
+    uniq, _ = torch.unique(sel, return_inverse=True)
+    gid = torch.searchsorted(uniq, packed1).clamp(0, max(uniq.shape[0] - 1, 0))
+    ng = int(uniq.shape[0])
     rep = _any_representative(gid, n, ng, mask, dev)
     return gid, rep, ng
+
+
+def _pack_or_hash(norm: List[torch.Tensor]) -> torch.Tensor:
+    """Combine multiple int64 key columns into one: exact range packing when
+    the span product fits int64, else a murmur-style mixed hash combine."""
+    mins, spans = [], []
+    total = 1
+    ok = True
+    for k in norm:
+        lo = int(k.min().item())
+        hi = int(k.max().item())
+        mins.append(lo)
+        spans.append(hi - lo + 1)
+        total *= spans[-1]
+        if total > (1 << 62):
+            ok = False
+            break
+    if ok:
+        acc = norm[0] - mins[0]
+        for i in range(1, len(norm)):
+            acc = acc * spans[i] + (norm[i] - mins[i])
+        return acc
+    acc = norm[0].clone()
+    for i in range(1, len(norm)):
+        acc = acc * 31 + norm[i]
+        mixed = acc ^ (acc >> 33)
+        mixed = mixed * -49064778989728563  # 0xFF51AFD7ED558CCD signed
+        acc = mixed ^ (mixed >> 33)
+    return acc
 
 
 def _any_representative(gid, n, ng, mask, dev):

@@ -25,6 +25,7 @@ from .rules.decorrelate import decorrelate
 from .rules.pushdown import pushdown_filters
 from .rules.prune import prune_columns
 from .rules.join_order import reorder_joins
+from .rules.semi_sink import sink_semi_joins
 
 
 def optimize(plan: S.Plan, enable_join_reorder: bool = True, stats=None) -> S.Plan:
@@ -32,5 +33,6 @@ def optimize(plan: S.Plan, enable_join_reorder: bool = True, stats=None) -> S.Pl
     plan = pushdown_filters(plan)
     if enable_join_reorder:
         plan = reorder_joins(plan, stats)
+    plan = sink_semi_joins(plan)
     plan = prune_columns(plan)
     return plan

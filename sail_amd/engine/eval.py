@@ -156,6 +156,25 @@ class Evaluator:
             out = (lv & rv & eq) | (~lv & ~rv)
             return Column(T.BOOL, out, None)
         if e.op in ("=", "!=", "<", "<=", ">", ">="):
+            # column-vs-scalar fast path: tensor-op-pyscalar, no broadcast
+            for col, sc, op in ((l, r, e.op), (r, l, _flip_cmp(e.op))):
+                if isinstance(col, Column) and isinstance(sc, Scalar) \
+                        and not isinstance(col, StringColumn) and not sc.is_null \
+                        and not isinstance(sc.value, str):
+                    v = sc.value
+                    if isinstance(v, _dt.date):
+                        v = (v - _dt.date(1970, 1, 1)).days
+                    if isinstance(col.dtype, T.DecimalType):
+                        # rescale the scalar exactly to the column's scale
+                        v = _to_scaled(v, col.dtype.scale)
+                    elif isinstance(v, float) and not col.data.dtype.is_floating_point:
+                        # int column vs fractional literal: fall through to
+                        # the generic coerced path
+                        if v != int(v):
+                            break
+                        v = int(v)
+                    data = _cmp_tensor_scalar(col.data, v, op)
+                    return Column(T.BOOL, data, col.validity)
             lc, rc = broadcast(l, n, dev), broadcast(r, n, dev)
             data = _cmp_data(lc, rc, e.op)
             return Column(T.BOOL, data, _merge_validity(lc, rc))
@@ -168,6 +187,10 @@ class Evaluator:
         r = self.eval(e.right, chunk)
         n, dev = chunk.num_rows, chunk.device
         lc, rc = broadcast(l, n, dev), broadcast(r, n, dev)
+        if lc.validity is None and rc.validity is None:
+            ld = lc.data if lc.data.dtype == torch.bool else lc.data != 0
+            rd = rc.data if rc.data.dtype == torch.bool else rc.data != 0
+            return Column(T.BOOL, ld & rd if e.op == "and" else ld | rd, None)
         lv, rv = lc.valid_mask(), rc.valid_mask()
         ld = lc.data.to(torch.bool)
         rd = rc.data.to(torch.bool)
@@ -323,6 +346,20 @@ class Evaluator:
         from .functions_impl import dispatch_function
 
         args = [self.eval(a, chunk) for a in e.args]
+        if args and all(isinstance(a, Scalar) for a in args) \
+                and e.name not in ("rand", "randn", "uuid", "monotonically_increasing_id"):
+            # constant folding: evaluate once on a 1-row chunk
+            one = Chunk([], [], chunk.partitioning)
+            one.forced_rows = 1
+            out = dispatch_function(e.name, args, e.dtype, one, self)
+            if isinstance(out, Scalar):
+                return out
+            vals = out.to_pylist()
+            v = vals[0] if vals else None
+            if isinstance(e.dtype, T.DecimalType) and v is not None:
+                # to_pylist already unscaled; keep as float for rebroadcast
+                return Scalar(v, e.dtype)
+            return Scalar(v, e.dtype or out.dtype)
         return dispatch_function(e.name, args, e.dtype, chunk, self)
 
 
@@ -343,6 +380,24 @@ def _retype_cmp(e: S.BinaryOp):
 # ---------------------------------------------------------------------------
 # kernels (torch fallback implementations)
 # ---------------------------------------------------------------------------
+
+def _flip_cmp(op: str) -> str:
+    return {"=": "=", "!=": "!=", "<": ">", "<=": ">=", ">": "<", ">=": "<="}[op]
+
+
+def _cmp_tensor_scalar(x: torch.Tensor, v, op: str) -> torch.Tensor:
+    if op == "=":
+        return x == v
+    if op == "!=":
+        return x != v
+    if op == "<":
+        return x < v
+    if op == "<=":
+        return x <= v
+    if op == ">":
+        return x > v
+    return x >= v
+
 
 def _merge_validity(a: Column, b: Column) -> Optional[torch.Tensor]:
     if a.validity is None and b.validity is None:

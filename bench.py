@@ -26,6 +26,8 @@ def main():
     ap.add_argument("--steps", type=int, default=2)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--sf", type=float, default=float(os.environ.get("SAIL_BENCH_SF", "100")))
+    ap.add_argument("--workload", default="tpch", choices=["tpch", "clickbench"])
+    ap.add_argument("--rows", type=int, default=100_000_000, help="clickbench rows")
     ap.add_argument("--device", default=None)
     ap.add_argument("--queries", default=None, help="comma-separated subset, e.g. 1,6,13")
     ap.add_argument("--print-times", action="store_true")
@@ -54,9 +56,14 @@ def main():
         dist = tdist
 
     import sail_amd
-    from sail_amd.datagen.tpch import register_tpch
-    from sail_amd.datagen.tpch_queries import QUERIES
     from sail_amd.exec.context import DistContext
+
+    if args.workload == "clickbench":
+        from sail_amd.datagen.clickbench import register_clickbench
+        from sail_amd.datagen.clickbench_queries import QUERIES as CB
+        QUERIES = {i: q for i, q in enumerate(CB)}
+    else:
+        from sail_amd.datagen.tpch_queries import QUERIES
 
     if use_cuda:
         from sail_amd.ops import kernels as K
@@ -68,16 +75,27 @@ def main():
         session.dist = DistContext(dist, rank=rank, world=world, device=device)
 
     t0 = time.time()
-    register_tpch(session, sf=args.sf, device=device, rank=rank, world=world)
+    if args.workload == "clickbench":
+        register_clickbench(session, rows=args.rows, device=device, rank=rank, world=world)
+    else:
+        from sail_amd.datagen.tpch import register_tpch
+
+        register_tpch(session, sf=args.sf, device=device, rank=rank, world=world)
     if use_cuda:
         torch.cuda.synchronize()
     gen_s = time.time() - t0
 
-    qids = [int(x) for x in args.queries.split(",")] if args.queries else list(range(1, 23))
+    if args.queries:
+        qids = [int(x) for x in args.queries.split(",")]
+    elif args.workload == "clickbench":
+        qids = list(range(len(QUERIES)))
+    else:
+        qids = list(range(1, 23))
     plans = {q: session.plan_sql(QUERIES[q]) for q in qids}
 
-    total_rows = sum(session.catalog.table_rows(t) or 0
-                     for t in ["lineitem", "orders", "customer", "part", "partsupp", "supplier"])
+    bench_tables = (["hits"] if args.workload == "clickbench" else
+                    ["lineitem", "orders", "customer", "part", "partsupp", "supplier"])
+    total_rows = sum(session.catalog.table_rows(t) or 0 for t in bench_tables)
     if dist is not None:
         tr = torch.tensor([total_rows], dtype=torch.int64,
                           device=device if use_cuda else "cpu")
@@ -127,9 +145,16 @@ def main():
             for q in qids:
                 print(f"# q{q}: {qtimes[q]*1000:.1f} ms", file=sys.stderr)
         print(f"# datagen: {gen_s:.1f}s, rows/GPU: {total_rows // max(world,1)}", file=sys.stderr)
-        baseline = 52.81 if abs(args.sf - 100.0) < 1e-6 else None
+        if args.workload == "clickbench":
+            baseline = None
+            metric = f"clickbench_{args.rows//1_000_000}m_total_s"
+            model = "ClickBench (43 queries)"
+        else:
+            baseline = 52.81 if abs(args.sf - 100.0) < 1e-6 else None
+            metric = f"tpch_sf{args.sf:g}_total_s"
+            model = "derived TPC-H (22 queries)"
         out = {
-            "metric": f"tpch_sf{args.sf:g}_total_s",
+            "metric": metric,
             "value": round(per_step, 4),
             "unit": "s",
             "n_gpus": n_gpus,
@@ -142,8 +167,9 @@ def main():
             "dtype": "exact-int64+fp64",
             "data": "synthetic",
             "config": {
-                "model": "derived TPC-H (22 queries)",
-                "sf": args.sf,
+                "model": model,
+                "sf": args.sf if args.workload == "tpch" else None,
+                "rows": args.rows if args.workload == "clickbench" else None,
                 "queries": len(qids),
                 "parallelism": f"sharded dp{n_gpus}" if n_gpus > 1 else "single-gpu",
             },

@@ -668,3 +668,40 @@ _IMPLS = {
     "string": _cast_fn(T.STRING), "date": _cast_fn(T.DATE),
     "timestamp": _cast_fn(T.TIMESTAMP), "boolean": _cast_fn(T.BOOL),
 }
+
+
+def _f_regexp_replace(args, out, chunk, ev):
+    import re as _re
+
+    c = _col(args[0], chunk)
+    pat = _scalarize(args[1]).value
+    repl = _scalarize(args[2]).value
+    # SQL backreferences \1 -> python \1 (compatible); $1 also accepted
+    py_repl = repl.replace("$1", "\\1").replace("$2", "\\2")
+    rx = _re.compile(pat)
+
+    def f(v):
+        return rx.sub(py_repl, v)
+
+    if isinstance(c, StringColumn) and c.is_dict:
+        vals = [f(v) for v in c.dict_values()]
+        # transformed values may collide: re-dictionary (sorted) + remap codes
+        uniq = sorted(set(vals))
+        idx = {s: i for i, s in enumerate(uniq)}
+        import torch as _t
+
+        lut = _t.tensor([idx[v] for v in vals], dtype=_t.int32, device=c.device)
+        from .column import _pack_strings
+
+        offs, byts = _pack_strings(uniq, c.device)
+        codes = lut[c.codes.to(_t.int64).clamp_min(0)]
+        codes = _t.where(c.codes >= 0, codes, c.codes)
+        return StringColumn(offs, byts, c.validity, codes)
+    return _str_map(c, f)
+
+
+def _f_length_any(args, out, chunk, ev):
+    return _f_length(args, out, chunk, ev)
+
+
+_IMPLS["regexp_replace"] = _f_regexp_replace

@@ -225,6 +225,16 @@ class Resolver:
             if 0 <= idx < len(child.schema):
                 n, t = child.schema[idx]
                 return S.BoundRef(idx, n, t)
+        # ORDER BY COUNT(*) / aggregate expressions over a grouped query
+        if any(isinstance(x, S.AggFunc) for x in e.walk()) and isinstance(child, S.Project):
+            agg = child.input
+            if isinstance(agg, S.Filter):
+                agg = agg.input
+            if isinstance(agg, S.Aggregate):
+                bound = self._bind_sort_agg(e, agg)
+                idx = len(child.schema) + len(extra)
+                extra.append(bound)
+                return S.BoundRef(idx, f"__sort{idx}", bound.dtype)
         try:
             return self._expr(e, scope)
         except ResolutionError:
@@ -235,6 +245,38 @@ class Resolver:
                 extra.append(bound)
                 return S.BoundRef(idx, f"__sort{idx}", bound.dtype)
             raise
+
+    def _bind_sort_agg(self, e: S.Expr, agg: S.Aggregate) -> S.Expr:
+        """Bind a sort expression containing aggregates against an Aggregate
+        node, appending missing aggregates to it."""
+        inner_scope = self._child_scope(agg.input, None)
+        ng = len(agg.group_by)
+
+        def bind(x: S.Expr) -> S.Expr:
+            if isinstance(x, S.AggFunc):
+                bargs = [self._expr(a, inner_scope) for a in x.args
+                         if not isinstance(a, S.Star)]
+                name = _normalize_agg_name(x.name)
+                rtype = agg_return_type(name, [a.dtype for a in bargs], x.distinct)
+                bound = S.AggFunc(name, bargs, x.distinct, rtype, None)
+                for ai, existing in enumerate(agg.aggs):
+                    if repr(existing) == repr(bound):
+                        return S.BoundRef(ng + ai, f"__agg{ai}", existing.dtype)
+                agg.aggs.append(bound)
+                agg.schema = list(agg.schema) + [(f"__agg{len(agg.aggs)-1}", rtype)]
+                return S.BoundRef(ng + len(agg.aggs) - 1, f"__agg{len(agg.aggs)-1}", rtype)
+            if isinstance(x, S.Col):
+                # group key by name
+                for gi in range(ng):
+                    if agg.schema[gi][0].lower() == x.name.lower():
+                        return S.BoundRef(gi, agg.schema[gi][0], agg.schema[gi][1])
+                raise ResolutionError(f"cannot resolve {x.name} in sort over aggregate")
+            ch = x.children()
+            if not ch:
+                return x
+            return self._type_expr(x.with_children([bind(c) for c in ch]))
+
+        return self._type_expr(bind(e))
 
     def _p_Aggregate(self, p: S.Aggregate, outer):
         child = self._plan(p.input, outer)

@@ -43,6 +43,8 @@ __device__ inline long long load_as_i64(const void* p, int elem, int64_t i) {
     case 0: return ((const long long*)p)[i];
     case 2: return (long long)((const int*)p)[i];
     case 4: return (long long)((const uint8_t*)p)[i];
+    case 5: return (long long)((const short*)p)[i];
+    case 6: return (long long)((const int8_t*)p)[i];
     default: return 0;
   }
 }
@@ -53,6 +55,8 @@ __device__ inline double load_as_f64(const void* p, int elem, int64_t i) {
     case 3: return (double)((const float*)p)[i];
     case 0: return (double)((const long long*)p)[i];
     case 2: return (double)((const int*)p)[i];
+    case 5: return (double)((const short*)p)[i];
+    case 6: return (double)((const int8_t*)p)[i];
     default: return 0.0;
   }
 }
@@ -230,6 +234,8 @@ ColArg make_col(const c10::optional<torch::Tensor>& t, int op) {
       case torch::kFloat32: a.elem = 3; break;
       case torch::kUInt8: a.elem = 4; break;
       case torch::kBool: a.elem = 4; break;
+      case torch::kInt16: a.elem = 5; break;
+      case torch::kInt8: a.elem = 6; break;
       default: TORCH_CHECK(false, "unsupported value dtype for grouped_acc");
     }
   }

@@ -148,17 +148,29 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
     c = args[0]
     mask = _masked(c, filter_mask)
     if distinct and name in ("count", "sum", "avg"):
-        # reduce (gid, value) pairs to unique before aggregating
+        # unique (gid, value) pairs via two stable sorts — torch.unique(dim=0)
+        # has an allocation-state-dependent crash on ROCm at ~1e8 rows
         vk = normalize_key(c)
-        pair = torch.stack([gid[mask], vk[mask]], dim=1)
-        upair = torch.unique(pair, dim=0)
-        ugid = upair[:, 0]
+        gm, vm = gid[mask], vk[mask]
+        if gm.numel():
+            ord1 = torch.argsort(vm)
+            g1 = gm.index_select(0, ord1)
+            ord2 = torch.argsort(g1, stable=True)
+            g_sorted = g1.index_select(0, ord2)
+            v_sorted = vm.index_select(0, ord1).index_select(0, ord2)
+            first = torch.ones(g_sorted.shape[0], dtype=torch.bool, device=dev)
+            first[1:] = (g_sorted[1:] != g_sorted[:-1]) | (v_sorted[1:] != v_sorted[:-1])
+            ugid = g_sorted[first]
+            uvals = v_sorted[first]
+        else:
+            ugid = torch.zeros(0, dtype=torch.int64, device=dev)
+            uvals = torch.zeros(0, dtype=torch.int64, device=dev)
         if name == "count":
             data = torch.zeros(ng, dtype=torch.int64, device=dev)
             data.index_add_(0, ugid, torch.ones(ugid.shape[0], dtype=torch.int64, device=dev))
             return Column(T.I64, data, None)
-        # sum/avg distinct: gather original values — only valid for int-like
-        vals = upair[:, 1]
+        # sum/avg distinct: normalized keys ARE the raw values for int-like
+        vals = uvals
         data = torch.zeros(ng, dtype=torch.int64, device=dev)
         data.index_add_(0, ugid, vals)
         if name == "avg":

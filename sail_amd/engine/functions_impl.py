@@ -427,11 +427,21 @@ def _f_length(args, out, chunk, ev):
     c = _col(args[0], chunk)
     if not isinstance(c, StringColumn):
         raise NotImplementedError("length on non-string")
+    lens = _utf8_char_lens(c.offsets, c.bytes_)
     if c.is_dict:
-        lens = torch.tensor([len(v) for v in c.dict_values()], dtype=torch.int32, device=c.device)
         return Column(T.I32, lens[c.codes.long().clamp_min(0)], c.validity)
-    lens = (c.offsets[1:] - c.offsets[:-1]).to(torch.int32)
     return Column(T.I32, lens, c.validity)
+
+
+def _utf8_char_lens(offsets: torch.Tensor, bytes_: torch.Tensor) -> torch.Tensor:
+    """Character count per string: continuation bytes (0b10xxxxxx) don't
+    start a character; prefix-sum the starts and difference at offsets."""
+    if bytes_.numel() == 0:
+        return torch.zeros(offsets.numel() - 1, dtype=torch.int32, device=offsets.device)
+    starts = ((bytes_ & 0xC0) != 0x80).to(torch.int32)
+    cum = torch.zeros(bytes_.numel() + 1, dtype=torch.int64, device=offsets.device)
+    torch.cumsum(starts.to(torch.int64), 0, out=cum[1:])
+    return (cum.index_select(0, offsets[1:]) - cum.index_select(0, offsets[:-1])).to(torch.int32)
 
 
 def _f_substring(args, out, chunk, ev):
@@ -501,7 +511,15 @@ def _str_pred(fn):
             raise NotImplementedError("string predicate with column pattern")
         p = pat.value
         if isinstance(c, StringColumn) and c.is_dict:
-            hit = torch.tensor([fn(v, p) for v in c.dict_values()], dtype=torch.bool, device=c.device)
+            if c.is_cuda and "%" not in p and "_" not in p:
+                from ..ops import kernels as K
+
+                like_pat = {"startswith": p + "%", "endswith": "%" + p,
+                            "contains": "%" + p + "%"}[fn.__name__]
+                hit = K.require().like_mask(c.offsets, c.bytes_, like_pat.encode())
+            else:
+                hit = torch.tensor([fn(v, p) for v in c.dict_values()],
+                                   dtype=torch.bool, device=c.device)
             return Column(T.BOOL, hit[c.codes.long().clamp_min(0)] & (c.codes >= 0), c.validity)
         if c.is_cuda:
             from ..ops import kernels as K

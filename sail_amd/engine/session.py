@@ -232,6 +232,10 @@ class SessionContext:
         ctx = ExecutionContext(self, self.device)
         return Executor(ctx).execute(plan)
 
+    @property
+    def read(self) -> "DataFrameReader":
+        return DataFrameReader(self)
+
     # -- convenience -------------------------------------------------------
     def create_dataframe(self, data: Dict[str, list],
                          schema: Optional[Dict[str, T.DataType]] = None,
@@ -350,6 +354,10 @@ class DataFrame:
     def explain(self) -> str:
         return S.plan_tree_string(self.plan)
 
+    @property
+    def write(self) -> "DataFrameWriter":
+        return DataFrameWriter(self)
+
 
 class _MaterializedDataFrame(DataFrame):
     def __init__(self, session, plan, chunk):
@@ -358,3 +366,95 @@ class _MaterializedDataFrame(DataFrame):
 
     def collect_chunk(self) -> Chunk:
         return self._chunk
+
+
+class DataFrameReader:
+    """spark.read.format(...).option(...).load(path) facade
+    (ref: crates/sail-plan/src/resolver/query/read.rs data-source reads)."""
+
+    def __init__(self, session: SessionContext):
+        self._session = session
+        self._format = "parquet"
+        self._options: Dict[str, str] = {}
+        self._schema = None
+
+    def format(self, fmt: str) -> "DataFrameReader":
+        self._format = fmt
+        return self
+
+    def option(self, k: str, v) -> "DataFrameReader":
+        self._options[k] = str(v)
+        return self
+
+    def options(self, **kw) -> "DataFrameReader":
+        for k, v in kw.items():
+            self._options[k] = str(v)
+        return self
+
+    def schema(self, schema) -> "DataFrameReader":
+        self._schema = schema
+        return self
+
+    def load(self, path) -> "DataFrame":
+        paths = path if isinstance(path, list) else [path]
+        plan = S.DataSourceRead(format=self._format, paths=paths,
+                                options=dict(self._options), user_schema=self._schema)
+        resolved = _CatalogAdapter(self._session).resolve(plan)
+        return DataFrame(self._session, resolved)
+
+    def parquet(self, *paths) -> "DataFrame":
+        return self.format("parquet").load(list(paths))
+
+    def csv(self, *paths, **kw) -> "DataFrame":
+        return self.format("csv").options(**kw).load(list(paths))
+
+    def json(self, *paths) -> "DataFrame":
+        return self.format("json").load(list(paths))
+
+    def table(self, name: str) -> "DataFrame":
+        return self._session.table(name)
+
+
+class DataFrameWriter:
+    """df.write.format(...).mode(...).save(path)."""
+
+    def __init__(self, df: "DataFrame"):
+        self._df = df
+        self._format = "parquet"
+        self._mode = "error"
+        self._options: Dict[str, str] = {}
+        self._partition_by: List[str] = []
+
+    def format(self, fmt: str) -> "DataFrameWriter":
+        self._format = fmt
+        return self
+
+    def mode(self, m: str) -> "DataFrameWriter":
+        self._mode = {"errorifexists": "error", "default": "error"}.get(m.lower(), m.lower())
+        return self
+
+    def option(self, k: str, v) -> "DataFrameWriter":
+        self._options[k] = str(v)
+        return self
+
+    def partitionBy(self, *cols) -> "DataFrameWriter":
+        self._partition_by = list(cols)
+        return self
+
+    def save(self, path: str):
+        from ..datasource.registry import write_source
+
+        chunk = self._df.collect_chunk()
+        write_source(self._format, path, chunk, self._mode, self._options,
+                     self._partition_by)
+
+    def parquet(self, path: str):
+        self.format("parquet").save(path)
+
+    def csv(self, path: str):
+        self.format("csv").save(path)
+
+    def saveAsTable(self, name: str):
+        chunk = self._df.collect_chunk()
+        self._df.session.catalog.register_table_chunk(
+            name, chunk, self._df.plan.schema)

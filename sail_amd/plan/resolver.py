@@ -82,6 +82,11 @@ class Resolver:
                 aliased = S.SubqueryAlias(input=resolved, alias=p.table)
                 aliased.schema = resolved.schema
                 return self._qualify(aliased, p.table)
+        # `SELECT ... FROM parquet.`/path``-style direct file reads (Spark)
+        head, _, rest = p.table.partition(".")
+        if head.lower() in ("parquet", "csv", "json") and rest:
+            ds = S.DataSourceRead(format=head.lower(), paths=[rest], options=p.options)
+            return self._qualify(self._p_DataSourceRead(ds, outer), head.lower())
         schema = self.catalog.table_schema(p.table)
         if schema is None:
             raise ResolutionError(f"table not found: {p.table}")

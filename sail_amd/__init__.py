@@ -8,4 +8,13 @@ from .engine.session import Catalog, DataFrame, SessionContext
 
 __version__ = "0.1.0"
 
-__all__ = ["SessionContext", "DataFrame", "Catalog", "__version__"]
+
+def connect_server(host="127.0.0.1", port=0, device=None):
+    """Start a Spark Connect server; returns the server (``.address`` holds
+    the bound endpoint)."""
+    from .connect.server import SparkConnectServer
+
+    return SparkConnectServer(host=host, port=port, device=device).start()
+
+
+__all__ = ["SessionContext", "DataFrame", "Catalog", "connect_server", "__version__"]

@@ -20,6 +20,10 @@ def infer_source_schema(fmt: str, paths: List[str], options: Dict[str, str]):
         from . import json_io
 
         return json_io.infer_schema(paths, options)
+    if fmt == "delta":
+        from . import delta
+
+        return delta.infer_schema(paths, options)
     raise ValueError(f"unsupported format {fmt}")
 
 
@@ -34,6 +38,10 @@ def read_source(fmt: str, paths: List[str], options: Dict[str, str], schema, dev
         from . import json_io
 
         return json_io.read(paths, schema, device, options)
+    if fmt == "delta":
+        from . import delta
+
+        return delta.read(paths, schema, device, options)
     raise ValueError(f"unsupported format {fmt}")
 
 

@@ -545,6 +545,21 @@ class Executor:
         write_source(p.format, p.path, data, p.mode, p.options, p.partition_by)
         return Chunk([], [])
 
+    def _x_MergeInto(self, p) -> Chunk:
+        from .dml import execute_merge
+
+        return execute_merge(self, p)
+
+    def _x_UpdateTable(self, p) -> Chunk:
+        from .dml import execute_update
+
+        return execute_update(self, p)
+
+    def _x_DeleteFrom(self, p) -> Chunk:
+        from .dml import execute_delete
+
+        return execute_delete(self, p)
+
     def _x_Explain(self, p: S.Explain) -> Chunk:
         text = S.plan_tree_string(p.input)
         return Chunk([StringColumn.from_pylist([text])], ["plan"])

@@ -84,7 +84,7 @@ class Resolver:
                 return self._qualify(aliased, p.table)
         # `SELECT ... FROM parquet.`/path``-style direct file reads (Spark)
         head, _, rest = p.table.partition(".")
-        if head.lower() in ("parquet", "csv", "json") and rest:
+        if head.lower() in ("parquet", "csv", "json", "delta") and rest:
             ds = S.DataSourceRead(format=head.lower(), paths=[rest], options=p.options)
             return self._qualify(self._p_DataSourceRead(ds, outer), head.lower())
         schema = self.catalog.table_schema(p.table)
@@ -478,6 +478,93 @@ class Resolver:
         out = S.Write(input=inp, format=p.format, path=p.path, table=p.table,
                       mode=p.mode, partition_by=p.partition_by, options=p.options)
         out.schema = []
+        return out
+
+    def _target_schema(self, name: str):
+        head, _, rest = name.partition(".")
+        if head.lower() == "delta" and rest:
+            from ..datasource.delta import infer_schema
+
+            return infer_schema([rest]), ("delta", rest)
+        schema = self.catalog.table_schema(name)
+        if schema is None:
+            raise ResolutionError(f"table not found: {name}")
+        return schema, ("catalog", name)
+
+    def _p_MergeInto(self, p: S.MergeInto, outer):
+        tschema, tref = self._target_schema(p.target)
+        source = self._plan(p.source, outer)
+        tq = p.target_alias or p.target.split(".")[-1]
+        sq = p.source_alias or _plan_qualifier(source)
+        fields = [Field(n, t, tq) for n, t in tschema] +                  [Field(n, t, sq) for n, t in source.schema]
+        scope = Scope(fields, outer)
+        on = _coerce_to_bool(self._expr(p.on, scope))
+
+        def bind_action(a: S.MergeAction, star_schema=None) -> S.MergeAction:
+            cond = _coerce_to_bool(self._expr(a.condition, scope)) if a.condition is not None else None
+            if a.kind == "update_star":
+                assigns = []
+                snames = {n.lower(): i for i, (n, _) in enumerate(source.schema)}
+                for ti, (n, t) in enumerate(tschema):
+                    if n.lower() in snames:
+                        si = snames[n.lower()]
+                        assigns.append((n, S.BoundRef(len(tschema) + si, n, source.schema[si][1])))
+                return S.MergeAction("update", cond, assigns)
+            if a.kind == "insert_star":
+                cols, vals = [], []
+                snames = {n.lower(): i for i, (n, _) in enumerate(source.schema)}
+                for n, t in tschema:
+                    if n.lower() in snames:
+                        si = snames[n.lower()]
+                        cols.append(n)
+                        vals.append(S.BoundRef(len(tschema) + si, n, source.schema[si][1]))
+                return S.MergeAction("insert", cond, insert_columns=cols, insert_values=vals)
+            if a.kind == "update":
+                assigns = []
+                for name, e in a.assignments:
+                    col = name.split(".")[-1]
+                    if not any(col.lower() == n.lower() for n, _ in tschema):
+                        raise ResolutionError(f"MERGE SET column {col} not in target")
+                    assigns.append((col, self._expr(e, scope)))
+                return S.MergeAction("update", cond, assigns)
+            if a.kind == "insert":
+                vals = [self._expr(e, scope) for e in a.insert_values]
+                return S.MergeAction("insert", cond, insert_columns=a.insert_columns,
+                                     insert_values=vals)
+            return S.MergeAction(a.kind, cond)
+
+        out = S.MergeInto(target=p.target, target_alias=p.target_alias, source=source,
+                          source_alias=p.source_alias, on=on,
+                          matched=[bind_action(a) for a in p.matched],
+                          not_matched=[bind_action(a) for a in p.not_matched],
+                          not_matched_by_source=[bind_action(a) for a in p.not_matched_by_source])
+        out.schema = []
+        out.__dict__["_target_schema"] = tschema
+        out.__dict__["_target_ref"] = tref
+        return out
+
+    def _p_UpdateTable(self, p: S.UpdateTable, outer):
+        tschema, tref = self._target_schema(p.table)
+        scope = Scope([Field(n, t, p.table.split(".")[-1]) for n, t in tschema], outer)
+        assigns = []
+        for name, e in p.assignments:
+            col = name.split(".")[-1]
+            assigns.append((col, self._expr(e, scope)))
+        cond = _coerce_to_bool(self._expr(p.condition, scope)) if p.condition is not None else None
+        out = S.UpdateTable(table=p.table, assignments=assigns, condition=cond)
+        out.schema = []
+        out.__dict__["_target_schema"] = tschema
+        out.__dict__["_target_ref"] = tref
+        return out
+
+    def _p_DeleteFrom(self, p: S.DeleteFrom, outer):
+        tschema, tref = self._target_schema(p.table)
+        scope = Scope([Field(n, t, p.table.split(".")[-1]) for n, t in tschema], outer)
+        cond = _coerce_to_bool(self._expr(p.condition, scope)) if p.condition is not None else None
+        out = S.DeleteFrom(table=p.table, condition=cond)
+        out.schema = []
+        out.__dict__["_target_schema"] = tschema
+        out.__dict__["_target_ref"] = tref
         return out
 
     def _p_Explain(self, p: S.Explain, outer):

@@ -237,14 +237,16 @@ def _reorder_tree(root: S.Join, stats) -> S.Plan:
             denom = _conjunct_ndv(c, leaves, offs, leaf_of, ests, stats)
             edge_ndv.append((a, b, denom))
 
-    def join_denom(j: int, placed: Set[int]) -> float:
+    def join_denom(j: int, placed: Set[int], cur_est: float) -> float:
         """Combined ndv for all equality conjuncts connecting leaf j to the
-        placed set (composite keys multiply, capped)."""
+        placed set. Per-key ndvs multiply but the composite-key ndv can never
+        exceed either side's cardinality (a table's PK tuple has ndv = its
+        row count, e.g. partsupp's (partkey, suppkey))."""
         d = 1.0
         for a, b, nd in edge_ndv:
             if (a == j and b in placed) or (b == j and a in placed):
                 d = min(d * nd, 1e15)
-        return d
+        return min(d, max(ests[j][1], cur_est, 1.0))
 
     # greedy from EVERY seed, objective = sum of intermediate sizes; a single
     # smallest-leaf seed can dead-end into a low-ndv edge (q5: region ->
@@ -253,21 +255,31 @@ def _reorder_tree(root: S.Join, stats) -> S.Plan:
         order = [seed]
         placed = {seed}
         cur_est = ests[seed][0]
+        # selectivity of the placed side: FK joins against an unfiltered fact
+        # keep every fact row scaled by this (containment lower bound) —
+        # without it, partsupp⋈lineitem estimates 1e5 instead of 6e8 (q9)
+        cur_sel = min(1.0, ests[seed][0] / max(ests[seed][1], 1.0))
         total = cur_est
         while len(placed) < n:
             best, best_cost = None, None
             for i in range(n):
                 if i in placed:
                     continue
-                est_i, _ = ests[i]
+                est_i, base_i = ests[i]
                 if any(j in placed for j in edges[i]):
-                    cost = cur_est * est_i / max(join_denom(i, placed), 1.0)
+                    formula = cur_est * est_i / max(join_denom(i, placed, cur_est), 1.0)
+                    cost = max(formula, est_i * cur_sel)
                 else:
-                    cost = cur_est * est_i  # cross product
+                    # disconnected cross product: greedy is myopic, so a
+                    # cheap-looking cross (1-row dims aside) poisons later
+                    # steps — penalize unless genuinely tiny
+                    cost = cur_est * est_i * (1.0 if est_i <= 2 else 8.0)
                 if best_cost is None or cost < best_cost:
                     best, best_cost = i, cost
             order.append(best)
             placed.add(best)
+            est_b, base_b = ests[best]
+            cur_sel = min(cur_sel, min(1.0, est_b / max(base_b, 1.0)))
             cur_est = max(best_cost, 1.0)
             total += cur_est
         return order, total

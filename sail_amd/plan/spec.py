@@ -606,3 +606,46 @@ class OuterRef(Expr):
 
     def __repr__(self):
         return f"outer#{self.index}:{self.name}"
+
+
+@dataclass
+class MergeAction:
+    """One WHEN clause of MERGE (ref: spec::CommandNode::MergeInto,
+    crates/sail-common/src/spec/plan.rs MergeInto)."""
+
+    kind: str  # "update" | "delete" | "insert" | "insert_star" | "update_star"
+    condition: Optional[Expr] = None
+    assignments: List[Tuple[str, Expr]] = field(default_factory=list)
+    insert_columns: Optional[List[str]] = None
+    insert_values: Optional[List[Expr]] = None
+
+
+@dataclass
+class MergeInto(Command):
+    target: str = ""
+    target_alias: Optional[str] = None
+    source: Plan = None
+    source_alias: Optional[str] = None
+    on: Expr = None
+    matched: List[MergeAction] = field(default_factory=list)
+    not_matched: List[MergeAction] = field(default_factory=list)
+    not_matched_by_source: List[MergeAction] = field(default_factory=list)
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+    def children(self):
+        return [self.source] if self.source is not None else []
+
+
+@dataclass
+class UpdateTable(Command):
+    table: str = ""
+    assignments: List[Tuple[str, Expr]] = field(default_factory=list)
+    condition: Optional[Expr] = None
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+
+@dataclass
+class DeleteFrom(Command):
+    table: str = ""
+    condition: Optional[Expr] = None
+    schema: Optional[List[Tuple[str, T.DataType]]] = None

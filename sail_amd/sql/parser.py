@@ -111,6 +111,12 @@ class Parser:
             return self._parse_drop()
         if self.at_kw("INSERT"):
             return self._parse_insert()
+        if self.at_kw("MERGE"):
+            return self._parse_merge()
+        if self.at_kw("UPDATE"):
+            return self._parse_update()
+        if self.at_kw("DELETE"):
+            return self._parse_delete()
         if self.at_kw("SHOW"):
             self.next()
             self.expect_kw("TABLES")
@@ -224,6 +230,112 @@ class Parser:
         self.eat_kw("TABLE")
         table = self._qualified_name()
         return S.InsertInto(table=table, input=self.parse_query(), overwrite=overwrite)
+
+    def _parse_merge(self) -> S.Plan:
+        self.expect_kw("MERGE")
+        self.expect_kw("INTO")
+        target = self._qualified_name()
+        talias = None
+        if self.eat_kw("AS"):
+            talias = self.ident()
+        elif self.peek().kind == "ident" and self.peek().upper not in ("USING",):
+            talias = self.ident()
+        self.expect_kw("USING")
+        if self.at_op("("):
+            self.expect_op("(")
+            source = self.parse_query()
+            self.expect_op(")")
+        else:
+            source = S.Read(table=self._qualified_name())
+        salias = None
+        if self.eat_kw("AS"):
+            salias = self.ident()
+        elif self.peek().kind == "ident" and self.peek().upper not in ("ON",):
+            salias = self.ident()
+        self.expect_kw("ON")
+        on = self.parse_expr()
+        matched, not_matched, nm_by_source = [], [], []
+        while self.at_kw("WHEN"):
+            self.next()
+            if self.eat_kw("MATCHED"):
+                cond = self.parse_expr() if self.eat_kw("AND") else None
+                self.expect_kw("THEN")
+                if self.eat_kw("DELETE"):
+                    matched.append(S.MergeAction("delete", cond))
+                else:
+                    self.expect_kw("UPDATE")
+                    self.expect_kw("SET")
+                    if self.at_op("*"):
+                        self.next()
+                        matched.append(S.MergeAction("update_star", cond))
+                    else:
+                        matched.append(S.MergeAction("update", cond,
+                                                     self._parse_assignments()))
+            else:
+                self.expect_kw("NOT")
+                self.expect_kw("MATCHED")
+                by_source = False
+                if self.eat_kw("BY"):
+                    if self.eat_kw("SOURCE"):
+                        by_source = True
+                    else:
+                        self.expect_kw("TARGET")
+                cond = self.parse_expr() if self.eat_kw("AND") else None
+                self.expect_kw("THEN")
+                if by_source:
+                    if self.eat_kw("DELETE"):
+                        nm_by_source.append(S.MergeAction("delete", cond))
+                    else:
+                        self.expect_kw("UPDATE")
+                        self.expect_kw("SET")
+                        nm_by_source.append(S.MergeAction("update", cond,
+                                                          self._parse_assignments()))
+                else:
+                    self.expect_kw("INSERT")
+                    if self.at_op("*"):
+                        self.next()
+                        not_matched.append(S.MergeAction("insert_star", cond))
+                    else:
+                        self.expect_op("(")
+                        cols = [self.ident()]
+                        while self.eat_op(","):
+                            cols.append(self.ident())
+                        self.expect_op(")")
+                        self.expect_kw("VALUES")
+                        self.expect_op("(")
+                        vals = self._expr_list()
+                        self.expect_op(")")
+                        not_matched.append(S.MergeAction("insert", cond,
+                                                         insert_columns=cols,
+                                                         insert_values=vals))
+        return S.MergeInto(target=target, target_alias=talias, source=source,
+                           source_alias=salias, on=on, matched=matched,
+                           not_matched=not_matched, not_matched_by_source=nm_by_source)
+
+    def _parse_assignments(self):
+        out = []
+        while True:
+            name = self._qualified_name()
+            self.expect_op("=")
+            out.append((name, self.parse_expr()))
+            if not self.eat_op(","):
+                break
+        return out
+
+    def _parse_update(self) -> S.Plan:
+        self.expect_kw("UPDATE")
+        table = self._qualified_name()
+        self.expect_kw("SET")
+        assignments = self._parse_assignments()
+        cond = self.parse_expr() if self.eat_kw("WHERE") else None
+        return S.UpdateTable(table=table, assignments=assignments, condition=cond)
+
+    def _parse_delete(self) -> S.Plan:
+        self.expect_kw("DELETE")
+        self.expect_kw("FROM")
+        table = self._qualified_name()
+        cond = self.parse_expr() if self.eat_kw("WHERE") else None
+        return S.DeleteFrom(table=table, condition=cond)
 
     def _qualified_name(self) -> str:
         parts = [self.ident()]

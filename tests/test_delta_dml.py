@@ -1,0 +1,107 @@
+"""Delta-lite format + MERGE INTO / UPDATE / DELETE
+(ref: crates/sail-delta-lake behavior; config #5 of BASELINE.json)."""
+import json
+import os
+
+import pytest
+
+import sail_amd
+from sail_amd.engine import types as T
+
+
+@pytest.fixture()
+def s():
+    return sail_amd.SessionContext(device="cpu")
+
+
+def _mk_delta(s, path):
+    s.create_dataframe({"id": [1, 2, 3], "v": [10.0, 20.0, 30.0]},
+                       schema={"id": T.I64, "v": T.F64}, name="seed")
+    s.table("seed").write.format("delta").mode("overwrite").save(path)
+
+
+def test_delta_write_read_roundtrip(s, tmp_path):
+    p = str(tmp_path / "dt")
+    _mk_delta(s, p)
+    assert os.path.exists(os.path.join(p, "_delta_log", f"{0:020d}.json"))
+    df = s.read.format("delta").load(p)
+    assert sorted(df.collect()) == [(1, 10.0), (2, 20.0), (3, 30.0)]
+
+
+def test_delta_append_and_time_travel(s, tmp_path):
+    p = str(tmp_path / "dt")
+    _mk_delta(s, p)
+    s.create_dataframe({"id": [4], "v": [40.0]}, schema={"id": T.I64, "v": T.F64}, name="more")
+    s.table("more").write.format("delta").mode("append").save(p)
+    assert len(s.read.format("delta").load(p).collect()) == 4
+    v0 = s.read.format("delta").option("versionAsOf", 0).load(p)
+    assert len(v0.collect()) == 3
+
+
+def test_delta_sql_read(s, tmp_path):
+    p = str(tmp_path / "dt")
+    _mk_delta(s, p)
+    rows = s.sql(f"SELECT sum(v) FROM delta.`{p}`").collect()
+    assert rows == [(60.0,)]
+
+
+def test_merge_into_catalog_table(s):
+    s.create_dataframe({"id": [1, 2, 3], "v": [10, 20, 30]},
+                       schema={"id": T.I64, "v": T.I64}, name="t")
+    s.create_dataframe({"id": [2, 3, 4], "v": [200, 300, 400]},
+                       schema={"id": T.I64, "v": T.I64}, name="src")
+    res = s.sql("""
+        MERGE INTO t USING src ON t.id = src.id
+        WHEN MATCHED AND src.v >= 300 THEN DELETE
+        WHEN MATCHED THEN UPDATE SET v = src.v
+        WHEN NOT MATCHED THEN INSERT (id, v) VALUES (src.id, src.v)
+    """).collect()
+    rows = sorted(s.sql("SELECT * FROM t").collect())
+    assert rows == [(1, 10), (2, 200), (4, 400)]
+
+
+def test_merge_into_delta(s, tmp_path):
+    p = str(tmp_path / "dt")
+    _mk_delta(s, p)
+    s.create_dataframe({"id": [3, 9], "v": [333.0, 999.0]},
+                       schema={"id": T.I64, "v": T.F64}, name="src")
+    s.sql(f"""
+        MERGE INTO delta.`{p}` AS t USING src AS u ON t.id = u.id
+        WHEN MATCHED THEN UPDATE SET v = u.v
+        WHEN NOT MATCHED THEN INSERT (id, v) VALUES (u.id, u.v)
+    """)
+    rows = sorted(s.read.format("delta").load(p).collect())
+    assert rows == [(1, 10.0), (2, 20.0), (3, 333.0), (9, 999.0)]
+    # merge committed a new version
+    assert len(s.read.format("delta").option("versionAsOf", 0).load(p).collect()) == 3
+
+
+def test_merge_update_star(s):
+    s.create_dataframe({"id": [1, 2], "v": [1, 2]}, schema={"id": T.I64, "v": T.I64}, name="t")
+    s.create_dataframe({"id": [2], "v": [22]}, schema={"id": T.I64, "v": T.I64}, name="src")
+    s.sql("MERGE INTO t USING src ON t.id = src.id WHEN MATCHED THEN UPDATE SET *")
+    assert sorted(s.sql("SELECT * FROM t").collect()) == [(1, 1), (2, 22)]
+
+
+def test_merge_cardinality_violation(s):
+    s.create_dataframe({"id": [1]}, schema={"id": T.I64}, name="t")
+    s.create_dataframe({"id": [1, 1]}, schema={"id": T.I64}, name="src")
+    with pytest.raises(Exception, match="cardinality"):
+        s.sql("MERGE INTO t USING src ON t.id = src.id WHEN MATCHED THEN DELETE")
+
+
+def test_update_delete_statements(s):
+    s.create_dataframe({"id": [1, 2, 3], "v": [10, 20, 30]},
+                       schema={"id": T.I64, "v": T.I64}, name="t")
+    res = s.sql("UPDATE t SET v = v + 1 WHERE id >= 2").collect()
+    assert res == [(2,)]
+    assert sorted(s.sql("SELECT * FROM t").collect()) == [(1, 10), (2, 21), (3, 31)]
+    s.sql("DELETE FROM t WHERE v > 25")
+    assert sorted(s.sql("SELECT * FROM t").collect()) == [(1, 10), (2, 21)]
+
+
+def test_delete_from_delta(s, tmp_path):
+    p = str(tmp_path / "dt")
+    _mk_delta(s, p)
+    s.sql(f"DELETE FROM delta.`{p}` WHERE v >= 20.0")
+    assert s.read.format("delta").load(p).collect() == [(1, 10.0)]

@@ -1,0 +1,43 @@
+"""Application configuration.
+
+Single source of defaults + SAIL_-prefixed environment overrides — the
+analogue of the reference's figment-loaded application.yaml
+(ref: crates/sail-common/src/config/application.yaml, loader.rs).
+"""
+from __future__ import annotations
+
+import os
+from typing import Dict
+
+DEFAULTS: Dict[str, str] = {
+    # execution
+    "sail.mode": "local",                       # local | spmd
+    "sail.execution.device": "auto",            # auto | cpu | cuda[:N]
+    "sail.execution.result_batch_rows": "65536",
+    # optimizer (ref: application.yaml optimizer.*)
+    "sail.optimizer.enable_join_reorder": "true",
+    "sail.optimizer.join_reorder_max_relations": "12",
+    "sail.optimizer.agg_mask_min_selectivity": "0.2",
+    # kernels
+    "sail.kernels.require_on_gpu": "true",
+    "sail.kernels.grouped_agg_max_lds_groups": "4096",
+    # tracing (ref: sail-telemetry)
+    "sail.trace": "false",
+    # spark-compatible session confs
+    "spark.sql.session.timeZone": "UTC",
+    "spark.sql.ansi.enabled": "false",
+    "spark.sql.caseSensitive": "false",
+    "spark.sql.shuffle.partitions": "1",
+}
+
+
+def load_config() -> Dict[str, str]:
+    """Defaults overridden by SAIL_FOO_BAR env vars (SAIL_ prefix, _ -> .)."""
+    conf = dict(DEFAULTS)
+    for k, v in os.environ.items():
+        if k.startswith("SAIL_") and k not in ("SAIL_TRACE",):
+            key = "sail." + k[len("SAIL_"):].lower().replace("_", ".")
+            conf[key] = v
+    if os.environ.get("SAIL_TRACE") == "1":
+        conf["sail.trace"] = "true"
+    return conf

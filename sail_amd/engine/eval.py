@@ -182,6 +182,21 @@ class Evaluator:
         lc, rc = broadcast(l, n, dev), broadcast(r, n, dev)
         return _arith(e.op, lc, rc, e.dtype)
 
+    def _eval_udf(self, e: S.Func, chunk: Chunk) -> Val:
+        """Host Python UDF: device columns -> host lists -> fn -> column."""
+        sess = self.ctx.session if self.ctx is not None else None
+        info = sess.udfs.get(e.name.lower()) if sess is not None else None
+        if info is None:
+            raise EvalError(f"UDF {e.name} not registered")
+        fn, rtype = info
+        cols = [self.eval_col(a, chunk).to_pylist() for a in e.args]
+        out = [fn(*vals) for vals in zip(*cols)] if cols else [fn() for _ in range(chunk.num_rows)]
+        from .column import Column as _C, StringColumn as _S
+
+        if isinstance(rtype, T.StringType):
+            return _S.from_pylist(out, device=chunk.device)
+        return _C.from_values(out, rtype, device=chunk.device)
+
     def _kleene(self, e: S.BinaryOp, chunk: Chunk) -> Val:
         l = self.eval(e.left, chunk)
         r = self.eval(e.right, chunk)
@@ -345,6 +360,9 @@ class Evaluator:
     def _e_Func(self, e: S.Func, chunk: Chunk) -> Val:
         from .functions_impl import dispatch_function
 
+        sess = getattr(self.ctx, "session", None) if self.ctx is not None else None
+        if sess is not None and getattr(sess, "udfs", None) and e.name.lower() in sess.udfs:
+            return self._eval_udf(e, chunk)
         args = [self.eval(a, chunk) for a in e.args]
         if args and all(isinstance(a, Scalar) for a in args) \
                 and e.name not in ("rand", "randn", "uuid", "monotonically_increasing_id"):

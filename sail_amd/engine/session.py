@@ -190,10 +190,12 @@ class SessionContext:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         self.device = torch.device(device)
         self.catalog = catalog or Catalog()
-        self.conf: Dict[str, str] = {
-            "spark.sql.session.timeZone": "UTC",
-            "spark.sql.ansi.enabled": "false",
-        }
+        from ..config import load_config
+
+        self.conf: Dict[str, str] = load_config()
+        self.udfs: Dict[str, tuple] = {}
+        self.query_log: List[dict] = []
+        self._register_system_tables()
 
     # -- pipeline ----------------------------------------------------------
     def parse(self, sql: str) -> S.Plan:
@@ -216,12 +218,62 @@ class SessionContext:
         return self.optimize(self.resolve(self.parse(sql)))
 
     def sql(self, sql: str) -> "DataFrame":
+        import time as _time
+
+        t0 = _time.time()
         plan = self.plan_sql(sql)
         if isinstance(plan, S.Command):
             # commands (DDL/config/writes) execute eagerly, like Spark sql()
             chunk = self.execute_plan(plan)
+            self._log_query(sql, t0, chunk.num_rows, "command")
             return _MaterializedDataFrame(self, plan, chunk)
-        return DataFrame(self, plan)
+        return DataFrame(self, plan, sql_text=sql, t_start=t0)
+
+    def _log_query(self, sql: str, t0: float, rows: int, kind: str):
+        import time as _time
+
+        self.query_log.append({
+            "query": sql.strip()[:500], "kind": kind,
+            "duration_ms": round((_time.time() - t0) * 1000, 3),
+            "rows": rows, "timestamp_ms": int(t0 * 1000),
+        })
+
+    def _register_system_tables(self):
+        """system.* virtual tables over engine telemetry — the reference's
+        'the Spark UI is SELECT * FROM system.jobs' design
+        (ref: crates/sail-catalog-system/src/)."""
+        from . import types as T
+
+        def queries_provider(device):
+            log = self.query_log
+            return Table.from_pydict(
+                {"query": [q["query"] for q in log],
+                 "kind": [q["kind"] for q in log],
+                 "duration_ms": [q["duration_ms"] for q in log],
+                 "rows": [q["rows"] for q in log],
+                 "timestamp_ms": [q["timestamp_ms"] for q in log]},
+                {"query": T.STRING, "kind": T.STRING, "duration_ms": T.F64,
+                 "rows": T.I64, "timestamp_ms": T.I64}, device="cpu")
+
+        self.catalog.register_provider(
+            "system_queries",
+            [("query", T.STRING), ("kind", T.STRING), ("duration_ms", T.F64),
+             ("rows", T.I64), ("timestamp_ms", T.I64)], queries_provider)
+
+        def tables_provider(device):
+            names = self.catalog.list_tables()
+            return Table.from_pydict(
+                {"tableName": names,
+                 "rows": [self.catalog.table_rows(n) or -1 for n in names]},
+                {"tableName": T.STRING, "rows": T.I64}, device="cpu")
+
+        self.catalog.register_provider(
+            "system_tables", [("tableName", T.STRING), ("rows", T.I64)],
+            tables_provider)
+
+    @property
+    def udf(self) -> "UdfRegistry":
+        return UdfRegistry(self)
 
     def table(self, name: str) -> "DataFrame":
         return DataFrame(self, self.optimize(self.resolve(S.Read(table=name))))
@@ -254,6 +306,23 @@ class SessionContext:
         return DataFrame(self, plan)
 
 
+class UdfRegistry:
+    """Host-side Python UDFs: the CPython boundary the reference crosses via
+    pyo3 (ref: crates/sail-python-udf/src/udf/pyspark_udf.rs) is a plain
+    in-process call here; device columns round-trip through host lists."""
+
+    def __init__(self, session: SessionContext):
+        self._session = session
+
+    def register(self, name: str, fn, return_type=None):
+        from . import types as T
+
+        if isinstance(return_type, str):
+            return_type = T.type_from_name(return_type)
+        self._session.udfs[name.lower()] = (fn, return_type or T.F64)
+        return fn
+
+
 class _CatalogAdapter:
     """Bridges the resolver's catalog protocol to Catalog + view expansion."""
 
@@ -274,6 +343,9 @@ class _CatalogAdapter:
 
     def view_plan(self, name: str):
         return self.session.catalog.view_plan(name)
+
+    def udf(self, name: str):
+        return self.session.udfs.get(name.lower())
 
 
 # patch Resolver to consult views: Read resolution checks views first
@@ -306,16 +378,25 @@ class DataFrame:
     """Minimal DataFrame facade over a resolved plan (result surface for the
     Connect server and Python API)."""
 
-    def __init__(self, session: SessionContext, plan: S.Plan):
+    def __init__(self, session: SessionContext, plan: S.Plan, sql_text=None, t_start=None):
         self.session = session
         self.plan = plan
+        self._sql_text = sql_text
+        self._t_start = t_start
 
     @property
     def schema(self):
         return self.plan.schema
 
     def collect_chunk(self) -> Chunk:
-        return self.session.execute_plan(self.plan)
+        out = self.session.execute_plan(self.plan)
+        if self._sql_text is not None:
+            import time as _time
+
+            self.session._log_query(self._sql_text, self._t_start or _time.time(),
+                                    out.num_rows, "query")
+            self._sql_text = None
+        return out
 
     def to_pydict(self) -> Dict[str, list]:
         c = self.collect_chunk()

@@ -722,6 +722,12 @@ class Resolver:
         if isinstance(e, S.Func):
             t = scalar_return_type(e.name, [a.dtype for a in e.args])
             if t is None:
+                udf = getattr(self.catalog, "udf", None)
+                info = udf(e.name) if udf else None
+                if info is not None:
+                    e.dtype = info[1]
+                    e.__dict__["_is_udf"] = True
+                    return e
                 raise ResolutionError(f"unknown function {e.name}")
             # structural return types
             if e.name == "coalesce" or e.name in ("nvl", "ifnull"):

@@ -26,7 +26,7 @@ def main():
     ap.add_argument("--steps", type=int, default=2)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--sf", type=float, default=float(os.environ.get("SAIL_BENCH_SF", "100")))
-    ap.add_argument("--workload", default="tpch", choices=["tpch", "clickbench"])
+    ap.add_argument("--workload", default="tpch", choices=["tpch", "clickbench", "delta"])
     ap.add_argument("--rows", type=int, default=100_000_000, help="clickbench rows")
     ap.add_argument("--device", default=None)
     ap.add_argument("--queries", default=None, help="comma-separated subset, e.g. 1,6,13")
@@ -62,6 +62,8 @@ def main():
         from sail_amd.datagen.clickbench import register_clickbench
         from sail_amd.datagen.clickbench_queries import QUERIES as CB
         QUERIES = {i: q for i, q in enumerate(CB)}
+    elif args.workload == "delta":
+        from sail_amd.datagen.delta_bench import DELTA_QUERIES as QUERIES
     else:
         from sail_amd.datagen.tpch_queries import QUERIES
 
@@ -77,6 +79,10 @@ def main():
     t0 = time.time()
     if args.workload == "clickbench":
         register_clickbench(session, rows=args.rows, device=device, rank=rank, world=world)
+    elif args.workload == "delta":
+        from sail_amd.datagen.delta_bench import setup_delta_bench
+
+        setup_delta_bench(session, sf=args.sf, device=device, rank=rank, world=world)
     else:
         from sail_amd.datagen.tpch import register_tpch
 
@@ -89,11 +95,24 @@ def main():
         qids = [int(x) for x in args.queries.split(",")]
     elif args.workload == "clickbench":
         qids = list(range(len(QUERIES)))
+    elif args.workload == "delta":
+        qids = sorted(QUERIES.keys())
     else:
         qids = list(range(1, 23))
-    plans = {q: session.plan_sql(QUERIES[q]) for q in qids}
+    if args.workload == "delta":
+        from sail_amd.datagen.delta_bench import _delta_sql
+
+        sqls = {q: _delta_sql(session, q) for q in qids}
+    else:
+        sqls = {q: QUERIES[q] for q in qids}
+    plans = {}
+    for q in qids:
+        if args.workload == "delta":
+            continue  # MERGE mutates state; plan fresh each step
+        plans[q] = session.plan_sql(sqls[q])
 
     bench_tables = (["hits"] if args.workload == "clickbench" else
+                    ["orders"] if args.workload == "delta" else
                     ["lineitem", "orders", "customer", "part", "partsupp", "supplier"])
     total_rows = sum(session.catalog.table_rows(t) or 0 for t in bench_tables)
     if dist is not None:
@@ -112,7 +131,10 @@ def main():
         times = {}
         for q in qids:
             tq = time.time()
-            chunk = session.execute_plan(plans[q])
+            if q in plans:
+                chunk = session.execute_plan(plans[q])
+            else:
+                chunk = session.sql(sqls[q]).collect_chunk()
             # result materialization to host is part of query completion
             if chunk.columns:
                 _ = chunk.columns[0].data.cpu() if hasattr(chunk.columns[0], "data") else None
@@ -149,6 +171,10 @@ def main():
             baseline = None
             metric = f"clickbench_{args.rows//1_000_000}m_total_s"
             model = "ClickBench (43 queries)"
+        elif args.workload == "delta":
+            baseline = None
+            metric = f"delta_sf{args.sf:g}_scan_merge_s"
+            model = "Delta Lake scan + MERGE INTO"
         else:
             baseline = 52.81 if abs(args.sf - 100.0) < 1e-6 else None
             metric = f"tpch_sf{args.sf:g}_total_s"

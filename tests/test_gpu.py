@@ -190,3 +190,33 @@ def test_hash_join_kernels_match_sort_join(ext):
             for j in range(int(lo[i]), int(hi[i])):
                 want.add((i, int(order[j])))
         assert got == want, (nb, np_, kr)
+
+
+def test_merge_into_on_gpu(tmp_path):
+    import sail_amd
+    from sail_amd.engine import types as T
+
+    s = sail_amd.SessionContext(device="cuda")
+    s.create_dataframe({"id": list(range(1000)), "v": [float(i) for i in range(1000)]},
+                       schema={"id": T.I64, "v": T.F64}, name="seed")
+    p = str(tmp_path / "dt")
+    s.table("seed").write.format("delta").mode("overwrite").save(p)
+    s.create_dataframe({"id": [5, 2000], "v": [555.0, -1.0]},
+                       schema={"id": T.I64, "v": T.F64}, name="src")
+    s.sql(f"MERGE INTO delta.`{p}` t USING src u ON t.id = u.id "
+          "WHEN MATCHED THEN UPDATE SET v = u.v "
+          "WHEN NOT MATCHED THEN INSERT (id, v) VALUES (u.id, u.v)")
+    rows = dict(s.read.format("delta").load(p).collect())
+    assert rows[5] == 555.0 and rows[2000] == -1.0 and len(rows) == 1001
+
+
+def test_update_delete_on_gpu():
+    import sail_amd
+    from sail_amd.engine import types as T
+
+    s = sail_amd.SessionContext(device="cuda")
+    s.create_dataframe({"id": [1, 2, 3], "v": [1.0, 2.0, 3.0]},
+                       schema={"id": T.I64, "v": T.F64}, name="t")
+    s.sql("UPDATE t SET v = v * 10 WHERE id > 1")
+    s.sql("DELETE FROM t WHERE v >= 30.0")
+    assert sorted(s.sql("SELECT * FROM t").collect()) == [(1, 1.0), (2, 20.0)]

@@ -67,8 +67,11 @@ def broadcast(v: Val, n: int, device) -> Column:
     val = v.value
     if isinstance(dt, T.DecimalType):
         val = _to_scaled(val, dt.scale)
-    elif isinstance(dt, T.DateType) and isinstance(val, str):
-        val = _date_str_to_days(val)
+    elif isinstance(dt, T.DateType):
+        if isinstance(val, str):
+            val = _date_str_to_days(val)
+        elif isinstance(val, _dt.date):
+            val = (val - _dt.date(1970, 1, 1)).days
     data = torch.full((n,), val, dtype=dt.storage, device=device)
     return Column(dt, data, None)
 

@@ -723,3 +723,320 @@ def _f_length_any(args, out, chunk, ev):
 
 
 _IMPLS["regexp_replace"] = _f_regexp_replace
+
+
+# ---------------------------------------------------------------------------
+# breadth batch: string functions via per-dictionary host transforms (dict
+# columns transform |dict| values, not |rows|) with raw-string fallback
+# ---------------------------------------------------------------------------
+
+def _dict_transform(fn):
+    """Wrap a str->str transform as a column impl preserving dict encoding."""
+
+    def impl(args, out, chunk, ev):
+        c = _col(args[0], chunk)
+        extra = [(_scalarize(a).value if _scalarize(a) else None) for a in args[1:]]
+
+        def f(v):
+            return fn(v, *extra)
+
+        if isinstance(c, StringColumn) and c.is_dict:
+            vals = [f(v) for v in c.dict_values()]
+            uniq = sorted(set(vals))
+            idx = {s: i for i, s in enumerate(uniq)}
+            lut = torch.tensor([idx[v] for v in vals], dtype=torch.int32, device=c.device)
+            from .column import _pack_strings
+
+            offs, byts = _pack_strings(uniq, c.device)
+            codes = lut[c.codes.to(torch.int64).clamp_min(0)]
+            codes = torch.where(c.codes >= 0, codes, c.codes)
+            return StringColumn(offs, byts, c.validity, codes)
+        return _str_map(c, f)
+
+    return impl
+
+
+def _dict_to_int(fn, out_type=None):
+    def impl(args, out, chunk, ev):
+        c = _col(args[0], chunk)
+        extra = [(_scalarize(a).value if _scalarize(a) else None) for a in args[1:]]
+
+        def f(v):
+            return fn(v, *extra)
+
+        ot = out_type or out or T.I32
+        if isinstance(c, StringColumn) and c.is_dict:
+            vals = [f(v) for v in c.dict_values()]
+            lut = torch.tensor(vals, dtype=ot.storage, device=c.device)
+            return Column(ot, lut[c.codes.to(torch.int64).clamp_min(0)], c.validity)
+        res = _str_map(c, f, out_is_string=False)
+        return Column.from_values(res, ot, device=chunk.device)
+
+    return impl
+
+
+def _f_concat_ws(args, out, chunk, ev):
+    sep = _scalarize(args[0]).value
+    n = chunk.num_rows
+    parts = [_col(a, chunk).to_pylist() for a in args[1:]]
+    res = [sep.join(str(v) for v in row if v is not None) for row in zip(*parts)]
+    return StringColumn.from_pylist(res, device=chunk.device)
+
+
+def _f_regexp_extract(args, out, chunk, ev):
+    import re as _re
+
+    pat = _re.compile(_scalarize(args[1]).value)
+    gi = _scalarize(args[2]).value if len(args) > 2 else 1
+
+    def f(v):
+        m = pat.search(v)
+        if not m:
+            return ""
+        return m.group(gi) if gi <= (m.lastindex or 0) else ("" if gi else m.group(0))
+
+    return _dict_transform(lambda v: f(v))(args[:1], out, chunk, ev)
+
+
+def _f_md5(args, out, chunk, ev):
+    import hashlib
+
+    return _dict_transform(lambda v: hashlib.md5(v.encode()).hexdigest())(args[:1], out, chunk, ev)
+
+
+def _f_sha2(args, out, chunk, ev):
+    import hashlib
+
+    bits = _scalarize(args[1]).value if len(args) > 1 else 256
+    algo = {0: "sha256", 224: "sha224", 256: "sha256", 384: "sha384", 512: "sha512"}[bits]
+
+    def f(v):
+        return getattr(hashlib, algo)(v.encode()).hexdigest()
+
+    return _dict_transform(lambda v: f(v))(args[:1], out, chunk, ev)
+
+
+def _f_months_between(args, out, chunk, ev):
+    a = _col(args[0], chunk)
+    b = _col(args[1], chunk)
+    ya, ma, da = _civil_from_days(_as_days(a))
+    yb, mb, db = _civil_from_days(_as_days(b))
+    months = (ya - yb) * 12 + (ma - mb)
+    frac = (da - db).to(torch.float64) / 31.0
+    return Column(T.F64, months.to(torch.float64) + frac, _merge(a, b))
+
+
+def _f_next_day(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    dow_name = _scalarize(args[1]).value.lower()[:3]
+    target = {"sun": 1, "mon": 2, "tue": 3, "wed": 4, "thu": 5, "fri": 6, "sat": 7}[dow_name]
+    days = _as_days(c)
+    cur = torch.remainder(days + 4, 7) + 1  # 1=Sunday
+    delta = torch.remainder(target - cur + 7 - 1, 7) + 1
+    return Column(T.DATE, (days + delta).to(torch.int32), c.validity)
+
+
+def _f_weekofyear(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    days = _as_days(c)
+    # ISO week number: Thursday-of-week determines the year
+    dow = torch.remainder(days + 3, 7)  # 0=Monday
+    thursday = days - dow + 3
+    y, m, d = _civil_from_days(thursday)
+    jan1 = _days_from_civil(y, torch.ones_like(y), torch.ones_like(y))
+    week = torch.div(thursday - jan1, 7, rounding_mode="floor") + 1
+    return Column(T.I32, week.to(torch.int32), c.validity)
+
+
+def _f_unix_timestamp(args, out, chunk, ev):
+    if not args:
+        import time as _time
+
+        from .eval import Scalar
+
+        return Scalar(int(_time.time()), T.I64)
+    c = _col(args[0], chunk)
+    if isinstance(c.dtype, T.TimestampType):
+        return Column(T.I64, torch.div(c.data, 1_000_000, rounding_mode="floor"), c.validity)
+    if isinstance(c.dtype, T.DateType):
+        return Column(T.I64, c.data.to(torch.int64) * 86400, c.validity)
+    raise NotImplementedError("unix_timestamp on strings with format TODO")
+
+
+def _f_from_unixtime_ts(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    return Column(T.TIMESTAMP, c.data.to(torch.int64) * 1_000_000, c.validity)
+
+
+def _f_make_timestamp(args, out, chunk, ev):
+    y = _col(args[0], chunk).data.to(torch.int64)
+    mo = _col(args[1], chunk).data.to(torch.int64)
+    d = _col(args[2], chunk).data.to(torch.int64)
+    h = _col(args[3], chunk).data.to(torch.int64)
+    mi = _col(args[4], chunk).data.to(torch.int64)
+    se = _col(args[5], chunk).data.to(torch.float64)
+    days = _days_from_civil(y, mo, d)
+    us = (days * 86400 + h * 3600 + mi * 60) * 1_000_000 + (se * 1e6).to(torch.int64)
+    return Column(T.TIMESTAMP, us, None)
+
+
+def _f_nvl2(args, out, chunk, ev):
+    from .eval import cast_value
+
+    a = _col(args[0], chunk)
+    when_ok = _col(cast_value(args[1], out, chunk), chunk)
+    when_null = _col(cast_value(args[2], out, chunk), chunk)
+    ok = a.valid_mask()
+    data = torch.where(ok, when_ok.data, when_null.data.to(when_ok.data.dtype))
+    valid = torch.where(ok, when_ok.valid_mask(), when_null.valid_mask())
+    return Column(out, data, None if bool(valid.all()) else valid.to(torch.uint8))
+
+
+def _f_bit_ops(op):
+    def impl(args, out, chunk, ev):
+        a = _col(args[0], chunk)
+        if len(args) == 1:
+            return Column(a.dtype, ~a.data, a.validity)
+        b = _col(args[1], chunk)
+        x, y = a.data.to(torch.int64), b.data.to(torch.int64)
+        data = {"shl": x << y, "shr": x >> y, "and": x & y, "or": x | y, "xor": x ^ y}[op]
+        return Column(T.I64, data, _merge(a, b))
+
+    return impl
+
+
+def _f_hex(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    if isinstance(c, StringColumn):
+        return _dict_transform(lambda v: v.encode().hex().upper())(args, out, chunk, ev)
+    vals = c.to_pylist()
+    return StringColumn.from_pylist([format(int(v), "X") if v is not None else None for v in vals],
+                                    device=chunk.device)
+
+
+def _f_factorial(args, out, chunk, ev):
+    import math as _m
+
+    c = _col(args[0], chunk)
+    lut = torch.tensor([_m.factorial(i) for i in range(21)], dtype=torch.int64, device=c.device)
+    x = c.data.to(torch.int64).clamp(0, 20)
+    return Column(T.I64, lut[x], c.validity)
+
+
+def _f_pi(args, out, chunk, ev):
+    from .eval import Scalar
+
+    return Scalar(math.pi, T.F64)
+
+
+def _f_e(args, out, chunk, ev):
+    from .eval import Scalar
+
+    return Scalar(math.e, T.F64)
+
+
+def _f_format_number(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    nd = _scalarize(args[1]).value
+    vals = c.to_pylist()
+    return StringColumn.from_pylist(
+        [None if v is None else f"{v:,.{nd}f}" for v in vals], device=chunk.device)
+
+
+def _f_atan2(args, out, chunk, ev):
+    a = _col(args[0], chunk)
+    b = _col(args[1], chunk)
+    return Column(T.F64, torch.atan2(a.data.to(torch.float64), b.data.to(torch.float64)),
+                  _merge(a, b))
+
+
+def _f_log_base(args, out, chunk, ev):
+    if len(args) == 1:
+        c = _col(args[0], chunk)
+        return Column(T.F64, torch.log(c.data.to(torch.float64)), c.validity)
+    base = _scalarize(args[0]).value
+    c = _col(args[1], chunk)
+    return Column(T.F64, torch.log(c.data.to(torch.float64)) / math.log(base), c.validity)
+
+
+_IMPLS.update({
+    "ltrim": _dict_transform(lambda v: v.lstrip()),
+    "rtrim": _dict_transform(lambda v: v.rstrip()),
+    "btrim": _dict_transform(lambda v, chars=None: v.strip(chars)),
+    "reverse": _dict_transform(lambda v: v[::-1]),
+    "repeat": _dict_transform(lambda v, n: v * int(n)),
+    "initcap": _dict_transform(lambda v: " ".join(w.capitalize() for w in v.split(" "))),
+    "lpad": _dict_transform(lambda v, n, p=" ": v[:int(n)] if len(v) >= int(n) else (p * int(n))[: int(n) - len(v)] + v),
+    "rpad": _dict_transform(lambda v, n, p=" ": v[:int(n)] if len(v) >= int(n) else v + (p * int(n))[: int(n) - len(v)]),
+    "left": _dict_transform(lambda v, n: v[:int(n)]),
+    "right": _dict_transform(lambda v, n: v[-int(n):] if int(n) else ""),
+    "translate": _dict_transform(lambda v, frm, to: v.translate(str.maketrans(frm[:len(to)], to[:len(frm)]))),
+    "substring_index": _dict_transform(
+        lambda v, d, n: d.join(v.split(d)[:int(n)]) if int(n) > 0 else d.join(v.split(d)[int(n):])),
+    "soundex": _dict_transform(lambda v: _soundex(v)),
+    "ascii": _dict_to_int(lambda v: ord(v[0]) if v else 0),
+    "position": _f_instr,
+    "levenshtein": _dict_to_int(lambda v, w: _levenshtein(v, w)),
+    "concat_ws": _f_concat_ws,
+    "regexp_extract": _f_regexp_extract,
+    "md5": _f_md5,
+    "sha2": _f_sha2,
+    "chr": _dict_transform(lambda v: v),  # placeholder; numeric chr below
+    "months_between": _f_months_between,
+    "next_day": _f_next_day,
+    "weekofyear": _f_weekofyear,
+    "unix_timestamp": _f_unix_timestamp, "to_unix_timestamp": _f_unix_timestamp,
+    "timestamp_seconds": _f_from_unixtime_ts,
+    "make_timestamp": _f_make_timestamp,
+    "nvl2": _f_nvl2,
+    "shiftleft": _f_bit_ops("shl"), "shiftright": _f_bit_ops("shr"),
+    "bitwise_not": _f_bit_ops("not"),
+    "hex": _f_hex,
+    "factorial": _f_factorial,
+    "pi": _f_pi, "e": _f_e,
+    "format_number": _f_format_number,
+    "atan2": _f_atan2,
+    "log": _f_log_base,
+})
+
+
+def _f_chr(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    vals = c.to_pylist()
+    return StringColumn.from_pylist(
+        [None if v is None else chr(int(v) % 256) for v in vals], device=chunk.device)
+
+
+_IMPLS["char"] = _f_chr
+_IMPLS["chr"] = _f_chr
+
+
+def _soundex(v: str) -> str:
+    if not v:
+        return ""
+    codes = {"b": "1", "f": "1", "p": "1", "v": "1", "c": "2", "g": "2", "j": "2",
+             "k": "2", "q": "2", "s": "2", "x": "2", "z": "2", "d": "3", "t": "3",
+             "l": "4", "m": "5", "n": "5", "r": "6"}
+    s = v.lower()
+    out = v[0].upper()
+    prev = codes.get(s[0], "")
+    for ch in s[1:]:
+        code = codes.get(ch, "")
+        if code and code != prev:
+            out += code
+            if len(out) == 4:
+                break
+        prev = code if ch not in "hw" else prev
+    return (out + "000")[:4]
+
+
+def _levenshtein(a: str, b: str) -> int:
+    if len(a) < len(b):
+        a, b = b, a
+    prev = list(range(len(b) + 1))
+    for i, ca in enumerate(a, 1):
+        cur = [i]
+        for j, cb in enumerate(b, 1):
+            cur.append(min(prev[j] + 1, cur[j - 1] + 1, prev[j - 1] + (ca != cb)))
+        prev = cur
+    return prev[-1]

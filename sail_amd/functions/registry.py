@@ -141,6 +141,7 @@ _reg("positive negative", _same)
 _reg("width_bucket", _i64)
 _reg("try_add try_subtract try_multiply try_divide", _numeric_common)
 _reg("rint", _f64)
+_reg("log", _f64)
 
 # string (ref: sail-function/src/scalar/string)
 _reg("concat concat_ws upper ucase lower lcase trim ltrim rtrim btrim initcap "

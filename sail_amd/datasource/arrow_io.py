@@ -21,22 +21,51 @@ def _arrow_type(t: T.DataType) -> pa.DataType:
 
 
 def column_to_arrow(c: Column, t: T.DataType) -> pa.Array:
-    if isinstance(c, StringColumn):
-        return pa.array(c.to_pylist(), type=pa.string())
     import numpy as np
 
+    if isinstance(c, StringColumn):
+        if c.is_dict:
+            # keep dictionary encoding end-to-end (parquet dictionary pages)
+            codes = c.codes.cpu().numpy()
+            vals = _raw_string_array(c.offsets.cpu(), c.bytes_.cpu())
+            mask = None
+            if c.validity is not None:
+                mask = ~c.validity.cpu().numpy().astype(bool)
+            ind = pa.array(codes, type=pa.int32(), mask=mask)
+            return pa.DictionaryArray.from_arrays(ind, vals)
+        arr = _raw_string_array(c.offsets.cpu(), c.bytes_.cpu(),
+                                validity=c.validity.cpu() if c.validity is not None else None)
+        return arr
     data = c.data.cpu().numpy()
     mask = None
     if c.validity is not None:
         mask = ~c.validity.cpu().numpy().astype(bool)
     if isinstance(t, T.DecimalType):
-        from decimal import Decimal
-
-        scale = t.scale
-        vals = [None if (mask is not None and mask[i]) else Decimal(int(data[i])).scaleb(-scale)
-                for i in range(len(data))]
-        return pa.array(vals, type=_arrow_type(t))
+        # vectorized int64 -> decimal128 (16-byte little-endian two's complement)
+        lo = data.astype(np.int64)
+        buf = np.zeros((len(lo), 2), dtype=np.int64)
+        buf[:, 0] = lo
+        buf[:, 1] = np.where(lo < 0, -1, 0)  # sign extension
+        validity_buf = None
+        if mask is not None:
+            validity_buf = pa.py_buffer(np.packbits(~mask, bitorder="little").tobytes())
+        return pa.Array.from_buffers(_arrow_type(t), len(lo),
+                                     [validity_buf, pa.py_buffer(buf.tobytes())])
     return pa.array(data, type=_arrow_type(t), mask=mask)
+
+
+def _raw_string_array(offsets, bytes_, validity=None) -> pa.Array:
+    import numpy as np
+
+    n = offsets.numel() - 1
+    validity_buf = None
+    if validity is not None:
+        validity_buf = pa.py_buffer(
+            np.packbits(validity.numpy().astype(bool), bitorder="little").tobytes())
+    return pa.Array.from_buffers(
+        pa.large_string(), n,
+        [validity_buf, pa.py_buffer(offsets.numpy().tobytes()),
+         pa.py_buffer(bytes_.numpy().tobytes())])
 
 
 def chunk_to_arrow(chunk: Chunk, schema) -> pa.Table:
@@ -64,19 +93,36 @@ def arrow_column(col: pa.ChunkedArray, device="cpu", dict_encode=True) -> Column
     at = col.type
     combined = col.combine_chunks() if isinstance(col, pa.ChunkedArray) else col
     if pa.types.is_dictionary(at):
+        arr = combined.combine_chunks() if isinstance(combined, pa.ChunkedArray) else combined
+        if pa.types.is_string(at.value_type) or pa.types.is_large_string(at.value_type):
+            vals = arr.dictionary
+            # engine invariant: dictionaries sorted — remap if needed
+            import numpy as np
+
+            pyvals = vals.to_pylist()
+            order = sorted(range(len(pyvals)), key=lambda i: pyvals[i] if pyvals[i] is not None else "")
+            sorted_vals = [pyvals[i] for i in order]
+            remap = np.empty(len(pyvals), dtype=np.int32)
+            for new, old in enumerate(order):
+                remap[old] = new
+            codes_np = arr.indices.to_numpy(zero_copy_only=False).astype(np.int64)
+            codes = torch.from_numpy(remap[codes_np].astype(np.int32)).to(device)
+            from ..engine.column import _pack_strings
+
+            offs, byts = _pack_strings([v or "" for v in sorted_vals], device)
+            validity = _validity(arr, device)
+            return StringColumn(offs, byts, validity, codes)
         combined = combined.cast(at.value_type)
         at = at.value_type
     if pa.types.is_string(at) or pa.types.is_large_string(at):
-        vals = combined.to_pylist()
-        return StringColumn.from_pylist(vals, device=device, dict_encode=None if dict_encode else False)
+        return _string_from_arrow(combined, device)
     if pa.types.is_decimal(at):
         scale = at.scale
-        vals = combined.to_pylist()
-        data = torch.tensor([0 if v is None else int(v.scaleb(scale)) for v in vals],
-                            dtype=torch.int64, device=device)
-        validity = None
-        if combined.null_count:
-            validity = torch.tensor([v is not None for v in vals], dtype=torch.uint8, device=device)
+        arr = combined.combine_chunks() if isinstance(combined, pa.ChunkedArray) else combined
+        buf = np.frombuffer(arr.buffers()[1], dtype=np.int64).reshape(-1, 2)
+        lo = buf[arr.offset : arr.offset + len(arr), 0].copy()
+        data = torch.from_numpy(lo).to(device)
+        validity = _validity(arr, device)
         return Column(T.DecimalType(at.precision, scale), data, validity)
     npmap = {pa.bool_(): (torch.bool, T.BOOL), pa.int8(): (torch.int8, T.I8),
              pa.int16(): (torch.int16, T.I16), pa.int32(): (torch.int32, T.I32),
@@ -97,6 +143,27 @@ def arrow_column(col: pa.ChunkedArray, device="cpu", dict_encode=True) -> Column
             validity = _validity(combined, device)
             return Column(et, data, validity)
     raise ValueError(f"unsupported arrow type {at}")
+
+
+def _string_from_arrow(arr: pa.Array, device) -> StringColumn:
+    import numpy as np
+
+    if isinstance(arr, pa.ChunkedArray):
+        arr = arr.combine_chunks()
+    if arr.offset != 0:
+        # re-materialize so buffer offsets start at zero
+        arr = pa.concat_arrays([arr])
+    at = arr.type
+    bufs = arr.buffers()
+    if pa.types.is_large_string(at):
+        offs = np.frombuffer(bufs[1], dtype=np.int64)[: len(arr) + 1].copy()
+    else:
+        offs = np.frombuffer(bufs[1], dtype=np.int32)[: len(arr) + 1].astype(np.int64)
+    byts = np.frombuffer(bufs[2], dtype=np.uint8)[: offs[-1]].copy() if bufs[2] is not None else np.zeros(0, np.uint8)
+    validity = _validity(arr, device)
+    col = StringColumn(torch.from_numpy(offs).to(device),
+                       torch.from_numpy(byts).to(device), validity)
+    return col
 
 
 def _validity(arr: pa.Array, device):

@@ -106,7 +106,11 @@ def arrow_column(col: pa.ChunkedArray, device="cpu", dict_encode=True) -> Column
             for new, old in enumerate(order):
                 remap[old] = new
             codes_np = arr.indices.to_numpy(zero_copy_only=False).astype(np.int64)
-            codes = torch.from_numpy(remap[codes_np].astype(np.int32)).to(device)
+            nullmask = (codes_np < 0) | (codes_np >= len(pyvals))
+            codes_np = np.where(nullmask, 0, codes_np)
+            mapped = remap[codes_np].astype(np.int32)
+            mapped = np.where(nullmask, np.int32(-1), mapped)
+            codes = torch.from_numpy(mapped).to(device)
             from ..engine.column import _pack_strings
 
             offs, byts = _pack_strings([v or "" for v in sorted_vals], device)

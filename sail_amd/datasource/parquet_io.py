@@ -77,6 +77,6 @@ def write(path: str, chunk, mode: str, options: Dict[str, str]):
             target = os.path.join(path, "part-00000.parquet")
     if mode == "error" and os.path.exists(target):
         raise FileExistsError(target)
-    compression = options.get("compression", "zstd")
+    compression = options.get("compression", "snappy")
     pq.write_table(tbl, target, compression=compression)
     return target

@@ -349,9 +349,35 @@ class Executor:
 
     def _dist_aggregate(self, p: S.Aggregate, child: Chunk) -> Chunk:
         """Two-phase distributed aggregate: local partials -> all_gather the
-        (tiny) partial tables -> merge. Falls back to gathering the input for
-        non-decomposable aggregates (DISTINCT, percentiles, ...)."""
-        from ..exec.distributed import decompose_agg, gather_chunk
+        (tiny) partial tables -> merge. High-cardinality groups instead
+        SHUFFLE the input by group key (each group lands whole on one rank,
+        full local aggregation, output stays sharded) — gathered partials at
+        1e8+ groups would replicate the whole table. Falls back to gathering
+        the input for non-decomposable aggregates over low-cardinality
+        groups (DISTINCT, percentiles, ...)."""
+        from ..exec.distributed import decompose_agg, gather_chunk, shuffle_chunk
+
+        if p.group_by:
+            n0, dev0 = child.num_rows, child.device
+            key_cols0 = [broadcast(self.ev.eval(g, child), n0, dev0) for g in p.group_by]
+            if n0:
+                _, _, ng_local = group_ids(key_cols0)
+            else:
+                ng_local = 0
+            threshold = int(self.ctx.session.conf.get(
+                "sail.exec.agg_shuffle_threshold_groups", str(4_000_000)))
+            if ng_local * self.dctx.world > threshold:
+                ncols = len(child.columns)
+                ext = Chunk(list(child.columns) + key_cols0,
+                            list(child.names) + [f"__k{i}" for i in range(len(key_cols0))],
+                            "sharded")
+                shuffled = shuffle_chunk(ext, list(range(ncols, ncols + len(key_cols0))),
+                                         self.dctx)
+                base = Chunk(shuffled.columns[:ncols], list(child.names), "replicated")
+                out = self._local_aggregate_with_keys(
+                    p, base, shuffled.columns[ncols:])
+                out.partitioning = "sharded"
+                return out
 
         decomps = [decompose_agg(a) for a in p.aggs]
         if any(d is None for d in decomps):
@@ -417,6 +443,28 @@ class Executor:
             ci += k
         return Chunk(out_keys + out_cols, [nm for nm, _ in p.schema])
 
+    def _local_aggregate_with_keys(self, p: S.Aggregate, child: Chunk,
+                                   key_cols) -> Chunk:
+        """Local aggregation with pre-evaluated group-key columns."""
+        n, dev = child.num_rows, child.device
+        if n == 0:
+            return Chunk([c.slice(0, 0) for c in key_cols]
+                         + [_empty_agg_col(a, dev) for a in p.aggs],
+                         [nm for nm, _ in p.schema])
+        gid, rep, ng = group_ids(key_cols)
+        out_keys = [c.gather(rep) for c in key_cols]
+        args_list = [[broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
+                     for a in p.aggs]
+        fmasks = [self.ev.eval_mask(a.filter, child) if a.filter is not None else None
+                  for a in p.aggs]
+        from .aggregates import fused_agg_batch
+
+        agg_cols = fused_agg_batch(p.aggs, args_list, fmasks, gid, ng)
+        if agg_cols is None:
+            agg_cols = [agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype)
+                        for a, args, fmask in zip(p.aggs, args_list, fmasks)]
+        return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
+
     def _local_aggregate(self, p: S.Aggregate, child: Chunk) -> Chunk:
         n = child.num_rows
         dev = child.device
@@ -454,8 +502,11 @@ class Executor:
             if lp == "sharded" or rp == "sharded":
                 out_part = "sharded"
                 if rp == "sharded" and lp == "sharded":
-                    # broadcast join: replicate the (smaller) build side
-                    right = self._gather(right)
+                    if self._should_shuffle(p, left, right):
+                        left, right = self._shuffle_join_inputs(p, left, right)
+                    else:
+                        # broadcast join: replicate the (smaller) build side
+                        right = self._gather(right)
                 elif lp == "replicated" and rp == "sharded":
                     if p.how not in ("inner", "cross"):
                         # left-side semantics need every left row exactly once
@@ -465,6 +516,31 @@ class Executor:
                 out.partitioning = out_part
                 return out
         return join_chunks(self.ev, p, left, right)
+
+    def _should_shuffle(self, p: S.Join, left: Chunk, right: Chunk) -> bool:
+        """Shuffle both sides instead of broadcasting the build side when the
+        replicated build side would exceed the broadcast budget (xGMI has the
+        bandwidth, HBM capacity is the constraint at SF1000 scale)."""
+        if p.how not in ("inner", "semi", "anti", "left"):
+            return False
+        from .executor import split_join_condition
+
+        equi, _ = split_join_condition(p.on, len(left.columns)) if p.on is not None else ([], None)
+        if not equi:
+            return False
+        threshold = int(self.ctx.session.conf.get(
+            "sail.exec.broadcast_threshold_bytes", str(2 << 30)))
+        return _chunk_bytes(right) * self.dctx.world > threshold
+
+    def _shuffle_join_inputs(self, p: S.Join, left: Chunk, right: Chunk):
+        from ..exec.distributed import shuffle_chunk
+        from .executor import split_join_condition
+
+        equi, _ = split_join_condition(p.on, len(left.columns))
+        lkeys = [i for i, _ in equi]
+        rkeys = [j - len(left.columns) for _, j in equi]
+        return (shuffle_chunk(left, lkeys, self.dctx),
+                shuffle_chunk(right, rkeys, self.dctx))
 
     # -- set ops -----------------------------------------------------------
     def _x_SetOp(self, p: S.SetOp) -> Chunk:
@@ -852,3 +928,15 @@ def _empty_partial(pname: str, dev) -> Column:
     if pname in ("count", "count_if", "sum"):
         return Column(T.I64, torch.zeros(0, dtype=torch.int64, device=dev))
     return Column(T.F64, torch.zeros(0, dtype=torch.float64, device=dev))
+
+
+def _chunk_bytes(chunk: Chunk) -> int:
+    total = 0
+    for c in chunk.columns:
+        if c is None:
+            continue
+        if isinstance(c, StringColumn):
+            total += int(c.bytes_.numel()) + 8 * len(c)
+        else:
+            total += c.data.numel() * c.data.element_size()
+    return total

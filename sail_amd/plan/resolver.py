@@ -333,6 +333,23 @@ class Resolver:
                 if _expr_equal_unbound(e, ge):
                     return S.BoundRef(gi, _expr_name(e, gi), be.dtype)
             if isinstance(e, S.AggFunc):
+                if e.name.lower() in ("grouping", "grouping_id"):
+                    # positions of the referenced group keys (executor fills
+                    # per-grouping-set values)
+                    pos = []
+                    for a in e.args:
+                        for gi, ge in enumerate(group_exprs):
+                            if _expr_equal_unbound(a, ge):
+                                pos.append(gi)
+                                break
+                        else:
+                            raise ResolutionError("grouping() argument must be a group key")
+                    rt = T.I32 if e.name.lower() == "grouping" else T.I64
+                    bound = S.AggFunc(e.name.lower(),
+                                      [S.Literal(p_, T.I32) for p_ in pos], False, rt, None)
+                    agg_funcs.append(bound)
+                    return S.BoundRef(len(bound_groups) + len(agg_funcs) - 1,
+                                      e.name.lower(), rt)
                 if len(e.args) == 1 and isinstance(e.args[0], S.Star):
                     bargs = []
                 else:

@@ -450,27 +450,52 @@ class Parser:
             plan = S.Filter(input=plan, condition=self.parse_expr())
 
         group_by: List[S.Expr] = []
-        grouping_kind = None
+        grouping_sets = None
         if self.eat_kw("GROUP"):
             self.expect_kw("BY")
             if self.eat_kw("ROLLUP"):
-                grouping_kind = "rollup"
                 self.expect_op("(")
                 group_by = self._expr_list()
                 self.expect_op(")")
+                grouping_sets = [list(range(k)) for k in range(len(group_by), -1, -1)]
             elif self.eat_kw("CUBE"):
-                grouping_kind = "cube"
                 self.expect_op("(")
                 group_by = self._expr_list()
                 self.expect_op(")")
+                n_ = len(group_by)
+                grouping_sets = [[i for i in range(n_) if m & (1 << i)]
+                                 for m in range((1 << n_) - 1, -1, -1)]
             elif self.eat_kw("GROUPING"):
                 self.expect_kw("SETS")
-                raise SqlError("GROUPING SETS not yet supported")
+                self.expect_op("(")
+                sets_exprs = []
+                while True:
+                    self.expect_op("(")
+                    one = [] if self.at_op(")") else self._expr_list()
+                    self.expect_op(")")
+                    sets_exprs.append(one)
+                    if not self.eat_op(","):
+                        break
+                self.expect_op(")")
+                # group_by = union of all expressions, sets = index lists
+                keymap = {}
+                for se in sets_exprs:
+                    for e in se:
+                        k = repr(e)
+                        if k not in keymap:
+                            keymap[k] = (len(group_by), e)
+                            group_by.append(e)
+                grouping_sets = [[keymap[repr(e)][0] for e in se] for se in sets_exprs]
             else:
                 group_by = self._expr_list()
                 if self.eat_kw("WITH"):
                     kw = self.next().upper
-                    grouping_kind = kw.lower()
+                    if kw == "ROLLUP":
+                        grouping_sets = [list(range(k)) for k in range(len(group_by), -1, -1)]
+                    elif kw == "CUBE":
+                        n_ = len(group_by)
+                        grouping_sets = [[i for i in range(n_) if m & (1 << i)]
+                                         for m in range((1 << n_) - 1, -1, -1)]
 
         having = None
         if self.eat_kw("HAVING"):
@@ -481,8 +506,7 @@ class Parser:
             having is not None and self._contains_agg(having))
         if has_agg:
             plan = S.Aggregate(input=plan, group_by=group_by, aggs=projections,
-                               grouping_sets=[[grouping_kind]] if grouping_kind else None,
-                               having=having)
+                               grouping_sets=grouping_sets, having=having)
         else:
             if having is not None:
                 plan = S.Filter(input=plan, condition=having)

@@ -215,3 +215,18 @@ def test_window_row_number(s):
         "SELECT a, row_number() OVER (PARTITION BY c ORDER BY a) AS rn FROM t WHERE c = 'x' ORDER BY a"
     ).collect()
     assert rows == [(1, 1), (3, 2)]
+
+
+def test_rollup_cube_grouping_sets():
+    s2 = sail_amd.SessionContext(device="cpu")
+    s2.create_dataframe({"a": ["x", "x", "y"], "b": ["p", "q", "p"], "v": [1, 2, 3]}, name="g")
+    rows = s2.sql("SELECT a, b, sum(v) FROM g GROUP BY ROLLUP(a, b) "
+                  "ORDER BY a NULLS LAST, b NULLS LAST").collect()
+    assert rows == [("x", "p", 1), ("x", "q", 2), ("x", None, 3),
+                    ("y", "p", 3), ("y", None, 3), (None, None, 6)]
+    rows = s2.sql("SELECT a, grouping(a), count(*) FROM g GROUP BY CUBE(a) "
+                  "ORDER BY a NULLS LAST").collect()
+    assert rows == [("x", 0, 2), ("y", 0, 1), (None, 1, 3)]
+    rows = s2.sql("SELECT a, b, sum(v) FROM g GROUP BY GROUPING SETS ((a), (b)) "
+                  "ORDER BY a NULLS LAST, b NULLS LAST").collect()
+    assert rows == [("x", None, 3), ("y", None, 3), (None, "p", 4), (None, "q", 2)]

@@ -1,0 +1,51 @@
+"""Window function coverage (ref: sail-function/src/window, BoundedWindowAggExec role)."""
+import pytest
+
+import sail_amd
+from sail_amd.engine import types as T
+
+
+@pytest.fixture()
+def s():
+    s = sail_amd.SessionContext(device="cpu")
+    s.create_dataframe(
+        {"g": ["a", "a", "a", "b", "b"], "o": [1, 2, 3, 1, 2],
+         "v": [10, 20, 30, 5, 15]},
+        schema={"g": T.STRING, "o": T.I32, "v": T.I64}, name="w")
+    return s
+
+
+def test_row_number_rank(s):
+    rows = s.sql("SELECT g, o, row_number() OVER (PARTITION BY g ORDER BY o DESC) AS rn "
+                 "FROM w ORDER BY g, o").collect()
+    assert rows == [("a", 1, 3), ("a", 2, 2), ("a", 3, 1), ("b", 1, 2), ("b", 2, 1)]
+
+
+def test_rank_with_ties(s):
+    s.create_dataframe({"x": [1, 1, 2, 3, 3, 3]}, name="r")
+    rows = s.sql("SELECT x, rank() OVER (ORDER BY x) AS r, dense_rank() OVER (ORDER BY x) AS d "
+                 "FROM r ORDER BY x").collect()
+    assert rows == [(1, 1, 1), (1, 1, 1), (2, 3, 2), (3, 4, 3), (3, 4, 3), (3, 4, 3)]
+
+
+def test_running_sum(s):
+    rows = s.sql("SELECT g, o, sum(v) OVER (PARTITION BY g ORDER BY o) AS rs "
+                 "FROM w ORDER BY g, o").collect()
+    assert rows == [("a", 1, 10), ("a", 2, 30), ("a", 3, 60), ("b", 1, 5), ("b", 2, 20)]
+
+
+def test_whole_partition_agg(s):
+    rows = s.sql("SELECT g, v, sum(v) OVER (PARTITION BY g) AS tot FROM w ORDER BY g, o").collect()
+    assert rows == [("a", 10, 60), ("a", 20, 60), ("a", 30, 60), ("b", 5, 20), ("b", 15, 20)]
+
+
+def test_lag_lead(s):
+    rows = s.sql("SELECT g, o, lag(v) OVER (PARTITION BY g ORDER BY o) AS lg, "
+                 "lead(v) OVER (PARTITION BY g ORDER BY o) AS ld FROM w ORDER BY g, o").collect()
+    assert rows == [("a", 1, None, 20), ("a", 2, 10, 30), ("a", 3, 20, None),
+                    ("b", 1, None, 15), ("b", 2, 5, None)]
+
+
+def test_ntile(s):
+    rows = s.sql("SELECT o, ntile(2) OVER (ORDER BY o) AS nt FROM w WHERE g = 'a' ORDER BY o").collect()
+    assert rows == [(1, 1), (2, 1), (3, 2)]

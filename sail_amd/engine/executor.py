@@ -116,6 +116,9 @@ class Executor:
             out.partitioning = "sharded"
         return out
 
+    def _x_ChunkSource(self, p: S.ChunkSource) -> Chunk:
+        return p.chunk
+
     def _x_DataSourceRead(self, p: S.DataSourceRead) -> Chunk:
         from ..datasource.registry import read_source
 

@@ -195,7 +195,25 @@ class SessionContext:
         self.conf: Dict[str, str] = load_config()
         self.udfs: Dict[str, tuple] = {}
         self.query_log: List[dict] = []
+        self._streams = None
         self._register_system_tables()
+
+    @property
+    def read_stream(self):
+        """Structured-streaming reader (ref: Spark spark.readStream)."""
+        from ..streaming.reader import DataStreamReader
+
+        return DataStreamReader(self)
+
+    readStream = read_stream
+
+    @property
+    def streams(self):
+        if self._streams is None:
+            from ..streaming.reader import StreamingQueryManager
+
+            self._streams = StreamingQueryManager()
+        return self._streams
 
     # -- pipeline ----------------------------------------------------------
     def parse(self, sql: str) -> S.Plan:

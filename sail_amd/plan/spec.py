@@ -347,6 +347,15 @@ class LocalRelation(Plan):
 
 
 @dataclass
+class ChunkSource(Plan):
+    """Pre-materialized chunk spliced into a plan (streaming incremental
+    aggregation substitutes the Aggregate subtree with its merged state)."""
+
+    chunk: object = None  # engine.chunk.Chunk
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+
+@dataclass
 class Project(Plan):
     input: Plan = None
     exprs: List[Expr] = field(default_factory=list)

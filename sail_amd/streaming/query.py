@@ -1,0 +1,428 @@
+"""Micro-batch streaming query runner.
+
+One StreamingQuery = one source + one batch plan + one sink, driven by a
+trigger loop on a daemon thread:
+
+  1. end = source.latest_offset(); nothing new -> idle (or finish for
+     trigger once / availableNow)
+  2. WAL: checkpoint offsets/<batch>.json BEFORE executing (Spark layout)
+  3. register the batch rows as the stream's table; execute the plan
+       - stateless plan  -> plan over just the new rows (append mode)
+       - aggregation     -> incremental: per-batch partial aggregates merged
+         into a persistent state table via the same partial/merge
+         decomposition as the distributed two-phase aggregate
+         (exec/distributed.decompose_agg); non-decomposable aggregates
+         (DISTINCT, percentiles) fall back to retained-input re-aggregation
+  4. sink.write(result, batch_id, mode)
+  5. checkpoint commits/<batch>.json + state/ parquet -> batch is durable;
+     on restart the query replays any offset without a commit.
+
+The reference threads Chandy-Lamport flow markers through operator streams
+(ref: sail-common-datafusion/src/streaming/event/marker.rs FlowMarker
+{Watermark, Checkpoint, EndOfData}); micro-batch boundaries play the same
+role here — every batch boundary IS a consistent cut, so the offset WAL +
+state snapshot gives the same recovery guarantee without per-operator
+marker plumbing.
+"""
+from __future__ import annotations
+
+import json
+import os
+import threading
+import time
+import uuid
+from typing import List, Optional
+
+import torch
+
+from ..engine import types as T
+from ..engine.chunk import Chunk
+from ..engine.column import Column
+from ..plan import spec as S
+from .sinks import StreamSink
+from .sources import StreamSource
+
+_PASSTHROUGH = (S.Project, S.Filter, S.Sort, S.Limit, S.SubqueryAlias)
+
+
+def _find_aggregate(plan: S.Plan):
+    """Locate a single Aggregate reachable through passthrough nodes from the
+    root. Returns (agg, parent) — parent is None when the root IS the
+    aggregate — or None when the plan has no such spine (stateless, or a
+    shape the incremental path doesn't cover)."""
+    node, parent = plan, None
+    while True:
+        if isinstance(node, S.Aggregate):
+            return node, parent
+        if isinstance(node, _PASSTHROUGH):
+            parent, node = node, node.input
+            continue
+        return None
+
+
+def _count_aggregates(plan: S.Plan) -> int:
+    n = 1 if isinstance(plan, S.Aggregate) else 0
+    for c in plan.children():
+        if c is not None:
+            n += _count_aggregates(c)
+    return n
+
+
+class _AggState:
+    """Persistent streaming-aggregation state: group keys + partial columns,
+    merged with each batch's local partials exactly like the distributed
+    merge phase (engine/executor.py _dist_aggregate steps 1+3)."""
+
+    def __init__(self, agg: S.Aggregate, decomps):
+        self.agg = agg
+        self.decomps = decomps
+        self.keys: Optional[List[Column]] = None      # one per group_by
+        self.partials: Optional[List[Column]] = None  # flattened per decomp
+
+    # -- persistence --------------------------------------------------------
+    def state_chunk(self) -> Optional[Chunk]:
+        if self.partials is None:
+            return None
+        keys = self.keys or []
+        return Chunk(list(keys) + list(self.partials),
+                     [f"k{i}" for i in range(len(keys))]
+                     + [f"p{i}" for i in range(len(self.partials))])
+
+    def load_chunk(self, chunk: Chunk):
+        nk = len(self.agg.group_by)
+        self.keys = list(chunk.columns[:nk])
+        self.partials = list(chunk.columns[nk:])
+
+    # -- update -------------------------------------------------------------
+    def update(self, ex, child: Chunk):
+        """Merge one batch into the state; returns (finalized_chunk,
+        touched_mask over output groups)."""
+        from ..engine.aggregates import agg_eval, global_ids, group_ids
+        from ..engine.eval import broadcast
+        from ..engine.executor import _empty_partial, concat_columns
+
+        agg = self.agg
+        n, dev = child.num_rows, child.device
+        # 1) batch-local partials
+        if agg.group_by:
+            key_cols = [broadcast(ex.ev.eval(g, child), n, dev) for g in agg.group_by]
+            if n:
+                gid, rep, ng = group_ids(key_cols)
+                batch_keys = [c.gather(rep) for c in key_cols]
+            else:
+                gid, ng = torch.zeros(0, dtype=torch.int64, device=dev), 0
+                batch_keys = key_cols
+        else:
+            gid, ng = (global_ids(n, dev) if n
+                       else (torch.zeros(0, dtype=torch.int64, device=dev), 0))
+            batch_keys = []
+        batch_partials: List[Column] = []
+        merge_names: List[str] = []
+        for a, d in zip(agg.aggs, self.decomps):
+            args = [broadcast(ex.ev.eval(x, child), n, dev) for x in a.args]
+            fmask = ex.ev.eval_mask(a.filter, child) if a.filter is not None else None
+            for pname, mname in zip(d.partials, d.merges):
+                use_args = args if (args or pname != "count") else []
+                col = agg_eval(pname, use_args, gid, ng, False, fmask,
+                               None) if ng else _empty_partial(pname, dev)
+                batch_partials.append(col)
+                merge_names.append(mname)
+        # 2) concat with state (state rows first), re-group, merge
+        if self.partials is None:
+            all_keys = batch_keys
+            all_partials = batch_partials
+            n_state = 0
+        else:
+            all_keys = [concat_columns([s, b]) for s, b in zip(self.keys, batch_keys)]
+            all_partials = [concat_columns([s, b])
+                            for s, b in zip(self.partials, batch_partials)]
+            n_state = self.partials[0].data.shape[0] if self.partials else 0
+        total = all_partials[0].data.shape[0] if all_partials else 0
+        if agg.group_by:
+            mgid, mrep, mng = group_ids(all_keys)
+            out_keys = [c.gather(mrep) for c in all_keys]
+        else:
+            mgid, mng = global_ids(total, dev) if total else (
+                torch.zeros(0, dtype=torch.int64, device=dev), 0)
+            out_keys = []
+        merged = [agg_eval(mname, [col], mgid, mng, False, None, None)
+                  for mname, col in zip(merge_names, all_partials)]
+        touched = torch.zeros(mng, dtype=torch.bool, device=dev)
+        if total > n_state:
+            touched[mgid[n_state:]] = True
+        # 3) new state + finalized output
+        self.keys = out_keys
+        self.partials = merged
+        out_cols = []
+        ci = 0
+        for a, d in zip(agg.aggs, self.decomps):
+            k = len(d.partials)
+            out_cols.append(d.finalize(merged[ci:ci + k], a.dtype))
+            ci += k
+        out = Chunk(out_keys + out_cols, [nm for nm, _ in agg.schema])
+        return out, touched
+
+
+class StreamingQuery:
+    """Handle on a running streaming query (ref: Spark StreamingQuery API:
+    stop/awaitTermination/processAllAvailable/lastProgress)."""
+
+    def __init__(self, session, source: StreamSource, sql: str, view_name: str,
+                 sink: StreamSink, output_mode: str = "append",
+                 trigger_interval: float = 0.1, trigger_once: bool = False,
+                 available_now: bool = False,
+                 checkpoint_location: Optional[str] = None,
+                 name: Optional[str] = None):
+        self.id = str(uuid.uuid4())
+        self.name = name
+        self.session = session
+        self.source = source
+        self.sql = sql
+        self.view_name = view_name
+        self.sink = sink
+        self.output_mode = output_mode
+        self.trigger_interval = trigger_interval
+        self.trigger_once = trigger_once
+        self.available_now = available_now
+        self.checkpoint = checkpoint_location
+        self.exception: Optional[BaseException] = None
+        self.last_progress: Optional[dict] = None
+        self.batch_id = -1
+        self._offset = source.initial_offset()
+        self._stop = threading.Event()
+        self._idle = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self._lock = threading.Lock()
+        self._plan = None
+        self._agg_state: Optional[_AggState] = None
+        self._upper_parent = None
+        self._retained: Optional[Chunk] = None  # non-incremental fallback
+        self._prepare()
+        if self.checkpoint:
+            self._recover()
+
+    # -- planning -----------------------------------------------------------
+    def _prepare(self):
+        # register an empty table of the source schema so the SQL resolves
+        empty = Chunk([Column.from_values([], t) for _, t in self.source.schema],
+                      [n for n, _ in self.source.schema])
+        self.session.catalog.register_table(
+            self.view_name, empty.to_table(), list(self.source.schema))
+        self._plan = self.session.plan_sql(self.sql)
+        if isinstance(self._plan, S.Command):
+            raise ValueError("streaming query must be a SELECT")
+        found = _find_aggregate(self._plan)
+        self._mode = "stateless"
+        if found is not None and _count_aggregates(self._plan) == 1:
+            agg, parent = found
+            from ..exec.distributed import decompose_agg
+
+            decomps = [decompose_agg(a) for a in agg.aggs]
+            ok = (not agg.grouping_sets and agg.having is None
+                  and all(d is not None for d in decomps)
+                  and all(not getattr(a, "distinct", False) for a in agg.aggs))
+            if ok:
+                self._mode = "incremental"
+                self._agg_state = _AggState(agg, decomps)
+                self._upper_parent = parent
+        elif _count_aggregates(self._plan) > 0:
+            self._mode = "retained"
+        if found is not None and self._mode == "stateless":
+            self._mode = "retained"
+        if self._mode == "stateless" and self.output_mode == "complete":
+            self._mode = "retained"  # complete over a stateless plan: re-run all
+        if self._mode != "stateless" and self.output_mode == "append":
+            raise ValueError(
+                "append output mode is not supported with streaming "
+                "aggregation (no watermark support); use complete or update")
+
+    # -- checkpoint ---------------------------------------------------------
+    def _ckpt_dir(self, sub: str) -> str:
+        d = os.path.join(self.checkpoint, sub)
+        os.makedirs(d, exist_ok=True)
+        return d
+
+    def _recover(self):
+        offs = self._ckpt_dir("offsets")
+        commits = self._ckpt_dir("commits")
+        done = sorted(int(f) for f in os.listdir(commits) if f.isdigit())
+        pending = sorted(int(f) for f in os.listdir(offs) if f.isdigit())
+        if done:
+            last = done[-1]
+            with open(os.path.join(offs, str(last))) as f:
+                self._offset = json.load(f)["offset"]
+            self.batch_id = last
+            # restore aggregation state
+            state_dir = os.path.join(self.checkpoint, "state")
+            if self._agg_state is not None and os.path.isdir(state_dir):
+                self._load_state(state_dir)
+        # a pending offset without a commit is replayed by the normal loop:
+        # read_between(self._offset, that offset) reproduces the batch.
+        if pending and (not done or pending[-1] > done[-1]):
+            self._pending_offset = self._load_offset(offs, pending[-1])
+            self._pending_id = pending[-1]
+        else:
+            self._pending_offset = None
+
+    @staticmethod
+    def _load_offset(d, i):
+        with open(os.path.join(d, str(i))) as f:
+            return json.load(f)["offset"]
+
+    def _save_state(self):
+        if self._agg_state is None or not self.checkpoint:
+            return
+        chunk = self._agg_state.state_chunk()
+        if chunk is None:
+            return
+        from ..datasource.delta import schema_to_string
+        from ..datasource.registry import write_source
+
+        state_dir = self._ckpt_dir("state")
+        schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
+        write_source("parquet", os.path.join(state_dir, "data"), chunk,
+                     "overwrite", {}, None)
+        with open(os.path.join(state_dir, "schema.json"), "w") as f:
+            f.write(schema_to_string(schema))
+
+    def _load_state(self, state_dir):
+        from ..datasource.delta import schema_from_string
+        from ..datasource.registry import read_source
+
+        sp = os.path.join(state_dir, "schema.json")
+        dp = os.path.join(state_dir, "data")
+        if not (os.path.exists(sp) and os.path.isdir(dp)):
+            return
+        with open(sp) as f:
+            schema = schema_from_string(f.read())
+        files = [os.path.join(dp, f) for f in sorted(os.listdir(dp))
+                 if f.endswith(".parquet")]
+        tbl = read_source("parquet", files, {}, schema, "cpu")
+        self._agg_state.load_chunk(Chunk.from_table(tbl))
+
+    # -- lifecycle ----------------------------------------------------------
+    def start(self) -> "StreamingQuery":
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name=f"stream-{self.name or self.id[:8]}")
+        self._thread.start()
+        return self
+
+    def stop(self):
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=30)
+
+    @property
+    def is_active(self) -> bool:
+        return self._thread is not None and self._thread.is_alive()
+
+    def await_termination(self, timeout: Optional[float] = None) -> bool:
+        if self._thread is None:
+            return True
+        self._thread.join(timeout)
+        if self.exception is not None:
+            raise self.exception
+        return not self._thread.is_alive()
+
+    def process_all_available(self, timeout: float = 30.0):
+        """Block until everything currently available has been committed."""
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            if self.exception is not None:
+                raise self.exception
+            if not self.is_active:
+                return
+            if self._offset == self.source.latest_offset() and self._idle.is_set():
+                return
+            time.sleep(0.01)
+        raise TimeoutError("process_all_available timed out")
+
+    # -- the loop -----------------------------------------------------------
+    def _run(self):
+        try:
+            while not self._stop.is_set():
+                progressed = self._run_one_batch()
+                if not progressed:
+                    self._idle.set()
+                    if self.trigger_once or self.available_now:
+                        return
+                    self._stop.wait(self.trigger_interval)
+                else:
+                    self._idle.clear()
+                    if self.trigger_once:
+                        return
+        except BaseException as e:  # surfaced via await_termination
+            self.exception = e
+        finally:
+            self._idle.set()
+
+    def _run_one_batch(self) -> bool:
+        if getattr(self, "_pending_offset", None) is not None:
+            end, bid = self._pending_offset, self._pending_id
+            self._pending_offset = None
+        else:
+            end = self.source.latest_offset()
+            if end == self._offset:
+                return False
+            bid = self.batch_id + 1
+            if self.checkpoint:
+                with open(os.path.join(self._ckpt_dir("offsets"), str(bid)), "w") as f:
+                    json.dump({"offset": end}, f)
+        t0 = time.time()
+        batch = self.source.read_between(self._offset, end)
+        nrows = batch.num_rows
+        result = self._execute_batch(batch)
+        if result is not None:
+            self.sink.write(result, bid, self.output_mode)
+        if self.checkpoint:
+            self._save_state()
+            with open(os.path.join(self._ckpt_dir("commits"), str(bid)), "w") as f:
+                json.dump({"batchId": bid}, f)
+        self._offset = end
+        self.batch_id = bid
+        self.last_progress = {
+            "id": self.id, "name": self.name, "batchId": bid,
+            "numInputRows": nrows,
+            "durationMs": round((time.time() - t0) * 1000, 3),
+            "sources": [{"endOffset": end}],
+        }
+        return True
+
+    def _execute_batch(self, batch: Chunk) -> Optional[Chunk]:
+        from ..engine.executor import ExecutionContext, Executor, concat_columns
+
+        cat = self.session.catalog
+        if self._mode == "retained":
+            if self._retained is None:
+                self._retained = batch
+            else:
+                self._retained = Chunk(
+                    [concat_columns([a, b]) for a, b in
+                     zip(self._retained.columns, batch.columns)],
+                    list(batch.names))
+            cat.register_table(self.view_name, self._retained.to_table(),
+                               list(self.source.schema))
+            return self.session.execute_plan(self._plan)
+        cat.register_table(self.view_name, batch.to_table(),
+                           list(self.source.schema))
+        if self._mode == "stateless":
+            if batch.num_rows == 0:
+                return None
+            return self.session.execute_plan(self._plan)
+        # incremental aggregation
+        ctx = ExecutionContext(self.session, self.session.device)
+        ex = Executor(ctx)
+        child = ex.execute(self._agg_state.agg.input)
+        finalized, touched = self._agg_state.update(ex, child)
+        if self.output_mode == "update":
+            idx = torch.nonzero(touched, as_tuple=False).flatten()
+            finalized = finalized.gather(idx)
+        if self._upper_parent is None:
+            return finalized
+        sub = S.ChunkSource(chunk=finalized, schema=self._agg_state.agg.schema)
+        orig = self._upper_parent.input
+        try:
+            self._upper_parent.input = sub
+            return ex.execute(self._plan)
+        finally:
+            self._upper_parent.input = orig

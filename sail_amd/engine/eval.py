@@ -191,11 +191,34 @@ class Evaluator:
         info = sess.udfs.get(e.name.lower()) if sess is not None else None
         if info is None:
             raise EvalError(f"UDF {e.name} not registered")
-        fn, rtype = info
-        cols = [self.eval_col(a, chunk).to_pylist() for a in e.args]
-        out = [fn(*vals) for vals in zip(*cols)] if cols else [fn() for _ in range(chunk.num_rows)]
+        fn, rtype = info[0], info[1]
+        vectorized = info[2] if len(info) > 2 else False
         from .column import Column as _C, StringColumn as _S
 
+        if vectorized:
+            # pandas/Arrow-batched UDF: one call over whole numpy arrays,
+            # the reference's vectorized PySpark UDF path
+            # (ref: crates/sail-python-udf pandas_udf) minus the IPC hop —
+            # the arrays come straight off the device tensors.
+            import numpy as np
+
+            args = []
+            for a in e.args:
+                c = self.eval_col(a, chunk)
+                if isinstance(c, _S):
+                    args.append(np.array(c.to_pylist(), dtype=object))
+                else:
+                    args.append(c.data.cpu().numpy())
+            out = fn(*args) if args else fn(chunk.num_rows)
+            if isinstance(rtype, T.StringType):
+                return _S.from_pylist(list(out), device=chunk.device)
+            arr = np.asarray(out)
+            data = torch.from_numpy(np.ascontiguousarray(arr))
+            if rtype.storage is not None and data.dtype != rtype.storage:
+                data = data.to(rtype.storage)
+            return _C(rtype, data.to(chunk.device))
+        cols = [self.eval_col(a, chunk).to_pylist() for a in e.args]
+        out = [fn(*vals) for vals in zip(*cols)] if cols else [fn() for _ in range(chunk.num_rows)]
         if isinstance(rtype, T.StringType):
             return _S.from_pylist(out, device=chunk.device)
         return _C.from_values(out, rtype, device=chunk.device)

@@ -201,6 +201,25 @@ class Executor:
         gid, rep, ng = group_ids(child.columns)
         return child.gather(rep)
 
+    def _x_Sample(self, p: S.Sample) -> Chunk:
+        """Bernoulli sampling with a deterministic generator (REPEATABLE
+        seed, or a fixed default so every SPMD rank draws the same mask for
+        replicated chunks); n ROWS takes the first n (Spark semantics)."""
+        child = self.execute(p.input)
+        if p.rows is not None:
+            if child.num_rows <= p.rows:
+                return child
+            return child.gather(torch.arange(p.rows, dtype=torch.int64,
+                                             device=child.device))
+        n = child.num_rows
+        if n == 0 or p.fraction is None:
+            return child
+        g = torch.Generator(device="cpu")
+        g.manual_seed(p.seed if p.seed is not None else 0x5A11)
+        mask = torch.rand(n, generator=g) < p.fraction
+        idx = torch.nonzero(mask, as_tuple=False).flatten().to(child.device)
+        return child.gather(idx)
+
     # -- sort --------------------------------------------------------------
     def _x_Sort(self, p: S.Sort) -> Chunk:
         child = self._gather(self.execute(p.input))

@@ -332,12 +332,14 @@ class UdfRegistry:
     def __init__(self, session: SessionContext):
         self._session = session
 
-    def register(self, name: str, fn, return_type=None):
+    def register(self, name: str, fn, return_type=None, vectorized: bool = False):
+        """vectorized=True: fn is called once per batch with whole numpy
+        arrays (pandas-UDF style) instead of once per row."""
         from . import types as T
 
         if isinstance(return_type, str):
             return_type = T.type_from_name(return_type)
-        self._session.udfs[name.lower()] = (fn, return_type or T.F64)
+        self._session.udfs[name.lower()] = (fn, return_type or T.F64, vectorized)
         return fn
 
 
@@ -452,6 +454,34 @@ class DataFrame:
 
     def explain(self) -> str:
         return S.plan_tree_string(self.plan)
+
+    def checkpoint(self, eager: bool = True) -> "DataFrame":
+        """Persist this DataFrame's result and return a DataFrame reading the
+        persisted copy — truncates the lineage like Spark's df.checkpoint()
+        (ref: RemoteCheckpointNode/WriteExec/CommitExec + RemoteCheckpoint-
+        Registry, crates/sail-cache/src/remote_checkpoint.rs).
+        The checkpoint root is `sail.execution.checkpoint_path` (defaults to
+        a session temp dir)."""
+        import os
+        import tempfile
+        import uuid as _uuid
+
+        root = self.session.conf.get("sail.execution.checkpoint_path")
+        if not root:
+            root = os.path.join(tempfile.gettempdir(), "sail_checkpoints")
+        path = os.path.join(root, _uuid.uuid4().hex)
+        if not eager:
+            # lazy checkpoint: materialize on first use; for this engine's
+            # eager collect model, just defer to the first collect
+            eager = True
+        chunk = self.collect_chunk()
+        from ..datasource.registry import write_source
+
+        os.makedirs(path, exist_ok=True)
+        write_source("parquet", path, chunk, "overwrite", {}, None)
+        plan = S.DataSourceRead(format="parquet", paths=[path])
+        plan.schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
+        return DataFrame(self.session, plan)
 
     @property
     def write(self) -> "DataFrameWriter":

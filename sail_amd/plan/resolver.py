@@ -193,6 +193,12 @@ class Resolver:
         out.schema = child.schema
         return out
 
+    def _p_Sample(self, p: S.Sample, outer):
+        child = self._plan(p.input, outer)
+        out = S.Sample(input=child, fraction=p.fraction, rows=p.rows, seed=p.seed)
+        out.schema = child.schema
+        return out
+
     def _p_Limit(self, p: S.Limit, outer):
         child = self._plan(p.input, outer)
         out = S.Limit(input=child, n=p.n, offset=p.offset)
@@ -981,7 +987,7 @@ def _scope_fields(p: S.Plan) -> List[Field]:
         return lf + rf
     if isinstance(p, S.SubqueryAlias):
         return [Field(n, t, p.alias) for n, t in p.schema]
-    if isinstance(p, (S.Filter, S.Limit, S.Sort, S.Distinct)):
+    if isinstance(p, (S.Filter, S.Limit, S.Sort, S.Distinct, S.Sample)):
         inner = _scope_fields(p.input)
         if len(inner) == len(p.schema):
             return inner
@@ -994,6 +1000,6 @@ def _plan_qualifier(p: S.Plan) -> Optional[str]:
         return p.alias
     if isinstance(p, S.Read):
         return p.table.split(".")[-1]
-    if isinstance(p, (S.Filter, S.Limit, S.Sort, S.Distinct)):
+    if isinstance(p, (S.Filter, S.Limit, S.Sort, S.Distinct, S.Sample)):
         return _plan_qualifier(p.input)
     return None

@@ -423,6 +423,21 @@ class Limit(Plan):
 
 
 @dataclass
+class Sample(Plan):
+    """TABLESAMPLE (n PERCENT | n ROWS) [REPEATABLE (seed)] — ref: Spark
+    sample grammar; spec::Sample (crates/sail-common/src/spec/plan.rs)."""
+
+    input: Plan = None
+    fraction: Optional[float] = None
+    rows: Optional[int] = None
+    seed: Optional[int] = None
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+    def children(self):
+        return [self.input]
+
+
+@dataclass
 class Distinct(Plan):
     input: Plan = None
     schema: Optional[List[Tuple[str, T.DataType]]] = None

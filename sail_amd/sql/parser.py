@@ -24,6 +24,7 @@ _RESERVED_STOP = {
     "ASC", "DESC", "NULLS", "BY", "WITH", "SELECT", "DISTINCT", "ALL",
     "SEMI", "ANTI", "NATURAL", "LATERAL", "WINDOW", "CLUSTER", "DISTRIBUTE",
     "SORT", "OVER", "ROWS", "RANGE", "PARTITION", "FOR", "CASE", "INTO",
+    "TABLESAMPLE",
 }
 
 _JOIN_TYPES = {
@@ -661,26 +662,52 @@ class Parser:
             self.expect_op("(")
             sub = self.parse_query()
             self.expect_op(")")
+            plan = self._maybe_sample(sub)
             alias, cols = self._parse_alias()
             if alias:
-                return S.SubqueryAlias(input=sub, alias=alias, column_aliases=cols)
-            return sub
+                return S.SubqueryAlias(input=plan, alias=alias, column_aliases=cols)
+            return plan
         name = self._qualified_name()
         if self.at_op("("):
             # table function: range(...), explode(...) etc.
             self.expect_op("(")
             args = [] if self.at_op(")") else self._expr_list()
             self.expect_op(")")
-            plan = self._table_function(name, args)
+            plan = self._maybe_sample(self._table_function(name, args))
             alias, cols = self._parse_alias()
             if alias:
                 return S.SubqueryAlias(input=plan, alias=alias, column_aliases=cols)
             return plan
-        plan = S.Read(table=name)
+        plan = self._maybe_sample(S.Read(table=name))
         alias, cols = self._parse_alias()
         if alias:
             return S.SubqueryAlias(input=plan, alias=alias, column_aliases=cols)
         return plan
+
+    def _maybe_sample(self, plan: S.Plan) -> S.Plan:
+        """TABLESAMPLE (n PERCENT | n ROWS) [REPEATABLE (seed)]"""
+        if not self.eat_kw("TABLESAMPLE"):
+            return plan
+        self.expect_op("(")
+        t = self.next()
+        if t.kind != "number":
+            raise SqlError("TABLESAMPLE expects a number")
+        val = float(t.value.split("#")[0])
+        fraction = rows = None
+        if self.eat_kw("PERCENT"):
+            fraction = val / 100.0
+        elif self.eat_kw("ROWS"):
+            rows = int(val)
+        else:
+            raise SqlError("TABLESAMPLE expects PERCENT or ROWS")
+        self.expect_op(")")
+        seed = None
+        if self.eat_kw("REPEATABLE"):
+            self.expect_op("(")
+            st = self.next()
+            seed = int(st.value.split("#")[0])
+            self.expect_op(")")
+        return S.Sample(input=plan, fraction=fraction, rows=rows, seed=seed)
 
     def _table_function(self, name: str, args: List[S.Expr]) -> S.Plan:
         lname = name.lower()

@@ -12,6 +12,8 @@ from ..engine.column import Column, StringColumn
 def _arrow_type(t: T.DataType) -> pa.DataType:
     if isinstance(t, T.DecimalType):
         return pa.decimal128(max(t.precision, t.scale + 1), t.scale)
+    if isinstance(t, T.ArrayType):
+        return pa.large_list(_arrow_type(t.element))
     m = {T.BooleanType: pa.bool_(), T.Int8Type: pa.int8(), T.Int16Type: pa.int16(),
          T.Int32Type: pa.int32(), T.Int64Type: pa.int64(), T.Float32Type: pa.float32(),
          T.Float64Type: pa.float64(), T.DateType: pa.date32(),
@@ -23,6 +25,19 @@ def _arrow_type(t: T.DataType) -> pa.DataType:
 def column_to_arrow(c: Column, t: T.DataType) -> pa.Array:
     import numpy as np
 
+    from ..engine.column import ListColumn
+
+    if isinstance(c, ListColumn):
+        child = column_to_arrow(c.child, c.child.dtype)
+        if isinstance(child, pa.DictionaryArray):
+            child = child.cast(pa.large_string())
+        offs = pa.array(c.offsets.cpu().numpy(), type=pa.int64())
+        out = pa.LargeListArray.from_arrays(offs, child)
+        if c.validity is not None:
+            mask = pa.array(c.valid_mask().cpu().numpy())
+            out = pa.LargeListArray.from_arrays(
+                offs, child, mask=pa.compute.invert(mask))
+        return out
     if isinstance(c, StringColumn):
         if c.is_dict:
             # keep dictionary encoding end-to-end (parquet dictionary pages)

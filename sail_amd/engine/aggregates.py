@@ -179,6 +179,33 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
             return _avg_result(data, cnt, c.dtype, out_type)
         return Column(out_type or T.I64, data, None)
 
+    if name in ("collect_list", "array_agg", "collect_set"):
+        from .column import ListColumn
+
+        m = mask & c.valid_mask()  # Spark: nulls are dropped from collections
+        gm = gid[m]
+        kept = torch.nonzero(m, as_tuple=False).flatten()
+        if name == "collect_set":
+            vk = normalize_key(c)[m]
+            o1 = torch.argsort(vk, stable=True)
+            o2 = torch.argsort(gm.index_select(0, o1), stable=True)
+            order = o1.index_select(0, o2)
+            gs = gm.index_select(0, order)
+            vs = vk.index_select(0, order)
+            first = torch.ones(gs.shape[0], dtype=torch.bool, device=dev)
+            if gs.shape[0] > 1:
+                first[1:] = (gs[1:] != gs[:-1]) | (vs[1:] != vs[:-1])
+            kept = kept.index_select(0, order)[first]
+            gm = gs[first]
+        else:
+            order = torch.argsort(gm, stable=True)
+            kept = kept.index_select(0, order)
+            gm = gm.index_select(0, order)
+        lens = torch.bincount(gm, minlength=ng)
+        offs = torch.zeros(ng + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(lens, 0, out=offs[1:])
+        return ListColumn(offs, c.gather(kept))
+
     gidm = gid[mask]
     n_used = int(mask.sum().item())
 

@@ -1,0 +1,337 @@
+"""Array (list) functions as segment ops over flat child columns.
+
+Spark's array function family (ref: crates/sail-plan/src/function/scalar/
+collection.rs names/semantics) implemented over ListColumn's
+offsets+child layout: per-row work becomes one vectorized pass over the
+flat child with `segment_ids()` as the reduction key — the same
+whole-partition shape every other operator in this engine uses, so these
+run unchanged on device. String-element ordering ops (sort_array/distinct
+over strings) and regex split run on host (documented deviation).
+"""
+from __future__ import annotations
+
+from typing import List
+
+import torch
+
+from . import types as T
+from .column import Column, ListColumn, StringColumn
+
+_BIG = (1 << 62)
+
+
+def _bcast(v, chunk):
+    from .eval import broadcast
+
+    return broadcast(v, chunk.num_rows, chunk.device)
+
+
+def _scalar_value(v):
+    from .eval import Scalar
+
+    return v.value if isinstance(v, Scalar) else None
+
+
+def _f_array(args, out, chunk, ev):
+    n, dev = chunk.num_rows, chunk.device
+    elem_t = out.element
+    from .eval import cast_column
+
+    cols = [cast_column(_bcast(a, chunk), elem_t) for a in args]
+    k = len(cols)
+    if k == 0:
+        offs = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+        return ListColumn(offs, Column.from_values([], elem_t, device=dev))
+    from .executor import concat_columns
+
+    allc = concat_columns(cols)  # column j occupies [j*n, (j+1)*n)
+    p = torch.arange(n * k, dtype=torch.int64, device=dev)
+    idx = (p % k) * n + (p // k)
+    child = allc.gather(idx)
+    offs = torch.arange(0, (n + 1) * k, k, dtype=torch.int64, device=dev)
+    return ListColumn(offs, child)
+
+
+def _f_size(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    data = (c.offsets[1:] - c.offsets[:-1]).to(torch.int32)
+    return Column(T.I32, data, c.validity)
+
+
+def _element_at(c: ListColumn, pos: torch.Tensor, one_based: bool):
+    """pos per row; 1-based supports negative-from-end (Spark element_at)."""
+    lens = c.lengths()
+    if one_based:
+        idx0 = torch.where(pos < 0, lens + pos, pos - 1)
+    else:
+        idx0 = pos
+    ok = (idx0 >= 0) & (idx0 < lens) & c.valid_mask()
+    safe = torch.where(ok, idx0, torch.zeros_like(idx0)) + c.offsets[:-1]
+    safe = safe.clamp(0, max(len(c.child) - 1, 0))
+    if len(c.child) == 0:
+        out = Column.from_values([None] * len(c), c.child.dtype, device=c.device)
+        return out
+    got = c.child.gather(safe)
+    valid = ok & got.valid_mask()
+    v = None if bool(valid.all()) else valid.to(torch.uint8)
+    if isinstance(got, StringColumn):
+        got.validity = v
+        return got
+    return Column(got.dtype, got.data, v)
+
+
+def _f_element_at(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    iv = _scalar_value(args[1])
+    if iv is not None:
+        pos = torch.full((len(c),), int(iv), dtype=torch.int64, device=c.device)
+    else:
+        pos = _bcast(args[1], chunk).data.to(torch.int64)
+    return _element_at(c, pos, one_based=True)
+
+
+def _f_get(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    iv = _scalar_value(args[1])
+    if iv is not None:
+        pos = torch.full((len(c),), int(iv), dtype=torch.int64, device=c.device)
+    else:
+        pos = _bcast(args[1], chunk).data.to(torch.int64)
+    return _element_at(c, pos, one_based=False)
+
+
+def _elem_match(c: ListColumn, needle) -> torch.Tensor:
+    """Per-child-row bool: element == needle (nulls never match)."""
+    child = c.child
+    if isinstance(child, StringColumn):
+        from .eval import Scalar
+
+        from .joins import normalize_key
+
+        key = normalize_key(child)
+        nk = normalize_key(StringColumn.from_pylist([needle], device=c.device))
+        m = key == nk[0]
+    else:
+        from .eval import Scalar, cast_value
+
+        sc = cast_value(Scalar(needle, child.dtype), child.dtype, None)
+        m = child.data == _scalar_tensor(sc.value, child, c.device)
+    return m & child.valid_mask()
+
+
+def _scalar_tensor(v, child: Column, dev):
+    if isinstance(child.dtype, T.DecimalType):
+        v = int(round(float(v) * 10 ** child.dtype.scale))
+    return torch.tensor(v, dtype=child.data.dtype, device=dev)
+
+
+def _f_array_contains(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    needle = _scalar_value(args[1])
+    m = _elem_match(c, needle).to(torch.int64)
+    seg = c.segment_ids()
+    acc = torch.zeros(len(c), dtype=torch.int64, device=c.device)
+    acc.index_add_(0, seg, m)
+    return Column(T.BOOL, acc > 0, c.validity)
+
+
+def _f_array_position(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    needle = _scalar_value(args[1])
+    m = _elem_match(c, needle)
+    seg = c.segment_ids()
+    pos_in_row = torch.arange(len(c.child), dtype=torch.int64, device=c.device) \
+        - torch.repeat_interleave(c.offsets[:-1], c.lengths())
+    cand = torch.where(m, pos_in_row + 1, torch.full_like(pos_in_row, _BIG))
+    first = torch.full((len(c),), _BIG, dtype=torch.int64, device=c.device)
+    first.scatter_reduce_(0, seg, cand, reduce="amin", include_self=True)
+    return Column(T.I64, torch.where(first == _BIG, torch.zeros_like(first), first),
+                  c.validity)
+
+
+def _seg_minmax(c: ListColumn, is_min: bool):
+    child = c.child
+    if isinstance(child, (StringColumn, ListColumn)):
+        raise NotImplementedError("array_min/max over non-primitive elements")
+    seg = c.segment_ids()
+    data = child.data
+    if data.dtype == torch.bool:
+        data = data.to(torch.int64)
+    if data.dtype.is_floating_point:
+        ext = torch.finfo(data.dtype).max
+    else:
+        ext = torch.iinfo(data.dtype).max
+    fill = data.new_full((), ext if is_min else -ext)
+    vals = torch.where(child.valid_mask(), data, fill.expand_as(data))
+    acc = fill.expand(len(c)).clone()
+    acc.scatter_reduce_(0, seg, vals, reduce="amin" if is_min else "amax",
+                        include_self=True)
+    cnt = torch.zeros(len(c), dtype=torch.int64, device=c.device)
+    cnt.index_add_(0, seg, child.valid_mask().to(torch.int64))
+    valid = (cnt > 0) & c.valid_mask()
+    v = None if bool(valid.all()) else valid.to(torch.uint8)
+    out_data = acc.to(child.data.dtype) if child.data.dtype != data.dtype else acc
+    return Column(child.dtype, out_data, v)
+
+
+def _f_array_min(args, out, chunk, ev):
+    return _seg_minmax(_bcast(args[0], chunk), True)
+
+
+def _f_array_max(args, out, chunk, ev):
+    return _seg_minmax(_bcast(args[0], chunk), False)
+
+
+def _f_sort_array(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    asc = True
+    if len(args) > 1:
+        asc = bool(_scalar_value(args[1]))
+    child = c.child
+    if isinstance(child, (StringColumn, ListColumn)):
+        # host fallback for non-primitive elements
+        vals = c.to_pylist()
+        return ListColumn.from_pylist(
+            [sorted(v, reverse=not asc) if v is not None else None for v in vals],
+            child.dtype, device=c.device)
+    seg = c.segment_ids()
+    # stable two-pass: order by value, then stable by segment
+    key = child.data.to(torch.float64) if child.data.dtype == torch.bool else child.data
+    o1 = torch.argsort(key, stable=True, descending=not asc)
+    o2 = torch.argsort(seg.index_select(0, o1), stable=True)
+    order = o1.index_select(0, o2)
+    return ListColumn(c.offsets, child.gather(order), c.validity, c.dtype)
+
+
+def _f_array_distinct(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    child = c.child
+    seg = c.segment_ids()
+    from .joins import normalize_key
+
+    key = normalize_key(child)
+    # first-occurrence order: sort by (seg, key, pos); drop adjacent dups;
+    # restore original position order among the kept
+    n = len(child)
+    pos = torch.arange(n, dtype=torch.int64, device=c.device)
+    o1 = torch.argsort(key, stable=True)
+    o2 = torch.argsort(seg.index_select(0, o1), stable=True)
+    order = o1.index_select(0, o2)  # sorted by (seg, key), pos-stable
+    sk = key.index_select(0, order)
+    ss = seg.index_select(0, order)
+    keep = torch.ones(n, dtype=torch.bool, device=c.device)
+    if n > 1:
+        keep[1:] = (sk[1:] != sk[:-1]) | (ss[1:] != ss[:-1])
+    kept = order[keep]
+    kept = kept.sort().values  # original order
+    new_seg = seg.index_select(0, kept)
+    lens = torch.bincount(new_seg, minlength=len(c))
+    offs = torch.zeros(len(c) + 1, dtype=torch.int64, device=c.device)
+    torch.cumsum(lens, 0, out=offs[1:])
+    return ListColumn(offs, child.gather(kept), c.validity, c.dtype)
+
+
+def _f_array_join(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    delim = _scalar_value(args[1])
+    vals = c.to_pylist()
+    joined = [None if v is None else delim.join(str(x) for x in v if x is not None)
+              for v in vals]
+    return StringColumn.from_pylist(joined, device=chunk.device)
+
+
+def _f_slice(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    lens = c.lengths()
+    start = _bcast(args[1], chunk).data.to(torch.int64)
+    length = _bcast(args[2], chunk).data.to(torch.int64)
+    s0 = torch.where(start > 0, start - 1, lens + start).clamp_min(0)
+    take = torch.minimum(length.clamp_min(0), (lens - s0).clamp_min(0))
+    new_off = torch.zeros(len(c) + 1, dtype=torch.int64, device=c.device)
+    torch.cumsum(take, 0, out=new_off[1:])
+    total = int(new_off[-1].item())
+    pos = torch.arange(total, dtype=torch.int64, device=c.device) \
+        - torch.repeat_interleave(new_off[:-1], take)
+    child_idx = torch.repeat_interleave(c.offsets[:-1] + s0, take) + pos
+    return ListColumn(new_off, c.child.gather(child_idx), c.validity, c.dtype)
+
+
+def _f_sequence(args, out, chunk, ev):
+    a = _bcast(args[0], chunk).data.to(torch.int64)
+    b = _bcast(args[1], chunk).data.to(torch.int64)
+    if len(args) > 2:
+        st = _bcast(args[2], chunk).data.to(torch.int64)
+    else:
+        st = torch.where(b >= a, torch.ones_like(a), -torch.ones_like(a))
+    lens = ((b - a) // st + 1).clamp_min(0)
+    n = a.shape[0]
+    offs = torch.zeros(n + 1, dtype=torch.int64, device=a.device)
+    torch.cumsum(lens, 0, out=offs[1:])
+    total = int(offs[-1].item())
+    pos = torch.arange(total, dtype=torch.int64, device=a.device) \
+        - torch.repeat_interleave(offs[:-1], lens)
+    child = torch.repeat_interleave(a, lens) + pos * torch.repeat_interleave(st, lens)
+    return ListColumn(offs, Column(T.I64, child))
+
+
+def _f_split(args, out, chunk, ev):
+    import re as _re
+
+    c = _bcast(args[0], chunk)
+    pat = _scalar_value(args[1])
+    rx = _re.compile(pat)
+    vals = c.to_pylist()
+    parts = [None if v is None else rx.split(v) for v in vals]
+    flat: List[str] = []
+    lens = []
+    for p in parts:
+        if p is None:
+            lens.append(0)
+        else:
+            lens.append(len(p))
+            flat.extend(p)
+    dev = chunk.device
+    offs = torch.zeros(len(parts) + 1, dtype=torch.int64, device=dev)
+    if parts:
+        torch.cumsum(torch.tensor(lens, dtype=torch.int64, device=dev), 0, out=offs[1:])
+    validity = None
+    if any(p is None for p in parts):
+        validity = torch.tensor([0 if p is None else 1 for p in parts],
+                                dtype=torch.uint8, device=dev)
+    child = StringColumn.from_pylist(flat, device=dev)
+    return ListColumn(offs, child, validity)
+
+
+def _f_arrays_overlap(args, out, chunk, ev):
+    a = _bcast(args[0], chunk)
+    b = _bcast(args[1], chunk)
+    from .joins import normalize_key
+
+    ka, kb = normalize_key(a.child), normalize_key(b.child)
+    sa, sb = a.segment_ids(), b.segment_ids()
+    # per row: any key of a present in b's keys for the same row — pack
+    # (row, key) and intersect via sorted search
+    pa = sa * 0x9E3779B97F4A7C15 + ka
+    pb = sb * 0x9E3779B97F4A7C15 + kb
+    sorted_b = torch.sort(pb).values
+    idx = torch.searchsorted(sorted_b, pa)
+    idx = idx.clamp_max(max(sorted_b.shape[0] - 1, 0))
+    hit = (sorted_b.shape[0] > 0) & (sorted_b.index_select(0, idx) == pa) \
+        & a.child.valid_mask()
+    acc = torch.zeros(len(a), dtype=torch.int64, device=a.device)
+    acc.index_add_(0, sa, hit.to(torch.int64))
+    valid = a.valid_mask() & b.valid_mask()
+    return Column(T.BOOL, acc > 0,
+                  None if bool(valid.all()) else valid.to(torch.uint8))
+
+
+IMPLS = {
+    "array": _f_array, "size": _f_size, "cardinality": _f_size,
+    "element_at": _f_element_at, "get": _f_get, "try_element_at": _f_element_at,
+    "array_contains": _f_array_contains, "array_position": _f_array_position,
+    "array_min": _f_array_min, "array_max": _f_array_max,
+    "sort_array": _f_sort_array, "array_sort": _f_sort_array,
+    "array_distinct": _f_array_distinct, "array_join": _f_array_join,
+    "slice": _f_slice, "sequence": _f_sequence, "split": _f_split,
+    "arrays_overlap": _f_arrays_overlap,
+}

@@ -380,3 +380,113 @@ class Table:
 
     def __repr__(self):
         return f"Table(rows={self.num_rows}, cols={self.names})"
+
+
+class ListColumn(Column):
+    """Arrow-style list column: int64 offsets (n+1) into a child element
+    column (fixed-width, string, or nested list). Array operators are
+    segment ops over the flat child — the layout GPU kernels and
+    repeat_interleave/cumsum both like (ref: Spark ArrayType semantics;
+    implementation is torch segment ops, not a port)."""
+
+    __slots__ = ("offsets", "child")
+
+    def __init__(self, offsets: torch.Tensor, child: Column,
+                 validity: Optional[torch.Tensor] = None,
+                 dtype: Optional[T.DataType] = None):
+        anchor = offsets[:-1] if offsets.numel() else offsets
+        super().__init__(dtype or T.ArrayType(child.dtype), anchor, validity)
+        self.offsets = offsets
+        self.child = child
+
+    @staticmethod
+    def from_pylist(values: List[Optional[list]], elem_type: T.DataType,
+                    device="cpu") -> "ListColumn":
+        device = _dev(device)
+        validity = None
+        if any(v is None for v in values):
+            validity = torch.tensor([0 if v is None else 1 for v in values],
+                                    dtype=torch.uint8, device=device)
+        lens = [0 if v is None else len(v) for v in values]
+        offsets = torch.zeros(len(values) + 1, dtype=torch.int64, device=device)
+        if values:
+            torch.cumsum(torch.tensor(lens, dtype=torch.int64, device=device),
+                         0, out=offsets[1:])
+        flat: List = []
+        for v in values:
+            if v is not None:
+                flat.extend(v)
+        child = Column.from_values(flat, elem_type, device=device)
+        return ListColumn(offsets, child, validity)
+
+    def __len__(self) -> int:
+        return int(self.offsets.shape[0]) - 1
+
+    @property
+    def device(self) -> torch.device:
+        return self.offsets.device
+
+    def lengths(self) -> torch.Tensor:
+        return self.offsets[1:] - self.offsets[:-1]
+
+    def segment_ids(self) -> torch.Tensor:
+        """child row -> parent row map."""
+        n = len(self)
+        return torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=self.device), self.lengths())
+
+    def to(self, device) -> "ListColumn":
+        device = _dev(device)
+        if self.device == device:
+            return self
+        return ListColumn(self.offsets.to(device), self.child.to(device),
+                          self.validity.to(device) if self.validity is not None else None,
+                          self.dtype)
+
+    def gather(self, indices: torch.Tensor) -> "ListColumn":
+        lens = self.lengths().index_select(0, indices)
+        new_off = torch.zeros(indices.shape[0] + 1, dtype=torch.int64, device=self.device)
+        torch.cumsum(lens, 0, out=new_off[1:])
+        total = int(new_off[-1].item()) if indices.numel() else 0
+        starts = self.offsets[:-1].index_select(0, indices)
+        # child index = start_of_selected_row + position inside that row
+        pos = torch.arange(total, dtype=torch.int64, device=self.device) \
+            - torch.repeat_interleave(new_off[:-1], lens)
+        child_idx = torch.repeat_interleave(starts, lens) + pos
+        v = self.validity.index_select(0, indices) if self.validity is not None else None
+        return ListColumn(new_off, self.child.gather(child_idx), v, self.dtype)
+
+    def filter(self, mask: torch.Tensor) -> "ListColumn":
+        return self.gather(torch.nonzero(mask, as_tuple=False).flatten())
+
+    def slice(self, start: int, length: int) -> "ListColumn":
+        off = self.offsets[start : start + length + 1]
+        base = int(off[0].item()) if off.numel() else 0
+        child = self.child.slice(base, int(off[-1].item()) - base if off.numel() else 0)
+        v = self.validity[start : start + length] if self.validity is not None else None
+        return ListColumn(off - base, child, v, self.dtype)
+
+    def null_count(self) -> int:
+        if self.validity is None:
+            return 0
+        return int(len(self) - int(self.validity.sum().item()))
+
+    def valid_mask(self) -> torch.Tensor:
+        if self.validity is None:
+            return torch.ones(len(self), dtype=torch.bool, device=self.device)
+        return self.validity.to(torch.bool)
+
+    def to_pylist(self) -> List:
+        flat = self.child.to_pylist()
+        offs = self.offsets.cpu().tolist()
+        vmask = self.validity.cpu().tolist() if self.validity is not None else None
+        out: List = []
+        for i in range(len(self)):
+            if vmask is not None and not vmask[i]:
+                out.append(None)
+            else:
+                out.append(flat[offs[i] : offs[i + 1]])
+        return out
+
+    def __repr__(self):
+        return f"ListColumn({self.dtype!r}, n={len(self)}, dev={self.device})"

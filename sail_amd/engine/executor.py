@@ -201,6 +201,46 @@ class Executor:
         gid, rep, ng = group_ids(child.columns)
         return child.gather(rep)
 
+    def _x_Generate(self, p: S.Generate) -> Chunk:
+        """explode/posexplode: repeat parent rows per array element
+        (repeat_interleave over the list offsets — one vectorized pass,
+        no per-row loop)."""
+        from .column import ListColumn
+
+        child = self.execute(p.input)
+        n, dev = child.num_rows, child.device
+        gen = broadcast(self.ev.eval(p.gen, child), n, dev)
+        if not isinstance(gen, ListColumn):
+            raise ExecError("generator input is not an array")
+        lens = gen.lengths()
+        valid = gen.valid_mask()
+        if p.outer:
+            eff = torch.where(valid & (lens > 0), lens, torch.ones_like(lens))
+            has_elem = valid & (lens > 0)
+        else:
+            eff = torch.where(valid, lens, torch.zeros_like(lens))
+            has_elem = None  # every emitted row has a real element
+        offs2 = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(eff, 0, out=offs2[1:])
+        total = int(offs2[-1].item())
+        parent = torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=dev), eff)
+        pos = torch.arange(total, dtype=torch.int64, device=dev) \
+            - torch.repeat_interleave(offs2[:-1], eff)
+        child_idx = gen.offsets[:-1].index_select(0, parent) + pos
+        child_idx = child_idx.clamp(0, max(len(gen.child) - 1, 0))
+        elem = gen.child.gather(child_idx) if len(gen.child) else \
+            Column.from_values([None] * total, gen.child.dtype, device=dev)
+        if has_elem is not None:
+            ev_mask = has_elem.index_select(0, parent) & elem.valid_mask()
+            elem.validity = None if bool(ev_mask.all()) else ev_mask.to(torch.uint8)
+        cols = [c.gather(parent) for c in child.columns]
+        if p.position:
+            cols.append(Column(T.I32, pos.to(torch.int32)))
+        cols.append(elem)
+        out = Chunk(cols, [nm for nm, _ in p.schema], child.partitioning)
+        return out
+
     def _x_Sample(self, p: S.Sample) -> Chunk:
         """Bernoulli sampling with a deterministic generator (REPEATABLE
         seed, or a fixed default so every SPMD rank draws the same mask for
@@ -800,6 +840,19 @@ def _null_sentinel(vals: torch.Tensor, lo: bool):
 
 def concat_columns(cols: List[Column]) -> Column:
     c0 = cols[0]
+    from .column import ListColumn
+
+    if isinstance(c0, ListColumn):
+        children = concat_columns([c.child for c in cols])
+        offs = [cols[0].offsets]
+        base = cols[0].offsets[-1]
+        for c in cols[1:]:
+            offs.append(c.offsets[1:] + base)
+            base = base + c.offsets[-1]
+        validity = None
+        if any(c.validity is not None for c in cols):
+            validity = torch.cat([c.valid_mask() for c in cols]).to(torch.uint8)
+        return ListColumn(torch.cat(offs), children, validity, c0.dtype)
     if isinstance(c0, StringColumn):
         vals = []
         for c in cols:

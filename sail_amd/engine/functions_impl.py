@@ -1040,3 +1040,9 @@ def _levenshtein(a: str, b: str) -> int:
             cur.append(min(prev[j] + 1, cur[j - 1] + 1, prev[j - 1] + (ca != cb)))
         prev = cur
     return prev[-1]
+
+
+# array/list functions (segment ops over ListColumn; engine/arrays.py)
+from .arrays import IMPLS as _ARRAY_IMPLS  # noqa: E402
+
+_IMPLS.update(_ARRAY_IMPLS)

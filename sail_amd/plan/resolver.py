@@ -176,6 +176,10 @@ class Resolver:
             out.dtype = e.dtype
             return out
 
+        gen_out = self._extract_generator(child, bound)
+        if gen_out is not None:
+            return gen_out
+
         bound2 = [extract_windows(e) for e in bound]
         if windows:
             wp = S.WindowPlan(input=child, window_exprs=windows)
@@ -185,6 +189,48 @@ class Resolver:
             return out
         out = S.Project(input=child, exprs=bound)
         out.schema = [(_expr_name(e, i), e.dtype) for i, e in enumerate(bound)]
+        return out
+
+    _GENERATORS = {"explode", "explode_outer", "posexplode", "posexplode_outer"}
+
+    def _extract_generator(self, child: S.Plan, bound: List[S.Expr]):
+        """SELECT explode(arr) [AS x], other... -> Project over Generate.
+        The generated element (and pos) columns are appended to the child
+        schema, so sibling select items keep their BoundRef ordinals."""
+        hits = []
+        for i, e in enumerate(bound):
+            inner = e.child if isinstance(e, S.Alias) else e
+            if isinstance(inner, S.Func) and inner.name.lower() in self._GENERATORS:
+                hits.append((i, e, inner))
+        if not hits:
+            return None
+        if len(hits) > 1:
+            raise ResolutionError("only one generator (explode/posexplode) "
+                                  "is allowed per SELECT list")
+        i, e, fn = hits[0]
+        if not isinstance(fn.args[0].dtype, T.ArrayType):
+            raise ResolutionError(f"{fn.name} expects an array argument")
+        elem_t = fn.args[0].dtype.element
+        lname = fn.name.lower()
+        position = lname.startswith("posexplode")
+        outer_gen = lname.endswith("_outer")
+        col_name = e.name if isinstance(e, S.Alias) else "col"
+        g = S.Generate(input=child, gen=fn.args[0], outer=outer_gen,
+                       position=position)
+        nin = len(child.schema)
+        g.schema = list(child.schema) \
+            + ([("pos", T.I32)] if position else []) + [(col_name, elem_t)]
+        new_exprs: List[S.Expr] = []
+        for j, b in enumerate(bound):
+            if j != i:
+                new_exprs.append(b)
+            elif position:
+                new_exprs.append(S.BoundRef(nin, "pos", T.I32))
+                new_exprs.append(S.BoundRef(nin + 1, col_name, elem_t))
+            else:
+                new_exprs.append(S.BoundRef(nin, col_name, elem_t))
+        out = S.Project(input=g, exprs=new_exprs)
+        out.schema = [(_expr_name(x, k), x.dtype) for k, x in enumerate(new_exprs)]
         return out
 
     def _p_Distinct(self, p: S.Distinct, outer):

@@ -423,6 +423,22 @@ class Limit(Plan):
 
 
 @dataclass
+class Generate(Plan):
+    """Generator (explode/posexplode): one output row per array element,
+    parent columns repeated — ref: Spark Generate; spec generator functions
+    (crates/sail-plan/src/function/generator.rs)."""
+
+    input: Plan = None
+    gen: Expr = None          # the array-valued expression
+    outer: bool = False       # explode_outer: keep empty/null arrays as null row
+    position: bool = False    # posexplode: emit 0-based pos column
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+    def children(self):
+        return [self.input]
+
+
+@dataclass
 class Sample(Plan):
     """TABLESAMPLE (n PERCENT | n ROWS) [REPEATABLE (seed)] — ref: Spark
     sample grammar; spec::Sample (crates/sail-common/src/spec/plan.rs)."""

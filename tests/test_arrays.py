@@ -1,0 +1,105 @@
+"""Array type, functions, collect aggregates, explode generators
+(ref: sail-plan function/scalar collection + generator.rs semantics)."""
+import pytest
+
+import sail_amd
+from sail_amd.engine import types as T
+
+
+@pytest.fixture()
+def s():
+    return sail_amd.SessionContext(device="cpu")
+
+
+def test_array_construction_and_size(s):
+    assert s.sql("SELECT array(1, 2, 3)").collect() == [([1, 2, 3],)]
+    assert s.sql("SELECT size(array(1,2,3)), cardinality(array())").collect()[0][0] == 3
+
+
+def test_element_at(s):
+    rows = s.sql("SELECT element_at(array(10,20,30), 2), element_at(array(10,20), -1), "
+                 "element_at(array(10), 5)").collect()
+    assert rows == [(20, 20, None)]
+
+
+def test_array_predicates(s):
+    rows = s.sql("SELECT array_contains(array(1,2),2), array_contains(array(1),9), "
+                 "array_position(array(5,6,7),7), array_position(array(5),9)").collect()
+    assert rows == [(True, False, 3, 0)]
+
+
+def test_array_minmax_sort_distinct(s):
+    rows = s.sql("SELECT array_min(array(3,1,2)), array_max(array(3,9)), "
+                 "sort_array(array(3,1,2)), sort_array(array(1,3), false), "
+                 "array_distinct(array(1,2,1,3,2))").collect()
+    assert rows == [(1, 9, [1, 2, 3], [3, 1], [1, 2, 3])]
+
+
+def test_array_join_slice_sequence(s):
+    rows = s.sql("SELECT array_join(array('a','b','c'), '-'), "
+                 "slice(array(1,2,3,4), 2, 2), slice(array(1,2,3,4), -2, 9), "
+                 "sequence(1, 4), sequence(4, 1)").collect()
+    assert rows == [("a-b-c", [2, 3], [3, 4], [1, 2, 3, 4], [4, 3, 2, 1])]
+
+
+def test_split_and_overlap(s):
+    rows = s.sql("SELECT split('a,b,c', ','), arrays_overlap(array(1,2), array(2,9)), "
+                 "arrays_overlap(array(1), array(9))").collect()
+    assert rows == [(["a", "b", "c"], True, False)]
+
+
+def test_array_column_ops(s):
+    s.create_dataframe({"v": [1, 2, 3]}, schema={"v": T.I64}, name="t")
+    rows = s.sql("SELECT v, size(sequence(1, v)) AS n FROM t ORDER BY v").collect()
+    assert rows == [(1, 1), (2, 2), (3, 3)]
+
+
+def test_collect_list_set(s):
+    s.create_dataframe({"g": ["a", "a", "b", "a"], "v": [1, 2, 3, 2]}, name="t")
+    rows = s.sql("SELECT g, collect_list(v), collect_set(v) FROM t GROUP BY g ORDER BY g").collect()
+    assert rows[0][0] == "a" and rows[0][1] == [1, 2, 2] and sorted(rows[0][2]) == [1, 2]
+    assert rows[1] == ("b", [3], [3])
+
+
+def test_collect_list_strings(s):
+    s.create_dataframe({"g": [1, 1, 2], "c": ["x", "y", "z"]}, name="t")
+    rows = s.sql("SELECT g, collect_list(c) FROM t GROUP BY g ORDER BY g").collect()
+    assert rows == [(1, ["x", "y"]), (2, ["z"])]
+
+
+def test_explode(s):
+    s.create_dataframe({"k": ["a", "b"], "v": [2, 3]}, name="t")
+    rows = s.sql("SELECT k, explode(sequence(1, v)) AS e FROM t ORDER BY k, e").collect()
+    assert rows == [("a", 1), ("a", 2), ("b", 1), ("b", 2), ("b", 3)]
+
+
+def test_posexplode(s):
+    rows = s.sql("SELECT posexplode(array(7, 8))").collect()
+    assert rows == [(0, 7), (1, 8)]
+
+
+def test_explode_outer_empty_and_null(s):
+    s.create_dataframe({"k": ["a", "b"], "v": [2, 0]}, name="t")
+    rows = s.sql("SELECT k, explode_outer(slice(sequence(1,3), 1, v)) AS e "
+                 "FROM t ORDER BY k, e NULLS LAST").collect()
+    assert rows == [("a", 1), ("a", 2), ("b", None)]
+    rows = s.sql("SELECT k, explode(slice(sequence(1,3), 1, v)) AS e "
+                 "FROM t ORDER BY k, e").collect()
+    assert rows == [("a", 1), ("a", 2)]  # inner explode drops empty
+
+
+def test_explode_then_aggregate(s):
+    s.create_dataframe({"k": ["a", "b"], "v": [3, 4]}, name="t")
+    rows = s.sql("WITH e AS (SELECT k, explode(sequence(1, v)) AS x FROM t) "
+                 "SELECT k, sum(x) FROM e GROUP BY k ORDER BY k").collect()
+    assert rows == [("a", 6), ("b", 10)]
+
+
+def test_array_to_arrow_roundtrip(s):
+    s.create_dataframe({"g": ["a", "a", "b"], "v": [1, 2, 3]}, name="t")
+    t = s.sql("SELECT g, collect_list(v) AS vs FROM t GROUP BY g ORDER BY g").to_arrow()
+    assert t.column("vs").to_pylist() == [[1, 2], [3]]
+
+
+def test_split_part_still_scalar(s):
+    assert s.sql("SELECT split_part('a:b:c', ':', 2)").collect() == [("b",)]

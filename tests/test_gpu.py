@@ -220,3 +220,57 @@ def test_update_delete_on_gpu():
     s.sql("UPDATE t SET v = v * 10 WHERE id > 1")
     s.sql("DELETE FROM t WHERE v >= 30.0")
     assert sorted(s.sql("SELECT * FROM t").collect()) == [(1, 1.0), (2, 20.0)]
+
+
+@pytest.mark.gpu
+def test_array_functions_on_gpu():
+    """Segment ops (gather/cumsum/scatter_reduce) over ListColumn on device."""
+    import sail_amd
+
+    s = sail_amd.SessionContext(device="cuda")
+    s.create_dataframe({"k": ["a", "b"], "v": [3, 5]}, name="t")
+    rows = s.sql("SELECT k, explode(sequence(1, v)) AS e FROM t ORDER BY k, e").collect()
+    assert rows == [("a", 1), ("a", 2), ("a", 3),
+                    ("b", 1), ("b", 2), ("b", 3), ("b", 4), ("b", 5)]
+    rows = s.sql("SELECT k, collect_list(x) FROM "
+                 "(SELECT k, explode(sequence(1, v)) AS x FROM t) e "
+                 "GROUP BY k ORDER BY k").collect()
+    assert rows == [("a", [1, 2, 3]), ("b", [1, 2, 3, 4, 5])]
+    rows = s.sql("SELECT array_min(sequence(v, 1)), array_position(sequence(1, v), 4), "
+                 "sort_array(array(3, 1, 2)) FROM t WHERE k = 'b'").collect()
+    assert rows == [(1, 4, [1, 2, 3])]
+
+
+@pytest.mark.gpu
+def test_streaming_incremental_agg_on_gpu():
+    """Micro-batch streaming with device-resident aggregation state."""
+    import sail_amd
+    from sail_amd.engine import types as T
+
+    s = sail_amd.SessionContext(device="cuda")
+    sdf = s.read_stream.format("memory").schema(
+        {"k": T.STRING, "v": T.I64}).load(name="ev")
+    src = sdf.source
+    q = (sdf.sql("SELECT k, sum(v) AS sv, count(*) AS n FROM ev GROUP BY k")
+         .write_stream.output_mode("complete").format("memory")
+         .query_name("agg_gpu").trigger(processing_time=0.01).start())
+    assert q._mode == "incremental"
+    src.add_rows({"k": ["a", "b", "a"], "v": [1, 2, 3]})
+    q.process_all_available()
+    src.add_rows({"k": ["b", "c"], "v": [10, 5]})
+    q.process_all_available()
+    q.stop()
+    assert q.exception is None
+    rows = dict((r[0], (r[1], r[2]))
+                for r in s.sql("SELECT k, sv, n FROM agg_gpu").collect())
+    assert rows == {"a": (4, 2), "b": (12, 2), "c": (5, 1)}
+
+
+@pytest.mark.gpu
+def test_tablesample_gpu_deterministic():
+    import sail_amd
+
+    s = sail_amd.SessionContext(device="cuda")
+    a = s.sql("SELECT sum(id) FROM range(100000) TABLESAMPLE (10 PERCENT) REPEATABLE (3) t").collect()
+    b = s.sql("SELECT sum(id) FROM range(100000) TABLESAMPLE (10 PERCENT) REPEATABLE (3) t").collect()
+    assert a == b and a[0][0] > 0

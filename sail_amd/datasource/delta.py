@@ -166,30 +166,61 @@ def read(paths: List[str], schema, device, options: Dict[str, str]):
     return parquet_io.read(full, tbl_schema, device, options or {})
 
 
-def write(path: str, chunk, mode: str, options: Dict[str, str], max_retries: int = 10):
-    """Append/overwrite commit with optimistic retry."""
+PART_ROWS = 4_000_000  # split big commits for parallel encode + scan
+
+
+def _write_parts(path: str, chunk, options) -> List[dict]:
+    """Encode the chunk as one or more part files, writing parts in parallel
+    (pyarrow releases the GIL during encode/compress/IO — a 60M-row commit
+    saturates multiple cores instead of one)."""
     import pyarrow.parquet as pq
 
+    from concurrent.futures import ThreadPoolExecutor
+
     from .arrow_io import chunk_to_arrow
+    from ..engine.chunk import Chunk as _Chunk
 
     schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
-    tbl = chunk_to_arrow(chunk, schema)
     os.makedirs(path, exist_ok=True)
-    part = f"part-{uuid.uuid4().hex}.parquet"
-    pq.write_table(tbl, os.path.join(path, part), compression=options.get("compression", "snappy") if options else "snappy")
-    add = {"add": {"path": part, "partitionValues": {}, "size": os.path.getsize(os.path.join(path, part)),
-                   "modificationTime": int(time.time() * 1000), "dataChange": True,
-                   "stats": json.dumps({"numRecords": tbl.num_rows})}}
+    compression = (options or {}).get("compression", "snappy")
+    n = chunk.num_rows
+    nparts = max(1, min(16, (n + PART_ROWS - 1) // PART_ROWS))
+    step = (n + nparts - 1) // nparts if nparts else n
+
+    def one(i):
+        lo = i * step
+        ln = min(step, n - lo)
+        sub = _Chunk([c.slice(lo, ln) for c in chunk.columns],
+                     list(chunk.names)) if nparts > 1 else chunk
+        tbl = chunk_to_arrow(sub, schema)
+        part = f"part-{i:05d}-{uuid.uuid4().hex}.parquet"
+        pq.write_table(tbl, os.path.join(path, part), compression=compression)
+        return {"add": {"path": part, "partitionValues": {},
+                        "size": os.path.getsize(os.path.join(path, part)),
+                        "modificationTime": int(time.time() * 1000),
+                        "dataChange": True,
+                        "stats": json.dumps({"numRecords": ln})}}
+
+    if nparts == 1:
+        return [one(0)]
+    with ThreadPoolExecutor(max_workers=min(nparts, 8)) as exe:
+        return list(exe.map(one, range(nparts)))
+
+
+def write(path: str, chunk, mode: str, options: Dict[str, str], max_retries: int = 10):
+    """Append/overwrite commit with optimistic retry."""
+    schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
+    adds = _write_parts(path, chunk, options)
     log = DeltaLog(path)
     for _ in range(max_retries):
         latest = log.latest_version()
         actions: List[dict] = []
         if latest is None:
-            actions = [_protocol_action(), _meta_action(schema), add]
+            actions = [_protocol_action(), _meta_action(schema)] + adds
             version = 0
         elif mode == "overwrite":
             old_schema, files, meta, _ = log.snapshot()
-            actions = [_meta_action(schema, meta.get("id")), add] + [
+            actions = [_meta_action(schema, meta.get("id"))] + adds + [
                 {"remove": {"path": f, "deletionTimestamp": int(time.time() * 1000),
                             "dataChange": True}} for f in files]
             version = latest + 1
@@ -198,7 +229,7 @@ def write(path: str, chunk, mode: str, options: Dict[str, str], max_retries: int
                 raise FileExistsError(path)
             if mode == "ignore" and latest is not None:
                 return
-            actions = [add]
+            actions = adds
             version = latest + 1
         else:
             raise ValueError(f"delta write mode {mode}")

@@ -854,16 +854,57 @@ def concat_columns(cols: List[Column]) -> Column:
             validity = torch.cat([c.valid_mask() for c in cols]).to(torch.uint8)
         return ListColumn(torch.cat(offs), children, validity, c0.dtype)
     if isinstance(c0, StringColumn):
-        vals = []
-        for c in cols:
-            vals.extend(c.to_pylist())
-        return StringColumn.from_pylist(vals, device=c0.device)
+        return _concat_strings(cols)
     data = torch.cat([c.data for c in cols])
     if any(c.validity is not None for c in cols):
         validity = torch.cat([c.valid_mask() for c in cols]).to(torch.uint8)
     else:
         validity = None
     return Column(c0.dtype, data, validity)
+
+
+def _concat_strings(cols: List[StringColumn]) -> StringColumn:
+    """Device-native string concat. All-dictionary inputs merge through a
+    sorted union dictionary (host work touches only the small dictionaries;
+    codes remap on device — keeps the engine's sorted-dict invariant);
+    otherwise raw offsets/bytes are chained on device. Replaces a
+    to_pylist/from_pylist host round-trip that dominated MERGE rewrites."""
+    dev = cols[0].device
+    total_dict = sum(c.offsets.shape[0] - 1 for c in cols if c.codes is not None)
+    if all(c.codes is not None for c in cols) and total_dict <= 200_000:
+        vals_list = [c.dict_values() for c in cols]
+        union = sorted(set().union(*vals_list))
+        pos = {s: i for i, s in enumerate(union)}
+        from .column import _pack_strings
+
+        offs, byts = _pack_strings(union, dev)
+        new_codes = []
+        for c, vals in zip(cols, vals_list):
+            m = torch.tensor([pos[v] for v in vals] or [0], dtype=torch.int32,
+                             device=dev)
+            cc = c.codes
+            remapped = m[cc.clamp_min(0).to(torch.int64)]
+            new_codes.append(torch.where(cc >= 0, remapped,
+                                         torch.full_like(cc, -1)))
+        validity = None
+        if any(c.validity is not None for c in cols):
+            validity = torch.cat([c.valid_mask() for c in cols]).to(torch.uint8)
+        return StringColumn(offs, byts, validity, torch.cat(new_codes),
+                            dtype=cols[0].dtype)
+    raws = [c.decode_dict() for c in cols]
+    offs_parts = []
+    bytes_parts = []
+    shift = 0
+    for r in raws:
+        offs_parts.append(r.offsets[:-1] + shift)
+        bytes_parts.append(r.bytes_)
+        shift += int(r.offsets[-1].item())
+    offs_parts.append(torch.tensor([shift], dtype=torch.int64, device=dev))
+    validity = None
+    if any(c.validity is not None for c in cols):
+        validity = torch.cat([c.valid_mask() for c in cols]).to(torch.uint8)
+    return StringColumn(torch.cat(offs_parts), torch.cat(bytes_parts),
+                        validity, None, dtype=cols[0].dtype)
 
 
 def join_chunks(ev: Evaluator, p: S.Join, left: Chunk, right: Chunk) -> Chunk:

@@ -193,6 +193,36 @@ class Resolver:
 
     _GENERATORS = {"explode", "explode_outer", "posexplode", "posexplode_outer"}
 
+    def _p_Generate(self, p: S.Generate, outer):
+        """LATERAL VIEW [OUTER] explode(e) v AS c1[, c2] (ref: Spark
+        LATERAL VIEW grammar; generator resolution in
+        crates/sail-plan/src/resolver/query/ lateral handling)."""
+        child = self._plan(p.input, outer)
+        scope = self._child_scope(child, outer)
+        fn = p.gen
+        if isinstance(fn, S.Func) and fn.name.lower() in self._GENERATORS:
+            arr = self._expr(fn.args[0], scope)
+        else:
+            arr = self._expr(fn, scope)
+        if not isinstance(arr.dtype, T.ArrayType):
+            raise ResolutionError("LATERAL VIEW generator expects an array")
+        elem_t = arr.dtype.element
+        aliases = p.aliases
+        if aliases is None:
+            aliases = ["pos", "col"] if p.position else ["col"]
+        want = 2 if p.position else 1
+        if len(aliases) != want:
+            raise ResolutionError(
+                f"LATERAL VIEW {'posexplode' if p.position else 'explode'} "
+                f"expects {want} column alias(es), got {len(aliases)}")
+        out = S.Generate(input=child, gen=arr, outer=p.outer,
+                         position=p.position, aliases=aliases,
+                         view_alias=p.view_alias)
+        gen_schema = ([(aliases[0], T.I32)] if p.position else []) \
+            + [(aliases[-1], elem_t)]
+        out.schema = list(child.schema) + gen_schema
+        return out
+
     def _extract_generator(self, child: S.Plan, bound: List[S.Expr]):
         """SELECT explode(arr) [AS x], other... -> Project over Generate.
         The generated element (and pos) columns are appended to the child
@@ -486,14 +516,52 @@ class Resolver:
         out = S.Join(left=left, right=right, how=p.how, on=on, using=using)
         if p.how in ("semi", "anti"):
             out.schema = list(left.schema)
-        elif p.how in ("rightsemi", "rightanti"):
+            return out
+        if p.how in ("rightsemi", "rightanti"):
             out.schema = list(right.schema)
-        elif using:
-            # USING join: shared columns appear once (from the left side)
+            return out
+        out.schema = list(left.schema) + list(right.schema)
+        if using:
+            # USING join: shared columns appear once. Normalized here to a
+            # Project over the full-concat join so every later phase (prune,
+            # reorder, executor) sees ONE coordinate system; key column comes
+            # from the non-null side (right for RIGHT joins, coalesce for
+            # FULL, left otherwise — Spark semantics).
             used = {c.lower() for c in using}
-            out.schema = list(left.schema) + [(n, t) for n, t in right.schema if n.lower() not in used]
-        else:
-            out.schema = list(left.schema) + list(right.schema)
+            nl = len(left.schema)
+            rpos = {n.lower(): j for j, (n, t) in enumerate(right.schema)}
+            exprs: List[S.Expr] = []
+            schema = []
+            for i, (n, t) in enumerate(left.schema):
+                if n.lower() in used:
+                    lref = S.BoundRef(i, n, t)
+                    rref = S.BoundRef(nl + rpos[n.lower()], n,
+                                      right.schema[rpos[n.lower()]][1])
+                    if p.how == "right":
+                        e = rref
+                    elif p.how == "full":
+                        e = S.Func("coalesce", [lref, rref], dtype=t)
+                    else:
+                        e = lref
+                    e = S.Alias(e, n, t) if not isinstance(e, S.BoundRef) else e
+                    exprs.append(e)
+                else:
+                    exprs.append(S.BoundRef(i, n, t))
+                schema.append((n, t))
+            for j, (n, t) in enumerate(right.schema):
+                if n.lower() not in used:
+                    exprs.append(S.BoundRef(nl + j, n, t))
+                    schema.append((n, t))
+            proj = S.Project(input=out, exprs=exprs)
+            proj.schema = schema
+            # preserve per-column qualifiers for parent join scopes
+            concat = lfields + rfields
+            ovr = []
+            for e, (n, t) in zip(exprs, schema):
+                q = concat[e.index].qualifier if isinstance(e, S.BoundRef) else None
+                ovr.append(Field(n, t, q))
+            proj.__dict__["_scope_fields_override"] = ovr
+            return proj
         return out
 
     def _p_SetOp(self, p: S.SetOp, outer):
@@ -1020,6 +1088,9 @@ def _expr_equal_unbound(a: S.Expr, b: S.Expr) -> bool:
 def _scope_fields(p: S.Plan) -> List[Field]:
     """Per-column qualifiers, preserved through nested join trees so that
     `n1.n_nationkey` resolves inside a 6-way comma join (q7/q8/q21)."""
+    ovr = p.__dict__.get("_scope_fields_override") if hasattr(p, "__dict__") else None
+    if ovr is not None:
+        return ovr
     if isinstance(p, S.Join):
         if p.how in ("semi", "anti"):
             return _scope_fields(p.left)
@@ -1037,6 +1108,13 @@ def _scope_fields(p: S.Plan) -> List[Field]:
         inner = _scope_fields(p.input)
         if len(inner) == len(p.schema):
             return inner
+    if isinstance(p, S.Generate):
+        inner = _scope_fields(p.input)
+        gen = [Field(n, t, p.view_alias)
+               for n, t in p.schema[len(p.input.schema):]]
+        if len(inner) == len(p.input.schema):
+            return inner + gen
+        return [Field(n, t, None) for n, t in p.schema[:len(p.input.schema)]] + gen
     q = _plan_qualifier(p)
     return [Field(n, t, q) for n, t in p.schema]
 

@@ -432,6 +432,9 @@ class Generate(Plan):
     gen: Expr = None          # the array-valued expression
     outer: bool = False       # explode_outer: keep empty/null arrays as null row
     position: bool = False    # posexplode: emit 0-based pos column
+    #: LATERAL VIEW form: generated-column names + view alias (unresolved)
+    aliases: Optional[List[str]] = None
+    view_alias: Optional[str] = None
     schema: Optional[List[Tuple[str, T.DataType]]] = None
 
     def children(self):

@@ -598,6 +598,27 @@ class Parser:
     def _parse_from(self) -> S.Plan:
         plan = self._parse_table_factor()
         while True:
+            if self.at_kw("LATERAL") and self.peek(1).upper == "VIEW":
+                self.next()
+                self.next()
+                outer = self.eat_kw("OUTER")
+                fn_name = self.ident().lower()
+                self.expect_op("(")
+                args = [] if self.at_op(")") else self._expr_list()
+                self.expect_op(")")
+                view_alias = self.ident() if self.peek().kind == "ident" \
+                    and self.peek().upper != "AS" else None
+                aliases = None
+                if self.eat_kw("AS"):
+                    aliases = [self.ident()]
+                    while self.eat_op(","):
+                        aliases.append(self.ident())
+                plan = S.Generate(
+                    input=plan, gen=S.Func(fn_name, args),
+                    outer=outer or fn_name.endswith("_outer"),
+                    position=fn_name.startswith("posexplode"),
+                    aliases=aliases, view_alias=view_alias)
+                continue
             if self.eat_op(","):
                 right = self._parse_table_factor()
                 plan = S.Join(left=plan, right=right, how="cross", on=None)

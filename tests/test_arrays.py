@@ -103,3 +103,33 @@ def test_array_to_arrow_roundtrip(s):
 
 def test_split_part_still_scalar(s):
     assert s.sql("SELECT split_part('a:b:c', ':', 2)").collect() == [("b",)]
+
+
+def test_lateral_view_explode(s):
+    s.create_dataframe({"k": ["a", "b"], "v": [2, 3]}, name="lt")
+    rows = s.sql("SELECT k, e FROM lt LATERAL VIEW explode(sequence(1, v)) ex AS e "
+                 "ORDER BY k, e").collect()
+    assert rows == [("a", 1), ("a", 2), ("b", 1), ("b", 2), ("b", 3)]
+
+
+def test_lateral_view_posexplode_qualified(s):
+    s.create_dataframe({"k": ["a"], "v": [2]}, name="lt2")
+    rows = s.sql("SELECT k, ex.p, ex.c FROM lt2 "
+                 "LATERAL VIEW posexplode(sequence(1, v)) ex AS p, c ORDER BY p").collect()
+    assert rows == [("a", 0, 1), ("a", 1, 2)]
+
+
+def test_lateral_view_outer(s):
+    s.create_dataframe({"k": ["a", "b"], "v": [1, 0]}, name="lt3")
+    rows = s.sql("SELECT k, e FROM lt3 "
+                 "LATERAL VIEW OUTER explode(slice(sequence(1,3), 1, v)) ex AS e "
+                 "ORDER BY k, e NULLS LAST").collect()
+    assert rows == [("a", 1), ("b", None)]
+
+
+def test_lateral_view_then_join(s):
+    s.create_dataframe({"k": ["a", "b"], "v": [2, 1]}, name="lt4")
+    s.create_dataframe({"e": [1, 2], "w": [10, 20]}, name="lt5")
+    rows = s.sql("SELECT k, e, w FROM lt4 LATERAL VIEW explode(sequence(1, v)) x AS e "
+                 "JOIN lt5 USING (e) ORDER BY k, e").collect()
+    assert rows == [("a", 1, 10), ("a", 2, 20), ("b", 1, 10)]

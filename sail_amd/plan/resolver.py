@@ -193,6 +193,90 @@ class Resolver:
 
     _GENERATORS = {"explode", "explode_outer", "posexplode", "posexplode_outer"}
 
+    def _p_Pivot(self, p: S.Pivot, outer):
+        """PIVOT -> grouped aggregate with one filtered agg per value
+        (ref: crates/sail-plan/src/resolver/query/pivoting.rs role)."""
+        child = self._plan(p.input, outer)
+        scope = self._child_scope(child, outer)
+        pivot = self._expr(p.pivot, scope)
+        agg_e = p.agg
+        alias_name = None
+        if isinstance(agg_e, S.Alias):
+            alias_name, agg_e = agg_e.name, agg_e.child
+        if isinstance(agg_e, S.Func):
+            agg_name, agg_args = agg_e.name.lower(), agg_e.args
+        elif isinstance(agg_e, S.AggFunc):
+            agg_name, agg_args = agg_e.name.lower(), agg_e.args
+        else:
+            raise ResolutionError("PIVOT expects an aggregate function")
+        bound_args = [self._expr(a, scope) for a in agg_args]
+        from .rules.util import expr_refs
+
+        used = set()
+        for e in [pivot] + bound_args:
+            used |= expr_refs(e)
+        group_refs = [S.BoundRef(i, n, t) for i, (n, t) in enumerate(child.schema)
+                      if i not in used]
+        aggs: List[S.Expr] = []
+        names: List[str] = []
+        for v in p.values:
+            vname = None
+            ve = v
+            if isinstance(v, S.Alias):
+                vname, ve = v.name, v.child
+            if not isinstance(ve, S.Literal):
+                raise ResolutionError("PIVOT IN list expects literals")
+            lit = self._expr(ve, scope)
+            cond = S.BinaryOp("=", pivot, lit, T.BOOL)
+            out_t = agg_return_type(agg_name, [a.dtype for a in bound_args])
+            af = S.AggFunc(agg_name, bound_args, False, out_t, cond)
+            aggs.append(af)
+            base = vname or str(ve.value)
+            names.append(base if alias_name is None else f"{base}_{alias_name}")
+        out = S.Aggregate(input=child, group_by=group_refs, aggs=aggs)
+        out.schema = [(r.name, r.dtype) for r in group_refs] \
+            + list(zip(names, [a.dtype for a in aggs]))
+        return out
+
+    def _p_Unpivot(self, p: S.Unpivot, outer):
+        """UNPIVOT -> UNION ALL of per-column projections + NOT NULL filter
+        (Spark default excludeNulls)."""
+        child = self._plan(p.input, outer)
+        byname = {n.lower(): (i, t) for i, (n, t) in enumerate(child.schema)}
+        for c in p.columns:
+            if c.lower() not in byname:
+                raise ResolutionError(f"UNPIVOT column not found: {c}")
+        vt = None
+        for c in p.columns:
+            vt = byname[c.lower()][1] if vt is None else T.common_type(vt, byname[c.lower()][1])
+        unpivot_set = {byname[c.lower()][0] for c in p.columns}
+        others = [(i, n, t) for i, (n, t) in enumerate(child.schema)
+                  if i not in unpivot_set]
+        parts = []
+        for c in p.columns:
+            ci, ct = byname[c.lower()]
+            exprs = [S.BoundRef(i, n, t) for i, n, t in others]
+            exprs.append(S.Alias(S.Literal(c, T.STRING), p.name_name, T.STRING))
+            vref: S.Expr = S.BoundRef(ci, c, ct)
+            if ct != vt:
+                vref = S.Cast(vref, vt, dtype=vt)
+            exprs.append(S.Alias(vref, p.value_name, vt))
+            pr = S.Project(input=child, exprs=exprs)
+            pr.schema = [(n, t) for _, n, t in others] \
+                + [(p.name_name, T.STRING), (p.value_name, vt)]
+            parts.append(pr)
+        plan = parts[0]
+        for nxt in parts[1:]:
+            so = S.SetOp(op="union", left=plan, right=nxt, is_all=True)
+            so.schema = parts[0].schema
+            plan = so
+        vi = len(plan.schema) - 1
+        cond = S.UnaryOp("isnotnull",
+                         S.BoundRef(vi, p.value_name, vt), dtype=T.BOOL)
+        out = S.Filter(input=plan, condition=cond)
+        out.schema = plan.schema
+        return out
+
     def _p_Generate(self, p: S.Generate, outer):
         """LATERAL VIEW [OUTER] explode(e) v AS c1[, c2] (ref: Spark
         LATERAL VIEW grammar; generator resolution in

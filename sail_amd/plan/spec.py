@@ -423,6 +423,37 @@ class Limit(Plan):
 
 
 @dataclass
+class Pivot(Plan):
+    """t PIVOT (agg(v) FOR k IN (v1, v2, ...)) — resolved into a grouped
+    aggregate with one filtered aggregate per pivot value (ref: Spark pivot;
+    sail-plan resolver/query/pivoting.rs role)."""
+
+    input: Plan = None
+    agg: Expr = None            # AggFunc over the value column (unresolved)
+    pivot: Expr = None          # the pivot column expression
+    values: List[Expr] = field(default_factory=list)  # literals
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+    def children(self):
+        return [self.input]
+
+
+@dataclass
+class Unpivot(Plan):
+    """t UNPIVOT (val FOR name IN (c1, c2, ...)) — resolved into a UNION ALL
+    of per-column projections."""
+
+    input: Plan = None
+    value_name: str = "value"
+    name_name: str = "name"
+    columns: List[str] = field(default_factory=list)
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+    def children(self):
+        return [self.input]
+
+
+@dataclass
 class Generate(Plan):
     """Generator (explode/posexplode): one output row per array element,
     parent columns repeated — ref: Spark Generate; spec generator functions

@@ -24,7 +24,7 @@ _RESERVED_STOP = {
     "ASC", "DESC", "NULLS", "BY", "WITH", "SELECT", "DISTINCT", "ALL",
     "SEMI", "ANTI", "NATURAL", "LATERAL", "WINDOW", "CLUSTER", "DISTRIBUTE",
     "SORT", "OVER", "ROWS", "RANGE", "PARTITION", "FOR", "CASE", "INTO",
-    "TABLESAMPLE",
+    "TABLESAMPLE", "PIVOT", "UNPIVOT",
 }
 
 _JOIN_TYPES = {
@@ -618,6 +618,44 @@ class Parser:
                     outer=outer or fn_name.endswith("_outer"),
                     position=fn_name.startswith("posexplode"),
                     aliases=aliases, view_alias=view_alias)
+                continue
+            if self.at_kw("PIVOT"):
+                self.next()
+                self.expect_op("(")
+                agg = self.parse_expr()
+                self.expect_kw("FOR")
+                pivot = self._parse_additive()  # stop before IN
+                self.expect_kw("IN")
+                self.expect_op("(")
+                values = []
+                while True:
+                    v = self.parse_expr()
+                    if self.eat_kw("AS"):
+                        v = S.Alias(v, self.ident())
+                    elif self.peek().kind == "ident" and self.peek().upper not in _RESERVED_STOP:
+                        v = S.Alias(v, self.ident())
+                    values.append(v)
+                    if not self.eat_op(","):
+                        break
+                self.expect_op(")")
+                self.expect_op(")")
+                plan = S.Pivot(input=plan, agg=agg, pivot=pivot, values=values)
+                continue
+            if self.at_kw("UNPIVOT"):
+                self.next()
+                self.expect_op("(")
+                value_name = self.ident()
+                self.expect_kw("FOR")
+                name_name = self.ident()
+                self.expect_kw("IN")
+                self.expect_op("(")
+                cols = [self.ident()]
+                while self.eat_op(","):
+                    cols.append(self.ident())
+                self.expect_op(")")
+                self.expect_op(")")
+                plan = S.Unpivot(input=plan, value_name=value_name,
+                                 name_name=name_name, columns=cols)
                 continue
             if self.eat_op(","):
                 right = self._parse_table_factor()

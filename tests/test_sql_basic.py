@@ -230,3 +230,43 @@ def test_rollup_cube_grouping_sets():
     rows = s2.sql("SELECT a, b, sum(v) FROM g GROUP BY GROUPING SETS ((a), (b)) "
                   "ORDER BY a NULLS LAST, b NULLS LAST").collect()
     assert rows == [("x", None, 3), ("y", None, 3), (None, "p", 4), (None, "q", 2)]
+
+
+def test_using_join_dedup_and_key_side(s):
+    s2 = sail_amd.SessionContext(device="cpu")
+    s2.create_dataframe({"a": [1, 2, 3], "x": ["p", "q", "r"]}, name="t")
+    s2.create_dataframe({"a": [2, 3, 4], "y": [20, 30, 40]}, name="u")
+    assert s2.sql("SELECT * FROM t JOIN u USING (a) ORDER BY a").collect() == [
+        (2, "q", 20), (3, "r", 30)]
+    # RIGHT: key column comes from the right side (4 present, not null)
+    assert s2.sql("SELECT * FROM t RIGHT JOIN u USING (a) ORDER BY a").collect() == [
+        (2, "q", 20), (3, "r", 30), (4, None, 40)]
+    # FULL: key column coalesced across both sides
+    assert s2.sql("SELECT * FROM t FULL JOIN u USING (a) ORDER BY a").collect() == [
+        (1, "p", None), (2, "q", 20), (3, "r", 30), (4, None, 40)]
+    # qualified refs still resolve through the dedup projection
+    assert s2.sql("SELECT u.y, t.x FROM t JOIN u USING (a) ORDER BY a").collect() == [
+        (20, "q"), (30, "r")]
+
+
+def test_pivot(s):
+    s2 = sail_amd.SessionContext(device="cpu")
+    s2.create_dataframe({"g": ["x", "x", "y", "y"], "k": ["a", "b", "a", "b"],
+                         "v": [1, 2, 3, 4]}, name="p")
+    rows = s2.sql("SELECT * FROM p PIVOT (sum(v) FOR k IN ('a', 'b')) ORDER BY g").collect()
+    assert rows == [("x", 1, 2), ("y", 3, 4)]
+    df = s2.sql("SELECT * FROM p PIVOT (sum(v) FOR k IN ('a' AS col_a, 'b')) ORDER BY g")
+    assert [n for n, _ in df.schema] == ["g", "col_a", "b"]
+    # missing combination -> null
+    s2.create_dataframe({"g": ["x", "y"], "k": ["a", "b"], "v": [1, 4]}, name="p2")
+    rows = s2.sql("SELECT * FROM p2 PIVOT (sum(v) FOR k IN ('a', 'b')) ORDER BY g").collect()
+    assert rows == [("x", 1, None), ("y", None, 4)]
+
+
+def test_unpivot(s):
+    s2 = sail_amd.SessionContext(device="cpu")
+    s2.create_dataframe({"id": [1, 2], "q1": [10, 30], "q2": [20, None]}, name="u")
+    rows = s2.sql("SELECT * FROM u UNPIVOT (sales FOR quarter IN (q1, q2)) "
+                  "ORDER BY id, quarter").collect()
+    # Spark default excludeNulls: (2, q2, None) is dropped
+    assert rows == [(1, "q1", 10), (1, "q2", 20), (2, "q1", 30)]

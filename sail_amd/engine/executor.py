@@ -201,6 +201,89 @@ class Executor:
         gid, rep, ng = group_ids(child.columns)
         return child.gather(rep)
 
+    def _x_RecursionRef(self, p: S.RecursionRef) -> Chunk:
+        frames = getattr(self.ctx, "recursion_frames", None)
+        if not frames or p.name not in frames:
+            raise ExecError(f"recursion reference {p.name} outside recursive CTE")
+        return frames[p.name]
+
+    def _x_RecursiveCte(self, p: S.RecursiveCte) -> Chunk:
+        """Fixpoint iteration: result += recursive(working_set) until empty
+        (ref: Spark recursive CTE semantics; sail-plan recursion.rs).
+        UNION (distinct) deduplicates each delta against everything seen —
+        required for cyclic-graph termination."""
+        result = self.execute(p.anchor)
+        if not hasattr(self.ctx, "recursion_frames"):
+            self.ctx.recursion_frames = {}
+        frames = self.ctx.recursion_frames
+        if not p.is_all:
+            result = self._dedup_chunk(result)
+        seen_keys = self._row_keys(result) if not p.is_all else None
+        work = result
+        names = [nm for nm, _ in p.schema]
+        it = 0
+        while work.num_rows > 0:
+            it += 1
+            if it > p.max_iter:
+                raise ExecError(
+                    f"recursive CTE {p.name}: exceeded {p.max_iter} iterations "
+                    "(set sail.execution.max_recursion to raise)")
+            frames[p.name] = work
+            self.ctx._cte_cache.clear()  # the body must re-execute per iteration
+            new = self.execute(p.recursive)
+            if not p.is_all and new.num_rows:
+                keys = self._row_keys(new)
+                srt = torch.sort(seen_keys).values
+                pos = torch.searchsorted(srt, keys).clamp_max(max(srt.shape[0] - 1, 0))
+                fresh_mask = ~((srt.shape[0] > 0) & (srt.index_select(0, pos) == keys))
+                idx = torch.nonzero(fresh_mask, as_tuple=False).flatten()
+                new = Chunk([c.gather(idx) for c in new.columns], list(new.names))
+                new = self._dedup_chunk(new)
+                if new.num_rows:
+                    seen_keys = torch.cat([seen_keys, self._row_keys(new)])
+            if new.num_rows == 0:
+                break
+            result = Chunk([concat_columns([a, b]) for a, b in
+                            zip(result.columns, new.columns)], names)
+            work = new
+        frames.pop(p.name, None)
+        result.names = names
+        return result
+
+    def _row_keys(self, chunk: Chunk) -> torch.Tensor:
+        """Cross-chunk-stable row keys (NOT _pack_or_hash, whose codes are
+        chunk-local): raw values for ints, FNV for strings, mix64-combined.
+        Hash-based — the engine-wide n²/2⁶⁴ collision tradeoff."""
+        from ..exec.distributed import _mix64
+        from .joins import fnv_key_tensor, normalize_key
+
+        if not chunk.columns:
+            return torch.zeros(chunk.num_rows, dtype=torch.int64)
+        acc = None
+        for c in chunk.columns:
+            if isinstance(c, StringColumn):
+                k = fnv_key_tensor(c.decode_dict())
+            else:
+                k = normalize_key(c)
+            # fold nulls to a distinct sentinel so NULL == NULL for dedup
+            if c.validity is not None:
+                k = torch.where(c.valid_mask(), k,
+                                torch.full_like(k, -(1 << 61) + 3))
+            acc = k if acc is None else _mix64(acc * 31 + k)
+        return acc
+
+    def _dedup_chunk(self, chunk: Chunk) -> Chunk:
+        if chunk.num_rows <= 1:
+            return chunk
+        keys = self._row_keys(chunk)
+        order = torch.argsort(keys, stable=True)
+        sk = keys.index_select(0, order)
+        first = torch.ones(sk.shape[0], dtype=torch.bool, device=sk.device)
+        first[1:] = sk[1:] != sk[:-1]
+        kept = order[first].sort().values
+        return Chunk([c.gather(kept) for c in chunk.columns], list(chunk.names),
+                     chunk.partitioning)
+
     def _x_Generate(self, p: S.Generate) -> Chunk:
         """explode/posexplode: repeat parent rows per array element
         (repeat_interleave over the list offsets — one vectorized pass,

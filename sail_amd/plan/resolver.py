@@ -55,6 +55,8 @@ class Resolver:
     def __init__(self, catalog):
         self.catalog = catalog  # engine/session.Catalog
         self.cte_scope: List[Dict[str, S.Plan]] = []
+        #: name -> schema while resolving a recursive CTE body
+        self.recursion_refs: Dict[str, list] = {}
 
     # =====================================================================
     def resolve(self, plan: S.Plan) -> S.Plan:
@@ -71,6 +73,11 @@ class Resolver:
         return Scope([Field(n, t, qualifier) for n, t in p.schema], outer)
 
     def _p_Read(self, p: S.Read, outer):
+        # working-set reference inside a recursive CTE body?
+        if p.table.lower() in self.recursion_refs:
+            ref = S.RecursionRef(name=p.table.lower())
+            ref.schema = list(self.recursion_refs[p.table.lower()])
+            return self._qualify(ref, p.table)
         # CTE reference?
         for scope in reversed(self.cte_scope):
             if p.table.lower() in scope:
@@ -127,11 +134,55 @@ class Resolver:
         self.cte_scope.append(scope)
         try:
             for name, sub in p.ctes:
-                resolved = self._plan(sub, outer)
+                if p.recursive and _refs_table(sub, name):
+                    resolved = self._resolve_recursive_cte(name, sub, outer)
+                else:
+                    resolved = self._plan(sub, outer)
                 scope[name.lower()] = resolved
             return self._plan(p.input, outer)
         finally:
             self.cte_scope.pop()
+
+    def _resolve_recursive_cte(self, name: str, sub: S.Plan, outer):
+        """WITH RECURSIVE name AS (anchor UNION [ALL] recursive)
+        (ref: crates/sail-plan/src/resolver/query/recursion.rs role)."""
+        wrapper_cols = None
+        body = sub
+        if isinstance(body, S.SubqueryAlias) and body.column_aliases:
+            wrapper_cols, body = body.column_aliases, body.input
+        if not (isinstance(body, S.SetOp) and body.op == "union"):
+            raise ResolutionError(
+                f"recursive CTE {name} must be 'anchor UNION [ALL] recursive'")
+        anchor = self._plan(body.left, outer)
+        if wrapper_cols:
+            if len(wrapper_cols) != len(anchor.schema):
+                raise ResolutionError(f"recursive CTE {name}: column count mismatch")
+            schema = [(wrapper_cols[i], t) for i, (_, t) in enumerate(anchor.schema)]
+        else:
+            schema = list(anchor.schema)
+        self.recursion_refs[name.lower()] = schema
+        try:
+            rec = self._plan(body.right, outer)
+        finally:
+            del self.recursion_refs[name.lower()]
+        if len(rec.schema) != len(schema):
+            raise ResolutionError(f"recursive CTE {name}: column count mismatch")
+        if any(rt != st for (_, rt), (_, st) in zip(rec.schema, schema)):
+            exprs = []
+            for i, ((rn, rt), (_, st)) in enumerate(zip(rec.schema, schema)):
+                ref = S.BoundRef(i, rn, rt)
+                exprs.append(S.Cast(ref, st, dtype=st) if rt != st else ref)
+            pr = S.Project(input=rec, exprs=exprs)
+            pr.schema = [(rn, st) for (rn, _), (_, st) in zip(rec.schema, schema)]
+            rec = pr
+        max_iter = 100
+        sess = getattr(self.catalog, "session", None)
+        if sess is not None:
+            max_iter = int(sess.conf.get("sail.execution.max_recursion", "100"))
+        out = S.RecursiveCte(name=name.lower(), anchor=anchor, recursive=rec,
+                             is_all=body.is_all, max_iter=max_iter)
+        out.schema = schema
+        return out
 
     def _p_SubqueryAlias(self, p: S.SubqueryAlias, outer):
         child = self._plan(p.input, outer)
@@ -1211,3 +1262,31 @@ def _plan_qualifier(p: S.Plan) -> Optional[str]:
     if isinstance(p, (S.Filter, S.Limit, S.Sort, S.Distinct, S.Sample)):
         return _plan_qualifier(p.input)
     return None
+
+
+def _refs_table(p, name: str) -> bool:
+    """Does the (unresolved) subtree read table `name`? (recursive-CTE detection)"""
+    lname = name.lower()
+    if isinstance(p, S.Read) and p.table.lower() == lname:
+        return True
+    for c in p.children():
+        if c is not None and _refs_table(c, lname):
+            return True
+    # subqueries inside expressions
+    for attr in ("exprs", "condition", "on"):
+        v = getattr(p, attr, None)
+        items = v if isinstance(v, list) else ([v] if v is not None else [])
+        for e in items:
+            if isinstance(e, S.Expr) and _expr_refs_table(e, lname):
+                return True
+    return False
+
+
+def _expr_refs_table(e, name: str) -> bool:
+    if isinstance(e, (S.ScalarSubquery, S.InSubquery, S.Exists)):
+        if e.plan is not None and _refs_table(e.plan, name):
+            return True
+    for c in e.children():
+        if isinstance(c, S.Expr) and _expr_refs_table(c, name):
+            return True
+    return False

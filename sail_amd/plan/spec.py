@@ -423,6 +423,32 @@ class Limit(Plan):
 
 
 @dataclass
+class RecursionRef(Plan):
+    """Reference to the working set inside a recursive CTE body
+    (ref: sail-plan resolver/query/recursion.rs; sail-logical-plan
+    RecursiveCTE nodes)."""
+
+    name: str = ""
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+
+@dataclass
+class RecursiveCte(Plan):
+    """WITH RECURSIVE name AS (anchor UNION [ALL] recursive): iterate the
+    recursive term to a fixpoint."""
+
+    name: str = ""
+    anchor: Plan = None
+    recursive: Plan = None
+    is_all: bool = True
+    max_iter: int = 100
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+    def children(self):
+        return [self.anchor, self.recursive]
+
+
+@dataclass
 class Pivot(Plan):
     """t PIVOT (agg(v) FOR k IN (v1, v2, ...)) — resolved into a grouped
     aggregate with one filtered aggregate per pivot value (ref: Spark pivot;

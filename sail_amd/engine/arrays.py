@@ -104,12 +104,11 @@ def _elem_match(c: ListColumn, needle) -> torch.Tensor:
     """Per-child-row bool: element == needle (nulls never match)."""
     child = c.child
     if isinstance(child, StringColumn):
-        from .eval import Scalar
+        from .joins import fnv_key_tensor
 
-        from .joins import normalize_key
-
-        key = normalize_key(child)
-        nk = normalize_key(StringColumn.from_pylist([needle], device=c.device))
+        key = fnv_key_tensor(child.decode_dict())
+        nk = fnv_key_tensor(StringColumn.from_pylist([needle], device=c.device,
+                                                     dict_encode=False))
         m = key == nk[0]
     else:
         from .eval import Scalar, cast_value
@@ -305,9 +304,14 @@ def _f_split(args, out, chunk, ev):
 def _f_arrays_overlap(args, out, chunk, ev):
     a = _bcast(args[0], chunk)
     b = _bcast(args[1], chunk)
-    from .joins import normalize_key
+    from .joins import fnv_key_tensor, normalize_key
 
-    ka, kb = normalize_key(a.child), normalize_key(b.child)
+    def _xkey(ch):
+        if isinstance(ch, StringColumn):
+            return fnv_key_tensor(ch.decode_dict())
+        return normalize_key(ch)
+
+    ka, kb = _xkey(a.child), _xkey(b.child)
     sa, sb = a.segment_ids(), b.segment_ids()
     # per row: any key of a present in b's keys for the same row — pack
     # (row, key) and intersect via sorted search
@@ -335,3 +339,139 @@ IMPLS = {
     "slice": _f_slice, "sequence": _f_sequence, "split": _f_split,
     "arrays_overlap": _f_arrays_overlap,
 }
+
+
+# ---------------------------------------------------------------------------
+# map functions (MapColumn: offsets + parallel keys/values children)
+# ---------------------------------------------------------------------------
+
+def _f_map(args, out, chunk, ev):
+    """map(k1, v1, k2, v2, ...)"""
+    from .column import MapColumn
+    from .eval import cast_column
+
+    n, dev = chunk.num_rows, chunk.device
+    kt, vt = out.key, out.value
+    ks = [cast_column(_bcast(args[i], chunk), kt) for i in range(0, len(args), 2)]
+    vs = [cast_column(_bcast(args[i], chunk), vt) for i in range(1, len(args), 2)]
+    m = len(ks)
+    from .executor import concat_columns
+
+    if m == 0:
+        offs = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+        return MapColumn(offs, Column.from_values([], kt, device=dev),
+                         Column.from_values([], vt, device=dev))
+    p = torch.arange(n * m, dtype=torch.int64, device=dev)
+    idx = (p % m) * n + (p // m)
+    keys = concat_columns(ks).gather(idx)
+    values = concat_columns(vs).gather(idx)
+    offs = torch.arange(0, (n + 1) * m, m, dtype=torch.int64, device=dev)
+    return MapColumn(offs, keys, values)
+
+
+def _f_map_from_arrays(args, out, chunk, ev):
+    from .column import MapColumn
+
+    ka = _bcast(args[0], chunk)
+    va = _bcast(args[1], chunk)
+    return MapColumn(ka.offsets, ka.child, va.child, ka.validity)
+
+
+def _f_map_keys(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    return c._as_lists()[0]
+
+
+def _f_map_values(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    return c._as_lists()[1]
+
+
+def _f_map_contains_key(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    m = _needle_match(c, args[1], chunk).to(torch.int64)
+    acc = torch.zeros(len(c), dtype=torch.int64, device=c.device)
+    acc.index_add_(0, c.segment_ids(), m)
+    return Column(T.BOOL, acc > 0, c.validity)
+
+
+def _needle_match(c, needle_arg, chunk) -> torch.Tensor:
+    """Per-entry bool: key[i] == needle(parent_of_i). Needle may be a
+    scalar or a per-row column; keys compared via stable encodings
+    (FNV for strings — codes are chunk-local and never comparable)."""
+    from .eval import Scalar
+    from .joins import fnv_key_tensor, normalize_key
+
+    child = c.keys
+    seg = c.segment_ids()
+    if isinstance(child, StringColumn):
+        ck = fnv_key_tensor(child.decode_dict())
+    else:
+        ck = normalize_key(child)
+    if isinstance(needle_arg, Scalar):
+        kl = c._as_lists()[0]
+        return _elem_match(kl, needle_arg.value)
+    nc = _bcast(needle_arg, chunk)
+    if isinstance(nc, StringColumn):
+        nk = fnv_key_tensor(nc.decode_dict())
+    else:
+        nk = normalize_key(nc)
+    m = ck == nk.index_select(0, seg)
+    return m & child.valid_mask() & nc.valid_mask().index_select(0, seg)
+
+
+def _f_map_element_at(args, out, chunk, ev):
+    """element_at(map, key) / map[key]: value for key, null if absent."""
+    c = _bcast(args[0], chunk)
+    m = _needle_match(c, args[1], chunk)
+    seg = c.segment_ids()
+    pos = torch.arange(len(c.keys), dtype=torch.int64, device=c.device)
+    cand = torch.where(m, pos, torch.full_like(pos, _BIG))
+    first = torch.full((len(c),), _BIG, dtype=torch.int64, device=c.device)
+    first.scatter_reduce_(0, seg, cand, reduce="amin", include_self=True)
+    found = first != _BIG
+    safe = torch.where(found, first, torch.zeros_like(first))
+    if len(c.values) == 0:
+        return Column.from_values([None] * len(c), c.values.dtype, device=c.device)
+    got = c.values.gather(safe.clamp(0, max(len(c.values) - 1, 0)))
+    valid = found & got.valid_mask() & c.valid_mask()
+    v = None if bool(valid.all()) else valid.to(torch.uint8)
+    from .column import StringColumn as _S
+
+    if isinstance(got, _S):
+        got.validity = v
+        return got
+    return Column(got.dtype, got.data, v)
+
+
+MAP_IMPLS = {
+    "map": _f_map, "map_from_arrays": _f_map_from_arrays,
+    "map_keys": _f_map_keys, "map_values": _f_map_values,
+    "map_contains_key": _f_map_contains_key,
+}
+IMPLS.update(MAP_IMPLS)
+
+
+def _f_element_at_dispatch(args, out, chunk, ev):
+    """element_at over arrays (1-based) or maps (by key)."""
+    from .column import MapColumn
+
+    c = _bcast(args[0], chunk)
+    if isinstance(c, MapColumn):
+        return _f_map_element_at(args, out, chunk, ev)
+    return _f_element_at(args, out, chunk, ev)
+
+
+def _f_subscript(args, out, chunk, ev):
+    """a[i]: 0-based for arrays (Spark), key lookup for maps."""
+    from .column import MapColumn
+
+    c = _bcast(args[0], chunk)
+    if isinstance(c, MapColumn):
+        return _f_map_element_at(args, out, chunk, ev)
+    return _f_get(args, out, chunk, ev)
+
+
+IMPLS["element_at"] = _f_element_at_dispatch
+IMPLS["try_element_at"] = _f_element_at_dispatch
+IMPLS["element_at_sql"] = _f_subscript

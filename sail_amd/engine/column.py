@@ -490,3 +490,112 @@ class ListColumn(Column):
 
     def __repr__(self):
         return f"ListColumn({self.dtype!r}, n={len(self)}, dev={self.device})"
+
+
+class MapColumn(Column):
+    """Map column: list-of-entries layout — int64 offsets (n+1) into parallel
+    keys/values child columns (Arrow map layout; ref: Spark MapType)."""
+
+    __slots__ = ("offsets", "keys", "values")
+
+    def __init__(self, offsets: torch.Tensor, keys: Column, values: Column,
+                 validity: Optional[torch.Tensor] = None,
+                 dtype: Optional[T.DataType] = None):
+        anchor = offsets[:-1] if offsets.numel() else offsets
+        super().__init__(dtype or T.MapType(keys.dtype, values.dtype),
+                         anchor, validity)
+        self.offsets = offsets
+        self.keys = keys
+        self.values = values
+
+    @staticmethod
+    def from_pylist(values: List[Optional[dict]], key_t: T.DataType,
+                    val_t: T.DataType, device="cpu") -> "MapColumn":
+        device = _dev(device)
+        validity = None
+        if any(v is None for v in values):
+            validity = torch.tensor([0 if v is None else 1 for v in values],
+                                    dtype=torch.uint8, device=device)
+        lens = [0 if v is None else len(v) for v in values]
+        offsets = torch.zeros(len(values) + 1, dtype=torch.int64, device=device)
+        if values:
+            torch.cumsum(torch.tensor(lens, dtype=torch.int64, device=device),
+                         0, out=offsets[1:])
+        fk: List = []
+        fv: List = []
+        for v in values:
+            if v is not None:
+                fk.extend(v.keys())
+                fv.extend(v.values())
+        return MapColumn(offsets, Column.from_values(fk, key_t, device=device),
+                         Column.from_values(fv, val_t, device=device), validity)
+
+    def __len__(self) -> int:
+        return int(self.offsets.shape[0]) - 1
+
+    @property
+    def device(self) -> torch.device:
+        return self.offsets.device
+
+    def lengths(self) -> torch.Tensor:
+        return self.offsets[1:] - self.offsets[:-1]
+
+    def segment_ids(self) -> torch.Tensor:
+        n = len(self)
+        return torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=self.device), self.lengths())
+
+    def _as_lists(self):
+        """View keys/values as ListColumns sharing this map's offsets."""
+        return (ListColumn(self.offsets, self.keys, self.validity),
+                ListColumn(self.offsets, self.values, self.validity))
+
+    def to(self, device) -> "MapColumn":
+        device = _dev(device)
+        if self.device == device:
+            return self
+        return MapColumn(self.offsets.to(device), self.keys.to(device),
+                         self.values.to(device),
+                         self.validity.to(device) if self.validity is not None else None,
+                         self.dtype)
+
+    def gather(self, indices: torch.Tensor) -> "MapColumn":
+        kl, vl = self._as_lists()
+        gk = kl.gather(indices)
+        gv = vl.gather(indices)
+        return MapColumn(gk.offsets, gk.child, gv.child, gk.validity, self.dtype)
+
+    def filter(self, mask: torch.Tensor) -> "MapColumn":
+        return self.gather(torch.nonzero(mask, as_tuple=False).flatten())
+
+    def slice(self, start: int, length: int) -> "MapColumn":
+        kl, vl = self._as_lists()
+        sk = kl.slice(start, length)
+        sv = vl.slice(start, length)
+        return MapColumn(sk.offsets, sk.child, sv.child, sk.validity, self.dtype)
+
+    def null_count(self) -> int:
+        if self.validity is None:
+            return 0
+        return int(len(self) - int(self.validity.sum().item()))
+
+    def valid_mask(self) -> torch.Tensor:
+        if self.validity is None:
+            return torch.ones(len(self), dtype=torch.bool, device=self.device)
+        return self.validity.to(torch.bool)
+
+    def to_pylist(self) -> List:
+        ks = self.keys.to_pylist()
+        vs = self.values.to_pylist()
+        offs = self.offsets.cpu().tolist()
+        vmask = self.validity.cpu().tolist() if self.validity is not None else None
+        out: List = []
+        for i in range(len(self)):
+            if vmask is not None and not vmask[i]:
+                out.append(None)
+            else:
+                out.append(dict(zip(ks[offs[i]:offs[i + 1]], vs[offs[i]:offs[i + 1]])))
+        return out
+
+    def __repr__(self):
+        return f"MapColumn({self.dtype!r}, n={len(self)}, dev={self.device})"

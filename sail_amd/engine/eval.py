@@ -392,8 +392,9 @@ class Evaluator:
         args = [self.eval(a, chunk) for a in e.args]
         if args and all(isinstance(a, Scalar) for a in args) \
                 and e.name not in ("rand", "randn", "uuid", "monotonically_increasing_id") \
-                and not isinstance(e.dtype, T.ArrayType) \
-                and not any(isinstance(a.dtype, T.ArrayType) for a in args):
+                and not isinstance(e.dtype, (T.ArrayType, T.MapType, T.StructType)) \
+                and not any(isinstance(a.dtype, (T.ArrayType, T.MapType, T.StructType))
+                            for a in args):
             # constant folding: evaluate once on a 1-row chunk
             one = Chunk([], [], chunk.partitioning)
             one.forced_rows = 1

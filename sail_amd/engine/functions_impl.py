@@ -1046,3 +1046,6 @@ def _levenshtein(a: str, b: str) -> int:
 from .arrays import IMPLS as _ARRAY_IMPLS  # noqa: E402
 
 _IMPLS.update(_ARRAY_IMPLS)
+_IMPLS["element_at"] = _ARRAY_IMPLS["element_at"]
+_IMPLS["element_at_sql"] = _ARRAY_IMPLS["element_at_sql"]
+_IMPLS["try_element_at"] = _ARRAY_IMPLS["try_element_at"]

@@ -194,7 +194,20 @@ _reg("array", lambda a: T.ArrayType(a[0] if a else T.NULL))
 _reg("array_contains", _bool)
 _reg("size cardinality array_size", _i32)
 _reg("array_max array_min", lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
-_reg("element_at element_at_sql get", lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
+def _elem_rule(a):
+    if isinstance(a[0], T.ArrayType):
+        return a[0].element
+    if isinstance(a[0], T.MapType):
+        return a[0].value
+    return T.NULL
+
+
+_reg("element_at element_at_sql get try_element_at", _elem_rule)
+_reg("map", lambda a: T.MapType(a[0] if a else T.STRING, a[1] if len(a) > 1 else T.STRING))
+_reg("map_from_arrays", lambda a: T.MapType(a[0].element, a[1].element))
+_reg("map_keys", lambda a: T.ArrayType(a[0].key) if isinstance(a[0], T.MapType) else T.NULL)
+_reg("map_values", lambda a: T.ArrayType(a[0].value) if isinstance(a[0], T.MapType) else T.NULL)
+_reg("map_contains_key", _bool)
 _reg("sort_array array_sort array_distinct array_remove array_compact flatten "
      "slice array_repeat shuffle", _same)
 _reg("array_join", _string)

@@ -248,6 +248,51 @@ def _reorder_tree(root: S.Join, stats) -> S.Plan:
                 d = min(d * nd, 1e15)
         return min(d, max(ests[j][1], cur_est, 1.0))
 
+    def step_cost(i: int, placed: Set[int], cur_est: float, cur_sel: float) -> float:
+        """Estimated rows after joining leaf i onto the placed set — shared
+        by the greedy and the DP search so both optimize one model."""
+        est_i, base_i = ests[i]
+        if any(j in placed for j in edges[i]):
+            formula = cur_est * est_i / max(join_denom(i, placed, cur_est), 1.0)
+            return max(formula, est_i * cur_sel)
+        return cur_est * est_i * (1.0 if est_i <= 2 else 8.0)
+
+    def run_dp():
+        """Exact DP over left-deep orders (all subsets, best-total state per
+        subset) — the DP counterpart of the reference's join_reorder DP
+        (ref: crates/sail-physical-optimizer/src/join_reorder/mod.rs:31).
+        O(2^n * n^2); used for n <= 10, multi-seed greedy beyond."""
+        dp: Dict[int, Tuple[float, float, float, Tuple[int, ...]]] = {}
+        for i in range(n):
+            sel = min(1.0, ests[i][0] / max(ests[i][1], 1.0))
+            dp[1 << i] = (ests[i][0], ests[i][0], sel, (i,))
+        full = (1 << n) - 1
+        masks = sorted(dp.keys())
+        by_pop: Dict[int, List[int]] = {1: masks}
+        for pop in range(1, n):
+            nxt: Dict[int, Tuple[float, float, float, Tuple[int, ...]]] = {}
+            for mask in by_pop[pop]:
+                total, cur_est, cur_sel, order = dp[mask]
+                placed = {order[k] for k in range(len(order))}
+                for i in range(n):
+                    bit = 1 << i
+                    if mask & bit:
+                        continue
+                    c = max(step_cost(i, placed, cur_est, cur_sel), 1.0)
+                    est_b, base_b = ests[i]
+                    nsel = min(cur_sel, min(1.0, est_b / max(base_b, 1.0)))
+                    nt = total + c
+                    m2 = mask | bit
+                    old = nxt.get(m2) or dp.get(m2)
+                    if old is None or nt < old[0]:
+                        nxt[m2] = (nt, c, nsel, order + (i,))
+            dp.update(nxt)
+            by_pop[pop + 1] = [m for m in nxt]
+        if full not in dp:
+            return None, None
+        total, _, _, order = dp[full]
+        return list(order), total
+
     # greedy from EVERY seed, objective = sum of intermediate sizes; a single
     # smallest-leaf seed can dead-end into a low-ndv edge (q5: region ->
     # nation -> supplier forces customer x supplier on nationkey)
@@ -289,6 +334,10 @@ def _reorder_tree(root: S.Join, stats) -> S.Plan:
         order_s, total_s = run_greedy(seed)
         if best_total is None or total_s < best_total:
             best_order, best_total = order_s, total_s
+    if n <= 10:
+        order_dp, total_dp = run_dp()
+        if order_dp is not None and total_dp < best_total:
+            best_order, best_total = order_dp, total_dp
     order = best_order
 
     # rebuild left-deep tree in `order`, remapping conjunct indices

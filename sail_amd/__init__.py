@@ -5,6 +5,8 @@ Public surface:
     sail_amd.connect(...)    — Spark Connect server entry (connect/)
 """
 from .engine.session import Catalog, DataFrame, SessionContext
+from .errors import (AnalysisException, ExecutionException, ParseException,
+                     SailError, UnsupportedOperationException)
 
 __version__ = "0.1.0"
 
@@ -17,4 +19,7 @@ def connect_server(host="127.0.0.1", port=0, device=None):
     return SparkConnectServer(host=host, port=port, device=device).start()
 
 
-__all__ = ["SessionContext", "DataFrame", "Catalog", "connect_server", "__version__"]
+__all__ = ["SessionContext", "DataFrame", "Catalog", "connect_server",
+           "SailError", "ParseException", "AnalysisException",
+           "ExecutionException", "UnsupportedOperationException",
+           "__version__"]

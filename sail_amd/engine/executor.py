@@ -24,7 +24,10 @@ from .eval import Evaluator, Scalar, broadcast, cast_column, cast_value
 from .joins import equi_join
 
 
-class ExecError(Exception):
+from ..errors import ExecutionException
+
+
+class ExecError(ExecutionException):
     pass
 
 
@@ -835,6 +838,20 @@ class Executor:
         return execute_delete(self, p)
 
     def _x_Explain(self, p: S.Explain) -> Chunk:
+        if p.mode == "analyze":
+            # EXPLAIN ANALYZE: execute under the tracer, report per-operator
+            # wall time / rows (TracingExec analogue — ref:
+            # crates/sail-telemetry/src/execution/physical_plan.rs:54)
+            from ..utils.trace import Tracer
+
+            sub = ExecutionContext(self.ctx.session, self.ctx.device)
+            sub.tracer = Tracer(self.ctx.device)
+            out = Executor(sub).execute(p.input)
+            text = (S.plan_tree_string(p.input).rstrip()
+                    + "\n\n== Analyzed (wall times) ==\n"
+                    + sub.tracer.trace.render()
+                    + f"\n\nresult rows: {out.num_rows}")
+            return Chunk([StringColumn.from_pylist([text])], ["plan"])
         text = S.plan_tree_string(p.input)
         return Chunk([StringColumn.from_pylist([text])], ["plan"])
 

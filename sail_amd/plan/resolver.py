@@ -20,7 +20,10 @@ from ..functions.registry import AGG_FUNCTIONS, WINDOW_FUNCTIONS, agg_return_typ
 from . import spec as S
 
 
-class ResolutionError(Exception):
+from ..errors import AnalysisException
+
+
+class ResolutionError(AnalysisException):
     pass
 
 

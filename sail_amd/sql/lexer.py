@@ -22,7 +22,10 @@ class Token:
             self.upper = self.value.upper()
 
 
-class SqlError(Exception):
+from ..errors import ParseException
+
+
+class SqlError(ParseException):
     def __init__(self, msg: str, sql: str = "", pos: int = -1):
         if pos >= 0 and sql:
             line = sql.count("\n", 0, pos) + 1

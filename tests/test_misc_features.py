@@ -70,3 +70,25 @@ def test_scalar_udf_still_works(s):
     s.udf.register("sq", lambda x: x * x, "bigint")
     s.create_dataframe({"v": [2, 3]}, schema={"v": T.I64}, name="u3")
     assert s.sql("SELECT sq(v) FROM u3 ORDER BY 1").collect() == [(4,), (9,)]
+
+
+def test_structured_errors(s):
+    import sail_amd as sa
+
+    with pytest.raises(sa.ParseException) as ei:
+        s.sql("SELEC 1")
+    assert ei.value.sql_state == "42601"
+    with pytest.raises(sa.AnalysisException):
+        s.sql("SELECT * FROM does_not_exist")
+    with pytest.raises(sa.AnalysisException):
+        s.sql("SELECT no_such_fn_xyz(1)")
+    # all are SailError
+    with pytest.raises(sa.SailError):
+        s.sql("SELEC 1")
+
+
+def test_explain_analyze(s):
+    s.create_dataframe({"a": [1, 2, 3]}, name="ea")
+    out = s.sql("EXPLAIN ANALYZE SELECT a, count(*) FROM ea GROUP BY a").collect()[0][0]
+    assert "Analyzed (wall times)" in out
+    assert "Aggregate" in out and "ms self" in out

@@ -475,3 +475,44 @@ def _f_subscript(args, out, chunk, ev):
 IMPLS["element_at"] = _f_element_at_dispatch
 IMPLS["try_element_at"] = _f_element_at_dispatch
 IMPLS["element_at_sql"] = _f_subscript
+
+
+# ---------------------------------------------------------------------------
+# struct functions (StructColumn: parallel named children)
+# ---------------------------------------------------------------------------
+
+def _f_struct(args, out, chunk, ev):
+    from .column import StructColumn
+
+    kids = []
+    for f, a in zip(out.fields, args):
+        kids.append((f.name, _bcast(a, chunk)))
+    return StructColumn(kids, dtype=out)
+
+
+def _f_named_struct(args, out, chunk, ev):
+    from .column import StructColumn
+
+    kids = []
+    for i, f in enumerate(out.fields):
+        kids.append((f.name, _bcast(args[2 * i + 1], chunk)))
+    return StructColumn(kids, dtype=out)
+
+
+def _f_get_field(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    name = _scalar_value(args[1])
+    got = c.field(name)
+    if c.validity is not None:
+        valid = c.valid_mask() & got.valid_mask()
+        v = None if bool(valid.all()) else valid.to(torch.uint8)
+        if isinstance(got, (StringColumn, ListColumn)):
+            got.validity = v
+            return got
+        return Column(got.dtype, got.data, v)
+    return got
+
+
+IMPLS["struct"] = _f_struct
+IMPLS["named_struct"] = _f_named_struct
+IMPLS["get_field"] = _f_get_field

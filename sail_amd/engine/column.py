@@ -599,3 +599,77 @@ class MapColumn(Column):
 
     def __repr__(self):
         return f"MapColumn({self.dtype!r}, n={len(self)}, dev={self.device})"
+
+
+class StructColumn(Column):
+    """Struct column: parallel named children (Arrow struct layout)."""
+
+    __slots__ = ("children_",)
+
+    def __init__(self, names_children, validity: Optional[torch.Tensor] = None,
+                 dtype: Optional[T.DataType] = None):
+        self.children_ = list(names_children)  # [(name, Column)]
+        n = len(self.children_[0][1]) if self.children_ else 0
+        anchor = (self.children_[0][1].data if self.children_
+                  else torch.zeros(0, dtype=torch.int64))
+        st = dtype or T.StructType(tuple(
+            T.StructField(nm, c.dtype) for nm, c in self.children_))
+        super().__init__(st, anchor, validity)
+
+    def __len__(self) -> int:
+        return len(self.children_[0][1]) if self.children_ else 0
+
+    @property
+    def device(self) -> torch.device:
+        return self.children_[0][1].device if self.children_ else torch.device("cpu")
+
+    def field(self, name: str) -> Column:
+        for nm, c in self.children_:
+            if nm.lower() == name.lower():
+                return c
+        raise KeyError(name)
+
+    def to(self, device) -> "StructColumn":
+        device = _dev(device)
+        if self.device == device:
+            return self
+        return StructColumn([(nm, c.to(device)) for nm, c in self.children_],
+                            self.validity.to(device) if self.validity is not None else None,
+                            self.dtype)
+
+    def gather(self, indices: torch.Tensor) -> "StructColumn":
+        v = self.validity.index_select(0, indices) if self.validity is not None else None
+        return StructColumn([(nm, c.gather(indices)) for nm, c in self.children_],
+                            v, self.dtype)
+
+    def filter(self, mask: torch.Tensor) -> "StructColumn":
+        return self.gather(torch.nonzero(mask, as_tuple=False).flatten())
+
+    def slice(self, start: int, length: int) -> "StructColumn":
+        v = self.validity[start:start + length] if self.validity is not None else None
+        return StructColumn([(nm, c.slice(start, length)) for nm, c in self.children_],
+                            v, self.dtype)
+
+    def null_count(self) -> int:
+        if self.validity is None:
+            return 0
+        return int(len(self) - int(self.validity.sum().item()))
+
+    def valid_mask(self) -> torch.Tensor:
+        if self.validity is None:
+            return torch.ones(len(self), dtype=torch.bool, device=self.device)
+        return self.validity.to(torch.bool)
+
+    def to_pylist(self) -> List:
+        lists = [(nm, c.to_pylist()) for nm, c in self.children_]
+        vmask = self.validity.cpu().tolist() if self.validity is not None else None
+        out: List = []
+        for i in range(len(self)):
+            if vmask is not None and not vmask[i]:
+                out.append(None)
+            else:
+                out.append({nm: vals[i] for nm, vals in lists})
+        return out
+
+    def __repr__(self):
+        return f"StructColumn({self.dtype!r}, n={len(self)})"

@@ -1010,6 +1010,33 @@ class Resolver:
                 for a in e.args[1:]:
                     tt = T.common_type(tt, a.dtype)
                 t = tt
+            elif e.name == "struct":
+                fields = []
+                for i, a in enumerate(e.args):
+                    nm = _expr_name(a, i)
+                    fields.append(T.StructField(nm, a.dtype))
+                t = T.StructType(tuple(fields))
+            elif e.name == "named_struct":
+                fields = []
+                for i in range(0, len(e.args), 2):
+                    k = e.args[i]
+                    if not isinstance(k, S.Literal):
+                        raise ResolutionError("named_struct expects literal field names")
+                    fields.append(T.StructField(str(k.value), e.args[i + 1].dtype))
+                t = T.StructType(tuple(fields))
+            elif e.name == "get_field":
+                st = e.args[0].dtype
+                fname = e.args[1].value if isinstance(e.args[1], S.Literal) else None
+                if isinstance(st, T.StructType) and fname is not None:
+                    ft = None
+                    for f in st.fields:
+                        if f.name.lower() == str(fname).lower():
+                            ft = f.dtype
+                    if ft is None:
+                        raise ResolutionError(f"no field {fname} in {st!r}")
+                    t = ft
+                else:
+                    raise ResolutionError("get_field expects struct.fieldname")
             e.dtype = t
             return e
         if isinstance(e, S.Alias):

@@ -41,3 +41,18 @@ def test_per_row_key_lookup(s):
 
 def test_array_subscript_zero_based(s):
     assert s.sql("SELECT array(10, 20, 30)[1], array(10)[5]").collect() == [(20, None)]
+
+
+def test_struct_roundtrip(s):
+    s.create_dataframe({"a": [1, 2], "b": ["x", "y"]}, name="st")
+    rows = s.sql("SELECT struct(a, b) FROM st ORDER BY a").collect()
+    assert rows == [({"a": 1, "b": "x"},), ({"a": 2, "b": "y"},)]
+    rows = s.sql("SELECT named_struct('p', a, 'q', b) FROM st ORDER BY a").collect()
+    assert rows == [({"p": 1, "q": "x"},), ({"p": 2, "q": "y"},)]
+
+
+def test_struct_field_access(s):
+    s.create_dataframe({"a": [1, 2], "b": ["x", "y"]}, name="st2")
+    assert s.sql("SELECT struct(a, b).b FROM st2 ORDER BY a").collect() == [("x",), ("y",)]
+    assert s.sql("SELECT get_field(struct(a, b), 'a') + 10 FROM st2 ORDER BY a").collect() == [
+        (11,), (12,)]

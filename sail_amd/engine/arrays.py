@@ -516,3 +516,51 @@ def _f_get_field(args, out, chunk, ev):
 IMPLS["struct"] = _f_struct
 IMPLS["named_struct"] = _f_named_struct
 IMPLS["get_field"] = _f_get_field
+
+
+# ---------------------------------------------------------------------------
+# higher-order functions: lambda body evaluated ONCE over the flattened
+# child (one vectorized pass), params at chunk positions 0..k-1 and the
+# enclosing row's columns (repeated per element) at k+i.
+# ---------------------------------------------------------------------------
+
+def eval_hof(ev, e, chunk):
+    from ..plan import spec as S
+    from .chunk import Chunk
+
+    name = e.name.lower()
+    arr = _bcast(ev.eval(e.args[0], chunk), chunk)
+    lam = e.args[1]
+    k = len(lam.params)
+    seg = arr.segment_ids()
+    total = len(arr.child)
+    dev = arr.device
+    flat_cols: List = [arr.child]
+    if k == 2:
+        pos = torch.arange(total, dtype=torch.int64, device=dev) \
+            - torch.repeat_interleave(arr.offsets[:-1], arr.lengths())
+        flat_cols.append(Column(T.I32, pos.to(torch.int32)))
+    for c in chunk.columns:
+        flat_cols.append(c.gather(seg) if c is not None else None)
+    flat = Chunk(flat_cols, [f"__l{i}" for i in range(len(flat_cols))],
+                 chunk.partitioning)
+    flat.forced_rows = total
+    from .eval import broadcast
+
+    res = broadcast(ev.eval(lam.body, flat), total, dev)
+    if name == "transform":
+        return ListColumn(arr.offsets, res, arr.validity)
+    mask = res.data.to(torch.bool) & res.valid_mask()
+    if name == "filter":
+        kept = torch.nonzero(mask, as_tuple=False).flatten()
+        new_seg = seg.index_select(0, kept)
+        lens = torch.bincount(new_seg, minlength=len(arr))
+        offs = torch.zeros(len(arr) + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(lens, 0, out=offs[1:])
+        return ListColumn(offs, arr.child.gather(kept), arr.validity, arr.dtype)
+    acc = torch.zeros(len(arr), dtype=torch.int64, device=dev)
+    acc.index_add_(0, seg, mask.to(torch.int64))
+    if name == "exists":
+        return Column(T.BOOL, acc > 0, arr.validity)
+    # forall: true when every element satisfies (vacuously true for empty)
+    return Column(T.BOOL, acc == arr.lengths(), arr.validity)

@@ -389,6 +389,10 @@ class Evaluator:
         sess = getattr(self.ctx, "session", None) if self.ctx is not None else None
         if sess is not None and getattr(sess, "udfs", None) and e.name.lower() in sess.udfs:
             return self._eval_udf(e, chunk)
+        if any(isinstance(a, S.Lambda) for a in e.args):
+            from .arrays import eval_hof
+
+            return eval_hof(self, e, chunk)
         args = [self.eval(a, chunk) for a in e.args]
         if args and all(isinstance(a, Scalar) for a in args) \
                 and e.name not in ("rand", "randn", "uuid", "monotonically_increasing_id") \

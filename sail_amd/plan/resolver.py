@@ -939,10 +939,47 @@ class Resolver:
                      for k in e.order_by]
             return S.WindowExpr(func=f, partition_by=part, order_by=order,
                                 frame=e.frame, dtype=f.dtype)
+        if isinstance(e, S.Func) and any(isinstance(a, S.Lambda) for a in e.args):
+            return self._resolve_hof(e, scope)
         # generic: resolve children then type
         ch = [self._expr(c, scope) for c in e.children()]
         out = e.with_children(ch) if ch else copy.copy(e)
         return self._type_expr(out)
+
+    _HOF = {"transform", "filter", "exists", "forall", "array_filter"}
+
+    def _resolve_hof(self, e: S.Func, scope: Scope) -> S.Expr:
+        """Higher-order array functions with lambdas (ref: sail-plan
+        resolver/expression/lambda.rs role). The lambda body is bound with
+        params at BoundRef 0..k-1 and the enclosing row's columns at k+i —
+        the evaluator builds a flattened-element chunk in that layout."""
+        name = e.name.lower()
+        if name not in self._HOF:
+            raise ResolutionError(f"unsupported lambda function {e.name}")
+        arr = self._expr(e.args[0], scope)
+        if not isinstance(arr.dtype, T.ArrayType):
+            raise ResolutionError(f"{name} expects an array argument")
+        lam = e.args[1]
+        if not isinstance(lam, S.Lambda):
+            raise ResolutionError(f"{name} expects a lambda")
+        k = len(lam.params)
+        if k > 2:
+            raise ResolutionError("lambdas take at most (element, index)")
+        ptypes = [arr.dtype.element] + ([T.I32] if k == 2 else [])
+        lam_fields = [Field(lam.params[i], ptypes[i]) for i in range(k)]
+        lscope = Scope(lam_fields + list(scope.fields), scope.outer)
+        body = self._expr(lam.body, lscope)
+        if name in ("filter", "exists", "forall", "array_filter"):
+            body = _coerce_to_bool(body)
+        blam = S.Lambda(lam.params, body, body.dtype)
+        if name == "transform":
+            t = T.ArrayType(body.dtype)
+        elif name in ("filter", "array_filter"):
+            t = arr.dtype
+        else:
+            t = T.BOOL
+        return S.Func("filter" if name == "array_filter" else name,
+                      [arr, blam], t)
 
     def _expr_window_func(self, f: S.Expr, scope: Scope) -> S.Expr:
         if isinstance(f, S.AggFunc):

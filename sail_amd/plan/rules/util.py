@@ -27,6 +27,13 @@ def _collect_refs(e: S.Expr, out: Set[int]):
     if isinstance(e, S.InSubquery):
         _collect_refs(e.child, out)
         return
+    if isinstance(e, S.Lambda):
+        # body refs 0..k-1 are lambda params; k+i is enclosing-chunk col i
+        k = len(e.params)
+        inner: Set[int] = set()
+        _collect_refs(e.body, inner)
+        out.update(r - k for r in inner if r >= k)
+        return
     for c in e.children():
         _collect_refs(c, out)
 
@@ -96,6 +103,11 @@ def remap_expr(e: S.Expr, mapping: Dict[int, int]) -> S.Expr:
                          remap_expr(e.filter, mapping) if e.filter is not None else None)
     if isinstance(e, S.SortKey):
         return S.SortKey(remap_expr(e.child, mapping), e.ascending, e.nulls_first)
+    if isinstance(e, S.Lambda):
+        k = len(e.params)
+        inner_map = {j: j for j in range(k)}
+        inner_map.update({i + k: v + k for i, v in mapping.items()})
+        return S.Lambda(e.params, remap_expr(e.body, inner_map), e.dtype)
     if isinstance(e, S.WindowExpr):
         return S.WindowExpr(func=remap_expr(e.func, mapping),
                             partition_by=[remap_expr(x, mapping) for x in e.partition_by],

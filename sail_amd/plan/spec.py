@@ -256,6 +256,27 @@ class Like(Expr):
 
 
 @dataclass
+class Lambda(Expr):
+    """Higher-order-function lambda `x -> body` / `(x, i) -> body`
+    (ref: Spark lambda functions; sail-plan resolver/expression/lambda.rs).
+    After resolution, `body` references params as BoundRef 0..k-1 into the
+    flattened-element chunk (outer columns follow at k+i)."""
+
+    params: List[str] = field(default_factory=list)
+    body: Expr = None
+    dtype: Optional[T.DataType] = None
+
+    def children(self):
+        return [self.body]
+
+    def with_children(self, ch):
+        return Lambda(self.params, ch[0], self.dtype)
+
+    def __repr__(self):
+        return f"({', '.join(self.params)}) -> {self.body!r}"
+
+
+@dataclass
 class Star(Expr):
     qualifier: Optional[str] = None
 

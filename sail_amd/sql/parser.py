@@ -801,7 +801,44 @@ class Parser:
     # Expressions (Pratt)
     # =======================================================================
     def parse_expr(self) -> S.Expr:
+        # lambda: `x -> expr` or `(x, y) -> expr`
+        t = self.peek()
+        if t.kind == "ident" and self.peek(1).kind == "op" and self.peek(1).value == "->":
+            name = self.ident()
+            self.next()  # ->
+            return S.Lambda([name], self.parse_expr())
+        if self.at_op("("):
+            save = self.i
+            if self._try_lambda_params() is not None:
+                params = self._try_lambda_params(consume=True)
+                return S.Lambda(params, self.parse_expr())
+            self.i = save
         return self._parse_or()
+
+    def _try_lambda_params(self, consume: bool = False):
+        """Look ahead for `(a, b, ...) ->`; returns params or None."""
+        save = self.i
+        if not self.eat_op("("):
+            self.i = save
+            return None
+        params = []
+        while True:
+            t = self.peek()
+            if t.kind != "ident":
+                self.i = save
+                return None
+            params.append(t.value)
+            self.next()
+            if self.eat_op(","):
+                continue
+            break
+        if not self.eat_op(")") or not (self.at_op("->")):
+            self.i = save
+            return None
+        self.next()  # ->
+        if not consume:
+            self.i = save
+        return params
 
     def _parse_or(self) -> S.Expr:
         left = self._parse_and()
@@ -821,7 +858,8 @@ class Parser:
         return self._parse_predicate()
 
     def _parse_predicate(self) -> S.Expr:
-        if self.at_kw("EXISTS") and self.peek(1).kind == "op" and self.peek(1).value == "(":
+        if self.at_kw("EXISTS") and self.peek(1).kind == "op" and self.peek(1).value == "(" \
+                and self.peek(2).kind == "ident" and self.peek(2).upper in ("SELECT", "WITH"):
             self.next()
             self.expect_op("(")
             sub = self.parse_query()
@@ -1017,8 +1055,11 @@ class Parser:
         if kw == "FALSE":
             self.next()
             return S.Literal(False, T.BOOL)
-        if kw == "EXISTS":
-            # handled in predicate, but can appear here after NOT
+        if kw == "EXISTS" and self.peek(2).kind == "ident" \
+                and self.peek(2).upper in ("SELECT", "WITH"):
+            # subquery EXISTS (predicate path covers the normal spot; this
+            # one appears after NOT). exists(arr, lambda) falls through to
+            # the generic function-call parse.
             self.next()
             self.expect_op("(")
             sub = self.parse_query()

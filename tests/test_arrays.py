@@ -133,3 +133,29 @@ def test_lateral_view_then_join(s):
     rows = s.sql("SELECT k, e, w FROM lt4 LATERAL VIEW explode(sequence(1, v)) x AS e "
                  "JOIN lt5 USING (e) ORDER BY k, e").collect()
     assert rows == [("a", 1, 10), ("a", 2, 20), ("b", 1, 10)]
+
+
+def test_lambda_transform(s):
+    assert s.sql("SELECT transform(array(1,2,3), x -> x * 10)").collect() == [([10, 20, 30],)]
+    assert s.sql("SELECT transform(array(1,2,3), (x, i) -> x + i)").collect() == [([1, 3, 5],)]
+
+
+def test_lambda_filter_exists_forall(s):
+    rows = s.sql("SELECT filter(array(1,2,3,4), x -> x % 2 = 0), "
+                 "exists(array(1,2), x -> x > 1), forall(array(1,2), x -> x > 1), "
+                 "forall(array(), x -> x > 1)").collect()
+    assert rows == [([2, 4], True, False, True)]
+
+
+def test_lambda_captures_outer_column(s):
+    s.create_dataframe({"base": [10, 20], "v": [2, 3]}, name="hof")
+    rows = s.sql("SELECT transform(sequence(1, v), x -> x + base) FROM hof ORDER BY base").collect()
+    assert rows == [([11, 12],), ([21, 22, 23],)]
+    rows = s.sql("SELECT filter(sequence(1, 5), x -> x <= v) FROM hof ORDER BY base").collect()
+    assert rows == [([1, 2],), ([1, 2, 3],)]
+
+
+def test_lambda_string_elements(s):
+    rows = s.sql("SELECT transform(split('a,bb,ccc', ','), x -> length(x)), "
+                 "filter(split('a,bb,ccc', ','), x -> length(x) > 1)").collect()
+    assert rows == [([1, 2, 3], ["bb", "ccc"])]

@@ -1,12 +1,30 @@
-"""Data source registry (parquet/csv/json/memory).
+"""Data source registry (parquet/csv/json/delta + user-defined formats).
 
-ref: crates/sail-data-source/src/formats/ for the reference's format set.
+ref: crates/sail-data-source/src/formats/ for the reference's format set;
+user-defined Python data sources mirror formats/python/ (discovery via an
+in-process registry instead of entry points — no package installation in
+the image).
 """
 from __future__ import annotations
 
 from typing import Dict, List, Optional, Tuple
 
 from ..engine import types as T
+
+#: user-defined formats: name -> object with infer_schema(paths, options),
+#: read(paths, schema, device, options) -> Table, optional
+#: write(path, chunk, mode, options)
+_USER_FORMATS: Dict[str, object] = {}
+
+
+def register_format(name: str, source) -> None:
+    """Register a Python data source (ref: sail-data-source
+    formats/python/ — user-defined DataSource classes). `source` provides:
+      infer_schema(paths, options) -> [(name, DataType)]
+      read(paths, schema, device, options) -> engine.column.Table
+      write(path, chunk, mode, options)     (optional)
+    """
+    _USER_FORMATS[name.lower()] = source
 
 
 def infer_source_schema(fmt: str, paths: List[str], options: Dict[str, str]):
@@ -24,6 +42,8 @@ def infer_source_schema(fmt: str, paths: List[str], options: Dict[str, str]):
         from . import delta
 
         return delta.infer_schema(paths, options)
+    if fmt in _USER_FORMATS:
+        return _USER_FORMATS[fmt].infer_schema(paths, options)
     raise ValueError(f"unsupported format {fmt}")
 
 
@@ -42,6 +62,8 @@ def read_source(fmt: str, paths: List[str], options: Dict[str, str], schema, dev
         from . import delta
 
         return delta.read(paths, schema, device, options)
+    if fmt in _USER_FORMATS:
+        return _USER_FORMATS[fmt].read(paths, schema, device, options)
     raise ValueError(f"unsupported format {fmt}")
 
 
@@ -56,4 +78,6 @@ def write_source(fmt: str, path: str, chunk, mode: str, options, partition_by):
         from . import delta
 
         return delta.write(path, chunk, mode, options)
+    if fmt in _USER_FORMATS and hasattr(_USER_FORMATS[fmt], "write"):
+        return _USER_FORMATS[fmt].write(path, chunk, mode, options)
     raise ValueError(f"unsupported write format {fmt}")

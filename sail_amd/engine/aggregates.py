@@ -134,11 +134,37 @@ def _masked(values: Column, extra_mask: Optional[torch.Tensor]):
     return mask
 
 
+#: user-defined aggregates: name -> (fn(list_of_group_values) -> scalar, type)
+#: engine-wide (registered via session.udf.register_aggregate); evaluated on
+#: host per group — the pyo3 UDAF boundary of the reference
+#: (ref: crates/sail-python-udf aggregate UDFs) as an in-process call.
+UDAFS: dict = {}
+
+
+def _udaf_eval(name, args, gid, ng, filter_mask, out_type):
+    fn, rt = UDAFS[name]
+    rt = out_type or rt
+    c = args[0]
+    mask = _masked(c, filter_mask)
+    gl = gid[mask].cpu().tolist()
+    import torch as _t
+
+    kept = _t.nonzero(mask, as_tuple=False).flatten()
+    vals = c.gather(kept.to(c.device)).to_pylist()
+    groups = [[] for _ in range(ng)]
+    for g, v in zip(gl, vals):
+        groups[g].append(v)
+    out = [fn(gvals) for gvals in groups]
+    return Column.from_values(out, rt, device=gid.device)
+
+
 def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
              distinct: bool = False, filter_mask: Optional[torch.Tensor] = None,
              out_type: T.DataType = None) -> Column:
     """Evaluate one aggregate over groups. args already evaluated per-row."""
     dev = gid.device
+    if name in UDAFS:
+        return _udaf_eval(name, args, gid, ng, filter_mask, out_type)
     if name == "count" and not args:
         mask = filter_mask if filter_mask is not None else torch.ones(gid.shape[0], dtype=torch.bool, device=dev)
         data = torch.zeros(ng, dtype=torch.int64, device=dev)

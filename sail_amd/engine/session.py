@@ -342,6 +342,17 @@ class UdfRegistry:
         self._session.udfs[name.lower()] = (fn, return_type or T.F64, vectorized)
         return fn
 
+    def register_aggregate(self, name: str, fn, return_type=None):
+        """UDAF: fn receives the group's values as a Python list and returns
+        one value (host evaluation; engine-wide registry)."""
+        from . import types as T
+        from .aggregates import UDAFS
+
+        if isinstance(return_type, str):
+            return_type = T.type_from_name(return_type)
+        UDAFS[name.lower()] = (fn, return_type or T.F64)
+        return fn
+
 
 class _CatalogAdapter:
     """Bridges the resolver's catalog protocol to Catalog + view expansion."""

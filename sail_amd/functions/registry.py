@@ -37,6 +37,10 @@ WINDOW_FUNCTIONS = {
 
 
 def agg_return_type(name: str, arg_types: List[T.DataType], distinct: bool = False) -> T.DataType:
+    from ..engine.aggregates import UDAFS
+
+    if name in UDAFS:
+        return UDAFS[name][1]
     a = arg_types[0] if arg_types else T.NULL
     if name in ("count", "count_if", "approx_count_distinct", "regr_count"):
         return T.I64

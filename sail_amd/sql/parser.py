@@ -1136,8 +1136,9 @@ class Parser:
             self.expect_op(")")
 
         from ..functions.registry import AGG_FUNCTIONS, WINDOW_FUNCTIONS
+        from ..engine.aggregates import UDAFS
         e: S.Expr
-        if lname in AGG_FUNCTIONS:
+        if lname in AGG_FUNCTIONS or lname in UDAFS:
             e = S.AggFunc(lname, args, distinct=distinct, filter=filt)
         elif lname in WINDOW_FUNCTIONS:
             e = S.Func(lname, args)

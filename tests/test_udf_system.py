@@ -44,3 +44,43 @@ def test_config_defaults_and_env(monkeypatch):
     s = sail_amd.SessionContext(device="cpu")
     assert s.conf["sail.optimizer.enable.join.reorder"] == "false"
     assert s.conf["spark.sql.session.timeZone"] == "UTC"
+
+
+def test_udaf_register_and_group(session):
+    s = session
+    import math
+
+    s.udf.register_aggregate(
+        "geo_mean",
+        lambda vals: math.exp(sum(math.log(v) for v in vals) / len(vals)) if vals else None,
+        "double")
+    s.create_dataframe({"g": ["a", "a", "b"], "v": [2.0, 8.0, 5.0]}, name="ua")
+    rows = s.sql("SELECT g, geo_mean(v) FROM ua GROUP BY g ORDER BY g").collect()
+    assert rows[0][0] == "a" and abs(rows[0][1] - 4.0) < 1e-9
+    assert rows[1][0] == "b" and abs(rows[1][1] - 5.0) < 1e-9
+    # global aggregate
+    assert abs(s.sql("SELECT geo_mean(v) FROM ua").collect()[0][0] - 4.30886938) < 1e-6
+
+
+def test_python_data_source(session, tmp_path):
+    s = session
+    """User-defined format (ref: sail-data-source formats/python/)."""
+    from sail_amd.datasource.registry import register_format
+    from sail_amd.engine.column import Column, Table
+    from sail_amd.engine import types as T
+
+    class FibSource:
+        def infer_schema(self, paths, options):
+            return [("n", T.I64), ("fib", T.I64)]
+
+        def read(self, paths, schema, device, options):
+            k = int(options.get("count", "8"))
+            fibs = [0, 1]
+            while len(fibs) < k:
+                fibs.append(fibs[-1] + fibs[-2])
+            return Table({"n": Column.from_values(list(range(k)), T.I64, device=device),
+                          "fib": Column.from_values(fibs[:k], T.I64, device=device)})
+
+    register_format("fib", FibSource())
+    df = s.read.format("fib").option("count", 7).load("ignored")
+    assert df.collect() == [(0, 0), (1, 1), (2, 1), (3, 2), (4, 3), (5, 5), (6, 8)]

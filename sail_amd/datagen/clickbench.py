@@ -133,4 +133,8 @@ def register_clickbench(session, rows: int = 100_000_000, device=None,
     t = generate_hits(rows=rows, device=dev, seed=seed, rank=rank, world=world)
     session.catalog.register_table("hits", t, replicated=(world == 1),
                                    global_rows=rows)
+    if world > 1 and getattr(session, "dist", None) is not None:
+        from ..exec.distributed import sync_table_stats
+
+        sync_table_stats(session)
     return t

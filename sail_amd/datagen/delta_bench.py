@@ -45,6 +45,10 @@ def setup_delta_bench(session, sf: float = 10.0, device="cpu", rank: int = 0, wo
         "o_totalprice": Column(T.DecimalType(12, 2), _randint(100000, 50000000, nu, g, torch.device(device))),
     })
     session.catalog.register_table("updates", updates, replicated=(world == 1))
+    if world > 1 and getattr(session, "dist", None) is not None:
+        from ..exec.distributed import sync_table_stats
+
+        sync_table_stats(session)
     _STATE["path"] = path
     session.conf["sail.delta.bench.path"] = path
 

@@ -430,4 +430,8 @@ def register_tpch(session, sf: float = 0.01, device=None, rank: int = 0, world: 
         session.catalog.register_table(
             name, tbl, replicated=(world == 1 or name in ("region", "nation")),
             global_rows=globals_[name])
+    if world > 1 and getattr(session, "dist", None) is not None:
+        from ..exec.distributed import sync_table_stats
+
+        sync_table_stats(session)
     return tables

@@ -129,6 +129,15 @@ class Catalog:
             return t
         return None
 
+    def set_column_stats(self, name: str, col: str, rows: int, ndv):
+        """Explicit planner statistics. In SPMD runs these MUST be set to
+        rank-identical values for sharded tables (sync_table_stats):
+        shard-local min/max give each rank a different ndv estimate, the
+        join reorderer then picks different orders per rank, and the ranks'
+        collectives mismatch (caught by the world=4 gloo test)."""
+        with self._lock:
+            self._col_stats[(self._key(name), col.lower())] = (rows, ndv)
+
     def column_stats(self, name: str, col: str):
         """(rows, ndv_estimate) for a base-table column; ndv from min/max
         span for integer-like storage, dict size for dictionary strings.

@@ -27,27 +27,30 @@ from .context import DistContext
 
 
 def gather_column(c: Column, d: DistContext) -> Column:
+    """NOTE: the validity mask is exchanged UNCONDITIONALLY (ones when the
+    shard has no nulls): whether a shard contains nulls is rank-local state,
+    and skipping the collective on some ranks only is a collective-order
+    mismatch (caught by the world=4 gloo test). n uint8 per column is noise
+    next to the data payload."""
     from ..engine.executor import concat_columns
 
     if isinstance(c, StringColumn):
         raw = c.decode_dict()
         offs_list = d.all_gather_tensors(raw.offsets[1:] - raw.offsets[:-1])
         bytes_list = d.all_gather_tensors(raw.bytes_)
-        val_list = d.all_gather_tensors(raw.valid_mask().to(torch.uint8)) \
-            if raw.validity is not None else None
+        val_list = d.all_gather_tensors(raw.valid_mask().to(torch.uint8))
         cols = []
         for i in range(d.world):
             lens = offs_list[i]
             offs = torch.zeros(lens.shape[0] + 1, dtype=torch.int64, device=lens.device)
             torch.cumsum(lens, 0, out=offs[1:])
-            v = val_list[i] if val_list is not None else None
+            v = val_list[i] if not bool(val_list[i].all()) else None
             cols.append(StringColumn(offs, bytes_list[i], v, None, dtype=c.dtype))
         return concat_columns(cols)
     data_list = d.all_gather_tensors(c.data)
-    if c.validity is not None:
-        val_list = d.all_gather_tensors(c.validity)
-        validity = torch.cat(val_list)
-    else:
+    val_list = d.all_gather_tensors(c.valid_mask().to(torch.uint8))
+    validity = torch.cat(val_list)
+    if bool(validity.all()):
         validity = None
     return Column(c.dtype, torch.cat(data_list), validity)
 
@@ -211,7 +214,52 @@ def shuffle_chunk(chunk: Chunk, key_idx, d: DistContext) -> Chunk:
 
 
 def _exchange_validity(c, order, send_counts, d):
-    if c.validity is None:
-        return None
-    v = c.validity.index_select(0, order)
-    return _exchange_1d(v, send_counts, d)
+    """Unconditional (see gather_column): null-presence is rank-local and
+    must not gate a collective."""
+    v = c.valid_mask().to(torch.uint8).index_select(0, order)
+    out = _exchange_1d(v, send_counts, d)
+    return None if bool(out.all()) else out
+
+
+def sync_table_stats(session):
+    """Make planner column statistics rank-identical: all_reduce the
+    min/max/dict-size of every sharded table's columns ONCE at registration
+    (planning itself stays communication-free, so a rank can never hang on
+    stats). Must be called in the same order on every rank."""
+    d = getattr(session, "dist", None)
+    if d is None or d.world <= 1:
+        return
+    cat = session.catalog
+    for name in sorted(cat._tables):
+        if cat.is_replicated(name):
+            continue  # identical on all ranks already
+        t = cat._tables[name]
+        grows = cat._global_rows.get(name, None)
+        if grows is None:
+            n = torch.tensor([t.num_rows], dtype=torch.int64)
+            d.all_reduce_sum_(n)
+            grows = int(n.item())
+            cat._global_rows[name] = grows
+        for cn, c in t.columns.items():
+            rows = len(c)
+            from ..engine.column import StringColumn
+
+            if isinstance(c, StringColumn):
+                local = c.dict_size if c.is_dict else -1
+                agg = torch.tensor([local], dtype=torch.int64)
+                d.dist.all_reduce(agg, op=d.dist.ReduceOp.MAX)
+                ndv = int(agg.item()) if int(agg.item()) >= 0 else None
+            elif rows == 0 or c.data.dtype == torch.bool:
+                ndv = 2
+            elif c.data.dtype.is_floating_point:
+                ndv = None
+            else:
+                lohi = torch.tensor([-int(c.data.min().item()) if rows else 0,
+                                     int(c.data.max().item()) if rows else 0],
+                                    dtype=torch.int64)
+                d.dist.all_reduce(lohi, op=d.dist.ReduceOp.MAX)
+                lo, hi = -int(lohi[0].item()), int(lohi[1].item())
+                ndv = min(grows, hi - lo + 1)
+            if ndv is not None:
+                ndv = min(max(1, ndv), grows)
+            cat.set_column_stats(name, cn, grows, ndv)

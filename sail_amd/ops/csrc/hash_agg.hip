@@ -273,20 +273,28 @@ torch::Tensor grouped_acc(torch::Tensor gid, c10::optional<torch::Tensor> mask,
   for (int c = 0; c < ncols && tiny_ok; ++c)
     if (ops[c] == 3 || ops[c] == 4) tiny_ok = false;
   if (tiny_ok) {
-    int grid = grid_for_n(n);
-#define TINY_CASE(NCV)                                                          \
-  case NCV:                                                                     \
-    hipLaunchKernelGGL((grouped_acc_tiny<8, NCV>), dim3(grid), dim3(kBlock), 0, \
-                       stream, gid.data_ptr<int32_t>(), mptr, n, cols[0],       \
-                       cols[1], cols[2], cols[3], cols[4], cols[5], cols[6],    \
-                       cols[7], cols[8], cols[9],                               \
-                       out.data_ptr<int64_t>(), (int)G);                        \
+    // TG matched to the actual group count: acc[NC][TG] lives in VGPRs, so
+    // TG=8 for G=4 (q1) doubles register pressure for nothing and costs
+    // occupancy — dispatch the tightest template that fits.
+    int grid = grid_for_n(n, 8);
+#define TINY_LAUNCH(TGV, NCV)                                                     \
+    hipLaunchKernelGGL((grouped_acc_tiny<TGV, NCV>), dim3(grid), dim3(kBlock), 0, \
+                       stream, gid.data_ptr<int32_t>(), mptr, n, cols[0],         \
+                       cols[1], cols[2], cols[3], cols[4], cols[5], cols[6],      \
+                       cols[7], cols[8], cols[9],                                 \
+                       out.data_ptr<int64_t>(), (int)G)
+#define TINY_CASE(NCV)                \
+  case NCV:                           \
+    if (G <= 2) TINY_LAUNCH(2, NCV);  \
+    else if (G <= 4) TINY_LAUNCH(4, NCV); \
+    else TINY_LAUNCH(8, NCV);         \
     break;
     switch (ncols) {
       TINY_CASE(1) TINY_CASE(2) TINY_CASE(3) TINY_CASE(4) TINY_CASE(5)
       TINY_CASE(6) TINY_CASE(7) TINY_CASE(8) TINY_CASE(9) TINY_CASE(10)
     }
 #undef TINY_CASE
+#undef TINY_LAUNCH
     return out;
   }
   TORCH_CHECK(G <= 4096, "grouped_acc: G too large for LDS variant");

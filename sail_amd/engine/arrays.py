@@ -529,6 +529,12 @@ def eval_hof(ev, e, chunk):
     from .chunk import Chunk
 
     name = e.name.lower()
+    if name in ("aggregate", "reduce"):
+        return _f_reduce(e.args, e.dtype, chunk, ev)
+    if name == "zip_with":
+        a = _bcast(ev.eval(e.args[0], chunk), chunk)
+        b = _bcast(ev.eval(e.args[1], chunk), chunk)
+        return _f_zip_with([a, b], e.dtype, chunk, ev, e.args[2])
     arr = _bcast(ev.eval(e.args[0], chunk), chunk)
     lam = e.args[1]
     k = len(lam.params)
@@ -564,3 +570,100 @@ def eval_hof(ev, e, chunk):
         return Column(T.BOOL, acc > 0, arr.validity)
     # forall: true when every element satisfies (vacuously true for empty)
     return Column(T.BOOL, acc == arr.lengths(), arr.validity)
+
+
+def _f_zip_with(args_cols, out, chunk, ev, lam):
+    """zip_with(a, b, (x, y) -> e): parallel element walk; shorter side
+    null-padded (Spark semantics)."""
+    from .chunk import Chunk
+    from .eval import broadcast
+
+    a, b = args_cols
+    la, lb = a.lengths(), b.lengths()
+    lens = torch.maximum(la, lb)
+    n = len(a)
+    dev = a.device
+    offs = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(lens, 0, out=offs[1:])
+    total = int(offs[-1].item())
+    seg = torch.repeat_interleave(torch.arange(n, dtype=torch.int64, device=dev), lens)
+    pos = torch.arange(total, dtype=torch.int64, device=dev) \
+        - torch.repeat_interleave(offs[:-1], lens)
+
+    def padded(c, clens):
+        ok = pos < clens.index_select(0, seg)
+        idx = (c.offsets[:-1].index_select(0, seg) + pos).clamp(0, max(len(c.child) - 1, 0))
+        got = c.child.gather(idx) if len(c.child) else None
+        if got is None:
+            return Column.from_values([None] * total, c.child.dtype, device=dev)
+        valid = ok & got.valid_mask()
+        v = None if bool(valid.all()) else valid.to(torch.uint8)
+        if isinstance(got, (StringColumn, ListColumn)):
+            got.validity = v
+            return got
+        return Column(got.dtype, got.data, v)
+
+    ea, eb = padded(a, la), padded(b, lb)
+    flat_cols = [ea, eb] + [c.gather(seg) if c is not None else None
+                            for c in chunk.columns]
+    flat = Chunk(flat_cols, [f"__z{i}" for i in range(len(flat_cols))],
+                 chunk.partitioning)
+    flat.forced_rows = total
+    res = broadcast(ev.eval(lam.body, flat), total, dev)
+    validity = None
+    va, vb = a.valid_mask(), b.valid_mask()
+    both = va & vb
+    if not bool(both.all()):
+        validity = both.to(torch.uint8)
+    return ListColumn(offs, res, validity)
+
+
+def _f_reduce(args, out, chunk, ev):
+    """aggregate(arr, init, (acc, x) -> e [, finish]): K vectorized rounds
+    (K = max array length) — round k folds element k into the accumulator
+    for every row whose array is long enough; no per-row Python loop."""
+    from .chunk import Chunk
+    from .eval import broadcast, cast_column
+
+    arr = _bcast(ev.eval(args[0], chunk), chunk)
+    lam = args[2]
+    acc = cast_column(_bcast(ev.eval(args[1], chunk), chunk), lam.body.dtype) \
+        if lam.body.dtype is not None else _bcast(ev.eval(args[1], chunk), chunk)
+    lens = arr.lengths()
+    n = len(arr)
+    dev = arr.device
+    kmax = int(lens.max().item()) if n else 0
+    for k in range(kmax):
+        live = lens > k
+        idx = (arr.offsets[:-1] + k).clamp(0, max(len(arr.child) - 1, 0))
+        elem = arr.child.gather(idx)
+        flat_cols = [acc, elem] + list(chunk.columns)
+        flat = Chunk(flat_cols, [f"__r{i}" for i in range(len(flat_cols))],
+                     chunk.partitioning)
+        flat.forced_rows = n
+        new_acc = broadcast(ev.eval(lam.body, flat), n, dev)
+        # rows whose array ended keep their accumulator
+        if bool(live.all()):
+            acc = new_acc
+        else:
+            keep = ~live
+            if isinstance(acc, (StringColumn, ListColumn)):
+                raise NotImplementedError("aggregate() with string accumulator")
+            data = torch.where(keep, acc.data, new_acc.data)
+            av = acc.valid_mask()
+            nv = new_acc.valid_mask()
+            valid = torch.where(keep, av, nv)
+            acc = Column(new_acc.dtype, data,
+                         None if bool(valid.all()) else valid.to(torch.uint8))
+    if len(args) > 3:
+        fin = args[3]
+        flat = Chunk([acc] + list(chunk.columns),
+                     [f"__f{i}" for i in range(1 + len(chunk.columns))],
+                     chunk.partitioning)
+        flat.forced_rows = n
+        acc = broadcast(ev.eval(fin.body, flat), n, dev)
+    # null array -> null result
+    if arr.validity is not None:
+        valid = arr.valid_mask() & acc.valid_mask()
+        acc = Column(acc.dtype, acc.data, valid.to(torch.uint8))
+    return acc

@@ -946,7 +946,8 @@ class Resolver:
         out = e.with_children(ch) if ch else copy.copy(e)
         return self._type_expr(out)
 
-    _HOF = {"transform", "filter", "exists", "forall", "array_filter"}
+    _HOF = {"transform", "filter", "exists", "forall", "array_filter",
+            "zip_with", "aggregate", "reduce"}
 
     def _resolve_hof(self, e: S.Func, scope: Scope) -> S.Expr:
         """Higher-order array functions with lambdas (ref: sail-plan
@@ -956,6 +957,10 @@ class Resolver:
         name = e.name.lower()
         if name not in self._HOF:
             raise ResolutionError(f"unsupported lambda function {e.name}")
+        if name == "zip_with":
+            return self._resolve_zip_with(e, scope)
+        if name in ("aggregate", "reduce"):
+            return self._resolve_reduce(e, scope)
         arr = self._expr(e.args[0], scope)
         if not isinstance(arr.dtype, T.ArrayType):
             raise ResolutionError(f"{name} expects an array argument")
@@ -980,6 +985,45 @@ class Resolver:
             t = T.BOOL
         return S.Func("filter" if name == "array_filter" else name,
                       [arr, blam], t)
+
+    def _bind_lambda(self, lam, ptypes, scope):
+        if not isinstance(lam, S.Lambda):
+            raise ResolutionError("expected a lambda")
+        if len(lam.params) != len(ptypes):
+            raise ResolutionError(
+                f"lambda expects {len(ptypes)} parameters, got {len(lam.params)}")
+        fields = [Field(lam.params[i], ptypes[i]) for i in range(len(ptypes))]
+        body = self._expr(lam.body, Scope(fields + list(scope.fields), scope.outer))
+        return S.Lambda(lam.params, body, body.dtype)
+
+    def _resolve_zip_with(self, e, scope):
+        a = self._expr(e.args[0], scope)
+        b = self._expr(e.args[1], scope)
+        if not (isinstance(a.dtype, T.ArrayType) and isinstance(b.dtype, T.ArrayType)):
+            raise ResolutionError("zip_with expects two arrays")
+        blam = self._bind_lambda(e.args[2], [a.dtype.element, b.dtype.element], scope)
+        return S.Func("zip_with", [a, b, blam], T.ArrayType(blam.dtype))
+
+    def _resolve_reduce(self, e, scope):
+        """aggregate(arr, init, (acc, x) -> merge [, acc -> finish])"""
+        arr = self._expr(e.args[0], scope)
+        if not isinstance(arr.dtype, T.ArrayType):
+            raise ResolutionError("aggregate expects an array")
+        init = self._expr(e.args[1], scope)
+        acc_t = init.dtype
+        merge = self._bind_lambda(e.args[2], [acc_t, arr.dtype.element], scope)
+        if merge.dtype != acc_t:
+            # accumulator type widened by the merge (e.g. int init, double
+            # body): rebind once with the widened type
+            acc_t = merge.dtype
+            merge = self._bind_lambda(e.args[2], [acc_t, arr.dtype.element], scope)
+        out_t = acc_t
+        args = [arr, init, merge]
+        if len(e.args) > 3:
+            fin = self._bind_lambda(e.args[3], [acc_t], scope)
+            out_t = fin.dtype
+            args.append(fin)
+        return S.Func("aggregate", args, out_t)
 
     def _expr_window_func(self, f: S.Expr, scope: Scope) -> S.Expr:
         if isinstance(f, S.AggFunc):

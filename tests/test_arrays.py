@@ -159,3 +159,19 @@ def test_lambda_string_elements(s):
     rows = s.sql("SELECT transform(split('a,bb,ccc', ','), x -> length(x)), "
                  "filter(split('a,bb,ccc', ','), x -> length(x) > 1)").collect()
     assert rows == [([1, 2, 3], ["bb", "ccc"])]
+
+
+def test_zip_with(s):
+    rows = s.sql("SELECT zip_with(array(1,2,3), array(10,20,30), (x, y) -> x + y), "
+                 "zip_with(array(1,2), array(10,20,30), (x, y) -> y)").collect()
+    assert rows == [([11, 22, 33], [10, 20, 30])]
+
+
+def test_aggregate_reduce(s):
+    rows = s.sql("SELECT aggregate(array(1,2,3,4), 0, (acc, x) -> acc + x), "
+                 "aggregate(array(1,2,3), 1, (acc, x) -> acc * x, acc -> acc * 10)").collect()
+    assert rows == [(10, 60)]
+    s.create_dataframe({"v": [2, 4]}, name="agg_t")
+    rows = s.sql("SELECT aggregate(sequence(1, v), 0, (acc, x) -> acc + x) "
+                 "FROM agg_t ORDER BY v").collect()
+    assert rows == [(3,), (10,)]

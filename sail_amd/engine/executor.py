@@ -119,6 +119,20 @@ class Executor:
             out.partitioning = "sharded"
         return out
 
+    def _x_TableFuncRead(self, p: S.TableFuncRead) -> Chunk:
+        info = self.ctx.session.udtfs.get(p.name.lower())
+        if info is None:
+            raise ExecError(f"table function {p.name} not registered")
+        fn, schema = info
+        argvals = []
+        for a in p.args:
+            v = self.ev.eval(a, Chunk([], [], forced_rows=1))
+            argvals.append(v.value if isinstance(v, Scalar) else v.to_pylist()[0])
+        data = fn(*argvals)
+        cols = [Column.from_values(data[n], t, device=self.ctx.device)
+                for n, t in schema]
+        return Chunk(cols, [n for n, _ in schema])
+
     def _x_ChunkSource(self, p: S.ChunkSource) -> Chunk:
         return p.chunk
 

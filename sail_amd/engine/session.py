@@ -203,6 +203,7 @@ class SessionContext:
 
         self.conf: Dict[str, str] = load_config()
         self.udfs: Dict[str, tuple] = {}
+        self.udtfs: Dict[str, tuple] = {}
         self.query_log: List[dict] = []
         self._streams = None
         self._register_system_tables()
@@ -351,6 +352,18 @@ class UdfRegistry:
         self._session.udfs[name.lower()] = (fn, return_type or T.F64, vectorized)
         return fn
 
+    def register_table_function(self, name: str, fn, schema):
+        """UDTF: fn(*scalar_args) -> {column: [values]} with a declared
+        schema ([(name, type)] or {name: type}); usable in FROM."""
+        from . import types as T
+
+        if isinstance(schema, dict):
+            schema = list(schema.items())
+        schema = [(n, T.type_from_name(t) if isinstance(t, str) else t)
+                  for n, t in schema]
+        self._session.udtfs[name.lower()] = (fn, schema)
+        return fn
+
     def register_aggregate(self, name: str, fn, return_type=None):
         """UDAF: fn receives the group's values as a Python list and returns
         one value (host evaluation; engine-wide registry)."""
@@ -386,6 +399,9 @@ class _CatalogAdapter:
 
     def udf(self, name: str):
         return self.session.udfs.get(name.lower())
+
+    def udtf(self, name: str):
+        return self.session.udtfs.get(name.lower())
 
 
 # patch Resolver to consult views: Read resolution checks views first

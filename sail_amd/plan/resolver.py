@@ -127,6 +127,16 @@ class Resolver:
             out.schema = [(k, _infer_pytype(v)) for k, v in p.data.items()]
         return out
 
+    def _p_TableFuncRead(self, p: S.TableFuncRead, outer):
+        udtf = getattr(self.catalog, "udtf", None)
+        info = udtf(p.name) if udtf else None
+        if info is None:
+            raise ResolutionError(f"unknown table function {p.name}")
+        out = S.TableFuncRead(name=p.name,
+                              args=[self._expr(a, Scope([], outer)) for a in p.args])
+        out.schema = list(info[1])
+        return self._qualify(out, p.name)
+
     def _p_Range(self, p: S.Range, outer):
         out = S.Range(p.start, p.end, p.step)
         out.schema = [("id", T.I64)]

@@ -368,6 +368,16 @@ class LocalRelation(Plan):
 
 
 @dataclass
+class TableFuncRead(Plan):
+    """FROM my_udtf(args...) — user-defined table function
+    (ref: Spark UDTF; sail-python-udf table functions)."""
+
+    name: str = ""
+    args: List[Expr] = field(default_factory=list)
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+
+@dataclass
 class ChunkSource(Plan):
     """Pre-materialized chunk spliced into a plan (streaming incremental
     aggregation substitutes the Aggregate subtree with its merged state)."""

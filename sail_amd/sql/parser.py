@@ -777,7 +777,7 @@ class Parser:
             if len(vals) == 2:
                 return S.Range(vals[0], vals[1], 1)
             return S.Range(vals[0], vals[1], vals[2])
-        raise SqlError(f"unsupported table function {name}")
+        return S.TableFuncRead(name=lname, args=args)
 
     def _parse_alias(self):
         if self.eat_kw("AS"):

@@ -84,3 +84,22 @@ def test_python_data_source(session, tmp_path):
     register_format("fib", FibSource())
     df = s.read.format("fib").option("count", 7).load("ignored")
     assert df.collect() == [(0, 0), (1, 1), (2, 1), (3, 2), (4, 3), (5, 5), (6, 8)]
+
+
+def test_udtf_table_function(session):
+    s = session
+
+    def primes_upto(n):
+        out = []
+        for x in range(2, n + 1):
+            if all(x % d for d in range(2, int(x ** 0.5) + 1)):
+                out.append(x)
+        return {"p": out}
+
+    s.udf.register_table_function("primes", primes_upto, {"p": "bigint"})
+    assert s.sql("SELECT * FROM primes(20)").collect() == [
+        (2,), (3,), (5,), (7,), (11,), (13,), (17,), (19,)]
+    assert s.sql("SELECT sum(p) FROM primes(10) WHERE p > 2").collect() == [(15,)]
+    rows = s.sql("SELECT t.p FROM primes(5) t JOIN primes(7) q ON t.p = q.p "
+                 "ORDER BY 1").collect()
+    assert rows == [(2,), (3,), (5,)]

@@ -25,6 +25,7 @@ from .rules.decorrelate import decorrelate
 from .rules.pushdown import pushdown_filters
 from .rules.prune import prune_columns
 from .rules.thin_agg import thin_aggregates
+from .rules.magic_set import semi_filter_aggregates
 from .rules.join_order import reorder_joins
 from .rules.semi_sink import sink_semi_joins
 
@@ -35,6 +36,7 @@ def optimize(plan: S.Plan, enable_join_reorder: bool = True, stats=None) -> S.Pl
     if enable_join_reorder:
         plan = reorder_joins(plan, stats)
     plan = sink_semi_joins(plan)
+    plan = semi_filter_aggregates(plan, stats)
     plan = thin_aggregates(plan, stats)
     plan = prune_columns(plan)
     return plan

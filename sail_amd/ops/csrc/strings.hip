@@ -96,11 +96,25 @@ __global__ void contains_kernel(const int64_t* __restrict__ offsets,
 __global__ void string_hash64_kernel(const int64_t* __restrict__ offsets,
                                      const uint8_t* __restrict__ bytes,
                                      int64_t* __restrict__ out, int64_t n) {
+  // FNV-1a is byte-serial, but the LOADS need not be: pull 8 bytes per
+  // unaligned dword2 load and fold them from the register — 8x fewer
+  // memory ops per row while producing the identical byte-order hash
+  // (must match fnv_key_tensor / the CPU reference bit-for-bit).
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * (int64_t)blockDim.x) {
     int64_t lo = offsets[i], hi = offsets[i + 1];
     uint64_t h = 14695981039346656037ull;
-    for (int64_t j = lo; j < hi; ++j) {
+    int64_t j = lo;
+    for (; j + 8 <= hi; j += 8) {
+      uint64_t w;
+      __builtin_memcpy(&w, bytes + j, 8);  // unaligned load, little-endian
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        h = (h ^ (w & 0xffull)) * 1099511628211ull;
+        w >>= 8;
+      }
+    }
+    for (; j < hi; ++j) {
       h = (h ^ bytes[j]) * 1099511628211ull;
     }
     // mix in length; keep sign bit clear so sentinels (<0) stay distinct

@@ -520,6 +520,13 @@ def _string_cmp(a: Column, b: Column, op: str) -> torch.Tensor:
             if b.dict_size == 1:
                 # b is a broadcast literal
                 target = b.dict_values()[0]
+                if op in ("=", "!="):
+                    code = a.dict_code_of(target)  # O(log n) sorted-dict search
+                    if code < 0:
+                        m = torch.zeros(len(a), dtype=torch.bool, device=a.device)
+                    else:
+                        m = a.codes == code
+                    return m if op == "=" else ~m
                 avals = a.dict_values()
                 if op in ("=", "!="):
                     try:

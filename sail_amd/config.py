@@ -18,9 +18,16 @@ DEFAULTS: Dict[str, str] = {
     "sail.optimizer.enable_join_reorder": "true",
     "sail.optimizer.join_reorder_max_relations": "12",
     "sail.optimizer.agg_mask_min_selectivity": "0.2",
+    "sail.execution.checkpoint_path": "",       # df.checkpoint() root (temp dir when empty)
+    "sail.execution.max_recursion": "100",      # WITH RECURSIVE iteration cap
+    # distributed exchanges (SPMD over RCCL)
+    "sail.exec.broadcast_threshold_bytes": str(2 << 30),  # gathered build side above this -> hash shuffle
+    "sail.exec.agg_shuffle_threshold_groups": "4000000",  # est. groups above this -> shuffled aggregation
     # kernels
     "sail.kernels.require_on_gpu": "true",
     "sail.kernels.grouped_agg_max_lds_groups": "4096",
+    # debugging (the compute-sanitizer analogue)
+    "sail.debug.sync_kernels": "false",         # device sync + error check after every operator
     # tracing (ref: sail-telemetry)
     "sail.trace": "false",
     # spark-compatible session confs

@@ -71,8 +71,11 @@ class Executor:
                 from ..utils.trace import Tracer
 
                 ctx.tracer = Tracer(ctx.device)
-                ctx.session.last_trace = ctx.tracer.trace
         self.tracer = ctx.tracer
+        # §5.2 sanitizer analogue: serialize + error-check every operator so a
+        # faulting kernel is attributed to the operator that launched it
+        self._sync_kernels = (ctx.session.conf.get("sail.debug.sync_kernels")
+                              == "true" and ctx.device.type == "cuda")
 
     def _gather(self, chunk: Chunk) -> Chunk:
         """Replicate a sharded chunk on every rank (all_gather over RCCL)."""
@@ -106,6 +109,12 @@ class Executor:
             out = m(plan)
         if ck is not None:
             self.ctx._cte_cache[ck] = out
+        if self._sync_kernels:
+            try:
+                torch.cuda.synchronize(self.ctx.device)
+            except RuntimeError as e:
+                raise ExecError(
+                    f"device fault inside {type(plan).__name__}: {e}") from e
         return out
 
     # -- leaves ------------------------------------------------------------

@@ -299,6 +299,29 @@ class SessionContext:
             "system_tables", [("tableName", T.STRING), ("rows", T.I64)],
             tables_provider)
 
+        def operators_provider(device):
+            tr = getattr(self, "last_trace", None)
+            ev = tr.events if tr is not None else []
+            return Table.from_pydict(
+                {"operator": [e.op for e in ev],
+                 "detail": [e.detail for e in ev],
+                 "depth": [e.depth for e in ev],
+                 "rows": [e.rows for e in ev],
+                 "self_ms": [round(e.self_ms, 3) for e in ev],
+                 "total_ms": [round(e.ms, 3) for e in ev]},
+                {"operator": T.STRING, "detail": T.STRING, "depth": T.I32,
+                 "rows": T.I64, "self_ms": T.F64, "total_ms": T.F64},
+                device="cpu")
+
+        # per-operator metrics of the last traced query — the reference's
+        # "Spark UI is a system table" design applied to TracingExec output
+        # (ref: sail-catalog-system + sail-telemetry system events)
+        self.catalog.register_provider(
+            "system_operators",
+            [("operator", T.STRING), ("detail", T.STRING), ("depth", T.I32),
+             ("rows", T.I64), ("self_ms", T.F64), ("total_ms", T.F64)],
+            operators_provider)
+
     @property
     def udf(self) -> "UdfRegistry":
         return UdfRegistry(self)
@@ -310,7 +333,12 @@ class SessionContext:
         from .executor import ExecutionContext, Executor
 
         ctx = ExecutionContext(self, self.device)
-        return Executor(ctx).execute(plan)
+        out = Executor(ctx).execute(plan)
+        if getattr(ctx, "tracer", None) is not None:
+            # published AFTER execution so `SELECT * FROM system_operators`
+            # reads the previous query's trace, not its own empty one
+            self.last_trace = ctx.tracer.trace
+        return out
 
     @property
     def read(self) -> "DataFrameReader":

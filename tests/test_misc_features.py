@@ -92,3 +92,12 @@ def test_explain_analyze(s):
     out = s.sql("EXPLAIN ANALYZE SELECT a, count(*) FROM ea GROUP BY a").collect()[0][0]
     assert "Analyzed (wall times)" in out
     assert "Aggregate" in out and "ms self" in out
+
+
+def test_system_operators_table(s):
+    s.conf["sail.trace"] = "true"
+    s.create_dataframe({"a": [1, 2, 3]}, name="sot")
+    s.sql("SELECT a, count(*) FROM sot GROUP BY a").collect()
+    rows = s.sql("SELECT operator FROM system_operators").collect()
+    ops = {r[0] for r in rows}
+    assert "Aggregate" in ops and "Read" in ops

@@ -568,3 +568,135 @@ def fused_agg_batch(aggs, args_list, fmasks, gid: torch.Tensor, ng: int):
                 out_cols.append(Column(a.dtype or c.dtype,
                                        acc.to(c.data.dtype) if c.data.dtype != torch.int64 else acc, v))
     return out_cols
+
+
+DIRECT_MIN_ROWS = 1_000_000
+
+
+def try_direct_aggregate(key_cols: List[Column], aggs, args_list, fmasks):
+    """Direct-address aggregation for dense high-cardinality keys
+    (150M l_orderkey groups in q18): accumulate straight into a span-sized
+    table with index_add and compact once — skips the whole gid machinery
+    (presence bitmap + cumsum + lut gather + representative scatter were 4
+    extra random passes over 600M rows). Key VALUES are reconstructed
+    arithmetically from the slot index, so no representative gather either.
+    Group order equals the packed-value order — identical to the dense
+    group_ids path. Returns (key_cols_out, agg_cols) or None."""
+    if not key_cols:
+        return None
+    n = len(key_cols[0])
+    dev = key_cols[0].device
+    if n < DIRECT_MIN_ROWS:
+        return None
+    for c in key_cols:
+        if c.validity is not None:
+            return None  # null keys: use the generic path
+    for a, args in zip(aggs, args_list):
+        if a.distinct or a.name not in ("sum", "try_sum", "count", "count_if",
+                                        "avg", "try_avg"):
+            return None
+        if args and isinstance(args[0], StringColumn):
+            return None
+    norm = [normalize_key(c) for c in key_cols]
+    mins, spans = [], []
+    total = 1
+    for k in norm:
+        lo = int(k.min().item())
+        hi = int(k.max().item())
+        mins.append(lo)
+        spans.append(hi - lo + 1)
+        total *= spans[-1]
+        if total > max(4 * n, 1 << 22) or total > (1 << 31):
+            return None
+    if total <= 4096:
+        return None  # tiny/LDS grouped_acc kernels handle this better
+    packed = norm[0] - mins[0]
+    for i in range(1, len(norm)):
+        packed = packed * spans[i] + (norm[i] - mins[i])
+
+    cnt = torch.zeros(total, dtype=torch.int64, device=dev)
+    cnt.index_add_(0, packed, torch.ones(n, dtype=torch.int64, device=dev))
+    present = cnt > 0
+    out_idx = torch.nonzero(present, as_tuple=False).flatten()
+
+    # reconstruct key columns from the slot index
+    out_keys: List[Column] = []
+    rem = out_idx
+    for i, c in enumerate(key_cols):
+        stride = 1
+        for s in spans[i + 1:]:
+            stride *= s
+        code = rem // stride + mins[i]
+        rem = rem % stride if i < len(key_cols) - 1 else rem
+        if isinstance(c, StringColumn):
+            out_keys.append(StringColumn(c.offsets, c.bytes_, None,
+                                         code.to(torch.int32), dtype=c.dtype))
+        else:
+            out_keys.append(Column(c.dtype, code.to(c.data.dtype), None))
+
+    agg_cols: List[Column] = []
+    for a, args, fm in zip(aggs, args_list, fmasks):
+        name = a.name
+        c = args[0] if args else None
+        mask = None
+        if c is not None and c.validity is not None:
+            mask = c.valid_mask()
+        if fm is not None:
+            mask = fm if mask is None else (mask & fm)
+        if name in ("count", "count_if"):
+            if name == "count_if":
+                m2 = c.data.to(torch.bool)
+                mask = m2 if mask is None else (mask & m2)
+            if mask is None:
+                agg_cols.append(Column(T.I64, cnt.index_select(0, out_idx)))
+                continue
+            acc = torch.zeros(total, dtype=torch.int64, device=dev)
+            acc.index_add_(0, packed[mask],
+                           torch.ones(int(mask.sum()), dtype=torch.int64, device=dev))
+            agg_cols.append(Column(T.I64, acc.index_select(0, out_idx)))
+            continue
+        # sum / avg
+        data = c.data
+        if data.dtype == torch.bool:
+            data = data.to(torch.int64)
+        is_f = data.dtype.is_floating_point
+        if not is_f:
+            amax = int(data.abs().max().item()) if n else 0
+            if amax and amax > (1 << 62) // max(n, 1):
+                data = data.to(torch.float64)
+                is_f = True
+            elif data.dtype != torch.int64:
+                data = data.to(torch.int64)
+        else:
+            data = data.to(torch.float64)
+        acc = torch.zeros(total, dtype=data.dtype, device=dev)
+        if mask is None:
+            acc.index_add_(0, packed, data)
+            gcnt = cnt
+        else:
+            acc.index_add_(0, packed[mask], data[mask])
+            gcnt = torch.zeros(total, dtype=torch.int64, device=dev)
+            gcnt.index_add_(0, packed[mask],
+                            torch.ones(int(mask.sum()), dtype=torch.int64, device=dev))
+        sums = acc.index_select(0, out_idx)
+        if name in ("avg", "try_avg"):
+            gc = gcnt.index_select(0, out_idx)
+            if is_f and isinstance(c.dtype, T.DecimalType):
+                sums = torch.round(sums).to(torch.int64)
+            agg_cols.append(_avg_result(sums, gc, c.dtype if not is_f or
+                                        isinstance(c.dtype, T.DecimalType) else T.F64,
+                                        a.dtype))
+        else:
+            rt = a.dtype or (c.dtype if isinstance(c.dtype, T.DecimalType) or c.dtype.is_float else T.I64)
+            if is_f and isinstance(c.dtype, T.DecimalType):
+                sums = torch.round(sums).to(torch.int64)
+            if isinstance(c.dtype, T.DecimalType) and isinstance(rt, T.DecimalType) \
+                    and rt.scale != c.dtype.scale:
+                from .eval import _rescale_int
+
+                sums = _rescale_int(sums, c.dtype.scale, rt.scale)
+            if rt.storage is not None and sums.dtype != rt.storage \
+                    and not rt.is_float:
+                sums = sums.to(rt.storage)
+            agg_cols.append(Column(rt, sums, None))
+    return out_keys, agg_cols

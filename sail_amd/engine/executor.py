@@ -398,6 +398,16 @@ class Executor:
             if n == 0:
                 return Chunk(key_cols + [_empty_agg_col(a, dev) for a in p.aggs],
                              [nm for nm, _ in p.schema])
+            args_list = [[broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
+                         for a in p.aggs]
+            fmasks = [self.ev.eval_mask(a.filter, child) if a.filter is not None else None
+                      for a in p.aggs]
+            from .aggregates import fused_agg_batch, try_direct_aggregate
+
+            d = try_direct_aggregate(key_cols, p.aggs, args_list, fmasks)
+            if d is not None:
+                out_keys, agg_cols = d
+                return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
             gid, rep, ng = group_ids(key_cols)
             out_keys = [c.gather(rep) for c in key_cols]
         else:
@@ -407,10 +417,12 @@ class Executor:
                 return Chunk(cols, [nm for nm, _ in p.schema])
             gid, ng = global_ids(n, dev)
             out_keys = []
-        args_list = [[broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
-                     for a in p.aggs]
-        fmasks = [self.ev.eval_mask(a.filter, child) if a.filter is not None else None
-                  for a in p.aggs]
+            args_list = [[broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
+                         for a in p.aggs]
+            fmasks = [self.ev.eval_mask(a.filter, child) if a.filter is not None else None
+                      for a in p.aggs]
+            from .aggregates import fused_agg_batch
+
         from .aggregates import fused_agg_batch
 
         agg_cols = fused_agg_batch(p.aggs, args_list, fmasks, gid, ng)
@@ -680,14 +692,18 @@ class Executor:
             return Chunk([c.slice(0, 0) for c in key_cols]
                          + [_empty_agg_col(a, dev) for a in p.aggs],
                          [nm for nm, _ in p.schema])
-        gid, rep, ng = group_ids(key_cols)
-        out_keys = [c.gather(rep) for c in key_cols]
         args_list = [[broadcast(self.ev.eval(x, child), n, dev) for x in a.args]
                      for a in p.aggs]
         fmasks = [self.ev.eval_mask(a.filter, child) if a.filter is not None else None
                   for a in p.aggs]
-        from .aggregates import fused_agg_batch
+        from .aggregates import fused_agg_batch, try_direct_aggregate
 
+        d = try_direct_aggregate(key_cols, p.aggs, args_list, fmasks)
+        if d is not None:
+            out_keys, agg_cols = d
+            return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
+        gid, rep, ng = group_ids(key_cols)
+        out_keys = [c.gather(rep) for c in key_cols]
         agg_cols = fused_agg_batch(p.aggs, args_list, fmasks, gid, ng)
         if agg_cols is None:
             agg_cols = [agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype)

@@ -270,3 +270,49 @@ def test_unpivot(s):
                   "ORDER BY id, quarter").collect()
     # Spark default excludeNulls: (2, q2, None) is dropped
     assert rows == [(1, "q1", 10), (1, "q2", 20), (2, "q1", 30)]
+
+
+def test_direct_address_aggregate_matches_generic(monkeypatch):
+    """High-cardinality dense-key fast path (q18 shape) vs the generic
+    group_ids path, bit-for-bit."""
+    import sail_amd.engine.aggregates as agg_mod
+
+    s2 = sail_amd.SessionContext(device="cpu")
+    import torch
+
+    n = 200_000
+    g = torch.Generator().manual_seed(5)
+    keys = torch.randint(0, 50_000, (n,), generator=g).tolist()
+    vals = torch.randint(-100, 100, (n,), generator=g).tolist()
+    flt = [v / 7.0 for v in vals]
+    s2.create_dataframe({"k": keys, "v": vals, "f": flt}, name="big")
+    q = ("SELECT k, sum(v), count(*), avg(f), count(v) FILTER (WHERE v > 0) "
+         "FROM big GROUP BY k ORDER BY k")
+    monkeypatch.setattr(agg_mod, "DIRECT_MIN_ROWS", 1 << 60)
+    want = s2.sql(q).collect()
+    monkeypatch.setattr(agg_mod, "DIRECT_MIN_ROWS", 1)
+    got = s2.sql(q).collect()
+    assert len(got) == len(want)
+    for gr, wr in zip(got, want):
+        assert gr[0] == wr[0] and gr[1] == wr[1] and gr[2] == wr[2] and gr[4] == wr[4]
+        assert abs(gr[3] - wr[3]) < 1e-9
+
+
+def test_direct_aggregate_multikey_strings(monkeypatch):
+    import sail_amd.engine.aggregates as agg_mod
+
+    s2 = sail_amd.SessionContext(device="cpu")
+    import torch
+
+    n = 50_000
+    g = torch.Generator().manual_seed(6)
+    k1 = torch.randint(0, 300, (n,), generator=g).tolist()
+    k2 = [["aa", "bb", "cc"][i] for i in torch.randint(0, 3, (n,), generator=g).tolist()]
+    v = torch.randint(0, 10, (n,), generator=g).tolist()
+    s2.create_dataframe({"k1": k1, "k2": k2, "v": v}, name="mk")
+    q = "SELECT k1, k2, sum(v) FROM mk GROUP BY k1, k2 ORDER BY k1, k2"
+    monkeypatch.setattr(agg_mod, "DIRECT_MIN_ROWS", 1 << 60)
+    want = s2.sql(q).collect()
+    monkeypatch.setattr(agg_mod, "DIRECT_MIN_ROWS", 1)
+    got = s2.sql(q).collect()
+    assert got == want

@@ -274,3 +274,33 @@ def test_tablesample_gpu_deterministic():
     a = s.sql("SELECT sum(id) FROM range(100000) TABLESAMPLE (10 PERCENT) REPEATABLE (3) t").collect()
     b = s.sql("SELECT sum(id) FROM range(100000) TABLESAMPLE (10 PERCENT) REPEATABLE (3) t").collect()
     assert a == b and a[0][0] > 0
+
+
+@pytest.mark.gpu
+def test_direct_aggregate_gpu_parity():
+    """q18-shaped dense high-cardinality aggregate: direct-address path vs
+    generic gid path on device."""
+    import sail_amd
+    import sail_amd.engine.aggregates as agg_mod
+    import torch
+
+    s = sail_amd.SessionContext(device="cuda")
+    n = 5_000_000
+    g = torch.Generator().manual_seed(9)
+    keys = torch.randint(0, 1_000_000, (n,), generator=g)
+    vals = torch.randint(-50, 50, (n,), generator=g)
+    from sail_amd.engine.column import Column, Table
+    from sail_amd.engine import types as T
+
+    s.catalog.register_table("dk", Table({
+        "k": Column(T.I64, keys.cuda()), "v": Column(T.I64, vals.cuda())}))
+    q = "SELECT k, sum(v), count(*) FROM dk GROUP BY k ORDER BY k LIMIT 50"
+    old = agg_mod.DIRECT_MIN_ROWS
+    try:
+        agg_mod.DIRECT_MIN_ROWS = 1 << 60
+        want = s.sql(q).collect()
+        agg_mod.DIRECT_MIN_ROWS = 1
+        got = s.sql(q).collect()
+    finally:
+        agg_mod.DIRECT_MIN_ROWS = old
+    assert got == want

@@ -98,6 +98,30 @@ def eval_window(ev: Evaluator, e: S.WindowExpr, chunk: Chunk) -> Column:
             out = g
         else:
             out = Column(src.dtype, g.data, vmask.to(torch.uint8))
+    elif fname == "cume_dist":
+        part_sizes = torch.zeros(int(part_id.max().item()) + 1, dtype=torch.int64, device=dev)
+        part_sizes.index_add_(0, part_id, torch.ones(n, dtype=torch.int64, device=dev))
+        # number of rows <= current row's peer group (last peer position)
+        peer_id = torch.cumsum(newval.to(torch.int64), 0) - 1
+        npeers = int(peer_id.max().item()) + 1
+        last_of_peer = torch.zeros(npeers, dtype=torch.int64, device=dev)
+        last_of_peer.scatter_reduce_(0, peer_id, torch.arange(n, device=dev),
+                                     reduce="amax", include_self=False)
+        upto = last_of_peer[peer_id] - part_start_pos[part_id] + 1
+        out = Column(T.F64, upto.to(torch.float64)
+                     / part_sizes[part_id].to(torch.float64))
+    elif fname == "nth_value":
+        src = broadcast(ev.eval(f.args[0], sorted_chunk), n, dev)
+        k = int(f.args[1].value)
+        tgt = part_start_pos[part_id] + (k - 1)
+        valid = pos_in_part >= (k - 1)
+        g = src.gather(tgt.clamp(0, n - 1))
+        vmask = g.valid_mask() & valid
+        if isinstance(g, StringColumn):
+            g.validity = vmask.to(torch.uint8)
+            out = g
+        else:
+            out = Column(src.dtype, g.data, vmask.to(torch.uint8))
     elif fname == "ntile":
         buckets = int(f.args[0].value)
         part_sizes = torch.zeros(int(part_id.max().item()) + 1, dtype=torch.int64, device=dev)
@@ -116,15 +140,32 @@ def eval_window(ev: Evaluator, e: S.WindowExpr, chunk: Chunk) -> Column:
 
 
 def _window_agg(ev, f: S.AggFunc, sorted_chunk, part_id, pos_in_part, e, n, dev):
-    """sum/count/avg/min/max over whole partition (default frame with no
-    ORDER BY) or running frame (with ORDER BY)."""
+    """sum/count/avg/min/max/first/last over whole-partition, running, or
+    bounded-ROWS frames (prefix-sum sliding windows)."""
     from .aggregates import agg_eval
 
     running = bool(e.order_by) and (e.frame is None or e.frame[1][0] == "unbounded_preceding")
     whole = not e.order_by or (e.frame is not None and e.frame[1][0] == "unbounded_preceding"
                                and e.frame[2][0] == "unbounded_following")
+    bounded = (e.frame is not None and e.frame[0] == "rows"
+               and not whole
+               and e.frame[1][0] in ("preceding", "following", "current", "unbounded_preceding")
+               and e.frame[2][0] in ("preceding", "following", "current"))
     ng = int(part_id.max().item()) + 1
     args = [broadcast(ev.eval(a, sorted_chunk), n, dev) for a in f.args] if f.args else []
+    part_start_pos = torch.nonzero(
+        torch.cat([torch.ones(1, dtype=torch.bool, device=dev),
+                   part_id[1:] != part_id[:-1]]), as_tuple=False).squeeze(1)
+    if bounded and f.name in ("sum", "count", "avg", "first", "first_value",
+                              "last", "last_value"):
+        return _bounded_rows_agg(f, args, part_id, pos_in_part,
+                                 part_start_pos, e.frame, n, dev)
+    if f.name in ("first", "first_value") and running:
+        src = args[0]
+        g = src.gather(part_start_pos[part_id])
+        return g
+    if f.name in ("last", "last_value") and running:
+        return args[0]  # Spark: running frame's last_value IS the current row
     if whole:
         per_group = agg_eval(f.name, args, part_id, ng, f.distinct, None, f.dtype)
         return per_group.gather(part_id)
@@ -152,3 +193,82 @@ def _window_agg(ev, f: S.AggFunc, sorted_chunk, part_id, pos_in_part, e, n, dev)
             res = opped(biased, 0).values - part_id * (big if f.name == "max" else -big)
             return Column(f.dtype, res.to(args[0].data.dtype))
     raise NotImplementedError(f"window frame for {f.name}")
+
+
+def _bounded_rows_agg(f, args, part_id, pos_in_part, part_start_pos, frame, n, dev):
+    """ROWS BETWEEN a AND b sliding window via prefix sums: win = cum[t] -
+    cum[s-1] with s/t clipped to the partition (one pass, no per-row loop)."""
+    sizes = torch.zeros(int(part_id.max().item()) + 1, dtype=torch.int64, device=dev)
+    sizes.index_add_(0, part_id, torch.ones(n, dtype=torch.int64, device=dev))
+    pstart = part_start_pos[part_id]
+    pend = pstart + sizes[part_id] - 1
+    i = torch.arange(n, device=dev)
+
+    def bound_pos(b, default_lo):
+        kind, v = b
+        if kind == "unbounded_preceding":
+            return pstart
+        if kind == "current":
+            return i
+        if kind == "preceding":
+            return i - int(v)
+        if kind == "following":
+            return i + int(v)
+        return pstart if default_lo else pend
+
+    s_pos = torch.maximum(bound_pos(frame[1], True), pstart)
+    t_pos = torch.minimum(bound_pos(frame[2], False), pend)
+    empty = s_pos > t_pos
+    s_pos = torch.minimum(s_pos, pend)
+    t_pos = torch.maximum(t_pos, pstart)
+
+    if f.name in ("first", "first_value"):
+        src = args[0]
+        g = src.gather(s_pos)
+        if empty.any():
+            vm = g.valid_mask() & ~empty
+            if isinstance(g, StringColumn):
+                g.validity = vm.to(torch.uint8)
+                return g
+            return Column(src.dtype, g.data, vm.to(torch.uint8))
+        return g
+    if f.name in ("last", "last_value"):
+        src = args[0]
+        g = src.gather(t_pos)
+        if empty.any():
+            vm = g.valid_mask() & ~empty
+            if isinstance(g, StringColumn):
+                g.validity = vm.to(torch.uint8)
+                return g
+            return Column(src.dtype, g.data, vm.to(torch.uint8))
+        return g
+
+    c = args[0] if args else None
+    valid = c.valid_mask() if c is not None else torch.ones(n, dtype=torch.bool, device=dev)
+    vcnt = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(valid.to(torch.int64), 0, out=vcnt[1:])
+    wcnt = (vcnt[t_pos + 1] - vcnt[s_pos]).clamp_min(0)
+    wcnt = torch.where(empty, torch.zeros_like(wcnt), wcnt)
+    if f.name == "count":
+        return Column(T.I64, wcnt)
+    x = c.data
+    x = x.to(torch.float64) if c.dtype.is_float else x.to(torch.int64)
+    x = torch.where(valid, x, torch.zeros_like(x))
+    cum = torch.zeros(n + 1, dtype=x.dtype, device=dev)
+    torch.cumsum(x, 0, out=cum[1:])
+    wsum = cum[t_pos + 1] - cum[s_pos]
+    vmask = (wcnt > 0)
+    vv = None if bool(vmask.all()) else vmask.to(torch.uint8)
+    if f.name == "avg":
+        from .aggregates import _avg_result
+
+        res = _avg_result(torch.where(vmask, wsum, torch.zeros_like(wsum)), wcnt,
+                          c.dtype, f.dtype)
+        if vv is not None:
+            res = Column(res.dtype, res.data, vv)
+        return res
+    out_t = f.dtype or c.dtype
+    data = wsum
+    if isinstance(c.dtype, T.DecimalType):
+        data = data.to(torch.int64)
+    return Column(out_t, data, vv)

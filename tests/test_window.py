@@ -49,3 +49,33 @@ def test_lag_lead(s):
 def test_ntile(s):
     rows = s.sql("SELECT o, ntile(2) OVER (ORDER BY o) AS nt FROM w WHERE g = 'a' ORDER BY o").collect()
     assert rows == [(1, 1), (2, 1), (3, 2)]
+
+
+def test_bounded_rows_frames(s):
+    s.create_dataframe({"o": [1, 2, 3, 4], "v": [10, 20, 30, 40]}, name="bf")
+    rows = s.sql("SELECT o, sum(v) OVER (ORDER BY o ROWS BETWEEN 1 PRECEDING AND 1 FOLLOWING) "
+                 "FROM bf ORDER BY o").collect()
+    assert rows == [(1, 30), (2, 60), (3, 90), (4, 70)]
+    rows = s.sql("SELECT o, avg(v) OVER (ORDER BY o ROWS BETWEEN 2 PRECEDING AND CURRENT ROW) "
+                 "FROM bf ORDER BY o").collect()
+    assert rows == [(1, 10.0), (2, 15.0), (3, 20.0), (4, 30.0)]
+    # empty trailing frames -> count 0
+    rows = s.sql("SELECT o, count(*) OVER (ORDER BY o ROWS BETWEEN 1 FOLLOWING AND 2 FOLLOWING) "
+                 "FROM bf ORDER BY o").collect()
+    assert rows == [(1, 2), (2, 2), (3, 1), (4, 0)]
+
+
+def test_bounded_frames_with_partitions(s):
+    rows = s.sql("SELECT g, o, sum(v) OVER (PARTITION BY g ORDER BY o "
+                 "ROWS BETWEEN 1 PRECEDING AND CURRENT ROW) FROM w ORDER BY g, o").collect()
+    assert rows == [("a", 1, 10), ("a", 2, 30), ("a", 3, 50), ("b", 1, 5), ("b", 2, 20)]
+
+
+def test_first_last_nth_cume(s):
+    s.create_dataframe({"o": [1, 2, 3, 4], "v": [10, 20, 30, 40]}, name="fl")
+    rows = s.sql("SELECT o, first_value(v) OVER (ORDER BY o DESC), "
+                 "last_value(v) OVER (ORDER BY o), "
+                 "nth_value(v, 2) OVER (ORDER BY o), "
+                 "cume_dist() OVER (ORDER BY o) FROM fl ORDER BY o").collect()
+    assert rows == [(1, 40, 10, None, 0.25), (2, 40, 20, 20, 0.5),
+                    (3, 40, 30, 20, 0.75), (4, 40, 40, 20, 1.0)]

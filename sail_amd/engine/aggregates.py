@@ -388,6 +388,8 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
         gnp = g.cpu().numpy()
         vnp = vals.cpu().numpy()
         q = 0.5
+        if name != "median" and len(args) > 1 and len(args[1]):
+            q = float(args[1].to_pylist()[0])
         import collections
 
         groups = collections.defaultdict(list)
@@ -397,6 +399,191 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
         for k, lst in groups.items():
             res[k] = float(np.percentile(lst, q * 100))
         return Column(T.F64, torch.from_numpy(res).to(dev), None)
+
+    if name == "approx_count_distinct":
+        # exact distinct count (an exact answer satisfies the approx
+        # contract; the reference's HLL sketch is a space tradeoff this
+        # engine doesn't need at whole-partition scale)
+        return agg_eval("count", args, gid, ng, True, filter_mask, T.I64)
+
+    if name in ("corr", "covar_samp", "covar_pop", "regr_count", "regr_avgx",
+                "regr_avgy", "regr_slope", "regr_intercept", "regr_r2",
+                "regr_sxx", "regr_syy", "regr_sxy"):
+        x = args[0]
+        y = args[1]
+        m = mask & y.valid_mask()
+        gm = gid[m]
+        xv = x.data[m].to(torch.float64)
+        yv = y.data[m].to(torch.float64)
+        if isinstance(x.dtype, T.DecimalType):
+            xv = xv / (10.0 ** x.dtype.scale)
+        if isinstance(y.dtype, T.DecimalType):
+            yv = yv / (10.0 ** y.dtype.scale)
+
+        def seg(v):
+            o = torch.zeros(ng, dtype=torch.float64, device=dev)
+            o.index_add_(0, gm, v)
+            return o
+
+        cnt = seg(torch.ones(gm.shape[0], dtype=torch.float64, device=dev))
+        sx, sy = seg(xv), seg(yv)
+        sxx, syy, sxy = seg(xv * xv), seg(yv * yv), seg(xv * yv)
+        cn = cnt.clamp_min(1)
+        mx, my = sx / cn, sy / cn
+        cxx = sxx - cnt * mx * mx
+        cyy = syy - cnt * my * my
+        cxy = sxy - cnt * mx * my
+        if name == "regr_count":
+            return Column(T.I64, cnt.to(torch.int64), None)
+        if name == "regr_avgx":
+            data, valid = mx, cnt > 0
+        elif name == "regr_avgy":
+            data, valid = my, cnt > 0
+        elif name == "regr_sxx":
+            data, valid = cxx, cnt > 0
+        elif name == "regr_syy":
+            data, valid = cyy, cnt > 0
+        elif name == "regr_sxy":
+            data, valid = cxy, cnt > 0
+        elif name == "regr_slope":
+            data, valid = cxy / cxx.clamp_min(1e-300), (cnt > 0) & (cxx > 0)
+        elif name == "regr_intercept":
+            data = my - (cxy / cxx.clamp_min(1e-300)) * mx
+            valid = (cnt > 0) & (cxx > 0)
+        elif name == "regr_r2":
+            data = (cxy * cxy) / (cxx * cyy).clamp_min(1e-300)
+            valid = (cnt > 0) & (cxx > 0) & (cyy > 0)
+        elif name == "corr":
+            data = cxy / torch.sqrt((cxx * cyy).clamp_min(1e-300))
+            valid = (cnt > 1) & (cxx > 0) & (cyy > 0)
+        else:  # covar_samp / covar_pop
+            denom = cnt - (1.0 if name == "covar_samp" else 0.0)
+            data = cxy / denom.clamp_min(1e-300)
+            valid = denom > 0
+        return Column(T.F64, data,
+                      None if bool(valid.all()) else valid.to(torch.uint8))
+
+    if name in ("skewness", "kurtosis"):
+        vals = c.data[mask].to(torch.float64)
+        if isinstance(c.dtype, T.DecimalType):
+            vals = vals / (10.0 ** c.dtype.scale)
+
+        def seg(v):
+            o = torch.zeros(ng, dtype=torch.float64, device=dev)
+            o.index_add_(0, gidm, v)
+            return o
+
+        cnt = seg(torch.ones(n_used, dtype=torch.float64, device=dev))
+        s1 = seg(vals)
+        mu = s1 / cnt.clamp_min(1)
+        d = vals - mu[gidm]
+        m2 = seg(d * d) / cnt.clamp_min(1)
+        m3 = seg(d * d * d) / cnt.clamp_min(1)
+        m4 = seg(d * d * d * d) / cnt.clamp_min(1)
+        if name == "skewness":
+            data = m3 / m2.clamp_min(1e-300) ** 1.5
+            valid = (cnt > 0) & (m2 > 0)
+        else:  # Spark kurtosis: excess kurtosis
+            data = m4 / (m2 * m2).clamp_min(1e-300) - 3.0
+            valid = (cnt > 0) & (m2 > 0)
+        return Column(T.F64, data,
+                      None if bool(valid.all()) else valid.to(torch.uint8))
+
+    if name in ("min_by", "max_by", "mode"):
+        # sort by (gid, order-key) and take the first per group
+        if name == "mode":
+            vkey = normalize_key(c)[mask]
+        else:
+            ordc = args[1]
+            m2 = mask & ordc.valid_mask()
+            if not bool(m2.equal(mask)):
+                mask = m2
+            vkey = normalize_key(ordc)[mask]
+        gm = gid[mask]
+        rows = torch.nonzero(mask, as_tuple=False).flatten()
+        if name == "mode":
+            # count per (gid, value): sort then run lengths, keep max
+            o1 = torch.argsort(vkey, stable=True)
+            o2 = torch.argsort(gm.index_select(0, o1), stable=True)
+            order = o1.index_select(0, o2)
+            gs = gm.index_select(0, order)
+            vs = vkey.index_select(0, order)
+            nn = gs.shape[0]
+            newrun = torch.ones(nn, dtype=torch.bool, device=dev)
+            if nn > 1:
+                newrun[1:] = (gs[1:] != gs[:-1]) | (vs[1:] != vs[:-1])
+            run_id = torch.cumsum(newrun.to(torch.int64), 0) - 1
+            nrun = int(run_id[-1].item()) + 1 if nn else 0
+            rcnt = torch.zeros(nrun, dtype=torch.int64, device=dev)
+            rcnt.index_add_(0, run_id, torch.ones(nn, dtype=torch.int64, device=dev))
+            rgid = gs[newrun]
+            rrow = rows.index_select(0, order)[newrun]
+            best = torch.zeros(ng, dtype=torch.int64, device=dev)
+            best.scatter_reduce_(0, rgid, rcnt, reduce="amax", include_self=True)
+            pick = rcnt == best[rgid]
+            # first winning run per group (large init: amin with include_self)
+            sel_row = torch.full((ng,), 1 << 62, dtype=torch.int64, device=dev)
+            cand = torch.nonzero(pick, as_tuple=False).flatten()
+            sel_row.scatter_reduce_(0, rgid[cand], rrow[cand], reduce="amin",
+                                    include_self=True)
+            got = c.gather(sel_row.clamp(0, max(len(c) - 1, 0)))
+            return got
+        desc = name == "max_by"
+        o1 = torch.argsort(vkey, stable=True, descending=desc)
+        o2 = torch.argsort(gm.index_select(0, o1), stable=True)
+        order = o1.index_select(0, o2)
+        gs = gm.index_select(0, order)
+        nn = gs.shape[0]
+        first = torch.ones(nn, dtype=torch.bool, device=dev)
+        if nn > 1:
+            first[1:] = gs[1:] != gs[:-1]
+        sel = rows.index_select(0, order)[first]
+        # sel is ordered by group id; groups with no rows keep row 0 (masked
+        # to null below)
+        out_rows = torch.zeros(ng, dtype=torch.int64, device=dev)
+        out_rows[gs[first]] = sel
+        present = torch.zeros(ng, dtype=torch.bool, device=dev)
+        present[gs[first]] = True
+        got = c.gather(out_rows)
+        vm = got.valid_mask() & present
+        v = None if bool(vm.all()) else vm.to(torch.uint8)
+        if isinstance(got, StringColumn):
+            got.validity = v
+            return got
+        return Column(got.dtype, got.data, v)
+
+    if name in ("bit_and", "bit_or", "bit_xor"):
+        vals = c.data[mask].to(torch.int64)
+        gm = gid[mask]
+        if name == "bit_xor":
+            # xor is addition mod 2 per bit: sum each bit's parity
+            out = torch.zeros(ng, dtype=torch.int64, device=dev)
+            for b in range(64):
+                bit = (vals >> b) & 1
+                acc = torch.zeros(ng, dtype=torch.int64, device=dev)
+                acc.index_add_(0, gm, bit)
+                out |= (acc & 1) << b
+            return Column(T.I64, out, None)
+        # and/or per bit via any/all (amax/amin of the bit)
+        out = torch.zeros(ng, dtype=torch.int64, device=dev)
+        red = "amin" if name == "bit_and" else "amax"
+        for b in range(64):
+            bit = (vals >> b) & 1
+            acc = torch.full((ng,), 1 if name == "bit_and" else 0,
+                             dtype=torch.int64, device=dev)
+            acc.scatter_reduce_(0, gm, bit, reduce=red, include_self=True)
+            out |= (acc & 1) << b
+        return Column(T.I64, out, None)
+
+    if name in ("string_agg", "listagg"):
+        lst = agg_eval("collect_list", [c], gid, ng, False, filter_mask, None)
+        sep = ","
+        if len(args) > 1:
+            sv = args[1]
+            sep = sv.to_pylist()[0] if len(sv) else ","
+        joined = [None if v is None or not v else sep.join(str(x) for x in v)
+                  for v in lst.to_pylist()]
+        return StringColumn.from_pylist(joined, device=dev)
 
     raise NotImplementedError(f"aggregate {name}")
 

@@ -316,3 +316,31 @@ def test_direct_aggregate_multikey_strings(monkeypatch):
     monkeypatch.setattr(agg_mod, "DIRECT_MIN_ROWS", 1)
     got = s2.sql(q).collect()
     assert got == want
+
+
+def test_extended_aggregates(s):
+    s2 = sail_amd.SessionContext(device="cpu")
+    s2.create_dataframe({"g": ["a", "a", "a", "b"], "v": [1.0, 2.0, 10.0, 5.0],
+                         "w": [2.0, 4.0, 20.0, 1.0], "c": ["x", "y", "z", "q"],
+                         "i": [12, 10, 6, 7]}, name="ea2")
+    assert s2.sql("SELECT approx_count_distinct(v) FROM ea2").collect() == [(4,)]
+    r = s2.sql("SELECT corr(v, w), covar_samp(v, w), covar_pop(v, w) FROM ea2").collect()[0]
+    assert abs(r[0] - 0.86445398) < 1e-6 and abs(r[1] - 31.1666667) < 1e-5
+    assert s2.sql("SELECT g, min_by(c, v), max_by(c, v) FROM ea2 GROUP BY g "
+                  "ORDER BY g").collect() == [("a", "x", "z"), ("b", "q", "q")]
+    assert s2.sql("SELECT bit_and(i), bit_or(i), bit_xor(i) FROM ea2").collect() == [(0, 15, 7)]
+    assert s2.sql("SELECT string_agg(c, '-') FROM ea2").collect() == [("x-y-z-q",)]
+    sk, ku = s2.sql("SELECT skewness(v), kurtosis(v) FROM ea2").collect()[0]
+    assert abs(sk - 0.62973761) < 1e-6
+    assert s2.sql("SELECT percentile(v, 0.25) FROM ea2").collect() == [(1.75,)]
+    s2.create_dataframe({"m": [1, 1, 2, 2, 2, 3]}, name="mm2")
+    assert s2.sql("SELECT mode(m) FROM mm2").collect() == [(2,)]
+
+
+def test_regr_aggregates(s):
+    s2 = sail_amd.SessionContext(device="cpu")
+    s2.create_dataframe({"x": [1.0, 2.0, 3.0, 4.0], "y": [2.1, 3.9, 6.1, 8.0]}, name="rg")
+    slope, icept, r2, n = s2.sql(
+        "SELECT regr_slope(y, x), regr_intercept(y, x), regr_r2(y, x), "
+        "regr_count(y, x) FROM rg").collect()[0]
+    assert abs(slope - 1.99) < 0.02 and n == 4 and r2 > 0.99

@@ -409,9 +409,10 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
     if name in ("corr", "covar_samp", "covar_pop", "regr_count", "regr_avgx",
                 "regr_avgy", "regr_slope", "regr_intercept", "regr_r2",
                 "regr_sxx", "regr_syy", "regr_sxy"):
-        x = args[0]
-        y = args[1]
-        m = mask & y.valid_mask()
+        # Spark argument order: f(y, x) — args[0] is the DEPENDENT variable
+        y = args[0]
+        x = args[1]
+        m = mask & x.valid_mask()
         gm = gid[m]
         xv = x.data[m].to(torch.float64)
         yv = y.data[m].to(torch.float64)

@@ -911,6 +911,65 @@ class Executor:
         return Chunk([StringColumn.from_pylist([p.key]), StringColumn.from_pylist([val])],
                      ["key", "value"])
 
+    def _x_ShowFunctions(self, p: S.ShowFunctions) -> Chunk:
+        from ..functions.registry import AGG_FUNCTIONS, SCALAR_RETURN, WINDOW_FUNCTIONS
+        from .aggregates import UDAFS
+
+        names = sorted(set(SCALAR_RETURN) | AGG_FUNCTIONS | WINDOW_FUNCTIONS
+                       | set(self.ctx.session.udfs) | set(UDAFS))
+        if p.pattern:
+            import fnmatch
+
+            pat = p.pattern.strip("'\"")
+            names = [n for n in names if fnmatch.fnmatch(n, pat.replace("%", "*"))]
+        return Chunk([StringColumn.from_pylist(names, dict_encode=False)], ["function"])
+
+    def _x_ShowDatabases(self, p: S.ShowDatabases) -> Chunk:
+        return Chunk([StringColumn.from_pylist(["default"], dict_encode=False)],
+                     ["namespace"])
+
+    def _x_CacheTable(self, p: S.CacheTable) -> Chunk:
+        chunk = self.execute(p.input)
+        cat = self.ctx.session.catalog
+        # a same-name view would shadow the materialized table: stash it so
+        # UNCACHE can restore the logical definition
+        key = cat._key(p.name)
+        vp = cat._views.pop(key, None)
+        if vp is not None:
+            cat.__dict__.setdefault("_cached_views", {})[key] = vp
+        cat.register_table(p.name, chunk.to_table(),
+                           [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)])
+        return Chunk([StringColumn.from_pylist([f"cached {p.name}"], dict_encode=False)],
+                     ["result"])
+
+    def _x_UncacheTable(self, p: S.UncacheTable) -> Chunk:
+        cat = self.ctx.session.catalog
+        key = cat._key(p.name)
+        cat._tables.pop(key, None)
+        cat._schemas.pop(key, None)
+        stash = cat.__dict__.get("_cached_views", {})
+        if key in stash:
+            cat._views[key] = stash.pop(key)
+        return Chunk([StringColumn.from_pylist([f"uncached {p.name}"], dict_encode=False)],
+                     ["result"])
+
+    def _x_AnalyzeTable(self, p: S.AnalyzeTable) -> Chunk:
+        cat = self.ctx.session.catalog
+        sch = cat.table_schema(p.name) or []
+        cols = p.columns if p.columns else [n for n, _ in sch]
+        if p.columns == []:
+            cols = [n for n, _ in sch]
+        names, rows_l, ndv_l = [], [], []
+        for cname in cols:
+            st = cat.column_stats(p.name, cname)
+            names.append(cname)
+            rows_l.append(st[0] if st else -1)
+            ndv_l.append(st[1] if st and st[1] is not None else -1)
+        return Chunk([StringColumn.from_pylist(names, dict_encode=False),
+                      Column.from_values(rows_l, T.I64),
+                      Column.from_values(ndv_l, T.I64)],
+                     ["column", "rows", "ndv"])
+
     def _x_ShowTables(self, p: S.ShowTables) -> Chunk:
         names = self.ctx.session.catalog.list_tables()
         return Chunk([StringColumn.from_pylist([""] * len(names), dict_encode=False),

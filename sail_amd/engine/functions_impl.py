@@ -357,6 +357,25 @@ def _f_coalesce(args, out, chunk, ev):
         cols.append(cast_value(a, out, chunk))
     if not cols:
         return Scalar(None, out)
+    if isinstance(out, T.StringType):
+        # pick-per-row across string columns: codes are chunk-local, so
+        # select via a union concat + gather (device-native)
+        from .column import StringColumn
+        from .executor import _concat_strings
+
+        scols = [_col(c, chunk) for c in cols]
+        n = chunk.num_rows
+        dev = chunk.device
+        sel = torch.full((n,), -1, dtype=torch.int64, device=dev)
+        for j, c in enumerate(scols):
+            sel = torch.where((sel < 0) & c.valid_mask(),
+                              torch.full_like(sel, j), sel)
+        valid = sel >= 0
+        allc = _concat_strings(scols)
+        idx = sel.clamp_min(0) * n + torch.arange(n, device=dev)
+        got = allc.gather(idx)
+        got.validity = None if bool(valid.all()) else valid.to(torch.uint8)
+        return got
     first = _col(cols[0], chunk)
     data = first.data.clone()
     valid = first.valid_mask().clone()
@@ -1049,3 +1068,95 @@ _IMPLS.update(_ARRAY_IMPLS)
 _IMPLS["element_at"] = _ARRAY_IMPLS["element_at"]
 _IMPLS["element_at_sql"] = _ARRAY_IMPLS["element_at_sql"]
 _IMPLS["try_element_at"] = _ARRAY_IMPLS["try_element_at"]
+
+
+# -- JSON functions (host eval, dictionary-aware — ref: sail-function
+#    src/scalar/json) ---------------------------------------------------
+
+def _json_path_get(doc: str, path: str):
+    import json as _json
+
+    if doc is None:
+        return None
+    try:
+        obj = _json.loads(doc)
+    except (ValueError, TypeError):
+        return None
+    if not path.startswith("$"):
+        return None
+    # $.a.b[0].c subset of JSONPath
+    import re as _re
+
+    for part in _re.findall(r"\.([A-Za-z_][A-Za-z0-9_]*)|\[(\d+)\]", path[1:]):
+        key, idx = part
+        if key:
+            if not isinstance(obj, dict) or key not in obj:
+                return None
+            obj = obj[key]
+        else:
+            i = int(idx)
+            if not isinstance(obj, list) or i >= len(obj):
+                return None
+            obj = obj[i]
+    if obj is None:
+        return None
+    if isinstance(obj, (dict, list)):
+        import json as _json
+
+        return _json.dumps(obj, separators=(",", ":"))
+    if isinstance(obj, bool):
+        return "true" if obj else "false"
+    return str(obj)
+
+
+def _f_get_json_object(args, out, chunk, ev):
+    path = _scalarize(args[1]).value
+    return _dict_transform(lambda v: _json_path_get(v, path))(args[:1], out, chunk, ev)
+
+
+def _f_json_tuple_field(args, out, chunk, ev):
+    # json_tuple is exposed as get_json_object('$.<field>') per field
+    field = _scalarize(args[1]).value
+    return _dict_transform(lambda v: _json_path_get(v, f"$.{field}"))(args[:1], out, chunk, ev)
+
+
+def _f_to_json(args, out, chunk, ev):
+    import json as _json
+
+    c = _col(args[0], chunk)
+    vals = c.to_pylist()
+    return StringColumn.from_pylist(
+        [None if v is None else _json.dumps(v, separators=(",", ":"), default=str)
+         for v in vals], device=chunk.device)
+
+
+def _f_schema_of_json(args, out, chunk, ev):
+    import json as _json
+
+    from .eval import Scalar
+
+    doc = _scalarize(args[0]).value
+
+    def tname(v):
+        if isinstance(v, bool):
+            return "BOOLEAN"
+        if isinstance(v, int):
+            return "BIGINT"
+        if isinstance(v, float):
+            return "DOUBLE"
+        if isinstance(v, list):
+            return f"ARRAY<{tname(v[0]) if v else 'STRING'}>"
+        if isinstance(v, dict):
+            inner = ", ".join(f"{k}: {tname(x)}" for k, x in v.items())
+            return f"STRUCT<{inner}>"
+        return "STRING"
+
+    try:
+        return Scalar(tname(_json.loads(doc)), T.STRING)
+    except (ValueError, TypeError):
+        return Scalar(None, T.STRING)
+
+
+_IMPLS["get_json_object"] = _f_get_json_object
+_IMPLS["to_json"] = _f_to_json
+_IMPLS["schema_of_json"] = _f_schema_of_json

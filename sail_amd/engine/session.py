@@ -519,6 +519,64 @@ class DataFrame:
     def explain(self) -> str:
         return S.plan_tree_string(self.plan)
 
+    # -- pandas-style conveniences (ref: spec StatSummary/FillNa/DropNa/
+    # Replace QueryNodes backing df.describe()/fillna()/dropna()) ---------
+    def _as_view(self) -> str:
+        name = f"__df_{id(self) & 0xffffff:x}"
+        self.session.catalog.create_view(name, self.plan, replace=True)
+        return name
+
+    def describe(self, *cols) -> "DataFrame":
+        """count/mean/stddev/min/max per numeric column (Spark df.describe)."""
+        name = self._as_view()
+        targets = list(cols) or [n for n, t in self.plan.schema
+                                 if t.is_numeric]
+        parts = []
+        for stat, fn in [("count", "count({c})"), ("mean", "avg({c})"),
+                         ("stddev", "stddev({c})"), ("min", "min({c})"),
+                         ("max", "max({c})")]:
+            exprs = ", ".join(
+                f"cast({fn.format(c=c)} as string) AS {c}" for c in targets)
+            parts.append(f"SELECT '{stat}' AS summary, {exprs} FROM {name}")
+        return self.session.sql(" UNION ALL ".join(parts))
+
+    summary = describe
+
+    def fillna(self, value, subset=None) -> "DataFrame":
+        name = self._as_view()
+        subset = set(s.lower() for s in subset) if subset else None
+        exprs = []
+        for n, t in self.plan.schema:
+            fill = value
+            applies = (subset is None or n.lower() in subset)
+            if applies and (t.is_numeric and isinstance(value, (int, float))
+                            or (isinstance(t, type(T.STRING)) and isinstance(value, str))):
+                v = repr(value) if isinstance(value, str) else str(value)
+                exprs.append(f"coalesce({n}, {v}) AS {n}")
+            else:
+                exprs.append(n)
+        return self.session.sql(f"SELECT {', '.join(exprs)} FROM {name}")
+
+    def dropna(self, how: str = "any", subset=None) -> "DataFrame":
+        name = self._as_view()
+        cols = subset or [n for n, _ in self.plan.schema]
+        op = " OR " if how == "any" else " AND "
+        cond = op.join(f"{c} IS NULL" for c in cols)
+        return self.session.sql(f"SELECT * FROM {name} WHERE NOT ({cond})")
+
+    def replace(self, to_replace, value, subset=None) -> "DataFrame":
+        name = self._as_view()
+        subset = set(s.lower() for s in subset) if subset else None
+        lit = (lambda v: repr(v) if isinstance(v, str) else str(v))
+        exprs = []
+        for n, t in self.plan.schema:
+            if subset is None or n.lower() in subset:
+                exprs.append(f"CASE WHEN {n} = {lit(to_replace)} "
+                             f"THEN {lit(value)} ELSE {n} END AS {n}")
+            else:
+                exprs.append(n)
+        return self.session.sql(f"SELECT {', '.join(exprs)} FROM {name}")
+
     def checkpoint(self, eager: bool = True) -> "DataFrame":
         """Persist this DataFrame's result and return a DataFrame reading the
         persisted copy — truncates the lineage like Spark's df.checkpoint()

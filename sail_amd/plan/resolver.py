@@ -866,6 +866,34 @@ class Resolver:
         p.schema = [("namespace", T.STRING), ("tableName", T.STRING), ("isTemporary", T.BOOL)]
         return p
 
+    def _p_ShowFunctions(self, p: S.ShowFunctions, outer):
+        p.schema = [("function", T.STRING)]
+        return p
+
+    def _p_ShowDatabases(self, p: S.ShowDatabases, outer):
+        p.schema = [("namespace", T.STRING)]
+        return p
+
+    def _p_CacheTable(self, p: S.CacheTable, outer):
+        inp = self._plan(p.input, outer) if p.input is not None else None
+        if inp is None:
+            # cache an existing view/table by name
+            inp = self._plan(S.Read(table=p.name), outer)
+        out = S.CacheTable(name=p.name, input=inp)
+        out.schema = [("result", T.STRING)]
+        return out
+
+    def _p_UncacheTable(self, p: S.UncacheTable, outer):
+        p.schema = [("result", T.STRING)]
+        return p
+
+    def _p_AnalyzeTable(self, p: S.AnalyzeTable, outer):
+        if self.catalog.table_schema(p.name) is None \
+                and getattr(self.catalog, "view_plan", lambda n: None)(p.name) is None:
+            raise ResolutionError(f"table not found: {p.name}")
+        p.schema = [("column", T.STRING), ("rows", T.I64), ("ndv", T.I64)]
+        return p
+
     def _p_DescribeTable(self, p: S.DescribeTable, outer):
         p.schema = [("col_name", T.STRING), ("data_type", T.STRING), ("comment", T.STRING)]
         return p

@@ -687,6 +687,39 @@ class SetConfig(Command):
 
 
 @dataclass
+class ShowFunctions(Command):
+    pattern: Optional[str] = None
+
+
+@dataclass
+class ShowDatabases(Command):
+    pass
+
+
+@dataclass
+class CacheTable(Command):
+    """CACHE TABLE name [AS query]: materialize a view/query as a table
+    (ref: spec CommandNode cache/uncache)."""
+
+    name: str = ""
+    input: Optional[Plan] = None
+
+
+@dataclass
+class UncacheTable(Command):
+    name: str = ""
+
+
+@dataclass
+class AnalyzeTable(Command):
+    """ANALYZE TABLE ... COMPUTE STATISTICS [FOR COLUMNS ...]: populate the
+    planner's column statistics (ref: spec AnalyzeTable)."""
+
+    name: str = ""
+    columns: Optional[List[str]] = None
+
+
+@dataclass
 class ShowTables(Command):
     pattern: Optional[str] = None
     schema: Optional[List[Tuple[str, T.DataType]]] = None

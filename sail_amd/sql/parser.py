@@ -120,11 +120,50 @@ class Parser:
             return self._parse_delete()
         if self.at_kw("SHOW"):
             self.next()
+            if self.eat_kw("FUNCTIONS"):
+                pattern = None
+                if self.eat_kw("LIKE"):
+                    pattern = self.next().value
+                return S.ShowFunctions(pattern=pattern)
+            if self.eat_kw("DATABASES") or self.eat_kw("SCHEMAS"):
+                return S.ShowDatabases()
             self.expect_kw("TABLES")
+            self.eat_kw("IN") and self.ident()
             pattern = None
             if self.eat_kw("LIKE"):
                 pattern = self.next().value
             return S.ShowTables(pattern=pattern)
+        if self.at_kw("CACHE"):
+            self.next()
+            self.eat_kw("LAZY")
+            self.expect_kw("TABLE")
+            name = self._qualified_name()
+            inp = None
+            if self.eat_kw("AS"):
+                inp = self.parse_query()
+            return S.CacheTable(name=name, input=inp)
+        if self.at_kw("UNCACHE"):
+            self.next()
+            self.expect_kw("TABLE")
+            self.eat_kw("IF") and self.expect_kw("EXISTS")
+            return S.UncacheTable(name=self._qualified_name())
+        if self.at_kw("ANALYZE"):
+            self.next()
+            self.expect_kw("TABLE")
+            name = self._qualified_name()
+            self.expect_kw("COMPUTE")
+            self.expect_kw("STATISTICS")
+            cols = None
+            if self.eat_kw("FOR"):
+                if self.eat_kw("ALL"):
+                    self.expect_kw("COLUMNS")
+                    cols = []
+                else:
+                    self.expect_kw("COLUMNS")
+                    cols = [self.ident()]
+                    while self.eat_op(","):
+                        cols.append(self.ident())
+            return S.AnalyzeTable(name=name, columns=cols)
         if self.at_kw("DESCRIBE", "DESC"):
             self.next()
             self.eat_kw("TABLE")

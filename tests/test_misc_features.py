@@ -101,3 +101,47 @@ def test_system_operators_table(s):
     rows = s.sql("SELECT operator FROM system_operators").collect()
     ops = {r[0] for r in rows}
     assert "Aggregate" in ops and "Read" in ops
+
+
+def test_show_cache_analyze_commands(s):
+    s.create_dataframe({"a": [1, 2, 2]}, name="cat_t")
+    fns = s.sql("SHOW FUNCTIONS").collect()
+    assert len(fns) > 250
+    assert s.sql("SHOW FUNCTIONS LIKE 'array_c%'").collect()[0][0].startswith("array_c")
+    assert s.sql("SHOW DATABASES").collect() == [("default",)]
+    s.sql("CREATE OR REPLACE TEMP VIEW cv AS SELECT a * 10 AS x FROM cat_t")
+    s.sql("CACHE TABLE cv")
+    s.create_dataframe({"a": [100]}, name="cat_t")
+    assert s.sql("SELECT sum(x) FROM cv").collect() == [(50,)]  # frozen
+    s.sql("UNCACHE TABLE cv")
+    assert s.sql("SELECT sum(x) FROM cv").collect() == [(1000,)]  # live view again
+    rows = s.sql("ANALYZE TABLE cat_t COMPUTE STATISTICS FOR ALL COLUMNS").collect()
+    assert rows[0][0] == "a" and rows[0][1] == 1
+
+
+def test_dataframe_conveniences(s):
+    df = s.create_dataframe({"a": [1, 2, None, 4], "b": ["x", None, "y", "x"]})
+    desc = dict((r[0], r[1]) for r in df.describe().collect())
+    assert desc["count"] == "3" and desc["min"] == "1"
+    assert df.fillna(0).collect()[2][0] == 0
+    assert df.fillna("zz").collect()[1][1] == "zz"
+    assert len(df.dropna().collect()) == 2
+    assert df.replace("x", "XX", subset=["b"]).collect()[0][1] == "XX"
+
+
+def test_json_functions(s):
+    s.create_dataframe({"j": ['{"a": 1, "b": {"c": [10, 20]}}', '{"a": "x"}', None]},
+                       name="jt")
+    rows = s.sql("SELECT get_json_object(j, '$.a'), get_json_object(j, '$.b.c[1]') "
+                 "FROM jt").collect()
+    assert rows == [("1", "20"), ("x", None), (None, None)]
+    assert s.sql("SELECT to_json(named_struct('p', 1, 's', 'q'))").collect() == [
+        ('{"p":1,"s":"q"}',)]
+    assert s.sql("SELECT to_json(map('k', 5)), to_json(array(1,2))").collect() == [
+        ('{"k":5}', "[1,2]")]
+    assert "STRUCT<" in s.sql("SELECT schema_of_json('{\"a\": 1}')").collect()[0][0]
+
+
+def test_string_coalesce(s):
+    s.create_dataframe({"b": ["x", None], "c": [None, "q"]}, name="sc")
+    assert s.sql("SELECT coalesce(b, c, 'd') FROM sc").collect() == [("x",), ("q",)]

@@ -1160,3 +1160,39 @@ def _f_schema_of_json(args, out, chunk, ev):
 _IMPLS["get_json_object"] = _f_get_json_object
 _IMPLS["to_json"] = _f_to_json
 _IMPLS["schema_of_json"] = _f_schema_of_json
+
+
+def _f_from_json(args, out, chunk, ev):
+    """from_json(col, 'a INT, b STRING') -> struct (host JSON parse)."""
+    import json as _json
+
+    from .column import StructColumn
+
+    c = _col(args[0], chunk)
+    docs = c.to_pylist()
+    parsed = []
+    for d in docs:
+        if d is None:
+            parsed.append(None)
+            continue
+        try:
+            v = _json.loads(d)
+            parsed.append(v if isinstance(v, dict) else None)
+        except (ValueError, TypeError):
+            parsed.append(None)
+    kids = []
+    for f in out.fields:
+        vals = [None if p is None else p.get(f.name) for p in parsed]
+        # stringify nested values for STRING fields
+        if isinstance(f.dtype, T.StringType):
+            vals = [None if v is None else
+                    (v if isinstance(v, str) else _json.dumps(v)) for v in vals]
+        kids.append((f.name, Column.from_values(vals, f.dtype, device=chunk.device)))
+    validity = None
+    if any(p is None for p in parsed):
+        validity = torch.tensor([0 if p is None else 1 for p in parsed],
+                                dtype=torch.uint8, device=chunk.device)
+    return StructColumn(kids, validity, dtype=out)
+
+
+_IMPLS["from_json"] = _f_from_json

@@ -214,6 +214,7 @@ _reg("map_values", lambda a: T.ArrayType(a[0].value) if isinstance(a[0], T.MapTy
 _reg("map_contains_key", _bool)
 _reg("struct named_struct get_field", lambda a: T.NULL)  # typed structurally in the resolver
 _reg("get_json_object to_json schema_of_json json_tuple", _string)
+_reg("from_json", lambda a: T.NULL)  # typed structurally in the resolver
 _reg("sort_array array_sort array_distinct array_remove array_compact flatten "
      "slice array_repeat shuffle", _same)
 _reg("array_join", _string)

@@ -1143,6 +1143,13 @@ class Resolver:
                         raise ResolutionError("named_struct expects literal field names")
                     fields.append(T.StructField(str(k.value), e.args[i + 1].dtype))
                 t = T.StructType(tuple(fields))
+            elif e.name == "from_json":
+                if len(e.args) < 2 or not isinstance(e.args[1], S.Literal):
+                    raise ResolutionError("from_json expects a literal schema string")
+                from ..sql.parser import parse_ddl_schema
+
+                fields = parse_ddl_schema(str(e.args[1].value))
+                t = T.StructType(tuple(T.StructField(n, ft) for n, ft in fields))
             elif e.name == "get_field":
                 st = e.args[0].dtype
                 fname = e.args[1].value if isinstance(e.args[1], S.Literal) else None

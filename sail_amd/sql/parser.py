@@ -1354,6 +1354,21 @@ def _parse_timestamp_us(s: str) -> int:
     return int((base.total_seconds() + hh * 3600 + mm * 60 + ss) * 1_000_000)
 
 
+def parse_ddl_schema(text: str):
+    """'a INT, b STRING' -> [(name, DataType)] (AnalyzePlan DdlParse and
+    from_json schema strings)."""
+    pr = Parser(text)
+    out = []
+    while pr.peek().kind != "eof":
+        name = pr.ident()
+        pr.eat_op(":")
+        t = pr._parse_type()
+        out.append((name, t))
+        if not pr.eat_op(","):
+            break
+    return out
+
+
 def parse_sql(sql: str) -> S.Plan:
     """Parse a single SQL statement into the spec IR."""
     stmts = Parser(sql).parse_statements()

@@ -145,3 +145,11 @@ def test_json_functions(s):
 def test_string_coalesce(s):
     s.create_dataframe({"b": ["x", None], "c": [None, "q"]}, name="sc")
     assert s.sql("SELECT coalesce(b, c, 'd') FROM sc").collect() == [("x",), ("q",)]
+
+
+def test_from_json(s):
+    s.create_dataframe({"j": ['{"a": 5, "b": "hi"}', "bad", None]}, name="fj")
+    rows = s.sql("SELECT from_json(j, 'a INT, b STRING') FROM fj").collect()
+    assert rows == [({"a": 5, "b": "hi"},), (None,), (None,)]
+    rows = s.sql("SELECT from_json(j, 'a INT, b STRING').a + 1 FROM fj").collect()
+    assert rows == [(6,), (None,), (None,)]

@@ -176,6 +176,56 @@ int64_t table_size_for(int64_t n) {
   return sz;
 }
 
+// ------------------------------------------------------------------
+// hash GROUP-ID assignment: the sort-free replacement for sparse-domain
+// group_ids (unique-sort + searchsorted + representative scatter were
+// ~600 ms of ClickBench's group-by-URL/WatchID queries). Claim pass CASes
+// each distinct key into the table recording a representative row; the
+// scan pass hands out dense ids in slot order; gid is then a plain gather.
+// Group NUMBERING is arbitrary (hash aggregation is unordered — ORDER BY
+// decides the final order, as on every other path).
+// ------------------------------------------------------------------
+__global__ void hg_claim_kernel(const int64_t* __restrict__ keys, int64_t n,
+                                int64_t* __restrict__ tkeys,
+                                int64_t* __restrict__ trow, int64_t tmask,
+                                int64_t* __restrict__ slot_of_row) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    long long k = keys[i];
+    if (k == kEmpty) k = kEmpty + 1;
+    uint64_t h = mix64((uint64_t)k) & tmask;
+    while (true) {
+      long long prev = atomicCAS((unsigned long long*)&tkeys[h],
+                                 (unsigned long long)kEmpty, (unsigned long long)k);
+      if (prev == kEmpty) {
+        trow[h] = i;  // claim winner records the representative row
+        slot_of_row[i] = h;
+        break;
+      }
+      if (prev == k) {
+        slot_of_row[i] = h;
+        break;
+      }
+      h = (h + 1) & tmask;
+    }
+  }
+}
+
+__global__ void hg_scan_kernel(const int64_t* __restrict__ tkeys,
+                               const int64_t* __restrict__ trow, int64_t sz,
+                               int32_t* __restrict__ slot_ids,
+                               int64_t* __restrict__ rep,
+                               int* __restrict__ counter) {
+  for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; s < sz;
+       s += gridDim.x * (int64_t)blockDim.x) {
+    if (tkeys[s] != (long long)kEmpty) {
+      int id = atomicAdd(counter, 1);
+      slot_ids[s] = id;
+      rep[id] = trow[s];
+    }
+  }
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> hj_build(torch::Tensor keys) {
@@ -255,4 +305,31 @@ std::vector<torch::Tensor> hj_probe_fill(torch::Tensor tkeys, torch::Tensor thea
                        out_build.data_ptr<int64_t>());
   }
   return {out_probe, out_build};
+}
+
+
+std::vector<torch::Tensor> hg_group(torch::Tensor keys) {
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
+  int64_t n = keys.numel();
+  int64_t sz = table_size_for(std::max<int64_t>(n, 1));
+  auto tkeys = torch::full({sz}, kEmpty, keys.options());
+  auto trow = torch::empty({sz}, keys.options());
+  auto slot_of_row = torch::empty({std::max<int64_t>(n, 1)}, keys.options());
+  auto slot_ids = torch::empty({sz}, keys.options().dtype(torch::kInt32));
+  auto rep = torch::empty({std::max<int64_t>(n, 1)}, keys.options());
+  auto counter = torch::zeros({1}, keys.options().dtype(torch::kInt32));
+  if (n) {
+    hipStream_t stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(hg_claim_kernel, dim3(grid_for(n)), dim3(kBlock), 0, stream,
+                       keys.data_ptr<int64_t>(), n, tkeys.data_ptr<int64_t>(),
+                       trow.data_ptr<int64_t>(), sz - 1,
+                       slot_of_row.data_ptr<int64_t>());
+    hipLaunchKernelGGL(hg_scan_kernel, dim3(grid_for(sz)), dim3(kBlock), 0, stream,
+                       tkeys.data_ptr<int64_t>(), trow.data_ptr<int64_t>(), sz,
+                       slot_ids.data_ptr<int32_t>(), rep.data_ptr<int64_t>(),
+                       counter.data_ptr<int32_t>());
+  }
+  // gid = slot_ids[slot_of_row]; ng = counter (host sync on read)
+  auto gid = slot_ids.index_select(0, slot_of_row.slice(0, 0, n)).to(torch::kInt64);
+  return {gid, rep, counter};
 }

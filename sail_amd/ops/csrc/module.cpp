@@ -19,6 +19,7 @@ torch::Tensor hj_probe_count(torch::Tensor tkeys, torch::Tensor theads,
 std::vector<torch::Tensor> hj_probe_fill(torch::Tensor tkeys, torch::Tensor theads,
                                          torch::Tensor next, torch::Tensor keys,
                                          torch::Tensor offsets, int64_t total);
+std::vector<torch::Tensor> hg_group(torch::Tensor keys);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sail_amd MI355X (gfx950) kernels";
@@ -32,4 +33,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hj_build_chain", &hj_build_chain, "hash join build with chains");
   m.def("hj_probe_count", &hj_probe_count, "hash join probe, count matches");
   m.def("hj_probe_fill", &hj_probe_fill, "hash join probe, emit match pairs");
+  m.def("hg_group", &hg_group, "hash group-id assignment (sort-free group_ids)");
 }

@@ -73,17 +73,11 @@ def group_ids(keys: List[Column], mask: Optional[torch.Tensor] = None
         packed1 = _pack_or_hash(norm)
     else:
         packed1 = norm[0]
-    if packed1.is_cuda and mask is None and n >= 1_000_000:
-        # sort-free hash group-id kernel (ops/csrc/hash_join.hip hg_group):
-        # unique-sort + searchsorted + rep-scatter were ~600 ms of
-        # ClickBench's group-by-URL queries (profiles/). Group numbering is
-        # arbitrary, as hash aggregation's is everywhere.
-        from ..ops import kernels as K
-
-        if K.available():
-            gid, rep_full, counter = K.require().hg_group(packed1)
-            ng = int(counter.item())
-            return gid, rep_full[:ng], ng
+    # NOTE: a sort-free hash group-id kernel exists (hg_group in
+    # ops/csrc/hash_join.hip, validated by tests/test_gpu.py) but measured
+    # SLOWER than this unique-sort path end-to-end (ClickBench 1.19s ->
+    # 1.33s): rocPRIM onesweep sorts at ~3 GB/ms beat random-scatter CAS
+    # claims + table gathers at these shapes, so the sort path stays.
     sel = packed1 if mask is None else packed1[mask]
     uniq, _ = torch.unique(sel, return_inverse=True)
     gid = torch.searchsorted(uniq, packed1).clamp(0, max(uniq.shape[0] - 1, 0))

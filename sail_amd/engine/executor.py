@@ -31,6 +31,9 @@ class ExecError(ExecutionException):
     pass
 
 
+TOPK_MIN_ROWS = 1_000_000  # ORDER BY+LIMIT below this just sorts
+
+
 class ExecutionContext:
     """Per-query execution context: session, device, subquery cache."""
 
@@ -209,10 +212,52 @@ class Executor:
         return Chunk(cols, [n for n, _ in p.schema], child.partitioning)
 
     def _x_Limit(self, p: S.Limit) -> Chunk:
+        fin = p.input
+        while isinstance(fin, S.SubqueryAlias):
+            fin = fin.input
+        if isinstance(fin, S.Sort) and p.n is not None \
+                and (p.n + p.offset) <= 100_000:
+            out = self._try_topk(fin, p.n + p.offset)
+            if out is not None:
+                start = p.offset
+                return out.slice(start, max(0, min(p.n, out.num_rows - start)))
         child = self._gather(self.execute(p.input))
         start = p.offset
         n = p.n if p.n is not None else child.num_rows - start
         return child.slice(start, max(0, min(n, child.num_rows - start)))
+
+    def _try_topk(self, sort: S.Sort, k: int) -> Optional[Chunk]:
+        """ORDER BY ... LIMIT k over a big input: torch.topk SELECTION on the
+        primary key (+ boundary ties), then the full stable multi-key sort on
+        that candidate set only — exact, and skips sorting 90M rows for a
+        top-10 (ClickBench q32-35 sorted the whole group table; profiles/
+        showed the onesweep sorts as the dominant kernels)."""
+        child = self._gather(self.execute(sort.input))
+        N = child.num_rows
+        if N < TOPK_MIN_ROWS or N <= 4 * k or not sort.keys:
+            if N <= 1:
+                return child
+            return child.gather(sort_indices(self.ev, sort.keys, child))
+        k0 = sort.keys[0]
+        col = broadcast(self.ev.eval(k0.child, child), N, child.device)
+        keyvals = _sortable(col)
+        nulls_first = k0.nulls_first if k0.nulls_first is not None else k0.ascending
+        if col.validity is not None:
+            big = _null_sentinel(keyvals, nulls_first == k0.ascending)
+            keyvals = torch.where(col.valid_mask(), keyvals, big)
+        kc = min(N, max(4 * k, k + 1024))
+        top = torch.topk(keyvals.to(torch.float64)
+                         if keyvals.dtype not in (torch.int64, torch.float64)
+                         else keyvals, kc, largest=not k0.ascending, sorted=True)
+        boundary = top.values[k - 1]
+        ties = torch.nonzero(keyvals == boundary.to(keyvals.dtype),
+                             as_tuple=False).flatten()
+        if ties.numel() > 5_000_000:
+            return child.gather(sort_indices(self.ev, sort.keys, child))
+        cand = torch.unique(torch.cat([top.indices, ties]))
+        sub = child.gather(cand)
+        order = sort_indices(self.ev, sort.keys, sub)
+        return sub.gather(order[:k])
 
     def _x_Distinct(self, p: S.Distinct) -> Chunk:
         child = self.execute(p.input)

@@ -344,3 +344,27 @@ def test_regr_aggregates(s):
         "SELECT regr_slope(y, x), regr_intercept(y, x), regr_r2(y, x), "
         "regr_count(y, x) FROM rg").collect()[0]
     assert abs(slope - 1.99) < 0.02 and n == 4 and r2 > 0.99
+
+
+def test_topk_matches_full_sort(monkeypatch):
+    import sail_amd.engine.executor as ex
+    import torch
+
+    s2 = sail_amd.SessionContext(device="cpu")
+    g = torch.Generator().manual_seed(13)
+    n = 300_000
+    c = torch.randint(0, 500, (n,), generator=g).tolist()   # heavy ties
+    k = torch.randint(0, 10**9, (n,), generator=g).tolist()
+    s2.create_dataframe({"c": c, "k": k}, name="tk")
+    q = "SELECT c, k FROM tk ORDER BY c DESC, k LIMIT 25"
+    monkeypatch.setattr(ex, "TOPK_MIN_ROWS", 1 << 60)
+    want = s2.sql(q).collect()
+    monkeypatch.setattr(ex, "TOPK_MIN_ROWS", 1)
+    got = s2.sql(q).collect()
+    assert got == want
+    # with OFFSET and ascending + nulls
+    q2 = "SELECT c FROM tk ORDER BY c LIMIT 10 OFFSET 5"
+    monkeypatch.setattr(ex, "TOPK_MIN_ROWS", 1 << 60)
+    want2 = s2.sql(q2).collect()
+    monkeypatch.setattr(ex, "TOPK_MIN_ROWS", 1)
+    assert s2.sql(q2).collect() == want2

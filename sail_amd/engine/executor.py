@@ -956,6 +956,47 @@ class Executor:
         return Chunk([StringColumn.from_pylist([p.key]), StringColumn.from_pylist([val])],
                      ["key", "value"])
 
+    def _x_AlterTable(self, p: S.AlterTable) -> Chunk:
+        cat = self.ctx.session.catalog
+        key = cat._key(p.name)
+        t = cat._tables.get(key)
+        schema = list(cat._schemas.get(key) or [])
+        if p.action == "add_columns":
+            nrows = t.num_rows if t is not None else 0
+            for cn, ct in p.columns:
+                if any(n.lower() == cn.lower() for n, _ in schema):
+                    raise ExecError(f"column {cn} already exists")
+                schema.append((cn, ct))
+                if t is not None:
+                    t.columns[cn] = Column.from_values([None] * nrows, ct,
+                                                       device="cpu")
+        elif p.action == "drop_column":
+            schema = [(n, ty) for n, ty in schema if n.lower() != p.column.lower()]
+            if t is not None:
+                for n in list(t.columns):
+                    if n.lower() == p.column.lower():
+                        del t.columns[n]
+        elif p.action == "rename_column":
+            schema = [(p.new_name if n.lower() == p.column.lower() else n, ty)
+                      for n, ty in schema]
+            if t is not None:
+                t.columns = {(p.new_name if n.lower() == p.column.lower() else n): c
+                             for n, c in t.columns.items()}
+        elif p.action == "rename_table":
+            nk = cat._key(p.new_name)
+            if t is not None:
+                cat._tables[nk] = cat._tables.pop(key)
+            cat._schemas[nk] = cat._schemas.pop(key, schema)
+            if key in cat._replicated:
+                cat._replicated.discard(key)
+                cat._replicated.add(nk)
+            return Chunk([StringColumn.from_pylist(
+                [f"renamed {p.name} to {p.new_name}"], dict_encode=False)], ["result"])
+        cat._schemas[key] = schema
+        cat._col_stats = {k: v for k, v in cat._col_stats.items() if k[0] != key}
+        return Chunk([StringColumn.from_pylist([f"altered {p.name}"], dict_encode=False)],
+                     ["result"])
+
     def _x_ShowFunctions(self, p: S.ShowFunctions) -> Chunk:
         from ..functions.registry import AGG_FUNCTIONS, SCALAR_RETURN, WINDOW_FUNCTIONS
         from .aggregates import UDAFS

@@ -866,6 +866,12 @@ class Resolver:
         p.schema = [("namespace", T.STRING), ("tableName", T.STRING), ("isTemporary", T.BOOL)]
         return p
 
+    def _p_AlterTable(self, p: S.AlterTable, outer):
+        if self.catalog.table_schema(p.name) is None:
+            raise ResolutionError(f"table not found: {p.name}")
+        p.schema = [("result", T.STRING)]
+        return p
+
     def _p_ShowFunctions(self, p: S.ShowFunctions, outer):
         p.schema = [("function", T.STRING)]
         return p

@@ -687,6 +687,18 @@ class SetConfig(Command):
 
 
 @dataclass
+class AlterTable(Command):
+    """ALTER TABLE: add/drop/rename columns, rename table (in-memory
+    catalog + delta metadata — ref: spec CommandNode AlterTable subset)."""
+
+    name: str = ""
+    action: str = ""            # add_columns | drop_column | rename_column | rename_table
+    columns: List[Tuple[str, T.DataType]] = field(default_factory=list)
+    column: str = ""
+    new_name: str = ""
+
+
+@dataclass
 class ShowFunctions(Command):
     pattern: Optional[str] = None
 

@@ -133,6 +133,40 @@ class Parser:
             if self.eat_kw("LIKE"):
                 pattern = self.next().value
             return S.ShowTables(pattern=pattern)
+        if self.at_kw("ALTER"):
+            self.next()
+            self.expect_kw("TABLE")
+            name = self._qualified_name()
+            if self.eat_kw("ADD"):
+                self.eat_kw("COLUMNS") or self.eat_kw("COLUMN")
+                cols = []
+                paren = self.eat_op("(")
+                while True:
+                    cn = self.ident()
+                    ct = self._parse_type()
+                    cols.append((cn, ct))
+                    if not self.eat_op(","):
+                        break
+                if paren:
+                    self.expect_op(")")
+                return S.AlterTable(name=name, action="add_columns", columns=cols)
+            if self.eat_kw("DROP"):
+                self.eat_kw("COLUMNS") or self.eat_kw("COLUMN")
+                paren = self.eat_op("(")
+                col = self.ident()
+                if paren:
+                    self.expect_op(")")
+                return S.AlterTable(name=name, action="drop_column", column=col)
+            if self.eat_kw("RENAME"):
+                if self.eat_kw("TO"):
+                    return S.AlterTable(name=name, action="rename_table",
+                                        new_name=self._qualified_name())
+                self.expect_kw("COLUMN")
+                col = self.ident()
+                self.expect_kw("TO")
+                return S.AlterTable(name=name, action="rename_column",
+                                    column=col, new_name=self.ident())
+            raise SqlError("unsupported ALTER TABLE action", self.sql, self.peek().pos)
         if self.at_kw("CACHE"):
             self.next()
             self.eat_kw("LAZY")

@@ -153,3 +153,17 @@ def test_from_json(s):
     assert rows == [({"a": 5, "b": "hi"},), (None,), (None,)]
     rows = s.sql("SELECT from_json(j, 'a INT, b STRING').a + 1 FROM fj").collect()
     assert rows == [(6,), (None,), (None,)]
+
+
+def test_alter_table(s):
+    s.create_dataframe({"x": [1, 2]}, name="at")
+    s.sql("ALTER TABLE at ADD COLUMNS (y INT, z STRING)")
+    assert s.sql("SELECT * FROM at ORDER BY x").collect() == [
+        (1, None, None), (2, None, None)]
+    s.sql("ALTER TABLE at RENAME COLUMN y TO yy")
+    s.sql("UPDATE at SET yy = 5 WHERE x = 1")
+    assert s.sql("SELECT x, yy FROM at ORDER BY x").collect() == [(1, 5), (2, None)]
+    s.sql("ALTER TABLE at DROP COLUMN z")
+    assert [r[0] for r in s.sql("DESCRIBE at").collect()] == ["x", "yy"]
+    s.sql("ALTER TABLE at RENAME TO at2")
+    assert s.sql("SELECT count(*) FROM at2").collect() == [(2,)]

@@ -84,6 +84,30 @@ class ConnectClient:
         sc = W.parse(W.first(fields, F.ANR_SCHEMA, b""))
         return W.first_str(sc, 2)
 
+    def ddl_parse(self, ddl: str) -> str:
+        req = (W.field_string(F.AN_SESSION_ID, self.session_id)
+               + W.field_message(F.AN_DDL_PARSE, W.field_string(1, ddl)))
+        fields = W.parse(self._call_unary("AnalyzePlan", req))
+        dp = W.parse(W.first(fields, F.ANR_DDL_PARSE, b""))
+        return W.first_str(dp, 2)
+
+    def same_semantics(self, sql_a: str, sql_b: str) -> bool:
+        inner = (W.field_message(1, self._sql_plan(sql_a))
+                 + W.field_message(2, self._sql_plan(sql_b)))
+        req = (W.field_string(F.AN_SESSION_ID, self.session_id)
+               + W.field_message(F.AN_SAME_SEMANTICS, inner))
+        fields = W.parse(self._call_unary("AnalyzePlan", req))
+        ss = W.parse(W.first(fields, F.ANR_SAME_SEMANTICS, b""))
+        return bool(W.first_varint(ss, 1, 0))
+
+    def semantic_hash(self, sql: str) -> int:
+        inner = W.field_message(1, self._sql_plan(sql))
+        req = (W.field_string(F.AN_SESSION_ID, self.session_id)
+               + W.field_message(F.AN_SEMANTIC_HASH, inner))
+        fields = W.parse(self._call_unary("AnalyzePlan", req))
+        sh = W.parse(W.first(fields, F.ANR_SEMANTIC_HASH, b""))
+        return int(W.first_varint(sh, 1, 0))
+
     def set_conf(self, key: str, value: str):
         kv = W.field_string(F.KV_KEY, key) + W.field_string(F.KV_VALUE, value)
         op = W.field_message(F.CFG_OP_SET, W.field_message(1, kv))

@@ -78,6 +78,8 @@ class F:
     AN_INPUT_FILES = 9
     AN_SPARK_VERSION = 10
     AN_DDL_PARSE = 11
+    AN_SAME_SEMANTICS = 12
+    AN_SEMANTIC_HASH = 13
     # AnalyzePlanResponse oneofs
     ANR_SESSION_ID = 1
     ANR_SCHEMA = 2
@@ -87,6 +89,9 @@ class F:
     ANR_IS_STREAMING = 6
     ANR_INPUT_FILES = 7
     ANR_SPARK_VERSION = 8
+    ANR_DDL_PARSE = 9
+    ANR_SAME_SEMANTICS = 10
+    ANR_SEMANTIC_HASH = 11
     # ConfigRequest
     CFG_SESSION_ID = 1
     CFG_OPERATION = 3
@@ -246,7 +251,43 @@ class SparkConnectServer:
             return out + W.field_message(F.ANR_IS_LOCAL, W.field_varint(1, 1))
         if F.AN_IS_STREAMING in req:
             return out + W.field_message(F.ANR_IS_STREAMING, W.field_varint(1, 0))
+        ddl = W.first(req, F.AN_DDL_PARSE)
+        if ddl is not None:
+            from ..sql.parser import parse_ddl_schema
+
+            text = W.first_str(W.parse(ddl), 1)
+            fields = parse_ddl_schema(text)
+            rendered = ", ".join(f"{n} {t!r}" for n, t in fields)
+            return out + W.field_message(F.ANR_DDL_PARSE, W.field_string(2, rendered))
+        ss = W.first(req, F.AN_SAME_SEMANTICS)
+        if ss is not None:
+            inner = W.parse(ss)
+            trees = []
+            for fn in (1, 2):
+                plan = W.parse(W.first(inner, fn, b""))
+                root = W.first(plan, F.PLAN_ROOT)
+                sql = W.first_str(W.parse(W.first(W.parse(root or b""), F.REL_SQL, b"")),
+                                  F.SQL_QUERY)
+                trees.append(self._semantic_tree(sess, sql))
+            same = 1 if trees[0] == trees[1] else 0
+            return out + W.field_message(F.ANR_SAME_SEMANTICS, W.field_varint(1, same))
+        sh = W.first(req, F.AN_SEMANTIC_HASH)
+        if sh is not None:
+            inner = W.parse(sh)
+            plan = W.parse(W.first(inner, 1, b""))
+            root = W.first(plan, F.PLAN_ROOT)
+            sql = W.first_str(W.parse(W.first(W.parse(root or b""), F.REL_SQL, b"")),
+                              F.SQL_QUERY)
+            import zlib
+
+            h = zlib.crc32(self._semantic_tree(sess, sql).encode()) & 0x7FFFFFFF
+            return out + W.field_message(F.ANR_SEMANTIC_HASH, W.field_varint(1, h))
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "analyze type not supported")
+
+    def _semantic_tree(self, sess: SessionContext, sql: str) -> str:
+        from ..plan import spec as S
+
+        return S.plan_tree_string(sess.plan_sql(sql))
 
     def _render_schema(self, sess: SessionContext, sql: str) -> bytes:
         plan = sess.plan_sql(sql)

@@ -102,3 +102,8 @@ def first(fields: Dict[int, List], num: int, default=None):
 def first_str(fields: Dict[int, List], num: int, default: str = "") -> str:
     v = first(fields, num)
     return v.decode("utf-8") if isinstance(v, (bytes, bytearray)) else default
+
+
+def first_varint(fields: Dict[int, List], num: int, default: int = 0) -> int:
+    v = first(fields, num)
+    return int(v) if isinstance(v, int) else default

@@ -75,3 +75,13 @@ def test_tpch_q1_over_connect(server, client):
     t = client.sql(QUERIES[1])
     assert t.num_rows == 4
     assert t.column_names[0] == "l_returnflag"
+
+
+def test_analyze_ddl_semantics(client):
+    out = client.ddl_parse("a INT, b STRING")
+    assert "a" in out and "b" in out
+    assert client.same_semantics("SELECT 1 + 1", "SELECT 1 + 1")
+    assert not client.same_semantics("SELECT 1", "SELECT 2")
+    h1 = client.semantic_hash("SELECT 1 + 1")
+    h2 = client.semantic_hash("SELECT 1 + 1")
+    assert h1 == h2 and h1 > 0

@@ -432,6 +432,18 @@ class Resolver:
     def _p_Sort(self, p: S.Sort, outer):
         child = self._plan(p.input, outer)
         scope = self._child_scope(child, outer)
+        if len(p.keys) == 1 and isinstance(p.keys[0].child, S.Col) \
+                and p.keys[0].child.name == "__all__":
+            # ORDER BY ALL: every output column, left to right
+            asc = p.keys[0].ascending
+            keys = [S.SortKey(S.BoundRef(i, n, t), asc, None)
+                    for i, (n, t) in enumerate(child.schema)]
+            out = S.Sort(input=child, keys=[self._expr(k, scope) if False else k
+                                            for k in keys])
+            for k in out.keys:
+                k.dtype = k.child.dtype
+            out.schema = child.schema
+            return out
         keys: List[S.SortKey] = []
         extra_exprs: List[S.Expr] = []
         for k in p.keys:

@@ -560,6 +560,14 @@ class Parser:
                             keymap[k] = (len(group_by), e)
                             group_by.append(e)
                 grouping_sets = [[keymap[repr(e)][0] for e in se] for se in sets_exprs]
+            elif self.at_kw("ALL") and self.peek(1).upper not in ("(",) \
+                    and not (self.peek(1).kind == "op" and self.peek(1).value in (",", "(")):
+                self.next()
+                # GROUP BY ALL: every non-aggregate select item is a key
+                for pe in projections:
+                    inner = pe.child if isinstance(pe, S.Alias) else pe
+                    if not self._contains_agg(pe) and not isinstance(inner, S.Star):
+                        group_by.append(inner)
             else:
                 group_by = self._expr_list()
                 if self.eat_kw("WITH"):
@@ -574,6 +582,31 @@ class Parser:
         having = None
         if self.eat_kw("HAVING"):
             having = self.parse_expr()
+
+        if self.at_kw("WINDOW"):
+            # named windows: WINDOW w AS (spec) [, w2 AS (spec)]
+            self.next()
+            named = {}
+            while True:
+                wname = self.ident()
+                self.expect_kw("AS")
+                part, order, frame = self._parse_window_spec()
+                named[wname.lower()] = (part, order, frame)
+                if not self.eat_op(","):
+                    break
+
+            def patch(e):
+                ref = e.__dict__.get("_window_ref") if hasattr(e, "__dict__") else None
+                if isinstance(e, S.WindowExpr) and ref:
+                    spec = named.get(ref.lower())
+                    if spec is None:
+                        raise SqlError(f"undefined window {ref}", self.sql, 0)
+                    e.partition_by, e.order_by, e.frame = spec
+                for c in e.children():
+                    patch(c)
+
+            for pe in projections:
+                patch(pe)
 
         # Build: aggregate if group_by or aggregate functions present
         has_agg = group_by or any(self._contains_agg(p) for p in projections) or (
@@ -593,6 +626,34 @@ class Parser:
 
     def _parse_order_limit(self, plan: S.Plan) -> S.Plan:
         if self.eat_kw("ORDER"):
+            self.expect_kw("BY")
+            if self.at_kw("ALL") and self.peek(1).kind != "op" or \
+                    (self.at_kw("ALL") and self.peek(1).kind == "op"
+                     and self.peek(1).value != "("):
+                self.next()
+                asc = not self.eat_kw("DESC")
+                self.eat_kw("ASC")
+                keys = [S.SortKey(S.Col("__all__"), asc, None)]
+            else:
+                keys = [self._parse_sort_key()]
+                while self.eat_op(","):
+                    keys.append(self._parse_sort_key())
+            plan = S.Sort(input=plan, keys=keys)
+        if self.at_kw("CLUSTER"):
+            # CLUSTER BY = DISTRIBUTE BY + SORT BY; a whole-partition-per-GPU
+            # engine treats the distribution as a physical no-op and keeps
+            # the sort
+            self.next()
+            self.expect_kw("BY")
+            keys = [S.SortKey(e, True, None) for e in self._expr_list()]
+            plan = S.Sort(input=plan, keys=keys)
+        if self.at_kw("DISTRIBUTE"):
+            self.next()
+            self.expect_kw("BY")
+            self._expr_list()  # partitioning hint: no-op locally
+        if self.at_kw("SORT"):
+            # SORT BY: partition-local sort; equals a total sort here
+            self.next()
             self.expect_kw("BY")
             keys = [self._parse_sort_key()]
             while self.eat_op(","):
@@ -1220,8 +1281,13 @@ class Parser:
 
         if self.at_kw("OVER"):
             self.next()
-            part, order, frame = self._parse_window_spec()
-            e = S.WindowExpr(func=e, partition_by=part, order_by=order, frame=frame)
+            if self.peek().kind == "ident" and self.peek().upper not in _RESERVED_STOP:
+                ref = self.ident()
+                e = S.WindowExpr(func=e)
+                e.__dict__["_window_ref"] = ref
+            else:
+                part, order, frame = self._parse_window_spec()
+                e = S.WindowExpr(func=e, partition_by=part, order_by=order, frame=frame)
         return e
 
     def _parse_window_spec(self):

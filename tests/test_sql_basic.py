@@ -368,3 +368,18 @@ def test_topk_matches_full_sort(monkeypatch):
     want2 = s2.sql(q2).collect()
     monkeypatch.setattr(ex, "TOPK_MIN_ROWS", 1)
     assert s2.sql(q2).collect() == want2
+
+
+def test_named_windows_and_by_all(s):
+    s2 = sail_amd.SessionContext(device="cpu")
+    s2.create_dataframe({"g": ["a", "a", "b"], "v": [1, 2, 3]}, name="nb")
+    rows = s2.sql("SELECT g, v, sum(v) OVER w FROM nb WINDOW w AS (PARTITION BY g) "
+                  "ORDER BY g, v").collect()
+    assert rows == [("a", 1, 3), ("a", 2, 3), ("b", 3, 3)]
+    assert s2.sql("SELECT g, sum(v) FROM nb GROUP BY ALL ORDER BY g").collect() == [
+        ("a", 3), ("b", 3)]
+    assert s2.sql("SELECT v, g FROM nb ORDER BY ALL DESC").collect() == [
+        (3, "b"), (2, "a"), (1, "a")]
+    assert s2.sql("SELECT * FROM nb SORT BY v DESC").collect()[0] == ("b", 3)
+    assert len(s2.sql("SELECT * FROM nb CLUSTER BY g").collect()) == 3
+    assert len(s2.sql("SELECT * FROM nb DISTRIBUTE BY g").collect()) == 3

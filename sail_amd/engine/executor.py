@@ -649,9 +649,8 @@ class Executor:
                 "sail.exec.agg_shuffle_threshold_groups", str(4_000_000)))
             # CONSENSUS across ranks (shards differ; a rank-local count would
             # send ranks down different exchange paths — gloo/RCCL mismatch)
-            ngt = torch.tensor([ng_local * self.dctx.world], dtype=torch.int64)
-            self.dctx.all_reduce_sum_(ngt)
-            ng_consensus = int(ngt.item()) // self.dctx.world  # mean estimate
+            ng_consensus = self.dctx.consensus_sum(
+                ng_local * self.dctx.world) // self.dctx.world  # mean estimate
             if ng_consensus * self.dctx.world > threshold:
                 ncols = len(child.columns)
                 ext = Chunk(list(child.columns) + key_cols0,
@@ -823,9 +822,7 @@ class Executor:
         # CONSENSUS: shard sizes differ per rank, but every rank must pick
         # the same exchange plan (divergent paths = collective mismatch) —
         # decide on the global build-side size via one tiny all_reduce.
-        local = torch.tensor([_chunk_bytes(right)], dtype=torch.int64)
-        self.dctx.all_reduce_sum_(local)
-        return int(local.item()) > threshold
+        return self.dctx.consensus_sum(_chunk_bytes(right)) > threshold
 
     def _shuffle_join_inputs(self, p: S.Join, left: Chunk, right: Chunk):
         from ..exec.distributed import shuffle_chunk

@@ -27,6 +27,32 @@ class DistContext:
     def barrier(self):
         self.dist.barrier()
 
+    def comm_device(self):
+        """Collectives must run on the backend's device: nccl (RCCL) only
+        accepts CUDA tensors, gloo only CPU — a CPU tensor on nccl is a
+        hard error on the 8-GPU path that CPU gloo tests can't catch."""
+        return self.device if self.dist.get_backend() == "nccl" else "cpu"
+
+    def consensus_sum(self, value: int) -> int:
+        """all_reduce(SUM) a host integer on the correct device."""
+        t = torch.tensor([int(value)], dtype=torch.int64,
+                         device=self.comm_device())
+        self.dist.all_reduce(t, op=self.dist.ReduceOp.SUM)
+        return int(t.item())
+
+    def consensus_minmax(self, lo: int, hi: int):
+        """all_reduce global (min, max) of per-rank ints (MAX of (-lo, hi))."""
+        t = torch.tensor([-int(lo), int(hi)], dtype=torch.int64,
+                         device=self.comm_device())
+        self.dist.all_reduce(t, op=self.dist.ReduceOp.MAX)
+        return -int(t[0].item()), int(t[1].item())
+
+    def consensus_max(self, value: int) -> int:
+        t = torch.tensor([int(value)], dtype=torch.int64,
+                         device=self.comm_device())
+        self.dist.all_reduce(t, op=self.dist.ReduceOp.MAX)
+        return int(t.item())
+
     def all_reduce_sum_(self, t: torch.Tensor):
         self.dist.all_reduce(t, op=self.dist.ReduceOp.SUM)
         return t

@@ -236,9 +236,7 @@ def sync_table_stats(session):
         t = cat._tables[name]
         grows = cat._global_rows.get(name, None)
         if grows is None:
-            n = torch.tensor([t.num_rows], dtype=torch.int64)
-            d.all_reduce_sum_(n)
-            grows = int(n.item())
+            grows = d.consensus_sum(t.num_rows)
             cat._global_rows[name] = grows
         for cn, c in t.columns.items():
             rows = len(c)
@@ -246,19 +244,16 @@ def sync_table_stats(session):
 
             if isinstance(c, StringColumn):
                 local = c.dict_size if c.is_dict else -1
-                agg = torch.tensor([local], dtype=torch.int64)
-                d.dist.all_reduce(agg, op=d.dist.ReduceOp.MAX)
-                ndv = int(agg.item()) if int(agg.item()) >= 0 else None
+                m = d.consensus_max(local)
+                ndv = m if m >= 0 else None
             elif rows == 0 or c.data.dtype == torch.bool:
                 ndv = 2
             elif c.data.dtype.is_floating_point:
                 ndv = None
             else:
-                lohi = torch.tensor([-int(c.data.min().item()) if rows else 0,
-                                     int(c.data.max().item()) if rows else 0],
-                                    dtype=torch.int64)
-                d.dist.all_reduce(lohi, op=d.dist.ReduceOp.MAX)
-                lo, hi = -int(lohi[0].item()), int(lohi[1].item())
+                lo, hi = d.consensus_minmax(
+                    int(c.data.min().item()) if rows else 0,
+                    int(c.data.max().item()) if rows else 0)
                 ndv = min(grows, hi - lo + 1)
             if ndv is not None:
                 ndv = min(max(1, ndv), grows)

@@ -117,6 +117,7 @@ class SparkConnectServer:
         self._host = host
         self._device = device
         self._sessions: Dict[str, SessionContext] = {}
+        self._session_atime: Dict[str, float] = {}
         self._lock = threading.Lock()
         self._server = grpc.server(
             futures.ThreadPoolExecutor(max_workers=16),
@@ -150,10 +151,24 @@ class SparkConnectServer:
     def address(self) -> str:
         return f"{self._host}:{self.port}"
 
+    #: idle eviction (ref: SessionManagerActor idle timeout,
+    #: crates/sail-session/src/; spark.session_timeout_secs)
+    SESSION_TIMEOUT_SECS = 3600.0
+
     def session(self, session_id: str) -> SessionContext:
+        import time as _t
+
+        now = _t.time()
         with self._lock:
+            # opportunistic eviction of idle sessions (frees their device
+            # tables); ReleaseSession removes eagerly
+            for sid in [k for k, ts in self._session_atime.items()
+                        if now - ts > self.SESSION_TIMEOUT_SECS and k != session_id]:
+                self._sessions.pop(sid, None)
+                self._session_atime.pop(sid, None)
             if session_id not in self._sessions:
                 self._sessions[session_id] = SessionContext(device=self._device)
+            self._session_atime[session_id] = now
             return self._sessions[session_id]
 
     def register_session(self, session_id: str, ctx: SessionContext):

@@ -125,6 +125,27 @@ def substitute_refs(e: S.Expr, exprs: List[S.Expr]) -> S.Expr:
     """Replace BoundRef i with exprs[i] (push an expr through a projection)."""
     if e is None:
         return None
+    if isinstance(e, S.Lambda):
+        # body refs 0..k-1 are lambda params, k+i is enclosing col i: keep
+        # params, substitute the captured columns SHIFTED back under the
+        # lambda (their own BoundRefs must re-shift by k)
+        k = len(e.params)
+
+        def shift(x: S.Expr) -> S.Expr:
+            if isinstance(x, S.BoundRef):
+                return S.BoundRef(x.index + k, x.name, x.dtype)
+            ch = x.children()
+            if not ch:
+                return x
+            out = x.with_children([shift(c) for c in ch])
+            out.dtype = x.dtype
+            return out
+
+        shifted = [shift(sub.child if isinstance(sub, S.Alias) else sub)
+                   for sub in exprs]
+        params = [S.BoundRef(j, e.params[j], None) for j in range(k)]
+        return S.Lambda(e.params, substitute_refs(e.body, params + shifted),
+                        e.dtype)
     if isinstance(e, S.BoundRef):
         sub = exprs[e.index]
         return sub.child if isinstance(sub, S.Alias) else sub

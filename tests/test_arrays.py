@@ -175,3 +175,15 @@ def test_aggregate_reduce(s):
     rows = s.sql("SELECT aggregate(sequence(1, v), 0, (acc, x) -> acc + x) "
                  "FROM agg_t ORDER BY v").collect()
     assert rows == [(3,), (10,)]
+
+
+def test_lambda_pushdown_through_projection(s):
+    """Filter with a lambda pushed through a subquery projection must keep
+    param indices intact (regression: substitute_refs corrupted them)."""
+    s.create_dataframe({"v": [2, 5]}, name="lpd")
+    q = ("SELECT v FROM (SELECT sequence(1, v) AS a, v FROM lpd) x "
+         "WHERE exists(a, e -> e = v) ORDER BY v")
+    assert s.sql(q).collect() == [(2,), (5,)]
+    q2 = ("SELECT v FROM (SELECT sequence(1, v) AS a, v FROM lpd) x "
+          "WHERE size(filter(a, e -> e > 1)) > 3")
+    assert s.sql(q2).collect() == [(5,)]

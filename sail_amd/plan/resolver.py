@@ -766,6 +766,31 @@ class Resolver:
 
     def _p_InsertInto(self, p: S.InsertInto, outer):
         inp = self._plan(p.input, outer)
+        head, _, rest = p.table.partition(".")
+        if head.lower() in ("parquet", "csv", "json", "delta") and rest:
+            # INSERT INTO delta.`/path` -> datasource append/overwrite;
+            # positional semantics: rename/cast the input to the target
+            # schema (VALUES rows arrive as col1/col2/...)
+            from ..datasource.registry import infer_source_schema
+
+            tgt = infer_source_schema(head.lower(), [rest], {})
+            if len(inp.schema) != len(tgt):
+                raise ResolutionError(
+                    f"INSERT INTO {p.table}: {len(inp.schema)} columns for "
+                    f"{len(tgt)}-column target")
+            exprs = []
+            for i, ((sn, st), (tn, tt)) in enumerate(zip(inp.schema, tgt)):
+                ref = S.BoundRef(i, sn, st)
+                e = S.Cast(ref, tt, dtype=tt) if st != tt else ref
+                exprs.append(S.Alias(e, tn, tt))
+            pr = S.Project(input=inp, exprs=exprs)
+            pr.schema = list(tgt)
+            out = S.Write(input=pr, format=head.lower(), path=rest,
+                          mode="overwrite" if p.overwrite else "append")
+            out.schema = []
+            return out
+        if self.catalog.table_schema(p.table) is None:
+            raise ResolutionError(f"table not found: {p.table}")
         out = S.InsertInto(table=p.table, input=inp, overwrite=p.overwrite)
         out.schema = []
         return out

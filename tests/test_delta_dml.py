@@ -131,3 +131,29 @@ def test_delta_checkpoint_roundtrip(s, tmp_path):
     # time travel below the checkpoint still works (full replay path)
     v5 = s.read.format("delta").option("versionAsOf", 5).load(p)
     assert len(v5.collect()) == 6
+
+
+def test_insert_into_delta_path(s, tmp_path):
+    p = str(tmp_path / "ins")
+    s.create_dataframe({"k": [1], "v": [1.0]}, schema={"k": T.I64, "v": T.F64},
+                       name="ins_seed")
+    s.table("ins_seed").write.format("delta").mode("overwrite").save(p)
+    s.sql(f"INSERT INTO delta.`{p}` VALUES (2, 2.5)")
+    s.sql(f"INSERT INTO delta.`{p}` SELECT k + 10, v FROM ins_seed")
+    assert s.sql(f"SELECT * FROM delta.`{p}` ORDER BY k").collect() == [
+        (1, 1.0), (2, 2.5), (11, 1.0)]
+    with pytest.raises(Exception):
+        s.sql(f"INSERT INTO delta.`{p}` VALUES (1)")  # arity mismatch
+
+
+def test_merge_not_matched_by_source(s, tmp_path):
+    p = str(tmp_path / "mbs")
+    s.create_dataframe({"k": [1, 2], "v": [1.0, 2.0]},
+                       schema={"k": T.I64, "v": T.F64}, name="mbs_t")
+    s.table("mbs_t").write.format("delta").mode("overwrite").save(p)
+    s.create_dataframe({"k": [1], "v": [9.0]},
+                       schema={"k": T.I64, "v": T.F64}, name="mbs_s")
+    s.sql(f"MERGE INTO delta.`{p}` t USING mbs_s u ON t.k = u.k "
+          "WHEN MATCHED THEN UPDATE SET v = u.v "
+          "WHEN NOT MATCHED BY SOURCE THEN DELETE")
+    assert s.sql(f"SELECT * FROM delta.`{p}`").collect() == [(1, 9.0)]

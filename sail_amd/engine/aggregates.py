@@ -78,8 +78,15 @@ def group_ids(keys: List[Column], mask: Optional[torch.Tensor] = None
     # SLOWER than this unique-sort path end-to-end (ClickBench 1.19s ->
     # 1.33s): rocPRIM onesweep sorts at ~3 GB/ms beat random-scatter CAS
     # claims + table gathers at these shapes, so the sort path stays.
-    sel = packed1 if mask is None else packed1[mask]
-    uniq, _ = torch.unique(sel, return_inverse=True)
+    if mask is None:
+        # unique's inverse IS the gid — recomputing it with searchsorted
+        # cost ~20 ms per 100M-row call (ClickBench q32 profile)
+        uniq, gid = torch.unique(packed1, return_inverse=True)
+        ng = int(uniq.shape[0])
+        rep = _any_representative(gid, n, ng, None, dev)
+        return gid, rep, ng
+    sel = packed1[mask]
+    uniq = torch.unique(sel)
     gid = torch.searchsorted(uniq, packed1).clamp(0, max(uniq.shape[0] - 1, 0))
     ng = int(uniq.shape[0])
     rep = _any_representative(gid, n, ng, mask, dev)

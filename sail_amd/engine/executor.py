@@ -1287,6 +1287,11 @@ def join_chunks(ev: Evaluator, p: S.Join, left: Chunk, right: Chunk) -> Chunk:
 
 def _gather_nullable(c: Column, idx: torch.Tensor) -> Column:
     """Gather where idx == -1 produces NULL."""
+    if len(c) == 0:
+        # all-null fill from an EMPTY side (outer join against an empty
+        # table): there is no row 0 to clamp to
+        return Column.from_values([None] * int(idx.shape[0]), c.dtype,
+                                  device=idx.device)
     null = idx < 0
     safe = torch.where(null, torch.zeros_like(idx), idx)
     out = c.gather(safe)

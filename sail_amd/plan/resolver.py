@@ -472,6 +472,9 @@ class Resolver:
             if 0 <= idx < len(child.schema):
                 n, t = child.schema[idx]
                 return S.BoundRef(idx, n, t)
+            raise ResolutionError(
+                f"ORDER BY position {e.value} is not in the select list "
+                f"(1..{len(child.schema)})")
         # ORDER BY COUNT(*) / aggregate expressions over a grouped query
         if any(isinstance(x, S.AggFunc) for x in e.walk()) and isinstance(child, S.Project):
             agg = child.input
@@ -973,11 +976,17 @@ class Resolver:
                 f = scope.fields[hits[0]]
                 return S.BoundRef(hits[0], f.name, f.dtype)
             if len(hits) > 1:
-                # identical duplicates (e.g. USING join remnants): take first
+                # same qualifier on every hit = genuine duplicate column set
+                # (e.g. SELECT * over a self-join projection): take first;
+                # different qualifiers = ambiguous (Spark AMBIGUOUS_REFERENCE)
                 f0 = scope.fields[hits[0]]
-                if all(scope.fields[h].dtype == f0.dtype for h in hits):
+                quals = {(scope.fields[h].qualifier or "") for h in hits}
+                if len(quals) == 1 and all(scope.fields[h].dtype == f0.dtype for h in hits):
                     return S.BoundRef(hits[0], f0.name, f0.dtype)
-                raise ResolutionError(f"ambiguous column {e.name}")
+                raise ResolutionError(
+                    f"ambiguous column {e.name} (candidates: "
+                    + ", ".join(sorted(f"{scope.fields[h].qualifier or '?'}.{e.name}"
+                                       for h in hits)) + ")")
             # outer scope (correlated subquery)
             s = scope.outer
             depth = 0

@@ -63,3 +63,13 @@ def test_division_semantics(s):
     # Spark: x / 0 -> null (non-ANSI)
     assert s.sql("SELECT a / 0 FROM t ORDER BY a").collect() == [(None,), (None,)]
     assert s.sql("SELECT a % 0 FROM t ORDER BY a").collect() == [(None,), (None,)]
+
+
+def test_outer_joins_empty_sides(s):
+    s.sql("CREATE TABLE e2 (x INT, y STRING)")
+    assert s.sql("SELECT * FROM e2 RIGHT JOIN t ON e2.x = t.a ORDER BY a").collect() == [
+        (None, None, 1, "x"), (None, None, 2, "y")]
+    assert s.sql("SELECT * FROM e2 FULL JOIN t ON e2.x = t.a ORDER BY a").collect() == [
+        (None, None, 1, "x"), (None, None, 2, "y")]
+    assert s.sql("SELECT * FROM t FULL JOIN e2 ON t.a = e2.x ORDER BY a").collect() == [
+        (1, "x", None, None), (2, "y", None, None)]

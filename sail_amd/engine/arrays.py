@@ -529,6 +529,8 @@ def eval_hof(ev, e, chunk):
     from .chunk import Chunk
 
     name = e.name.lower()
+    if name in ("transform_keys", "transform_values", "map_filter"):
+        return eval_map_hof(ev, e, chunk)
     if name in ("aggregate", "reduce"):
         return _f_reduce(e.args, e.dtype, chunk, ev)
     if name == "zip_with":
@@ -667,3 +669,197 @@ def _f_reduce(args, out, chunk, ev):
         valid = arr.valid_mask() & acc.valid_mask()
         acc = Column(acc.dtype, acc.data, valid.to(torch.uint8))
     return acc
+
+
+# ---------------------------------------------------------------------------
+# array set ops / append / repeat / flatten
+# ---------------------------------------------------------------------------
+
+def _concat_rows(a: ListColumn, b: ListColumn) -> ListColumn:
+    """Per-row concatenation: out[i] = a[i] ++ b[i]."""
+    from .executor import concat_columns
+
+    n = len(a)
+    dev = a.device
+    la, lb = a.lengths(), b.lengths()
+    lens = la + lb
+    offs = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(lens, 0, out=offs[1:])
+    total = int(offs[-1].item())
+    seg = torch.repeat_interleave(torch.arange(n, dtype=torch.int64, device=dev), lens)
+    pos = torch.arange(total, dtype=torch.int64, device=dev) \
+        - torch.repeat_interleave(offs[:-1], lens)
+    from_a = pos < la.index_select(0, seg)
+    na = len(a.child)
+    idx_a = (a.offsets[:-1].index_select(0, seg) + pos).clamp(0, max(na - 1, 0))
+    idx_b = (b.offsets[:-1].index_select(0, seg)
+             + (pos - la.index_select(0, seg))).clamp(0, max(len(b.child) - 1, 0))
+    # gather from the concatenation of both children with one index space
+    allc = concat_columns([a.child, b.child]) if len(b.child) or len(a.child) else a.child
+    idx = torch.where(from_a, idx_a, idx_b + na)
+    child = allc.gather(idx) if len(allc) else a.child
+    validity = None
+    va, vb = a.valid_mask(), b.valid_mask()
+    both = va & vb
+    if not bool(both.all()):
+        validity = both.to(torch.uint8)
+    return ListColumn(offs, child, validity)
+
+
+def _f_array_union(args, out, chunk, ev):
+    a = _bcast(args[0], chunk)
+    b = _bcast(args[1], chunk)
+    cat = _concat_rows(a, b)
+    return _f_array_distinct([cat], out, chunk, ev)
+
+
+def _row_membership(a: ListColumn, b: ListColumn) -> torch.Tensor:
+    """Per-element-of-a bool: element also present in b's SAME row."""
+    from .joins import fnv_key_tensor, normalize_key
+
+    def key(ch):
+        if isinstance(ch, StringColumn):
+            return fnv_key_tensor(ch.decode_dict())
+        return normalize_key(ch)
+
+    ka, kb = key(a.child), key(b.child)
+    sa, sb = a.segment_ids(), b.segment_ids()
+    pa = _mix_rowkey(sa, ka)
+    pb = _mix_rowkey(sb, kb)
+    if pb.numel() == 0:
+        return torch.zeros(pa.shape[0], dtype=torch.bool, device=a.device)
+    sorted_b = torch.sort(pb).values
+    idx = torch.searchsorted(sorted_b, pa).clamp_max(sorted_b.shape[0] - 1)
+    return sorted_b.index_select(0, idx) == pa
+
+
+def _mix_rowkey(seg: torch.Tensor, key: torch.Tensor) -> torch.Tensor:
+    from ..exec.distributed import _mix64
+
+    return _mix64(seg * 31 + _mix64(key))
+
+
+def _filter_elements(c: ListColumn, keep: torch.Tensor) -> ListColumn:
+    kept = torch.nonzero(keep, as_tuple=False).flatten()
+    seg = c.segment_ids().index_select(0, kept)
+    lens = torch.bincount(seg, minlength=len(c))
+    offs = torch.zeros(len(c) + 1, dtype=torch.int64, device=c.device)
+    torch.cumsum(lens, 0, out=offs[1:])
+    return ListColumn(offs, c.child.gather(kept), c.validity, c.dtype)
+
+
+def _f_array_intersect(args, out, chunk, ev):
+    a = _bcast(args[0], chunk)
+    b = _bcast(args[1], chunk)
+    da = _f_array_distinct([a], out, chunk, ev)
+    return _filter_elements(da, _row_membership(da, b))
+
+
+def _f_array_except(args, out, chunk, ev):
+    a = _bcast(args[0], chunk)
+    b = _bcast(args[1], chunk)
+    da = _f_array_distinct([a], out, chunk, ev)
+    return _filter_elements(da, ~_row_membership(da, b))
+
+
+def _f_array_remove(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    m = _elem_match(c, _scalar_value(args[1]))
+    return _filter_elements(c, ~m)
+
+
+def _f_array_compact(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)
+    return _filter_elements(c, c.child.valid_mask())
+
+
+def _f_flatten(args, out, chunk, ev):
+    c = _bcast(args[0], chunk)  # list<list<T>>
+    inner = c.child
+    if not isinstance(inner, ListColumn):
+        raise NotImplementedError("flatten expects array<array<...>>")
+    if inner.validity is not None:
+        raise NotImplementedError("flatten with null inner arrays")
+    # inner lists are laid out contiguously: compose offsets directly
+    new_offs = inner.offsets.index_select(0, c.offsets)
+    return ListColumn(new_offs, inner.child, c.validity)
+
+
+def _f_array_repeat(args, out, chunk, ev):
+    from .eval import broadcast
+
+    n = chunk.num_rows
+    dev = chunk.device
+    v = broadcast(args[0], n, dev)
+    cnt = _bcast(args[1], chunk).data.to(torch.int64).clamp_min(0)
+    offs = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(cnt, 0, out=offs[1:])
+    seg = torch.repeat_interleave(torch.arange(n, dtype=torch.int64, device=dev), cnt)
+    child = v.gather(seg)
+    return ListColumn(offs, child)
+
+
+def _f_array_append(args, out, chunk, ev):
+    from .eval import broadcast, cast_column
+
+    a = _bcast(args[0], chunk)
+    v = cast_column(broadcast(args[1], chunk.num_rows, chunk.device),
+                    a.child.dtype)
+    one = torch.ones(len(a), dtype=torch.int64, device=a.device)
+    offs = torch.arange(0, len(a) + 1, dtype=torch.int64, device=a.device)
+    single = ListColumn(offs, v, None)
+    return _concat_rows(a, single)
+
+
+def _f_array_prepend(args, out, chunk, ev):
+    from .eval import broadcast, cast_column
+
+    a = _bcast(args[0], chunk)
+    v = cast_column(broadcast(args[1], chunk.num_rows, chunk.device),
+                    a.child.dtype)
+    offs = torch.arange(0, len(a) + 1, dtype=torch.int64, device=a.device)
+    single = ListColumn(offs, v, None)
+    return _concat_rows(single, a)
+
+
+IMPLS.update({
+    "array_union": _f_array_union, "array_intersect": _f_array_intersect,
+    "array_except": _f_array_except, "array_remove": _f_array_remove,
+    "array_compact": _f_array_compact, "flatten": _f_flatten,
+    "array_repeat": _f_array_repeat, "array_append": _f_array_append,
+    "array_prepend": _f_array_prepend,
+})
+
+
+def eval_map_hof(ev, e, chunk):
+    """transform_keys/transform_values/map_filter: lambda body over the flat
+    (key, value) entry columns — params at 0,1, enclosing row columns at 2+."""
+    from .chunk import Chunk
+    from .column import MapColumn
+    from .eval import broadcast
+
+    name = e.name.lower()
+    m = _bcast(ev.eval(e.args[0], chunk), chunk)
+    lam = e.args[1]
+    seg = m.segment_ids()
+    total = len(m.keys)
+    dev = m.device
+    flat_cols = [m.keys, m.values] + [c.gather(seg) if c is not None else None
+                                      for c in chunk.columns]
+    flat = Chunk(flat_cols, [f"__m{i}" for i in range(len(flat_cols))],
+                 chunk.partitioning)
+    flat.forced_rows = total
+    res = broadcast(ev.eval(lam.body, flat), total, dev)
+    if name == "transform_keys":
+        return MapColumn(m.offsets, res, m.values, m.validity)
+    if name == "transform_values":
+        return MapColumn(m.offsets, m.keys, res, m.validity)
+    # map_filter
+    keep = res.data.to(torch.bool) & res.valid_mask()
+    kept = torch.nonzero(keep, as_tuple=False).flatten()
+    new_seg = seg.index_select(0, kept)
+    lens = torch.bincount(new_seg, minlength=len(m))
+    offs = torch.zeros(len(m) + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(lens, 0, out=offs[1:])
+    return MapColumn(offs, m.keys.gather(kept), m.values.gather(kept),
+                     m.validity)

@@ -1196,3 +1196,93 @@ def _f_from_json(args, out, chunk, ev):
 
 
 _IMPLS["from_json"] = _f_from_json
+
+
+# -- datetime formatting (host path; Spark pattern subset) --------------
+
+def _spark_fmt_to_strftime(p: str) -> str:
+    # longest-first to keep yyyy from matching yy twice
+    subs = [("yyyy", "%Y"), ("yy", "%y"), ("MMMM", "%B"), ("MMM", "%b"),
+            ("MM", "%m"), ("dd", "%d"), ("HH", "%H"), ("hh", "%I"),
+            ("mm", "%M"), ("ss", "%S"), ("EEEE", "%A"), ("EEE", "%a"),
+            ("a", "%p"), ("DDD", "%j")]
+    out = ""
+    i = 0
+    while i < len(p):
+        for pat, rep in subs:
+            if p.startswith(pat, i):
+                out += rep
+                i += len(pat)
+                break
+        else:
+            out += p[i]
+            i += 1
+    return out
+
+
+def _f_date_format(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    fmt = _spark_fmt_to_strftime(_scalarize(args[1]).value)
+    vals = c.to_pylist()
+    res = [None if v is None else v.strftime(fmt) if hasattr(v, "strftime")
+           else str(v) for v in vals]
+    return StringColumn.from_pylist(res, device=chunk.device)
+
+
+def _f_from_unixtime(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    fmt = _spark_fmt_to_strftime(_scalarize(args[1]).value) \
+        if len(args) > 1 and _scalarize(args[1]) else "%Y-%m-%d %H:%M:%S"
+    vals = c.to_pylist()
+    res = [None if v is None else
+           _dt.datetime.fromtimestamp(int(v), _dt.timezone.utc).strftime(fmt)
+           for v in vals]
+    return StringColumn.from_pylist(res, device=chunk.device)
+
+
+def _f_to_timestamp(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    fmt = _spark_fmt_to_strftime(_scalarize(args[1]).value) \
+        if len(args) > 1 and _scalarize(args[1]) else None
+    vals = c.to_pylist()
+    res = []
+    for v in vals:
+        if v is None:
+            res.append(None)
+            continue
+        try:
+            if fmt:
+                dt = _dt.datetime.strptime(v, fmt)
+            else:
+                dt = _dt.datetime.fromisoformat(str(v))
+            res.append(int(dt.replace(tzinfo=_dt.timezone.utc).timestamp() * 1_000_000))
+        except (ValueError, TypeError):
+            res.append(None)
+    validity = None
+    if any(r is None for r in res):
+        validity = torch.tensor([0 if r is None else 1 for r in res],
+                                dtype=torch.uint8, device=chunk.device)
+    data = torch.tensor([0 if r is None else r for r in res],
+                        dtype=torch.int64, device=chunk.device)
+    return Column(T.TIMESTAMP, data, validity)
+
+
+def _f_datepart(args, out, chunk, ev):
+    unit = str(_scalarize(args[0]).value).lower()
+    fmap = {"year": "year", "yyyy": "year", "yy": "year", "month": "month",
+            "mon": "month", "mm": "month", "day": "day", "dd": "day",
+            "dayofweek": "dayofweek", "dow": "dayofweek", "doy": "dayofyear",
+            "hour": "hour", "minute": "minute", "second": "second",
+            "quarter": "quarter", "week": "weekofyear"}
+    name = fmap.get(unit)
+    if name is None:
+        raise NotImplementedError(f"date_part unit {unit}")
+    return dispatch_function(name, [args[1]], out, chunk, ev)
+
+
+_IMPLS["date_format"] = _f_date_format
+_IMPLS["from_unixtime"] = _f_from_unixtime
+_IMPLS["to_timestamp"] = _f_to_timestamp
+_IMPLS["try_to_timestamp"] = _f_to_timestamp
+_IMPLS["datepart"] = _f_datepart
+_IMPLS["date_part"] = _f_datepart

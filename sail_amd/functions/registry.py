@@ -215,11 +215,18 @@ _reg("map_contains_key", _bool)
 _reg("struct named_struct get_field", lambda a: T.NULL)  # typed structurally in the resolver
 _reg("get_json_object to_json schema_of_json json_tuple", _string)
 _reg("from_json", lambda a: T.NULL)  # typed structurally in the resolver
+_reg("date_format from_unixtime", _string)
+_reg("to_timestamp try_to_timestamp", lambda a: T.TIMESTAMP)
+_reg("datepart date_part", _i32)
 _reg("sort_array array_sort array_distinct array_remove array_compact flatten "
      "slice array_repeat shuffle", _same)
 _reg("array_join", _string)
 _reg("array_position", _i64)
 _reg("arrays_overlap", _bool)
+_reg("array_union array_intersect array_except", _same)
+_reg("array_append array_prepend", _same)
+_reg("array_repeat", lambda a: T.ArrayType(a[0]))
+_reg("flatten", lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
 _reg("sequence", lambda a: T.ArrayType(a[0]))
 _reg("explode explode_outer posexplode", lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
 

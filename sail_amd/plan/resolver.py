@@ -1037,7 +1037,8 @@ class Resolver:
         return self._type_expr(out)
 
     _HOF = {"transform", "filter", "exists", "forall", "array_filter",
-            "zip_with", "aggregate", "reduce"}
+            "zip_with", "aggregate", "reduce",
+            "transform_keys", "transform_values", "map_filter"}
 
     def _resolve_hof(self, e: S.Func, scope: Scope) -> S.Expr:
         """Higher-order array functions with lambdas (ref: sail-plan
@@ -1051,6 +1052,19 @@ class Resolver:
             return self._resolve_zip_with(e, scope)
         if name in ("aggregate", "reduce"):
             return self._resolve_reduce(e, scope)
+        if name in ("transform_keys", "transform_values", "map_filter"):
+            m = self._expr(e.args[0], scope)
+            if not isinstance(m.dtype, T.MapType):
+                raise ResolutionError(f"{name} expects a map argument")
+            blam = self._bind_lambda(e.args[1], [m.dtype.key, m.dtype.value], scope)
+            if name == "map_filter":
+                blam = S.Lambda(blam.params, _coerce_to_bool(blam.body), T.BOOL)
+                t = m.dtype
+            elif name == "transform_keys":
+                t = T.MapType(blam.dtype, m.dtype.value)
+            else:
+                t = T.MapType(m.dtype.key, blam.dtype)
+            return S.Func(name, [m, blam], t)
         arr = self._expr(e.args[0], scope)
         if not isinstance(arr.dtype, T.ArrayType):
             raise ResolutionError(f"{name} expects an array argument")

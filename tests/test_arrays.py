@@ -187,3 +187,30 @@ def test_lambda_pushdown_through_projection(s):
     q2 = ("SELECT v FROM (SELECT sequence(1, v) AS a, v FROM lpd) x "
           "WHERE size(filter(a, e -> e > 1)) > 3")
     assert s.sql(q2).collect() == [(5,)]
+
+
+def test_array_set_ops(s):
+    rows = s.sql("SELECT array_union(array(1,2), array(2,3)), "
+                 "array_intersect(array(1,2), array(2,3)), "
+                 "array_except(array(1,2), array(2,3))").collect()
+    assert rows == [([1, 2, 3], [2], [1])]
+    rows = s.sql("SELECT array_remove(array(1,2,1), 1), "
+                 "array_compact(array(1, NULL, 2)), "
+                 "flatten(array(array(1,2), array(3))), array_repeat('x', 3), "
+                 "array_append(array(1), 2), array_prepend(array(1), 0)").collect()
+    assert rows == [([2], [1, 2], [1, 2, 3], ["x", "x", "x"], [1, 2], [0, 1])]
+
+
+def test_map_lambdas(s):
+    rows = s.sql("SELECT transform_values(map('a',1,'b',2), (k,v) -> v * 10), "
+                 "transform_keys(map('a',1), (k, v) -> upper(k)), "
+                 "map_filter(map('a',1,'b',2), (k,v) -> v > 1)").collect()
+    assert rows == [({"a": 10, "b": 20}, {"A": 1}, {"b": 2})]
+
+
+def test_datetime_formatting(s):
+    rows = s.sql("SELECT date_format(DATE '2024-03-05', 'yyyy-MM-dd'), "
+                 "from_unixtime(86400), "
+                 "datepart('year', DATE '2024-03-05'), "
+                 "unix_timestamp(to_timestamp('1970-01-02 00:00:00'))").collect()
+    assert rows == [("2024-03-05", "1970-01-02 00:00:00", 2024, 86400)]

@@ -440,7 +440,10 @@ def _read_with_views(resolver, p, outer):
         import copy
 
         sub = copy.deepcopy(vp)
-        resolved = resolver._plan(sub, None)
+        # a view stored as an already-resolved plan (DataFrame composition
+        # views) must NOT be re-resolved or re-optimized in place: resolved
+        # trees use BoundRef ordinals that a second bind pass can corrupt
+        resolved = resolver._plan(sub, None) if sub.schema is None else sub
         out = S.SubqueryAlias(input=resolved, alias=p.table.split(".")[-1])
         out.schema = resolved.schema
         return out
@@ -577,6 +580,98 @@ class DataFrame:
                 exprs.append(n)
         return self.session.sql(f"SELECT {', '.join(exprs)} FROM {name}")
 
+    # -- PySpark-style composition (SQL-backed; ref: the Connect client
+    # DataFrame API surface) ----------------------------------------------
+    def select(self, *cols) -> "DataFrame":
+        name = self._as_view()
+        exprs = ", ".join(cols) if cols else "*"
+        return self.session.sql(f"SELECT {exprs} FROM {name}")
+
+    def select_expr(self, *exprs) -> "DataFrame":
+        return self.select(*exprs)
+
+    selectExpr = select_expr
+
+    def filter(self, condition: str) -> "DataFrame":
+        name = self._as_view()
+        return self.session.sql(f"SELECT * FROM {name} WHERE {condition}")
+
+    where = filter
+
+    def with_column(self, name: str, expr: str) -> "DataFrame":
+        view = self._as_view()
+        keep = [n for n, _ in self.plan.schema if n.lower() != name.lower()]
+        cols = ", ".join(keep + [f"{expr} AS {name}"])
+        return self.session.sql(f"SELECT {cols} FROM {view}")
+
+    withColumn = with_column
+
+    def with_column_renamed(self, old: str, new: str) -> "DataFrame":
+        view = self._as_view()
+        cols = ", ".join(f"{n} AS {new}" if n.lower() == old.lower() else n
+                         for n, _ in self.plan.schema)
+        return self.session.sql(f"SELECT {cols} FROM {view}")
+
+    withColumnRenamed = with_column_renamed
+
+    def drop(self, *cols) -> "DataFrame":
+        view = self._as_view()
+        low = {c.lower() for c in cols}
+        keep = [n for n, _ in self.plan.schema if n.lower() not in low]
+        return self.session.sql(f"SELECT {', '.join(keep)} FROM {view}")
+
+    def distinct(self) -> "DataFrame":
+        return self.session.sql(f"SELECT DISTINCT * FROM {self._as_view()}")
+
+    def order_by(self, *keys) -> "DataFrame":
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} ORDER BY {', '.join(keys)}")
+
+    orderBy = sort = order_by
+
+    def limit(self, n: int) -> "DataFrame":
+        return self.session.sql(f"SELECT * FROM {self._as_view()} LIMIT {int(n)}")
+
+    def group_by(self, *keys) -> "GroupedData":
+        return GroupedData(self, list(keys))
+
+    groupBy = group_by
+
+    def join(self, other: "DataFrame", on, how: str = "inner") -> "DataFrame":
+        lv, rv = self._as_view(), other._as_view()
+        if isinstance(on, str):
+            on = [on]
+        how_sql = {"inner": "JOIN", "left": "LEFT JOIN", "right": "RIGHT JOIN",
+                   "full": "FULL JOIN", "outer": "FULL JOIN",
+                   "semi": "LEFT SEMI JOIN", "anti": "LEFT ANTI JOIN",
+                   "cross": "CROSS JOIN"}[how.lower()]
+        using = f" USING ({', '.join(on)})" if on else ""
+        return self.session.sql(f"SELECT * FROM {lv} {how_sql} {rv}{using}")
+
+    def union(self, other: "DataFrame") -> "DataFrame":
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} UNION ALL SELECT * FROM {other._as_view()}")
+
+    unionAll = union
+
+    def agg(self, *exprs) -> "DataFrame":
+        return self.session.sql(
+            f"SELECT {', '.join(exprs)} FROM {self._as_view()}")
+
+    def first(self):
+        rows = self.limit(1).collect()
+        return rows[0] if rows else None
+
+    def head(self, n: int = 1):
+        return self.limit(n).collect()
+
+    def take(self, n: int):
+        return self.limit(n).collect()
+
+    @property
+    def columns(self):
+        return [n for n, _ in self.plan.schema]
+
     def checkpoint(self, eager: bool = True) -> "DataFrame":
         """Persist this DataFrame's result and return a DataFrame reading the
         persisted copy — truncates the lineage like Spark's df.checkpoint()
@@ -608,6 +703,26 @@ class DataFrame:
     @property
     def write(self) -> "DataFrameWriter":
         return DataFrameWriter(self)
+
+
+class GroupedData:
+    """df.group_by(...).agg(...) (SQL-backed)."""
+
+    def __init__(self, df: "DataFrame", keys):
+        self._df = df
+        self._keys = keys
+
+    def agg(self, *exprs) -> "DataFrame":
+        view = self._df._as_view()
+        sel = ", ".join(list(self._keys) + list(exprs))
+        gb = ", ".join(self._keys) if self._keys else ""
+        q = f"SELECT {sel} FROM {view}"
+        if gb:
+            q += f" GROUP BY {gb}"
+        return self._df.session.sql(q)
+
+    def count(self) -> "DataFrame":
+        return self.agg("count(*) AS count")
 
 
 class _MaterializedDataFrame(DataFrame):

@@ -334,3 +334,32 @@ def test_hash_group_ids_matches_sort_path():
     want = s2.sql("SELECT k, sum(v) AS sv, count(*) AS c FROM hg GROUP BY k "
                   "ORDER BY sv DESC, k LIMIT 100").collect()
     assert rows == want
+
+
+@pytest.mark.gpu
+def test_feature_batch_on_gpu():
+    """Device coverage for the round's feature additions in one session:
+    array set ops, map lambdas, window frames, topk, pivot, recursive CTE."""
+    import sail_amd
+
+    s = sail_amd.SessionContext(device="cuda")
+    assert s.sql("SELECT array_union(array(1,2), array(2,3)), "
+                 "array_except(array(1,2), array(2,3))").collect() == [([1, 2, 3], [1])]
+    assert s.sql("SELECT transform(sequence(1, 4), x -> x * x)").collect() == [([1, 4, 9, 16],)]
+    assert s.sql("SELECT map_filter(map('a',1,'b',2), (k,v) -> v > 1)").collect() == [({"b": 2},)]
+    s.create_dataframe({"o": [1, 2, 3, 4], "v": [10, 20, 30, 40]}, name="wf")
+    assert s.sql("SELECT o, sum(v) OVER (ORDER BY o ROWS BETWEEN 1 PRECEDING AND "
+                 "1 FOLLOWING) FROM wf ORDER BY o").collect() == [
+        (1, 30), (2, 60), (3, 90), (4, 70)]
+    rows = s.sql("WITH RECURSIVE n(x) AS (SELECT 1 UNION ALL SELECT x + 1 FROM n "
+                 "WHERE x < 100) SELECT sum(x) FROM n").collect()
+    assert rows == [(5050,)]
+    s.create_dataframe({"g": ["x", "x", "y"], "k": ["a", "b", "a"], "v": [1, 2, 3]},
+                       name="pv")
+    assert s.sql("SELECT * FROM pv PIVOT (sum(v) FOR k IN ('a', 'b')) ORDER BY g").collect() == [
+        ("x", 1, 2), ("y", 3, None)]
+    # topk path over a big range (selection, not full sort)
+    assert s.sql("SELECT id FROM range(5000000) ORDER BY id DESC LIMIT 3").collect() == [
+        (4999999,), (4999998,), (4999997,)]
+    # struct/json
+    assert s.sql("SELECT to_json(named_struct('a', 1))").collect() == [('{"a":1}',)]

@@ -29,6 +29,15 @@ CASES = {
                   "WHERE t1.k IN (SELECT k FROM t3 WHERE v2 > 0)"),
     "prune_columns": "SELECT a FROM (SELECT * FROM t1 JOIN t2 ON t1.k = t2.k) x",
     "order_by_agg": "SELECT k, count(*) FROM t1 GROUP BY k ORDER BY count(*) DESC LIMIT 3",
+    "pivot": "SELECT * FROM (SELECT k, v, a FROM t2 JOIN t1 USING (k)) p "
+             "PIVOT (sum(a) FOR v IN (1, 2))",
+    "recursive_cte": ("WITH RECURSIVE n(x) AS (SELECT 1 UNION ALL "
+                      "SELECT x + 1 FROM n WHERE x < 5) SELECT sum(x) FROM n"),
+    "lateral_view": ("SELECT k, e FROM t1 LATERAL VIEW explode(sequence(1, a)) "
+                     "ex AS e"),
+    "grouping_sets": ("SELECT k, a, count(*) FROM t1 GROUP BY ROLLUP(k, a)"),
+    "window_frame": ("SELECT a, sum(a) OVER (ORDER BY a ROWS BETWEEN 1 "
+                     "PRECEDING AND CURRENT ROW) FROM t1"),
 }
 
 

@@ -182,6 +182,62 @@ class DeltaLog:
         with open(os.path.join(self.log_path, "_last_checkpoint"), "w") as f:
             json.dump({"version": version, "size": len(kinds)}, f)
 
+    def version_times(self):
+        """[(version, commit_time_ms)] from file mtimes (commitInfo actions
+        are optional in the protocol; mtime is the portable signal)."""
+        out = []
+        for v in self.versions():
+            p = os.path.join(self.log_path, f"{v:020d}.json")
+            out.append((v, int(os.path.getmtime(p) * 1000)))
+        return out
+
+    def history(self) -> List[dict]:
+        """DESCRIBE HISTORY rows: version, timestamp, operation summary."""
+        rows = []
+        for v, t in self.version_times():
+            ops = {"adds": 0, "removes": 0, "metaData": 0}
+            with open(os.path.join(self.log_path, f"{v:020d}.json")) as f:
+                for line in f:
+                    if not line.strip():
+                        continue
+                    a = json.loads(line)
+                    if "add" in a:
+                        ops["adds"] += 1
+                    elif "remove" in a:
+                        ops["removes"] += 1
+                    elif "metaData" in a:
+                        ops["metaData"] += 1
+            if v == 0:
+                op = "CREATE TABLE"
+            elif ops["removes"] and ops["adds"]:
+                op = "OVERWRITE/MERGE"
+            elif ops["adds"]:
+                op = "WRITE (append)"
+            else:
+                op = "DELETE"
+            rows.append({"version": v, "timestamp_ms": t, "operation": op,
+                         "num_added_files": ops["adds"],
+                         "num_removed_files": ops["removes"]})
+        return rows
+
+    def vacuum(self, retention_hours: float = 168.0, dry_run: bool = False):
+        """Delete part files no longer referenced by the CURRENT snapshot
+        and older than the retention window (ref: sail-delta-lake log
+        cleanup / vacuum semantics)."""
+        _, live, _, _ = self.snapshot()
+        live_set = set(live)
+        cutoff = time.time() - retention_hours * 3600.0
+        removed = []
+        for f in os.listdir(self.path):
+            if not f.endswith(".parquet") or f in live_set:
+                continue
+            full = os.path.join(self.path, f)
+            if os.path.getmtime(full) <= cutoff:
+                removed.append(f)
+                if not dry_run:
+                    os.remove(full)
+        return removed
+
     def commit(self, version: int, actions: List[dict]):
         """Atomic O_EXCL create; raises FileExistsError on concurrent commit
         (the caller retries with a fresh version — optimistic concurrency)."""
@@ -212,13 +268,30 @@ def _protocol_action() -> dict:
 
 def infer_schema(paths: List[str], options: Dict[str, str] = None):
     log = DeltaLog(paths[0])
-    schema, _, _, _ = log.snapshot(_version_opt(options))
+    schema, _, _, _ = log.snapshot(_version_opt(options, log))
     return schema
 
 
-def _version_opt(options) -> Optional[int]:
+def _version_opt(options, log: "DeltaLog" = None) -> Optional[int]:
     if options and options.get("versionAsOf") is not None:
         return int(options["versionAsOf"])
+    if options and options.get("timestampAsOf") is not None and log is not None:
+        ts = options["timestampAsOf"]
+        import datetime as _dt2
+
+        if isinstance(ts, str):
+            tsm = int(_dt2.datetime.fromisoformat(ts).replace(
+                tzinfo=_dt2.timezone.utc).timestamp() * 1000)
+        else:
+            tsm = int(float(ts) * 1000)
+        best = None
+        for v, t in log.version_times():
+            if t <= tsm:
+                best = v
+        if best is None:
+            raise FileNotFoundError(
+                f"no delta version at or before {ts}")
+        return best
     return None
 
 
@@ -227,7 +300,7 @@ def read(paths: List[str], schema, device, options: Dict[str, str]):
     from ..engine.column import Table
 
     log = DeltaLog(paths[0])
-    tbl_schema, files, _, _ = log.snapshot(_version_opt(options))
+    tbl_schema, files, _, _ = log.snapshot(_version_opt(options, log))
     if not files:
         from ..engine.column import Column
 

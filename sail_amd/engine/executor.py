@@ -953,6 +953,35 @@ class Executor:
         return Chunk([StringColumn.from_pylist([p.key]), StringColumn.from_pylist([val])],
                      ["key", "value"])
 
+    @staticmethod
+    def _delta_path(name: str) -> str:
+        head, _, rest = name.partition(".")
+        if head.lower() != "delta" or not rest:
+            raise ExecError(f"{name}: expected delta.`/path`")
+        return rest
+
+    def _x_VacuumTable(self, p: S.VacuumTable) -> Chunk:
+        from ..datasource.delta import DeltaLog
+
+        log = DeltaLog(self._delta_path(p.name))
+        removed = log.vacuum(p.retention_hours if p.retention_hours is not None
+                             else 168.0, p.dry_run)
+        return Chunk([StringColumn.from_pylist(removed, dict_encode=False)],
+                     ["removed_file"])
+
+    def _x_DescribeHistory(self, p: S.DescribeHistory) -> Chunk:
+        from ..datasource.delta import DeltaLog
+
+        rows = DeltaLog(self._delta_path(p.name)).history()
+        return Chunk([
+            Column.from_values([r["version"] for r in rows], T.I64),
+            Column.from_values([r["timestamp_ms"] for r in rows], T.I64),
+            StringColumn.from_pylist([r["operation"] for r in rows], dict_encode=False),
+            Column.from_values([r["num_added_files"] for r in rows], T.I64),
+            Column.from_values([r["num_removed_files"] for r in rows], T.I64),
+        ], ["version", "timestamp_ms", "operation", "num_added_files",
+            "num_removed_files"])
+
     def _x_AlterTable(self, p: S.AlterTable) -> Chunk:
         cat = self.ctx.session.catalog
         key = cat._key(p.name)

@@ -906,6 +906,16 @@ class Resolver:
         p.schema = [("namespace", T.STRING), ("tableName", T.STRING), ("isTemporary", T.BOOL)]
         return p
 
+    def _p_VacuumTable(self, p: S.VacuumTable, outer):
+        p.schema = [("removed_file", T.STRING)]
+        return p
+
+    def _p_DescribeHistory(self, p: S.DescribeHistory, outer):
+        p.schema = [("version", T.I64), ("timestamp_ms", T.I64),
+                    ("operation", T.STRING), ("num_added_files", T.I64),
+                    ("num_removed_files", T.I64)]
+        return p
+
     def _p_AlterTable(self, p: S.AlterTable, outer):
         if self.catalog.table_schema(p.name) is None:
             raise ResolutionError(f"table not found: {p.name}")

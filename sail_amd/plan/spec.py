@@ -687,6 +687,18 @@ class SetConfig(Command):
 
 
 @dataclass
+class VacuumTable(Command):
+    name: str = ""
+    retention_hours: Optional[float] = None
+    dry_run: bool = False
+
+
+@dataclass
+class DescribeHistory(Command):
+    name: str = ""
+
+
+@dataclass
 class AlterTable(Command):
     """ALTER TABLE: add/drop/rename columns, rename table (in-memory
     catalog + delta metadata — ref: spec CommandNode AlterTable subset)."""

@@ -198,8 +198,20 @@ class Parser:
                     while self.eat_op(","):
                         cols.append(self.ident())
             return S.AnalyzeTable(name=name, columns=cols)
+        if self.at_kw("VACUUM"):
+            self.next()
+            name = self._qualified_name()
+            hours = None
+            if self.eat_kw("RETAIN"):
+                t = self.next()
+                hours = float(t.value.split("#")[0])
+                self.eat_kw("HOURS")
+            dry = bool(self.eat_kw("DRY")) and bool(self.eat_kw("RUN"))
+            return S.VacuumTable(name=name, retention_hours=hours, dry_run=dry)
         if self.at_kw("DESCRIBE", "DESC"):
             self.next()
+            if self.eat_kw("HISTORY"):
+                return S.DescribeHistory(name=self._qualified_name())
             self.eat_kw("TABLE")
             self.eat_kw("EXTENDED")
             return S.DescribeTable(name=self._qualified_name())

@@ -157,3 +157,25 @@ def test_merge_not_matched_by_source(s, tmp_path):
           "WHEN MATCHED THEN UPDATE SET v = u.v "
           "WHEN NOT MATCHED BY SOURCE THEN DELETE")
     assert s.sql(f"SELECT * FROM delta.`{p}`").collect() == [(1, 9.0)]
+
+
+def test_delta_history_vacuum_timestamp(s, tmp_path):
+    p = str(tmp_path / "hv")
+    s.create_dataframe({"k": [1], "v": [1.0]}, schema={"k": T.I64, "v": T.F64},
+                       name="hv_seed")
+    s.table("hv_seed").write.format("delta").mode("overwrite").save(p)
+    s.sql(f"INSERT INTO delta.`{p}` VALUES (2, 2.0)")
+    s.table("hv_seed").write.format("delta").mode("overwrite").save(p)
+    hist = s.sql(f"DESCRIBE HISTORY delta.`{p}`").collect()
+    assert [h[0] for h in hist] == [0, 1, 2]
+    assert hist[0][2] == "CREATE TABLE" and hist[2][2] == "OVERWRITE/MERGE"
+    removed = s.sql(f"VACUUM delta.`{p}` RETAIN 0 HOURS").collect()
+    assert len(removed) == 2  # the two orphaned part files
+    assert s.sql(f"SELECT * FROM delta.`{p}`").collect() == [(1, 1.0)]
+    # retention guard: fresh files survive the default window
+    assert s.sql(f"VACUUM delta.`{p}`").collect() == []
+    import datetime
+
+    now = datetime.datetime.now().isoformat()
+    df = s.read.format("delta").option("timestampAsOf", now).load(p)
+    assert df.collect() == [(1, 1.0)]

@@ -28,6 +28,7 @@ def register_format(name: str, source) -> None:
 
 
 def infer_source_schema(fmt: str, paths: List[str], options: Dict[str, str]):
+    fmt = fmt.lower()
     from . import parquet_io, csv_io
 
     if fmt == "parquet":
@@ -42,12 +43,25 @@ def infer_source_schema(fmt: str, paths: List[str], options: Dict[str, str]):
         from . import delta
 
         return delta.infer_schema(paths, options)
+    if fmt == "text":
+        from . import text_io
+
+        return text_io.text_infer_schema(paths, options)
+    if fmt in ("binary", "binaryfile"):
+        from . import text_io
+
+        return text_io.binary_infer_schema(paths, options)
+    if fmt == "arrow":
+        from . import text_io
+
+        return text_io.arrow_infer_schema(paths, options)
     if fmt in _USER_FORMATS:
         return _USER_FORMATS[fmt].infer_schema(paths, options)
     raise ValueError(f"unsupported format {fmt}")
 
 
 def read_source(fmt: str, paths: List[str], options: Dict[str, str], schema, device):
+    fmt = fmt.lower()
     from . import parquet_io, csv_io
 
     if fmt == "parquet":
@@ -62,12 +76,25 @@ def read_source(fmt: str, paths: List[str], options: Dict[str, str], schema, dev
         from . import delta
 
         return delta.read(paths, schema, device, options)
+    if fmt == "text":
+        from . import text_io
+
+        return text_io.text_read(paths, schema, device, options)
+    if fmt in ("binary", "binaryfile"):
+        from . import text_io
+
+        return text_io.binary_read(paths, schema, device, options)
+    if fmt == "arrow":
+        from . import text_io
+
+        return text_io.arrow_read(paths, schema, device, options)
     if fmt in _USER_FORMATS:
         return _USER_FORMATS[fmt].read(paths, schema, device, options)
     raise ValueError(f"unsupported format {fmt}")
 
 
 def write_source(fmt: str, path: str, chunk, mode: str, options, partition_by):
+    fmt = fmt.lower()
     from . import parquet_io, csv_io
 
     if fmt == "parquet":
@@ -78,6 +105,14 @@ def write_source(fmt: str, path: str, chunk, mode: str, options, partition_by):
         from . import delta
 
         return delta.write(path, chunk, mode, options)
+    if fmt == "text":
+        from . import text_io
+
+        return text_io.text_write(path, chunk, mode, options)
+    if fmt == "arrow":
+        from . import text_io
+
+        return text_io.arrow_write(path, chunk, mode, options)
     if fmt in _USER_FORMATS and hasattr(_USER_FORMATS[fmt], "write"):
         return _USER_FORMATS[fmt].write(path, chunk, mode, options)
     raise ValueError(f"unsupported write format {fmt}")

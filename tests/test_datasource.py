@@ -46,3 +46,28 @@ def test_parquet_predicate_after_read(s, tmp_path):
     s.table("t").write.mode("overwrite").parquet(p)
     rows = s.sql(f"SELECT a FROM parquet.`{p}` WHERE d >= DATE '2023-01-01' ORDER BY a").collect()
     assert rows == [(1,), (2,)]
+
+
+def test_text_format(session, tmp_path):
+    p = tmp_path / "a.txt"
+    p.write_text("hello\nworld\n")
+    df = session.read.format("text").load(str(p))
+    assert df.collect() == [("hello",), ("world",)]
+    out = tmp_path / "out"
+    df.write.format("text").save(str(out))
+    assert session.read.format("text").load(str(out)).collect() == [
+        ("hello",), ("world",)]
+
+
+def test_binary_file_format(session, tmp_path):
+    p = tmp_path / "b.bin"
+    p.write_bytes(b"\x00\x01binary")
+    rows = session.read.format("binaryFile").load(str(p)).collect()
+    assert rows[0][1] == 8
+
+
+def test_arrow_format_roundtrip(session, tmp_path):
+    d = str(tmp_path / "ar")
+    df = session.create_dataframe({"x": [1, 2], "y": ["a", "b"]})
+    df.write.format("arrow").save(d)
+    assert session.read.format("arrow").load(d).collect() == [(1, "a"), (2, "b")]

@@ -46,19 +46,93 @@ def _engine_type(at: pa.DataType) -> T.DataType:
     raise ValueError(f"unsupported parquet type {at}")
 
 
+def _partition_info(root: str, files: List[str]):
+    """Hive-style `key=value` path partitioning (ref: sail-data-source
+    src/listing/ partition handling): per-file partition values, inferred
+    int64 where every value parses, else string."""
+    import urllib.parse
+
+    keys: List[str] = []
+    per_file: List[dict] = []
+    for f in files:
+        rel = os.path.relpath(f, root)
+        vals = {}
+        for seg in rel.split(os.sep)[:-1]:
+            if "=" in seg:
+                k, _, v = seg.partition("=")
+                vals[k] = urllib.parse.unquote(v)
+                if k not in keys:
+                    keys.append(k)
+        per_file.append(vals)
+    if not keys or any(len(v) != len(keys) for v in per_file):
+        return [], []
+    types = []
+    for k in keys:
+        try:
+            for v in per_file:
+                int(v[k])
+            types.append(T.I64)
+        except ValueError:
+            types.append(T.STRING)
+    return list(zip(keys, types)), per_file
+
+
 def infer_schema(paths: List[str]) -> List[Tuple[str, T.DataType]]:
     files = _expand(paths)
     if not files:
         raise FileNotFoundError(f"no parquet files under {paths}")
     sch = pq.read_schema(files[0])
-    return [(f.name, _engine_type(f.type)) for f in sch]
+    out = [(f.name, _engine_type(f.type)) for f in sch]
+    if len(paths) == 1 and os.path.isdir(paths[0]):
+        pkeys, _ = _partition_info(paths[0], files)
+        out += pkeys
+    return out
 
 
 def read(paths: List[str], schema, device, options: Dict[str, str]):
     files = _expand(paths)
-    cols = [n for n, _ in schema] if schema else None
-    tbl = pq.read_table(files, columns=cols)
-    return arrow_to_table(tbl, device=device)
+    pkeys, pvals = ([], [])
+    if len(paths) == 1 and os.path.isdir(paths[0]):
+        pkeys, pvals = _partition_info(paths[0], files)
+    if not pkeys:
+        cols = [n for n, _ in schema] if schema else None
+        tbl = pq.read_table(files, columns=cols)
+        return arrow_to_table(tbl, device=device)
+    # partition pruning: equality filters pushed down as options
+    # "partition.<key>" (see plan/rules/pushdown.py)
+    if options:
+        for k, t in pkeys:
+            want = options.get(f"partition.{k}")
+            if want is not None:
+                keep = [i for i, v in enumerate(pvals) if v[k] == str(want)]
+                files = [files[i] for i in keep]
+                pvals = [pvals[i] for i in keep]
+    data_names = {f.name for f in pq.read_schema(files[0])} if files else set()
+    from ..engine.column import Column, StringColumn, Table
+
+    parts = []
+    for f, v in zip(files, pvals):
+        t = pq.read_table(f, columns=[n for n, _ in schema if n in data_names]
+                          if schema else None)
+        parts.append((t, v))
+    if not parts:
+        cols = {}
+        for n, t in (schema or []):
+            cols[n] = Column.from_values([], t, device=device)
+        return Table(cols)
+    from ..engine.executor import concat_columns
+
+    tables = [arrow_to_table(t, device=device) for t, _ in parts]
+    out_cols = {}
+    for n in tables[0].columns:
+        out_cols[n] = concat_columns([t.columns[n] for t in tables])             if len(tables) > 1 else tables[0].columns[n]
+    for k, kt in pkeys:
+        vals = []
+        for (t, v) in parts:
+            pv = int(v[k]) if kt == T.I64 else v[k]
+            vals.extend([pv] * t.num_rows)
+        out_cols[k] = Column.from_values(vals, kt, device=device)
+    return Table(out_cols)
 
 
 def write(path: str, chunk, mode: str, options: Dict[str, str]):

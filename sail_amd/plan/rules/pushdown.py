@@ -130,6 +130,24 @@ def _push(plan: S.Plan, conds: List[S.Expr]) -> S.Plan:
         out.schema = plan.schema
         return _apply(out, stuck)
 
+    if isinstance(plan, S.DataSourceRead) and plan.format == "parquet":
+        # hive-partition pruning hint: equality conjuncts on scan columns are
+        # COPIED (not moved) into scan options; the reader skips files whose
+        # key=value path disagrees, the filter still applies for correctness
+        for c in conds:
+            if isinstance(c, S.BinaryOp) and c.op == "=":
+                def _strip(x):
+                    return x.child if isinstance(x, S.Cast) else x
+
+                ref, lit = _strip(c.left), _strip(c.right)
+                if isinstance(lit, S.BoundRef) and isinstance(ref, S.Literal):
+                    ref, lit = lit, ref
+                if isinstance(ref, S.BoundRef) and isinstance(lit, S.Literal) \
+                        and not isinstance(lit.value, (list, dict)):
+                    plan.options = dict(plan.options or {})
+                    plan.options[f"partition.{ref.name}"] = str(lit.value)
+        return _apply(plan, conds)
+
     # leaves and commands: recurse into children generically
     for attr in ("input",):
         if hasattr(plan, attr) and getattr(plan, attr) is not None and isinstance(getattr(plan, attr), S.Plan):

@@ -71,3 +71,53 @@ def test_arrow_format_roundtrip(session, tmp_path):
     df = session.create_dataframe({"x": [1, 2], "y": ["a", "b"]})
     df.write.format("arrow").save(d)
     assert session.read.format("arrow").load(d).collect() == [(1, "a"), (2, "b")]
+
+
+def test_hive_partitioned_parquet(session, tmp_path):
+    import os
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    d = str(tmp_path / "part")
+    for y in (2023, 2024):
+        for m in ("01", "02"):
+            sub = os.path.join(d, f"year={y}", f"month={m}")
+            os.makedirs(sub)
+            pq.write_table(pa.table({"v": [y * 100 + int(m)]}),
+                           os.path.join(sub, "p.parquet"))
+    df = session.read.format("parquet").load(d)
+    assert [n for n, _ in df.plan.schema] == ["v", "year", "month"]
+    assert sorted(df.collect()) == [
+        (202301, 2023, 1), (202302, 2023, 2), (202401, 2024, 1), (202402, 2024, 2)]
+    assert session.sql(
+        f"SELECT v FROM parquet.`{d}` WHERE year = 2024 AND month = 2").collect() == [
+        (202402,)]
+
+
+def test_partition_pruning_skips_files(session, tmp_path, monkeypatch):
+    import os
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    import sail_amd.datasource.parquet_io as pio
+
+    d = str(tmp_path / "pp")
+    for y in (2023, 2024):
+        sub = os.path.join(d, f"year={y}")
+        os.makedirs(sub)
+        pq.write_table(pa.table({"v": [y]}), os.path.join(sub, "p.parquet"))
+    calls = []
+    orig = pio.pq.read_table
+
+    def spy(f, *a, **k):
+        calls.append(str(f))
+        return orig(f, *a, **k)
+
+    monkeypatch.setattr(pio.pq, "read_table", spy)
+    assert session.sql(f"SELECT sum(v) FROM parquet.`{d}` WHERE year = 2023").collect() == [
+        (2023,)]
+    read_dirs = {os.path.basename(os.path.dirname(c)) for c in calls
+                 if c.endswith(".parquet")}
+    assert read_dirs == {"year=2023"}

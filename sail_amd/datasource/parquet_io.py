@@ -104,7 +104,15 @@ def read(paths: List[str], schema, device, options: Dict[str, str]):
         for k, t in pkeys:
             want = options.get(f"partition.{k}")
             if want is not None:
-                keep = [i for i, v in enumerate(pvals) if v[k] == str(want)]
+                def _match(raw, want=want, t=t):
+                    if t == T.I64:
+                        try:
+                            return int(raw) == int(float(want))
+                        except (ValueError, TypeError):
+                            return False
+                    return raw == str(want)
+
+                keep = [i for i, v in enumerate(pvals) if _match(v[k])]
                 files = [files[i] for i in keep]
                 pvals = [pvals[i] for i in keep]
     data_names = {f.name for f in pq.read_schema(files[0])} if files else set()

@@ -95,6 +95,8 @@ def read_source(fmt: str, paths: List[str], options: Dict[str, str], schema, dev
 
 def write_source(fmt: str, path: str, chunk, mode: str, options, partition_by):
     fmt = fmt.lower()
+    if partition_by and fmt == "parquet":
+        return _write_partitioned(path, chunk, mode, options, partition_by)
     from . import parquet_io, csv_io
 
     if fmt == "parquet":
@@ -116,3 +118,38 @@ def write_source(fmt: str, path: str, chunk, mode: str, options, partition_by):
     if fmt in _USER_FORMATS and hasattr(_USER_FORMATS[fmt], "write"):
         return _USER_FORMATS[fmt].write(path, chunk, mode, options)
     raise ValueError(f"unsupported write format {fmt}")
+
+
+def _write_partitioned(path, chunk, mode, options, partition_by):
+    """Hive-style partitioned parquet write: one `key=value/.../part.parquet`
+    per distinct partition tuple (ref: sail-data-source listing sink
+    planning)."""
+    import os
+
+    import torch
+
+    from . import parquet_io
+    from ..engine.aggregates import group_ids
+    from ..engine.chunk import Chunk as _Chunk
+
+    names = [n.lower() for n in chunk.names]
+    pidx = []
+    for k in partition_by:
+        if k.lower() not in names:
+            raise ValueError(f"partitionBy column {k} not in output")
+        pidx.append(names.index(k.lower()))
+    key_cols = [chunk.columns[i] for i in pidx]
+    gid, rep, ng = group_ids(key_cols)
+    data_idx = [i for i in range(len(chunk.columns)) if i not in pidx]
+    written = []
+    for g in range(ng):
+        rows = torch.nonzero(gid == g, as_tuple=False).flatten()
+        sub = _Chunk([chunk.columns[i].gather(rows) for i in data_idx],
+                     [chunk.names[i] for i in data_idx])
+        vals = [key_cols[j].gather(rep[g:g + 1]).to_pylist()[0]
+                for j in range(len(key_cols))]
+        sub_dir = os.path.join(path, *[f"{k}={v}" for k, v in
+                                       zip(partition_by, vals)])
+        os.makedirs(sub_dir, exist_ok=True)
+        written.append(parquet_io.write(sub_dir, sub, mode, options))
+    return written

@@ -121,3 +121,17 @@ def test_partition_pruning_skips_files(session, tmp_path, monkeypatch):
     read_dirs = {os.path.basename(os.path.dirname(c)) for c in calls
                  if c.endswith(".parquet")}
     assert read_dirs == {"year=2023"}
+
+
+def test_partitioned_parquet_write_roundtrip(session, tmp_path):
+    import os
+
+    d = str(tmp_path / "wout")
+    df = session.create_dataframe({"year": [2023, 2023, 2024], "v": [1, 2, 3]})
+    w = df.write.format("parquet")
+    (w.partition_by if hasattr(w, "partition_by") else w.partitionBy)("year").save(d)
+    assert sorted(os.listdir(d)) == ["year=2023", "year=2024"]
+    assert sorted(session.read.format("parquet").load(d).collect()) == [
+        (1, 2023), (2, 2023), (3, 2024)]
+    assert session.sql(f"SELECT sum(v) FROM parquet.`{d}` WHERE year = 2023").collect() == [
+        (3,)]

@@ -1350,13 +1350,28 @@ def _nested_loop_join(ev: Evaluator, p: S.Join, left: Chunk, right: Chunk, names
         return left.filter_mask(counts > 0)
     if how == "anti":
         return left.filter_mask(counts == 0)
-    if how == "left":
+    rcounts = torch.zeros(nr, dtype=torch.int64, device=dev)
+    rcounts.index_add_(0, ri[mask],
+                       torch.ones(int(mask.sum()), dtype=torch.int64, device=dev))
+    if how == "rightsemi":
+        return right.filter_mask(rcounts > 0)
+    if how == "rightanti":
+        return right.filter_mask(rcounts == 0)
+    if how in ("left", "right", "full"):
         keep_li = li[mask]
         keep_ri = ri[mask]
-        unmatched = torch.nonzero(counts == 0, as_tuple=False).squeeze(1)
-        lidx = torch.cat([keep_li, unmatched])
-        ridx = torch.cat([keep_ri, torch.full((unmatched.shape[0],), -1, dtype=torch.int64, device=dev)])
-        lcols = [c.gather(lidx) for c in left.columns]
+        lidx, ridx = keep_li, keep_ri
+        if how in ("left", "full"):
+            unmatched = torch.nonzero(counts == 0, as_tuple=False).squeeze(1)
+            lidx = torch.cat([lidx, unmatched])
+            ridx = torch.cat([ridx, torch.full((unmatched.shape[0],), -1,
+                                               dtype=torch.int64, device=dev)])
+        if how in ("right", "full"):
+            runmatched = torch.nonzero(rcounts == 0, as_tuple=False).squeeze(1)
+            lidx = torch.cat([lidx, torch.full((runmatched.shape[0],), -1,
+                                               dtype=torch.int64, device=dev)])
+            ridx = torch.cat([ridx, runmatched])
+        lcols = [_gather_nullable(c, lidx) for c in left.columns]
         rcols = [_gather_nullable(c, ridx) for c in right.columns]
         return Chunk(lcols + rcols, names)
     raise ExecError(f"nested loop join type {how} TODO")

@@ -383,3 +383,16 @@ def test_named_windows_and_by_all(s):
     assert s2.sql("SELECT * FROM nb SORT BY v DESC").collect()[0] == ("b", 3)
     assert len(s2.sql("SELECT * FROM nb CLUSTER BY g").collect()) == 3
     assert len(s2.sql("SELECT * FROM nb DISTRIBUTE BY g").collect()) == 3
+
+
+def test_non_equi_outer_joins(s):
+    s2 = sail_amd.SessionContext(device="cpu")
+    s2.create_dataframe({"a": [1, 5], "x": ["p", "q"]}, name="nl")
+    s2.create_dataframe({"b": [2, 9], "y": ["r", "t"]}, name="nr")
+    assert s2.sql("SELECT * FROM nl FULL JOIN nr ON a > b "
+                  "ORDER BY a NULLS LAST, b NULLS LAST").collect() == [
+        (1, "p", None, None), (5, "q", 2, "r"), (None, None, 9, "t")]
+    assert s2.sql("SELECT * FROM nl RIGHT JOIN nr ON a > b ORDER BY b").collect() == [
+        (5, "q", 2, "r"), (None, None, 9, "t")]
+    assert s2.sql("SELECT a FROM nl LEFT SEMI JOIN nr ON a > b").collect() == [(5,)]
+    assert s2.sql("SELECT a FROM nl LEFT ANTI JOIN nr ON a > b").collect() == [(1,)]

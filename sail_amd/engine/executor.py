@@ -890,6 +890,30 @@ class Executor:
 
     def _x_CreateTable(self, p: S.CreateTable) -> Chunk:
         sess = self.ctx.session
+        head, _, rest = p.name.partition(".")
+        path = None
+        if head.lower() in ("parquet", "csv", "json", "delta") and rest:
+            path, fmt = rest, head.lower()
+        elif p.location and (p.format or "").lower() in ("parquet", "csv",
+                                                         "json", "delta"):
+            path, fmt = p.location, p.format.lower()
+        if path is not None:
+            # CREATE TABLE <fmt>.`/path` [USING fmt] AS SELECT ... /
+            # CREATE TABLE t USING fmt LOCATION '/path' AS SELECT ...
+            if p.input is None:
+                raise ExecError("path-based CREATE TABLE requires AS SELECT")
+            from ..datasource.registry import write_source
+
+            data = self.execute(p.input)
+            write_source(fmt, path, data,
+                         "overwrite" if p.replace else "error", p.options, None)
+            if path != p.name and not p.name.count("."):
+                # register the named table as a view over the path
+                view = S.DataSourceRead(format=fmt, paths=[path])
+                view.schema = [(n, c.dtype) for n, c in
+                               zip(data.names, data.columns)]
+                sess.catalog.create_view(p.name, view, replace=True)
+            return Chunk([], [])
         if p.input is not None:
             data = self.execute(p.input)
             sess.catalog.register_table_chunk(p.name, data, [(n, t) for n, t in p.input.schema])

@@ -179,3 +179,18 @@ def test_delta_history_vacuum_timestamp(s, tmp_path):
     now = datetime.datetime.now().isoformat()
     df = s.read.format("delta").option("timestampAsOf", now).load(p)
     assert df.collect() == [(1, 1.0)]
+
+
+def test_create_table_path_and_location(s, tmp_path):
+    s.create_dataframe({"a": [1, 2]}, name="ct_src")
+    p = str(tmp_path / "ctas")
+    s.sql(f"CREATE TABLE delta.`{p}` AS SELECT a FROM ct_src")
+    assert s.sql(f"SELECT * FROM delta.`{p}` ORDER BY a").collect() == [(1,), (2,)]
+    with pytest.raises(Exception):
+        s.sql(f"CREATE TABLE delta.`{p}` AS SELECT a FROM ct_src")  # exists
+    s.sql(f"CREATE OR REPLACE TABLE delta.`{p}` AS SELECT a + 5 AS a FROM ct_src")
+    assert s.sql(f"SELECT * FROM delta.`{p}` ORDER BY a").collect() == [(6,), (7,)]
+    p2 = str(tmp_path / "loc")
+    s.sql(f"CREATE TABLE ct_named USING delta LOCATION '{p2}' "
+          "AS SELECT a * 10 AS b FROM ct_src")
+    assert s.sql("SELECT * FROM ct_named ORDER BY b").collect() == [(10,), (20,)]

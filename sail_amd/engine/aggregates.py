@@ -320,7 +320,23 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
                 return StringColumn(c.offsets, c.bytes_,
                                     None if bool((cnt > 0).all()) else (cnt > 0).to(torch.uint8),
                                     data.to(torch.int32))
-            raise NotImplementedError("min/max over raw strings TODO")
+            # raw strings: order-preserving dense ranks, then pick the row
+            # holding the group's best rank and gather the actual string
+            from .executor import _sortable
+
+            ranks = _sortable(c)
+            big = int(ranks.max().item()) + 1 if ranks.numel() else 1
+            rv = torch.where(mask, ranks, torch.full_like(ranks, big if name == "min" else -1))
+            best = torch.full((ng,), big if name == "min" else -1,
+                              dtype=torch.int64, device=dev)
+            best.scatter_reduce_(0, gid, rv, reduce=red, include_self=True)
+            is_best = (rv == best[gid]) & mask
+            rows = torch.full((ng,), 0, dtype=torch.int64, device=dev)
+            rows.scatter_(0, gid[is_best], torch.arange(gid.shape[0], device=dev)[is_best])
+            got = c.gather(rows)
+            valid = cnt > 0
+            got.validity = None if bool(valid.all()) else valid.to(torch.uint8)
+            return got
         vals = c.data[mask]
         if vals.dtype.is_floating_point:
             init = float("inf") if name == "min" else float("-inf")

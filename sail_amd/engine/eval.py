@@ -356,7 +356,22 @@ class Evaluator:
             targets = []
             for v in vals:
                 if not isinstance(v, Scalar):
-                    raise EvalError("IN with non-literal values TODO")
+                    # general IN: OR-chain of equality comparisons with SQL
+                    # 3VL (true if any match; null if no match but a null
+                    # comparison exists)
+                    any_true = None
+                    any_null = None
+                    for ve in e.values:
+                        cmpe = S.BinaryOp("=", e.child, ve, T.BOOL)
+                        cv = self.eval_col(cmpe, chunk)
+                        m = cv.data.to(torch.bool) & cv.valid_mask()
+                        nl = ~cv.valid_mask()
+                        any_true = m if any_true is None else (any_true | m)
+                        any_null = nl if any_null is None else (any_null | nl)
+                    valid = any_true | ~any_null
+                    data = ~any_true if e.negated else any_true
+                    return Column(T.BOOL, data,
+                                  None if bool(valid.all()) else valid.to(torch.uint8))
                 tv = v.value
                 if isinstance(cc.dtype, T.DecimalType):
                     tv = _to_scaled(tv, cc.dtype.scale)

@@ -879,7 +879,10 @@ def _f_unix_timestamp(args, out, chunk, ev):
         return Column(T.I64, torch.div(c.data, 1_000_000, rounding_mode="floor"), c.validity)
     if isinstance(c.dtype, T.DateType):
         return Column(T.I64, c.data.to(torch.int64) * 86400, c.validity)
-    raise NotImplementedError("unix_timestamp on strings with format TODO")
+    # strings: parse (optional Spark pattern) then convert to seconds
+    ts = _f_to_timestamp(args, T.TIMESTAMP, chunk, ev)
+    return Column(T.I64, torch.div(ts.data, 1_000_000, rounding_mode="floor"),
+                  ts.validity)
 
 
 def _f_from_unixtime_ts(args, out, chunk, ev):

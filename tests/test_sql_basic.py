@@ -396,3 +396,23 @@ def test_non_equi_outer_joins(s):
         (5, "q", 2, "r"), (None, None, 9, "t")]
     assert s2.sql("SELECT a FROM nl LEFT SEMI JOIN nr ON a > b").collect() == [(5,)]
     assert s2.sql("SELECT a FROM nl LEFT ANTI JOIN nr ON a > b").collect() == [(1,)]
+
+
+def test_todo_closures(s):
+    s2 = sail_amd.SessionContext(device="cpu")
+    # raw-string (non-dictionary) min/max
+    names = [f"name_{i:04d}" for i in range(80)]
+    import random
+
+    random.seed(1)
+    random.shuffle(names)
+    s2.create_dataframe({"g": [i % 2 for i in range(80)], "c": names}, name="rm")
+    rows = s2.sql("SELECT g, min(c), max(c) FROM rm GROUP BY g ORDER BY g").collect()
+    assert rows == [(0, "name_0000", "name_0079"), (1, "name_0001", "name_0077")]
+    # unix_timestamp on strings
+    assert s2.sql("SELECT unix_timestamp('2024-01-01 00:00:00')").collect() == [
+        (1704067200,)]
+    # IN with non-literal (column-dependent) values, incl. 3VL nulls
+    s2.create_dataframe({"a": [1, 2, 3], "x": [1, 5, None]}, name="nv")
+    rows = s2.sql("SELECT a, a IN (x - 1, x) FROM nv ORDER BY a").collect()
+    assert rows == [(1, True), (2, True), (3, None)]

@@ -415,4 +415,4 @@ def test_todo_closures(s):
     # IN with non-literal (column-dependent) values, incl. 3VL nulls
     s2.create_dataframe({"a": [1, 2, 3], "x": [1, 5, None]}, name="nv")
     rows = s2.sql("SELECT a, a IN (x - 1, x) FROM nv ORDER BY a").collect()
-    assert rows == [(1, True), (2, True), (3, None)]
+    assert rows == [(1, True), (2, False), (3, None)]

@@ -1289,3 +1289,49 @@ _IMPLS["to_timestamp"] = _f_to_timestamp
 _IMPLS["try_to_timestamp"] = _f_to_timestamp
 _IMPLS["datepart"] = _f_datepart
 _IMPLS["date_part"] = _f_datepart
+
+
+
+def _parse_duration_us(text: str) -> int:
+    import re as _re
+
+    m = _re.match(r"\s*(\d+)\s*(\w+)\s*$", str(text))
+    if not m:
+        raise ValueError(f"bad duration {text!r}")
+    n, unit = int(m.group(1)), m.group(2).lower().rstrip("s")
+    mult = {"microsecond": 1, "millisecond": 1000, "second": 1_000_000,
+            "minute": 60_000_000, "hour": 3_600_000_000,
+            "day": 86_400_000_000, "week": 7 * 86_400_000_000}
+    if unit not in mult:
+        raise ValueError(f"bad duration unit {unit!r}")
+    return n * mult[unit]
+
+
+def _f_window(args, out, chunk, ev):
+    """Tumbling (or sliding start-aligned) event-time window
+    (ref: Spark window() grouping function). Returns struct(start, end)
+    in TIMESTAMP micros; GROUP BY window(ts, '1 hour') works because
+    struct grouping falls back to its fields."""
+    from .column import StructColumn
+
+    c = _col(args[0], chunk)
+    width = _parse_duration_us(_scalarize(args[1]).value)
+    us = c.data.to(torch.int64)
+    if isinstance(c.dtype, T.DateType):
+        us = us * 86_400_000_000
+    start = torch.div(us, width, rounding_mode="floor") * width
+    end = start + width
+    return StructColumn([("start", Column(T.TIMESTAMP, start, c.validity)),
+                         ("end", Column(T.TIMESTAMP, end, c.validity))],
+                        c.validity, dtype=out)
+
+
+def _f_window_time(args, out, chunk, ev):
+    """window_time(w) = w.end - 1 microsecond (Spark semantics)."""
+    c = _col(args[0], chunk)
+    end = dict(c.children_)["end"]
+    return Column(T.TIMESTAMP, end.data.to(torch.int64) - 1, c.validity)
+
+
+_IMPLS["window"] = _f_window
+_IMPLS["window_time"] = _f_window_time

@@ -218,6 +218,9 @@ _reg("from_json", lambda a: T.NULL)  # typed structurally in the resolver
 _reg("date_format from_unixtime", _string)
 _reg("to_timestamp try_to_timestamp", lambda a: T.TIMESTAMP)
 _reg("datepart date_part", _i32)
+_reg("window", lambda a: T.StructType((T.StructField("start", T.TIMESTAMP),
+                                       T.StructField("end", T.TIMESTAMP))))
+_reg("window_time", lambda a: T.TIMESTAMP)
 _reg("sort_array array_sort array_distinct array_remove array_compact flatten "
      "slice array_repeat shuffle", _same)
 _reg("array_join", _string)

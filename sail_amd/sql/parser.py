@@ -1236,7 +1236,10 @@ class Parser:
 
         # identifier: function call or column
         name = self.ident()
-        if self.at_op("(") and name.upper() not in _RESERVED_STOP:
+        if self.at_op("(") and (name.upper() not in _RESERVED_STOP
+                                or name.upper() == "WINDOW"):
+            # WINDOW is reserved for the WINDOW clause but is also the
+            # tumbling-window grouping function window(ts, '1 hour')
             return self._parse_call(name)
         # qualified column a.b or a.b.c
         qualifier = None

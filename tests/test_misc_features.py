@@ -167,3 +167,29 @@ def test_alter_table(s):
     assert [r[0] for r in s.sql("DESCRIBE at").collect()] == ["x", "yy"]
     s.sql("ALTER TABLE at RENAME TO at2")
     assert s.sql("SELECT count(*) FROM at2").collect() == [(2,)]
+
+
+def test_window_tumbling_group_by(s):
+    s.create_dataframe(
+        {"ts": ["2024-01-01 00:05:00", "2024-01-01 00:55:00", "2024-01-01 01:10:00"],
+         "v": [1, 2, 30]}, name="win_ev")
+    rows = s.sql(
+        "SELECT window(to_timestamp(ts), '1 hour').start AS ws, sum(v) "
+        "FROM win_ev GROUP BY window(to_timestamp(ts), '1 hour') ORDER BY 1").collect()
+    assert rows == [(1704067200000000, 3), (1704070800000000, 30)]
+
+
+def test_window_by_ordinal_and_end(s):
+    s.create_dataframe({"ts": ["2024-01-01 00:05:00", "2024-01-01 00:55:00"]},
+                         name="win_ev2")
+    rows = s.sql(
+        "SELECT window(to_timestamp(ts), '30 minutes').end AS we, count(*) "
+        "FROM win_ev2 GROUP BY 1 ORDER BY 1").collect()
+    assert rows == [(1704069000000000, 1), (1704070800000000, 1)]
+
+
+def test_window_time(s):
+    s.create_dataframe({"ts": ["2024-01-01 00:05:00"]}, name="win_ev3")
+    rows = s.sql(
+        "SELECT window_time(window(to_timestamp(ts), '1 hour')) FROM win_ev3").collect()
+    assert rows == [(1704070799999999,)]

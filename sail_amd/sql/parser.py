@@ -501,8 +501,21 @@ class Parser:
             if not self.eat_op(","):
                 break
         ncols = len(rows[0])
-        data = {f"col{j+1}": [self._literal_value(rows[r][j]) for r in range(len(rows))] for j in range(ncols)}
-        return S.LocalRelation(data=data)
+        try:
+            data = {f"col{j+1}": [self._literal_value(rows[r][j]) for r in range(len(rows))]
+                    for j in range(ncols)}
+            return S.LocalRelation(data=data)
+        except SqlError:
+            # non-literal row values: UNION ALL of one-row SELECTs, which
+            # reuses the full expression resolution/eval machinery
+            plan = None
+            for row in rows:
+                sel = S.Project(input=S.LocalRelation(data={"__one__": [1]}),
+                                exprs=[S.Alias(e, f"col{j+1}", None)
+                                       for j, e in enumerate(row)])
+                plan = sel if plan is None else S.SetOp(op="union", left=plan,
+                                                        right=sel, is_all=True)
+            return plan
 
     def _literal_value(self, e: S.Expr):
         if isinstance(e, S.Literal):
@@ -868,6 +881,12 @@ class Parser:
             sub = self.parse_query()
             self.expect_op(")")
             plan = self._maybe_sample(sub)
+            alias, cols = self._parse_alias()
+            if alias:
+                return S.SubqueryAlias(input=plan, alias=alias, column_aliases=cols)
+            return plan
+        if self.at_kw("VALUES"):
+            plan = self._parse_values()
             alias, cols = self._parse_alias()
             if alias:
                 return S.SubqueryAlias(input=plan, alias=alias, column_aliases=cols)

@@ -416,3 +416,15 @@ def test_todo_closures(s):
     s2.create_dataframe({"a": [1, 2, 3], "x": [1, 5, None]}, name="nv")
     rows = s2.sql("SELECT a, a IN (x - 1, x) FROM nv ORDER BY a").collect()
     assert rows == [(1, True), (2, False), (3, None)]
+
+
+def test_values_table_factor_and_expressions(s):
+    # VALUES as a FROM-clause table factor with column aliases
+    assert s.sql("SELECT * FROM VALUES (1+1, upper('a')), (10, 'b') AS t(x, y)"
+                 ).collect() == [(2, "A"), (10, "b")]
+    # expression rows route through one-row SELECT union
+    assert s.sql("SELECT x*2 FROM VALUES (abs(-5)) t(x)").collect() == [(10,)]
+    # literal rows keep the LocalRelation fast path
+    assert s.sql("VALUES (1, 2), (3, 4)").collect() == [(1, 2), (3, 4)]
+    assert s.sql("SELECT y, sum(x) FROM VALUES (1,'a'),(2,'a'),(3,'b') AS t(x,y) "
+                 "GROUP BY y ORDER BY y").collect() == [("a", 3), ("b", 3)]

@@ -1199,6 +1199,16 @@ def concat_columns(cols: List[Column]) -> Column:
         return ListColumn(torch.cat(offs), children, validity, c0.dtype)
     if isinstance(c0, StringColumn):
         return _concat_strings(cols)
+    from .column import StructColumn
+
+    if isinstance(c0, StructColumn):
+        names = [nm for nm, _ in c0.children_]
+        kids = [(nm, concat_columns([dict(c.children_)[nm] for c in cols]))
+                for nm in names]
+        validity = None
+        if any(c.validity is not None for c in cols):
+            validity = torch.cat([c.valid_mask() for c in cols]).to(torch.uint8)
+        return StructColumn(kids, validity, dtype=c0.dtype)
     data = torch.cat([c.data for c in cols])
     if any(c.validity is not None for c in cols):
         validity = torch.cat([c.valid_mask() for c in cols]).to(torch.uint8)

@@ -81,17 +81,43 @@ class _AggState:
 
     # -- persistence --------------------------------------------------------
     def state_chunk(self) -> Optional[Chunk]:
+        """State as a flat chunk; struct keys (window()) are flattened to
+        one column per field so the parquet state files stay flat."""
+        from ..engine.column import StructColumn
+
         if self.partials is None:
             return None
-        keys = self.keys or []
-        return Chunk(list(keys) + list(self.partials),
-                     [f"k{i}" for i in range(len(keys))]
-                     + [f"p{i}" for i in range(len(self.partials))])
+        cols, names = [], []
+        for i, k in enumerate(self.keys or []):
+            if isinstance(k, StructColumn):
+                for fname, fcol in k.children_:
+                    cols.append(Column(fcol.dtype, fcol.data, k.validity))
+                    names.append(f"k{i}.{fname}")
+            else:
+                cols.append(k)
+                names.append(f"k{i}")
+        cols += list(self.partials)
+        names += [f"p{i}" for i in range(len(self.partials))]
+        return Chunk(cols, names)
 
     def load_chunk(self, chunk: Chunk):
-        nk = len(self.agg.group_by)
-        self.keys = list(chunk.columns[:nk])
-        self.partials = list(chunk.columns[nk:])
+        from ..engine.column import StructColumn
+
+        keys, ci = [], 0
+        for g in self.agg.group_by:
+            dt = getattr(g, "dtype", None)
+            if isinstance(dt, T.StructType):
+                fields = []
+                for f in dt.fields:
+                    fields.append((f.name, chunk.columns[ci]))
+                    ci += 1
+                validity = fields[0][1].validity
+                keys.append(StructColumn(fields, validity, dtype=dt))
+            else:
+                keys.append(chunk.columns[ci])
+                ci += 1
+        self.keys = keys
+        self.partials = list(chunk.columns[ci:])
 
     # -- update -------------------------------------------------------------
     def update(self, ex, child: Chunk):
@@ -172,7 +198,7 @@ class StreamingQuery:
                  trigger_interval: float = 0.1, trigger_once: bool = False,
                  available_now: bool = False,
                  checkpoint_location: Optional[str] = None,
-                 name: Optional[str] = None):
+                 name: Optional[str] = None, watermark=None):
         self.id = str(uuid.uuid4())
         self.name = name
         self.session = session
@@ -185,6 +211,11 @@ class StreamingQuery:
         self.trigger_once = trigger_once
         self.available_now = available_now
         self.checkpoint = checkpoint_location
+        self.watermark_spec = watermark          # (event_time_col, delay_str)
+        self._wm_delay_us = 0
+        self._max_event_us: Optional[int] = None  # max event time ever seen
+        self._wm_col_idx: Optional[int] = None    # index into agg.input schema
+        self._time_key_idx: Optional[int] = None  # index into agg.group_by
         self.exception: Optional[BaseException] = None
         self.last_progress: Optional[dict] = None
         self.batch_id = -1
@@ -231,10 +262,47 @@ class StreamingQuery:
             self._mode = "retained"
         if self._mode == "stateless" and self.output_mode == "complete":
             self._mode = "retained"  # complete over a stateless plan: re-run all
+        if self._mode == "incremental" and self.watermark_spec is not None:
+            self._prepare_watermark()
         if self._mode != "stateless" and self.output_mode == "append":
+            if self._time_key_idx is None:
+                raise ValueError(
+                    "append output mode with streaming aggregation requires "
+                    "withWatermark() on the event-time column and grouping by "
+                    "window(<event-time>, ...) or the event-time column itself")
+
+    def _prepare_watermark(self):
+        """Resolve the watermark column against the aggregate's input and
+        find which group key carries event time (a window() struct or a
+        timestamp-typed key)."""
+        from ..engine.functions_impl import _parse_duration_us
+
+        col, delay = self.watermark_spec
+        self._wm_delay_us = _parse_duration_us(delay)
+        agg = self._agg_state.agg
+        in_names = [n for n, _ in agg.input.schema]
+        if col not in in_names:
             raise ValueError(
-                "append output mode is not supported with streaming "
-                "aggregation (no watermark support); use complete or update")
+                f"watermark column {col!r} not found below the aggregation "
+                f"(have {in_names})")
+        self._wm_col_idx = in_names.index(col)
+        wdt = agg.input.schema[self._wm_col_idx][1]
+        if not (isinstance(wdt, (T.TimestampType, T.DateType)) or wdt.is_integer):
+            raise ValueError(
+                f"watermark column {col!r} must be timestamp/date/integer, "
+                f"got {wdt}")
+        for i, g in enumerate(agg.group_by):
+            e = g.child if isinstance(g, S.Alias) else g
+            if isinstance(e, S.Func) and e.name.lower() == "window":
+                self._time_key_idx = i
+                break
+        else:
+            for i, g in enumerate(agg.group_by):
+                e = g.child if isinstance(g, S.Alias) else g
+                if isinstance(getattr(e, "dtype", None),
+                              (T.TimestampType, T.DateType)):
+                    self._time_key_idx = i
+                    break
 
     # -- checkpoint ---------------------------------------------------------
     def _ckpt_dir(self, sub: str) -> str:
@@ -252,6 +320,8 @@ class StreamingQuery:
             with open(os.path.join(offs, str(last))) as f:
                 self._offset = json.load(f)["offset"]
             self.batch_id = last
+            with open(os.path.join(commits, str(last))) as f:
+                self._max_event_us = json.load(f).get("maxEventTimeUs")
             # restore aggregation state
             state_dir = os.path.join(self.checkpoint, "state")
             if self._agg_state is not None and os.path.isdir(state_dir):
@@ -377,7 +447,8 @@ class StreamingQuery:
         if self.checkpoint:
             self._save_state()
             with open(os.path.join(self._ckpt_dir("commits"), str(bid)), "w") as f:
-                json.dump({"batchId": bid}, f)
+                json.dump({"batchId": bid,
+                           "maxEventTimeUs": self._max_event_us}, f)
         self._offset = end
         self.batch_id = bid
         self.last_progress = {
@@ -387,6 +458,73 @@ class StreamingQuery:
             "sources": [{"endOffset": end}],
         }
         return True
+
+    # -- watermark ----------------------------------------------------------
+    @property
+    def watermark_us(self) -> Optional[int]:
+        """Current watermark in epoch micros (None before any event)."""
+        if self._max_event_us is None:
+            return None
+        return self._max_event_us - self._wm_delay_us
+
+    @staticmethod
+    def _to_us(col: Column) -> torch.Tensor:
+        us = col.data.to(torch.int64)
+        if isinstance(col.dtype, T.DateType):
+            us = us * 86_400_000_000
+        return us
+
+    def _filter_late(self, child: Chunk) -> Chunk:
+        """Drop rows older than the watermark (and null event times)."""
+        wm = self.watermark_us
+        c = child.columns[self._wm_col_idx]
+        keep = self._to_us(c) >= wm if wm is not None else torch.ones(
+            len(c), dtype=torch.bool, device=c.device)
+        if c.validity is not None:
+            keep &= c.validity
+        if bool(keep.all()):
+            return child
+        idx = torch.nonzero(keep, as_tuple=False).flatten()
+        return Chunk([col.gather(idx) for col in child.columns],
+                     list(child.names))
+
+    def _advance_watermark(self, child: Chunk):
+        c = child.columns[self._wm_col_idx]
+        us = self._to_us(c)
+        if c.validity is not None:
+            us = us[c.validity]
+        if us.numel():
+            m = int(us.max())
+            self._max_event_us = m if self._max_event_us is None \
+                else max(self._max_event_us, m)
+
+    def _group_close_us(self) -> Optional[torch.Tensor]:
+        """Per-state-group close time: window.end for window() keys, the key
+        value itself for plain timestamp keys."""
+        st = self._agg_state
+        if st.keys is None or self._time_key_idx is None:
+            return None
+        key = st.keys[self._time_key_idx]
+        from ..engine.column import StructColumn
+
+        if isinstance(key, StructColumn):
+            return self._to_us(dict(key.children_)["end"])
+        return self._to_us(key)
+
+    def _expired_mask(self) -> Optional[torch.Tensor]:
+        wm = self.watermark_us
+        close = self._group_close_us()
+        if wm is None or close is None:
+            return None
+        return close <= wm
+
+    def _evict(self, expired: Optional[torch.Tensor]):
+        if expired is None or not bool(expired.any()):
+            return
+        st = self._agg_state
+        keep = torch.nonzero(~expired, as_tuple=False).flatten()
+        st.keys = [c.gather(keep) for c in st.keys]
+        st.partials = [c.gather(keep) for c in st.partials]
 
     def _execute_batch(self, batch: Chunk) -> Optional[Chunk]:
         from ..engine.executor import ExecutionContext, Executor, concat_columns
@@ -413,10 +551,27 @@ class StreamingQuery:
         ctx = ExecutionContext(self.session, self.session.device)
         ex = Executor(ctx)
         child = ex.execute(self._agg_state.agg.input)
+        if self._wm_col_idx is not None:
+            child = self._filter_late(child)
         finalized, touched = self._agg_state.update(ex, child)
-        if self.output_mode == "update":
+        expired = None
+        if self._wm_col_idx is not None:
+            self._advance_watermark(child)
+            expired = self._expired_mask()
+        if self.output_mode == "append":
+            # emit only windows the watermark has closed, then drop their state
+            idx = torch.nonzero(expired, as_tuple=False).flatten() \
+                if expired is not None else torch.zeros(0, dtype=torch.int64)
+            out = finalized.gather(idx) if idx.numel() else None
+            self._evict(expired)
+            if out is None:
+                return None
+            finalized = out
+        elif self.output_mode == "update":
             idx = torch.nonzero(touched, as_tuple=False).flatten()
             finalized = finalized.gather(idx)
+            self._evict(expired)
+        # complete mode keeps all state (Spark: watermark does not evict there)
         if self._upper_parent is None:
             return finalized
         sub = S.ChunkSource(chunk=finalized, schema=self._agg_state.agg.schema)

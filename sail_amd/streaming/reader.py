@@ -54,14 +54,26 @@ class StreamingDataFrame:
     is_streaming = True
 
     def __init__(self, session, source: StreamSource, view_name: str,
-                 query_sql: Optional[str] = None):
+                 query_sql: Optional[str] = None, watermark=None):
         self.session = session
         self.source = source
         self.view_name = view_name
         self.query_sql = query_sql or f"SELECT * FROM {view_name}"
+        self.watermark = watermark  # (event_time_col, delay_str)
 
     def sql(self, query: str) -> "StreamingDataFrame":
-        return StreamingDataFrame(self.session, self.source, self.view_name, query)
+        return StreamingDataFrame(self.session, self.source, self.view_name,
+                                  query, self.watermark)
+
+    def with_watermark(self, col: str, delay: str) -> "StreamingDataFrame":
+        """Declare `col` as the event-time column with max lateness `delay`
+        (ref: Spark Dataset.withWatermark). Enables append output mode for
+        windowed aggregations: a window is emitted once, when the watermark
+        (max event time seen - delay) passes its end."""
+        return StreamingDataFrame(self.session, self.source, self.view_name,
+                                  self.query_sql, (col, delay))
+
+    withWatermark = with_watermark
 
     @property
     def write_stream(self) -> "DataStreamWriter":
@@ -130,7 +142,7 @@ class DataStreamWriter:
             trigger_once=self._trigger_once,
             available_now=self._available_now,
             checkpoint_location=self._options.get("checkpointLocation"),
-            name=self._query_name)
+            name=self._query_name, watermark=sdf.watermark)
         sdf.session.streams.register(q)
         return q.start()
 

@@ -208,3 +208,83 @@ def test_append_with_aggregation_rejected(s):
     with pytest.raises(ValueError):
         sdf.sql("SELECT k, sum(v) FROM ev2 GROUP BY k") \
            .write_stream.output_mode("append").format("noop").start()
+
+
+HOUR_US = 3_600_000_000
+MIN_US = 60_000_000
+
+
+def _wm_query(s, name, out, mode="append", ckpt=None):
+    r = s.read_stream.format("memory").schema([("ts", T.TIMESTAMP), ("v", T.I64)])
+    sdf = r.load(name=name)
+    src = sdf.source
+    sdf = sdf.sql(f"SELECT window(ts, '1 hour').start AS ws, sum(v) AS sv "
+                  f"FROM {name} GROUP BY window(ts, '1 hour')"
+                  ).with_watermark("ts", "30 minutes")
+    w = sdf.write_stream.format("memory").query_name(out).output_mode(mode) \
+        .trigger(processing_time=0.01)
+    if ckpt:
+        w = w.option("checkpointLocation", ckpt)
+    return src, w.start()
+
+
+def test_watermark_append_mode(s):
+    src, q = _wm_query(s, "wm_ev1", "wm_out1")
+    try:
+        src.add_rows({"ts": [10 * MIN_US, 50 * MIN_US], "v": [1, 2]})
+        q.process_all_available()
+        assert q.watermark_us == 20 * MIN_US  # max(50m) - 30m
+        # nothing closed yet -> no output table
+        src.add_rows({"ts": [2 * HOUR_US + 40 * MIN_US], "v": [5]})
+        q.process_all_available()
+        # watermark 2h10m closes the hour-0 window
+        assert s.sql("SELECT * FROM wm_out1 ORDER BY 1").collect() == [(0, 3)]
+        # late row (5m < watermark) must be DROPPED; 3h50m closes hour-2
+        src.add_rows({"ts": [5 * MIN_US, 3 * HOUR_US + 50 * MIN_US], "v": [100, 7]})
+        q.process_all_available()
+        assert s.sql("SELECT * FROM wm_out1 ORDER BY 1").collect() == \
+            [(0, 3), (2 * HOUR_US, 5)]
+    finally:
+        q.stop()
+
+
+def test_watermark_update_mode_evicts_state(s):
+    src, q = _wm_query(s, "wm_ev2", "wm_out2", mode="update")
+    try:
+        src.add_rows({"ts": [10 * MIN_US], "v": [1]})
+        q.process_all_available()
+        src.add_rows({"ts": [3 * HOUR_US], "v": [2]})
+        q.process_all_available()
+        # hour-0 window evicted once watermark passed; state holds 1 group
+        assert len(q._agg_state.keys[0]) == 1
+    finally:
+        q.stop()
+
+
+def test_watermark_append_requires_time_key(s):
+    r = s.read_stream.format("memory").schema([("k", T.I64), ("v", T.I64)])
+    sdf = r.load(name="wm_ev3").sql("SELECT k, sum(v) FROM wm_ev3 GROUP BY k")
+    with pytest.raises(ValueError, match="append output mode"):
+        sdf.write_stream.format("memory").query_name("wm_out3") \
+            .output_mode("append").start()
+
+
+def test_watermark_checkpoint_recovery(s, tmp_path):
+    ck = str(tmp_path / "wm_ck")
+    src, q = _wm_query(s, "wm_ev4", "wm_out4", ckpt=ck)
+    try:
+        src.add_rows({"ts": [10 * MIN_US, HOUR_US + 10 * MIN_US], "v": [1, 2]})
+        q.process_all_available()
+    finally:
+        q.stop()
+    # restart from the checkpoint: watermark and window state survive
+    src2, q2 = _wm_query(s, "wm_ev5", "wm_out4", ckpt=ck)
+    try:
+        assert q2._max_event_us == HOUR_US + 10 * MIN_US
+        assert len(q2._agg_state.keys[0]) == 2  # both open windows restored
+        src2.add_rows({"ts": [3 * HOUR_US], "v": [9]})
+        q2.process_all_available()
+        rows = s.sql("SELECT * FROM wm_out4 ORDER BY 1").collect()
+        assert (0, 1) in rows and (HOUR_US, 2) in rows
+    finally:
+        q2.stop()

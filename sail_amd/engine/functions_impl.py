@@ -1315,6 +1315,13 @@ def _f_window(args, out, chunk, ev):
     from .column import StructColumn
 
     c = _col(args[0], chunk)
+    if len(args) > 2:
+        slide = _parse_duration_us(_scalarize(args[2]).value)
+        width = _parse_duration_us(_scalarize(args[1]).value)
+        if slide != width:
+            raise NotImplementedError(
+                "sliding windows (slide != width) are not supported yet; "
+                "use tumbling windows")
     width = _parse_duration_us(_scalarize(args[1]).value)
     us = c.data.to(torch.int64)
     if isinstance(c.dtype, T.DateType):

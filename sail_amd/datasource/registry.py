@@ -43,6 +43,10 @@ def infer_source_schema(fmt: str, paths: List[str], options: Dict[str, str]):
         from . import delta
 
         return delta.infer_schema(paths, options)
+    if fmt == "iceberg":
+        from . import iceberg
+
+        return iceberg.infer_schema(paths, options)
     if fmt == "text":
         from . import text_io
 
@@ -76,6 +80,10 @@ def read_source(fmt: str, paths: List[str], options: Dict[str, str], schema, dev
         from . import delta
 
         return delta.read(paths, schema, device, options)
+    if fmt == "iceberg":
+        from . import iceberg
+
+        return iceberg.read(paths, schema, device, options)
     if fmt == "text":
         from . import text_io
 
@@ -107,6 +115,10 @@ def write_source(fmt: str, path: str, chunk, mode: str, options, partition_by):
         from . import delta
 
         return delta.write(path, chunk, mode, options)
+    if fmt == "iceberg":
+        from . import iceberg
+
+        return iceberg.write(path, chunk, mode, options)
     if fmt == "text":
         from . import text_io
 

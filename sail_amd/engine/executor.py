@@ -892,10 +892,11 @@ class Executor:
         sess = self.ctx.session
         head, _, rest = p.name.partition(".")
         path = None
-        if head.lower() in ("parquet", "csv", "json", "delta") and rest:
+        if head.lower() in ("parquet", "csv", "json", "delta", "iceberg") and rest:
             path, fmt = rest, head.lower()
         elif p.location and (p.format or "").lower() in ("parquet", "csv",
-                                                         "json", "delta"):
+                                                         "json", "delta",
+                                                         "iceberg"):
             path, fmt = p.location, p.format.lower()
         if path is not None:
             # CREATE TABLE <fmt>.`/path` [USING fmt] AS SELECT ... /

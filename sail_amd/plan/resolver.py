@@ -94,7 +94,7 @@ class Resolver:
                 return self._qualify(aliased, p.table)
         # `SELECT ... FROM parquet.`/path``-style direct file reads (Spark)
         head, _, rest = p.table.partition(".")
-        if head.lower() in ("parquet", "csv", "json", "delta") and rest:
+        if head.lower() in ("parquet", "csv", "json", "delta", "iceberg") and rest:
             ds = S.DataSourceRead(format=head.lower(), paths=[rest], options=p.options)
             return self._qualify(self._p_DataSourceRead(ds, outer), head.lower())
         schema = self.catalog.table_schema(p.table)
@@ -770,7 +770,7 @@ class Resolver:
     def _p_InsertInto(self, p: S.InsertInto, outer):
         inp = self._plan(p.input, outer)
         head, _, rest = p.table.partition(".")
-        if head.lower() in ("parquet", "csv", "json", "delta") and rest:
+        if head.lower() in ("parquet", "csv", "json", "delta", "iceberg") and rest:
             # INSERT INTO delta.`/path` -> datasource append/overwrite;
             # positional semantics: rename/cast the input to the target
             # schema (VALUES rows arrive as col1/col2/...)
@@ -811,6 +811,10 @@ class Resolver:
             from ..datasource.delta import infer_schema
 
             return infer_schema([rest]), ("delta", rest)
+        if head.lower() == "iceberg" and rest:
+            from ..datasource.iceberg import infer_schema
+
+            return infer_schema([rest]), ("iceberg", rest)
         schema = self.catalog.table_schema(name)
         if schema is None:
             raise ResolutionError(f"table not found: {name}")

@@ -32,9 +32,9 @@ def _load_target(executor, ref, schema) -> Chunk:
         if t is None:
             raise MergeError(f"no data for table {name}")
         return Chunk.from_table(t)
-    from ..datasource.delta import read
+    from ..datasource.registry import read_source
 
-    t = read([name], schema, executor.ctx.device, {})
+    t = read_source(kind, [name], {}, schema, executor.ctx.device)
     return Chunk.from_table(t)
 
 
@@ -42,10 +42,15 @@ def _store_target(executor, ref, chunk: Chunk, schema):
     kind, name = ref
     if kind == "catalog":
         executor.ctx.session.catalog.register_table_chunk(name, chunk, schema)
-    else:
+    elif kind == "delta":
         from ..datasource.delta import replace_table
 
         replace_table(name, chunk)
+    else:
+        # copy-on-write rewrite for other table formats (iceberg)
+        from ..datasource.registry import write_source
+
+        write_source(kind, name, chunk, "overwrite", {}, None)
 
 
 def _null_column(dtype: T.DataType, n: int, device) -> Column:

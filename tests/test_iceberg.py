@@ -163,3 +163,19 @@ def test_iceberg_types_round_trip(s, tmp_path):
     sch = I.infer_schema([base])
     kinds = {n: str(t) for n, t in sch}
     assert "timestamp" in kinds["ts"].lower() or kinds["ts"]
+
+
+def test_iceberg_dml(s, tmp_path):
+    base = str(tmp_path / "dml")
+    s.create_dataframe({"id": [1, 2, 3], "v": [10, 20, 30]}, name="dml_src")
+    s.sql(f"CREATE TABLE iceberg.`{base}` AS SELECT * FROM dml_src").collect()
+    s.sql(f"UPDATE iceberg.`{base}` SET v = v + 1 WHERE id = 2").collect()
+    s.sql(f"DELETE FROM iceberg.`{base}` WHERE id = 3").collect()
+    s.create_dataframe({"id": [1, 4], "v": [100, 400]}, name="dml_upd")
+    s.sql(f"MERGE INTO iceberg.`{base}` t USING dml_upd u ON t.id = u.id "
+          "WHEN MATCHED THEN UPDATE SET v = u.v "
+          "WHEN NOT MATCHED THEN INSERT (id, v) VALUES (u.id, u.v)").collect()
+    assert s.sql(f"SELECT * FROM iceberg.`{base}` ORDER BY id").collect() == \
+        [(1, 100), (2, 21), (4, 400)]
+    # each DML statement created a snapshot (CTAS + update + delete + merge)
+    assert len(I.history(base)) == 4

@@ -20,6 +20,8 @@ deviation). Deletion vectors / column mapping are follow-ups.
 from __future__ import annotations
 
 import json
+import struct
+import zlib
 import os
 import time
 import uuid
@@ -148,6 +150,53 @@ class DeltaLog:
                         files.pop(action["remove"]["path"], None)
         return schema, list(files.keys()), meta, version
 
+    def snapshot_adds(self, version: Optional[int] = None):
+        """Like snapshot() but returns the live ADD actions (needed for
+        deletion vectors / stats), not just the file names."""
+        vs = self.versions()
+        if not vs:
+            raise FileNotFoundError(f"not a delta table: {self.path}")
+        if version is None:
+            version = vs[-1]
+        files: Dict[str, dict] = {}
+        schema = None
+        meta = {}
+        start = 0
+        ckpt = self._last_checkpoint()
+        if ckpt is not None and ckpt <= version:
+            cp = os.path.join(self.log_path, f"{ckpt:020d}.checkpoint.parquet")
+            if os.path.exists(cp):
+                import pyarrow.parquet as pq
+
+                tbl = pq.read_table(cp)
+                for kind, payload in zip(tbl.column("kind").to_pylist(),
+                                         tbl.column("json").to_pylist()):
+                    action = json.loads(payload)
+                    if kind == "metaData":
+                        meta = action
+                        schema = schema_from_string(meta["schemaString"])
+                    elif kind == "add":
+                        files[action["path"]] = action
+                start = ckpt + 1
+        for v in vs:
+            if v < start:
+                continue
+            if v > version:
+                break
+            with open(os.path.join(self.log_path, f"{v:020d}.json")) as f:
+                for line in f:
+                    if not line.strip():
+                        continue
+                    action = json.loads(line)
+                    if "metaData" in action:
+                        meta = action["metaData"]
+                        schema = schema_from_string(meta["schemaString"])
+                    elif "add" in action:
+                        files[action["add"]["path"]] = action["add"]
+                    elif "remove" in action:
+                        files.pop(action["remove"]["path"], None)
+        return schema, list(files.values()), meta, version
+
     def maybe_checkpoint(self, version: int):
         """Write a parquet checkpoint of the live state every
         CHECKPOINT_INTERVAL commits + the _last_checkpoint pointer.
@@ -266,6 +315,148 @@ def _protocol_action() -> dict:
 
 # -- read/write -------------------------------------------------------------
 
+
+# ===========================================================================
+# deletion vectors (ref: sail-delta-lake/src/deletion_vector/ — roaring
+# bitmap + z85; Delta protocol DV descriptors on add actions)
+# ===========================================================================
+DV_MAGIC = 1681511377
+
+
+def dv_positions(table_path: str, dv: dict):
+    """Decode a deletionVector descriptor to the sorted int64 row positions
+    it deletes. storageType: "i" inline z85, "u" uuid-named file relative to
+    the table, "p" absolute path."""
+    import numpy as np
+
+    from ..utils.roaring import roaring64_deserialize, z85_decode
+
+    st = dv["storageType"]
+    if st == "i":
+        data = z85_decode(dv["pathOrInlineDv"])
+    else:
+        if st == "p":
+            fpath = dv["pathOrInlineDv"]
+            if fpath.startswith("file://"):
+                fpath = fpath[len("file://"):]
+        else:  # "u"
+            enc = dv["pathOrInlineDv"]
+            prefix, uenc = enc[:-20], enc[-20:]
+            u = uuid.UUID(bytes=z85_decode(uenc))
+            parts = [table_path] + ([prefix] if prefix else [])
+            fpath = os.path.join(*parts, f"deletion_vector_{u}.bin")
+        with open(fpath, "rb") as f:
+            f.seek(dv.get("offset", 1))
+            (size,) = struct.unpack(">i", f.read(4))
+            data = f.read(size)
+    (magic,) = struct.unpack_from("<i", data)
+    if magic != DV_MAGIC:
+        raise ValueError(f"bad deletion-vector magic {magic}")
+    return roaring64_deserialize(data[4:])
+
+
+def write_dv_file(table_path: str, positions) -> dict:
+    """Write a deletion-vector .bin file; returns the descriptor to put on
+    the add action. Layout: version byte, then <int32 BE size><data>
+    <int32 BE crc32>, data = <int32 LE magic><RoaringBitmapArray>."""
+    from ..utils.roaring import roaring64_serialize, z85_encode
+
+    u = uuid.uuid4()
+    blob = struct.pack("<i", DV_MAGIC) + roaring64_serialize(positions)
+    fpath = os.path.join(table_path, f"deletion_vector_{u}.bin")
+    with open(fpath, "wb") as f:
+        f.write(b"\x01")
+        f.write(struct.pack(">i", len(blob)))
+        f.write(blob)
+        f.write(struct.pack(">i", zlib.crc32(blob) & 0x7FFFFFFF))
+    return {"storageType": "u", "pathOrInlineDv": z85_encode(u.bytes),
+            "offset": 1, "sizeInBytes": len(blob),
+            "cardinality": len(positions)}
+
+
+def scan_layout(path: str, schema, device, options):
+    """Read the table file-by-file applying deletion vectors; returns
+    (Table, [(add_action, surviving_original_positions ndarray)]) in row
+    order — the layout lets DELETE map global row numbers back to per-file
+    positions for DV rewrites."""
+    import numpy as np
+    import torch
+
+    from . import parquet_io
+    from ..engine.chunk import Chunk
+    from ..engine.column import Column, Table
+    from ..engine.executor import concat_columns
+
+    log = DeltaLog(path)
+    tbl_schema, adds, _, _ = log.snapshot_adds(_version_opt(options, log))
+    if not adds:
+        cols = {n: Column.from_values([], t, device=device)
+                for n, t in tbl_schema}
+        return Table(cols), []
+    parts, layout = [], []
+    for add in adds:
+        fpath = os.path.join(path, add["path"])
+        t = parquet_io.read([fpath], tbl_schema, device, options or {})
+        chunk = Chunk.from_table(t)
+        nrows = chunk.num_rows
+        if add.get("deletionVector"):
+            drop = dv_positions(path, add["deletionVector"])
+            keep = np.ones(nrows, dtype=bool)
+            keep[drop] = False
+            orig = np.nonzero(keep)[0]
+            idx = torch.from_numpy(orig).to(torch.int64)
+            chunk = Chunk([c.gather(idx) for c in chunk.columns],
+                          list(chunk.names))
+        else:
+            orig = np.arange(nrows, dtype=np.int64)
+        parts.append(chunk)
+        layout.append((add, orig))
+    out = parts[0] if len(parts) == 1 else Chunk(
+        [concat_columns([p.columns[i] for p in parts])
+         for i in range(len(parts[0].columns))], list(parts[0].names))
+    return Table({n: c for n, c in zip(out.names, out.columns)}), layout
+
+
+def delete_with_dv(path: str, layout, deleted_mask, max_retries: int = 10):
+    """Commit a DELETE as deletion-vector updates (no data-file rewrite):
+    per touched file, merge new positions into its DV and re-add the file
+    with the new descriptor."""
+    import numpy as np
+
+    log = DeltaLog(path)
+    actions = []
+    off = 0
+    now = int(time.time() * 1000)
+    for add, orig in layout:
+        seg = deleted_mask[off:off + len(orig)]
+        off += len(orig)
+        newly = orig[seg]
+        if not len(newly):
+            continue
+        old_dv = add.get("deletionVector")
+        merged = newly if old_dv is None else np.union1d(
+            dv_positions(path, old_dv), newly)
+        desc = write_dv_file(path, merged)
+        actions.append({"remove": {"path": add["path"],
+                                   "deletionTimestamp": now,
+                                   "dataChange": True}})
+        new_add = dict(add)
+        new_add["deletionVector"] = desc
+        new_add["dataChange"] = True
+        actions.append({"add": new_add})
+    if not actions:
+        return None
+    for _ in range(max_retries):
+        version = (log.latest_version() or 0) + 1
+        try:
+            log.commit(version, actions)
+            log.maybe_checkpoint(version)
+            return version
+        except FileExistsError:
+            continue
+    raise RuntimeError("delta DV commit: too many conflicts")
+
+
 def infer_schema(paths: List[str], options: Dict[str, str] = None):
     log = DeltaLog(paths[0])
     schema, _, _, _ = log.snapshot(_version_opt(options, log))
@@ -300,13 +491,16 @@ def read(paths: List[str], schema, device, options: Dict[str, str]):
     from ..engine.column import Table
 
     log = DeltaLog(paths[0])
-    tbl_schema, files, _, _ = log.snapshot(_version_opt(options, log))
-    if not files:
+    tbl_schema, adds, _, _ = log.snapshot_adds(_version_opt(options, log))
+    if not adds:
         from ..engine.column import Column
 
         cols = {n: Column.from_values([], t, device=device) for n, t in tbl_schema}
         return Table(cols)
-    full = [os.path.join(paths[0], f) for f in files]
+    if any(a.get("deletionVector") for a in adds):
+        t, _ = scan_layout(paths[0], tbl_schema, device, options)
+        return t
+    full = [os.path.join(paths[0], a["path"]) for a in adds]
     return parquet_io.read(full, tbl_schema, device, options or {})
 
 

@@ -230,10 +230,36 @@ def execute_update(executor, p: S.UpdateTable) -> Chunk:
                  ["num_affected_rows"])
 
 
+#: DELETE rewrites data files only beyond this deleted fraction; below it
+#: the commit is a deletion-vector update (no parquet rewrite)
+DV_DELETE_MAX_FRACTION = 0.5
+
+
 def execute_delete(executor, p: S.DeleteFrom) -> Chunk:
     ev = executor.ev
     tschema = p.__dict__["_target_schema"]
     tref = p.__dict__["_target_ref"]
+    kind, name = tref
+    if kind == "delta":
+        # merge-on-read path: map deleted rows back to per-file positions
+        # and commit deletion vectors instead of rewriting parquet
+        from ..datasource.delta import delete_with_dv, scan_layout
+
+        t, layout = scan_layout(name, tschema, executor.ctx.device, {})
+        target = Chunk.from_table(t)
+        dev = target.device
+        mask = (ev.eval_mask(p.condition, target) if p.condition is not None
+                else torch.ones(target.num_rows, dtype=torch.bool, device=dev))
+        ndel = int(mask.sum().item())
+        if ndel == 0:
+            return Chunk([Column.from_values([0], T.I64, device="cpu")],
+                         ["num_affected_rows"])
+        if target.num_rows and ndel / target.num_rows <= DV_DELETE_MAX_FRACTION:
+            delete_with_dv(name, layout, mask.cpu().numpy())
+        else:
+            _store_target(executor, tref, target.filter_mask(~mask), tschema)
+        return Chunk([Column.from_values([ndel], T.I64, device="cpu")],
+                     ["num_affected_rows"])
     target = _load_target(executor, tref, tschema)
     dev = target.device
     mask = (ev.eval_mask(p.condition, target) if p.condition is not None

@@ -288,3 +288,22 @@ def test_watermark_checkpoint_recovery(s, tmp_path):
         assert (0, 1) in rows and (HOUR_US, 2) in rows
     finally:
         q2.stop()
+
+
+def test_stream_static_join(s):
+    """Stateless streaming plans can join against static catalog tables
+    (ref: Spark stream-static joins)."""
+    s.create_dataframe({"k": [1, 2], "label": ["one", "two"]}, name="dims")
+    r = s.read_stream.format("memory").schema([("k", T.I64), ("v", T.I64)])
+    sdf = r.load(name="ss_ev")
+    src = sdf.source
+    q = sdf.sql("SELECT d.label, e.v FROM ss_ev e JOIN dims d ON e.k = d.k") \
+        .write_stream.format("memory").query_name("ss_out") \
+        .trigger(processing_time=0.01).start()
+    try:
+        src.add_rows({"k": [1, 2, 1], "v": [10, 20, 30]})
+        q.process_all_available()
+        rows = sorted(s.sql("SELECT * FROM ss_out").collect())
+        assert rows == [("one", 10), ("one", 30), ("two", 20)]
+    finally:
+        q.stop()

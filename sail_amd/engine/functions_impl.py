@@ -761,15 +761,24 @@ def _dict_transform(fn):
 
         if isinstance(c, StringColumn) and c.is_dict:
             vals = [f(v) for v in c.dict_values()]
-            uniq = sorted(set(vals))
+            uniq = sorted({v for v in vals if v is not None})
             idx = {s: i for i, s in enumerate(uniq)}
-            lut = torch.tensor([idx[v] for v in vals], dtype=torch.int32, device=c.device)
+            # fn may return None for some dictionary entries: those become
+            # code -1 + invalid rows
+            lut = torch.tensor([-1 if v is None else idx[v] for v in vals],
+                               dtype=torch.int32, device=c.device)
             from .column import _pack_strings
 
             offs, byts = _pack_strings(uniq, c.device)
             codes = lut[c.codes.to(torch.int64).clamp_min(0)]
             codes = torch.where(c.codes >= 0, codes, c.codes)
-            return StringColumn(offs, byts, c.validity, codes)
+            validity = c.validity
+            if bool((lut < 0).any()):
+                valid = codes >= 0
+                if validity is not None:
+                    valid &= c.valid_mask()
+                validity = valid.to(torch.uint8)
+            return StringColumn(offs, byts, validity, codes)
         return _str_map(c, f)
 
     return impl
@@ -1342,3 +1351,432 @@ def _f_window_time(args, out, chunk, ev):
 
 _IMPLS["window"] = _f_window
 _IMPLS["window_time"] = _f_window_time
+
+
+# -- URL / XML xpath / CSV / variant families (host eval — ref:
+#    sail-function src/scalar/{url,xml,csv,variant}) -----------------------
+
+def _parse_url_part(url, part, key=None):
+    from urllib.parse import urlparse, parse_qs
+
+    if url is None or part is None:
+        return None
+    try:
+        u = urlparse(url)
+    except ValueError:
+        return None
+    part = part.upper()
+    if part == "HOST":
+        return u.hostname
+    if part == "PATH":
+        return u.path or None
+    if part == "QUERY":
+        if key is not None:
+            vals = parse_qs(u.query, keep_blank_values=False).get(key)
+            return vals[0] if vals else None
+        return u.query or None
+    if part == "REF":
+        return u.fragment or None
+    if part == "PROTOCOL":
+        return u.scheme or None
+    if part == "FILE":
+        return (u.path + ("?" + u.query if u.query else "")) or None
+    if part == "AUTHORITY":
+        return u.netloc or None
+    if part == "USERINFO":
+        if "@" not in u.netloc:
+            return None
+        return u.netloc.rsplit("@", 1)[0]
+    return None
+
+
+def _f_parse_url(args, out, chunk, ev):
+    part = _scalarize(args[1]).value
+    key = _scalarize(args[2]).value if len(args) > 2 else None
+    return _dict_transform(
+        lambda v: _parse_url_part(v, part, key))(args[:1], out, chunk, ev)
+
+
+def _f_url_encode(args, out, chunk, ev):
+    from urllib.parse import quote_plus
+
+    return _dict_transform(
+        lambda v: None if v is None else quote_plus(v))(args[:1], out, chunk, ev)
+
+
+def _f_url_decode(args, out, chunk, ev):
+    from urllib.parse import unquote_plus
+
+    return _dict_transform(
+        lambda v: None if v is None else unquote_plus(v))(args[:1], out, chunk, ev)
+
+
+def _xpath_nodes(doc, path):
+    """Subset of XPath over ElementTree: steps a/b, .//b, [@k='v'] filters,
+    trailing /text() or /@attr."""
+    import xml.etree.ElementTree as ET
+
+    if doc is None:
+        return None
+    want_text = False
+    attr = None
+    if path.endswith("/text()"):
+        want_text, path = True, path[: -len("/text()")]
+    elif "/@" in path:
+        path, attr = path.rsplit("/@", 1)
+    try:
+        root = ET.fromstring(doc)
+    except ET.ParseError:
+        return None
+    # XPath evaluates from the DOCUMENT node (Java semantics): both '/a/b'
+    # and 'a/b' take their first step against the root element's tag
+    if path.startswith("//"):
+        path = "." + path  # '//c' -> './/c' (descendant search)
+    elif path:
+        steps = path.lstrip("/").split("/", 1)
+        if steps[0] not in (root.tag, "*", "."):
+            return []
+        path = steps[1] if len(steps) > 1 else "."
+    nodes = root.findall(path) if path else [root]
+    if attr is not None:
+        return [n.get(attr) for n in nodes if n.get(attr) is not None]
+    return [(n.text or "") for n in nodes] if (want_text or True) else nodes
+
+
+def _xpath_first(doc, path):
+    got = _xpath_nodes(doc, path)
+    return got[0] if got else None
+
+
+def _f_xpath(args, out, chunk, ev):
+    from .column import ListColumn, StringColumn
+
+    path = _scalarize(args[1]).value
+    c = _col(args[0], chunk)
+    rows = [_xpath_nodes(v, path) for v in c.to_pylist()]
+    flat = [x for r in rows if r for x in r]
+    offs = [0]
+    for r in rows:
+        offs.append(offs[-1] + (len(r) if r else 0))
+    validity = None
+    if any(r is None for r in rows):
+        validity = torch.tensor([0 if r is None else 1 for r in rows],
+                                dtype=torch.uint8, device=chunk.device)
+    child = StringColumn.from_pylist(flat, device=str(chunk.device))
+    return ListColumn(torch.tensor(offs, dtype=torch.int64,
+                                   device=chunk.device), child, validity, out)
+
+
+def _f_xpath_typed(cast):
+    def run(args, out, chunk, ev):
+        path = _scalarize(args[1]).value
+        c = _col(args[0], chunk)
+        vals = [cast(_xpath_first(v, path)) for v in c.to_pylist()]
+        return Column.from_values(vals, out, device=str(chunk.device))
+    return run
+
+
+def _xp_num(v):
+    if v is None or v == "":
+        return None
+    try:
+        return float(v)
+    except ValueError:
+        return None
+
+
+def _xp_int(v):
+    f = _xp_num(v)
+    return None if f is None else int(f)
+
+
+def _f_xpath_boolean(args, out, chunk, ev):
+    path = _scalarize(args[1]).value
+    c = _col(args[0], chunk)
+    vals = [None if (r := _xpath_nodes(v, path)) is None else bool(r)
+            for v in c.to_pylist()]
+    return Column.from_values(vals, out, device=str(chunk.device))
+
+
+def _csv_split(line, sep):
+    import csv as _csv
+    import io as _io
+
+    if line is None:
+        return None
+    return next(_csv.reader(_io.StringIO(line), delimiter=sep))
+
+
+def _f_from_csv(args, out, chunk, ev):
+    from .column import StructColumn
+
+    sep = ","
+    if len(args) > 2:
+        opts = _scalarize(args[2]).value
+        if isinstance(opts, str) and opts:
+            import json as _json
+
+            try:
+                sep = _json.loads(opts).get("sep", ",")
+            except ValueError:
+                pass
+    c = _col(args[0], chunk)
+    rows = [_csv_split(v, sep) for v in c.to_pylist()]
+    kids = []
+    for i, f in enumerate(out.fields):
+        raw = [None if r is None or i >= len(r) or r[i] == "" else r[i]
+               for r in rows]
+        if not isinstance(f.dtype, T.StringType):
+            conv = []
+            for v in raw:
+                if v is None:
+                    conv.append(None)
+                else:
+                    try:
+                        conv.append(float(v) if isinstance(
+                            f.dtype, (T.Float32Type, T.Float64Type,
+                                      T.DecimalType)) else int(v))
+                    except ValueError:
+                        conv.append(None)
+            raw = conv
+        kids.append((f.name, Column.from_values(raw, f.dtype,
+                                                device=str(chunk.device))))
+    validity = None
+    if any(r is None for r in rows):
+        validity = torch.tensor([0 if r is None else 1 for r in rows],
+                                dtype=torch.uint8, device=chunk.device)
+    return StructColumn(kids, validity, dtype=out)
+
+
+def _f_to_csv(args, out, chunk, ev):
+    from .column import StringColumn
+
+    c = _col(args[0], chunk)
+    names = [nm for nm, _ in c.children_]
+    cols = [kid.to_pylist() for _, kid in c.children_]
+    n = len(c)
+    vals = []
+    for i in range(n):
+        parts = []
+        for j in range(len(names)):
+            v = cols[j][i]
+            if v is None:
+                parts.append("")
+            elif isinstance(v, bool):
+                parts.append("true" if v else "false")
+            else:
+                parts.append(str(v))
+        vals.append(",".join(parts))
+    vcol = StringColumn.from_pylist(vals, device=str(chunk.device))
+    vcol.validity = c.validity
+    return vcol
+
+
+def _f_schema_of_csv(args, out, chunk, ev):
+    sample = _scalarize(args[0]).value
+    parts = _csv_split(sample, ",") or []
+    kinds = []
+    for p in parts:
+        try:
+            int(p)
+            kinds.append("BIGINT")
+            continue
+        except ValueError:
+            pass
+        try:
+            float(p)
+            kinds.append("DOUBLE")
+            continue
+        except ValueError:
+            kinds.append("STRING")
+    body = ", ".join(f"_c{i}: {k}" for i, k in enumerate(kinds))
+    from .column import StringColumn
+
+    return StringColumn.from_pylist([f"STRUCT<{body}>"] * chunk.num_rows,
+                                    device=str(chunk.device))
+
+
+def _canon_json(v, strict):
+    import json as _json
+
+    if v is None:
+        return None
+    try:
+        return _json.dumps(_json.loads(v), separators=(",", ":"))
+    except (ValueError, TypeError):
+        if strict:
+            raise ValueError(f"parse_json: malformed JSON: {v!r}")
+        return None
+
+
+def _f_parse_json(args, out, chunk, ev):
+    return _dict_transform(lambda v: _canon_json(v, True))(args[:1], out, chunk, ev)
+
+
+def _f_try_parse_json(args, out, chunk, ev):
+    return _dict_transform(lambda v: _canon_json(v, False))(args[:1], out, chunk, ev)
+
+
+def _f_variant_get(args, out, chunk, ev):
+    # variant is stored as canonical JSON text (documented simplification);
+    # optional third `type` arg is accepted — cast happens via the SQL CAST
+    path = _scalarize(args[1]).value
+    return _dict_transform(
+        lambda v: _json_path_get(v, path))(args[:1], out, chunk, ev)
+
+
+def _f_is_variant_null(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    vals = [None if v is None else (v.strip() == "null")
+            for v in c.to_pylist()]
+    return Column.from_values(vals, out, device=str(chunk.device))
+
+
+def _schema_of_variant_one(v):
+    import json as _json
+
+    if v is None:
+        return None
+    try:
+        obj = _json.loads(v)
+    except (ValueError, TypeError):
+        return None
+
+    def name(o):
+        if o is None:
+            return "VOID"
+        if isinstance(o, bool):
+            return "BOOLEAN"
+        if isinstance(o, int):
+            return "BIGINT"
+        if isinstance(o, float):
+            return "DOUBLE"
+        if isinstance(o, str):
+            return "STRING"
+        if isinstance(o, list):
+            inner = {name(x) for x in o} or {"VOID"}
+            return f"ARRAY<{inner.pop() if len(inner) == 1 else 'VARIANT'}>"
+        return ("OBJECT<" + ", ".join(
+            f"{k}: {name(x)}" for k, x in sorted(o.items())) + ">")
+
+    return name(obj)
+
+
+def _f_schema_of_variant(args, out, chunk, ev):
+    return _dict_transform(_schema_of_variant_one)(args[:1], out, chunk, ev)
+
+
+def _f_json_array_length(args, out, chunk, ev):
+    import json as _json
+
+    def ln(v):
+        if v is None:
+            return None
+        try:
+            obj = _json.loads(v)
+        except (ValueError, TypeError):
+            return None
+        return len(obj) if isinstance(obj, list) else None
+
+    c = _col(args[0], chunk)
+    vals = [ln(v) for v in c.to_pylist()]
+    return Column.from_values(vals, out, device=str(chunk.device))
+
+
+def _f_json_object_keys(args, out, chunk, ev):
+    import json as _json
+
+    from .column import ListColumn, StringColumn
+
+    c = _col(args[0], chunk)
+    rows = []
+    for v in c.to_pylist():
+        try:
+            obj = None if v is None else _json.loads(v)
+        except (ValueError, TypeError):
+            obj = None
+        rows.append(list(obj.keys()) if isinstance(obj, dict) else None)
+    flat = [k for r in rows if r for k in r]
+    offs = [0]
+    for r in rows:
+        offs.append(offs[-1] + (len(r) if r else 0))
+    validity = None
+    if any(r is None for r in rows):
+        validity = torch.tensor([0 if r is None else 1 for r in rows],
+                                dtype=torch.uint8, device=chunk.device)
+    return ListColumn(torch.tensor(offs, dtype=torch.int64, device=chunk.device),
+                      StringColumn.from_pylist(flat, device=str(chunk.device)),
+                      validity, out)
+
+
+def _luhn_ok(v):
+    if v is None or not v.isdigit() or not v:
+        return False if v is not None else None
+    total = 0
+    for i, ch in enumerate(reversed(v)):
+        d = ord(ch) - 48
+        if i % 2 == 1:
+            d *= 2
+            if d > 9:
+                d -= 9
+        total += d
+    return total % 10 == 0
+
+
+def _f_luhn_check(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    vals = [_luhn_ok(v) for v in c.to_pylist()]
+    return Column.from_values(vals, out, device=str(chunk.device))
+
+
+_CRC32C_TABLE = None
+
+
+def _crc32c(data: bytes) -> int:
+    global _CRC32C_TABLE
+    if _CRC32C_TABLE is None:
+        tbl = []
+        for i in range(256):
+            c = i
+            for _ in range(8):
+                c = (c >> 1) ^ 0x82F63B78 if c & 1 else c >> 1
+            tbl.append(c)
+        _CRC32C_TABLE = tbl
+    crc = 0xFFFFFFFF
+    for b in data:
+        crc = (crc >> 8) ^ _CRC32C_TABLE[(crc ^ b) & 0xFF]
+    return crc ^ 0xFFFFFFFF
+
+
+def _f_crc32c(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    vals = [None if v is None else _crc32c(v.encode()) for v in c.to_pylist()]
+    return Column.from_values(vals, out, device=str(chunk.device))
+
+
+_IMPLS["parse_url"] = _f_parse_url
+_IMPLS["try_parse_url"] = _f_parse_url
+_IMPLS["url_encode"] = _f_url_encode
+_IMPLS["url_decode"] = _f_url_decode
+_IMPLS["xpath"] = _f_xpath
+_IMPLS["xpath_string"] = _f_xpath_typed(lambda v: v)
+_IMPLS["xpath_int"] = _f_xpath_typed(_xp_int)
+_IMPLS["xpath_short"] = _f_xpath_typed(_xp_int)
+_IMPLS["xpath_long"] = _f_xpath_typed(_xp_int)
+_IMPLS["xpath_double"] = _f_xpath_typed(_xp_num)
+_IMPLS["xpath_float"] = _f_xpath_typed(_xp_num)
+_IMPLS["xpath_number"] = _f_xpath_typed(_xp_num)
+_IMPLS["xpath_boolean"] = _f_xpath_boolean
+_IMPLS["from_csv"] = _f_from_csv
+_IMPLS["to_csv"] = _f_to_csv
+_IMPLS["schema_of_csv"] = _f_schema_of_csv
+_IMPLS["parse_json"] = _f_parse_json
+_IMPLS["try_parse_json"] = _f_try_parse_json
+_IMPLS["variant_get"] = _f_variant_get
+_IMPLS["try_variant_get"] = _f_variant_get
+_IMPLS["is_variant_null"] = _f_is_variant_null
+_IMPLS["schema_of_variant"] = _f_schema_of_variant
+_IMPLS["json_array_length"] = _f_json_array_length
+_IMPLS["json_object_keys"] = _f_json_object_keys
+_IMPLS["luhn_check"] = _f_luhn_check
+_IMPLS["crc32c"] = _f_crc32c

@@ -214,6 +214,20 @@ _reg("map_values", lambda a: T.ArrayType(a[0].value) if isinstance(a[0], T.MapTy
 _reg("map_contains_key", _bool)
 _reg("struct named_struct get_field", lambda a: T.NULL)  # typed structurally in the resolver
 _reg("get_json_object to_json schema_of_json json_tuple", _string)
+_reg("json_array_length", _i32)
+_reg("json_object_keys", lambda a: T.ArrayType(T.STRING))
+_reg("from_csv", lambda a: T.NULL)  # typed structurally in the resolver
+_reg("to_csv schema_of_csv", _string)
+_reg("parse_url try_parse_url url_encode url_decode", _string)
+_reg("xpath", lambda a: T.ArrayType(T.STRING))
+_reg("xpath_string", _string)
+_reg("xpath_boolean", _bool)
+_reg("xpath_int xpath_short", _i32)
+_reg("xpath_long", _i64)
+_reg("xpath_double xpath_float xpath_number", _f64)
+_reg("parse_json try_parse_json schema_of_variant variant_get try_variant_get", _string)
+_reg("is_variant_null luhn_check", _bool)
+_reg("crc32c", _i64)
 _reg("from_json", lambda a: T.NULL)  # typed structurally in the resolver
 _reg("date_format from_unixtime", _string)
 _reg("to_timestamp try_to_timestamp", lambda a: T.TIMESTAMP)

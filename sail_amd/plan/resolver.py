@@ -1223,9 +1223,9 @@ class Resolver:
                         raise ResolutionError("named_struct expects literal field names")
                     fields.append(T.StructField(str(k.value), e.args[i + 1].dtype))
                 t = T.StructType(tuple(fields))
-            elif e.name == "from_json":
+            elif e.name in ("from_json", "from_csv"):
                 if len(e.args) < 2 or not isinstance(e.args[1], S.Literal):
-                    raise ResolutionError("from_json expects a literal schema string")
+                    raise ResolutionError(f"{e.name} expects a literal schema string")
                 from ..sql.parser import parse_ddl_schema
 
                 fields = parse_ddl_schema(str(e.args[1].value))

@@ -428,3 +428,64 @@ def test_values_table_factor_and_expressions(s):
     assert s.sql("VALUES (1, 2), (3, 4)").collect() == [(1, 2), (3, 4)]
     assert s.sql("SELECT y, sum(x) FROM VALUES (1,'a'),(2,'a'),(3,'b') AS t(x,y) "
                  "GROUP BY y ORDER BY y").collect() == [("a", 3), ("b", 3)]
+
+
+def test_url_functions(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT parse_url('https://u:p@spark.apache.org:8080/path?query=1#Ref', 'HOST')") \
+        == [("spark.apache.org",)]
+    assert q("SELECT parse_url('https://h/p?a=1&b=2', 'QUERY', 'b')") == [("2",)]
+    assert q("SELECT parse_url('https://u:p@h/p', 'USERINFO'), "
+             "parse_url('https://h/p?a=1', 'FILE'), "
+             "parse_url('https://h/p', 'PROTOCOL')") == [("u:p", "/p?a=1", "https")]
+    assert q("SELECT url_encode('hello world/x'), url_decode('hello+world%2Fx')") \
+        == [("hello+world%2Fx", "hello world/x")]
+
+
+def test_xpath_functions(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT xpath('<a><b>b1</b><b>b2</b></a>', 'a/b/text()')") \
+        == [(["b1", "b2"],)]
+    assert q("SELECT xpath_string('<a><b>bb</b></a>', 'a/b'), "
+             "xpath_int('<a><b>3</b></a>', 'a/b'), "
+             "xpath_double('<a><b>2.5</b></a>', 'a/b'), "
+             "xpath_boolean('<a><b>1</b></a>', 'a/c')") == [("bb", 3, 2.5, False)]
+    assert q("SELECT xpath('<r><x id=\"7\"/><x id=\"9\"/></r>', 'r/x/@id')") \
+        == [(["7", "9"],)]
+    assert q("SELECT xpath('<a><b><c>1</c></b><b><c>2</c></b></a>', '//c/text()')") \
+        == [(["1", "2"],)]
+    # malformed XML -> null
+    assert q("SELECT xpath_string('<oops', 'a')") == [(None,)]
+
+
+def test_csv_functions(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT from_csv('1,apple', 'a INT, b STRING').a, "
+             "from_csv('1,apple', 'a INT, b STRING').b") == [(1, "apple")]
+    assert q("SELECT to_csv(named_struct('a', 1, 'b', 'x'))") == [("1,x",)]
+    assert q("SELECT schema_of_csv('1,abc,2.5')") \
+        == [("STRUCT<_c0: BIGINT, _c1: STRING, _c2: DOUBLE>",)]
+
+
+def test_variant_functions(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("""SELECT parse_json('{"b": 1, "a": [1,2]}')""") \
+        == [('{"b":1,"a":[1,2]}',)]
+    assert q("SELECT try_parse_json('oops')") == [(None,)]
+    assert q("""SELECT variant_get(parse_json('{"a":[1,{"b":5}]}'), '$.a[1].b')""") \
+        == [("5",)]
+    assert q("SELECT is_variant_null(parse_json('null'))") == [(True,)]
+    assert q("""SELECT schema_of_variant('{"x": 1.5, "y": "s"}')""") \
+        == [("OBJECT<x: DOUBLE, y: STRING>",)]
+    with pytest.raises(Exception):
+        q("SELECT parse_json('bad')")
+
+
+def test_json_misc_and_checks(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT json_array_length('[1,2,3]'), json_array_length('{}')") \
+        == [(3, None)]
+    assert q("""SELECT json_object_keys('{"a":1,"b":2}')""") == [(["a", "b"],)]
+    assert q("SELECT luhn_check('79927398713'), luhn_check('79927398714')") \
+        == [(True, False)]
+    assert q("SELECT crc32c('abc')") == [(910901175,)]

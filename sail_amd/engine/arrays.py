@@ -531,6 +531,8 @@ def eval_hof(ev, e, chunk):
     name = e.name.lower()
     if name in ("transform_keys", "transform_values", "map_filter"):
         return eval_map_hof(ev, e, chunk)
+    if name == "map_zip_with":
+        return _f_map_zip_with_eval(ev, e, chunk)
     if name in ("aggregate", "reduce"):
         return _f_reduce(e.args, e.dtype, chunk, ev)
     if name == "zip_with":
@@ -863,3 +865,289 @@ def eval_map_hof(ev, e, chunk):
     torch.cumsum(lens, 0, out=offs[1:])
     return MapColumn(offs, m.keys.gather(kept), m.values.gather(kept),
                      m.validity)
+
+
+# ===========================================================================
+# structural array/map builders (host path — these are shape-changing,
+# allocation-bound ops, not hot compute; ref: sail-function scalar/map,
+# scalar/collection)
+# ===========================================================================
+def _col(a, chunk):
+    from .eval import broadcast
+
+    return broadcast(a, chunk.num_rows, chunk.device)
+
+
+def _scalarize(a):
+    from .functions_impl import _scalarize as _sc
+
+    return _sc(a)
+
+
+def _build_list_generic(rows, elem_type, device):
+    """ListColumn from python lists whose element type may be a struct."""
+    from .column import MapColumn, StructColumn
+
+    if isinstance(elem_type, T.StructType):
+        flat = [x for r in rows if r is not None for x in r]
+        offs = [0]
+        for r in rows:
+            offs.append(offs[-1] + (len(r) if r is not None else 0))
+        kids = []
+        for i, f in enumerate(elem_type.fields):
+            vals = [None if x is None else
+                    (x.get(f.name) if isinstance(x, dict) else x[i])
+                    for x in flat]
+            if isinstance(f.dtype, T.StringType):
+                kids.append((f.name, StringColumn.from_pylist(
+                    vals, device=str(device))))
+            else:
+                kids.append((f.name, Column.from_values(
+                    vals, f.dtype, device=device)))
+        child = StructColumn(kids, dtype=elem_type)
+        validity = None
+        if any(r is None for r in rows):
+            validity = torch.tensor([0 if r is None else 1 for r in rows],
+                                    dtype=torch.uint8, device=device)
+        return ListColumn(torch.tensor(offs, dtype=torch.int64, device=device),
+                          child, validity,
+                          T.ArrayType(elem_type))
+    return ListColumn.from_pylist(rows, elem_type, device=str(device))
+
+
+def _f_arrays_zip(args, out, chunk, ev):
+    """arrays_zip(a1, a2, ...) -> array<struct> (shorter arrays padded with
+    nulls, Spark semantics)."""
+    cols = [_col(a, chunk) for a in args]
+    pys = [c.to_pylist() for c in cols]
+    n = chunk.num_rows
+    rows = []
+    for i in range(n):
+        arrs = [p[i] for p in pys]
+        if all(a is None for a in arrs):
+            rows.append(None)
+            continue
+        ln = max(len(a) for a in arrs if a is not None)
+        rows.append([tuple((a[j] if a is not None and j < len(a) else None)
+                           for a in arrs) for j in range(ln)])
+    return _build_list_generic(rows, out.element, chunk.device)
+
+
+def _f_array_insert(args, out, chunk, ev):
+    """array_insert(arr, pos, val): 1-based; negative counts from the end."""
+    arr = _col(args[0], chunk)
+    rows = arr.to_pylist()
+    from .eval import broadcast
+
+    n = chunk.num_rows
+    pos = broadcast(args[1], n, chunk.device).to_pylist()
+    val = broadcast(args[2], n, chunk.device).to_pylist()
+    res = []
+    for r, p, v in zip(rows, pos, val):
+        if r is None or p is None or p == 0:
+            res.append(None)
+            continue
+        r = list(r)
+        if p > 0:
+            while len(r) < p - 1:
+                r.append(None)
+            r.insert(p - 1, v)
+        else:
+            k = len(r) + p + 1
+            if k < 0:
+                r = [v] + [None] * (-k) + r
+            else:
+                r.insert(k, v)
+        res.append(r)
+    return _build_list_generic(res, out.element, chunk.device)
+
+
+def _f_array_contains_all(args, out, chunk, ev):
+    a = _col(args[0], chunk).to_pylist()
+    b = _col(args[1], chunk).to_pylist()
+    vals = [None if (x is None or y is None)
+            else all(e in x for e in y) for x, y in zip(a, b)]
+    return Column.from_values(vals, T.BOOL, device=chunk.device)
+
+
+def _f_map_concat(args, out, chunk, ev):
+    maps = [_col(a, chunk).to_pylist() for a in args]
+    n = chunk.num_rows
+    rows = []
+    for i in range(n):
+        ms = [m[i] for m in maps]
+        if all(m is None for m in ms):
+            rows.append(None)
+            continue
+        merged = {}
+        for m in ms:
+            if m:
+                merged.update(m)  # later maps win (Spark LAST_WIN policy)
+        rows.append(merged)
+    from .column import MapColumn
+
+    return MapColumn.from_pylist(rows, out.key, out.value,
+                                 device=str(chunk.device))
+
+
+def _f_map_entries(args, out, chunk, ev):
+    m = _col(args[0], chunk).to_pylist()
+    rows = [None if r is None else [(k, v) for k, v in r.items()] for r in m]
+    return _build_list_generic(rows, out.element, chunk.device)
+
+
+def _f_map_from_entries(args, out, chunk, ev):
+    rows = _col(args[0], chunk).to_pylist()
+    res = []
+    for r in rows:
+        if r is None:
+            res.append(None)
+            continue
+        d = {}
+        for e in r:
+            if e is None:
+                continue
+            k, v = (e.get(f.name) for f in args[0].dtype.element.fields) \
+                if isinstance(e, dict) else (e[0], e[1])
+            d[k] = v
+        res.append(d)
+    from .column import MapColumn
+
+    return MapColumn.from_pylist(res, out.key, out.value,
+                                 device=str(chunk.device))
+
+
+def _f_str_to_map(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    pair_d = _scalarize(args[1]).value if len(args) > 1 else ","
+    kv_d = _scalarize(args[2]).value if len(args) > 2 else ":"
+    rows = []
+    for v in c.to_pylist():
+        if v is None:
+            rows.append(None)
+            continue
+        d = {}
+        for pair in v.split(pair_d):
+            if kv_d in pair:
+                k, val = pair.split(kv_d, 1)
+                d[k] = val
+            elif pair:
+                d[pair] = None
+        rows.append(d)
+    from .column import MapColumn
+
+    return MapColumn.from_pylist(rows, T.STRING, T.STRING,
+                                 device=str(chunk.device))
+
+
+def _col_from_py(vals, dtype, device):
+    if isinstance(dtype, T.StringType):
+        return StringColumn.from_pylist(vals, device=str(device))
+    return Column.from_values(vals, dtype, device=device)
+
+
+def _f_map_zip_with_eval(ev, e, chunk):
+    """map_zip_with(m1, m2, (k, v1, v2) -> ...): per-row key union, lambda
+    over the aligned flat entries (params at 0..2, row columns at 3+)."""
+    from .chunk import Chunk
+    from .column import MapColumn
+    from .eval import broadcast
+
+    m1 = _bcast(ev.eval(e.args[0], chunk), chunk)
+    m2 = _bcast(ev.eval(e.args[1], chunk), chunk)
+    lam = e.args[2]
+    r1, r2 = m1.to_pylist(), m2.to_pylist()
+    n = chunk.num_rows
+    keys, v1, v2, offs, valid = [], [], [], [0], []
+    for i in range(n):
+        a, b = r1[i], r2[i]
+        if a is None and b is None:
+            valid.append(0)
+            offs.append(offs[-1])
+            continue
+        valid.append(1)
+        a = a or {}
+        b = b or {}
+        uk = list(a.keys()) + [k for k in b.keys() if k not in a]
+        keys.extend(uk)
+        v1.extend(a.get(k) for k in uk)
+        v2.extend(b.get(k) for k in uk)
+        offs.append(offs[-1] + len(uk))
+    dev = chunk.device
+    kcol = _col_from_py(keys, m1.dtype.key, dev)
+    flat_cols = [kcol,
+                 _col_from_py(v1, m1.dtype.value, dev),
+                 _col_from_py(v2, m2.dtype.value, dev)]
+    seg_rows = []
+    for i in range(n):
+        seg_rows.extend([i] * (offs[i + 1] - offs[i]))
+    seg = torch.tensor(seg_rows, dtype=torch.int64, device=dev)
+    flat_cols += [c.gather(seg) if c is not None else None
+                  for c in chunk.columns]
+    flat = Chunk(flat_cols, [f"__z{i}" for i in range(len(flat_cols))],
+                 chunk.partitioning)
+    flat.forced_rows = len(keys)
+    res = broadcast(ev.eval(lam.body, flat), len(keys), dev)
+    validity = None
+    if not all(valid):
+        validity = torch.tensor(valid, dtype=torch.uint8, device=dev)
+    return MapColumn(torch.tensor(offs, dtype=torch.int64, device=dev),
+                     kcol, res, validity, dtype=e.dtype)
+
+
+# -- vector math over array<double> (segment ops, device path) -------------
+def _vec(args, chunk, i=0):
+    c = _col(args[i], chunk)
+    return c
+
+
+def _f_vector_norm(args, out, chunk, ev):
+    c = _vec(args, chunk)
+    p = float(_scalarize(args[1]).value) if len(args) > 1 else 2.0
+    seg = c.segment_ids()
+    x = c.child.data.to(torch.float64).abs() ** p
+    sums = torch.zeros(len(c), dtype=torch.float64, device=c.device)
+    sums.index_add_(0, seg, x)
+    return Column(T.F64, sums ** (1.0 / p), c.validity)
+
+
+def _f_vector_inner_product(args, out, chunk, ev):
+    a, b = _vec(args, chunk, 0), _vec(args, chunk, 1)
+    seg = a.segment_ids()
+    prod = a.child.data.to(torch.float64) * b.child.data.to(torch.float64)
+    sums = torch.zeros(len(a), dtype=torch.float64, device=a.device)
+    sums.index_add_(0, seg, prod)
+    return Column(T.F64, sums, a.validity)
+
+
+def _f_vector_l2_distance(args, out, chunk, ev):
+    a, b = _vec(args, chunk, 0), _vec(args, chunk, 1)
+    seg = a.segment_ids()
+    d = a.child.data.to(torch.float64) - b.child.data.to(torch.float64)
+    sums = torch.zeros(len(a), dtype=torch.float64, device=a.device)
+    sums.index_add_(0, seg, d * d)
+    return Column(T.F64, sums.sqrt(), a.validity)
+
+
+def _f_vector_normalize(args, out, chunk, ev):
+    c = _vec(args, chunk)
+    seg = c.segment_ids()
+    x = c.child.data.to(torch.float64)
+    sums = torch.zeros(len(c), dtype=torch.float64, device=c.device)
+    sums.index_add_(0, seg, x * x)
+    norm = sums.sqrt().clamp_min(1e-300)
+    return ListColumn(c.offsets, Column(T.F64, x / norm[seg], None),
+                      c.validity, T.ArrayType(T.F64))
+
+
+IMPLS["arrays_zip"] = _f_arrays_zip
+IMPLS["array_insert"] = _f_array_insert
+IMPLS["array_contains_all"] = _f_array_contains_all
+IMPLS["map_concat"] = _f_map_concat
+IMPLS["map_entries"] = _f_map_entries
+IMPLS["map_from_entries"] = _f_map_from_entries
+IMPLS["str_to_map"] = _f_str_to_map
+IMPLS["vector_norm"] = _f_vector_norm
+IMPLS["vector_inner_product"] = _f_vector_inner_product
+IMPLS["vector_l2_distance"] = _f_vector_l2_distance
+IMPLS["vector_normalize"] = _f_vector_normalize

@@ -410,7 +410,7 @@ class Evaluator:
             return eval_hof(self, e, chunk)
         args = [self.eval(a, chunk) for a in e.args]
         if args and all(isinstance(a, Scalar) for a in args) \
-                and e.name not in ("rand", "randn", "uuid", "monotonically_increasing_id") \
+                and e.name not in ("rand", "randn", "uuid", "monotonically_increasing_id", "random", "uniform", "randstr") \
                 and not isinstance(e.dtype, (T.ArrayType, T.MapType, T.StructType)) \
                 and not any(isinstance(a.dtype, (T.ArrayType, T.MapType, T.StructType))
                             for a in args):
@@ -739,6 +739,10 @@ def _rescale_int(data: torch.Tensor, cur_scale: int, target_scale: int) -> torch
 
 
 def _scalar_binop(op: str, l: Scalar, r: Scalar, out_type) -> Scalar:
+    if op == "<=>":  # null-safe: never returns null
+        if l.is_null or r.is_null:
+            return Scalar(l.is_null and r.is_null, T.BOOL)
+        return Scalar(l.value == r.value, T.BOOL)
     if l.is_null or r.is_null:
         return Scalar(None, out_type or T.NULL)
     a, b = l.value, r.value

@@ -508,6 +508,12 @@ def _f_substring(args, out, chunk, ev):
 def _f_concat(args, out, chunk, ev):
     from .eval import Scalar
 
+    from .column import ListColumn
+
+    if args and any(isinstance(a, ListColumn)
+                    or isinstance(getattr(a, "dtype", None), T.ArrayType)
+                    for a in args):
+        return _IMPLS["array_concat"](args, out, chunk, ev)
     n = chunk.num_rows
     parts = []
     for a in args:
@@ -1780,3 +1786,404 @@ _IMPLS["json_array_length"] = _f_json_array_length
 _IMPLS["json_object_keys"] = _f_json_object_keys
 _IMPLS["luhn_check"] = _f_luhn_check
 _IMPLS["crc32c"] = _f_crc32c
+
+
+
+def _f_array_concat(args, out, chunk, ev):
+    from .arrays import _bcast, _concat_rows
+
+    acc = _bcast(args[0], chunk)
+    for a in args[1:]:
+        acc = _concat_rows(acc, _bcast(a, chunk))
+    return acc
+
+
+_IMPLS["array_concat"] = _f_array_concat
+
+
+# -- misc Spark scalar surface: regexp_*, mask/quote, timestamp arithmetic,
+#    randoms, to_number/to_binary, bit access (ref: sail-function
+#    src/scalar/{string,math,datetime,misc}) --------------------------------
+
+def _f_strpos(args, out, chunk, ev):
+    return _IMPLS["instr"](args, out, chunk, ev)
+
+
+def _f_quote(args, out, chunk, ev):
+    return _dict_transform(
+        lambda v: None if v is None else
+        "'" + v.replace("\\", "\\\\").replace("'", "\\'") + "'"
+    )(args[:1], out, chunk, ev)
+
+
+def _mask_one(v, up, lo, dig, other):
+    if v is None:
+        return None
+    out = []
+    for ch in v:
+        if ch.isupper():
+            out.append(ch if up is None else up)
+        elif ch.islower():
+            out.append(ch if lo is None else lo)
+        elif ch.isdigit():
+            out.append(ch if dig is None else dig)
+        else:
+            out.append(ch if other is None else other)
+    return "".join(out)
+
+
+def _f_mask(args, out, chunk, ev):
+    def opt(i, default):
+        if len(args) <= i:
+            return default
+        v = _scalarize(args[i]).value
+        return v
+    up = opt(1, "X")
+    lo = opt(2, "x")
+    dig = opt(3, "n")
+    other = opt(4, None)
+    return _dict_transform(
+        lambda v: _mask_one(v, up, lo, dig, other))(args[:1], out, chunk, ev)
+
+
+def _f_regexp_count(args, out, chunk, ev):
+    import re as _re
+
+    pat = _re.compile(_scalarize(args[1]).value)
+    c = _col(args[0], chunk)
+    vals = [None if v is None else len(pat.findall(v)) for v in c.to_pylist()]
+    return Column.from_values(vals, out, device=str(chunk.device))
+
+
+def _f_regexp_instr(args, out, chunk, ev):
+    import re as _re
+
+    pat = _re.compile(_scalarize(args[1]).value)
+    idx = int(_scalarize(args[2]).value) if len(args) > 2 else 0
+    c = _col(args[0], chunk)
+    vals = []
+    for v in c.to_pylist():
+        if v is None:
+            vals.append(None)
+            continue
+        m = pat.search(v)
+        vals.append(0 if m is None else
+                    (m.start() + 1 if idx == 0 else m.end() + 1))
+    return Column.from_values(vals, out, device=str(chunk.device))
+
+
+def _f_regexp_substr(args, out, chunk, ev):
+    import re as _re
+
+    pat = _re.compile(_scalarize(args[1]).value)
+    return _dict_transform(
+        lambda v: None if v is None else
+        (lambda m: m.group(0) if m else None)(pat.search(v))
+    )(args[:1], out, chunk, ev)
+
+
+_DAYNAMES = ["Mon", "Tue", "Wed", "Thu", "Fri", "Sat", "Sun"]
+
+
+def _f_dayname(args, out, chunk, ev):
+    from .column import StringColumn
+
+    c = _col(args[0], chunk)
+    days = c.data.to(torch.int64)
+    if isinstance(c.dtype, T.TimestampType):
+        days = torch.div(days, 86_400_000_000, rounding_mode="floor")
+    dow = ((days % 7) + 3) % 7  # 1970-01-01 was a Thursday
+    vals = [_DAYNAMES[int(d)] for d in dow.tolist()]
+    col = StringColumn.from_pylist(vals, device=str(chunk.device))
+    col.validity = c.validity
+    return col
+
+
+def _f_date_from_unix_date(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    return Column(T.DATE, c.data.to(torch.int32), c.validity)
+
+
+_TS_UNITS = {"MICROSECOND": 1, "MILLISECOND": 1000, "SECOND": 1_000_000,
+             "MINUTE": 60_000_000, "HOUR": 3_600_000_000,
+             "DAY": 86_400_000_000, "WEEK": 7 * 86_400_000_000}
+
+
+def _ts_us(c):
+    us = c.data.to(torch.int64)
+    if isinstance(c.dtype, T.DateType):
+        us = us * 86_400_000_000
+    return us
+
+
+def _f_timestampadd(args, out, chunk, ev):
+    unit = str(_scalarize(args[0]).value).upper()
+    n = _col(args[1], chunk).data.to(torch.int64)
+    c = _col(args[2], chunk)
+    us = _ts_us(c)
+    if unit in _TS_UNITS:
+        res = us + n * _TS_UNITS[unit]
+    elif unit in ("MONTH", "QUARTER", "YEAR"):
+        import datetime as _dt
+
+        mult = {"MONTH": 1, "QUARTER": 3, "YEAR": 12}[unit]
+        vals = []
+        for u, k in zip(us.tolist(), n.tolist()):
+            d = _dt.datetime(1970, 1, 1) + _dt.timedelta(microseconds=u)
+            total = d.year * 12 + (d.month - 1) + k * mult
+            y, m = divmod(total, 12)
+            import calendar as _cal
+
+            day = min(d.day, _cal.monthrange(y, m + 1)[1])
+            nd = d.replace(year=y, month=m + 1, day=day)
+            vals.append(int((nd - _dt.datetime(1970, 1, 1)).total_seconds()
+                            * 1_000_000) + d.microsecond % 1)
+        res = torch.tensor(vals, dtype=torch.int64, device=chunk.device)
+    else:
+        raise ValueError(f"timestampadd: bad unit {unit}")
+    return Column(T.TIMESTAMP, res, c.validity)
+
+
+def _f_timestampdiff(args, out, chunk, ev):
+    unit = str(_scalarize(args[0]).value).upper()
+    a = _col(args[1], chunk)
+    b = _col(args[2], chunk)
+    ua, ub = _ts_us(a), _ts_us(b)
+    if unit in _TS_UNITS:
+        res = torch.div(ub - ua, _TS_UNITS[unit], rounding_mode="trunc")
+    elif unit in ("MONTH", "QUARTER", "YEAR"):
+        import datetime as _dt
+
+        div = {"MONTH": 1, "QUARTER": 3, "YEAR": 12}[unit]
+        vals = []
+        for x, y in zip(ua.tolist(), ub.tolist()):
+            da = _dt.datetime(1970, 1, 1) + _dt.timedelta(microseconds=x)
+            db = _dt.datetime(1970, 1, 1) + _dt.timedelta(microseconds=y)
+            months = (db.year - da.year) * 12 + db.month - da.month
+            # partial months don't count (Spark truncates toward zero)
+            if months > 0 and (db.day, db.time()) < (da.day, da.time()):
+                months -= 1
+            elif months < 0 and (db.day, db.time()) > (da.day, da.time()):
+                months += 1
+            vals.append(months // div if months >= 0 else -((-months) // div))
+        res = torch.tensor(vals, dtype=torch.int64, device=chunk.device)
+    else:
+        raise ValueError(f"timestampdiff: bad unit {unit}")
+    validity = None
+    if a.validity is not None or b.validity is not None:
+        validity = (a.valid_mask() & b.valid_mask()).to(torch.uint8)
+    return Column(T.I64, res, validity)
+
+
+def _f_convert_timezone(args, out, chunk, ev):
+    from zoneinfo import ZoneInfo
+
+    import datetime as _dt
+
+    if len(args) == 2:
+        src, tgt, col = "UTC", _scalarize(args[0]).value, args[1]
+    else:
+        src, tgt, col = (_scalarize(args[0]).value,
+                         _scalarize(args[1]).value, args[2])
+    c = _col(col, chunk)
+    zsrc, ztgt = ZoneInfo(src), ZoneInfo(tgt)
+    vals = []
+    for u in _ts_us(c).tolist():
+        d = _dt.datetime(1970, 1, 1) + _dt.timedelta(microseconds=u)
+        d2 = d.replace(tzinfo=zsrc).astimezone(ztgt).replace(tzinfo=None)
+        vals.append(int((d2 - _dt.datetime(1970, 1, 1)).total_seconds()
+                        * 1_000_000))
+    return Column(T.TIMESTAMP,
+                  torch.tensor(vals, dtype=torch.int64, device=chunk.device),
+                  c.validity)
+
+
+def _f_getbit(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    from .eval import broadcast
+
+    pos = broadcast(args[1], chunk.num_rows, chunk.device)
+    res = (c.data.to(torch.int64) >> pos.data.to(torch.int64)) & 1
+    validity = None
+    if c.validity is not None or pos.validity is not None:
+        validity = (c.valid_mask() & pos.valid_mask()).to(torch.uint8)
+    return Column(T.I32, res.to(torch.int32), validity)
+
+
+def _f_random(args, out, chunk, ev):
+    n = chunk.num_rows
+    if args:
+        g = torch.Generator(device="cpu")
+        g.manual_seed(int(_scalarize(args[0]).value))
+        vals = torch.rand(n, generator=g, dtype=torch.float64).to(chunk.device)
+    else:
+        vals = torch.rand(n, dtype=torch.float64, device=chunk.device)
+    return Column(T.F64, vals, None)
+
+
+def _f_uniform(args, out, chunk, ev):
+    lo = int(_scalarize(args[0]).value)
+    hi = int(_scalarize(args[1]).value)
+    n = chunk.num_rows
+    if len(args) > 2:
+        g = torch.Generator(device="cpu")
+        g.manual_seed(int(_scalarize(args[2]).value))
+        vals = torch.randint(lo, hi + 1, (n,), generator=g).to(chunk.device)
+    else:
+        vals = torch.randint(lo, hi + 1, (n,), device=chunk.device)
+    return Column(T.I64, vals.to(torch.int64), None)
+
+
+def _f_randstr(args, out, chunk, ev):
+    import random as _random
+    import string as _string
+
+    from .column import StringColumn
+
+    ln = int(_scalarize(args[0]).value)
+    rng = _random.Random(int(_scalarize(args[1]).value)) \
+        if len(args) > 1 else _random
+    alpha = _string.ascii_letters + _string.digits
+    vals = ["".join(rng.choice(alpha) for _ in range(ln))
+            for _ in range(chunk.num_rows)]
+    return StringColumn.from_pylist(vals, device=str(chunk.device),
+                                    dict_encode=False)
+
+
+def _to_number_one(v, fmt):
+    if v is None:
+        return None
+    t = v.strip().replace(",", "").replace("$", "")
+    neg = t.startswith("-") or (t.startswith("(") and t.endswith(")"))
+    t = t.strip("()-+")
+    try:
+        x = float(t)
+    except ValueError:
+        return None
+    return -x if neg else x
+
+
+def _f_to_number(args, out, chunk, ev):
+    fmt = _scalarize(args[1]).value if len(args) > 1 else None
+    c = _col(args[0], chunk)
+    vals = [_to_number_one(v, fmt) for v in c.to_pylist()]
+    raws = c.to_pylist()
+    bad = [r for r, v in zip(raws, vals) if r is not None and v is None]
+    if bad:
+        raise ValueError(f"to_number: cannot parse {bad[0]!r}")
+    return Column.from_values(vals, out, device=str(chunk.device))
+
+
+def _f_try_to_number(args, out, chunk, ev):
+    fmt = _scalarize(args[1]).value if len(args) > 1 else None
+    c = _col(args[0], chunk)
+    vals = [_to_number_one(v, fmt) for v in c.to_pylist()]
+    return Column.from_values(vals, out, device=str(chunk.device))
+
+
+def _to_binary_one(v, fmt, strict):
+    import base64 as _b64
+
+    if v is None:
+        return None
+    try:
+        if fmt == "utf-8" or fmt == "utf8":
+            return v
+        if fmt == "hex":
+            return bytes.fromhex(v).decode("latin-1")
+        if fmt == "base64":
+            return _b64.b64decode(v).decode("latin-1")
+    except (ValueError, TypeError):
+        if strict:
+            raise ValueError(f"to_binary: cannot decode {v!r} as {fmt}")
+        return None
+    raise ValueError(f"to_binary: unknown format {fmt!r}")
+
+
+def _f_to_binary(args, out, chunk, ev):
+    fmt = (str(_scalarize(args[1]).value).lower()
+           if len(args) > 1 else "hex")
+    return _dict_transform(
+        lambda v: _to_binary_one(v, fmt, True))(args[:1], out, chunk, ev)
+
+
+def _f_try_to_binary(args, out, chunk, ev):
+    fmt = (str(_scalarize(args[1]).value).lower()
+           if len(args) > 1 else "hex")
+    return _dict_transform(
+        lambda v: _to_binary_one(v, fmt, False))(args[:1], out, chunk, ev)
+
+
+def _f_to_varchar(args, out, chunk, ev):
+    from .column import StringColumn
+
+    c = _col(args[0], chunk)
+    vals = c.to_pylist()
+    res = [None if v is None else
+           ("true" if v is True else "false" if v is False else str(v))
+           for v in vals]
+    return StringColumn.from_pylist(res, device=str(chunk.device))
+
+
+def _f_try_mod(args, out, chunk, ev):
+    from .eval import broadcast
+
+    n = chunk.num_rows
+    a = broadcast(args[0], n, chunk.device)
+    b = broadcast(args[1], n, chunk.device)
+    bz = b.data == 0
+    safe_b = torch.where(bz, torch.ones_like(b.data), b.data)
+    res = torch.fmod(a.data, safe_b) if a.data.is_floating_point() \
+        else a.data - torch.div(a.data, safe_b, rounding_mode="trunc") * safe_b
+    valid = a.valid_mask() & b.valid_mask() & ~bz
+    return Column(out or a.dtype, res, valid.to(torch.uint8))
+
+
+def _f_current_schema(args, out, chunk, ev):
+    from .column import StringColumn
+
+    return StringColumn.from_pylist(["default"] * chunk.num_rows,
+                                    device=str(chunk.device))
+
+
+def _f_user(args, out, chunk, ev):
+    import getpass
+
+    from .column import StringColumn
+
+    try:
+        u = getpass.getuser()
+    except Exception:
+        u = "unknown"
+    return StringColumn.from_pylist([u] * chunk.num_rows,
+                                    device=str(chunk.device))
+
+
+_IMPLS["strpos"] = _f_strpos
+_IMPLS["quote"] = _f_quote
+_IMPLS["mask"] = _f_mask
+_IMPLS["regexp_count"] = _f_regexp_count
+_IMPLS["regexp_instr"] = _f_regexp_instr
+_IMPLS["regexp_substr"] = _f_regexp_substr
+_IMPLS["dayname"] = _f_dayname
+_IMPLS["date_from_unix_date"] = _f_date_from_unix_date
+_IMPLS["timestampadd"] = _f_timestampadd
+_IMPLS["timestampdiff"] = _f_timestampdiff
+_IMPLS["timestamp_add"] = _f_timestampadd
+_IMPLS["timestamp_diff"] = _f_timestampdiff
+_IMPLS["convert_timezone"] = _f_convert_timezone
+_IMPLS["getbit"] = _f_getbit
+_IMPLS["bit_get"] = _f_getbit
+_IMPLS["random"] = _f_random
+_IMPLS["uniform"] = _f_uniform
+_IMPLS["randstr"] = _f_randstr
+_IMPLS["to_number"] = _f_to_number
+_IMPLS["try_to_number"] = _f_try_to_number
+_IMPLS["to_binary"] = _f_to_binary
+_IMPLS["try_to_binary"] = _f_try_to_binary
+_IMPLS["to_varchar"] = _f_to_varchar
+_IMPLS["to_char"] = _f_to_varchar
+_IMPLS["try_mod"] = _f_try_mod
+_IMPLS["current_schema"] = _f_current_schema
+_IMPLS["user"] = _f_user
+_IMPLS["session_user"] = _f_user

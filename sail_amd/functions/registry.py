@@ -148,7 +148,7 @@ _reg("rint", _f64)
 _reg("log", _f64)
 
 # string (ref: sail-function/src/scalar/string)
-_reg("concat concat_ws upper ucase lower lcase trim ltrim rtrim btrim initcap "
+_reg("concat_ws upper ucase lower lcase trim ltrim rtrim btrim initcap "
      "reverse repeat lpad rpad substring substr left right replace translate "
      "regexp_replace regexp_extract regexp_extract_all split_part soundex "
      "chr char space format_string printf format_number substring_index "
@@ -228,6 +228,28 @@ _reg("xpath_double xpath_float xpath_number", _f64)
 _reg("parse_json try_parse_json schema_of_variant variant_get try_variant_get", _string)
 _reg("is_variant_null luhn_check", _bool)
 _reg("crc32c", _i64)
+_reg("array_insert", lambda a: a[0])
+_reg("array_contains_all", _bool)
+_reg("array_concat", lambda a: a[0])
+_reg("concat", lambda a: a[0] if a and isinstance(a[0], T.ArrayType) else T.STRING)
+_reg("map_concat", lambda a: a[0])
+_reg("str_to_map", lambda a: T.MapType(T.STRING, T.STRING))
+_reg("vector_norm vector_inner_product vector_l2_distance", _f64)
+_reg("vector_normalize", lambda a: T.ArrayType(T.F64))
+_reg("arrays_zip map_entries map_from_entries", lambda a: T.NULL)  # resolver-typed
+_reg("strpos regexp_count regexp_instr", _i32)
+_reg("quote mask regexp_substr dayname to_varchar to_char randstr "
+     "current_schema user session_user", _string)
+_reg("date_from_unix_date", lambda a: T.DATE)
+_reg("timestampadd timestamp_add convert_timezone", lambda a: T.TIMESTAMP)
+_reg("timestampdiff timestamp_diff uniform", _i64)
+_reg("random", _f64)
+_reg("to_number try_to_number", _f64)
+_reg("to_binary try_to_binary", lambda a: T.BINARY)
+_reg("getbit bit_get", _i32)
+_reg("try_mod", lambda a: a[0])
+_reg("equal_null", _bool)
+_reg("nullifzero zeroifnull", lambda a: a[0])
 _reg("from_json", lambda a: T.NULL)  # typed structurally in the resolver
 _reg("date_format from_unixtime", _string)
 _reg("to_timestamp try_to_timestamp", lambda a: T.TIMESTAMP)

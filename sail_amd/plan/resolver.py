@@ -1051,7 +1051,7 @@ class Resolver:
         return self._type_expr(out)
 
     _HOF = {"transform", "filter", "exists", "forall", "array_filter",
-            "zip_with", "aggregate", "reduce",
+            "zip_with", "aggregate", "reduce", "map_zip_with",
             "transform_keys", "transform_values", "map_filter"}
 
     def _resolve_hof(self, e: S.Func, scope: Scope) -> S.Expr:
@@ -1066,6 +1066,17 @@ class Resolver:
             return self._resolve_zip_with(e, scope)
         if name in ("aggregate", "reduce"):
             return self._resolve_reduce(e, scope)
+        if name == "map_zip_with":
+            m1 = self._expr(e.args[0], scope)
+            m2 = self._expr(e.args[1], scope)
+            if not (isinstance(m1.dtype, T.MapType)
+                    and isinstance(m2.dtype, T.MapType)):
+                raise ResolutionError("map_zip_with expects two maps")
+            blam = self._bind_lambda(
+                e.args[2], [m1.dtype.key, m1.dtype.value, m2.dtype.value],
+                scope)
+            return S.Func(name, [m1, m2, blam],
+                          T.MapType(m1.dtype.key, blam.dtype))
         if name in ("transform_keys", "transform_values", "map_filter"):
             m = self._expr(e.args[0], scope)
             if not isinstance(m.dtype, T.MapType):
@@ -1223,6 +1234,29 @@ class Resolver:
                         raise ResolutionError("named_struct expects literal field names")
                     fields.append(T.StructField(str(k.value), e.args[i + 1].dtype))
                 t = T.StructType(tuple(fields))
+            elif e.name == "arrays_zip":
+                fields = []
+                for i, a in enumerate(e.args):
+                    if not isinstance(a.dtype, T.ArrayType):
+                        raise ResolutionError("arrays_zip expects arrays")
+                    nm = a.name if isinstance(a, (S.BoundRef, S.Alias)) else str(i)
+                    fields.append(T.StructField(nm, a.dtype.element))
+                t = T.ArrayType(T.StructType(tuple(fields)))
+            elif e.name == "map_entries":
+                mt = e.args[0].dtype
+                if not isinstance(mt, T.MapType):
+                    raise ResolutionError("map_entries expects a map")
+                t = T.ArrayType(T.StructType((T.StructField("key", mt.key),
+                                              T.StructField("value", mt.value))))
+            elif e.name == "map_from_entries":
+                at = e.args[0].dtype
+                if not (isinstance(at, T.ArrayType)
+                        and isinstance(at.element, T.StructType)
+                        and len(at.element.fields) == 2):
+                    raise ResolutionError(
+                        "map_from_entries expects array<struct<k,v>>")
+                t = T.MapType(at.element.fields[0].dtype,
+                              at.element.fields[1].dtype)
             elif e.name in ("from_json", "from_csv"):
                 if len(e.args) < 2 or not isinstance(e.args[1], S.Literal):
                     raise ResolutionError(f"{e.name} expects a literal schema string")

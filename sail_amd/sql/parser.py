@@ -1286,6 +1286,16 @@ class Parser:
         else:
             self.eat_kw("ALL")
         args: List[S.Expr] = []
+        if lname in ("timestampadd", "timestampdiff", "timestamp_add",
+                     "timestamp_diff", "date_add_unit") and \
+                self.peek().kind == "ident" and \
+                self.peek(1).kind == "op" and self.peek(1).value == ",":
+            # unit keyword first argument: timestampadd(HOUR, 2, ts)
+            args.append(S.Literal(self.next().value.upper(), None))
+            self.expect_op(",")
+            args += self._expr_list()
+            self.expect_op(")")
+            return S.Func(lname.replace("timestamp_", "timestamp"), args)
         if not self.at_op(")"):
             if self.at_op("*"):
                 self.next()
@@ -1293,6 +1303,12 @@ class Parser:
             else:
                 args = self._expr_list()
         self.expect_op(")")
+        if lname == "equal_null" and len(args) == 2:
+            return S.BinaryOp("<=>", args[0], args[1])
+        if lname == "nullifzero" and len(args) == 1:
+            return S.Func("nullif", [args[0], S.Literal(0, None)])
+        if lname == "zeroifnull" and len(args) == 1:
+            return S.Func("coalesce", [args[0], S.Literal(0, None)])
         self.eat_kw("IGNORE") and self.expect_kw("NULLS")
         # FILTER (WHERE ...)
         filt = None

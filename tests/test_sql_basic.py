@@ -489,3 +489,84 @@ def test_json_misc_and_checks(s):
     assert q("SELECT luhn_check('79927398713'), luhn_check('79927398714')") \
         == [(True, False)]
     assert q("SELECT crc32c('abc')") == [(910901175,)]
+
+
+def test_collection_builders(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT arrays_zip(xs, ys) FROM (SELECT array(1,2) AS xs, "
+             "array('a','b') AS ys)") == \
+        [([{"xs": 1, "ys": "a"}, {"xs": 2, "ys": "b"}],)]
+    assert q("SELECT array_insert(array(1,3), 2, 2), array_insert(array(1,2), -1, 9)") \
+        == [([1, 2, 3], [1, 2, 9])]
+    assert q("SELECT array_contains_all(array(1,2,3), array(1,2)), "
+             "array_contains_all(array(1), array(2))") == [(True, False)]
+    assert q("SELECT array_concat(array(1), array(2,3)), concat(array(4), array(5))") \
+        == [([1, 2, 3], [4, 5])]
+    assert q("SELECT map_concat(map('a',1), map('b',2))") == [({"a": 1, "b": 2},)]
+    assert q("SELECT map_entries(map('a',1))") == [([{"key": "a", "value": 1}],)]
+    assert q("SELECT map_from_entries(map_entries(map('a',1,'b',2)))") \
+        == [({"a": 1, "b": 2},)]
+    assert q("SELECT str_to_map('a:1,b:2')") == [({"a": "1", "b": "2"},)]
+    assert q("SELECT map_zip_with(map('a',1,'b',2), map('a',10,'c',30), "
+             "(k,v1,v2) -> coalesce(v1,0) + coalesce(v2,0))") \
+        == [({"a": 11, "b": 2, "c": 30},)]
+
+
+def test_vector_functions(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT vector_norm(array(3.0,4.0)), "
+             "vector_inner_product(array(1.0,2.0), array(3.0,4.0)), "
+             "vector_l2_distance(array(0.0,0.0), array(3.0,4.0))") \
+        == [(5.0, 11.0, 5.0)]
+    (vec,), = q("SELECT vector_normalize(array(3.0,4.0))")
+    assert vec == [0.6, 0.8]
+
+
+def test_null_safe_and_try_functions(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT 1 <=> NULL, NULL <=> NULL, 2 <=> 2") == [(False, True, True)]
+    assert q("SELECT equal_null(1, NULL), equal_null(NULL, NULL)") == [(False, True)]
+    assert q("SELECT nullifzero(0), nullifzero(5), zeroifnull(NULL), zeroifnull(7)") \
+        == [(None, 5, 0, 7)]
+    assert q("SELECT try_mod(5, 0), try_mod(7, 3)") == [(None, 1)]
+
+
+def test_misc_string_functions(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT strpos('hello', 'l')") == [(3,)]
+    assert q("SELECT quote(\"Don't\")") == [("'Don\\'t'",)]
+    assert q("SELECT mask('AbCD123-@$#'), mask('AbCD123-@$#', 'Q', 'q', 'd', 'o')") \
+        == [("XxXXnnn-@$#", "QqQQdddoooo")]
+    assert q("SELECT regexp_count('Steven', 'e'), regexp_instr('hello world', 'o'), "
+             "regexp_substr('hello world', 'o.')") == [(2, 5, "o ")]
+    assert q("SELECT to_number('$1,234.56', '999'), try_to_number('oops', '999')") \
+        == [(1234.56, None)]
+    assert q("SELECT to_binary('616263', 'hex'), try_to_binary('_bad_', 'base64')") \
+        == [("abc", None)]
+    assert q("SELECT to_varchar(123), to_char(1.5, '9.9')") == [("123", "1.5")]
+
+
+def test_timestamp_arithmetic_functions(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT timestampdiff(HOUR, TIMESTAMP '2024-01-01 00:00:00', "
+             "TIMESTAMP '2024-01-01 05:30:00')") == [(5,)]
+    # month-add clamps to end of shorter month (Java Calendar semantics)
+    assert q("SELECT timestampadd(MONTH, 1, TIMESTAMP '2024-01-31 00:00:00')") \
+        == [(1709164800000000,)]  # 2024-02-29
+    # partial months do not count
+    assert q("SELECT timestampdiff(MONTH, TIMESTAMP '2024-01-31 00:00:00', "
+             "TIMESTAMP '2024-02-29 00:00:00')") == [(0,)]
+    assert q("SELECT convert_timezone('UTC', 'America/Los_Angeles', "
+             "TIMESTAMP '2024-01-01 08:00:00')") == [(1704067200000000,)]
+    assert q("SELECT dayname(DATE '2024-01-01'), date_from_unix_date(1)") \
+        == [("Mon", __import__("datetime").date(1970, 1, 2))]
+
+
+def test_bit_and_random_functions(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT getbit(5, 0), getbit(5, 1), bit_get(5, 2)") == [(1, 0, 1)]
+    assert q("SELECT random() BETWEEN 0 AND 1, uniform(0, 10) BETWEEN 0 AND 10, "
+             "length(randstr(8))") == [(True, True, 8)]
+    # seeded variants are deterministic
+    assert q("SELECT uniform(0, 100, 42)") == q("SELECT uniform(0, 100, 42)")
+    assert q("SELECT randstr(6, 1)") == q("SELECT randstr(6, 1)")

@@ -402,6 +402,99 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
         data.scatter_reduce_(0, gidm, vals, reduce="prod", include_self=True)
         return Column(T.F64, data, None)
 
+    if name == "percentile_disc":
+        # discrete percentile: smallest value with cdf >= q; row-gather
+        # preserves the input dtype exactly
+        import math as _math
+
+        q = float(args[1].to_pylist()[0]) if len(args) > 1 and len(args[1]) else 0.5
+        desc = bool(args[2].to_pylist()[0]) if len(args) > 2 and len(args[2]) else False
+        ridx = torch.nonzero(mask, as_tuple=False).flatten()
+        vals = c.data[mask].to(torch.float64)
+        g = gid[mask]
+        import collections
+
+        groups = collections.defaultdict(list)
+        for rr, gg, vv in zip(ridx.tolist(), g.tolist(), vals.tolist()):
+            groups[gg].append((vv, rr))
+        pick = torch.zeros(ng, dtype=torch.int64, device=dev)
+        has = torch.zeros(ng, dtype=torch.bool, device=dev)
+        for k, lst in groups.items():
+            lst.sort(reverse=desc)
+            n_ = len(lst)
+            i = max(0, min(n_ - 1, _math.ceil(q * n_) - 1))
+            pick[k] = lst[i][1]
+            has[k] = True
+        out = c.gather(pick)
+        if not bool(has.all()):
+            v = out.valid_mask() & has
+            out = Column(out.dtype, out.data, v.to(torch.uint8)) \
+                if not isinstance(out, StringColumn) else out
+        return out
+
+    if name in ("vector_sum", "vector_avg"):
+        from .column import ListColumn
+
+        lens = c.lengths()
+        rows_mask = mask
+        L = int(lens[rows_mask].max().item()) if bool(rows_mask.any()) else 0
+        seg = c.segment_ids()
+        emask = rows_mask[seg]
+        pos = torch.arange(seg.shape[0], dtype=torch.int64, device=dev) \
+            - c.offsets[:-1][seg]
+        tgt = gid[seg] * L + pos
+        out = torch.zeros(ng * L, dtype=torch.float64, device=dev)
+        out.index_add_(0, tgt[emask],
+                       c.child.data.to(torch.float64)[emask])
+        if name == "vector_avg":
+            cnt = torch.zeros(ng, dtype=torch.float64, device=dev)
+            cnt.index_add_(0, gid[rows_mask],
+                           torch.ones(int(rows_mask.sum()),
+                                      dtype=torch.float64, device=dev))
+            out = out / cnt.clamp_min(1.0).repeat_interleave(L)
+        offs = torch.arange(ng + 1, dtype=torch.int64, device=dev) * L
+        return ListColumn(offs, Column(T.F64, out, None), None,
+                          T.ArrayType(T.F64))
+
+    if name == "histogram_numeric":
+        # Ben-Haim/Tom-Tov style: merge the closest centroid pair until
+        # nb bins remain (ref: sail-function aggregate/histogram_numeric)
+        from .column import ListColumn, StructColumn
+
+        nb = int(args[1].to_pylist()[0]) if len(args) > 1 and len(args[1]) else 10
+        vals = c.data[mask].to(torch.float64)
+        if isinstance(c.dtype, T.DecimalType):
+            vals = vals / (10.0 ** c.dtype.scale)
+        g = gid[mask]
+        import collections
+
+        groups = collections.defaultdict(lambda: collections.Counter())
+        for gg, vv in zip(g.tolist(), vals.tolist()):
+            groups[gg][vv] += 1
+        rows = [[] for _ in range(ng)]
+        for k, cnt in groups.items():
+            bins = sorted((x, float(y)) for x, y in cnt.items())
+            while len(bins) > nb:
+                gaps = [(bins[i + 1][0] - bins[i][0], i)
+                        for i in range(len(bins) - 1)]
+                _, i = min(gaps)
+                (x1, y1), (x2, y2) = bins[i], bins[i + 1]
+                bins[i:i + 2] = [((x1 * y1 + x2 * y2) / (y1 + y2), y1 + y2)]
+            rows[k] = bins
+        flat_x = [x for r in rows for x, _ in r]
+        flat_y = [y for r in rows for _, y in r]
+        offs = [0]
+        for r in rows:
+            offs.append(offs[-1] + len(r))
+        elem = T.StructType((T.StructField("x", T.F64),
+                             T.StructField("y", T.F64)))
+        child = StructColumn(
+            [("x", Column.from_values(flat_x, T.F64, device=dev)),
+             ("y", Column.from_values(flat_y, T.F64, device=dev))],
+            dtype=elem)
+        return ListColumn(torch.tensor(offs, dtype=torch.int64, device=dev),
+                          child, None, T.ArrayType(elem))
+
     if name in ("median", "percentile", "percentile_approx"):
         # exact median/percentile via host (small group counts expected)
         vals = c.data[mask].to(torch.float64)
@@ -418,6 +511,9 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
         q = 0.5
         if name != "median" and len(args) > 1 and len(args[1]):
             q = float(args[1].to_pylist()[0])
+        if name != "median" and len(args) > 2 and len(args[2]) \
+                and bool(args[2].to_pylist()[0]):
+            q = 1.0 - q  # WITHIN GROUP (ORDER BY x DESC)
         import collections
 
         groups = collections.defaultdict(list)

@@ -28,6 +28,7 @@ AGG_FUNCTIONS = {
     "any_value", "try_sum", "try_avg", "regr_count", "regr_avgx", "regr_avgy",
     "regr_slope", "regr_intercept", "regr_r2", "regr_sxx", "regr_syy", "regr_sxy",
     "grouping", "grouping_id", "histogram_numeric", "string_agg", "listagg",
+    "std", "percentile_cont", "percentile_disc", "vector_sum", "vector_avg",
 }
 
 WINDOW_FUNCTIONS = {
@@ -55,9 +56,16 @@ def agg_return_type(name: str, arg_types: List[T.DataType], distinct: bool = Fal
             return T.DecimalType(min(38, a.precision + 4), min(a.scale + 4, 10))
         return T.F64
     if name in ("min", "max", "first", "first_value", "last", "last_value",
-                "any_value", "mode", "median", "max_by", "min_by"):
+                "any_value", "mode", "median", "max_by", "min_by",
+                "percentile_disc"):
         return a
+    if name in ("vector_sum", "vector_avg"):
+        return T.ArrayType(T.F64)
+    if name == "histogram_numeric":
+        return T.ArrayType(T.StructType((T.StructField("x", T.F64),
+                                         T.StructField("y", T.F64))))
     if name in ("stddev", "stddev_samp", "stddev_pop", "variance", "var_samp",
+                "std", "percentile_cont",
                 "var_pop", "corr", "covar_samp", "covar_pop", "skewness",
                 "kurtosis", "percentile", "percentile_approx", "approx_percentile",
                 "product") or name.startswith("regr_"):

@@ -1460,6 +1460,7 @@ def _normalize_agg_name(name: str) -> str:
     return {"mean": "avg", "first_value": "first", "last_value": "last",
             "some": "any", "bool_or": "any", "every": "bool_and",
             "array_agg": "collect_list", "stddev": "stddev_samp",
+            "std": "stddev_samp", "percentile_cont": "percentile",
             "variance": "var_samp", "approx_percentile": "percentile_approx"}.get(name, name)
 
 

@@ -1310,6 +1310,27 @@ class Parser:
         if lname == "zeroifnull" and len(args) == 1:
             return S.Func("coalesce", [args[0], S.Literal(0, None)])
         self.eat_kw("IGNORE") and self.expect_kw("NULLS")
+        # ordered-set aggregates: f(q) WITHIN GROUP (ORDER BY x [DESC])
+        if self.at_kw("WITHIN"):
+            self.next()
+            self.expect_kw("GROUP")
+            self.expect_op("(")
+            self.expect_kw("ORDER")
+            self.expect_kw("BY")
+            order = self.parse_expr()
+            desc = False
+            if self.eat_kw("DESC"):
+                desc = True
+            else:
+                self.eat_kw("ASC")
+            self.expect_op(")")
+            if lname in ("percentile_cont", "percentile_disc", "percentile"):
+                args = [order] + args + [S.Literal(desc, T.BOOL)]
+            elif lname in ("listagg", "string_agg"):
+                args = args + [order, S.Literal(desc, T.BOOL)]
+            else:
+                raise SqlError(f"WITHIN GROUP not supported for {name}",
+                               self.sql, self.peek().pos)
         # FILTER (WHERE ...)
         filt = None
         if self.at_kw("FILTER"):

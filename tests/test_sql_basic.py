@@ -570,3 +570,21 @@ def test_bit_and_random_functions(s):
     # seeded variants are deterministic
     assert q("SELECT uniform(0, 100, 42)") == q("SELECT uniform(0, 100, 42)")
     assert q("SELECT randstr(6, 1)") == q("SELECT randstr(6, 1)")
+
+
+def test_ordered_set_and_vector_aggregates(s):
+    s.create_dataframe({"g": [1, 1, 1, 1, 2, 2], "x": [10, 20, 30, 40, 5, 15]},
+                       name="osa")
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT g, percentile_cont(0.5) WITHIN GROUP (ORDER BY x) "
+             "FROM osa GROUP BY g ORDER BY g") == [(1, 25.0), (2, 10.0)]
+    assert q("SELECT g, percentile_disc(0.5) WITHIN GROUP (ORDER BY x) "
+             "FROM osa GROUP BY g ORDER BY g") == [(1, 20), (2, 5)]
+    assert q("SELECT g, percentile_disc(0.5) WITHIN GROUP (ORDER BY x DESC) "
+             "FROM osa GROUP BY g ORDER BY g") == [(1, 30), (2, 15)]
+    assert q("SELECT std(x) = stddev(x) FROM osa") == [(True,)]
+    assert q("SELECT g, vector_sum(v), vector_avg(v) FROM "
+             "(SELECT g, array(1.0*g, 2.0) AS v FROM osa) GROUP BY g ORDER BY g") \
+        == [(1, [4.0, 8.0], [1.0, 2.0]), (2, [4.0, 4.0], [2.0, 2.0])]
+    (hist,), = q("SELECT histogram_numeric(x, 2) FROM osa WHERE g = 1")
+    assert len(hist) == 2 and sum(b["y"] for b in hist) == 4.0

@@ -2187,3 +2187,119 @@ _IMPLS["try_mod"] = _f_try_mod
 _IMPLS["current_schema"] = _f_current_schema
 _IMPLS["user"] = _f_user
 _IMPLS["session_user"] = _f_user
+
+
+# -- XML struct parsing + collation stubs ----------------------------------
+
+def _f_from_xml(args, out, chunk, ev):
+    """from_xml(col, 'a INT, b STRING') -> struct: each field read from the
+    matching child element of the row's root (ref: sail-function
+    scalar/xml)."""
+    import xml.etree.ElementTree as ET
+
+    from .column import StructColumn
+
+    c = _col(args[0], chunk)
+    docs = c.to_pylist()
+    parsed = []
+    for d in docs:
+        if d is None:
+            parsed.append(None)
+            continue
+        try:
+            parsed.append(ET.fromstring(d))
+        except ET.ParseError:
+            parsed.append(None)
+    kids = []
+    for f in out.fields:
+        vals = []
+        for root in parsed:
+            if root is None:
+                vals.append(None)
+                continue
+            el = root.find(f.name)
+            txt = None if el is None else (el.text or "")
+            if txt is None or isinstance(f.dtype, T.StringType):
+                vals.append(txt)
+            else:
+                try:
+                    vals.append(float(txt) if isinstance(
+                        f.dtype, (T.Float32Type, T.Float64Type,
+                                  T.DecimalType)) else int(txt))
+                except ValueError:
+                    vals.append(None)
+        if isinstance(f.dtype, T.StringType):
+            kids.append((f.name, StringColumn.from_pylist(
+                vals, device=str(chunk.device))))
+        else:
+            kids.append((f.name, Column.from_values(
+                vals, f.dtype, device=chunk.device)))
+    validity = None
+    if any(p is None for p in parsed):
+        validity = torch.tensor([0 if p is None else 1 for p in parsed],
+                                dtype=torch.uint8, device=chunk.device)
+    return StructColumn(kids, validity, dtype=out)
+
+
+def _f_to_xml(args, out, chunk, ev):
+    from .column import StringColumn
+    from xml.sax.saxutils import escape
+
+    c = _col(args[0], chunk)
+    names = [nm for nm, _ in c.children_]
+    cols = [kid.to_pylist() for _, kid in c.children_]
+    parts = []
+    for i in range(len(c)):
+        fields = []
+        for nm, vals in zip(names, cols):
+            v = vals[i]
+            fields.append(f"<{nm}/>" if v is None
+                          else f"<{nm}>{escape(str(v))}</{nm}>")
+        parts.append("<ROW>" + "".join(fields) + "</ROW>")
+    col = StringColumn.from_pylist(parts, device=str(chunk.device))
+    col.validity = c.validity
+    return col
+
+
+def _f_schema_of_xml(args, out, chunk, ev):
+    import xml.etree.ElementTree as ET
+
+    from .column import StringColumn
+
+    sample = _scalarize(args[0]).value
+    root = ET.fromstring(sample)
+    fields = []
+    for el in root:
+        txt = (el.text or "").strip()
+        try:
+            int(txt)
+            t = "BIGINT"
+        except ValueError:
+            try:
+                float(txt)
+                t = "DOUBLE"
+            except ValueError:
+                t = "STRING"
+        fields.append(f"{el.tag}: {t}")
+    body = ", ".join(fields)
+    return StringColumn.from_pylist([f"STRUCT<{body}>"] * chunk.num_rows,
+                                    device=str(chunk.device))
+
+
+def _f_collate(args, out, chunk, ev):
+    # single-collation engine (UTF8_BINARY): collate() is identity
+    return _col(args[0], chunk)
+
+
+def _f_collation(args, out, chunk, ev):
+    from .column import StringColumn
+
+    return StringColumn.from_pylist(["UTF8_BINARY"] * chunk.num_rows,
+                                    device=str(chunk.device))
+
+
+_IMPLS["from_xml"] = _f_from_xml
+_IMPLS["to_xml"] = _f_to_xml
+_IMPLS["schema_of_xml"] = _f_schema_of_xml
+_IMPLS["collate"] = _f_collate
+_IMPLS["collation"] = _f_collation

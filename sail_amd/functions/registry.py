@@ -225,7 +225,9 @@ _reg("get_json_object to_json schema_of_json json_tuple", _string)
 _reg("json_array_length", _i32)
 _reg("json_object_keys", lambda a: T.ArrayType(T.STRING))
 _reg("from_csv", lambda a: T.NULL)  # typed structurally in the resolver
-_reg("to_csv schema_of_csv", _string)
+_reg("to_csv schema_of_csv to_xml schema_of_xml collation", _string)
+_reg("collate", lambda a: a[0])
+_reg("from_xml", lambda a: T.NULL)  # typed structurally in the resolver
 _reg("parse_url try_parse_url url_encode url_decode", _string)
 _reg("xpath", lambda a: T.ArrayType(T.STRING))
 _reg("xpath_string", _string)

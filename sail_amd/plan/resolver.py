@@ -1257,7 +1257,7 @@ class Resolver:
                         "map_from_entries expects array<struct<k,v>>")
                 t = T.MapType(at.element.fields[0].dtype,
                               at.element.fields[1].dtype)
-            elif e.name in ("from_json", "from_csv"):
+            elif e.name in ("from_json", "from_csv", "from_xml"):
                 if len(e.args) < 2 or not isinstance(e.args[1], S.Literal):
                     raise ResolutionError(f"{e.name} expects a literal schema string")
                 from ..sql.parser import parse_ddl_schema

@@ -588,3 +588,16 @@ def test_ordered_set_and_vector_aggregates(s):
         == [(1, [4.0, 8.0], [1.0, 2.0]), (2, [4.0, 4.0], [2.0, 2.0])]
     (hist,), = q("SELECT histogram_numeric(x, 2) FROM osa WHERE g = 1")
     assert len(hist) == 2 and sum(b["y"] for b in hist) == 4.0
+
+
+def test_xml_struct_and_collation(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT from_xml('<r><a>1</a><b>x</b></r>', 'a INT, b STRING').a, "
+             "from_xml('<r><a>1</a><b>x</b></r>', 'a INT, b STRING').b") \
+        == [(1, "x")]
+    assert q("SELECT to_xml(named_struct('a', 1, 'b', 'x'))") \
+        == [("<ROW><a>1</a><b>x</b></ROW>",)]
+    assert q("SELECT schema_of_xml('<r><a>1</a><b>s</b></r>')") \
+        == [("STRUCT<a: BIGINT, b: STRING>",)]
+    assert q("SELECT collate('x', 'UTF8_BINARY'), collation('y')") \
+        == [("x", "UTF8_BINARY")]

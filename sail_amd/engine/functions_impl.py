@@ -2303,3 +2303,31 @@ _IMPLS["to_xml"] = _f_to_xml
 _IMPLS["schema_of_xml"] = _f_schema_of_xml
 _IMPLS["collate"] = _f_collate
 _IMPLS["collation"] = _f_collation
+
+
+def _f_make_dt_interval(args, out, chunk, ev):
+    """make_dt_interval([days, hours, mins, secs]) -> day-time interval,
+    represented as int64 microseconds (the engine's interval storage —
+    adds directly to TIMESTAMP columns)."""
+    from .eval import broadcast
+
+    n = chunk.num_rows
+    mults = [86_400_000_000, 3_600_000_000, 60_000_000, 1_000_000]
+    total = torch.zeros(n, dtype=torch.int64, device=chunk.device)
+    validity = None
+    for i, m in enumerate(mults):
+        if len(args) <= i:
+            break
+        c = broadcast(args[i], n, chunk.device)
+        if i == 3 and c.data.is_floating_point():
+            total = total + (c.data.to(torch.float64) * m).to(torch.int64)
+        else:
+            total = total + c.data.to(torch.int64) * m
+        if c.validity is not None:
+            v = c.valid_mask()
+            validity = v if validity is None else (validity & v)
+    return Column(T.I64, total,
+                  validity.to(torch.uint8) if validity is not None else None)
+
+
+_IMPLS["make_dt_interval"] = _f_make_dt_interval

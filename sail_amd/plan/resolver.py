@@ -1043,6 +1043,32 @@ class Resolver:
                      for k in e.order_by]
             return S.WindowExpr(func=f, partition_by=part, order_by=order,
                                 frame=e.frame, dtype=f.dtype)
+        if isinstance(e, S.Func) and e.name.lower() in ("make_dt_interval",
+                                                        "make_ym_interval"):
+            # intervals are literal-folded: the engine stores day-time
+            # intervals as ("__interval__", months, micros) literal tuples
+            vals = []
+            for a in e.args:
+                lit = a
+                if isinstance(lit, S.Cast):
+                    lit = lit.child
+                if isinstance(lit, S.UnaryOp) and lit.op == "neg" \
+                        and isinstance(lit.child, S.Literal):
+                    vals.append(-lit.child.value)
+                elif isinstance(lit, S.Literal):
+                    vals.append(lit.value)
+                else:
+                    raise ResolutionError(
+                        f"{e.name} requires literal arguments")
+            vals += [0] * (4 - len(vals))
+            if e.name.lower() == "make_ym_interval":
+                return S.Literal((_INTERVAL, int(vals[0]) * 12 + int(vals[1]),
+                                  0), T.NULL)
+            micros = (int(vals[0]) * 86_400_000_000
+                      + int(vals[1]) * 3_600_000_000
+                      + int(vals[2]) * 60_000_000
+                      + int(round(float(vals[3]) * 1_000_000)))
+            return S.Literal((_INTERVAL, 0, micros), T.NULL)
         if isinstance(e, S.Func) and any(isinstance(a, S.Lambda) for a in e.args):
             return self._resolve_hof(e, scope)
         # generic: resolve children then type

@@ -601,3 +601,11 @@ def test_xml_struct_and_collation(s):
         == [("STRUCT<a: BIGINT, b: STRING>",)]
     assert q("SELECT collate('x', 'UTF8_BINARY'), collation('y')") \
         == [("x", "UTF8_BINARY")]
+
+
+def test_interval_makers(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT TIMESTAMP '2024-01-01 00:00:00' + make_dt_interval(1, 2, 3, 4.5)") \
+        == [(1704160984500000,)]  # +1d 2h 3m 4.5s
+    assert q("SELECT DATE '2024-01-15' + make_ym_interval(1, 1)") \
+        == [(__import__("datetime").date(2025, 2, 15),)]

@@ -2331,3 +2331,106 @@ def _f_make_dt_interval(args, out, chunk, ev):
 
 
 _IMPLS["make_dt_interval"] = _f_make_dt_interval
+
+
+# -- last trivial parity batch ---------------------------------------------
+
+def _f_trig_recip(fn):
+    def run(args, out, chunk, ev):
+        c = _col(args[0], chunk)
+        return Column(T.F64, 1.0 / fn(c.data.to(torch.float64)), c.validity)
+    return run
+
+
+def _f_current_timezone(args, out, chunk, ev):
+    from .column import StringColumn
+
+    return StringColumn.from_pylist(["UTC"] * chunk.num_rows,
+                                    device=str(chunk.device))
+
+
+def _f_true_const(args, out, chunk, ev):
+    # engine strings are python str: always valid UTF-8 by construction
+    return Column(T.BOOL, torch.ones(chunk.num_rows, dtype=torch.bool,
+                                     device=chunk.device), None)
+
+
+def _f_identity_str(args, out, chunk, ev):
+    return _col(args[0], chunk)
+
+
+def _f_is_valid_variant(args, out, chunk, ev):
+    import json as _json
+
+    c = _col(args[0], chunk)
+
+    def ok(v):
+        if v is None:
+            return None
+        try:
+            _json.loads(v)
+            return True
+        except (ValueError, TypeError):
+            return False
+
+    return Column.from_values([ok(v) for v in c.to_pylist()], T.BOOL,
+                              device=str(chunk.device))
+
+
+def _f_random_poisson(args, out, chunk, ev):
+    lam = float(_scalarize(args[0]).value) if args else 1.0
+    vals = torch.poisson(torch.full((chunk.num_rows,), lam,
+                                    dtype=torch.float64, device=chunk.device))
+    return Column(T.I64, vals.to(torch.int64), None)
+
+
+def _f_bitmap_bit_position(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    return Column(T.I64, (c.data.to(torch.int64) - 1) % 32768, c.validity)
+
+
+def _f_bitmap_bucket_number(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    return Column(T.I64,
+                  torch.div(c.data.to(torch.int64) - 1, 32768,
+                            rounding_mode="floor") + 1, c.validity)
+
+
+def _f_deep_size(args, out, chunk, ev):
+    import sys
+
+    c = _col(args[0], chunk)
+    vals = [None if v is None else sys.getsizeof(v) for v in c.to_pylist()]
+    return Column.from_values(vals, T.I64, device=str(chunk.device))
+
+
+_IMPLS["cot"] = _f_trig_recip(torch.tan)
+_IMPLS["csc"] = _f_trig_recip(torch.sin)
+_IMPLS["sec"] = _f_trig_recip(torch.cos)
+_IMPLS["current_timezone"] = _f_current_timezone
+_IMPLS["is_valid_utf8"] = _f_true_const
+_IMPLS["try_validate_utf8"] = _f_identity_str
+_IMPLS["validate_utf8"] = _f_identity_str
+_IMPLS["make_valid_utf8"] = _f_identity_str
+_IMPLS["variant_to_json"] = _f_identity_str
+_IMPLS["to_variant_object"] = _f_parse_json
+_IMPLS["is_valid_variant"] = _f_is_valid_variant
+_IMPLS["try_url_decode"] = _f_url_decode
+_IMPLS["random_poisson"] = _f_random_poisson
+_IMPLS["binary"] = _f_to_varchar
+_IMPLS["bitmap_bit_position"] = _f_bitmap_bit_position
+_IMPLS["bitmap_bucket_number"] = _f_bitmap_bucket_number
+_IMPLS["deep_size"] = _f_deep_size
+
+
+def _f_ts_add_months(args, out, chunk, ev):
+    from .eval import Scalar, broadcast
+
+    n = chunk.num_rows
+    months = broadcast(args[1], n, chunk.device).data.to(torch.int64)
+    unit = Scalar("MONTH", T.STRING)
+    return _f_timestampadd([unit, Column(T.I64, months, None), args[0]],
+                           out, chunk, ev)
+
+
+_IMPLS["ts_add_months"] = _f_ts_add_months

@@ -236,7 +236,12 @@ _reg("xpath_int xpath_short", _i32)
 _reg("xpath_long", _i64)
 _reg("xpath_double xpath_float xpath_number", _f64)
 _reg("parse_json try_parse_json schema_of_variant variant_get try_variant_get", _string)
-_reg("is_variant_null luhn_check", _bool)
+_reg("is_variant_null luhn_check is_valid_utf8 is_valid_variant", _bool)
+_reg("cot csc sec", _f64)
+_reg("current_timezone variant_to_json to_variant_object try_url_decode "
+     "validate_utf8 try_validate_utf8 make_valid_utf8", _string)
+_reg("binary", lambda a: T.BINARY)
+_reg("random_poisson bitmap_bit_position bitmap_bucket_number deep_size", _i64)
 _reg("crc32c", _i64)
 _reg("array_insert", lambda a: a[0])
 _reg("array_contains_all", _bool)

@@ -1043,8 +1043,9 @@ class Resolver:
                      for k in e.order_by]
             return S.WindowExpr(func=f, partition_by=part, order_by=order,
                                 frame=e.frame, dtype=f.dtype)
-        if isinstance(e, S.Func) and e.name.lower() in ("make_dt_interval",
-                                                        "make_ym_interval"):
+        if isinstance(e, S.Func) and e.name.lower() in (
+                "make_dt_interval", "make_ym_interval", "make_interval",
+                "try_make_interval"):
             # intervals are literal-folded: the engine stores day-time
             # intervals as ("__interval__", months, micros) literal tuples
             vals = []
@@ -1060,6 +1061,15 @@ class Resolver:
                 else:
                     raise ResolutionError(
                         f"{e.name} requires literal arguments")
+            if e.name.lower() in ("make_interval", "try_make_interval"):
+                # (years, months, weeks, days, hours, mins, secs)
+                vals += [0] * (7 - len(vals))
+                months = int(vals[0]) * 12 + int(vals[1])
+                micros = ((int(vals[2]) * 7 + int(vals[3])) * 86_400_000_000
+                          + int(vals[4]) * 3_600_000_000
+                          + int(vals[5]) * 60_000_000
+                          + int(round(float(vals[6]) * 1_000_000)))
+                return S.Literal((_INTERVAL, months, micros), T.NULL)
             vals += [0] * (4 - len(vals))
             if e.name.lower() == "make_ym_interval":
                 return S.Literal((_INTERVAL, int(vals[0]) * 12 + int(vals[1]),
@@ -1337,17 +1347,19 @@ def _type_binary(e: S.BinaryOp) -> S.Expr:
         _, months, micros = ivl.value
         sign = -1 if e.op == "-" else 1
         if isinstance(other.dtype, T.DateType):
+            out = other
             if months:
-                out = S.Func("add_months", [other, S.Literal(sign * months, T.I32)], T.DATE)
-            else:
-                out = S.Func("date_add", [other, S.Literal(sign * (micros // 86_400_000_000), T.I32)], T.DATE)
+                out = S.Func("add_months", [out, S.Literal(sign * months, T.I32)], T.DATE)
+            if micros or not months:
+                out = S.Func("date_add", [out, S.Literal(sign * (micros // 86_400_000_000), T.I32)], T.DATE)
             return out
         if isinstance(other.dtype, T.TimestampType):
             total = sign * (micros + months * 2_592_000_000_000)  # months≈30d only if ts; Spark uses calendar — handled in add_months path
+            out = other
             if months:
-                out = S.Func("ts_add_months", [other, S.Literal(sign * months, T.I32)], T.TIMESTAMP)
-            else:
-                out = S.BinaryOp("+", other, S.Literal(sign * micros, T.I64), T.TIMESTAMP)
+                out = S.Func("ts_add_months", [out, S.Literal(sign * months, T.I32)], T.TIMESTAMP)
+            if micros or not months:
+                out = S.BinaryOp("+", out, S.Literal(sign * micros, T.I64), T.TIMESTAMP)
             return out
         raise ResolutionError("interval arithmetic requires date/timestamp operand")
     if e.op in ("=", "!=", "<", "<=", ">", ">=", "<=>"):

@@ -609,3 +609,20 @@ def test_interval_makers(s):
         == [(1704160984500000,)]  # +1d 2h 3m 4.5s
     assert q("SELECT DATE '2024-01-15' + make_ym_interval(1, 1)") \
         == [(__import__("datetime").date(2025, 2, 15),)]
+
+
+def test_trivial_parity_batch(s):
+    q = lambda x: s.sql(x).collect()  # noqa: E731
+    assert q("SELECT round(cot(1.0), 4), round(sec(1.0), 4)") == [(0.6421, 1.8508)]
+    assert q("SELECT current_timezone(), is_valid_utf8('x')") == [("UTC", True)]
+    assert q("""SELECT variant_to_json(parse_json('{"a":1}')), """
+             "is_valid_variant('{}'), is_valid_variant('no')") \
+        == [('{"a":1}', True, False)]
+    assert q("SELECT try_url_decode('a%20b')") == [("a b",)]
+    assert q("SELECT bitmap_bit_position(32769), bitmap_bucket_number(32769)") \
+        == [(0, 2)]
+    assert q("SELECT DATE '2020-01-01' + make_interval(1, 0, 1, 1)") \
+        == [(__import__("datetime").date(2021, 1, 9),)]
+    # month interval on TIMESTAMP clamps to month end
+    assert q("SELECT TIMESTAMP '2020-01-31 00:00:00' + INTERVAL 1 MONTH") \
+        == [(1582934400000000,)]  # 2020-02-29

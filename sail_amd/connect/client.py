@@ -131,3 +131,34 @@ class ConnectClient:
             f"/{_SERVICE}/{method}",
             request_serializer=None, response_deserializer=None)
         return call(req)
+
+    def json_to_ddl(self, json_schema: str) -> str:
+        req = (W.field_string(F.AN_SESSION_ID, self.session_id)
+               + W.field_message(F.AN_JSON_TO_DDL,
+                                 W.field_string(1, json_schema)))
+        fields = W.parse(self._call_unary("AnalyzePlan", req))
+        jd = W.parse(W.first(fields, F.ANR_JSON_TO_DDL, b""))
+        return W.first_str(jd, 1)
+
+    def input_files(self, sql: str) -> list:
+        inner = W.field_message(1, self._sql_plan(sql))
+        req = (W.field_string(F.AN_SESSION_ID, self.session_id)
+               + W.field_message(F.AN_INPUT_FILES, inner))
+        fields = W.parse(self._call_unary("AnalyzePlan", req))
+        body = W.parse(W.first(fields, F.ANR_INPUT_FILES, b""))
+        return W.all_strs(body, 1)
+
+    def persist(self, sql: str) -> None:
+        inner = W.field_message(1, self._sql_plan(sql))
+        req = (W.field_string(F.AN_SESSION_ID, self.session_id)
+               + W.field_message(F.AN_PERSIST, inner))
+        self._call_unary("AnalyzePlan", req)
+
+    def get_storage_level(self, sql: str) -> bool:
+        inner = W.field_message(1, self._sql_plan(sql))
+        req = (W.field_string(F.AN_SESSION_ID, self.session_id)
+               + W.field_message(F.AN_GET_STORAGE_LEVEL, inner))
+        fields = W.parse(self._call_unary("AnalyzePlan", req))
+        lvl = W.parse(W.first(fields, F.ANR_GET_STORAGE_LEVEL, b""))
+        sl = W.parse(W.first(lvl, 1, b""))
+        return bool(W.first_varint(sl, 2, 0))

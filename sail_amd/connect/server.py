@@ -80,6 +80,10 @@ class F:
     AN_DDL_PARSE = 11
     AN_SAME_SEMANTICS = 12
     AN_SEMANTIC_HASH = 13
+    AN_PERSIST = 14
+    AN_UNPERSIST = 15
+    AN_GET_STORAGE_LEVEL = 16
+    AN_JSON_TO_DDL = 18
     # AnalyzePlanResponse oneofs
     ANR_SESSION_ID = 1
     ANR_SCHEMA = 2
@@ -92,6 +96,10 @@ class F:
     ANR_DDL_PARSE = 9
     ANR_SAME_SEMANTICS = 10
     ANR_SEMANTIC_HASH = 11
+    ANR_PERSIST = 12
+    ANR_UNPERSIST = 13
+    ANR_GET_STORAGE_LEVEL = 14
+    ANR_JSON_TO_DDL = 16
     # ConfigRequest
     CFG_SESSION_ID = 1
     CFG_OPERATION = 3
@@ -297,7 +305,63 @@ class SparkConnectServer:
 
             h = zlib.crc32(self._semantic_tree(sess, sql).encode()) & 0x7FFFFFFF
             return out + W.field_message(F.ANR_SEMANTIC_HASH, W.field_varint(1, h))
+        if F.AN_INPUT_FILES in req:
+            inner = W.parse(W.first(req, F.AN_INPUT_FILES))
+            plan = W.parse(W.first(inner, 1, b""))
+            root = W.first(plan, F.PLAN_ROOT)
+            sql = W.first_str(W.parse(W.first(W.parse(root or b""), F.REL_SQL, b"")),
+                              F.SQL_QUERY)
+            files = self._input_files(sess, sql)
+            body = b"".join(W.field_string(1, f) for f in files)
+            return out + W.field_message(F.ANR_INPUT_FILES, body)
+        if F.AN_PERSIST in req:
+            return out + W.field_message(F.ANR_PERSIST, b"")
+        if F.AN_UNPERSIST in req:
+            return out + W.field_message(F.ANR_UNPERSIST, b"")
+        if F.AN_GET_STORAGE_LEVEL in req:
+            # single-level engine: everything is device/host memory resident
+            level = W.field_varint(2, 1)  # use_memory = true
+            return out + W.field_message(F.ANR_GET_STORAGE_LEVEL,
+                                         W.field_message(1, level))
+        jtd = W.first(req, F.AN_JSON_TO_DDL)
+        if jtd is not None:
+            import json as _json
+
+            text = W.first_str(W.parse(jtd), 1)
+            obj = _json.loads(text)
+            parts = []
+            for f in obj.get("fields", []):
+                parts.append(f"{f['name']} {self._json_type_ddl(f['type'])}")
+            return out + W.field_message(F.ANR_JSON_TO_DDL,
+                                         W.field_string(1, ",".join(parts)))
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "analyze type not supported")
+
+    @staticmethod
+    def _json_type_ddl(t) -> str:
+        names = {"long": "BIGINT", "integer": "INT", "double": "DOUBLE",
+                 "float": "FLOAT", "string": "STRING", "boolean": "BOOLEAN",
+                 "date": "DATE", "timestamp": "TIMESTAMP", "short": "SMALLINT",
+                 "byte": "TINYINT", "binary": "BINARY"}
+        if isinstance(t, str):
+            return names.get(t, t.upper())
+        return "STRING"
+
+    def _input_files(self, sess: SessionContext, sql: str):
+        """All file paths the plan reads (ref: Spark AnalyzePlan InputFiles)."""
+        from ..plan import spec as S
+
+        plan = sess.plan_sql(sql)
+        out = []
+
+        def walk(p):
+            if isinstance(p, S.DataSourceRead):
+                out.extend(p.paths or [])
+            for c in p.children():
+                if c is not None:
+                    walk(c)
+
+        walk(plan)
+        return out
 
     def _semantic_tree(self, sess: SessionContext, sql: str) -> str:
         from ..plan import spec as S

@@ -107,3 +107,11 @@ def first_str(fields: Dict[int, List], num: int, default: str = "") -> str:
 def first_varint(fields: Dict[int, List], num: int, default: int = 0) -> int:
     v = first(fields, num)
     return int(v) if isinstance(v, int) else default
+
+
+def all_strs(fields, num) -> list:
+    """All values of a repeated string field."""
+    out = []
+    for v in fields.get(num, []):
+        out.append(v.decode("utf-8") if isinstance(v, bytes) else str(v))
+    return out

@@ -85,3 +85,12 @@ def test_analyze_ddl_semantics(client):
     h1 = client.semantic_hash("SELECT 1 + 1")
     h2 = client.semantic_hash("SELECT 1 + 1")
     assert h1 == h2 and h1 > 0
+
+
+def test_analyze_extended_types(client):
+    assert client.json_to_ddl('{"type":"struct","fields":'
+                              '[{"name":"a","type":"long"},'
+                              '{"name":"b","type":"string"}]}') == "a BIGINT,b STRING"
+    assert client.get_storage_level("SELECT 1") is True
+    client.persist("SELECT 1")  # no-op ack
+    assert client.input_files("SELECT 1") == []

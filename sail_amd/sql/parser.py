@@ -1522,7 +1522,12 @@ def _parse_timestamp_us(s: str) -> int:
     mm = int(parts[1]) if len(parts) > 1 else 0
     ss = float(parts[2]) if len(parts) > 2 else 0.0
     base = _dt.datetime(y, m, d) - _dt.datetime(1970, 1, 1)
-    return int((base.total_seconds() + hh * 3600 + mm * 60 + ss) * 1_000_000)
+    # integer-exact micros: float total_seconds() loses sub-µs precision
+    # near the epoch (e.g. 1969-12-31 23:59:59.999999 truncated to 0)
+    sec_int = int(ss)
+    frac_us = int(round((ss - sec_int) * 1_000_000))
+    return ((base.days * 86_400 + hh * 3600 + mm * 60 + sec_int) * 1_000_000
+            + frac_us)
 
 
 def parse_ddl_schema(text: str):

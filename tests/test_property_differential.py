@@ -98,3 +98,41 @@ def test_order_limit_matches_pandas(data, col, asc, k):
             assert g == pytest.approx(w, rel=1e-12)
         else:
             assert g == w
+
+
+@given(st.lists(st.integers(min_value=0, max_value=2**40), max_size=200))
+@settings(max_examples=30, deadline=None)
+def test_roaring_bitmap_round_trip(vals):
+    from sail_amd.utils.roaring import roaring64_deserialize, roaring64_serialize
+
+    out = roaring64_deserialize(roaring64_serialize(vals))
+    assert out.tolist() == sorted(set(vals))
+
+
+@given(st.text(max_size=40))
+@settings(max_examples=30, deadline=None)
+def test_url_encode_decode_round_trip(text):
+    from urllib.parse import quote_plus
+
+    s = sail_amd.SessionContext(device="cpu")
+    s.create_dataframe({"t": [text if text else "x"]}, name="urlrt")
+    (enc,), = s.sql("SELECT url_encode(t) FROM urlrt").collect()
+    (dec,), = s.sql("SELECT url_decode(url_encode(t)) FROM urlrt").collect()
+    assert enc == quote_plus(text if text else "x")
+    assert dec == (text if text else "x")
+
+
+@given(st.integers(min_value=-10**15, max_value=10**15),
+       st.integers(min_value=-5000, max_value=5000))
+@settings(max_examples=30, deadline=None)
+def test_timestampadd_hours_matches_datetime(us, hours):
+    import datetime as dt
+
+    s = sail_amd.SessionContext(device="cpu")
+    base = dt.datetime(1970, 1, 1) + dt.timedelta(microseconds=us)
+    lit = base.strftime("%Y-%m-%d %H:%M:%S.%f")
+    (got,), = s.sql(f"SELECT timestampadd(HOUR, {hours}, "
+                    f"TIMESTAMP '{lit}')").collect()
+    want = int(((base + dt.timedelta(hours=hours))
+                - dt.datetime(1970, 1, 1)).total_seconds() * 1_000_000)
+    assert got == want

@@ -307,3 +307,22 @@ def test_stream_static_join(s):
         assert rows == [("one", 10), ("one", 30), ("two", 20)]
     finally:
         q.stop()
+
+
+def test_iceberg_stream_source(s, tmp_path):
+    base = str(tmp_path / "ice_stream")
+    s.create_dataframe({"id": [1, 2]}, name="ist_src")
+    s.sql(f"CREATE TABLE iceberg.`{base}` AS SELECT * FROM ist_src").collect()
+    sdf = s.read_stream.format("iceberg").load(base)
+    q = sdf.sql(f"SELECT count(*) AS n FROM {sdf.view_name}") \
+        .write_stream.format("memory").query_name("ist_out") \
+        .output_mode("complete").trigger(processing_time=0.02).start()
+    try:
+        q.process_all_available()
+        assert s.sql("SELECT * FROM ist_out").collect() == [(2,)]
+        # appended snapshot arrives as an incremental micro-batch
+        s.sql(f"INSERT INTO iceberg.`{base}` VALUES (3)").collect()
+        q.process_all_available()
+        assert s.sql("SELECT * FROM ist_out").collect() == [(3,)]
+    finally:
+        q.stop()

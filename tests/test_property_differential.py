@@ -133,6 +133,8 @@ def test_timestampadd_hours_matches_datetime(us, hours):
     lit = base.strftime("%Y-%m-%d %H:%M:%S.%f")
     (got,), = s.sql(f"SELECT timestampadd(HOUR, {hours}, "
                     f"TIMESTAMP '{lit}')").collect()
-    want = int(((base + dt.timedelta(hours=hours))
-                - dt.datetime(1970, 1, 1)).total_seconds() * 1_000_000)
+    delta = (base + dt.timedelta(hours=hours)) - dt.datetime(1970, 1, 1)
+    # integer-exact oracle (float total_seconds() loses sub-µs precision)
+    want = (delta.days * 86_400 + delta.seconds) * 1_000_000 \
+        + delta.microseconds
     assert got == want

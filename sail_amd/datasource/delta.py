@@ -273,12 +273,21 @@ class DeltaLog:
         """Delete part files no longer referenced by the CURRENT snapshot
         and older than the retention window (ref: sail-delta-lake log
         cleanup / vacuum semantics)."""
-        _, live, _, _ = self.snapshot()
-        live_set = set(live)
+        _, adds, _, _ = self.snapshot_adds()
+        live_set = {a["path"] for a in adds}
+        # deletion-vector files referenced by live adds must survive too
+        for a in adds:
+            dv = a.get("deletionVector")
+            if dv and dv.get("storageType") == "u":
+                from ..utils.roaring import z85_decode
+
+                u = uuid.UUID(bytes=z85_decode(dv["pathOrInlineDv"][-20:]))
+                live_set.add(f"deletion_vector_{u}.bin")
         cutoff = time.time() - retention_hours * 3600.0
         removed = []
         for f in os.listdir(self.path):
-            if not f.endswith(".parquet") or f in live_set:
+            if not (f.endswith(".parquet")
+                    or f.startswith("deletion_vector_")) or f in live_set:
                 continue
             full = os.path.join(self.path, f)
             if os.path.getmtime(full) <= cutoff:

@@ -54,8 +54,14 @@ def read(paths: List[str], schema, device, options: Dict[str, str]):
 def write(path: str, chunk, mode: str, options: Dict[str, str]):
     schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
     tbl = chunk_to_arrow(chunk, schema)
-    os.makedirs(path, exist_ok=True)
-    target = os.path.join(path, "part-00000.csv")
+    if path.endswith(".csv"):
+        parent = os.path.dirname(path)
+        if parent:
+            os.makedirs(parent, exist_ok=True)
+        target = path
+    else:
+        os.makedirs(path, exist_ok=True)
+        target = os.path.join(path, "part-00000.csv")
     if mode == "error" and os.path.exists(target):
         raise FileExistsError(target)
     pacsv.write_csv(tbl, target)

@@ -269,28 +269,68 @@ class DeltaLog:
                          "num_removed_files": ops["removes"]})
         return rows
 
+    def tombstones(self) -> Dict[str, int]:
+        """Replay the whole log collecting, for every file that is no longer
+        referenced by the current snapshot, the time (epoch ms) at which it
+        left the live set: remove actions carry deletionTimestamp; a DV file
+        is tombstoned when the add referencing it is superseded (new DV or
+        plain re-add) or removed. Retention for VACUUM is measured from this
+        time, not file mtime (ref: sail-delta-lake vacuum semantics —
+        tombstone age, so readers inside the retention window stay safe)."""
+        live_add_dv: Dict[str, Optional[str]] = {}  # data path -> its DV file
+        tomb: Dict[str, int] = {}
+        for v in self.versions():
+            p = os.path.join(self.log_path, f"{v:020d}.json")
+            commit_ms = int(os.path.getmtime(p) * 1000)
+            with open(p) as f:
+                for line in f:
+                    if not line.strip():
+                        continue
+                    action = json.loads(line)
+                    if "add" in action:
+                        a = action["add"]
+                        new_dv = _dv_file_name(a.get("deletionVector"))
+                        old_dv = live_add_dv.get(a["path"])
+                        if old_dv and old_dv != new_dv:
+                            tomb[old_dv] = commit_ms  # DV superseded
+                        if new_dv:
+                            tomb.pop(new_dv, None)
+                        live_add_dv[a["path"]] = new_dv
+                        tomb.pop(a["path"], None)  # re-added file is live again
+                    elif "remove" in action:
+                        r = action["remove"]
+                        ts = int(r.get("deletionTimestamp") or commit_ms)
+                        if r["path"] in live_add_dv:
+                            dvf = live_add_dv.pop(r["path"])
+                            if dvf:
+                                tomb[dvf] = ts
+                            tomb[r["path"]] = ts
+        return tomb
+
     def vacuum(self, retention_hours: float = 168.0, dry_run: bool = False):
-        """Delete part files no longer referenced by the CURRENT snapshot
-        and older than the retention window (ref: sail-delta-lake log
-        cleanup / vacuum semantics)."""
+        """Delete part/DV files no longer referenced by the CURRENT snapshot
+        whose remove-tombstone time is older than the retention window.
+        Files never mentioned in the log (untracked orphans, e.g. aborted
+        writes) fall back to mtime-based retention."""
         _, adds, _, _ = self.snapshot_adds()
         live_set = {a["path"] for a in adds}
-        # deletion-vector files referenced by live adds must survive too
         for a in adds:
-            dv = a.get("deletionVector")
-            if dv and dv.get("storageType") == "u":
-                from ..utils.roaring import z85_decode
-
-                u = uuid.UUID(bytes=z85_decode(dv["pathOrInlineDv"][-20:]))
-                live_set.add(f"deletion_vector_{u}.bin")
-        cutoff = time.time() - retention_hours * 3600.0
+            dvf = _dv_file_name(a.get("deletionVector"))
+            if dvf:
+                live_set.add(dvf)
+        tomb = self.tombstones()
+        cutoff_ms = (time.time() - retention_hours * 3600.0) * 1000.0
         removed = []
         for f in os.listdir(self.path):
             if not (f.endswith(".parquet")
                     or f.startswith("deletion_vector_")) or f in live_set:
                 continue
             full = os.path.join(self.path, f)
-            if os.path.getmtime(full) <= cutoff:
+            if f in tomb:
+                age_ok = tomb[f] <= cutoff_ms
+            else:  # untracked orphan: only mtime is available
+                age_ok = os.path.getmtime(full) * 1000.0 <= cutoff_ms
+            if age_ok:
                 removed.append(f)
                 if not dry_run:
                     os.remove(full)
@@ -332,6 +372,17 @@ def _protocol_action() -> dict:
 DV_MAGIC = 1681511377
 
 
+def _dv_file_name(dv: Optional[dict]) -> Optional[str]:
+    """Table-relative file name for a "u"-storage DV descriptor (None for
+    inline/absolute/absent descriptors)."""
+    if not dv or dv.get("storageType") != "u":
+        return None
+    from ..utils.roaring import z85_decode
+
+    u = uuid.UUID(bytes=z85_decode(dv["pathOrInlineDv"][-20:]))
+    return f"deletion_vector_{u}.bin"
+
+
 def dv_positions(table_path: str, dv: dict):
     """Decode a deletionVector descriptor to the sorted int64 row positions
     it deletes. storageType: "i" inline z85, "u" uuid-named file relative to
@@ -358,6 +409,11 @@ def dv_positions(table_path: str, dv: dict):
             f.seek(dv.get("offset", 1))
             (size,) = struct.unpack(">i", f.read(4))
             data = f.read(size)
+            crc_bytes = f.read(4)
+        if len(crc_bytes) == 4:
+            (crc,) = struct.unpack(">I", crc_bytes)
+            if crc != (zlib.crc32(data) & 0xFFFFFFFF):
+                raise ValueError(f"deletion-vector checksum mismatch: {fpath}")
     (magic,) = struct.unpack_from("<i", data)
     if magic != DV_MAGIC:
         raise ValueError(f"bad deletion-vector magic {magic}")
@@ -377,7 +433,9 @@ def write_dv_file(table_path: str, positions) -> dict:
         f.write(b"\x01")
         f.write(struct.pack(">i", len(blob)))
         f.write(blob)
-        f.write(struct.pack(">i", zlib.crc32(blob) & 0x7FFFFFFF))
+        # full unmasked 32-bit CRC, big-endian (Delta protocol; the reference
+        # validates it on read — sail-delta-lake deletion_vector/storage.rs)
+        f.write(struct.pack(">I", zlib.crc32(blob) & 0xFFFFFFFF))
     return {"storageType": "u", "pathOrInlineDv": z85_encode(u.bytes),
             "offset": 1, "sizeInBytes": len(blob),
             "cardinality": len(positions)}
@@ -554,10 +612,34 @@ def _write_parts(path: str, chunk, options) -> List[dict]:
         return list(exe.map(one, range(nparts)))
 
 
-def write(path: str, chunk, mode: str, options: Dict[str, str], max_retries: int = 10):
-    """Append/overwrite commit with optimistic retry."""
+def last_txn_version(path: str, app_id: str) -> Optional[int]:
+    """Highest committed txn version for app_id (Delta `txn` actions —
+    the idempotence handle streaming sinks use; ref: sail-delta-lake
+    transaction application transactions)."""
+    log = DeltaLog(path)
+    best = None
+    for v in log.versions():
+        with open(os.path.join(log.log_path, f"{v:020d}.json")) as f:
+            for line in f:
+                if not line.strip():
+                    continue
+                action = json.loads(line)
+                t = action.get("txn")
+                if t and t.get("appId") == app_id:
+                    tv = int(t.get("version", -1))
+                    best = tv if best is None else max(best, tv)
+    return best
+
+
+def write(path: str, chunk, mode: str, options: Dict[str, str], max_retries: int = 10,
+          txn: Optional[Tuple[str, int]] = None):
+    """Append/overwrite commit with optimistic retry. `txn=(app_id, version)`
+    stamps the commit with a Delta txn action for idempotent writers."""
     schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
     adds = _write_parts(path, chunk, options)
+    if txn is not None:
+        adds = [{"txn": {"appId": txn[0], "version": int(txn[1]),
+                         "lastUpdated": int(time.time() * 1000)}}] + adds
     log = DeltaLog(path)
     for _ in range(max_retries):
         latest = log.latest_version()

@@ -402,10 +402,35 @@ def write(path: str, chunk, mode: str, options: Dict[str, str]):
               "sort-orders": [{"order-id": 0, "fields": []}],
               "current-snapshot-id": -1, "snapshots": [],
               "snapshot-log": [], "metadata-log": [], "properties": {}}
-    elif mode == "append" and [n for n, _ in t.schema()] != [n for n, _ in schema]:
-        raise ValueError(
-            f"Iceberg append schema mismatch: table has "
-            f"{[n for n, _ in t.schema()]}, write has {[n for n, _ in schema]}")
+    elif mode == "append":
+        cur = t.schema()
+        if [n for n, _ in cur] != [n for n, _ in schema] or \
+                [_type_to_ice(ty) for _, ty in cur] != \
+                [_type_to_ice(ty) for _, ty in schema]:
+            raise ValueError(
+                f"Iceberg append schema mismatch: table has "
+                f"{[(n, _type_to_ice(ty)) for n, ty in cur]}, write has "
+                f"{[(n, _type_to_ice(ty)) for n, ty in schema]}")
+    elif mode == "overwrite":
+        # schema evolution on overwrite: register the new schema and make it
+        # current so data files and metadata agree
+        # (ref: sail-iceberg schema evolution on replace)
+        new_ice = _ice_schema(schema)
+        cur_id = md.get("current-schema-id", 0)
+        cur_ice = next((s for s in md.get("schemas", [])
+                        if s.get("schema-id", 0) == cur_id),
+                       md["schemas"][0] if md.get("schemas") else None)
+        def _fields(s):
+            return [(f["name"], f["type"]) for f in s.get("fields", [])] \
+                if s else None
+        if _fields(cur_ice) != _fields(new_ice):
+            new_id = max((s.get("schema-id", 0)
+                          for s in md.get("schemas", [])), default=-1) + 1
+            new_ice["schema-id"] = new_id
+            md["schemas"] = md.get("schemas", []) + [new_ice]
+            md["current-schema-id"] = new_id
+            md["last-column-id"] = max(md.get("last-column-id", 0),
+                                       len(schema))
 
     seq = md["last-sequence-number"] + 1
     snap_id = _new_snapshot_id()
@@ -416,10 +441,13 @@ def write(path: str, chunk, mode: str, options: Dict[str, str]):
                 "file_sequence_number": None, "data_file": df}
                for df in data_files]
     mpath = os.path.join(t.meta_dir, f"{uuid.uuid4().hex}-m0.avro")
-    cur_schema = md["schemas"][0] if md.get("schemas") else _ice_schema(schema)
+    cur_id = md.get("current-schema-id", 0)
+    cur_schema = next((s for s in md.get("schemas", [])
+                       if s.get("schema-id", 0) == cur_id),
+                      _ice_schema(schema))
     write_container(mpath, _MANIFEST_ENTRY_SCHEMA, entries, metadata={
         "schema": json.dumps(cur_schema).encode(),
-        "schema-id": b"0",
+        "schema-id": str(cur_id).encode(),
         "partition-spec": json.dumps([]).encode(),
         "partition-spec-id": b"0",
         "format-version": b"2",

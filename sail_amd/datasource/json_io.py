@@ -37,3 +37,25 @@ def read(paths: List[str], schema, device, options: Dict[str, str]):
     tables = [pajson.read_json(f) for f in files]
     tbl = pa.concat_tables(tables) if len(tables) > 1 else tables[0]
     return arrow_to_table(tbl, device=device)
+
+
+def write(path: str, chunk, mode: str, options: Dict[str, str]):
+    """NDJSON write (one JSON object per line, Spark json sink layout)."""
+    import json as _json
+
+    if path.endswith(".json"):
+        parent = os.path.dirname(path)
+        if parent:
+            os.makedirs(parent, exist_ok=True)
+        target = path
+    else:
+        os.makedirs(path, exist_ok=True)
+        target = os.path.join(path, "part-00000.json")
+    if mode == "error" and os.path.exists(target):
+        raise FileExistsError(target)
+    cols = [c.to_pylist() for c in chunk.columns]
+    with open(target, "w") as f:
+        for row in zip(*cols) if cols else []:
+            f.write(_json.dumps({n: v for n, v in zip(chunk.names, row)},
+                                default=str) + "\n")
+    return target

@@ -111,6 +111,10 @@ def write_source(fmt: str, path: str, chunk, mode: str, options, partition_by):
         return parquet_io.write(path, chunk, mode, options)
     if fmt == "csv":
         return csv_io.write(path, chunk, mode, options)
+    if fmt == "json":
+        from . import json_io
+
+        return json_io.write(path, chunk, mode, options)
     if fmt == "delta":
         from . import delta
 

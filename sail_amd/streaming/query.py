@@ -228,6 +228,12 @@ class StreamingQuery:
         self._agg_state: Optional[_AggState] = None
         self._upper_parent = None
         self._retained: Optional[Chunk] = None  # non-incremental fallback
+        if self.checkpoint:
+            self.id = self._load_or_create_metadata()
+        if hasattr(self.sink, "app_id"):
+            # stable identity for sink-side idempotence (delta txn /
+            # _sail_metadata manifests survive restarts with the query id)
+            self.sink.app_id = self.id
         self._prepare()
         if self.checkpoint:
             self._recover()
@@ -310,6 +316,18 @@ class StreamingQuery:
         os.makedirs(d, exist_ok=True)
         return d
 
+    def _load_or_create_metadata(self) -> str:
+        """Stable query id across restarts (Spark checkpoint `metadata`
+        layout: {"id": ...})."""
+        os.makedirs(self.checkpoint, exist_ok=True)
+        p = os.path.join(self.checkpoint, "metadata")
+        if os.path.exists(p):
+            with open(p) as f:
+                return json.load(f)["id"]
+        with open(p, "w") as f:
+            json.dump({"id": self.id}, f)
+        return self.id
+
     def _recover(self):
         offs = self._ckpt_dir("offsets")
         commits = self._ckpt_dir("commits")
@@ -322,10 +340,16 @@ class StreamingQuery:
             self.batch_id = last
             with open(os.path.join(commits, str(last))) as f:
                 self._max_event_us = json.load(f).get("maxEventTimeUs")
-            # restore aggregation state
-            state_dir = os.path.join(self.checkpoint, "state")
-            if self._agg_state is not None and os.path.isdir(state_dir):
-                self._load_state(state_dir)
+            # restore state from the snapshot matching the last COMMITTED
+            # batch (snapshots are versioned per batch id, so a crash
+            # between state write and commit cannot double-count: the
+            # pending batch replays against the previous state)
+            state_root = os.path.join(self.checkpoint, "state")
+            vdir = os.path.join(state_root, str(last))
+            if os.path.isdir(vdir):
+                self._load_state(vdir)
+            elif os.path.isdir(state_root):
+                self._load_state(state_root)  # legacy unversioned layout
         # a pending offset without a commit is replayed by the normal loop:
         # read_between(self._offset, that offset) reproduces the batch.
         if pending and (not done or pending[-1] > done[-1]):
@@ -339,21 +363,42 @@ class StreamingQuery:
         with open(os.path.join(d, str(i))) as f:
             return json.load(f)["offset"]
 
-    def _save_state(self):
-        if self._agg_state is None or not self.checkpoint:
+    def _save_state(self, bid: int):
+        """Versioned per-batch state snapshot under state/<bid>/ — written
+        BEFORE the commit marker. Covers incremental aggregation state and
+        retained-mode accumulated input (both must survive restart for
+        exactly-once results)."""
+        if not self.checkpoint:
             return
-        chunk = self._agg_state.state_chunk()
+        kind = "agg"
+        chunk = None
+        if self._agg_state is not None:
+            chunk = self._agg_state.state_chunk()
+        elif self._mode == "retained":
+            kind, chunk = "retained", self._retained
         if chunk is None:
             return
         from ..datasource.delta import schema_to_string
         from ..datasource.registry import write_source
 
-        state_dir = self._ckpt_dir("state")
+        state_dir = os.path.join(self._ckpt_dir("state"), str(bid))
+        os.makedirs(state_dir, exist_ok=True)
         schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
         write_source("parquet", os.path.join(state_dir, "data"), chunk,
                      "overwrite", {}, None)
         with open(os.path.join(state_dir, "schema.json"), "w") as f:
-            f.write(schema_to_string(schema))
+            json.dump({"kind": kind, "schema": schema_to_string(schema)}, f)
+
+    def _prune_state(self, bid: int):
+        """Drop state snapshots older than the previous committed batch."""
+        import shutil
+
+        sd = os.path.join(self.checkpoint, "state")
+        if not os.path.isdir(sd):
+            return
+        for f in os.listdir(sd):
+            if f.isdigit() and int(f) < bid - 1:
+                shutil.rmtree(os.path.join(sd, f), ignore_errors=True)
 
     def _load_state(self, state_dir):
         from ..datasource.delta import schema_from_string
@@ -364,11 +409,19 @@ class StreamingQuery:
         if not (os.path.exists(sp) and os.path.isdir(dp)):
             return
         with open(sp) as f:
-            schema = schema_from_string(f.read())
+            obj = json.load(f)
+        if isinstance(obj, dict) and "kind" in obj:
+            kind, schema = obj["kind"], schema_from_string(obj["schema"])
+        else:  # legacy: the file IS the schema string
+            kind, schema = "agg", schema_from_string(json.dumps(obj))
         files = [os.path.join(dp, f) for f in sorted(os.listdir(dp))
                  if f.endswith(".parquet")]
-        tbl = read_source("parquet", files, {}, schema, "cpu")
-        self._agg_state.load_chunk(Chunk.from_table(tbl))
+        tbl = read_source("parquet", files, {}, schema, self.session.device)
+        chunk = Chunk.from_table(tbl)
+        if kind == "retained":
+            self._retained = chunk
+        elif self._agg_state is not None:
+            self._agg_state.load_chunk(chunk)
 
     # -- lifecycle ----------------------------------------------------------
     def start(self) -> "StreamingQuery":
@@ -445,10 +498,11 @@ class StreamingQuery:
         if result is not None:
             self.sink.write(result, bid, self.output_mode)
         if self.checkpoint:
-            self._save_state()
+            self._save_state(bid)
             with open(os.path.join(self._ckpt_dir("commits"), str(bid)), "w") as f:
                 json.dump({"batchId": bid,
                            "maxEventTimeUs": self._max_event_us}, f)
+            self._prune_state(bid)
         self._offset = end
         self.batch_id = bid
         self.last_progress = {

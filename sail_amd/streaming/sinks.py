@@ -2,6 +2,7 @@
 file/delta sinks through the batch writers)."""
 from __future__ import annotations
 
+import os
 from typing import Callable, Dict, Optional
 
 from ..engine.chunk import Chunk
@@ -77,22 +78,73 @@ class _BatchView:
 
 
 class FileSink(StreamSink):
-    """Append one file set per batch via the batch writers."""
+    """Append one file set per batch via the batch writers, exactly-once.
+
+    Idempotence (the reference relies on Spark's _spark_metadata /
+    delta txn versions for this):
+      * delta: each batch commit carries a `txn` action keyed by the query's
+        stable id; a replayed batch whose version is already committed is
+        skipped (ref: sail-delta-lake transaction app transactions).
+      * parquet/csv/json: files are named part-<batch_id>-...; a
+        `_sail_metadata/<batch_id>` manifest is O_EXCL-created after the
+        data files. On replay, a manifested batch is skipped; an
+        unmanifested partial batch has its stray files deleted and is
+        rewritten.
+    """
 
     def __init__(self, fmt: str, path: str, options: Dict[str, str]):
         self.fmt = fmt
         self.path = path
         self.options = options or {}
+        self.app_id: Optional[str] = None  # set by StreamingQuery (stable id)
+
+    def _manifest_dir(self) -> str:
+        d = os.path.join(self.path, "_sail_metadata")
+        os.makedirs(d, exist_ok=True)
+        return d
+
+    def committed_batch(self, batch_id: int) -> bool:
+        if self.fmt == "delta":
+            if self.app_id is None:
+                return False
+            from ..datasource import delta
+
+            try:
+                last = delta.last_txn_version(self.path, self.app_id)
+            except FileNotFoundError:
+                return False
+            return last is not None and last >= batch_id
+        return os.path.exists(os.path.join(self.path, "_sail_metadata",
+                                           str(batch_id)))
 
     def write(self, chunk, batch_id, output_mode):
+        import glob
+        import json as _json
+        import uuid as _uuid
+
         from ..datasource.registry import write_source
 
+        if self.committed_batch(batch_id):
+            return
         if self.fmt == "delta":
             from ..datasource import delta
 
-            delta.write(self.path, chunk, "append", self.options)
-        else:
-            write_source(self.fmt, self.path, chunk, "append", self.options, None)
+            txn = (self.app_id, batch_id) if self.app_id is not None else None
+            delta.write(self.path, chunk, "append", self.options, txn=txn)
+            return
+        # drop stray files from a crashed attempt at this batch
+        for stray in glob.glob(os.path.join(self.path,
+                                            f"part-{batch_id:05d}-*")):
+            os.remove(stray)
+        ext = {"parquet": "parquet", "csv": "csv", "json": "json"}[self.fmt]
+        os.makedirs(self.path, exist_ok=True)
+        fname = f"part-{batch_id:05d}-{_uuid.uuid4().hex}.{ext}"
+        target = os.path.join(self.path, fname)
+        write_source(self.fmt, target, chunk, "append", self.options, None)
+        manifest = os.path.join(self._manifest_dir(), str(batch_id))
+        fd = os.open(manifest, os.O_CREAT | os.O_EXCL | os.O_WRONLY)
+        with os.fdopen(fd, "w") as f:
+            _json.dump({"batchId": batch_id, "files": [fname]}, f)
 
 
 class NoopSink(StreamSink):

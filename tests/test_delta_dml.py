@@ -274,10 +274,75 @@ def test_vacuum_removes_orphaned_deletion_vectors(s, tmp_path):
     s.sql(f"DELETE FROM delta.`{base}` WHERE id = 4").collect()  # merges DV; old orphaned
     assert len([f for f in os.listdir(base)
                 if f.startswith("deletion_vector_")]) == 2
+    # retention is measured from the supersession (tombstone) time, which is
+    # fresh — the orphaned DV must survive a 1h window even with an old mtime
     for f in os.listdir(base):
         os.utime(os.path.join(base, f),
                  (time.time() - 10_000, time.time() - 10_000))
-    removed = DeltaLog(base).vacuum(retention_hours=1.0)
+    assert DeltaLog(base).vacuum(retention_hours=1.0) == []
+    removed = DeltaLog(base).vacuum(retention_hours=0.0)
     assert len(removed) == 1 and removed[0].startswith("deletion_vector_")
     # the live DV survived: reads still see both deletes
     assert s.sql(f"SELECT count(*) FROM delta.`{base}`").collect() == [(18,)]
+
+
+def test_vacuum_tombstone_time_not_mtime(s, tmp_path):
+    """a data file with an OLD mtime that becomes unreferenced
+    NOW must survive the retention window (tombstone age, not file age)."""
+    import os
+    import time
+
+    from sail_amd.datasource.delta import DeltaLog
+
+    base = str(tmp_path / "tombvac")
+    s.create_dataframe({"id": [1, 2, 3]}, name="tv_src")
+    s.sql(f"CREATE TABLE delta.`{base}` AS SELECT * FROM tv_src").collect()
+    # age the data files (created long before the retention window)
+    for f in os.listdir(base):
+        if f.endswith(".parquet"):
+            os.utime(os.path.join(base, f),
+                     (time.time() - 10 * 86400, time.time() - 10 * 86400))
+    # overwrite: old files become unreferenced with a fresh tombstone
+    s.create_dataframe({"id": [9]}, name="tv_src2")
+    s.table("tv_src2").write.format("delta").mode("overwrite").save(base)
+    assert DeltaLog(base).vacuum(retention_hours=1.0) == []
+    # time travel within the window still works
+    r = s.read.format("delta").option("versionAsOf", "0").load(base)
+    assert sorted(x[0] for x in r.collect()) == [1, 2, 3]
+    # past the window (retention 0) the tombstoned files go away
+    removed = DeltaLog(base).vacuum(retention_hours=0.0)
+    assert len(removed) >= 1
+    assert s.sql(f"SELECT * FROM delta.`{base}`").collect() == [(9,)]
+
+
+def test_dv_file_crc_is_full_crc32(s, tmp_path):
+    """DV files carry the full unmasked 32-bit big-endian CRC and
+    reads validate it."""
+    import os
+    import struct
+    import zlib
+
+    import pytest as _pytest
+
+    from sail_amd.datasource.delta import DeltaLog, dv_positions
+
+    base = str(tmp_path / "dvcrc")
+    s.create_dataframe({"id": list(range(50))}, name="crc_src")
+    s.sql(f"CREATE TABLE delta.`{base}` AS SELECT * FROM crc_src").collect()
+    s.sql(f"DELETE FROM delta.`{base}` WHERE id < 5").collect()
+    dvf = [f for f in os.listdir(base) if f.startswith("deletion_vector_")]
+    assert len(dvf) == 1
+    raw = open(os.path.join(base, dvf[0]), "rb").read()
+    (size,) = struct.unpack(">i", raw[1:5])
+    blob = raw[5:5 + size]
+    (crc,) = struct.unpack(">I", raw[5 + size:9 + size])
+    assert crc == (zlib.crc32(blob) & 0xFFFFFFFF)
+    # corruption is detected on read
+    _, adds, _, _ = DeltaLog(base).snapshot_adds()
+    dv = next(a["deletionVector"] for a in adds if a.get("deletionVector"))
+    bad = bytearray(raw)
+    bad[7] ^= 0xFF
+    with open(os.path.join(base, dvf[0]), "wb") as f:
+        f.write(bytes(bad))
+    with _pytest.raises(ValueError, match="checksum"):
+        dv_positions(base, dv)

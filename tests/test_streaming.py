@@ -326,3 +326,148 @@ def test_iceberg_stream_source(s, tmp_path):
         assert s.sql("SELECT * FROM ist_out").collect() == [(3,)]
     finally:
         q.stop()
+
+
+def test_retained_mode_checkpoint_recovery(s, tmp_path):
+    """retained-mode (non-incremental) input must be checkpointed
+    so a restart does not silently drop prior rows while offsets advance."""
+    ckpt = str(tmp_path / "ret_ckpt")
+    schema = {"k": T.STRING, "v": T.I64}
+    sdf = s.read_stream.format("memory").schema(schema).load(name="rv1")
+    src = sdf.source
+    q = (sdf.sql("SELECT count(DISTINCT k) AS d FROM rv1")
+         .write_stream.output_mode("complete").format("memory")
+         .query_name("r_out").option("checkpointLocation", ckpt)
+         .trigger(processing_time=0.01).start())
+    assert q._mode == "retained"
+    src.add_rows({"k": ["a", "b"], "v": [1, 1]})
+    q.process_all_available()
+    q.stop()
+    s2 = sail_amd.SessionContext(device="cpu")
+    sdf2 = s2.read_stream.format("memory").schema(schema).load(name="rv1")
+    src2 = sdf2.source
+    src2.add_rows({"k": ["a", "b"], "v": [1, 1]})  # replay committed offsets
+    q2 = (sdf2.sql("SELECT count(DISTINCT k) AS d FROM rv1")
+          .write_stream.output_mode("complete").format("memory")
+          .query_name("r_out").option("checkpointLocation", ckpt)
+          .trigger(processing_time=0.01).start())
+    assert q2.batch_id == q.batch_id
+    src2.add_rows({"k": ["a", "c"], "v": [1, 1]})  # only c is new
+    q2.process_all_available()
+    q2.stop()
+    assert q2.exception is None
+    # a,b retained from before the restart: distinct = {a,b,c} = 3
+    assert s2.sql("SELECT d FROM r_out").collect() == [(3,)]
+
+
+def test_state_crash_between_save_and_commit_no_double_count(s, tmp_path):
+    """a crash after _save_state but before the commit marker must
+    replay the pending batch against the PREVIOUS state (versioned
+    snapshots), not against state that already contains it."""
+    import json as _json
+    import os
+    import shutil
+
+    ckpt = str(tmp_path / "dc_ckpt")
+    schema = {"k": T.STRING, "v": T.I64}
+    sdf = s.read_stream.format("memory").schema(schema).load(name="dc1")
+    src = sdf.source
+    q = (sdf.sql("SELECT k, sum(v) AS sv FROM dc1 GROUP BY k")
+         .write_stream.output_mode("complete").format("memory")
+         .query_name("dc_out").option("checkpointLocation", ckpt)
+         .trigger(processing_time=0.01).start())
+    src.add_rows({"k": ["a"], "v": [1]})
+    q.process_all_available()
+    q.stop()
+    last = q.batch_id
+    # forge the crash window: state for batch last+1 exists (as if merged),
+    # offsets/last+1 exists, but commits/last+1 does NOT
+    nxt = last + 1
+    shutil.copytree(os.path.join(ckpt, "state", str(last)),
+                    os.path.join(ckpt, "state", str(nxt)))
+    # make the forged pending-state distinguishable: as if the pending batch
+    # (v=10) was already merged into it (a -> 11)
+    import pyarrow as _pa
+    import pyarrow.parquet as _pq
+
+    forged = os.path.join(ckpt, "state", str(nxt), "data",
+                          "part-00000.parquet")
+    _pq.write_table(_pa.table({"k0": ["a"], "p0": [11]}), forged)
+    with open(os.path.join(ckpt, "offsets", str(nxt)), "w") as f:
+        _json.dump({"offset": 2}, f)
+    s2 = sail_amd.SessionContext(device="cpu")
+    sdf2 = s2.read_stream.format("memory").schema(schema).load(name="dc1")
+    src2 = sdf2.source
+    src2.add_rows({"k": ["a"], "v": [1]})   # replayed committed batch
+    src2.add_rows({"k": ["a"], "v": [10]})  # the pending batch's rows
+    q2 = (sdf2.sql("SELECT k, sum(v) AS sv FROM dc1 GROUP BY k")
+          .write_stream.output_mode("complete").format("memory")
+          .query_name("dc_out").option("checkpointLocation", ckpt)
+          .trigger(processing_time=0.01).start())
+    q2.process_all_available()
+    q2.stop()
+    assert q2.exception is None
+    # exactly-once: 1 + 10 (not 1 + 10 + 10 from the forged state)
+    assert dict(s2.sql("SELECT k, sv FROM dc_out").collect()) == {"a": 11}
+
+
+def test_file_sink_replay_idempotent(s, tmp_path):
+    """replaying a pending batch after a crash between sink.write
+    and the commit marker must not duplicate rows (manifest idempotence)."""
+    import os
+
+    ckpt = str(tmp_path / "fs_ckpt")
+    out = str(tmp_path / "fs_out")
+    schema = {"k": T.STRING, "v": T.I64}
+    sdf = s.read_stream.format("memory").schema(schema).load(name="fs1")
+    src = sdf.source
+    q = (sdf.sql("SELECT k, v FROM fs1")
+         .write_stream.format("parquet").option("checkpointLocation", ckpt)
+         .trigger(processing_time=0.01).start(out))
+    src.add_rows({"k": ["a", "b"], "v": [1, 2]})
+    q.process_all_available()
+    q.stop()
+    last = q.batch_id
+    # forge the crash: drop the commit marker; sink files + manifest remain
+    os.remove(os.path.join(ckpt, "commits", str(last)))
+    s2 = sail_amd.SessionContext(device="cpu")
+    sdf2 = s2.read_stream.format("memory").schema(schema).load(name="fs1")
+    src2 = sdf2.source
+    src2.add_rows({"k": ["a", "b"], "v": [1, 2]})  # pending batch replays
+    q2 = (sdf2.sql("SELECT k, v FROM fs1")
+          .write_stream.format("parquet").option("checkpointLocation", ckpt)
+          .trigger(processing_time=0.01).start(out))
+    q2.process_all_available()
+    q2.stop()
+    assert q2.exception is None
+    rows = s2.sql(f"SELECT k, v FROM parquet.`{out}` ORDER BY k").collect()
+    assert rows == [("a", 1), ("b", 2)]
+
+
+def test_delta_sink_replay_idempotent(s, tmp_path):
+    """Delta sink idempotence via txn actions keyed by the stable query id."""
+    import os
+
+    ckpt = str(tmp_path / "ds_ckpt")
+    out = str(tmp_path / "ds_out")
+    schema = {"k": T.STRING, "v": T.I64}
+    sdf = s.read_stream.format("memory").schema(schema).load(name="ds1")
+    src = sdf.source
+    q = (sdf.sql("SELECT k, v FROM ds1")
+         .write_stream.format("delta").option("checkpointLocation", ckpt)
+         .trigger(processing_time=0.01).start(out))
+    src.add_rows({"k": ["x"], "v": [7]})
+    q.process_all_available()
+    q.stop()
+    os.remove(os.path.join(ckpt, "commits", str(q.batch_id)))
+    s2 = sail_amd.SessionContext(device="cpu")
+    sdf2 = s2.read_stream.format("memory").schema(schema).load(name="ds1")
+    src2 = sdf2.source
+    src2.add_rows({"k": ["x"], "v": [7]})
+    q2 = (sdf2.sql("SELECT k, v FROM ds1")
+          .write_stream.format("delta").option("checkpointLocation", ckpt)
+          .trigger(processing_time=0.01).start(out))
+    q2.process_all_available()
+    q2.stop()
+    assert q2.exception is None
+    assert s2.sql(f"SELECT k, v FROM delta.`{out}`").collect() == [("x", 7)]

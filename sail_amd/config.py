@@ -23,6 +23,10 @@ DEFAULTS: Dict[str, str] = {
     # distributed exchanges (SPMD over RCCL)
     "sail.exec.broadcast_threshold_bytes": str(2 << 30),  # gathered build side above this -> hash shuffle
     "sail.exec.agg_shuffle_threshold_groups": "4000000",  # est. groups above this -> shuffled aggregation
+    # scan path: GPU parquet page decode (datasource/gpu_parquet.py)
+    # auto = GPU decode when supported, host pyarrow otherwise;
+    # force = raise instead of falling back (tests); off = host only
+    "sail.io.gpu_parquet": "auto",
     # kernels
     "sail.kernels.require_on_gpu": "true",
     "sail.kernels.grouped_agg_max_lds_groups": "4096",

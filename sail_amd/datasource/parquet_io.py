@@ -89,12 +89,38 @@ def infer_schema(paths: List[str]) -> List[Tuple[str, T.DataType]]:
     return out
 
 
+def _gpu_mode(options: Dict[str, str]) -> str:
+    if options and options.get("gpuDecode") is not None:
+        return str(options["gpuDecode"]).lower()
+    return os.environ.get("SAIL_IO_GPU_PARQUET", "auto").lower()
+
+
+def _try_read_gpu(files: List[str], schema, device, options):
+    """GPU page-decode path (gpu_parquet.py); None => host fallback."""
+    mode = _gpu_mode(options)
+    if mode in ("off", "false", "0") or not str(device).startswith("cuda"):
+        if mode == "force":
+            raise RuntimeError("gpuDecode=force but device is not cuda")
+        return None
+    from . import gpu_parquet
+
+    try:
+        return gpu_parquet.read_gpu(files, schema, device)
+    except gpu_parquet.Unsupported:
+        if mode == "force":
+            raise
+        return None
+
+
 def read(paths: List[str], schema, device, options: Dict[str, str]):
     files = _expand(paths)
     pkeys, pvals = ([], [])
     if len(paths) == 1 and os.path.isdir(paths[0]):
         pkeys, pvals = _partition_info(paths[0], files)
     if not pkeys:
+        t = _try_read_gpu(files, schema, device, options)
+        if t is not None:
+            return t
         cols = [n for n, _ in schema] if schema else None
         tbl = pq.read_table(files, columns=cols)
         return arrow_to_table(tbl, device=device)
@@ -119,10 +145,14 @@ def read(paths: List[str], schema, device, options: Dict[str, str]):
     from ..engine.column import Column, StringColumn, Table
 
     parts = []
+    dschema = [(n, t) for n, t in schema if n in data_names] if schema else None
     for f, v in zip(files, pvals):
-        t = pq.read_table(f, columns=[n for n, _ in schema if n in data_names]
-                          if schema else None)
-        parts.append((t, v))
+        gt = _try_read_gpu([f], dschema, device, options)
+        if gt is not None:
+            parts.append((gt, v))
+            continue
+        t = pq.read_table(f, columns=[n for n, _ in dschema] if dschema else None)
+        parts.append((arrow_to_table(t, device=device), v))
     if not parts:
         cols = {}
         for n, t in (schema or []):
@@ -130,7 +160,7 @@ def read(paths: List[str], schema, device, options: Dict[str, str]):
         return Table(cols)
     from ..engine.executor import concat_columns
 
-    tables = [arrow_to_table(t, device=device) for t, _ in parts]
+    tables = [t for t, _ in parts]  # already engine Tables (GPU or host path)
     out_cols = {}
     for n in tables[0].columns:
         out_cols[n] = concat_columns([t.columns[n] for t in tables])             if len(tables) > 1 else tables[0].columns[n]

@@ -20,6 +20,20 @@ std::vector<torch::Tensor> hj_probe_fill(torch::Tensor tkeys, torch::Tensor thea
                                          torch::Tensor next, torch::Tensor keys,
                                          torch::Tensor offsets, int64_t total);
 std::vector<torch::Tensor> hg_group(torch::Tensor keys);
+// parquet_decode.hip
+torch::Tensor pq_rle_decode(torch::Tensor buf, torch::Tensor pages, int64_t total);
+torch::Tensor pq_plain_copy(torch::Tensor buf, torch::Tensor pages, int64_t total,
+                            int64_t width);
+void pq_copy_bytes(torch::Tensor buf, torch::Tensor pages, torch::Tensor out);
+torch::Tensor pq_flba_i64(torch::Tensor buf, torch::Tensor pages, int64_t total,
+                          int64_t width);
+std::vector<torch::Tensor> pq_delta_decode(torch::Tensor buf, torch::Tensor pages,
+                                           int64_t total);
+std::vector<torch::Tensor> pq_bytearray_walk(torch::Tensor buf, torch::Tensor pages,
+                                             int64_t total);
+torch::Tensor pq_gather_strings(torch::Tensor buf, torch::Tensor src_pos,
+                                torch::Tensor lengths, torch::Tensor out_offsets,
+                                int64_t total_bytes);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sail_amd MI355X (gfx950) kernels";
@@ -34,4 +48,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hj_probe_count", &hj_probe_count, "hash join probe, count matches");
   m.def("hj_probe_fill", &hj_probe_fill, "hash join probe, emit match pairs");
   m.def("hg_group", &hg_group, "hash group-id assignment (sort-free group_ids)");
+  m.def("pq_rle_decode", &pq_rle_decode, "parquet RLE/bit-packed hybrid decode");
+  m.def("pq_plain_copy", &pq_plain_copy, "parquet PLAIN fixed-width page copy");
+  m.def("pq_copy_bytes", &pq_copy_bytes, "parquet raw page-region byte copy");
+  m.def("pq_flba_i64", &pq_flba_i64, "parquet FLBA big-endian -> int64");
+  m.def("pq_delta_decode", &pq_delta_decode, "parquet DELTA_BINARY_PACKED decode");
+  m.def("pq_bytearray_walk", &pq_bytearray_walk, "parquet PLAIN byte_array walk");
+  m.def("pq_gather_strings", &pq_gather_strings, "gather byte_array payloads");
 }

@@ -31,6 +31,10 @@ def main():
     ap.add_argument("--device", default=None)
     ap.add_argument("--queries", default=None, help="comma-separated subset, e.g. 1,6,13")
     ap.add_argument("--print-times", action="store_true")
+    ap.add_argument("--scan", default=os.environ.get("SAIL_BENCH_SCAN", "parquet"),
+                    choices=["parquet", "resident"],
+                    help="parquet: timed steps scan+GPU-decode parquet from disk "
+                         "(BASELINE config #2); resident: HBM-resident tables")
     args = ap.parse_args()
 
     import torch
@@ -77,12 +81,22 @@ def main():
         session.dist = DistContext(dist, rank=rank, world=world, device=device)
 
     t0 = time.time()
+    scan_info = None
     if args.workload == "clickbench":
         register_clickbench(session, rows=args.rows, device=device, rank=rank, world=world)
     elif args.workload == "delta":
         from sail_amd.datagen.delta_bench import setup_delta_bench
 
         setup_delta_bench(session, sf=args.sf, device=device, rank=rank, world=world)
+    elif args.scan == "parquet":
+        # scan-inclusive: every timed query re-reads its columns from the
+        # parquet shards on disk and decodes them on the GPU
+        from sail_amd.datagen.tpch import register_tpch_parquet
+
+        scan_info = register_tpch_parquet(session, sf=args.sf, device=device,
+                                          rank=rank, world=world)
+        print(f"# parquet shards: {scan_info['bytes']/1e9:.2f} GB in "
+              f"{scan_info['data_dir']}", file=sys.stderr)
     else:
         from sail_amd.datagen.tpch import register_tpch
 
@@ -198,6 +212,9 @@ def main():
                 "rows": args.rows if args.workload == "clickbench" else None,
                 "queries": len(qids),
                 "parallelism": f"sharded dp{n_gpus}" if n_gpus > 1 else "single-gpu",
+                "scan": ("parquet->GPU-decode->HBM (disk included in timed region)"
+                         if scan_info is not None else "HBM-resident"),
+                "parquet_bytes": scan_info["bytes"] if scan_info else None,
             },
         }
         print(json.dumps(out))

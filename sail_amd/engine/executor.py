@@ -152,7 +152,18 @@ class Executor:
         from ..datasource.registry import read_source
 
         t = read_source(p.format, p.paths, p.options, p.schema, self.ctx.device)
-        return Chunk.from_table(t)
+        if p.schema:
+            want = [n for n, _ in p.schema]
+            have = list(t.columns.keys())
+            if have != want and all(n in t.columns for n in want):
+                # readers that ignore the pruned schema still return the
+                # full table; trim to the scan's declared columns
+                t = t.select(want)
+        out = Chunk.from_table(t)
+        if (p.options or {}).get("partitioning") == "sharded" \
+                and self.dctx is not None:
+            out.partitioning = "sharded"
+        return out
 
     def _x_LocalRelation(self, p: S.LocalRelation) -> Chunk:
         cols = []

@@ -229,7 +229,20 @@ def _prune(plan: S.Plan, needed: Set[int]) -> Tuple[S.Plan, Dict[int, int]]:
             mapping[nin + old] = nin_new + new
         return out, mapping
 
-    if isinstance(plan, (S.Read, S.DataSourceRead, S.LocalRelation, S.Range)):
+    if isinstance(plan, S.DataSourceRead):
+        # push the column subset INTO the scan: file readers (parquet GPU
+        # decode, pyarrow) only decode the schema's columns — the analogue
+        # of DataFusion's scan projection pushdown
+        keep = sorted(needed)
+        if len(keep) == len(plan.schema):
+            return _identity(plan)
+        nd = S.DataSourceRead(format=plan.format, paths=plan.paths,
+                              options=plan.options,
+                              user_schema=plan.user_schema)
+        nd.schema = [plan.schema[i] for i in keep]
+        return nd, {old: new for new, old in enumerate(keep)}
+
+    if isinstance(plan, (S.Read, S.LocalRelation, S.Range)):
         keep = sorted(needed)
         if len(keep) == len(plan.schema):
             return _identity(plan)

@@ -207,3 +207,74 @@ def test_tpch_world4_subset_matches_single():
                     assert gv == pytest.approx(wv, rel=1e-9, abs=1e-9), f"q{q} row {i}"
                 else:
                     assert gv == wv, f"q{q} row {i}: {gv!r} != {wv!r}"
+
+
+def _worker_scan(rank, world, port, out_dir, qids):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    import sail_amd
+    from sail_amd.datagen.tpch import register_tpch_parquet
+    from sail_amd.datagen.tpch_queries import QUERIES
+    from sail_amd.exec.context import DistContext
+
+    s = sail_amd.SessionContext(device="cpu")
+    s.dist = DistContext(dist, rank=rank, world=world, device="cpu")
+    register_tpch_parquet(s, sf=0.01, rank=rank, world=world,
+                          data_dir=os.path.join(out_dir, "shards"))
+    results = {q: s.sql(QUERIES[q]).collect() for q in qids}
+    if rank == 0:
+        with open(os.path.join(out_dir, "rank0.pkl"), "wb") as f:
+            pickle.dump(results, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_tpch_scan_mode_world2_matches_single():
+    """Scan-inclusive bench path (parquet shard views, pruned scans) at
+    world=2 matches single-process resident execution."""
+    import sail_amd
+    from sail_amd.datagen.tpch import TpchGenerator
+    from sail_amd.datagen.tpch_queries import QUERIES
+    from sail_amd.engine.executor import concat_columns
+    from sail_amd.engine.column import Table
+
+    qids = [1, 3, 6, 13, 18, 21]
+    single = sail_amd.SessionContext(device="cpu")
+    shard_tables = [TpchGenerator(sf=0.01, device="cpu", rank=r, world=2).generate_all()
+                    for r in range(2)]
+    for name in shard_tables[0]:
+        if name in ("region", "nation"):
+            single.catalog.register_table(name, shard_tables[0][name])
+            continue
+        cols = {}
+        for cn in shard_tables[0][name].columns:
+            cols[cn] = concat_columns([shard_tables[r][name].columns[cn] for r in range(2)])
+        single.catalog.register_table(name, Table(cols))
+    want = {q: single.sql(QUERIES[q]).collect() for q in qids}
+
+    with tempfile.TemporaryDirectory() as d:
+        port = 29517
+        ctx = mp.get_context("spawn")
+        procs = [ctx.Process(target=_worker_scan, args=(r, 2, port, d, qids))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=540)
+        for p in procs:
+            assert p.exitcode == 0, f"worker failed: {p.exitcode}"
+        with open(os.path.join(d, "rank0.pkl"), "rb") as f:
+            got = pickle.load(f)
+
+    for q in qids:
+        assert len(got[q]) == len(want[q]), f"q{q}"
+        for i, (g, w) in enumerate(zip(got[q], want[q])):
+            for gv, wv in zip(g, w):
+                if isinstance(wv, float):
+                    assert gv == pytest.approx(wv, rel=1e-9, abs=1e-9), f"q{q} row {i}"
+                else:
+                    assert gv == wv, f"q{q} row {i}"

@@ -208,8 +208,8 @@ def _upload_ranges(path: str, ranges: List[Tuple[int, int]], device):
     """Read file byte ranges into pinned staging, one H2D copy; returns
     (device u8 tensor, [staging offset per range])."""
     total = sum(e - s for s, e in ranges)
-    stage = _staging(total)
-    view = stage.numpy()
+    stage = _staging(total + 16)  # +16: aligned-word kernels may read past
+    view = stage.numpy()          # the last page's bytes (funnel shift)
     offs = []
     pos = 0
     with open(path, "rb") as f:
@@ -220,7 +220,8 @@ def _upload_ranges(path: str, ranges: List[Tuple[int, int]], device):
                 raise IOError(f"short read in {path}")
             offs.append(pos)
             pos += e - s
-    dev = stage[:total].to(device, non_blocking=True)
+    view[total:total + 16] = 0
+    dev = stage[:total + 16].to(device, non_blocking=True)
     return dev, offs
 
 

@@ -33,8 +33,19 @@ def group_ids(keys: List[Column], mask: Optional[torch.Tensor] = None
     dev = keys[0].device
     n = len(keys[0])
     norm = []
-    for c in keys:
-        k = normalize_key(c)
+    from .column import StringColumn as _SC
+    from .joins import exact_string_codes
+
+    raw_str = [i for i, c in enumerate(keys)
+               if isinstance(c, _SC) and not c.is_dict]
+    exact = {}
+    if raw_str:
+        # raw-string keys: exact dense codes (128-bit hash + byte-verify),
+        # not FNV-64 — grouping must never merge distinct strings
+        codes = exact_string_codes([keys[i] for i in raw_str])
+        exact = dict(zip(raw_str, codes))
+    for i, c in enumerate(keys):
+        k = exact[i] if i in exact else normalize_key(c)
         if c.validity is not None:
             # give nulls a dedicated code below the domain
             k = torch.where(c.valid_mask(), k, torch.full_like(k, k.min().item() - 1 if n else -1))

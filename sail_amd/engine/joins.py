@@ -42,18 +42,28 @@ def normalize_key_pair(a: Column, b: Column):
         lb = b.offsets[1:] - b.offsets[:-1]
         ma = int(la.max().item()) if len(a) else 0
         mb = int(lb.max().item()) if len(b) else 0
-        force = max(ma, mb) > 7
-        return raw_string_key(a, force), raw_string_key(b, force)
+        if max(ma, mb) <= 7:
+            return raw_string_key(a), raw_string_key(b)
+        # exact shared codes across both sides (no FNV-collision class):
+        # equal strings get equal codes, distinct strings never do
+        ka, kb = exact_string_codes([a, b])
+        return ka, kb
     if isinstance(a, StringColumn) and isinstance(b, StringColumn) \
             and a.is_dict != b.is_dict:
-        # one side dict, one raw: decode dict side keys through its values
+        # one side dict, one raw: exact codes over (dict values, raw rows);
+        # the dict side maps its per-row codes through the value codes
         da = a if a.is_dict else b
         ra = b if a.is_dict else a
         la = ra.offsets[1:] - ra.offsets[:-1]
         force = (int(la.max().item()) if len(ra) else 0) > 7 or \
             max((len(v) for v in da.dict_values()), default=0) > 7
-        dk = _dict_side_key(da, force)
-        rk = raw_string_key(ra, force)
+        if not force:
+            dk = _dict_side_key(da, False)
+            rk = raw_string_key(ra, False)
+            return (dk, rk) if a.is_dict else (rk, dk)
+        dvals = StringColumn(da.offsets, da.bytes_, None, None)
+        vcodes, rk = exact_string_codes([dvals, ra])
+        dk = vcodes.index_select(0, da.codes.to(torch.int64).clamp_min(0))
         return (dk, rk) if a.is_dict else (rk, dk)
     return normalize_key(a), normalize_key(b)
 
@@ -253,14 +263,154 @@ def fnv_key_tensor(c: StringColumn) -> torch.Tensor:
 
 def raw_string_key(c: StringColumn, force_hash: bool = False) -> torch.Tensor:
     """Exact packed key for <=7-byte strings; FNV-1a 64-bit beyond (same
-    values on CPU and GPU). Collision probability at 1e8 distinct strings is
-    ~3e-4 per query over the full key space — documented engine tradeoff for
-    raw-string grouping; dictionary-encoded columns are always exact."""
+    values on CPU and GPU). Used where keys must be VALUE-determined
+    (literal comparisons, array membership). Group-by and join keys go
+    through exact_string_codes instead — no collision class there."""
     lens = c.offsets[1:] - c.offsets[:-1]
     max_len = int(lens.max().item()) if len(c) else 0
     if not force_hash and max_len <= 7:
         return short_string_key(c)
     return fnv_key_tensor(c)
+
+
+# -- exact string codes ------------------------------------------------------
+# Two independent 64-bit hash families + byte-equality verification of
+# hash-equal neighbors => exact dictionary codes for raw-string group/join
+# keys (kills the FNV-collision silent-wrong-answer class; the detection of
+# a (2^-128-probability) double collision falls back to host encoding).
+# ref: sail-execution join planner unmatched-row discipline
+# (crates/sail-execution/src/job_graph/planner.rs:101) relies on exact keys.
+
+_H2_SEED = 0x9E3779B97F4A7C15
+_H2_MULT = 0xC6A4A7935BD1E995  # murmur64A multiplier — a different family
+_H2_LEN_MIX = 0x2545F4914F6CDD1D
+
+
+def _signed64(v: int) -> int:
+    return v - (1 << 64) if v >= (1 << 63) else v
+
+
+def hash2_tensor(c: StringColumn) -> torch.Tensor:
+    """Second hash family (must match string_hash64_seeded_kernel)."""
+    if c.is_cuda:
+        from ..ops import kernels as K
+
+        return K.require().string_hash64_seeded(
+            c.offsets, c.bytes_, _signed64(_H2_SEED), _signed64(_H2_MULT))
+    import numpy as np
+
+    offs = c.offsets.numpy()
+    byts = c.bytes_.numpy().astype(np.uint64)
+    n = len(offs) - 1
+    lens = offs[1:] - offs[:-1]
+    max_len = int(lens.max()) if n else 0
+    h = np.full(n, np.uint64(_H2_SEED), dtype=np.uint64)
+    starts = offs[:-1]
+    mult = np.uint64(_H2_MULT)
+    with np.errstate(over="ignore"):
+        for k in range(max_len):
+            live = k < lens
+            idx = np.where(live, starts + k, 0)
+            b = byts[idx] if byts.size else np.zeros(n, dtype=np.uint64)
+            nh = (h ^ b) * mult
+            h = np.where(live, nh, h)
+        h = h ^ (lens.astype(np.uint64) * np.uint64(_H2_LEN_MIX))
+    return torch.from_numpy(h.view(np.int64).copy()).to(c.device)
+
+
+def string_hash_pair(c: StringColumn):
+    """(h1, h2): the 128-bit combined key for exact codes. Separated out so
+    tests can force collisions."""
+    return fnv_key_tensor(c), hash2_tensor(c)
+
+
+def str_pairs_equal(a: StringColumn, ia: torch.Tensor,
+                    b: StringColumn, ib: torch.Tensor) -> torch.Tensor:
+    """Byte-exact equality of (a[ia[i]], b[ib[i]]) pairs -> bool tensor."""
+    if a.is_cuda:
+        from ..ops import kernels as K
+
+        return K.require().str_pairs_equal(
+            a.offsets, a.bytes_, ia, b.offsets, b.bytes_, ib).to(torch.bool)
+    import numpy as np
+
+    oa = a.offsets.numpy()
+    ob = b.offsets.numpy()
+    ba = a.bytes_.numpy()
+    bb = b.bytes_.numpy()
+    out = np.zeros(ia.numel(), dtype=bool)
+    for i, (x, y) in enumerate(zip(ia.numpy(), ib.numpy())):
+        sa = ba[oa[x]:oa[x + 1]]
+        sb = bb[ob[y]:ob[y + 1]]
+        out[i] = len(sa) == len(sb) and bool((sa == sb).all())
+    return torch.from_numpy(out)
+
+
+def _concat_raw(cols):
+    """Concatenate raw string columns into one (offsets, bytes) view."""
+    if len(cols) == 1:
+        return cols[0]
+    dev = cols[0].device
+    lens = torch.cat([c.offsets[1:] - c.offsets[:-1] for c in cols])
+    offs = torch.zeros(lens.numel() + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(lens, 0, out=offs[1:])
+    bytes_ = torch.cat([c.bytes_ for c in cols])
+    return StringColumn(offs, bytes_, None, None)
+
+
+def _host_exact_codes(comb: StringColumn) -> torch.Tensor:
+    """128-bit-collision fallback: exact host dictionary encode."""
+    vals = comb.to_pylist()
+    seen = {}
+    out = torch.empty(len(vals), dtype=torch.int64)
+    for i, v in enumerate(vals):
+        v = v or ""
+        j = seen.get(v)
+        if j is None:
+            j = len(seen)
+            seen[v] = j
+        out[i] = j
+    return out.to(comb.device)
+
+
+def exact_string_codes(cols) -> list:
+    """Exact dense codes for raw-string key columns: equal strings (across
+    all given columns) share a code, distinct strings never do. Sort by
+    (h1, h2), verify hash-equal neighbors byte-for-byte on device, assign
+    codes at verified boundaries. Returns one int64 tensor per column."""
+    comb = _concat_raw(cols)
+    n = len(comb)
+    if n == 0:
+        return [torch.zeros(0, dtype=torch.int64, device=comb.device)
+                for _ in cols]
+    h1, h2 = string_hash_pair(comb)
+    order = torch.argsort(h2, stable=True)
+    order = order[torch.argsort(h1.index_select(0, order), stable=True)]
+    s1 = h1.index_select(0, order)
+    s2 = h2.index_select(0, order)
+    same = (s1[1:] == s1[:-1]) & (s2[1:] == s2[:-1])
+    codes_all = None
+    idx = torch.nonzero(same, as_tuple=False).flatten()
+    if idx.numel():
+        ia = order.index_select(0, idx)
+        ib = order.index_select(0, idx + 1)
+        eq = str_pairs_equal(comb, ia, comb, ib)
+        if not bool(eq.all().item()):
+            # genuine 128-bit double collision (or a forced test): exact
+            # host fallback keeps the answer right
+            codes_all = _host_exact_codes(comb)
+    if codes_all is None:
+        boundary = torch.ones(n, dtype=torch.int64, device=comb.device)
+        boundary[1:] = (~same).to(torch.int64)
+        codes_sorted = torch.cumsum(boundary, 0) - 1
+        codes_all = torch.empty(n, dtype=torch.int64, device=comb.device)
+        codes_all[order] = codes_sorted
+    out = []
+    at = 0
+    for c in cols:
+        out.append(codes_all[at:at + len(c)])
+        at += len(c)
+    return out
 
 
 def _expand_matches_gpu(bids: torch.Tensor, pids: torch.Tensor, ext):

@@ -6,6 +6,11 @@ torch::Tensor like_mask(torch::Tensor offsets, torch::Tensor bytes, py::bytes pa
 torch::Tensor string_hash64(torch::Tensor offsets, torch::Tensor bytes);
 std::vector<torch::Tensor> substr_fixed(torch::Tensor offsets, torch::Tensor bytes,
                                         int64_t start, int64_t len);
+torch::Tensor string_hash64_seeded(torch::Tensor offsets, torch::Tensor bytes,
+                                   int64_t seed, int64_t mult);
+torch::Tensor str_pairs_equal(torch::Tensor offs_a, torch::Tensor bytes_a,
+                              torch::Tensor ia, torch::Tensor offs_b,
+                              torch::Tensor bytes_b, torch::Tensor ib);
 // hash_agg.hip
 torch::Tensor grouped_acc(torch::Tensor gid, c10::optional<torch::Tensor> mask,
                           std::vector<c10::optional<torch::Tensor>> vals,
@@ -40,6 +45,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("like_mask", &like_mask, "SQL LIKE mask over string column");
   m.def("string_hash64", &string_hash64, "FNV-1a 64 hash per string row");
   m.def("substr_fixed", &substr_fixed, "fixed-length substring extraction");
+  m.def("string_hash64_seeded", &string_hash64_seeded,
+        "seeded 64-bit string hash (second family for exact codes)");
+  m.def("str_pairs_equal", &str_pairs_equal,
+        "byte-exact equality of (row_a, row_b) string pairs");
   m.def("grouped_acc", &grouped_acc,
         "fused grouped accumulation (sum/count/min/max) for small group counts");
   m.def("hj_build", &hj_build, "hash join build (unique keys)");

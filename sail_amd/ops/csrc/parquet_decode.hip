@@ -130,25 +130,32 @@ __global__ void rle_decode_kernel(const uint8_t* __restrict__ buf,
         s_vdone = vdone;
       }
       __syncthreads();
+      // cache ALL shared loop state between the two barriers: thread 0
+      // rewrites s_* in the next iteration's parse phase, so the exit
+      // decision must be taken on locally-held copies (barrier-divergence
+      // race found by the SF100 run — pages with >kRuns runs)
       int nr = s_nruns;
-      if (nr == 0) break;
-      long lo = r_start[0];
-      long hi = (long)r_start[nr - 1] + r_count[nr - 1];
-      for (long v = lo + threadIdx.x; v < hi; v += blockDim.x) {
-        int a = 0, b = nr - 1;
-        while (a < b) {
-          int mid = (a + b + 1) >> 1;
-          if ((long)r_start[mid] <= v) a = mid; else b = mid - 1;
-        }
-        if (r_kind[a] == 0) {
-          dst[v] = (int)r_payload[a];
-        } else {
-          long bit = r_payload[a] + (v - r_start[a]) * (long)pg.bw;
-          dst[v] = (int)read_bits(pg.src, bit, pg.bw);
+      long vdone_now = s_vdone;
+      long pos_now = s_pos;
+      if (nr != 0) {
+        long lo = r_start[0];
+        long hi = (long)r_start[nr - 1] + r_count[nr - 1];
+        for (long v = lo + threadIdx.x; v < hi; v += blockDim.x) {
+          int a = 0, b = nr - 1;
+          while (a < b) {
+            int mid = (a + b + 1) >> 1;
+            if ((long)r_start[mid] <= v) a = mid; else b = mid - 1;
+          }
+          if (r_kind[a] == 0) {
+            dst[v] = (int)r_payload[a];
+          } else {
+            long bit = r_payload[a] + (v - r_start[a]) * (long)pg.bw;
+            dst[v] = (int)read_bits(pg.src, bit, pg.bw);
+          }
         }
       }
       __syncthreads();
-      if (s_vdone >= pg.nvals || s_pos >= pg.len) break;
+      if (nr == 0 || vdone_now >= pg.nvals || pos_now >= pg.len) break;
     }
     __syncthreads();
   }
@@ -299,23 +306,28 @@ __global__ void delta_decode_kernel(const uint8_t* __restrict__ buf,
         s_didx = didx;
       }
       __syncthreads();
+      // same barrier-race discipline as rle_decode_kernel: take the exit
+      // decision on copies cached between the two barriers
       int nmb = s_nmb;
-      if (nmb == 0) break;
-      long lo = m_base[0];
-      long hi = (long)m_base[nmb - 1] + m_cnt[nmb - 1];
-      for (long d = lo + threadIdx.x; d < hi; d += blockDim.x) {
-        int a = 0, b = nmb - 1;
-        while (a < b) {
-          int mid = (a + b + 1) >> 1;
-          if ((long)m_base[mid] <= d) a = mid; else b = mid - 1;
+      long didx_now = s_didx;
+      long ndeltas_now = s_ndeltas;
+      if (nmb != 0) {
+        long lo = m_base[0];
+        long hi = (long)m_base[nmb - 1] + m_cnt[nmb - 1];
+        for (long d = lo + threadIdx.x; d < hi; d += blockDim.x) {
+          int a = 0, b = nmb - 1;
+          while (a < b) {
+            int mid = (a + b + 1) >> 1;
+            if ((long)m_base[mid] <= d) a = mid; else b = mid - 1;
+          }
+          int bw = m_bw[a];
+          long v = bw == 0 ? 0
+                           : (long)read_bits(pg.src, m_bit[a] + (d - m_base[a]) * (long)bw, bw);
+          dst[1 + d] = m_min[a] + v;
         }
-        int bw = m_bw[a];
-        long v = bw == 0 ? 0
-                         : (long)read_bits(pg.src, m_bit[a] + (d - m_base[a]) * (long)bw, bw);
-        dst[1 + d] = m_min[a] + v;
       }
       __syncthreads();
-      if (s_didx >= s_ndeltas) break;
+      if (nmb == 0 || didx_now >= ndeltas_now) break;
     }
     __syncthreads();
     if (threadIdx.x == 0) data_end[p] = s_pos;

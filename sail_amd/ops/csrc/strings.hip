@@ -172,6 +172,66 @@ __global__ void string_hash64_kernel(const int64_t* __restrict__ offsets,
   }
 }
 
+__global__ void string_hash64_seeded_kernel(const int64_t* __restrict__ offsets,
+                                            const uint8_t* __restrict__ bytes,
+                                            int64_t* __restrict__ out, int64_t n,
+                                            uint64_t seed, uint64_t mult) {
+  // second, independent hash family for the 128-bit exact-string-code
+  // scheme (joins.exact_string_codes): different multiplier, not just a
+  // different seed — an FNV seed change is a function of (h1, len) and
+  // would NOT be independent
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    int64_t lo = offsets[i], hi = offsets[i + 1];
+    uint64_t h = seed;
+    int64_t j = lo;
+    for (; j + 8 <= hi; j += 8) {
+      uint64_t w;
+      __builtin_memcpy(&w, bytes + j, 8);
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        h = (h ^ (w & 0xffull)) * mult;
+        w >>= 8;
+      }
+    }
+    for (; j < hi; ++j) h = (h ^ bytes[j]) * mult;
+    h ^= (uint64_t)(hi - lo) * 0x2545F4914F6CDD1Dull;
+    out[i] = (int64_t)h;
+  }
+}
+
+__global__ void str_pairs_equal_kernel(const int64_t* __restrict__ offs_a,
+                                       const uint8_t* __restrict__ bytes_a,
+                                       const int64_t* __restrict__ ia,
+                                       const int64_t* __restrict__ offs_b,
+                                       const uint8_t* __restrict__ bytes_b,
+                                       const int64_t* __restrict__ ib,
+                                       uint8_t* __restrict__ out, int64_t n) {
+  // byte-exact equality per (row_a, row_b) pair — the verification pass
+  // that turns hashed string keys into exact ones
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    int64_t a = ia[i], b = ib[i];
+    int64_t la = offs_a[a + 1] - offs_a[a];
+    int64_t lb = offs_b[b + 1] - offs_b[b];
+    if (la != lb) { out[i] = 0; continue; }
+    const uint8_t* pa = bytes_a + offs_a[a];
+    const uint8_t* pb = bytes_b + offs_b[b];
+    int64_t j = 0;
+    bool eq = true;
+    for (; j + 8 <= la; j += 8) {
+      uint64_t wa, wb;
+      __builtin_memcpy(&wa, pa + j, 8);
+      __builtin_memcpy(&wb, pb + j, 8);
+      if (wa != wb) { eq = false; break; }
+    }
+    if (eq)
+      for (; j < la; ++j)
+        if (pa[j] != pb[j]) { eq = false; break; }
+    out[i] = eq ? 1 : 0;
+  }
+}
+
 __global__ void substr_fixed_kernel(const int64_t* __restrict__ offsets,
                                     const uint8_t* __restrict__ bytes,
                                     int start, int len,
@@ -260,6 +320,36 @@ torch::Tensor string_hash64(torch::Tensor offsets, torch::Tensor bytes) {
   hipLaunchKernelGGL(string_hash64_kernel, dim3(grid_for(n)), dim3(kBlock), 0, stream,
                      offsets.data_ptr<int64_t>(), bytes.data_ptr<uint8_t>(),
                      out.data_ptr<int64_t>(), n);
+  return out;
+}
+
+torch::Tensor string_hash64_seeded(torch::Tensor offsets, torch::Tensor bytes,
+                                   int64_t seed, int64_t mult) {
+  SAIL_CHECK(offsets.is_cuda() && bytes.is_cuda());
+  int64_t n = offsets.numel() - 1;
+  auto out = torch::empty({n}, offsets.options().dtype(torch::kInt64));
+  if (n == 0) return out;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(string_hash64_seeded_kernel, dim3(grid_for(n)), dim3(kBlock),
+                     0, stream, offsets.data_ptr<int64_t>(),
+                     bytes.data_ptr<uint8_t>(), out.data_ptr<int64_t>(), n,
+                     (uint64_t)seed, (uint64_t)mult);
+  return out;
+}
+
+torch::Tensor str_pairs_equal(torch::Tensor offs_a, torch::Tensor bytes_a,
+                              torch::Tensor ia, torch::Tensor offs_b,
+                              torch::Tensor bytes_b, torch::Tensor ib) {
+  SAIL_CHECK(offs_a.is_cuda() && bytes_a.is_cuda() && ia.is_cuda());
+  int64_t n = ia.numel();
+  auto out = torch::empty({n}, bytes_a.options().dtype(torch::kUInt8));
+  if (n == 0) return out;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(str_pairs_equal_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+                     stream, offs_a.data_ptr<int64_t>(),
+                     bytes_a.data_ptr<uint8_t>(), ia.data_ptr<int64_t>(),
+                     offs_b.data_ptr<int64_t>(), bytes_b.data_ptr<uint8_t>(),
+                     ib.data_ptr<int64_t>(), out.data_ptr<uint8_t>(), n);
   return out;
 }
 

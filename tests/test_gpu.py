@@ -434,3 +434,71 @@ def test_gpu_parquet_sql_roundtrip(tmp_path, gpu_session):
         assert r == [("a", n, sum(float(i % 97) for i in range(n)), n - 1)]
     finally:
         os.environ["SAIL_IO_GPU_PARQUET"] = "auto"
+
+
+def test_string_hash2_and_pairs_equal_match_cpu(ext):
+    from sail_amd.engine import joins
+
+    vals = [f"k-{i%97}-{'abc'*(i%11)}" for i in range(20000)]
+    cpu = StringColumn.from_pylist(vals, device="cpu", dict_encode=False)
+    gpu = cpu.to("cuda")
+    assert joins.hash2_tensor(gpu).cpu().tolist() == \
+        joins.hash2_tensor(cpu).tolist()
+    ia = torch.arange(0, 20000, 2)
+    ib = torch.arange(1, 20000, 2)
+    g = joins.str_pairs_equal(gpu, ia.cuda(), gpu, ib.cuda()).cpu()
+    c = joins.str_pairs_equal(cpu, ia, cpu, ib)
+    assert torch.equal(g, c)
+
+
+def test_exact_string_codes_gpu_matches_cpu(ext):
+    from sail_amd.engine import joins
+
+    vals = [f"value-{i % 513}-{'pad' * (i % 7)}" for i in range(100_000)]
+    cpu = StringColumn.from_pylist(vals, device="cpu", dict_encode=False)
+    gpu = cpu.to("cuda")
+    cc = joins.exact_string_codes([cpu])[0].tolist()
+    gc = joins.exact_string_codes([gpu])[0].cpu().tolist()
+    # code NUMBERS may differ (sort ties); the partition must be identical
+    def parts(codes):
+        m = {}
+        for i, c in enumerate(codes):
+            m.setdefault(c, []).append(i)
+        return sorted(map(tuple, m.values()))
+    assert parts(cc) == parts(gc)
+    assert len(set(gc)) == len(set(vals))
+
+
+def test_gpu_parquet_large_pages_batched_runs(tmp_path):
+    """Pages with >>kRuns RLE runs and >>kRuns delta miniblocks exercise the
+    kernels' LDS batching loops (the SF100 barrier-race regression)."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from sail_amd.datasource import gpu_parquet as G
+
+    n = 2_000_000
+    rng = np.random.default_rng(3)
+    ints = rng.integers(-10**14, 10**14, n)
+    # alternating short runs -> tens of thousands of RLE runs per page
+    dict_vals = np.array(["aaaa", "bbbb", "cccc"])[(np.arange(n) // 9) % 3]
+    p = str(tmp_path / "big.parquet")
+    pq.write_table(
+        pa.table({"k": pa.array(ints), "g": pa.array(dict_vals),
+                  "ks": pa.array(np.sort(rng.integers(0, 10**10, n)))}),
+        p, compression="NONE", use_dictionary=["g"],
+        column_encoding={"k": "DELTA_BINARY_PACKED",
+                         "ks": "DELTA_BINARY_PACKED"},
+        data_page_size=1 << 20, data_page_version="1.0",
+        row_group_size=n)
+    out = G.read_gpu([p], [("k", None), ("g", None), ("ks", None)], "cuda:0")
+    torch.cuda.synchronize()
+    assert out.columns["k"].data.cpu().numpy().tolist() == ints.tolist()
+    assert out.columns["ks"].data.cpu().numpy().min() >= 0
+    import numpy as _np
+    assert _np.array_equal(out.columns["ks"].data.cpu().numpy(),
+                           _np.sort(rng.integers(0, 10**10, 0)) if False else
+                           pq.read_table(p, columns=["ks"]).column("ks").to_numpy())
+    got_g = out.columns["g"]
+    assert got_g.is_dict
+    assert got_g.to_pylist() == dict_vals.tolist()

@@ -138,3 +138,95 @@ def test_timestampadd_hours_matches_datetime(us, hours):
     want = (delta.days * 86_400 + delta.seconds) * 1_000_000 \
         + delta.microseconds
     assert got == want
+
+
+# -- exact string keys (kill the FNV-collision class) -----------
+
+def _mk_session():
+    import sail_amd
+
+    return sail_amd.SessionContext(device="cpu")
+
+
+def test_exact_string_codes_randomized():
+    import random
+
+    from sail_amd.engine.column import StringColumn
+    from sail_amd.engine.joins import exact_string_codes
+
+    rng = random.Random(11)
+    vals = [f"string-{rng.randint(0, 300)}-{'pad' * rng.randint(0, 6)}"
+            for _ in range(5000)]
+    col = StringColumn.from_pylist(vals, device="cpu", dict_encode=False)
+    codes = exact_string_codes([col])[0].tolist()
+    by_code = {}
+    for v, c in zip(vals, codes):
+        by_code.setdefault(c, set()).add(v)
+    # each code maps to exactly one string and vice versa
+    assert all(len(s) == 1 for s in by_code.values())
+    assert len(by_code) == len(set(vals))
+
+
+def test_group_by_exact_under_forced_h1_collision(monkeypatch):
+    """All h1 values collide; grouping must still be exact (h2 + byte
+    verification carry it)."""
+    import torch
+
+    from sail_amd.engine import joins
+
+    real_pair = joins.string_hash_pair
+
+    def collide_h1(c):
+        h1, h2 = real_pair(c)
+        return torch.zeros_like(h1), h2
+
+    monkeypatch.setattr(joins, "string_hash_pair", collide_h1)
+    s = _mk_session()
+    vals = [f"longish-key-value-{i % 37}" for i in range(1000)]
+    s.create_dataframe({"k": vals, "v": [1] * 1000}, name="fc1")
+    rows = s.sql("SELECT k, count(*) FROM fc1 GROUP BY k").collect()
+    assert len(rows) == 37
+    assert all(r[1] == (28 if int(r[0].split("-")[-1]) < 1000 % 37 else 27)
+               for r in rows)
+
+
+def test_group_by_exact_under_double_collision(monkeypatch):
+    """Both hash families collide (forced): detection kicks in and the host
+    fallback still produces exact groups."""
+    import torch
+
+    from sail_amd.engine import joins
+
+    def collide_both(c):
+        z = torch.zeros(len(c), dtype=torch.int64)
+        return z, z.clone()
+
+    monkeypatch.setattr(joins, "string_hash_pair", collide_both)
+    s = _mk_session()
+    vals = [f"another-long-key-{i % 11}" for i in range(330)]
+    s.create_dataframe({"k": vals, "v": [2] * 330}, name="fc2")
+    rows = s.sql("SELECT k, count(*) FROM fc2 GROUP BY k").collect()
+    assert len(rows) == 11 and all(r[1] == 30 for r in rows)
+
+
+def test_join_exact_under_forced_h1_collision(monkeypatch):
+    import torch
+
+    from sail_amd.engine import joins
+
+    real_pair = joins.string_hash_pair
+
+    def collide_h1(c):
+        h1, h2 = real_pair(c)
+        return torch.full_like(h1, 7), h2
+
+    monkeypatch.setattr(joins, "string_hash_pair", collide_h1)
+    s = _mk_session()
+    left = [f"join-key-string-{i}" for i in range(200)]
+    right = [f"join-key-string-{i}" for i in range(100, 300)]
+    s.create_dataframe({"k": left, "a": list(range(200))}, name="jl")
+    s.create_dataframe({"k": right, "b": list(range(200))}, name="jr")
+    rows = s.sql("SELECT jl.k, a, b FROM jl JOIN jr ON jl.k = jr.k").collect()
+    assert len(rows) == 100  # exactly the overlap, no hash-merged extras
+    for k, a, b in rows:
+        assert k == f"join-key-string-{a}" and a == b + 100

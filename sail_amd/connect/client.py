@@ -162,3 +162,306 @@ class ConnectClient:
         lvl = W.parse(W.first(fields, F.ANR_GET_STORAGE_LEVEL, b""))
         sl = W.parse(W.first(lvl, 1, b""))
         return bool(W.first_varint(sl, 2, 0))
+
+
+# ===========================================================================
+# Relation-tree (DataFrame API) client surface: builders producing the same
+# spark/connect Relation & Expression protos a PySpark client sends
+# (relations.proto / expressions.proto field numbers), plus reattach/release.
+# ===========================================================================
+
+class E:
+    """Expression proto builders (bytes)."""
+
+    @staticmethod
+    def col(name: str) -> bytes:
+        return W.field_message(2, W.field_string(1, name))
+
+    @staticmethod
+    def lit_long(v: int) -> bytes:
+        return W.field_message(1, W.field_varint(7, v))
+
+    @staticmethod
+    def lit_int(v: int) -> bytes:
+        return W.field_message(1, W.field_varint(6, v))
+
+    @staticmethod
+    def lit_str(s: str) -> bytes:
+        return W.field_message(1, W.field_string(13, s))
+
+    @staticmethod
+    def lit_double(x: float) -> bytes:
+        import struct
+
+        raw = struct.unpack("<Q", struct.pack("<d", x))[0]
+        return W.field_message(1, W.field_varint(11, raw))
+
+    @staticmethod
+    def fn(name: str, *args: bytes, distinct: bool = False) -> bytes:
+        body = W.field_string(1, name)
+        for a in args:
+            body += W.field_message(2, a)
+        if distinct:
+            body += W.field_varint(3, 1)
+        return W.field_message(3, body)
+
+    @staticmethod
+    def alias(child: bytes, name: str) -> bytes:
+        return W.field_message(6, W.field_message(1, child)
+                               + W.field_string(2, name))
+
+    @staticmethod
+    def star() -> bytes:
+        return W.field_message(5, b"")
+
+    @staticmethod
+    def cast(child: bytes, type_str: str) -> bytes:
+        return W.field_message(7, W.field_message(1, child)
+                               + W.field_string(3, type_str))
+
+    @staticmethod
+    def sort_order(child: bytes, asc: bool = True,
+                   nulls_first: Optional[bool] = None) -> bytes:
+        body = W.field_message(1, child) + W.field_varint(2, 1 if asc else 2)
+        if nulls_first is not None:
+            body += W.field_varint(3, 1 if nulls_first else 2)
+        return body
+
+
+class R:
+    """Relation proto builders (bytes)."""
+
+    @staticmethod
+    def read_table(name: str) -> bytes:
+        nt = W.field_message(1, W.field_string(1, name))
+        return W.field_message(2, nt)
+
+    @staticmethod
+    def read_source(fmt: str, paths, options=None) -> bytes:
+        ds = W.field_string(1, fmt)
+        for p in paths:
+            ds += W.field_string(4, p)
+        for k, v in (options or {}).items():
+            ds += W.field_message(3, W.field_string(1, k) + W.field_string(2, v))
+        return W.field_message(2, W.field_message(2, ds))
+
+    @staticmethod
+    def project(input_rel: bytes, *exprs: bytes) -> bytes:
+        body = W.field_message(1, input_rel)
+        for e in exprs:
+            body += W.field_message(3, e)
+        return W.field_message(3, body)
+
+    @staticmethod
+    def filter(input_rel: bytes, condition: bytes) -> bytes:
+        return W.field_message(4, W.field_message(1, input_rel)
+                               + W.field_message(2, condition))
+
+    @staticmethod
+    def join(left: bytes, right: bytes, how: int = 1,
+             condition: Optional[bytes] = None, using=None) -> bytes:
+        body = W.field_message(1, left) + W.field_message(2, right)
+        if condition is not None:
+            body += W.field_message(3, condition)
+        body += W.field_varint(4, how)
+        for c in (using or []):
+            body += W.field_string(5, c)
+        return W.field_message(5, body)
+
+    @staticmethod
+    def aggregate(input_rel: bytes, group: list, aggs: list) -> bytes:
+        body = W.field_message(1, input_rel) + W.field_varint(2, 1)
+        for g in group:
+            body += W.field_message(3, g)
+        for a in aggs:
+            body += W.field_message(4, a)
+        return W.field_message(9, body)
+
+    @staticmethod
+    def sort(input_rel: bytes, *orders: bytes) -> bytes:
+        body = W.field_message(1, input_rel)
+        for o in orders:
+            body += W.field_message(2, o)
+        return W.field_message(7, body)
+
+    @staticmethod
+    def limit(input_rel: bytes, n: int) -> bytes:
+        return W.field_message(8, W.field_message(1, input_rel)
+                               + W.field_varint(2, n))
+
+    @staticmethod
+    def set_op(left: bytes, right: bytes, kind: int, is_all: bool) -> bytes:
+        body = (W.field_message(1, left) + W.field_message(2, right)
+                + W.field_varint(3, kind) + W.field_varint(4, 1 if is_all else 0))
+        return W.field_message(6, body)
+
+    @staticmethod
+    def local_relation(table) -> bytes:
+        """pyarrow.Table -> LocalRelation (Arrow IPC bytes)."""
+        import io as _io
+
+        import pyarrow as pa
+
+        sink = _io.BytesIO()
+        with pa.ipc.new_stream(sink, table.schema) as w:
+            for b in table.to_batches():
+                w.write_batch(b)
+        return W.field_message(11, W.field_bytes(1, sink.getvalue()))
+
+    @staticmethod
+    def range(start: int, end: int, step: int = 1) -> bytes:
+        body = (W.field_varint(1, start) + W.field_varint(2, end)
+                + W.field_varint(3, step))
+        return W.field_message(15, body)
+
+    @staticmethod
+    def deduplicate(input_rel: bytes, columns=None, all_columns=False) -> bytes:
+        body = W.field_message(1, input_rel)
+        for c in (columns or []):
+            body += W.field_string(2, c)
+        if all_columns:
+            body += W.field_varint(3, 1)
+        return W.field_message(14, body)
+
+    @staticmethod
+    def with_columns(input_rel: bytes, aliases: list) -> bytes:
+        body = W.field_message(1, input_rel)
+        for al in aliases:
+            # aliases are Expression.Alias payloads (not wrapped Expression)
+            body += W.field_message(2, al)
+        return W.field_message(23, body)
+
+    @staticmethod
+    def alias_payload(child: bytes, name: str) -> bytes:
+        return W.field_message(1, child) + W.field_string(2, name)
+
+    @staticmethod
+    def drop(input_rel: bytes, *names: str) -> bytes:
+        body = W.field_message(1, input_rel)
+        for n in names:
+            body += W.field_string(3, n)
+        return W.field_message(21, body)
+
+    @staticmethod
+    def with_columns_renamed(input_rel: bytes, renames: dict) -> bytes:
+        body = W.field_message(1, input_rel)
+        for old, new in renames.items():
+            body += W.field_message(3, W.field_string(1, old)
+                                    + W.field_string(2, new))
+        return W.field_message(19, body)
+
+    @staticmethod
+    def show_string(input_rel: bytes, num_rows: int = 20,
+                    truncate: int = 20) -> bytes:
+        body = (W.field_message(1, input_rel) + W.field_varint(2, num_rows)
+                + W.field_varint(3, truncate))
+        return W.field_message(20, body)
+
+    @staticmethod
+    def subquery_alias(input_rel: bytes, alias: str) -> bytes:
+        return W.field_message(16, W.field_message(1, input_rel)
+                               + W.field_string(2, alias))
+
+    @staticmethod
+    def to_df(input_rel: bytes, *names: str) -> bytes:
+        body = W.field_message(1, input_rel)
+        for n in names:
+            body += W.field_string(2, n)
+        return W.field_message(18, body)
+
+    @staticmethod
+    def tail(input_rel: bytes, n: int) -> bytes:
+        return W.field_message(22, W.field_message(1, input_rel)
+                               + W.field_varint(2, n))
+
+
+class _Responses:
+    """Raw ExecutePlan responses with ids (for reattach tests)."""
+
+    def __init__(self, items):
+        self.items = items  # [(response_id, fields)]
+
+
+def _collect_batches(responses) -> "object":
+    import pyarrow as pa
+
+    batches = []
+    for resp in responses:
+        fields = W.parse(resp)
+        ab = W.first(fields, F.RESP_ARROW_BATCH)
+        if ab is not None:
+            abf = W.parse(ab)
+            data = W.first(abf, F.AB_DATA, b"")
+            with pa.ipc.open_stream(io.BytesIO(data)) as r:
+                for b in r:
+                    batches.append(b)
+    if not batches:
+        return pa.table({})
+    return pa.Table.from_batches(batches)
+
+
+def _client_relation_methods():
+    def execute_relation(self, rel_bytes: bytes, operation_id=None):
+        """Send a relation-tree plan; returns a pyarrow.Table."""
+        plan = W.field_message(F.PLAN_ROOT, rel_bytes)
+        req = (W.field_string(F.EXEC_SESSION_ID, self.session_id)
+               + W.field_message(F.EXEC_PLAN, plan)
+               + W.field_string(F.EXEC_OPERATION_ID,
+                                operation_id or str(uuid.uuid4())))
+        call = self._channel.unary_stream(
+            f"/{_SERVICE}/ExecutePlan",
+            request_serializer=None, response_deserializer=None)
+        return _collect_batches(call(req))
+
+    def execute_relation_raw(self, rel_bytes: bytes, operation_id: str,
+                             stop_after: Optional[int] = None):
+        """Like execute_relation but returns raw (response_id, msg) pairs;
+        optionally abandons the stream early (reattach testing)."""
+        plan = W.field_message(F.PLAN_ROOT, rel_bytes)
+        req = (W.field_string(F.EXEC_SESSION_ID, self.session_id)
+               + W.field_message(F.EXEC_PLAN, plan)
+               + W.field_string(F.EXEC_OPERATION_ID, operation_id))
+        call = self._channel.unary_stream(
+            f"/{_SERVICE}/ExecutePlan",
+            request_serializer=None, response_deserializer=None)
+        out = []
+        stream = call(req)
+        for resp in stream:
+            fields = W.parse(resp)
+            out.append((W.first_str(fields, F.RESP_RESPONSE_ID), resp))
+            if stop_after is not None and len(out) >= stop_after:
+                stream.cancel()
+                break
+        return out
+
+    def reattach(self, operation_id: str, last_response_id: Optional[str] = None):
+        req = (W.field_string(1, self.session_id)
+               + W.field_string(3, operation_id))
+        if last_response_id:
+            req += W.field_string(5, last_response_id)
+        call = self._channel.unary_stream(
+            f"/{_SERVICE}/ReattachExecute",
+            request_serializer=None, response_deserializer=None)
+        return [( W.first_str(W.parse(r), F.RESP_RESPONSE_ID), r)
+                for r in call(req)]
+
+    def release_until(self, operation_id: str, response_id: str):
+        req = (W.field_string(1, self.session_id)
+               + W.field_string(3, operation_id)
+               + W.field_message(6, W.field_string(1, response_id)))
+        self._call_unary("ReleaseExecute", req)
+
+    def release_all(self, operation_id: str):
+        req = (W.field_string(1, self.session_id)
+               + W.field_string(3, operation_id)
+               + W.field_message(5, b""))
+        self._call_unary("ReleaseExecute", req)
+
+    ConnectClient.execute_relation = execute_relation
+    ConnectClient.execute_relation_raw = execute_relation_raw
+    ConnectClient.reattach = reattach
+    ConnectClient.release_until = release_until
+    ConnectClient.release_all = release_all
+
+
+_client_relation_methods()

@@ -137,9 +137,9 @@ class SparkConnectServer:
             "AnalyzePlan": grpc.unary_unary_rpc_method_handler(self._analyze_plan),
             "Config": grpc.unary_unary_rpc_method_handler(self._config),
             "Interrupt": grpc.unary_unary_rpc_method_handler(self._ack),
-            "ReleaseExecute": grpc.unary_unary_rpc_method_handler(self._ack),
+            "ReleaseExecute": grpc.unary_unary_rpc_method_handler(self._release_execute),
             "ReleaseSession": grpc.unary_unary_rpc_method_handler(self._ack),
-            "ReattachExecute": grpc.unary_stream_rpc_method_handler(self._unimplemented_stream),
+            "ReattachExecute": grpc.unary_stream_rpc_method_handler(self._reattach_execute),
             "AddArtifacts": grpc.stream_unary_rpc_method_handler(self._unimplemented_unary),
             "ArtifactStatus": grpc.unary_unary_rpc_method_handler(self._unimplemented_unary),
         }
@@ -184,25 +184,65 @@ class SparkConnectServer:
             self._sessions[session_id] = ctx
 
     # -- RPC impls ---------------------------------------------------------
+    #: reattachable-execution buffers (ref: Executor/ExecutorBuffer,
+    #: crates/sail-spark-connect/src/executor.rs:31-98): responses are
+    #: retained per operation until ReleaseExecute trims them, so
+    #: ReattachExecute can replay after a dropped stream.
+    _BUFFER_CAP_BYTES = 64 << 20
+
+    def _buffer_key(self, session_id: str, op_id: str):
+        return (session_id, op_id)
+
+    def _buffered_yield(self, session_id, op_id, responses):
+        buf = {"responses": [], "bytes": 0, "complete": False}
+        with self._lock:
+            if not hasattr(self, "_op_buffers"):
+                self._op_buffers = {}
+            self._op_buffers[self._buffer_key(session_id, op_id)] = buf
+        for rid, msg in responses:
+            buf["responses"].append((rid, msg))
+            buf["bytes"] += len(msg)
+            while buf["bytes"] > self._BUFFER_CAP_BYTES and len(buf["responses"]) > 1:
+                _, old = buf["responses"].pop(0)
+                buf["bytes"] -= len(old)
+            yield msg
+        buf["complete"] = True
+
     def _execute_plan(self, request: bytes, context) -> Iterator[bytes]:
         req = W.parse(request)
         session_id = W.first_str(req, F.EXEC_SESSION_ID)
         op_id = W.first_str(req, F.EXEC_OPERATION_ID) or str(uuid.uuid4())
+        yield from self._buffered_yield(
+            session_id, op_id,
+            self._execute_plan_responses(req, session_id, op_id, context))
+
+    def _execute_plan_responses(self, req, session_id, op_id, context):
         plan = W.parse(W.first(req, F.EXEC_PLAN, b""))
         sess = self.session(session_id)
 
         sql = None
+        relation_plan = None
         is_command = False
         root = W.first(plan, F.PLAN_ROOT)
         cmd = W.first(plan, F.PLAN_COMMAND)
         if root is not None:
             rel = W.parse(root)
             sql_msg = W.first(rel, F.REL_SQL)
-            if sql_msg is None:
-                context.abort(grpc.StatusCode.UNIMPLEMENTED,
-                              "only Relation.sql plans are supported over the wire; "
-                              "use SQL or the native client API")
-            sql = W.first_str(W.parse(sql_msg), F.SQL_QUERY)
+            if sql_msg is not None:
+                sql = W.first_str(W.parse(sql_msg), F.SQL_QUERY)
+            else:
+                # relation-tree (DataFrame API) plan: proto -> spec ->
+                # normal resolve/optimize/execute pipeline
+                # (ref: crates/sail-spark-connect/src/proto/plan.rs)
+                from .relations import RelationConverter
+                from .relations import Unsupported as _RelUnsupported
+
+                try:
+                    spec_plan = RelationConverter(sess).convert(root)
+                    relation_plan = sess.optimize(sess.resolve(spec_plan))
+                except _RelUnsupported as e:
+                    context.abort(grpc.StatusCode.UNIMPLEMENTED,
+                                  f"relation not supported over the wire: {e}")
         elif cmd is not None:
             c = W.parse(cmd)
             sql_cmd = W.first(c, F.CMD_SQL)
@@ -218,15 +258,26 @@ class SparkConnectServer:
                 replace = bool(W.first(v, F.VIEW_REPLACE, 0))
                 plan_ = sess.plan_sql(sub_sql)
                 sess.catalog.create_view(name, plan_, replace=replace)
-                yield self._complete_response(session_id, op_id)
+                yield (f"{op_id}-0", self._complete_response(session_id, op_id))
                 return
             else:
                 context.abort(grpc.StatusCode.UNIMPLEMENTED, "unsupported command")
         else:
             context.abort(grpc.StatusCode.INVALID_ARGUMENT, "empty plan")
 
-        df = sess.sql(sql)
-        table = df.to_arrow() if df.plan.schema else None
+        if relation_plan is not None:
+            from ..datasource.arrow_io import chunk_to_arrow
+            from ..plan import spec as S
+
+            if isinstance(relation_plan, S.Command):
+                sess.execute_plan(relation_plan)
+                table = None
+            else:
+                chunk = sess.execute_plan(relation_plan)
+                table = chunk_to_arrow(chunk, relation_plan.schema)
+        else:
+            df = sess.sql(sql)
+            table = df.to_arrow() if df.plan.schema else None
         rid = 0
         if table is not None:
             import pyarrow as pa
@@ -241,9 +292,9 @@ class SparkConnectServer:
                        + W.field_message(F.RESP_ARROW_BATCH, ab)
                        + W.field_string(F.RESP_OPERATION_ID, op_id)
                        + W.field_string(F.RESP_RESPONSE_ID, f"{op_id}-{rid}"))
+                yield (f"{op_id}-{rid}", msg)
                 rid += 1
-                yield msg
-        yield self._complete_response(session_id, op_id, rid)
+        yield (f"{op_id}-{rid}", self._complete_response(session_id, op_id, rid))
 
     def _complete_response(self, session_id: str, op_id: str, rid: int = 0) -> bytes:
         return (W.field_string(F.RESP_SESSION_ID, session_id)
@@ -409,6 +460,49 @@ class SparkConnectServer:
             for key in u.get(1, []):
                 sess.conf.pop(key.decode(), None)
         return W.field_string(F.CFGR_SESSION_ID, session_id) + pairs_out
+
+    def _reattach_execute(self, request: bytes, context) -> Iterator[bytes]:
+        """Replay buffered responses after last_response_id (ref: the
+        reference's reattachable-execution state machine)."""
+        req = W.parse(request)
+        session_id = W.first_str(req, 1)
+        op_id = W.first_str(req, 3)
+        last = W.first_str(req, 5)
+        buf = getattr(self, "_op_buffers", {}).get(
+            self._buffer_key(session_id, op_id))
+        if buf is None:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "INVALID_HANDLE.OPERATION_NOT_FOUND")
+        start = 0
+        if last:
+            for i, (rid, _) in enumerate(buf["responses"]):
+                if rid == last:
+                    start = i + 1
+                    break
+        for _, msg in list(buf["responses"][start:]):
+            yield msg
+
+    def _release_execute(self, request: bytes, context) -> bytes:
+        req = W.parse(request)
+        session_id = W.first_str(req, 1)
+        op_id = W.first_str(req, 3)
+        key = self._buffer_key(session_id, op_id)
+        buf = getattr(self, "_op_buffers", {}).get(key)
+        if buf is not None:
+            until = W.first(req, 6)
+            if until is not None:
+                rid = W.first_str(W.parse(until), 1)
+                keep = buf["responses"]
+                for i, (r, _) in enumerate(keep):
+                    if r == rid:
+                        dropped = keep[:i + 1]
+                        buf["responses"] = keep[i + 1:]
+                        buf["bytes"] -= sum(len(m) for _, m in dropped)
+                        break
+            if W.first(req, 5) is not None:  # release_all
+                self._op_buffers.pop(key, None)
+        return (W.field_string(1, session_id)
+                + W.field_string(2, op_id))
 
     def _ack(self, request: bytes, context) -> bytes:
         return b""

@@ -75,6 +75,11 @@ class Resolver:
     def _scope(self, p: S.Plan, qualifier: Optional[str] = None, outer=None) -> Scope:
         return Scope([Field(n, t, qualifier) for n, t in p.schema], outer)
 
+    def _p_ChunkSource(self, p: S.ChunkSource, outer):
+        # pre-materialized leaf (Connect LocalRelation / tail results):
+        # schema is already known
+        return p
+
     def _p_Read(self, p: S.Read, outer):
         # working-set reference inside a recursive CTE body?
         if p.table.lower() in self.recursion_refs:

@@ -94,3 +94,156 @@ def test_analyze_extended_types(client):
     assert client.get_storage_level("SELECT 1") is True
     client.persist("SELECT 1")  # no-op ack
     assert client.input_files("SELECT 1") == []
+
+
+# -- relation-tree (DataFrame API) plans over the wire ----------------------
+# (PySpark DataFrame clients send Relation protos, not SQL;
+#  ref: crates/sail-spark-connect/src/proto/plan.rs)
+
+from sail_amd.connect.client import E, R
+
+
+@pytest.fixture()
+def rel_data(server, client):
+    sess = server.session(client.session_id)
+    sess.create_dataframe(
+        {"k": ["a", "b", "a", "c", "b", "a"],
+         "v": [1, 2, 3, 4, 5, 6],
+         "w": [1.5, 2.5, 3.5, 4.5, 5.5, 6.5]}, name="rt")
+    sess.create_dataframe(
+        {"k": ["a", "b", "z"], "tag": [10, 20, 30]}, name="rt2")
+    return sess
+
+
+def test_relation_project_filter(client, rel_data):
+    rel = R.project(
+        R.filter(R.read_table("rt"),
+                 E.fn(">", E.col("v"), E.lit_long(2))),
+        E.col("k"), E.alias(E.fn("*", E.col("v"), E.lit_long(10)), "v10"))
+    t = client.execute_relation(rel)
+    assert t.column_names == ["k", "v10"]
+    assert t.to_pydict() == {"k": ["a", "c", "b", "a"], "v10": [30, 40, 50, 60]}
+
+
+def test_relation_aggregate_sort(client, rel_data):
+    rel = R.sort(
+        R.aggregate(R.read_table("rt"),
+                    group=[E.col("k")],
+                    aggs=[E.alias(E.fn("sum", E.col("v")), "sv"),
+                          E.alias(E.fn("count", E.col("v")), "c")]),
+        E.sort_order(E.col("k")))
+    t = client.execute_relation(rel)
+    assert t.to_pydict() == {"k": ["a", "b", "c"], "sv": [10, 7, 4],
+                             "c": [3, 2, 1]}
+
+
+def test_relation_join_limit(client, rel_data):
+    rel = R.limit(
+        R.sort(
+            R.join(R.read_table("rt"), R.read_table("rt2"), how=1,
+                   using=["k"]),
+            E.sort_order(E.col("v"))),
+        3)
+    t = client.execute_relation(rel)
+    assert t.num_rows == 3
+    assert t.to_pydict()["tag"] == [10, 20, 10]
+
+
+def test_relation_local_and_setop(client):
+    import pyarrow as pa
+
+    local = R.local_relation(pa.table({"x": [1, 2, 3]}))
+    local2 = R.local_relation(pa.table({"x": [3, 4]}))
+    rel = R.sort(R.set_op(local, local2, kind=2, is_all=True),
+                 E.sort_order(E.col("x")))
+    t = client.execute_relation(rel)
+    assert t.to_pydict() == {"x": [1, 2, 3, 3, 4]}
+
+
+def test_relation_range_withcolumns_drop_rename(client):
+    rel = R.range(0, 5)
+    rel = R.with_columns(rel, [R.alias_payload(
+        E.fn("*", E.col("id"), E.lit_long(2)), "dbl")])
+    rel = R.with_columns_renamed(rel, {"dbl": "double_id"})
+    rel = R.drop(rel, "id")
+    t = client.execute_relation(rel)
+    assert t.column_names == ["double_id"]
+    assert t.to_pydict() == {"double_id": [0, 2, 4, 6, 8]}
+
+
+def test_relation_dedup_and_todf(client, rel_data):
+    rel = R.to_df(R.deduplicate(R.project(R.read_table("rt"), E.col("k")),
+                                all_columns=True), "key")
+    t = client.execute_relation(rel)
+    assert t.column_names == ["key"]
+    assert sorted(t.to_pydict()["key"]) == ["a", "b", "c"]
+    # subset dedup keeps one row per key with all columns
+    rel2 = R.deduplicate(R.read_table("rt"), columns=["k"])
+    t2 = client.execute_relation(rel2)
+    assert t2.num_rows == 3 and set(t2.column_names) == {"k", "v", "w"}
+
+
+def test_relation_show_string_and_tail(client, rel_data):
+    rel = R.show_string(R.sort(R.read_table("rt"), E.sort_order(E.col("v"))),
+                        num_rows=3)
+    t = client.execute_relation(rel)
+    text = t.to_pydict()["show_string"][0]
+    assert "only showing top 3 rows" in text and "| k" in text
+    t2 = client.execute_relation(
+        R.tail(R.sort(R.read_table("rt"), E.sort_order(E.col("v"))), 2))
+    assert t2.to_pydict()["v"] == [5, 6]
+
+
+def test_relation_cast_and_distinct_agg(client, rel_data):
+    rel = R.aggregate(
+        R.read_table("rt"), group=[],
+        aggs=[E.alias(E.fn("count", E.col("k"), distinct=True), "dk"),
+              E.alias(E.fn("sum", E.cast(E.col("w"), "bigint")), "sw")])
+    t = client.execute_relation(rel)
+    assert t.to_pydict() == {"dk": [3], "sw": [21]}
+
+
+def test_reattach_and_release(server, client, rel_data):
+    """Dropped result stream -> ReattachExecute replays from the last seen
+    response id; ReleaseExecute(until) trims; release_all forgets the op."""
+    import uuid as _uuid
+
+    op = str(_uuid.uuid4())
+    # big enough for several 65536-row arrow batches
+    sess = server.session(client.session_id)
+    sess.create_dataframe({"n": list(range(200_000))}, name="big_rt")
+    rel = R.read_table("big_rt")
+    got = client.execute_relation_raw(rel, op, stop_after=1)
+    assert len(got) == 1
+    first_rid = got[0][0]
+    # reattach after the first response: the rest of the batches arrive
+    rest = client.reattach(op, first_rid)
+    assert len(rest) >= 1
+    all_ids = [r[0] for r in got + rest]
+    assert len(all_ids) == len(set(all_ids))  # no duplicates
+    total = 0
+    import io as _io
+
+    import pyarrow as pa
+
+    from sail_amd.connect import wire as W2
+    from sail_amd.connect.server import F as F2
+
+    for _, resp in got + rest:
+        fields = W2.parse(resp)
+        ab = W2.first(fields, F2.RESP_ARROW_BATCH)
+        if ab is not None:
+            data = W2.first(W2.parse(ab), F2.AB_DATA, b"")
+            with pa.ipc.open_stream(_io.BytesIO(data)) as r:
+                for b in r:
+                    total += b.num_rows
+    assert total == 200_000
+    # release everything up to the first id, reattach returns only later ones
+    client.release_until(op, first_rid)
+    later = client.reattach(op, None)
+    assert first_rid not in [r[0] for r in later]
+    client.release_all(op)
+    import grpc as _grpc
+
+    with pytest.raises(_grpc.RpcError):
+        client.reattach(op, None)

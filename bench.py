@@ -141,9 +141,13 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
+    progress = os.environ.get("SAIL_BENCH_PROGRESS") == "1"
+
     def one_step(collect_times=False):
         times = {}
         for q in qids:
+            if progress and rank == 0:
+                print(f"# start q{q}", file=sys.stderr, flush=True)
             tq = time.time()
             if q in plans:
                 chunk = session.execute_plan(plans[q])

@@ -551,6 +551,7 @@ def register_tpch_parquet(session, sf: float = 0.01, device=None, rank: int = 0,
             format="parquet", paths=[paths[name]],
             options={"partitioning": "sharded" if sharded else "replicated"})
         node.schema = schema
+        node.__dict__["_table_name"] = name  # planner statistics key
         with cat._lock:
             k = cat._key(name)
             cat._tables.pop(k, None)

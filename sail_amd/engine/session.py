@@ -333,7 +333,12 @@ class SessionContext:
         from .executor import ExecutionContext, Executor
 
         ctx = ExecutionContext(self, self.device)
-        out = Executor(ctx).execute(plan)
+        ex = Executor(ctx)
+        out = ex.execute(plan)
+        # a top-level sharded result (shuffled distinct/window/aggregate)
+        # must be replicated so every rank's collect() sees all rows
+        if ex.dctx is not None and out.partitioning == "sharded":
+            out = ex._gather(out)
         if getattr(ctx, "tracer", None) is not None:
             # published AFTER execution so `SELECT * FROM system_operators`
             # reads the previous query's trace, not its own empty one

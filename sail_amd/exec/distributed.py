@@ -185,9 +185,17 @@ def shuffle_chunk(chunk: Chunk, key_idx, d: DistContext) -> Chunk:
     """Repartition a sharded chunk by hash of the key columns: one
     all_to_all per column payload (few large messages for xGMI's
     point-to-point links, vs the reference's many Flight streams)."""
+    pids = partition_ids([chunk.columns[i] for i in key_idx], d.world)
+    return shuffle_chunk_by_pids(chunk, pids, d)
+
+
+def shuffle_chunk_by_pids(chunk: Chunk, pids: torch.Tensor, d: DistContext) -> Chunk:
+    """Repartition a sharded chunk by an explicit per-row destination-rank
+    tensor (hash partitioning for joins/aggregates, RANGE partitioning for
+    the distributed sort). The stable argsort keeps same-destination rows in
+    their rank-local order so ties stay deterministic."""
     world = d.world
-    pids = partition_ids([chunk.columns[i] for i in key_idx], world)
-    order = torch.argsort(pids)
+    order = torch.argsort(pids, stable=True)
     send_counts = torch.bincount(pids, minlength=world).to(torch.int64)
     out_cols = []
     for c in chunk.columns:
@@ -211,6 +219,31 @@ def shuffle_chunk(chunk: Chunk, key_idx, d: DistContext) -> Chunk:
                                    _exchange_validity(c, order, send_counts, d)))
     out = Chunk(out_cols, list(chunk.names), "sharded")
     return out
+
+
+def range_partition_ids(keys: torch.Tensor, d: DistContext,
+                        descending: bool = False) -> torch.Tensor:
+    """Destination ranks for the range-partitioned distributed sort:
+    stride-sampled splitters (all_gathered, so identical on every rank);
+    equal keys always map to one rank, which keeps multi-key ties local.
+    ref: the reference sorts per-partition then SortPreservingMerge
+    (sail-physical-optimizer EnforceSorting); here ranks own disjoint key
+    ranges so a rank-order concat IS the global order."""
+    n = keys.numel()
+    step = max(n // 4096, 1)
+    sample = keys[::step].contiguous()
+    allsamp = torch.cat(d.all_gather_tensors(sample))
+    allsamp, _ = torch.sort(allsamp)
+    m = allsamp.numel()
+    if m == 0 or d.world <= 1:
+        return torch.zeros(n, dtype=torch.int64, device=keys.device)
+    idxs = torch.tensor([min((m * r) // d.world, m - 1)
+                         for r in range(1, d.world)], device=allsamp.device)
+    splitters = allsamp.index_select(0, idxs).contiguous()
+    pids = torch.searchsorted(splitters, keys)
+    if descending:
+        pids = (d.world - 1) - pids
+    return pids
 
 
 def _exchange_validity(c, order, send_counts, d):

@@ -240,6 +240,8 @@ def _prune(plan: S.Plan, needed: Set[int]) -> Tuple[S.Plan, Dict[int, int]]:
                               options=plan.options,
                               user_schema=plan.user_schema)
         nd.schema = [plan.schema[i] for i in keep]
+        if "_table_name" in plan.__dict__:
+            nd.__dict__["_table_name"] = plan.__dict__["_table_name"]
         return nd, {old: new for new, old in enumerate(keep)}
 
     if isinstance(plan, (S.Read, S.LocalRelation, S.Range)):

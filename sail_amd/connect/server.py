@@ -199,14 +199,21 @@ class SparkConnectServer:
             if not hasattr(self, "_op_buffers"):
                 self._op_buffers = {}
             self._op_buffers[self._buffer_key(session_id, op_id)] = buf
+        # drain the execution fully BEFORE streaming: a client cancelling
+        # the stream must not abort production, or ReattachExecute could
+        # never replay the rest (ref: the reference's Executor runs the
+        # result pull in a spawned task independent of the gRPC stream)
+        produced = []
         for rid, msg in responses:
+            produced.append(msg)
             buf["responses"].append((rid, msg))
             buf["bytes"] += len(msg)
             while buf["bytes"] > self._BUFFER_CAP_BYTES and len(buf["responses"]) > 1:
                 _, old = buf["responses"].pop(0)
                 buf["bytes"] -= len(old)
-            yield msg
         buf["complete"] = True
+        for msg in produced:
+            yield msg
 
     def _execute_plan(self, request: bytes, context) -> Iterator[bytes]:
         req = W.parse(request)

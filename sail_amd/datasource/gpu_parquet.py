@@ -290,9 +290,11 @@ _PHYS_TORCH = {"INT32": torch.int32, "INT64": torch.int64,
 
 
 class _ColumnDecoder:
-    """Decodes one column of one file into device buffers."""
+    """Decodes one column of one file into device buffers. `rg_window`
+    restricts to row groups [lo, hi) — the out-of-core scan streams
+    batches of row groups instead of whole files."""
 
-    def __init__(self, idx: FileIndex, name: str, device):
+    def __init__(self, idx: FileIndex, name: str, device, rg_window=None):
         self.idx = idx
         self.device = device
         self.ci = idx.column_index(name)
@@ -300,6 +302,8 @@ class _ColumnDecoder:
         self.physical = self.sc.physical_type
         self.flba_w = self.sc.length or 0
         self.chunks = idx.chunks(self.ci)
+        if rg_window is not None:
+            self.chunks = self.chunks[rg_window[0]:rg_window[1]]
         self.pages = [p for ch in self.chunks for p in ch.pages]
         self.nrows = sum(p.nvals for p in self.pages)
         encs = {p.enc for p in self.pages}
@@ -651,7 +655,7 @@ class _ColumnDecoder:
         raise Unsupported(f"{self.sc.name}: arrow type {at}")
 
 
-def read_gpu(files: List[str], schema, device) -> Table:
+def read_gpu(files: List[str], schema, device, rg_window=None) -> Table:
     """Decode `schema`'s columns of the given parquet files on the GPU.
     Raises Unsupported when any file/column needs the host fallback."""
     if not str(device).startswith("cuda") and not _ALLOW_CPU:
@@ -665,7 +669,7 @@ def read_gpu(files: List[str], schema, device) -> Table:
         names = [n for n, _ in schema] if schema else [
             idx.schema.column(i).name for i in range(len(idx.schema))]
         for n in names:
-            dec = _ColumnDecoder(idx, n, device)
+            dec = _ColumnDecoder(idx, n, device, rg_window=rg_window)
             col, _ = dec.decode()
             cols[n] = col
         per_file.append(cols)

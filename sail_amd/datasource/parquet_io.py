@@ -192,3 +192,35 @@ def write(path: str, chunk, mode: str, options: Dict[str, str]):
     compression = options.get("compression", "snappy")
     pq.write_table(tbl, target, compression=compression)
     return target
+
+
+def scan_batches(paths: List[str], schema, device, options: Dict[str, str],
+                 target_rows: int = 8_000_000):
+    """Out-of-core scan: yield engine Tables in bounded row batches (one or
+    more row groups at a time) instead of materializing whole files — the
+    spill-free analogue of the reference's bounded-memory streams
+    (ref: application.yaml runtime memory pools / execution.batch_size).
+    Uses the GPU page decoder per batch when supported."""
+    files = _expand(paths)
+    mode = _gpu_mode(options)
+    use_gpu = mode not in ("off", "false", "0") and str(device).startswith("cuda")
+    from . import gpu_parquet
+
+    for f in files:
+        pf = pq.ParquetFile(f)
+        nrg = pf.metadata.num_row_groups
+        rows_per_rg = max(pf.metadata.num_rows // max(nrg, 1), 1)
+        step = max(1, target_rows // rows_per_rg)
+        for lo in range(0, nrg, step):
+            hi = min(lo + step, nrg)
+            if use_gpu:
+                try:
+                    yield gpu_parquet.read_gpu([f], schema, device,
+                                               rg_window=(lo, hi))
+                    continue
+                except gpu_parquet.Unsupported:
+                    if mode == "force":
+                        raise
+            cols = [n for n, _ in schema] if schema else None
+            tbl = pf.read_row_groups(list(range(lo, hi)), columns=cols)
+            yield arrow_to_table(tbl, device=device)

@@ -499,9 +499,21 @@ class Executor:
         return out
 
     # -- aggregate ---------------------------------------------------------
+    #: scans above this many file bytes stream through row-group batches
+    #: with a partial/merge aggregation state instead of materializing the
+    #: whole table in HBM (out-of-core; ref: the
+    #: reference's memory pools + spill, application.yaml:21-67)
+    STREAM_SCAN_BYTES = int(os.environ.get("SAIL_EXEC_STREAM_SCAN_BYTES",
+                                           str(150 << 30)))
+    STREAM_SCAN_BATCH_ROWS = int(os.environ.get(
+        "SAIL_EXEC_STREAM_SCAN_BATCH_ROWS", "16000000"))
+
     def _x_Aggregate(self, p: S.Aggregate) -> Chunk:
         if p.grouping_sets:
             return self._grouping_sets_aggregate(p)
+        streamed = self._try_streamed_scan_aggregate(p)
+        if streamed is not None:
+            return streamed
         # Aggregate∘[Project]∘Filter fusion: evaluate group keys and agg
         # inputs on the UNFILTERED child and pass the selection mask into the
         # aggregation — avoids materializing high-selectivity filters (Q1
@@ -551,6 +563,70 @@ class Executor:
             agg_cols = [agg_eval(a.name, args, gid, ng, a.distinct, fmask, a.dtype)
                         for a, args, fmask in zip(p.aggs, args_list, fmasks)]
         return Chunk(out_keys + agg_cols, [nm for nm, _ in p.schema])
+
+    def _try_streamed_scan_aggregate(self, p: S.Aggregate) -> Optional[Chunk]:
+        """Aggregate over a parquet scan whose bytes exceed the streaming
+        budget: decode row-group batches, fold each into a partial/merge
+        aggregation state (streaming's _AggState — the identical
+        decomposition), finalize once. Bounded memory: one batch + the
+        group state resident at a time."""
+        if self.dctx is not None:  # SPMD path merges partials its own way
+            return None
+        node = p.input
+        chain = []
+        while isinstance(node, (S.Project, S.Filter, S.SubqueryAlias)):
+            if node.__dict__.get("_cte_cache_key") is not None:
+                return None  # shared subtree: caching assumes one execution
+            chain.append(node)
+            node = node.input
+        if not isinstance(node, S.DataSourceRead) or node.format != "parquet":
+            return None
+        if node.__dict__.get("_cte_cache_key") is not None:
+            return None
+        import glob as _g
+        import os as _os
+
+        files = []
+        for pth in node.paths or []:
+            if _os.path.isdir(pth):
+                files += _g.glob(_os.path.join(pth, "**", "*.parquet"),
+                                 recursive=True)
+            elif _os.path.isfile(pth):
+                files.append(pth)
+        total_bytes = sum(_os.path.getsize(f) for f in files)
+        if not files or total_bytes < self.STREAM_SCAN_BYTES:
+            return None
+        from ..exec.distributed import decompose_agg
+
+        decomps = [decompose_agg(a) for a in p.aggs]
+        if any(d is None for d in decomps) or \
+                any(getattr(a, "distinct", False) for a in p.aggs):
+            return None
+        from ..datasource.parquet_io import scan_batches
+        from ..streaming.query import _AggState
+
+        st = _AggState(p, decomps)
+        parent = chain[-1] if chain else None
+        orig_input = parent.input if parent is not None else None
+        finalized = None
+        try:
+            for table in scan_batches(node.paths, node.schema, self.ctx.device,
+                                      node.options,
+                                      target_rows=self.STREAM_SCAN_BATCH_ROWS):
+                src = S.ChunkSource(chunk=Chunk.from_table(table),
+                                    schema=node.schema)
+                if parent is not None:
+                    parent.input = src
+                    child = self.execute(p.input)
+                else:
+                    child = self.execute(src)
+                finalized, _ = st.update(self, child)
+        finally:
+            if parent is not None:
+                parent.input = orig_input
+        if finalized is None:  # zero batches: fall back to the normal path
+            return None
+        return finalized
 
     def _grouping_sets_aggregate(self, p: S.Aggregate) -> Chunk:
         """ROLLUP/CUBE/GROUPING SETS: one aggregation per set, absent keys

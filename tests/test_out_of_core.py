@@ -1,0 +1,83 @@
+"""Out-of-core streamed scan-aggregate: aggregates over
+a parquet scan bigger than the streaming budget run row-group-batch-wise
+with a partial/merge state and bounded memory, producing identical results
+to whole-table execution."""
+import os
+
+import pytest
+
+import sail_amd
+from sail_amd.engine import types as T
+
+
+@pytest.fixture()
+def big_parquet(tmp_path):
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    rng = np.random.default_rng(3)
+    n = 200_000
+    t = pa.table({
+        "k": pa.array((np.arange(n) % 7).astype("int64")),
+        "v": pa.array(rng.integers(0, 1000, n)),
+        "w": pa.array(rng.standard_normal(n)),
+        "flag": pa.array(np.array(["x", "y"])[np.arange(n) % 2]),
+    })
+    p = str(tmp_path / "big.parquet")
+    pq.write_table(t, p, compression="NONE", row_group_size=10_000,
+                   data_page_version="1.0")
+    return p
+
+
+def _fresh_session():
+    return sail_amd.SessionContext(device="cpu")
+
+
+def test_streamed_scan_aggregate_matches_whole(big_parquet, monkeypatch):
+    from sail_amd.engine.executor import Executor
+
+    sql = (f"SELECT k, count(*) c, sum(v) sv, avg(w) aw, min(v) mv, max(v) xv "
+           f"FROM parquet.`{big_parquet}` WHERE flag = 'x' GROUP BY k ORDER BY k")
+    want = _fresh_session().sql(sql).collect()
+
+    calls = {"n": 0}
+    import sail_amd.datasource.parquet_io as pio
+
+    orig = pio.scan_batches
+
+    def counting(*a, **kw):
+        for t in orig(*a, **kw):
+            calls["n"] += 1
+            yield t
+
+    monkeypatch.setattr(pio, "scan_batches", counting)
+    monkeypatch.setattr(Executor, "STREAM_SCAN_BYTES", 1)  # force streaming
+    monkeypatch.setattr(Executor, "STREAM_SCAN_BATCH_ROWS", 25_000)
+    got = _fresh_session().sql(sql).collect()
+    assert calls["n"] >= 8  # genuinely ran in several bounded batches
+    assert len(got) == len(want)
+    for g, w in zip(got, want):
+        assert g[0] == w[0] and g[1] == w[1] and g[2] == w[2]
+        assert g[3] == pytest.approx(w[3]) and g[4] == w[4] and g[5] == w[5]
+
+
+def test_streamed_global_aggregate(big_parquet, monkeypatch):
+    from sail_amd.engine.executor import Executor
+
+    sql = f"SELECT count(*), sum(v), avg(w) FROM parquet.`{big_parquet}`"
+    want = _fresh_session().sql(sql).collect()
+    monkeypatch.setattr(Executor, "STREAM_SCAN_BYTES", 1)
+    monkeypatch.setattr(Executor, "STREAM_SCAN_BATCH_ROWS", 30_000)
+    got = _fresh_session().sql(sql).collect()
+    assert got[0][0] == want[0][0] and got[0][1] == want[0][1]
+    assert got[0][2] == pytest.approx(want[0][2])
+
+
+def test_streamed_distinct_agg_falls_back(big_parquet, monkeypatch):
+    """count(DISTINCT) has no partial decomposition: whole-table path."""
+    from sail_amd.engine.executor import Executor
+
+    monkeypatch.setattr(Executor, "STREAM_SCAN_BYTES", 1)
+    sql = f"SELECT count(DISTINCT k) FROM parquet.`{big_parquet}`"
+    assert _fresh_session().sql(sql).collect() == [(7,)]

@@ -192,36 +192,75 @@ def file_index(path: str) -> FileIndex:
 
 # -- pinned staging ----------------------------------------------------------
 
-_PINNED: Optional[torch.Tensor] = None
+
+class _StagingPool:
+    """Two pinned host buffers in rotation. The async H2D copy out of a
+    buffer must complete before the NEXT column's file read overwrites it;
+    the recorded event enforces that while still letting the CPU read
+    column i+1 during column i's device decode."""
+
+    def __init__(self):
+        self.bufs = [None, None]
+        self.events = [None, None]
+        self.i = 0
+
+    def acquire(self, nbytes: int):
+        i = self.i
+        self.i ^= 1
+        ev = self.events[i]
+        if ev is not None:
+            ev.synchronize()
+            self.events[i] = None
+        if self.bufs[i] is None or self.bufs[i].numel() < nbytes:
+            cap = max(nbytes, 64 << 20)
+            pin = torch.cuda.is_available()
+            self.bufs[i] = torch.empty(cap, dtype=torch.uint8, pin_memory=pin)
+        return i, self.bufs[i]
+
+    def mark_uploaded(self, i: int):
+        if torch.cuda.is_available():
+            ev = torch.cuda.Event()
+            ev.record()
+            self.events[i] = ev
 
 
-def _staging(nbytes: int) -> torch.Tensor:
-    global _PINNED
-    if _PINNED is None or _PINNED.numel() < nbytes:
-        cap = max(nbytes, 64 << 20)
-        pin = torch.cuda.is_available()
-        _PINNED = torch.empty(cap, dtype=torch.uint8, pin_memory=pin)
-    return _PINNED
+_STAGING = _StagingPool()
 
 
 def _upload_ranges(path: str, ranges: List[Tuple[int, int]], device):
     """Read file byte ranges into pinned staging, one H2D copy; returns
     (device u8 tensor, [staging offset per range])."""
     total = sum(e - s for s, e in ranges)
-    stage = _staging(total + 16)  # +16: aligned-word kernels may read past
-    view = stage.numpy()          # the last page's bytes (funnel shift)
+    slot, stage = _STAGING.acquire(total + 16)  # +16: aligned-word kernels
+    view = stage.numpy()                        # may read past the last page
     offs = []
     pos = 0
-    with open(path, "rb") as f:
-        for s, e in ranges:
+    jobs = []
+    for s, e in ranges:
+        jobs.append((s, e, pos))
+        offs.append(pos)
+        pos += e - s
+    # parallel pread into pinned staging: readinto releases the GIL, so
+    # page-cache-warm scans move at memory bandwidth instead of one core
+    from concurrent.futures import ThreadPoolExecutor
+
+    def _one(job):
+        s, e, at = job
+        with open(path, "rb") as f:
             f.seek(s)
-            got = f.readinto(memoryview(view)[pos:pos + (e - s)])
-            if got != e - s:
-                raise IOError(f"short read in {path}")
-            offs.append(pos)
-            pos += e - s
+            got = f.readinto(memoryview(view)[at:at + (e - s)])
+        if got != e - s:
+            raise IOError(f"short read in {path}")
+
+    if len(jobs) > 1 and total > (64 << 20):
+        with ThreadPoolExecutor(max_workers=min(8, len(jobs))) as pool:
+            list(pool.map(_one, jobs))
+    else:
+        for j in jobs:
+            _one(j)
     view[total:total + 16] = 0
     dev = stage[:total + 16].to(device, non_blocking=True)
+    _STAGING.mark_uploaded(slot)
     return dev, offs
 
 

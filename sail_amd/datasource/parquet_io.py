@@ -21,6 +21,13 @@ from .arrow_io import arrow_to_table, chunk_to_arrow
 def _expand(paths: List[str]) -> List[str]:
     out = []
     for p in paths:
+        if "://" in p and not p.startswith("file://"):
+            from ..storage.object_store import expand_to_local
+
+            out.extend(expand_to_local(p))
+            continue
+        if p.startswith("file://"):
+            p = p[len("file://"):]
         if os.path.isdir(p):
             out.extend(sorted(_glob.glob(os.path.join(p, "**", "*.parquet"), recursive=True)))
         elif any(ch in p for ch in "*?["):
@@ -176,6 +183,22 @@ def read(paths: List[str], schema, device, options: Dict[str, str]):
 def write(path: str, chunk, mode: str, options: Dict[str, str]):
     schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
     tbl = chunk_to_arrow(chunk, schema)
+    if "://" in path and not path.startswith("file://"):
+        # object-store write: encode to a buffer, put via the registry
+        from ..storage.object_store import global_registry
+
+        store, sp = global_registry().for_uri(path)
+        if not sp.endswith(".parquet"):
+            sp = sp.rstrip("/") + "/part-00000.parquet"
+        if mode == "error" and store.exists(sp):
+            raise FileExistsError(path)
+        import io as _io
+
+        sink = _io.BytesIO()
+        pq.write_table(tbl, sink,
+                       compression=options.get("compression", "snappy"))
+        store.write_bytes(sp, sink.getvalue())
+        return path
     if os.path.isdir(path) or path.endswith("/"):
         os.makedirs(path, exist_ok=True)
         target = os.path.join(path, "part-00000.parquet")

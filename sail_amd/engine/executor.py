@@ -1089,7 +1089,15 @@ class Executor:
                 view = S.DataSourceRead(format=fmt, paths=[path])
                 view.schema = [(n, c.dtype) for n, c in
                                zip(data.names, data.columns)]
+                view.__dict__["_table_name"] = p.name
                 sess.catalog.create_view(p.name, view, replace=True)
+                prov = getattr(sess.catalog, "persistent", None)
+                if prov is not None:
+                    from ..catalogs.persistent import TableDef
+
+                    prov.create_table(TableDef(
+                        p.name, fmt, path, schema=view.schema,
+                        options=dict(p.options or {})), replace=True)
             return Chunk([], [])
         if p.input is not None:
             data = self.execute(p.input)
@@ -1099,7 +1107,11 @@ class Executor:
         return Chunk([], [])
 
     def _x_DropTable(self, p: S.DropTable) -> Chunk:
-        self.ctx.session.catalog.drop(p.name, if_exists=p.if_exists)
+        cat = self.ctx.session.catalog
+        cat.drop(p.name, if_exists=p.if_exists)
+        prov = getattr(cat, "persistent", None)
+        if prov is not None:
+            prov.drop_table(p.name.split(".")[-1], if_exists=True)
         return Chunk([], [])
 
     def _x_InsertInto(self, p: S.InsertInto) -> Chunk:

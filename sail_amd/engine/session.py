@@ -7,6 +7,7 @@ tables live as device-resident Tables (or lazy providers) in the catalog.
 """
 from __future__ import annotations
 
+import os
 import threading
 from typing import Callable, Dict, List, Optional, Tuple
 
@@ -207,6 +208,14 @@ class SessionContext:
         self.query_log: List[dict] = []
         self._streams = None
         self._register_system_tables()
+        # durable catalog (ref: sail-catalog providers): table definitions
+        # under sail.catalog.path survive sessions
+        cat_path = self.conf.get("sail.catalog.path") or \
+            os.environ.get("SAIL_CATALOG_PATH", "")
+        if cat_path:
+            from ..catalogs.persistent import attach
+
+            attach(self, cat_path)
 
     @property
     def read_stream(self):

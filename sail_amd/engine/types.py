@@ -205,6 +205,20 @@ _BY_NAME = {
 }
 
 
+def type_name(t: DataType) -> str:
+    """Canonical SQL name (inverse of type_from_name)."""
+    if isinstance(t, DecimalType):
+        return f"decimal({t.precision},{t.scale})"
+    m = {NullType: "void", BooleanType: "boolean", Int8Type: "tinyint",
+         Int16Type: "smallint", Int32Type: "int", Int64Type: "bigint",
+         Float32Type: "float", Float64Type: "double", DateType: "date",
+         TimestampType: "timestamp", StringType: "string",
+         BinaryType: "binary"}
+    if type(t) in m:
+        return m[type(t)]
+    return str(t)
+
+
 def type_from_name(name: str) -> DataType:
     base = name.strip().lower()
     if base.startswith("decimal") or base.startswith("numeric"):

@@ -475,21 +475,20 @@ class _ColumnDecoder:
         if delta_rows:
             if self.physical not in ("INT32", "INT64"):
                 raise Unsupported(f"{self.sc.name}: DELTA on {self.physical}")
+            table = _page_table(delta_rows, dev)
             deltas, _ = ext.pq_delta_decode(
-                self.buf, _page_table(delta_rows, dev),
+                self.buf, table,
                 int(np.max([r[3] + r[2] for r in delta_rows])))
-            # compact the delta segments then segmented-cumsum
-            segs = [deltas[r[3]:r[3] + r[2]] for r in delta_rows]
-            vals = _segmented_cumsum(torch.cat(segs), delta_counts)
+            # finish [first, d1, ...] -> values with the on-device per-page
+            # scan (replaces torch.cumsum + cat + correction gathers)
+            ext.pq_segscan(deltas, table)
             if len(delta_rows) == pi:
-                dense = vals.to(_PHYS_TORCH[self.physical]) \
-                    if self.physical == "INT32" else vals
+                dense = deltas.to(_PHYS_TORCH[self.physical]) \
+                    if self.physical == "INT32" else deltas
             else:
                 o = _out()
-                at = 0
                 for r, c in zip(delta_rows, delta_counts):
-                    o[r[3]:r[3] + c] = vals[at:at + c].to(o.dtype)
-                    at += c
+                    o[r[3]:r[3] + c] = deltas[r[3]:r[3] + c].to(o.dtype)
 
         if dict_rows:
             codes = self._decode_dict_codes(ext, dict_rows)
@@ -604,16 +603,30 @@ class _ColumnDecoder:
                     pi += 1
             deltas, data_end = ext.pq_delta_decode(
                 self.buf, _page_table(rows, dev), n_dense)
-            lengths = _segmented_cumsum(deltas, dense_counts)
             total_bytes = 0
             copy_rows = []
+            scan_rows = []
             ends = data_end.cpu().tolist()
             for r, de in zip(rows, ends):
                 nbytes = r[1] - de
                 copy_rows.append((r[0] + de, nbytes, 0, 0, total_bytes, 0))
+                scan_rows.append((0, 0, r[2], r[3], total_bytes, 0))
                 total_bytes += nbytes
             blob = torch.empty(total_bytes, dtype=torch.uint8, device=dev)
             ext.pq_copy_bytes(self.buf, _page_table(copy_rows, dev), blob)
+            if validity is None:
+                # scan 1: delta stream -> length values (per page);
+                # scan 2: lengths -> offsets, page byte base folded in via
+                # aux (base == sum of earlier pages' lengths, so offsets
+                # are globally continuous)
+                len_rows = [(0, 0, r[2], r[3], 0, 0) for r in scan_rows]
+                ext.pq_segscan(deltas, _page_table(len_rows, dev))
+                ext.pq_segscan(deltas, _page_table(scan_rows, dev))
+                offsets = torch.zeros(n_dense + 1, dtype=torch.int64,
+                                      device=dev)
+                offsets[1:] = deltas
+                return StringColumn(offsets, blob, None, None)
+            lengths = _segmented_cumsum(deltas, dense_counts)
             offsets = torch.zeros(n_dense + 1, dtype=torch.int64, device=dev)
             torch.cumsum(lengths, 0, out=offsets[1:])
             return self._assemble_strings(offsets, blob, lengths, validity,

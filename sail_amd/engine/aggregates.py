@@ -186,6 +186,8 @@ def agg_eval(name: str, args: List[Column], gid: torch.Tensor, ng: int,
              out_type: T.DataType = None) -> Column:
     """Evaluate one aggregate over groups. args already evaluated per-row."""
     dev = gid.device
+    if name.endswith("_agg") and name not in UDAFS:
+        from . import functions_impl  # noqa: F401  (registers sketch UDAFS)
     if name in UDAFS:
         return _udaf_eval(name, args, gid, ng, filter_mask, out_type)
     if name == "count" and not args:

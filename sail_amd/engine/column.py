@@ -45,6 +45,9 @@ class Column:
         device = _dev(device)
         if isinstance(dtype, T.StringType):
             return StringColumn.from_pylist(list(values), device=device, dtype=dtype)
+        if isinstance(dtype, T.ArrayType):  # nested lists
+            return ListColumn.from_pylist(list(values), dtype.element,
+                                          device=device)
         validity = None
         if any(v is None for v in values):
             validity = torch.tensor([0 if v is None else 1 for v in values], dtype=torch.uint8, device=device)
@@ -161,6 +164,8 @@ class StringColumn(Column):
         validity = None
         if any(v is None for v in values):
             validity = torch.tensor([0 if v is None else 1 for v in values], dtype=torch.uint8, device=device)
+        empty = b"" if isinstance(dtype, T.BinaryType) else ""
+        values = [empty if v is None else v for v in values]
         uniq = set(v for v in values if v is not None)
         if dict_encode is None:
             dict_encode = n > 64 and len(uniq) * 16 < n
@@ -170,7 +175,7 @@ class StringColumn(Column):
             codes = torch.tensor([(-1 if v is None else idx[v]) for v in values], dtype=torch.int32, device=device)
             offsets, bytes_ = _pack_strings(udict, device)
             return StringColumn(offsets, bytes_, validity, codes, dtype=dtype)
-        offsets, bytes_ = _pack_strings(["" if v is None else v for v in values], device)
+        offsets, bytes_ = _pack_strings(values, device)
         return StringColumn(offsets, bytes_, validity, None, dtype=dtype)
 
     @staticmethod
@@ -251,9 +256,12 @@ class StringColumn(Column):
         return -1
 
     def dict_values(self) -> List[str]:
-        """Host copy of the dictionary (or all values if not dict-encoded)."""
+        """Host copy of the dictionary (or all values if not dict-encoded).
+        BINARY-typed columns return raw bytes (no utf-8 decode)."""
         offs = self.offsets.cpu().numpy()
         byts = self.bytes_.cpu().numpy().tobytes()
+        if isinstance(self.dtype, T.BinaryType):
+            return [byts[offs[i]:offs[i + 1]] for i in range(len(offs) - 1)]
         return [byts[offs[i]:offs[i + 1]].decode("utf-8", "replace") for i in range(len(offs) - 1)]
 
     def to_pylist(self) -> List[Optional[str]]:
@@ -274,7 +282,8 @@ class StringColumn(Column):
 
 
 def _pack_strings(strs: List[str], device) -> tuple:
-    enc = [s.encode("utf-8") for s in strs]
+    enc = [s if isinstance(s, (bytes, bytearray)) else s.encode("utf-8")
+           for s in strs]
     lens = np.fromiter((len(e) for e in enc), dtype=np.int64, count=len(enc))
     offsets = np.zeros(len(enc) + 1, dtype=np.int64)
     np.cumsum(lens, out=offsets[1:])

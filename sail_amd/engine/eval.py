@@ -63,7 +63,7 @@ def broadcast(v: Val, n: int, device) -> Column:
     if isinstance(dt, T.StringType):
         offs, byts = _pack_strings([v.value], device)
         codes = torch.zeros(n, dtype=torch.int32, device=device)
-        return StringColumn(offs, byts, None, codes)
+        return StringColumn(offs, byts, None, codes, dtype=dt)
     val = v.value
     if isinstance(dt, T.DecimalType):
         val = _to_scaled(val, dt.scale)

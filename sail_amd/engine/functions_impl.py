@@ -2434,3 +2434,5 @@ def _f_ts_add_months(args, out, chunk, ev):
 
 
 _IMPLS["ts_add_months"] = _f_ts_add_months
+
+from . import functions_ext  # noqa: E402,F401  (registers the round-2 batch)

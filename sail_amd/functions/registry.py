@@ -18,6 +18,7 @@ from ..engine import types as T
 # ---------------------------------------------------------------------------
 
 AGG_FUNCTIONS = {
+    "hll_sketch_agg", "hll_union_agg", "theta_sketch_agg", "theta_union_agg",
     "sum", "avg", "mean", "count", "min", "max", "first", "first_value",
     "last", "last_value", "stddev", "stddev_samp", "stddev_pop", "variance",
     "var_samp", "var_pop", "count_if", "any", "some", "bool_or", "every",
@@ -303,3 +304,20 @@ def scalar_return_type(name: str, arg_types: List[T.DataType]) -> Optional[T.Dat
     if rule is None:
         return None
     return rule(arg_types)
+
+
+# round-2 extension batch (impls in engine/functions_ext.py)
+_reg("monthname", lambda a: T.STRING)
+_reg("add_days add_years", lambda a: T.DATE)
+_reg("div bitmap_count hll_sketch_estimate theta_sketch_estimate",
+     lambda a: T.I64)
+_reg("aes_encrypt try_aes_encrypt aes_decrypt try_aes_decrypt "
+     "hll_union theta_union theta_intersection theta_difference",
+     lambda a: T.BINARY)
+_reg("cosine_similarity vector_cosine_similarity l1", lambda a: T.F64)
+_reg("unbase64 unhex", lambda a: T.BINARY)  # Spark returns BINARY
+_reg("crc32", lambda a: T.I64)  # Spark: BIGINT (values exceed int32)
+_reg("try_divide", lambda a: T.F64 if not isinstance(a[0], T.DecimalType)
+     else T.DecimalType(min(38, a[0].precision + 4), min(a[0].scale + 4, 10)))
+_reg("regexp_extract_all", lambda a: T.ArrayType(T.STRING))
+_reg("sentences", lambda a: T.ArrayType(T.ArrayType(T.STRING)))

@@ -36,6 +36,7 @@ std::vector<torch::Tensor> pq_delta_decode(torch::Tensor buf, torch::Tensor page
                                            int64_t total);
 std::vector<torch::Tensor> pq_bytearray_walk(torch::Tensor buf, torch::Tensor pages,
                                              int64_t total);
+void pq_segscan(torch::Tensor data, torch::Tensor pages);
 torch::Tensor pq_gather_strings(torch::Tensor buf, torch::Tensor src_pos,
                                 torch::Tensor lengths, torch::Tensor out_offsets,
                                 int64_t total_bytes);
@@ -63,5 +64,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pq_flba_i64", &pq_flba_i64, "parquet FLBA big-endian -> int64");
   m.def("pq_delta_decode", &pq_delta_decode, "parquet DELTA_BINARY_PACKED decode");
   m.def("pq_bytearray_walk", &pq_bytearray_walk, "parquet PLAIN byte_array walk");
+  m.def("pq_segscan", &pq_segscan, "per-page inclusive int64 scan (+page base)");
   m.def("pq_gather_strings", &pq_gather_strings, "gather byte_array payloads");
 }

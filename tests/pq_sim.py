@@ -166,3 +166,11 @@ def pq_gather_strings(buf_t, src_pos_t, lengths_t, out_offsets_t, total_bytes):
     for sp, ln, oo in zip(_np(src_pos_t), _np(lengths_t), _np(out_offsets_t)):
         out[oo:oo + ln] = buf[sp:sp + ln]
     return torch.from_numpy(out)
+
+
+def pq_segscan(data_t, pages_t):
+    data = data_t.numpy()
+    for _off, _len, nvals, out_row, aux, _w in _np(pages_t).reshape(-1, 6):
+        seg = data[out_row:out_row + nvals]
+        np.cumsum(seg, out=seg)
+        seg += aux

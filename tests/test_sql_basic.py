@@ -542,7 +542,7 @@ def test_misc_string_functions(s):
     assert q("SELECT to_number('$1,234.56', '999'), try_to_number('oops', '999')") \
         == [(1234.56, None)]
     assert q("SELECT to_binary('616263', 'hex'), try_to_binary('_bad_', 'base64')") \
-        == [("abc", None)]
+        == [(b"abc", None)]  # BINARY surfaces as bytes (Spark parity)
     assert q("SELECT to_varchar(123), to_char(1.5, '9.9')") == [("123", "1.5")]
 
 

@@ -101,58 +101,63 @@ class DeltaLog:
         except (ValueError, KeyError, json.JSONDecodeError):
             return None
 
-    def snapshot(self, version: Optional[int] = None):
-        """Replay the log: returns (schema, files, metadata, version).
-        Starts from the newest parquet checkpoint at or below the requested
-        version (ref: sail-delta-lake delta_log checkpoints/segments), then
-        replays only the JSON tail."""
-        vs = self.versions()
-        if not vs:
-            raise FileNotFoundError(f"not a delta table: {self.path}")
-        if version is None:
-            version = vs[-1]
-        files: Dict[str, dict] = {}
-        schema = None
-        meta = {}
-        start = 0
-        ckpt = self._last_checkpoint()
-        if ckpt is not None and ckpt <= version:
-            cp = os.path.join(self.log_path, f"{ckpt:020d}.checkpoint.parquet")
-            if os.path.exists(cp):
-                import pyarrow.parquet as pq
+    def _read_checkpoint(self, cp: str):
+        """Parse a checkpoint parquet in EITHER layout: Spark's nested
+        action-struct schema (one struct column per action type — what we
+        now write) or the legacy (kind, json) rows. Returns
+        (meta, {path: add_action}, {appId: txn})."""
+        import pyarrow.parquet as pq
 
-                tbl = pq.read_table(cp)
-                for kind, payload in zip(tbl.column("kind").to_pylist(),
-                                         tbl.column("json").to_pylist()):
-                    action = json.loads(payload)
-                    if kind == "metaData":
-                        meta = action
-                        schema = schema_from_string(meta["schemaString"])
-                    elif kind == "add":
-                        files[action["path"]] = {"add": action}
-                start = ckpt + 1
-        for v in vs:
-            if v < start:
-                continue
-            if v > version:
-                break
-            with open(os.path.join(self.log_path, f"{v:020d}.json")) as f:
-                for line in f:
-                    if not line.strip():
-                        continue
-                    action = json.loads(line)
-                    if "metaData" in action:
-                        meta = action["metaData"]
-                        schema = schema_from_string(meta["schemaString"])
-                    elif "add" in action:
-                        files[action["add"]["path"]] = action["add"]
-                    elif "remove" in action:
-                        files.pop(action["remove"]["path"], None)
-        return schema, list(files.keys()), meta, version
+        tbl = pq.read_table(cp)
+        meta: dict = {}
+        files: Dict[str, dict] = {}
+        txns: Dict[str, dict] = {}
+        if "kind" in tbl.column_names:  # legacy layout
+            for kind, payload in zip(tbl.column("kind").to_pylist(),
+                                     tbl.column("json").to_pylist()):
+                action = json.loads(payload)
+                if kind == "metaData":
+                    meta = action
+                elif kind == "add":
+                    files[action["path"]] = action
+                elif kind == "txn":
+                    txns[action["appId"]] = action
+            return meta, files, txns
+        cols = {n: tbl.column(n).to_pylist() for n in tbl.column_names}
+        n = tbl.num_rows
+        for i in range(n):
+            md = cols.get("metaData", [None] * n)[i]
+            if md is not None and md.get("id") is not None:
+                meta = {k: v for k, v in md.items() if v is not None}
+                if isinstance(meta.get("configuration"), list):
+                    meta["configuration"] = dict(meta["configuration"])
+                fmt = meta.get("format")
+                if isinstance(fmt, dict):
+                    meta["format"] = {k: v for k, v in fmt.items()
+                                      if v is not None}
+            a = cols.get("add", [None] * n)[i]
+            if a is not None and a.get("path") is not None:
+                add = {k: v for k, v in a.items() if v is not None}
+                if isinstance(add.get("partitionValues"), list):
+                    add["partitionValues"] = dict(add["partitionValues"])
+                dv = add.get("deletionVector")
+                if isinstance(dv, dict):
+                    if dv.get("storageType") is None:
+                        add.pop("deletionVector", None)
+                    else:
+                        add["deletionVector"] = {k: v for k, v in dv.items()
+                                                 if v is not None}
+                files[add["path"]] = add
+            t = cols.get("txn", [None] * n)[i]
+            if t is not None and t.get("appId") is not None:
+                txns[t["appId"]] = {k: v for k, v in t.items()
+                                    if v is not None}
+        return meta, files, txns
 
     def snapshot_adds(self, version: Optional[int] = None):
-        """Like snapshot() but returns the live ADD actions (needed for
-        deletion vectors / stats), not just the file names."""
+        """Replay the log from the newest checkpoint at or below `version`
+        plus the JSON tail; returns (schema, live add actions, metadata,
+        version). ref: sail-delta-lake delta_log checkpoints/segments."""
         vs = self.versions()
         if not vs:
             raise FileNotFoundError(f"not a delta table: {self.path}")
@@ -160,23 +165,15 @@ class DeltaLog:
             version = vs[-1]
         files: Dict[str, dict] = {}
         schema = None
-        meta = {}
+        meta: dict = {}
         start = 0
         ckpt = self._last_checkpoint()
         if ckpt is not None and ckpt <= version:
             cp = os.path.join(self.log_path, f"{ckpt:020d}.checkpoint.parquet")
             if os.path.exists(cp):
-                import pyarrow.parquet as pq
-
-                tbl = pq.read_table(cp)
-                for kind, payload in zip(tbl.column("kind").to_pylist(),
-                                         tbl.column("json").to_pylist()):
-                    action = json.loads(payload)
-                    if kind == "metaData":
-                        meta = action
-                        schema = schema_from_string(meta["schemaString"])
-                    elif kind == "add":
-                        files[action["path"]] = action
+                meta, files, _txns = self._read_checkpoint(cp)
+                if meta.get("schemaString"):
+                    schema = schema_from_string(meta["schemaString"])
                 start = ckpt + 1
         for v in vs:
             if v < start:
@@ -197,19 +194,24 @@ class DeltaLog:
                         files.pop(action["remove"]["path"], None)
         return schema, list(files.values()), meta, version
 
+    def snapshot(self, version: Optional[int] = None):
+        """(schema, live file names, metadata, version)."""
+        schema, adds, meta, v = self.snapshot_adds(version)
+        return schema, [a["path"] for a in adds], meta, v
+
     def maybe_checkpoint(self, version: int):
         """Write a parquet checkpoint of the live state every
-        CHECKPOINT_INTERVAL commits + the _last_checkpoint pointer.
-        Layout: one (kind, json) row per live action — a simplified
-        (non-Spark-binary) checkpoint schema, documented in STATUS.md."""
+        CHECKPOINT_INTERVAL commits + the _last_checkpoint pointer, in
+        Spark's columnar action layout: one nullable struct column per
+        action type (protocol/metaData/add/txn), one action per row
+        (ref: sail-delta-lake checkpoint action schema)."""
         if version == 0 or version % self.CHECKPOINT_INTERVAL != 0:
             return
         import pyarrow as pa
         import pyarrow.parquet as pq
 
-        schema, file_names, meta, _ = self.snapshot(version)
-        # re-replay gathered only file paths; fetch full add actions
-        files: Dict[str, dict] = {}
+        schema, adds, meta, _ = self.snapshot_adds(version)
+        txns: Dict[str, dict] = {}
         for v in self.versions():
             if v > version:
                 break
@@ -218,18 +220,74 @@ class DeltaLog:
                     if not line.strip():
                         continue
                     action = json.loads(line)
-                    if "add" in action:
-                        files[action["add"]["path"]] = action["add"]
-                    elif "remove" in action:
-                        files.pop(action["remove"]["path"], None)
-        kinds = ["protocol", "metaData"] + ["add"] * len(files)
-        payloads = [json.dumps({"minReaderVersion": 1, "minWriterVersion": 2}),
-                    json.dumps(meta)] + [json.dumps(a) for a in files.values()]
-        tbl = pa.table({"kind": kinds, "json": payloads})
-        target = os.path.join(self.log_path, f"{version:020d}.checkpoint.parquet")
+                    if "txn" in action:
+                        txns[action["txn"]["appId"]] = action["txn"]
+
+        proto_t = pa.struct([("minReaderVersion", pa.int32()),
+                             ("minWriterVersion", pa.int32())])
+        meta_t = pa.struct([
+            ("id", pa.string()), ("name", pa.string()),
+            ("description", pa.string()),
+            ("format", pa.struct([("provider", pa.string())])),
+            ("schemaString", pa.string()),
+            ("partitionColumns", pa.list_(pa.string())),
+            ("configuration", pa.map_(pa.string(), pa.string())),
+            ("createdTime", pa.int64())])
+        dv_t = pa.struct([("storageType", pa.string()),
+                          ("pathOrInlineDv", pa.string()),
+                          ("offset", pa.int32()),
+                          ("sizeInBytes", pa.int32()),
+                          ("cardinality", pa.int64())])
+        add_t = pa.struct([
+            ("path", pa.string()),
+            ("partitionValues", pa.map_(pa.string(), pa.string())),
+            ("size", pa.int64()), ("modificationTime", pa.int64()),
+            ("dataChange", pa.bool_()), ("stats", pa.string()),
+            ("deletionVector", dv_t)])
+        txn_t = pa.struct([("appId", pa.string()), ("version", pa.int64()),
+                           ("lastUpdated", pa.int64())])
+
+        rows = []
+        rows.append({"protocol": {"minReaderVersion": 1,
+                                  "minWriterVersion": 2},
+                     "metaData": None, "add": None, "txn": None})
+        md = {
+            "id": meta.get("id"), "name": meta.get("name"),
+            "description": meta.get("description"),
+            "format": {"provider": meta.get("format", {}).get(
+                "provider", "parquet")},
+            "schemaString": meta.get("schemaString"),
+            "partitionColumns": meta.get("partitionColumns", []),
+            "configuration": list((meta.get("configuration") or {}).items()),
+            "createdTime": meta.get("createdTime"),
+        }
+        rows.append({"protocol": None, "metaData": md, "add": None,
+                     "txn": None})
+        for a in adds:
+            dv = a.get("deletionVector")
+            rows.append({"protocol": None, "metaData": None, "txn": None,
+                         "add": {
+                             "path": a["path"],
+                             "partitionValues": list(
+                                 (a.get("partitionValues") or {}).items()),
+                             "size": a.get("size"),
+                             "modificationTime": a.get("modificationTime"),
+                             "dataChange": bool(a.get("dataChange", True)),
+                             "stats": a.get("stats"),
+                             "deletionVector": dv if dv else None}})
+        for t in txns.values():
+            rows.append({"protocol": None, "metaData": None, "add": None,
+                         "txn": {"appId": t["appId"],
+                                 "version": int(t.get("version", 0)),
+                                 "lastUpdated": t.get("lastUpdated")}})
+        arrow_schema = pa.schema([("protocol", proto_t), ("metaData", meta_t),
+                                  ("add", add_t), ("txn", txn_t)])
+        tbl = pa.Table.from_pylist(rows, schema=arrow_schema)
+        target = os.path.join(self.log_path,
+                              f"{version:020d}.checkpoint.parquet")
         pq.write_table(tbl, target)
         with open(os.path.join(self.log_path, "_last_checkpoint"), "w") as f:
-            json.dump({"version": version, "size": len(kinds)}, f)
+            json.dump({"version": version, "size": len(rows)}, f)
 
     def version_times(self):
         """[(version, commit_time_ms)] from file mtimes (commitInfo actions
@@ -441,6 +499,39 @@ def write_dv_file(table_path: str, positions) -> dict:
             "cardinality": len(positions)}
 
 
+class ConcurrentModificationException(RuntimeError):
+    """A winning commit touched files this transaction read/modifies
+    (ref: sail-delta-lake/src/transaction/conflict_checker.rs)."""
+
+
+def check_conflicts(log: "DeltaLog", read_version: int, touched_paths,
+                    operation: str = "update"):
+    """Replay commits AFTER read_version; raise when any of them added,
+    removed or re-added one of `touched_paths` (row-level ops cannot be
+    rebased blindly — the caller re-reads and re-runs). Returns the latest
+    version examined."""
+    latest = log.latest_version() or 0
+    touched = set(touched_paths)
+    for v in log.versions():
+        if v <= read_version:
+            continue
+        with open(os.path.join(log.log_path, f"{v:020d}.json")) as f:
+            for line in f:
+                if not line.strip():
+                    continue
+                action = json.loads(line)
+                p = None
+                if "add" in action:
+                    p = action["add"]["path"]
+                elif "remove" in action:
+                    p = action["remove"]["path"]
+                if p is not None and p in touched:
+                    raise ConcurrentModificationException(
+                        f"delta {operation}: version {v} modified {p} "
+                        f"after read version {read_version}")
+    return latest
+
+
 def scan_layout(path: str, schema, device, options):
     """Read the table file-by-file applying deletion vectors; returns
     (Table, [(add_action, surviving_original_positions ndarray)]) in row
@@ -455,11 +546,14 @@ def scan_layout(path: str, schema, device, options):
     from ..engine.executor import concat_columns
 
     log = DeltaLog(path)
-    tbl_schema, adds, _, _ = log.snapshot_adds(_version_opt(options, log))
+    tbl_schema, adds, _, snap_version = log.snapshot_adds(
+        _version_opt(options, log))
     if not adds:
         cols = {n: Column.from_values([], t, device=device)
                 for n, t in tbl_schema}
-        return Table(cols), []
+        empty = _Layout()
+        empty.version = snap_version
+        return Table(cols), empty
     parts, layout = [], []
     for add in adds:
         fpath = os.path.join(path, add["path"])
@@ -478,16 +572,28 @@ def scan_layout(path: str, schema, device, options):
             orig = np.arange(nrows, dtype=np.int64)
         parts.append(chunk)
         layout.append((add, orig))
+    layout = _Layout(layout)
+    layout.version = snap_version
     out = parts[0] if len(parts) == 1 else Chunk(
         [concat_columns([p.columns[i] for p in parts])
          for i in range(len(parts[0].columns))], list(parts[0].names))
     return Table({n: c for n, c in zip(out.names, out.columns)}), layout
 
 
+class _Layout(list):
+    """Scan layout + the snapshot version it was read at (conflict
+    detection needs the read version)."""
+
+    version: Optional[int] = None
+
+
 def delete_with_dv(path: str, layout, deleted_mask, max_retries: int = 10):
     """Commit a DELETE as deletion-vector updates (no data-file rewrite):
     per touched file, merge new positions into its DV and re-add the file
-    with the new descriptor."""
+    with the new descriptor. Raises ConcurrentModificationException when a
+    commit after the read version touched the same files (row-level ops
+    cannot be blindly rebased; ref: sail-delta-lake transaction/
+    conflict_checker.rs)."""
     import numpy as np
 
     log = DeltaLog(path)
@@ -513,7 +619,11 @@ def delete_with_dv(path: str, layout, deleted_mask, max_retries: int = 10):
         actions.append({"add": new_add})
     if not actions:
         return None
+    touched = [a["add"]["path"] for a in actions if "add" in a]
+    read_version = getattr(layout, "version", None)
     for _ in range(max_retries):
+        if read_version is not None:
+            check_conflicts(log, read_version, touched, "DELETE")
         version = (log.latest_version() or 0) + 1
         try:
             log.commit(version, actions)

@@ -346,3 +346,75 @@ def test_dv_file_crc_is_full_crc32(s, tmp_path):
         f.write(bytes(bad))
     with _pytest.raises(ValueError, match="checksum"):
         dv_positions(base, dv)
+
+
+def test_concurrent_dv_delete_conflict(s, tmp_path):
+    """ADVICE/VERDICT r1: row-level ops must detect that a winning commit
+    touched the same files instead of blindly re-committing stale DVs."""
+    from sail_amd.datasource import delta
+    from sail_amd.datasource.delta import ConcurrentModificationException
+
+    base = str(tmp_path / "cc")
+    s.create_dataframe({"id": list(range(40))}, name="cc_src")
+    s.sql(f"CREATE TABLE delta.`{base}` AS SELECT * FROM cc_src").collect()
+    # two readers scan the same version
+    t1, lay1 = delta.scan_layout(base, None, "cpu", {})
+    t2, lay2 = delta.scan_layout(base, None, "cpu", {})
+    import numpy as np
+
+    mask1 = np.zeros(40, dtype=bool)
+    mask1[:5] = True
+    mask2 = np.zeros(40, dtype=bool)
+    mask2[10:15] = True
+    assert delta.delete_with_dv(base, lay1, mask1) is not None
+    # the second transaction read the pre-delete version of the same file
+    with pytest.raises(ConcurrentModificationException):
+        delta.delete_with_dv(base, lay2, mask2)
+    # a FRESH read sees version after commit 1; its delete succeeds
+    _, lay3 = delta.scan_layout(base, None, "cpu", {})
+    mask3 = np.zeros(35, dtype=bool)
+    mask3[:3] = True
+    assert delta.delete_with_dv(base, lay3, mask3) is not None
+    assert s.sql(f"SELECT count(*) FROM delta.`{base}`").collect() == [(32,)]
+
+
+def test_spark_layout_checkpoint_roundtrip(s, tmp_path):
+    """Checkpoints use Spark's nested action-struct parquet layout; replay
+    reads through them (metaData + adds + DV descriptors + txn actions)."""
+    import pyarrow.parquet as pq
+
+    from sail_amd.datasource import delta
+    from sail_amd.datasource.delta import DeltaLog
+
+    base = str(tmp_path / "ckpt_v2")
+    s.create_dataframe({"id": list(range(30)), "v": [i * 2 for i in range(30)]},
+                       name="cp_src")
+    s.sql(f"CREATE TABLE delta.`{base}` AS SELECT * FROM cp_src").collect()
+    s.sql(f"DELETE FROM delta.`{base}` WHERE id < 3").collect()  # DV add
+    log = DeltaLog(base)
+    # drive to a checkpoint boundary with a txn-stamped append in the mix
+    from sail_amd.engine.chunk import Chunk
+    from sail_amd.engine.column import Column, Table
+    from sail_amd.engine import types as T
+
+    while (log.latest_version() or 0) < DeltaLog.CHECKPOINT_INTERVAL:
+        chunk = Chunk([Column.from_values([100], T.I64),
+                       Column.from_values([0], T.I64)], ["id", "v"])
+        delta.write(base, chunk, "append", {},
+                    txn=("sinkapp", log.latest_version() or 0))
+    cp = f"{DeltaLog.CHECKPOINT_INTERVAL:020d}.checkpoint.parquet"
+    import os
+    cp_path = os.path.join(base, "_delta_log", cp)
+    assert os.path.exists(cp_path)
+    t = pq.read_table(cp_path)
+    assert {"protocol", "metaData", "add", "txn"} <= set(t.column_names)
+    assert "kind" not in t.column_names  # nested layout, not (kind, json)
+    # force replay THROUGH the checkpoint: delete the early json versions
+    for v in range(0, 3):
+        os.remove(os.path.join(base, "_delta_log", f"{v:020d}.json"))
+    schema, adds, meta, _ = log.snapshot_adds()
+    assert meta.get("schemaString")
+    assert any(a.get("deletionVector") for a in adds)  # DV survived
+    assert delta.last_txn_version(base, "sinkapp") is not None
+    n = s.sql(f"SELECT count(*) FROM delta.`{base}`").collect()[0][0]
+    assert n == 27 + (log.latest_version() - 1)  # 30 - 3 deleted + appends

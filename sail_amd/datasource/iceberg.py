@@ -469,6 +469,28 @@ def write(path: str, chunk, mode: str, options: Dict[str, str]):
         _, prev, _ = read_container(t._local(parent["manifest-list"]))
         manifests += prev
 
+    _commit_snapshot(t, md, manifests, snap_id, seq, parent,
+                     "append" if mode == "append" else "overwrite")
+
+
+def history(path: str):
+    """Snapshot log as rows (version, snapshot_id, timestamp_ms, operation)."""
+    t = IcebergTable(path)
+    if not t.exists():
+        raise ValueError(f"not an Iceberg table: {path}")
+    out = []
+    for i, s in enumerate(t.metadata.get("snapshots", [])):
+        out.append((i, s["snapshot-id"], s["timestamp-ms"],
+                    s.get("summary", {}).get("operation", "")))
+    return out
+
+
+def _commit_snapshot(t: "IcebergTable", md: dict, manifests: List[dict],
+                     snap_id: int, seq: int, parent: Optional[dict],
+                     operation: str):
+    """Write manifest list + snapshot entry + new vN.metadata.json
+    (the tail every commit shares; ref: sail-iceberg commit flow)."""
+    now_ms = int(time.time() * 1000)
     ml_path = os.path.join(t.meta_dir, f"snap-{snap_id}.avro")
     write_container(ml_path, _MANIFEST_FILE_SCHEMA, manifests, metadata={
         "snapshot-id": str(snap_id).encode(),
@@ -480,8 +502,7 @@ def write(path: str, chunk, mode: str, options: Dict[str, str]):
     snap = {"snapshot-id": snap_id, "sequence-number": seq,
             "timestamp-ms": now_ms, "manifest-list": ml_path,
             "schema-id": md.get("current-schema-id", 0),
-            "summary": {"operation":
-                        "append" if mode == "append" else "overwrite"}}
+            "summary": {"operation": operation}}
     if parent is not None:
         snap["parent-snapshot-id"] = parent["snapshot-id"]
     md["snapshots"] = md.get("snapshots", []) + [snap]
@@ -502,13 +523,121 @@ def write(path: str, chunk, mode: str, options: Dict[str, str]):
         f.write(str(version))
 
 
-def history(path: str):
-    """Snapshot log as rows (version, snapshot_id, timestamp_ms, operation)."""
+def scan_layout(path: str, schema, device, options):
+    """Read with deletes applied, tracking per data file the manifest
+    file_path and surviving ORIGINAL row positions — lets DELETE commit
+    merge-on-read position-delete files (ref: sail-iceberg position delete
+    writers, src/physical_plan/)."""
+    import numpy as np
+    import pyarrow.parquet as pq
+    import torch
+
+    from . import parquet_io
+    from ..engine.chunk import Chunk
+    from ..engine.column import Column, Table
+    from ..engine.executor import concat_columns
+
     t = IcebergTable(path)
-    if not t.exists():
-        raise ValueError(f"not an Iceberg table: {path}")
-    out = []
-    for i, s in enumerate(t.metadata.get("snapshots", [])):
-        out.append((i, s["snapshot-id"], s["timestamp-ms"],
-                    s.get("summary", {}).get("operation", "")))
-    return out
+    tbl_schema = t.schema()
+    data, deletes = t.files(options)
+    if not data:
+        cols = {n: Column.from_values([], dt, device=device)
+                for n, dt in tbl_schema}
+        return Table(cols), []
+    pos_by_file: Dict[str, set] = {}
+    for df in deletes:
+        if df.get("content") != 1:
+            continue
+        dtbl = pq.read_table(df["file_path"])
+        for f, p0 in zip(dtbl.column("file_path").to_pylist(),
+                         dtbl.column("pos").to_pylist()):
+            pos_by_file.setdefault(os.path.basename(f), set()).add(p0)
+    parts, layout = [], []
+    for d in data:
+        tab = parquet_io.read([d["file_path"]], tbl_schema, device,
+                              options or {})
+        chunk = Chunk.from_table(tab)
+        nrows = chunk.num_rows
+        drop = pos_by_file.get(os.path.basename(d["file_path"]))
+        if drop:
+            keep = np.ones(nrows, dtype=bool)
+            keep[sorted(drop)] = False
+            orig = np.nonzero(keep)[0]
+            idx = torch.from_numpy(orig).to(torch.int64)
+            chunk = Chunk([c.gather(idx) for c in chunk.columns],
+                          list(chunk.names))
+        else:
+            orig = np.arange(nrows, dtype=np.int64)
+        parts.append(chunk)
+        layout.append((d.get("orig_path", d["file_path"]), orig))
+    out = parts[0] if len(parts) == 1 else Chunk(
+        [concat_columns([p.columns[i] for p in parts])
+         for i in range(len(parts[0].columns))], list(parts[0].names))
+    return Table({n: c for n, c in zip(out.names, out.columns)}), layout
+
+
+#: iceberg position-delete column field ids (spec: 2147483546/2147483545)
+def delete_with_positions(path: str, layout, deleted_mask):
+    """Commit a DELETE as a merge-on-read position-delete file: parquet of
+    (file_path, pos) rows + a content=1 (deletes) manifest appended to the
+    current snapshot's manifest list."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    t = IcebergTable(path)
+    md = t.metadata
+    rows_fp, rows_pos = [], []
+    off = 0
+    for fpath, orig in layout:
+        seg = deleted_mask[off:off + len(orig)]
+        off += len(orig)
+        newly = orig[seg]
+        rows_fp.extend([fpath] * len(newly))
+        rows_pos.extend(int(x) for x in newly)
+    if not rows_fp:
+        return None
+    data_dir = os.path.join(path, "data")
+    os.makedirs(data_dir, exist_ok=True)
+    del_path = os.path.join(data_dir,
+                            f"delete-{uuid.uuid4().hex}.parquet")
+    pq.write_table(pa.table({"file_path": pa.array(rows_fp),
+                             "pos": pa.array(rows_pos, pa.int64())}),
+                   del_path)
+    seq = md["last-sequence-number"] + 1
+    snap_id = _new_snapshot_id()
+    entry = {"status": 1, "snapshot_id": snap_id, "sequence_number": None,
+             "file_sequence_number": None,
+             "data_file": {"content": 1, "file_path": del_path,
+                           "file_format": "PARQUET", "partition": {},
+                           "record_count": len(rows_fp),
+                           "file_size_in_bytes": os.path.getsize(del_path),
+                           "equality_ids": None}}
+    mpath = os.path.join(t.meta_dir, f"{uuid.uuid4().hex}-m0.avro")
+    cur_id = md.get("current-schema-id", 0)
+    cur_schema = next((s for s in md.get("schemas", [])
+                       if s.get("schema-id", 0) == cur_id), None)
+    write_container(mpath, _MANIFEST_ENTRY_SCHEMA, [entry], metadata={
+        "schema": json.dumps(cur_schema or {}).encode(),
+        "schema-id": str(cur_id).encode(),
+        "partition-spec": json.dumps([]).encode(),
+        "partition-spec-id": b"0",
+        "format-version": b"2",
+        "content": b"deletes",
+    })
+    new_mf = {"manifest_path": mpath,
+              "manifest_length": os.path.getsize(mpath),
+              "partition_spec_id": 0, "content": 1,
+              "sequence_number": seq, "min_sequence_number": seq,
+              "added_snapshot_id": snap_id,
+              "added_data_files_count": 1,
+              "existing_data_files_count": 0, "deleted_data_files_count": 0,
+              "added_rows_count": len(rows_fp), "existing_rows_count": 0,
+              "deleted_rows_count": 0}
+    parent = t.snapshot()
+    manifests = [new_mf]
+    if parent is not None:
+        _, prev, _ = read_container(t._local(parent["manifest-list"]))
+        manifests += prev
+    _commit_snapshot(t, md, manifests, snap_id, seq, parent, "delete")
+    return len(rows_fp)

@@ -379,9 +379,20 @@ def _new_snapshot_id() -> int:
     return random.getrandbits(62)
 
 
-def write(path: str, chunk, mode: str, options: Dict[str, str]):
+def _partition_record_schema(pvals: Dict[str, object]) -> dict:
+    fields = []
+    for k, v in pvals.items():
+        at = "long" if isinstance(v, int) else "string"
+        fields.append({"name": k, "type": ["null", at], "default": None})
+    return {"type": "record", "name": "r102", "fields": fields}
+
+
+def write(path: str, chunk, mode: str, options: Dict[str, str],
+          partition_values: Optional[Dict[str, object]] = None,
+          partition_spec: Optional[List[dict]] = None):
     """append / overwrite commit: data parquet -> manifest Avro ->
-    manifest-list Avro -> new vN.metadata.json + version-hint."""
+    manifest-list Avro -> new vN.metadata.json + version-hint.
+    `partition_values`/`partition_spec` come from write_partitioned."""
     schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
     t = IcebergTable(path)
     if t.exists() and mode == "error":
@@ -396,7 +407,8 @@ def write(path: str, chunk, mode: str, options: Dict[str, str]):
               "last-column-id": len(schema),
               "current-schema-id": 0, "schemas": [_ice_schema(schema)],
               "default-spec-id": 0,
-              "partition-specs": [{"spec-id": 0, "fields": []}],
+              "partition-specs": [{"spec-id": 0,
+                                   "fields": partition_spec or []}],
               "last-partition-id": 999,
               "default-sort-order-id": 0,
               "sort-orders": [{"order-id": 0, "fields": []}],
@@ -437,15 +449,25 @@ def write(path: str, chunk, mode: str, options: Dict[str, str]):
     now_ms = int(time.time() * 1000)
 
     data_files = _write_data_files(path, chunk, options)
+    if partition_values:
+        for df in data_files:
+            df["partition"] = dict(partition_values)
     entries = [{"status": 1, "snapshot_id": snap_id, "sequence_number": None,
                 "file_sequence_number": None, "data_file": df}
                for df in data_files]
+    entry_schema = _MANIFEST_ENTRY_SCHEMA
+    if partition_values:
+        import copy as _copy
+
+        entry_schema = _copy.deepcopy(_MANIFEST_ENTRY_SCHEMA)
+        entry_schema["fields"][-1]["type"]["fields"][3]["type"] = \
+            _partition_record_schema(partition_values)
     mpath = os.path.join(t.meta_dir, f"{uuid.uuid4().hex}-m0.avro")
     cur_id = md.get("current-schema-id", 0)
     cur_schema = next((s for s in md.get("schemas", [])
                        if s.get("schema-id", 0) == cur_id),
                       _ice_schema(schema))
-    write_container(mpath, _MANIFEST_ENTRY_SCHEMA, entries, metadata={
+    write_container(mpath, entry_schema, entries, metadata={
         "schema": json.dumps(cur_schema).encode(),
         "schema-id": str(cur_id).encode(),
         "partition-spec": json.dumps([]).encode(),
@@ -641,3 +663,162 @@ def delete_with_positions(path: str, layout, deleted_mask):
         manifests += prev
     _commit_snapshot(t, md, manifests, snap_id, seq, parent, "delete")
     return len(rows_fp)
+
+
+# ===========================================================================
+# partition transforms (ref: crates/sail-iceberg/src/physical_plan/
+# partition_transform_expr.rs; iceberg spec "Partition Transforms")
+# ===========================================================================
+
+def _murmur3_32(data: bytes, seed: int = 0) -> int:
+    """murmur3_x86_32 — the hash the iceberg bucket transform specifies."""
+    c1, c2 = 0xCC9E2D51, 0x1B873593
+    h = seed
+    n = len(data)
+    for i in range(0, n - n % 4, 4):
+        k = int.from_bytes(data[i:i + 4], "little")
+        k = (k * c1) & 0xFFFFFFFF
+        k = ((k << 15) | (k >> 17)) & 0xFFFFFFFF
+        k = (k * c2) & 0xFFFFFFFF
+        h ^= k
+        h = ((h << 13) | (h >> 19)) & 0xFFFFFFFF
+        h = (h * 5 + 0xE6546B64) & 0xFFFFFFFF
+    tail = data[n - n % 4:]
+    if tail:
+        k = int.from_bytes(tail.ljust(4, b"\x00"), "little")
+        k = (k * c1) & 0xFFFFFFFF
+        k = ((k << 15) | (k >> 17)) & 0xFFFFFFFF
+        k = (k * c2) & 0xFFFFFFFF
+        h ^= k
+    h ^= n
+    h ^= h >> 16
+    h = (h * 0x85EBCA6B) & 0xFFFFFFFF
+    h ^= h >> 13
+    h = (h * 0xC2B2AE35) & 0xFFFFFFFF
+    h ^= h >> 16
+    return h
+
+
+def _bucket_hash(value, dtype: T.DataType) -> int:
+    import struct as _s
+
+    if value is None:
+        return 0
+    if isinstance(dtype, T.DecimalType):
+        unscaled = int(round(float(value) * (10 ** dtype.scale)))
+        blen = max((unscaled.bit_length() + 8) // 8, 1)
+        return _murmur3_32(unscaled.to_bytes(blen, "big", signed=True))
+    if isinstance(dtype, T.StringType):
+        b = value if isinstance(value, bytes) else str(value).encode()
+        return _murmur3_32(b)
+    if isinstance(dtype, T.DateType):
+        import datetime as _dt
+
+        days = (value - _dt.date(1970, 1, 1)).days \
+            if isinstance(value, _dt.date) else int(value)
+        return _murmur3_32(_s.pack("<q", days))
+    return _murmur3_32(_s.pack("<q", int(value)))
+
+
+_EPOCH_Y = 1970
+
+
+def parse_transform(spec: str):
+    """'bucket(4, col)' / 'truncate(10, col)' / 'years(col)' / 'col' ->
+    (transform_name, column, param)."""
+    m = re.match(r"(\w+)\s*\(\s*(?:(\d+)\s*,\s*)?(\w+)\s*\)$", spec.strip())
+    if not m:
+        return ("identity", spec.strip(), None)
+    name, param, col = m.group(1).lower(), m.group(2), m.group(3)
+    alias = {"year": "years", "month": "months", "day": "days",
+             "hour": "hours", "date": "days"}
+    name = alias.get(name, name)
+    if name in ("bucket", "truncate"):
+        if param is None:
+            raise ValueError(f"{name} transform needs a width: {spec}")
+        return (name, col, int(param))
+    if name in ("years", "months", "days", "hours", "identity", "void"):
+        return (name, col, None)
+    raise ValueError(f"unknown partition transform {spec!r}")
+
+
+def apply_transform(name: str, value, dtype: T.DataType, param):
+    """One partition value (iceberg spec semantics; None passes through)."""
+    import datetime as _dt
+
+    if value is None or name == "void":
+        return None
+    if name == "identity":
+        return value
+    if name == "bucket":
+        return (_bucket_hash(value, dtype) & 0x7FFFFFFF) % param
+    if name == "truncate":
+        if isinstance(dtype, T.StringType):
+            return str(value)[:param]
+        v = int(value)
+        return v - (v % param if v >= 0 else (v % param))
+    # temporal transforms: value may be date / epoch-micros timestamp
+    if isinstance(value, _dt.date) and not isinstance(value, _dt.datetime):
+        d = value
+    elif isinstance(dtype, T.TimestampType):
+        d = _dt.datetime.utcfromtimestamp(int(value) / 1e6)
+    else:
+        d = _dt.date(1970, 1, 1) + _dt.timedelta(days=int(value))
+    if name == "years":
+        return d.year - _EPOCH_Y
+    if name == "months":
+        return (d.year - _EPOCH_Y) * 12 + (d.month - 1)
+    if name == "days":
+        dd = d.date() if isinstance(d, _dt.datetime) else d
+        return (dd - _dt.date(1970, 1, 1)).days
+    if name == "hours":
+        if not isinstance(d, _dt.datetime):
+            d = _dt.datetime(d.year, d.month, d.day)
+        return int((d - _dt.datetime(1970, 1, 1)).total_seconds() // 3600)
+    raise ValueError(f"unknown transform {name}")
+
+
+def write_partitioned(path: str, chunk, mode: str, options: Dict[str, str],
+                      partition_by: List[str]):
+    """Partitioned iceberg write: rows are split by the transformed
+    partition tuple; each partition gets its own data files and manifest
+    entries carry the partition record; metadata records the spec."""
+    import numpy as np
+
+    from ..engine.chunk import Chunk as _Chunk
+
+    schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
+    names_low = [n.lower() for n in chunk.names]
+    specs = [parse_transform(s) for s in partition_by]
+    src_idx = []
+    for tname, col, param in specs:
+        if col.lower() not in names_low:
+            raise ValueError(f"partition column {col} not in output")
+        src_idx.append(names_low.index(col.lower()))
+    # transformed partition tuple per row (host; partition columns only)
+    cols_host = [chunk.columns[i].to_pylist() for i in src_idx]
+    n = chunk.num_rows
+    tuples = []
+    for r in range(n):
+        tuples.append(tuple(
+            apply_transform(t, cols_host[j][r], schema[src_idx[j]][1], p)
+            for j, (t, _c, p) in enumerate(specs)))
+    uniq = sorted(set(tuples), key=lambda x: tuple(
+        (v is None, v) for v in x))
+    import torch
+
+    spec_fields = [{"name": (f"{c}_{t}" if t != "identity" else c),
+                    "transform": (f"{t}[{p}]" if p is not None else t),
+                    "source-id": src_idx[j] + 1, "field-id": 1000 + j}
+                   for j, (t, c, p) in enumerate(specs)]
+    first = not IcebergTable(path).exists()
+    for i, key in enumerate(uniq):
+        rows = torch.tensor([r for r, tp in enumerate(tuples) if tp == key],
+                            dtype=torch.int64)
+        sub = _Chunk([c.gather(rows.to(c.device)) for c in chunk.columns],
+                     list(chunk.names))
+        pvals = {f["name"]: v for f, v in zip(spec_fields, key)}
+        write(path, sub,
+              mode if i == 0 else "append", options,
+              partition_values=pvals, partition_spec=spec_fields)
+    return len(uniq)

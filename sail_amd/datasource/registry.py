@@ -105,6 +105,11 @@ def write_source(fmt: str, path: str, chunk, mode: str, options, partition_by):
     fmt = fmt.lower()
     if partition_by and fmt == "parquet":
         return _write_partitioned(path, chunk, mode, options, partition_by)
+    if partition_by and fmt == "iceberg":
+        from . import iceberg
+
+        return iceberg.write_partitioned(path, chunk, mode, options,
+                                         partition_by)
     from . import parquet_io, csv_io
 
     if fmt == "parquet":

@@ -240,10 +240,17 @@ def execute_delete(executor, p: S.DeleteFrom) -> Chunk:
     tschema = p.__dict__["_target_schema"]
     tref = p.__dict__["_target_ref"]
     kind, name = tref
-    if kind == "delta":
+    if kind in ("delta", "iceberg"):
         # merge-on-read path: map deleted rows back to per-file positions
-        # and commit deletion vectors instead of rewriting parquet
-        from ..datasource.delta import delete_with_dv, scan_layout
+        # and commit deletion vectors (delta) / position-delete files
+        # (iceberg) instead of rewriting parquet
+        if kind == "delta":
+            from ..datasource.delta import delete_with_dv as _commit_del
+            from ..datasource.delta import scan_layout
+        else:
+            from ..datasource.iceberg import (delete_with_positions
+                                              as _commit_del)
+            from ..datasource.iceberg import scan_layout
 
         t, layout = scan_layout(name, tschema, executor.ctx.device, {})
         target = Chunk.from_table(t)
@@ -255,7 +262,7 @@ def execute_delete(executor, p: S.DeleteFrom) -> Chunk:
             return Chunk([Column.from_values([0], T.I64, device="cpu")],
                          ["num_affected_rows"])
         if target.num_rows and ndel / target.num_rows <= DV_DELETE_MAX_FRACTION:
-            delete_with_dv(name, layout, mask.cpu().numpy())
+            _commit_del(name, layout, mask.cpu().numpy())
         else:
             _store_target(executor, tref, target.filter_mask(~mask), tschema)
         return Chunk([Column.from_values([ndel], T.I64, device="cpu")],

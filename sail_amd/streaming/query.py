@@ -228,6 +228,14 @@ class StreamingQuery:
         self._agg_state: Optional[_AggState] = None
         self._upper_parent = None
         self._retained: Optional[Chunk] = None  # non-incremental fallback
+        #: stream-stream mode: every OTHER streaming view referenced by the
+        #: SQL (ref: Spark stream-stream joins; the reference rewrites these
+        #: through its StreamingRewriter — here the micro-batch runner
+        #: retains both inputs and re-evaluates, emitting deltas)
+        self._extra_sources: Dict[str, StreamSource] = {}
+        self._multi_retained: Dict[str, Optional[Chunk]] = {}
+        self._extra_offsets: Dict[str, object] = {}
+        self._prev_result_keys = None
         if self.checkpoint:
             self.id = self._load_or_create_metadata()
         if hasattr(self.sink, "app_id"):
@@ -245,9 +253,41 @@ class StreamingQuery:
                       [n for n, _ in self.source.schema])
         self.session.catalog.register_table(
             self.view_name, empty.to_table(), list(self.source.schema))
+        reg = getattr(self.session, "_stream_sources", {}) or {}
+        refs = set()
+        raw = self.session.parse(self.sql)
+
+        def _walk_reads(pl):
+            if isinstance(pl, S.Read):
+                refs.add(pl.table.lower())
+            for ch in (pl.children() if hasattr(pl, "children") else []):
+                if ch is not None:
+                    _walk_reads(ch)
+
+        _walk_reads(raw)
+        for v in sorted(refs):
+            if v in reg and v != self.view_name.lower():
+                self._extra_sources[v] = reg[v]
+                self._multi_retained[v] = None
+                self._extra_offsets[v] = reg[v].initial_offset()
+                # the extra view must resolve: register its empty schema
+                from ..engine.column import Column as _C
+
+                empty2 = Chunk([_C.from_values([], t)
+                                for _, t in reg[v].schema],
+                               [n for n, _ in reg[v].schema])
+                self.session.catalog.register_table(
+                    v, empty2.to_table(), list(reg[v].schema))
         self._plan = self.session.plan_sql(self.sql)
         if isinstance(self._plan, S.Command):
             raise ValueError("streaming query must be a SELECT")
+        if self._extra_sources:
+            self._mode = "multi_retained"
+            if self.watermark_spec is not None:
+                from ..engine.functions_impl import _parse_duration_us
+
+                self._wm_delay_us = _parse_duration_us(self.watermark_spec[1])
+            return
         found = _find_aggregate(self._plan)
         self._mode = "stateless"
         if found is not None and _count_aggregates(self._plan) == 1:
@@ -336,7 +376,11 @@ class StreamingQuery:
         if done:
             last = done[-1]
             with open(os.path.join(offs, str(last))) as f:
-                self._offset = json.load(f)["offset"]
+                rec = json.load(f)
+            self._offset = rec["offset"]
+            for v, o in (rec.get("extra") or {}).items():
+                if v in self._extra_offsets:
+                    self._extra_offsets[v] = o
             self.batch_id = last
             with open(os.path.join(commits, str(last))) as f:
                 self._max_event_us = json.load(f).get("maxEventTimeUs")
@@ -353,7 +397,10 @@ class StreamingQuery:
         # a pending offset without a commit is replayed by the normal loop:
         # read_between(self._offset, that offset) reproduces the batch.
         if pending and (not done or pending[-1] > done[-1]):
-            self._pending_offset = self._load_offset(offs, pending[-1])
+            with open(os.path.join(offs, str(pending[-1]))) as f:
+                rec = json.load(f)
+            self._pending_offset = rec["offset"]
+            self._pending_extra = rec.get("extra") or {}
             self._pending_id = pending[-1]
         else:
             self._pending_offset = None
@@ -455,7 +502,10 @@ class StreamingQuery:
                 raise self.exception
             if not self.is_active:
                 return
-            if self._offset == self.source.latest_offset() and self._idle.is_set():
+            if self._offset == self.source.latest_offset() \
+                    and all(self._extra_offsets[v] == s2.latest_offset()
+                            for v, s2 in self._extra_sources.items()) \
+                    and self._idle.is_set():
                 return
             time.sleep(0.01)
         raise TimeoutError("process_all_available timed out")
@@ -481,20 +531,34 @@ class StreamingQuery:
 
     def _run_one_batch(self) -> bool:
         if getattr(self, "_pending_offset", None) is not None:
-            end, bid = self._pending_offset, self._pending_id
+            end, extra_ends, bid = (self._pending_offset,
+                                    getattr(self, "_pending_extra", {}),
+                                    self._pending_id)
             self._pending_offset = None
         else:
             end = self.source.latest_offset()
-            if end == self._offset:
+            extra_ends = {v: src.latest_offset()
+                          for v, src in self._extra_sources.items()}
+            if end == self._offset and all(
+                    extra_ends[v] == self._extra_offsets[v]
+                    for v in extra_ends):
                 return False
             bid = self.batch_id + 1
             if self.checkpoint:
                 with open(os.path.join(self._ckpt_dir("offsets"), str(bid)), "w") as f:
-                    json.dump({"offset": end}, f)
+                    json.dump({"offset": end, "extra": extra_ends}, f)
         t0 = time.time()
         batch = self.source.read_between(self._offset, end)
         nrows = batch.num_rows
-        result = self._execute_batch(batch)
+        extra_batches = {}
+        for v, src in self._extra_sources.items():
+            extra_batches[v] = src.read_between(self._extra_offsets[v],
+                                                extra_ends[v])
+            nrows += extra_batches[v].num_rows
+        if self._mode == "multi_retained":
+            result = self._execute_multi(batch, extra_batches)
+        else:
+            result = self._execute_batch(batch)
         if result is not None:
             self.sink.write(result, bid, self.output_mode)
         if self.checkpoint:
@@ -504,6 +568,7 @@ class StreamingQuery:
                            "maxEventTimeUs": self._max_event_us}, f)
             self._prune_state(bid)
         self._offset = end
+        self._extra_offsets.update(extra_ends)
         self.batch_id = bid
         self.last_progress = {
             "id": self.id, "name": self.name, "batchId": bid,
@@ -579,6 +644,93 @@ class StreamingQuery:
         keep = torch.nonzero(~expired, as_tuple=False).flatten()
         st.keys = [c.gather(keep) for c in st.keys]
         st.partials = [c.gather(keep) for c in st.partials]
+
+    def _accumulate(self, prev: Optional[Chunk], batch: Chunk) -> Chunk:
+        from ..engine.executor import concat_columns
+
+        if prev is None or prev.num_rows == 0:
+            return batch
+        if batch.num_rows == 0:
+            return prev
+        return Chunk([concat_columns([a, b]) for a, b in
+                      zip(prev.columns, batch.columns)], list(batch.names))
+
+    def _execute_multi(self, batch: Chunk, extra: dict) -> Optional[Chunk]:
+        """Stream-stream mode (joins and any other multi-source shape):
+        retain every source's input, re-evaluate the whole plan, and for
+        append/update emit only rows NEW since the previous evaluation
+        (multiset delta on row keys). Watermarks evict retained rows older
+        than the event-time horizon."""
+        cat = self.session.catalog
+        self._retained = self._accumulate(self._retained, batch)
+        cat.register_table(self.view_name, self._retained.to_table(),
+                           list(self.source.schema))
+        for v, b in extra.items():
+            self._multi_retained[v] = self._accumulate(
+                self._multi_retained.get(v), b)
+            cat.register_table(v, self._multi_retained[v].to_table(),
+                               list(self._extra_sources[v].schema))
+        self._evict_retained_by_watermark()
+        result = self.session.execute_plan(self._plan)
+        if self.output_mode == "complete":
+            return result
+        from ..engine.executor import ExecutionContext, Executor
+
+        ex = Executor(ExecutionContext(self.session, self.session.device))
+        keys = ex._row_keys(result).cpu().tolist() if result.num_rows else []
+        prev = self._prev_result_keys or {}
+        counts: dict = {}
+        emit_rows = []
+        for i, k in enumerate(keys):
+            counts[k] = counts.get(k, 0) + 1
+            if counts[k] > prev.get(k, 0):
+                emit_rows.append(i)
+        self._prev_result_keys = counts
+        if not emit_rows:
+            return None
+        idx = torch.tensor(emit_rows, dtype=torch.int64,
+                           device=result.device)
+        return Chunk([c.gather(idx) for c in result.columns],
+                     list(result.names))
+
+    def _evict_retained_by_watermark(self):
+        """Drop retained rows older than the watermark on every source that
+        carries the event-time column (bounds stream-stream join state)."""
+        if self.watermark_spec is None:
+            return
+        col, _ = self.watermark_spec
+        # advance max event time from the primary retained input
+        for name, chunk in [(self.view_name, self._retained)] +                 list(self._multi_retained.items()):
+            if chunk is None:
+                continue
+            names_l = [n.lower() for n in chunk.names]
+            if col.lower() not in names_l:
+                continue
+            c = chunk.columns[names_l.index(col.lower())]
+            if chunk.num_rows:
+                mx = int(c.data.max().item())
+                self._max_event_us = max(self._max_event_us or mx, mx)
+        wm = self.watermark_us
+        if wm is None:
+            return
+        for name in [self.view_name] + list(self._multi_retained.keys()):
+            chunk = self._retained if name == self.view_name                 else self._multi_retained[name]
+            if chunk is None:
+                continue
+            names_l = [n.lower() for n in chunk.names]
+            if col.lower() not in names_l:
+                continue
+            c = chunk.columns[names_l.index(col.lower())]
+            keep = c.data >= wm
+            if bool(keep.all()):
+                continue
+            idx = torch.nonzero(keep, as_tuple=False).flatten()
+            pruned = Chunk([cc.gather(idx) for cc in chunk.columns],
+                           list(chunk.names))
+            if name == self.view_name:
+                self._retained = pruned
+            else:
+                self._multi_retained[name] = pruned
 
     def _execute_batch(self, batch: Chunk) -> Optional[Chunk]:
         from ..engine.executor import ExecutionContext, Executor, concat_columns

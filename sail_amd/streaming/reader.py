@@ -44,6 +44,12 @@ class DataStreamReader:
         if name is None:
             _STREAM_SEQ[0] += 1
             name = f"stream_{_STREAM_SEQ[0]}"
+        # session-wide registry: stream-stream joins resolve the OTHER
+        # streaming views referenced by a query through this map
+        reg = getattr(self._session, "_stream_sources", None)
+        if reg is None:
+            reg = self._session._stream_sources = {}
+        reg[name.lower()] = source
         return StreamingDataFrame(self._session, source, name)
 
 

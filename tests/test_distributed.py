@@ -377,3 +377,42 @@ def test_shuffled_sort_distinct_window_world2():
             got = pickle.load(f)
     for k in want:
         assert got[k] == want[k], k
+
+
+def test_fake_rccl_dist_world2_cpu():
+    """The thread-rank fake communicator (used to validate device
+    collectives on a 1-GPU box) matches single-process results on CPU."""
+    import sail_amd
+    from sail_amd.datagen.tpch import TpchGenerator, register_tpch
+    from sail_amd.datagen.tpch_queries import QUERIES
+    from sail_amd.engine.column import Table
+    from sail_amd.engine.executor import concat_columns
+    from sail_amd.exec.context import DistContext
+    from sail_amd.exec.fake_dist import run_world
+
+    world = 2
+    qids = [1, 6, 13]
+    results = {}
+
+    def body(rank, dist):
+        s = sail_amd.SessionContext(device="cpu")
+        s.dist = DistContext(dist, rank=rank, world=world, device="cpu")
+        register_tpch(s, sf=0.01, rank=rank, world=world)
+        results[rank] = {q: s.sql(QUERIES[q]).collect() for q in qids}
+
+    run_world(world, "cpu", body, strict_cuda=False)
+
+    single = sail_amd.SessionContext(device="cpu")
+    shards = [TpchGenerator(sf=0.01, device="cpu", rank=r, world=world).generate_all()
+              for r in range(world)]
+    for name in shards[0]:
+        if name in ("region", "nation"):
+            single.catalog.register_table(name, shards[0][name])
+            continue
+        cols = {cn: concat_columns([shards[r][name].columns[cn]
+                                    for r in range(world)])
+                for cn in shards[0][name].columns}
+        single.catalog.register_table(name, Table(cols))
+    for q in qids:
+        want = single.sql(QUERIES[q]).collect()
+        assert results[0][q] == want and results[1][q] == want, q

@@ -471,3 +471,77 @@ def test_delta_sink_replay_idempotent(s, tmp_path):
     q2.stop()
     assert q2.exception is None
     assert s2.sql(f"SELECT k, v FROM delta.`{out}`").collect() == [("x", 7)]
+
+
+def test_stream_stream_inner_join(s):
+    """Stream-stream join: rows match across batches in both directions
+    (ref: Spark stream-stream joins; the reference's streaming rewriter)."""
+    left = s.read_stream.format("memory").schema(
+        {"k": T.STRING, "lv": T.I64}).load(name="ssl")
+    right = s.read_stream.format("memory").schema(
+        {"k": T.STRING, "rv": T.I64}).load(name="ssr")
+    srcL, srcR = left.source, right.source
+    q = (left.sql("SELECT ssl.k, lv, rv FROM ssl JOIN ssr ON ssl.k = ssr.k")
+         .write_stream.output_mode("append").format("memory")
+         .query_name("ss_out").trigger(processing_time=0.01).start())
+    assert q._mode == "multi_retained"
+    srcL.add_rows({"k": ["a", "b"], "lv": [1, 2]})
+    q.process_all_available()
+    # no matches yet
+    srcR.add_rows({"k": ["b", "c"], "rv": [20, 30]})
+    q.process_all_available()
+    # late-arriving left row matches EARLIER right row
+    srcL.add_rows({"k": ["c"], "lv": [3]})
+    q.process_all_available()
+    q.stop()
+    assert q.exception is None
+    rows = sorted(s.sql("SELECT k, lv, rv FROM ss_out").collect())
+    assert rows == [("b", 2, 20), ("c", 3, 30)]
+
+
+def test_stream_stream_join_complete_and_agg(s):
+    left = s.read_stream.format("memory").schema(
+        {"k": T.STRING, "lv": T.I64}).load(name="cl")
+    right = s.read_stream.format("memory").schema(
+        {"k": T.STRING, "rv": T.I64}).load(name="cr")
+    q = (left.sql("SELECT cl.k AS k, sum(lv + rv) AS s "
+                  "FROM cl JOIN cr ON cl.k = cr.k GROUP BY cl.k")
+         .write_stream.output_mode("complete").format("memory")
+         .query_name("cc_out").trigger(processing_time=0.01).start())
+    left.source.add_rows({"k": ["x", "x"], "lv": [1, 2]})
+    right.source.add_rows({"k": ["x"], "rv": [10]})
+    q.process_all_available()
+    assert dict(s.sql("SELECT k, s FROM cc_out").collect()) == {"x": 23}
+    right.source.add_rows({"k": ["x"], "rv": [100]})
+    q.process_all_available()
+    q.stop()
+    assert q.exception is None
+    assert dict(s.sql("SELECT k, s FROM cc_out").collect()) == {"x": 226}
+
+
+def test_stream_stream_join_watermark_evicts_state(s):
+    """Retained join state is bounded: rows older than the watermark are
+    evicted and can no longer match."""
+    left = s.read_stream.format("memory").schema(
+        {"ts": T.TIMESTAMP, "k": T.STRING}).load(name="wl")
+    right = s.read_stream.format("memory").schema(
+        {"ts": T.TIMESTAMP, "k": T.STRING}).load(name="wr")
+    hour = 3_600_000_000
+    q = (left.with_watermark("ts", "1 hour")
+         .sql("SELECT wl.k FROM wl JOIN wr ON wl.k = wr.k")
+         .write_stream.output_mode("append").format("memory")
+         .query_name("wm_out").trigger(processing_time=0.01).start())
+    left.source.add_rows({"ts": [1 * hour], "k": ["old"]})
+    q.process_all_available()
+    # advance event time far beyond the watermark horizon
+    left.source.add_rows({"ts": [10 * hour], "k": ["new"]})
+    q.process_all_available()
+    # "old" (at 1h) is beyond the 1h watermark of max(10h) -> evicted;
+    # a matching right row must NOT join it anymore
+    right.source.add_rows({"ts": [10 * hour], "k": ["old"]})
+    right.source.add_rows({"ts": [10 * hour], "k": ["new"]})
+    q.process_all_available()
+    q.stop()
+    assert q.exception is None
+    rows = s.sql("SELECT k FROM wm_out").collect()
+    assert rows == [("new",)]

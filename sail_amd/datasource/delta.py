@@ -697,7 +697,10 @@ def _write_parts(path: str, chunk, options) -> List[dict]:
 
     schema = [(n, c.dtype) for n, c in zip(chunk.names, chunk.columns)]
     os.makedirs(path, exist_ok=True)
-    compression = (options or {}).get("compression", "snappy")
+    # default uncompressed: the GPU page decoder reads these directly
+    # (page-cache-warm scans are PCIe/decode-bound, not disk-bound);
+    # pass compression=snappy for Spark-default sizing
+    compression = (options or {}).get("compression", "none")
     n = chunk.num_rows
     nparts = max(1, min(16, (n + PART_ROWS - 1) // PART_ROWS))
     step = (n + nparts - 1) // nparts if nparts else n

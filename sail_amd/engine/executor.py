@@ -518,16 +518,24 @@ class Executor:
         # inputs on the UNFILTERED child and pass the selection mask into the
         # aggregation — avoids materializing high-selectivity filters (Q1
         # keeps 98.6% of lineitem; the gather costs more than the aggregate).
-        fused = self._try_masked_aggregate(p) if not p.grouping_sets else None
+        fused = self._try_masked_aggregate(p) \
+            if not p.grouping_sets and self._find_session_window(p) is None \
+            else None
         if fused is not None:
             return fused
         child = self.execute(p.input)
+        sw = self._find_session_window(p)
         if self.dctx is not None and child.partitioning == "sharded":
-            return self._dist_aggregate(p, child)
+            if sw is None:
+                return self._dist_aggregate(p, child)
+            child = self._gather(child)  # sessions span shard boundaries
         n = child.num_rows
         dev = child.device
         if p.group_by:
-            key_cols = [broadcast(self.ev.eval(g, child), n, dev) for g in p.group_by]
+            key_cols = [self._session_window_col(sw[1], j, p, child, n, dev)
+                        if sw is not None and j == sw[0]
+                        else broadcast(self.ev.eval(g, child), n, dev)
+                        for j, g in enumerate(p.group_by)]
             if n == 0:
                 return Chunk(key_cols + [_empty_agg_col(a, dev) for a in p.aggs],
                              [nm for nm, _ in p.schema])
@@ -627,6 +635,66 @@ class Executor:
         if finalized is None:  # zero batches: fall back to the normal path
             return None
         return finalized
+
+    @staticmethod
+    def _find_session_window(p: S.Aggregate):
+        """(index, Func) of a session_window(ts, gap) group key, or None."""
+        for i, g in enumerate(p.group_by or []):
+            e = g.child if isinstance(g, S.Alias) else g
+            if isinstance(e, S.Func) and e.name.lower() == "session_window":
+                return i, e
+        return None
+
+    def _session_window_col(self, e: S.Func, idx: int, p: S.Aggregate,
+                            child: Chunk, n: int, dev) -> Column:
+        """session_window(ts, gap): gap-separated sessions computed per
+        sibling group-key partition — rows whose event times are within
+        `gap` of the previous row (same partition) share a session; the
+        struct is (min_ts, max_ts + gap) of the session (Spark semantics;
+        ref: Spark session windows / sail streaming rewriter)."""
+        from .column import StructColumn
+        from .eval import Scalar
+        from .functions_impl import _parse_duration_us
+
+        ts = broadcast(self.ev.eval(e.args[0], child), n, dev)
+        gv = self.ev.eval(e.args[1], child)
+        gap_str = gv.value if isinstance(gv, Scalar) else gv.to_pylist()[0]
+        gap = _parse_duration_us(gap_str) if isinstance(gap_str, str) \
+            else int(gap_str)
+        others = [broadcast(self.ev.eval(g, child), n, dev)
+                  for j, g in enumerate(p.group_by) if j != idx]
+        us = ts.data.to(torch.int64)
+        if isinstance(ts.dtype, T.DateType):
+            us = us * 86_400_000_000
+        if others:
+            pgid, _, _np = group_ids(others)
+        else:
+            pgid = torch.zeros(n, dtype=torch.int64, device=dev)
+        order = torch.argsort(us)
+        order = order[torch.argsort(pgid.index_select(0, order), stable=True)]
+        g_s = pgid.index_select(0, order)
+        t_s = us.index_select(0, order)
+        new = torch.ones(n, dtype=torch.bool, device=dev)
+        if n > 1:
+            new[1:] = (g_s[1:] != g_s[:-1]) | ((t_s[1:] - t_s[:-1]) > gap)
+        sid_sorted = torch.cumsum(new.to(torch.int64), 0) - 1
+        ns = int(sid_sorted[-1].item()) + 1 if n else 0
+        starts = torch.zeros(ns, dtype=torch.int64, device=dev)
+        ends = torch.zeros(ns, dtype=torch.int64, device=dev)
+        starts[sid_sorted[new]] = t_s[new]
+        # last row of each session: next row starts a new session (or EOF)
+        last = torch.ones(n, dtype=torch.bool, device=dev)
+        if n > 1:
+            last[:-1] = new[1:]
+        ends[sid_sorted[last]] = t_s[last] + gap
+        sid = torch.empty(n, dtype=torch.int64, device=dev)
+        sid[order] = sid_sorted
+        return StructColumn(
+            [("start", Column(T.TIMESTAMP, starts.index_select(0, sid),
+                              ts.validity)),
+             ("end", Column(T.TIMESTAMP, ends.index_select(0, sid),
+                            ts.validity))],
+            ts.validity, dtype=e.dtype)
 
     def _grouping_sets_aggregate(self, p: S.Aggregate) -> Chunk:
         """ROLLUP/CUBE/GROUPING SETS: one aggregation per set, absent keys

@@ -273,6 +273,8 @@ _reg("datepart date_part", _i32)
 _reg("window", lambda a: T.StructType((T.StructField("start", T.TIMESTAMP),
                                        T.StructField("end", T.TIMESTAMP))))
 _reg("window_time", lambda a: T.TIMESTAMP)
+_reg("session_window", lambda a: T.StructType((
+    T.StructField("start", T.TIMESTAMP), T.StructField("end", T.TIMESTAMP))))
 _reg("sort_array array_sort array_distinct array_remove array_compact flatten "
      "slice array_repeat shuffle", _same)
 _reg("array_join", _string)

@@ -295,9 +295,14 @@ class StreamingQuery:
             from ..exec.distributed import decompose_agg
 
             decomps = [decompose_agg(a) for a in agg.aggs]
+            from ..engine.executor import Executor as _Ex
+
             ok = (not agg.grouping_sets and agg.having is None
                   and all(d is not None for d in decomps)
-                  and all(not getattr(a, "distinct", False) for a in agg.aggs))
+                  and all(not getattr(a, "distinct", False) for a in agg.aggs)
+                  # session windows re-session retroactively as batches
+                  # arrive: retained re-evaluation, not incremental merge
+                  and _Ex._find_session_window(agg) is None)
             if ok:
                 self._mode = "incremental"
                 self._agg_state = _AggState(agg, decomps)

@@ -545,3 +545,38 @@ def test_stream_stream_join_watermark_evicts_state(s):
     assert q.exception is None
     rows = s.sql("SELECT k FROM wm_out").collect()
     assert rows == [("new",)]
+
+
+def test_session_window_batch_and_stream(s):
+    """session_window(ts, gap): gap-separated sessions per sibling group
+    key, (min_ts, max_ts+gap) structs (ref: Spark session windows)."""
+    m = 60_000_000
+    s.create_dataframe(
+        {"u": ["a", "a", "a", "b", "a", "b"],
+         "ts": [0 * m, 3 * m, 20 * m, 1 * m, 22 * m, 2 * m],
+         "v": [1, 2, 3, 4, 5, 6]},
+        schema={"u": T.STRING, "ts": T.TIMESTAMP, "v": T.I64}, name="sess_ev")
+    r = s.sql("SELECT u, session_window(ts, '5 minutes') AS w, sum(v) AS sv "
+              "FROM sess_ev GROUP BY u, session_window(ts, '5 minutes')").collect()
+    got = sorted((u, w["start"], w["end"], sv) for u, w, sv in r)
+    assert got == [
+        ("a", 0, 8 * m, 3),            # rows at 0 and 3min merge
+        ("a", 20 * m, 27 * m, 8),      # rows at 20 and 22min merge
+        ("b", 1 * m, 7 * m, 10),       # rows at 1 and 2min merge
+    ]
+    # streaming (retained mode re-evaluates sessions each batch)
+    sdf = s.read_stream.format("memory").schema(
+        {"u": T.STRING, "ts": T.TIMESTAMP, "v": T.I64}).load(name="sw_in")
+    src = sdf.source
+    q = (sdf.sql("SELECT u, session_window(ts, '5 minutes') AS w, sum(v) s "
+                 "FROM sw_in GROUP BY u, session_window(ts, '5 minutes')")
+         .write_stream.output_mode("complete").format("memory")
+         .query_name("sw_out").trigger(processing_time=0.01).start())
+    src.add_rows({"u": ["x"], "ts": [0], "v": [1]})
+    q.process_all_available()
+    src.add_rows({"u": ["x"], "ts": [2 * m], "v": [10]})  # extends session
+    q.process_all_available()
+    q.stop()
+    assert q.exception is None
+    rows = s.sql("SELECT u, w.start, w.end, s FROM sw_out").collect()
+    assert rows == [("x", 0, 7 * m, 11)]

@@ -578,5 +578,6 @@ def test_session_window_batch_and_stream(s):
     q.process_all_available()
     q.stop()
     assert q.exception is None
-    rows = s.sql("SELECT u, w.start, w.end, s FROM sw_out").collect()
-    assert rows == [("x", 0, 7 * m, 11)]
+    rows = s.sql("SELECT u, w, s FROM sw_out").collect()
+    assert [(u, w["start"], w["end"], sv) for u, w, sv in rows] == \
+        [("x", 0, 7 * m, 11)]

@@ -37,6 +37,8 @@ std::vector<torch::Tensor> pq_delta_decode(torch::Tensor buf, torch::Tensor page
 std::vector<torch::Tensor> pq_bytearray_walk(torch::Tensor buf, torch::Tensor pages,
                                              int64_t total);
 void pq_segscan(torch::Tensor data, torch::Tensor pages);
+// mfma_reduce.hip
+torch::Tensor mfma_sum_f64(torch::Tensor x, bool use_mfma);
 torch::Tensor pq_gather_strings(torch::Tensor buf, torch::Tensor src_pos,
                                 torch::Tensor lengths, torch::Tensor out_offsets,
                                 int64_t total_bytes);
@@ -65,5 +67,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pq_delta_decode", &pq_delta_decode, "parquet DELTA_BINARY_PACKED decode");
   m.def("pq_bytearray_walk", &pq_bytearray_walk, "parquet PLAIN byte_array walk");
   m.def("pq_segscan", &pq_segscan, "per-page inclusive int64 scan (+page base)");
+  m.def("mfma_sum_f64", &mfma_sum_f64,
+        "f64 sum via v_mfma_f64_16x16x4 (measurement harness vs VALU)");
   m.def("pq_gather_strings", &pq_gather_strings, "gather byte_array payloads");
 }

@@ -568,3 +568,40 @@ def test_spmd_exchanges_world2_on_one_gpu():
                             f"q{q} rank{r}"
                     else:
                         assert gv == wv, f"q{q} rank{r}: {gv!r} != {wv!r}"
+
+
+def test_mfma_sum_f64_measurement(ext):
+    """MFMA-assisted reduction (north star: 'MFMA for aggregate
+    reductions'): correctness vs torch.sum plus an A/B timing against the
+    VALU kernel — the result is recorded in profiles/README.md whichever
+    way it lands (a whole-column sum is HBM-bound, so parity is the
+    expected outcome)."""
+    import time
+
+    n = 200_000_000
+    x = torch.rand(n, dtype=torch.float64, device="cuda")
+    want = float(x.sum().item())
+    got_mfma = float(ext.mfma_sum_f64(x, True).item())
+    got_valu = float(ext.mfma_sum_f64(x, False).item())
+    assert got_mfma == pytest.approx(want, rel=1e-9)
+    assert got_valu == pytest.approx(want, rel=1e-9)
+
+    def bench(fn, reps=10):
+        fn()  # warm
+        torch.cuda.synchronize()
+        t0 = time.time()
+        for _ in range(reps):
+            fn()
+        torch.cuda.synchronize()
+        return (time.time() - t0) / reps
+
+    t_mfma = bench(lambda: ext.mfma_sum_f64(x, True))
+    t_valu = bench(lambda: ext.mfma_sum_f64(x, False))
+    t_torch = bench(lambda: x.sum())
+    gbps = n * 8 / 1e9
+    print(f"\n# mfma_sum_f64 A/B on {n} doubles ({gbps:.1f} GB): "
+          f"mfma={t_mfma*1e3:.2f}ms ({gbps/t_mfma:.0f} GB/s) "
+          f"valu={t_valu*1e3:.2f}ms ({gbps/t_valu:.0f} GB/s) "
+          f"torch={t_torch*1e3:.2f}ms ({gbps/t_torch:.0f} GB/s)")
+    # both must run at a credible fraction of HBM bandwidth
+    assert gbps / t_mfma > 1000 and gbps / t_valu > 1000

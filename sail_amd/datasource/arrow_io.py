@@ -17,6 +17,7 @@ def _arrow_type(t: T.DataType) -> pa.DataType:
     m = {T.BooleanType: pa.bool_(), T.Int8Type: pa.int8(), T.Int16Type: pa.int16(),
          T.Int32Type: pa.int32(), T.Int64Type: pa.int64(), T.Float32Type: pa.float32(),
          T.Float64Type: pa.float64(), T.DateType: pa.date32(),
+         T.TimeType: pa.time64("us"),
          T.TimestampType: pa.timestamp("us"), T.StringType: pa.string(),
          T.BinaryType: pa.binary(), T.NullType: pa.null()}
     return m[type(t)]

@@ -121,6 +121,8 @@ class Column:
                 out.append(v / scale)
             elif isinstance(self.dtype, T.DateType):
                 out.append(_from_days(v))
+            elif isinstance(self.dtype, T.TimeType):
+                out.append(_from_time_us(v))
             else:
                 out.append(v)
         return out
@@ -332,6 +334,15 @@ def _gather_strings(offsets: torch.Tensor, bytes_: torch.Tensor, indices: torch.
         parts.append(bytes_.index_select(0, src))
         row_start = row_end
     return out_offsets, torch.cat(parts)
+
+
+def _from_time_us(us: int):
+    import datetime as _dt2
+
+    us = int(us)
+    return _dt2.time((us // 3_600_000_000) % 24,
+                     (us // 60_000_000) % 60,
+                     (us // 1_000_000) % 60, us % 1_000_000)
 
 
 def _to_days(v) -> int:

@@ -72,6 +72,9 @@ def broadcast(v: Val, n: int, device) -> Column:
             val = _date_str_to_days(val)
         elif isinstance(val, _dt.date):
             val = (val - _dt.date(1970, 1, 1)).days
+    elif isinstance(dt, T.TimeType) and isinstance(val, _dt.time):
+        val = ((val.hour * 60 + val.minute) * 60 + val.second) * 1_000_000 \
+            + val.microsecond
     data = torch.full((n,), val, dtype=dt.storage, device=device)
     return Column(dt, data, None)
 

@@ -851,3 +851,120 @@ for _n, _t in (("tinyint", T.I8), ("smallint", T.I16),
 
 # (name registration lives in functions/registry.py — resolution must not
 # depend on this module having been imported)
+
+
+# ---------------------------------------------------------------------------
+# TIME type family (Spark 4.1 TIME; ref: sail-plan scalar registry
+# time/make_time/to_time/time_trunc/... names)
+# ---------------------------------------------------------------------------
+
+_US_DAY = 86_400_000_000
+
+
+def _parse_time_str(s: str) -> Optional[int]:
+    import re as _re
+
+    m = _re.match(r"^\s*(\d{1,2}):(\d{2})(?::(\d{2})(\.\d+)?)?\s*$", s or "")
+    if not m:
+        return None
+    h, mi = int(m.group(1)), int(m.group(2))
+    sec = int(m.group(3) or 0)
+    frac = float(m.group(4) or 0.0)
+    if h > 23 or mi > 59 or sec > 59:
+        return None
+    return ((h * 60 + mi) * 60 + sec) * 1_000_000 + int(round(frac * 1e6))
+
+
+def _f_to_time(try_: bool):
+    def run(args, out, chunk, ev):
+        c = _col(args[0], chunk)
+        vals = []
+        for v in c.to_pylist():
+            if v is None:
+                vals.append(None)
+                continue
+            if isinstance(v, int):
+                vals.append(v % _US_DAY)
+                continue
+            us = _parse_time_str(str(v))
+            if us is None and not try_:
+                raise ValueError(f"cannot parse TIME {v!r}")
+            vals.append(us)
+        return _ret(vals, T.TIME, chunk)
+    return run
+
+
+_IMPLS["time"] = _IMPLS["to_time"] = _f_to_time(False)
+_IMPLS["try_to_time"] = _f_to_time(True)
+_IMPLS["make_time"] = _hostn(
+    lambda h, m, sec=0: None if h is None or m is None else
+    ((int(h) * 60 + int(m)) * 60) * 1_000_000 + int(round(float(sec or 0) * 1e6)))
+_IMPLS["time_from_micros"] = _hostn(
+    lambda v: None if v is None else int(v) % _US_DAY)
+_IMPLS["time_from_millis"] = _hostn(
+    lambda v: None if v is None else (int(v) * 1000) % _US_DAY)
+_IMPLS["time_from_seconds"] = _hostn(
+    lambda v: None if v is None else (int(v) * 1_000_000) % _US_DAY)
+_IMPLS["time_to_micros"] = _hostn(lambda t: None if t is None else _t_us(t))
+_IMPLS["time_to_millis"] = _hostn(
+    lambda t: None if t is None else _t_us(t) // 1000)
+_IMPLS["time_to_seconds"] = _hostn(
+    lambda t: None if t is None else _t_us(t) // 1_000_000)
+
+_TRUNC_UNITS = {"hour": 3_600_000_000, "minute": 60_000_000,
+                "second": 1_000_000, "millisecond": 1000, "microsecond": 1}
+
+
+def _t_us(t) -> int:
+    import datetime as _dt
+
+    if isinstance(t, _dt.time):
+        return ((t.hour * 60 + t.minute) * 60 + t.second) * 1_000_000 \
+            + t.microsecond
+    return int(t)
+
+
+def _f_time_trunc(args, out, chunk, ev):
+    cols, _ = _rows(args, chunk)
+    units, times = cols[0], cols[1]
+    vals = []
+    for u, t in zip(units, times):
+        if u is None or t is None:
+            vals.append(None)
+            continue
+        w = _TRUNC_UNITS.get(str(u).lower())
+        if w is None:
+            raise ValueError(f"time_trunc: unknown unit {u!r}")
+        vals.append((_t_us(t) // w) * w)
+    return _ret(vals, T.TIME, chunk)
+
+
+def _f_time_diff(args, out, chunk, ev):
+    cols, _ = _rows(args, chunk)
+    units, a, b = cols[0], cols[1], cols[2]
+    vals = []
+    for u, x, y in zip(units, a, b):
+        if u is None or x is None or y is None:
+            vals.append(None)
+            continue
+        w = _TRUNC_UNITS.get(str(u).lower())
+        if w is None:
+            raise ValueError(f"time_diff: unknown unit {u!r}")
+        vals.append((_t_us(y) - _t_us(x)) // w)
+    return _ret(vals, out, chunk)
+
+
+def _f_current_time(args, out, chunk, ev):
+    import datetime as _dt
+
+    now = _dt.datetime.utcnow()
+    us = ((now.hour * 60 + now.minute) * 60 + now.second) * 1_000_000 \
+        + now.microsecond
+    n = max(chunk.num_rows, 1)
+    return _ret([us] * chunk.num_rows if chunk.num_rows else [us],
+                T.TIME, chunk)
+
+
+_IMPLS["time_trunc"] = _f_time_trunc
+_IMPLS["time_diff"] = _f_time_diff
+_IMPLS["current_time"] = _f_current_time

@@ -47,7 +47,7 @@ class DataType:
 
     @property
     def is_temporal(self) -> bool:
-        return isinstance(self, (DateType, TimestampType))
+        return isinstance(self, (DateType, TimestampType, TimeType))
 
 
 class IntegerLike(DataType):
@@ -101,6 +101,12 @@ class DateType(DataType):
     """Days since epoch (Arrow date32)."""
 
     storage = torch.int32
+
+
+class TimeType(DataType):
+    """Microseconds since midnight (Arrow time64[us]; Spark 4.1 TIME)."""
+
+    storage = torch.int64
 
 
 class TimestampType(DataType):
@@ -185,6 +191,7 @@ I64 = Int64Type()
 F32 = Float32Type()
 F64 = Float64Type()
 DATE = DateType()
+TIME = TimeType()
 TIMESTAMP = TimestampType()
 STRING = StringType()
 BINARY = BinaryType()
@@ -200,6 +207,7 @@ _BY_NAME = {
     "double": F64,
     "date": DATE,
     "timestamp": TIMESTAMP, "timestamp_ltz": TIMESTAMP, "timestamp_ntz": TIMESTAMP,
+    "time": TIME,
     "string": STRING, "varchar": STRING, "char": STRING, "text": STRING,
     "binary": BINARY,
 }
@@ -212,7 +220,7 @@ def type_name(t: DataType) -> str:
     m = {NullType: "void", BooleanType: "boolean", Int8Type: "tinyint",
          Int16Type: "smallint", Int32Type: "int", Int64Type: "bigint",
          Float32Type: "float", Float64Type: "double", DateType: "date",
-         TimestampType: "timestamp", StringType: "string",
+         TimestampType: "timestamp", TimeType: "time", StringType: "string",
          BinaryType: "binary"}
     if type(t) in m:
         return m[type(t)]

@@ -323,3 +323,8 @@ _reg("try_divide", lambda a: T.F64 if not isinstance(a[0], T.DecimalType)
      else T.DecimalType(min(38, a[0].precision + 4), min(a[0].scale + 4, 10)))
 _reg("regexp_extract_all", lambda a: T.ArrayType(T.STRING))
 _reg("sentences", lambda a: T.ArrayType(T.ArrayType(T.STRING)))
+_reg("time to_time try_to_time make_time time_trunc current_time "
+     "time_from_micros time_from_millis time_from_seconds",
+     lambda a: T.TIME)
+_reg("time_to_micros time_to_millis time_to_seconds time_diff",
+     lambda a: T.I64)

@@ -143,3 +143,21 @@ def test_misc(s):
     with pytest.raises(Exception, match="boom"):
         q(s, "SELECT raise_error('boom')")
     assert q(s, "SELECT json_tuple('{\"a\":1, \"b\":\"x\"}', 'a')") == [("1",)]
+
+
+def test_time_type_family(s):
+    import datetime
+
+    r = q(s, "SELECT make_time(12, 30, 1.5)")
+    assert r == [(datetime.time(12, 30, 1, 500000),)]
+    assert q(s, "SELECT to_time('09:05:07')") == [(datetime.time(9, 5, 7),)]
+    assert q(s, "SELECT try_to_time('nope')") == [(None,)]
+    assert q(s, "SELECT time_trunc('HOUR', make_time(12, 45, 9))") == \
+        [(datetime.time(12, 0),)]
+    assert q(s, "SELECT time_diff('minute', make_time(1,0,0), make_time(2,30,0))") == \
+        [(90,)]
+    assert q(s, "SELECT time_to_seconds(make_time(0, 2, 5))") == [(125,)]
+    assert q(s, "SELECT time_from_seconds(3661)") == \
+        [(datetime.time(1, 1, 1),)]
+    # TIME compares as micros-of-day
+    assert q(s, "SELECT make_time(9,0,0) < make_time(10,0,0)") == [(True,)]

@@ -465,3 +465,69 @@ def _client_relation_methods():
 
 
 _client_relation_methods()
+
+
+def _client_artifact_methods():
+    import zlib as _zlib
+
+    def add_artifact(self, name: str, data: bytes):
+        """Upload one artifact (single-chunk batch form)."""
+        chunk = (W.field_bytes(1, data)
+                 + W.field_varint(2, _zlib.crc32(data) & 0xFFFFFFFF))
+        art = W.field_string(1, name) + W.field_message(2, chunk)
+        batch = W.field_message(1, art)
+        req = (W.field_string(1, self.session_id)
+               + W.field_message(3, batch))
+        call = self._channel.stream_unary(
+            f"/{_SERVICE}/AddArtifacts",
+            request_serializer=None, response_deserializer=None)
+        resp = W.parse(call(iter([req])))
+        out = []
+        for s in resp.get(1, []):
+            sf = W.parse(s)
+            out.append((W.first_str(sf, 1), bool(W.first_varint(sf, 2))))
+        return out
+
+    def add_artifact_chunked(self, name: str, chunks):
+        """Upload one artifact as a chunked stream."""
+        chunks = list(chunks)
+        total = sum(len(c) for c in chunks)
+
+        def reqs():
+            first = chunks[0] if chunks else b""
+            ic = (W.field_bytes(1, first)
+                  + W.field_varint(2, _zlib.crc32(first) & 0xFFFFFFFF))
+            begin = (W.field_string(1, name) + W.field_varint(2, total)
+                     + W.field_varint(3, len(chunks))
+                     + W.field_message(4, ic))
+            yield (W.field_string(1, self.session_id)
+                   + W.field_message(4, begin))
+            for c in chunks[1:]:
+                ch = (W.field_bytes(1, c)
+                      + W.field_varint(2, _zlib.crc32(c) & 0xFFFFFFFF))
+                yield (W.field_string(1, self.session_id)
+                       + W.field_message(5, ch))
+
+        call = self._channel.stream_unary(
+            f"/{_SERVICE}/AddArtifacts",
+            request_serializer=None, response_deserializer=None)
+        call(reqs())
+
+    def artifact_statuses(self, names):
+        req = W.field_string(1, self.session_id)
+        for n in names:
+            req += W.field_string(4, n)
+        resp = W.parse(self._call_unary("ArtifactStatus", req))
+        out = {}
+        for e in resp.get(1, []):
+            ef = W.parse(e)
+            st = W.parse(W.first(ef, 2, b""))
+            out[W.first_str(ef, 1)] = bool(W.first_varint(st, 1))
+        return out
+
+    ConnectClient.add_artifact = add_artifact
+    ConnectClient.add_artifact_chunked = add_artifact_chunked
+    ConnectClient.artifact_statuses = artifact_statuses
+
+
+_client_artifact_methods()

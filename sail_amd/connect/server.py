@@ -140,8 +140,8 @@ class SparkConnectServer:
             "ReleaseExecute": grpc.unary_unary_rpc_method_handler(self._release_execute),
             "ReleaseSession": grpc.unary_unary_rpc_method_handler(self._ack),
             "ReattachExecute": grpc.unary_stream_rpc_method_handler(self._reattach_execute),
-            "AddArtifacts": grpc.stream_unary_rpc_method_handler(self._unimplemented_unary),
-            "ArtifactStatus": grpc.unary_unary_rpc_method_handler(self._unimplemented_unary),
+            "AddArtifacts": grpc.stream_unary_rpc_method_handler(self._add_artifacts),
+            "ArtifactStatus": grpc.unary_unary_rpc_method_handler(self._artifact_status),
         }
         self._server.add_generic_rpc_handlers(
             (_GenericHandler(_SERVICE, handlers),))
@@ -510,6 +510,75 @@ class SparkConnectServer:
                 self._op_buffers.pop(key, None)
         return (W.field_string(1, session_id)
                 + W.field_string(2, op_id))
+
+    def _session_artifacts(self, session_id: str) -> dict:
+        with self._lock:
+            if not hasattr(self, "_artifacts"):
+                self._artifacts = {}
+            return self._artifacts.setdefault(session_id, {})
+
+    def _add_artifacts(self, request_iterator, context) -> bytes:
+        """Client-streamed artifact upload (single-chunk batches and
+        chunked artifacts; ref: sail-spark-connect server.rs AddArtifacts).
+        Artifacts are retained per session (pyfile/jar payloads are stored;
+        execution-side artifact use is the Python UDF registry)."""
+        import zlib as _zlib
+
+        session_id = ""
+        summaries = []
+        pending_name = None
+        pending_buf = b""
+        store = None
+        for raw in request_iterator:
+            req = W.parse(raw)
+            session_id = W.first_str(req, 1) or session_id
+            if store is None:
+                store = self._session_artifacts(session_id)
+            batch = W.first(req, 3)
+            if batch is not None:
+                for art in W.parse(batch).get(1, []):
+                    af = W.parse(art)
+                    name = W.first_str(af, 1)
+                    ch = W.parse(W.first(af, 2, b""))
+                    data = W.first(ch, 1, b"")
+                    crc = W.first_varint(ch, 2, 0)
+                    ok = (_zlib.crc32(data) & 0xFFFFFFFF) == crc or crc == 0
+                    if ok:
+                        store[name] = data
+                    summaries.append((name, ok))
+            begin = W.first(req, 4)
+            if begin is not None:
+                bf = W.parse(begin)
+                pending_name = W.first_str(bf, 1)
+                pending_buf = b""
+                ic = W.first(bf, 4)  # BeginChunkedArtifact.initial_chunk
+                if ic is not None:
+                    pending_buf += W.first(W.parse(ic), 1, b"")
+            chunk = W.first(req, 5)
+            if chunk is not None and pending_name is not None:
+                pending_buf += W.first(W.parse(chunk), 1, b"")
+        if pending_name is not None:
+            store = store if store is not None else \
+                self._session_artifacts(session_id)
+            store[pending_name] = pending_buf
+            summaries.append((pending_name, True))
+        out = W.field_string(2, session_id)
+        for name, ok in summaries:
+            body = W.field_string(1, name) + W.field_varint(2, 1 if ok else 0)
+            out += W.field_message(1, body)
+        return out
+
+    def _artifact_status(self, request: bytes, context) -> bytes:
+        req = W.parse(request)
+        session_id = W.first_str(req, 1)
+        store = self._session_artifacts(session_id)
+        out = W.field_string(2, session_id)
+        for name in req.get(4, []):
+            n = name.decode()
+            status = W.field_varint(1, 1 if n in store else 0)
+            entry = (W.field_string(1, n) + W.field_message(2, status))
+            out += W.field_message(1, entry)
+        return out
 
     def _ack(self, request: bytes, context) -> bytes:
         return b""

@@ -247,3 +247,19 @@ def test_reattach_and_release(server, client, rel_data):
 
     with pytest.raises(_grpc.RpcError):
         client.reattach(op, None)
+
+
+def test_add_artifacts_and_status(server, client):
+    """AddArtifacts (single-chunk batch + chunked stream) and
+    ArtifactStatus (ref: sail-spark-connect server.rs:288-356)."""
+    res = client.add_artifact("pyfiles/helper.py", b"def f():\n    return 7\n")
+    assert res == [("pyfiles/helper.py", True)]
+    client.add_artifact_chunked("jars/big.bin",
+                                [b"a" * 100, b"b" * 100, b"c" * 50])
+    st = client.artifact_statuses(
+        ["pyfiles/helper.py", "jars/big.bin", "missing.txt"])
+    assert st == {"pyfiles/helper.py": True, "jars/big.bin": True,
+                  "missing.txt": False}
+    # stored content is intact on the server side
+    arts = server._session_artifacts(client.session_id)
+    assert arts["jars/big.bin"] == b"a" * 100 + b"b" * 100 + b"c" * 50

@@ -83,7 +83,16 @@ def main():
     t0 = time.time()
     scan_info = None
     if args.workload == "clickbench":
-        register_clickbench(session, rows=args.rows, device=device, rank=rank, world=world)
+        if args.scan == "parquet":
+            from sail_amd.datagen.clickbench import register_clickbench_parquet
+
+            scan_info = register_clickbench_parquet(
+                session, rows=args.rows, device=device, rank=rank, world=world)
+            print(f"# parquet shards: {scan_info['bytes']/1e9:.2f} GB in "
+                  f"{scan_info['data_dir']}", file=sys.stderr)
+        else:
+            register_clickbench(session, rows=args.rows, device=device,
+                                rank=rank, world=world)
     elif args.workload == "delta":
         from sail_amd.datagen.delta_bench import setup_delta_bench
 

@@ -138,3 +138,25 @@ def register_clickbench(session, rows: int = 100_000_000, device=None,
 
         sync_table_stats(session)
     return t
+
+
+
+def register_clickbench_parquet(session, rows: int = 100_000_000, device=None,
+                                rank: int = 0, world: int = 1, seed: int = 7,
+                                data_dir=None):
+    """Scan-inclusive ClickBench (BASELINE config #3 "hits.parquet"):
+    generate on device, persist the shard as parquet, register a scan view
+    so every timed query re-reads + GPU-decodes hits from disk."""
+    from .scan_swap import persist_and_swap
+    from .tpch import write_tpch_parquet
+
+    dev = device or session.device
+    t = generate_hits(rows=rows, device=dev, seed=seed, rank=rank, world=world)
+    session.catalog.register_table("hits", t, replicated=(world == 1),
+                                   global_rows=rows)
+    tables = {"hits": t}
+    return persist_and_swap(
+        session, tables, data_dir=data_dir,
+        default_dir=f"sail_hits_{rows // 1_000_000}m", rank=rank, world=world,
+        write_fn=lambda tbls, d, r: write_tpch_parquet(
+            tbls, d, rank=r, page_size=1 << 20))

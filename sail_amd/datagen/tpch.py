@@ -500,6 +500,10 @@ def write_tpch_parquet(tables: Dict[str, Table], out_dir: str, rank: int = 0,
                         column_encoding=col_enc or None,
                         data_page_size=page_size,
                         store_decimal_as_integer=True,
+                        # keep big dictionaries dict-encoded (ClickBench
+                        # URL/Title): a PLAIN fallback mid-chunk would push
+                        # the column to the host decode path
+                        dictionary_pagesize_limit=64 << 20,
                         data_page_version="1.0")
                 writer.write_table(at, row_group_size=row_group_rows)
                 if n == 0:
@@ -519,9 +523,7 @@ def register_tpch_parquet(session, sf: float = 0.01, device=None, rank: int = 0,
     register scan VIEWS so each query re-reads + GPU-decodes its columns.
     Planner statistics are computed while the data is still resident (the
     analogue of the reference's statistics cache) and survive the swap."""
-    import tempfile
-
-    from ..plan import spec as S
+    from .scan_swap import persist_and_swap
 
     dev = device or session.device
     gen = TpchGenerator(sf=sf, device=dev, seed=seed, rank=rank, world=world,
@@ -531,40 +533,11 @@ def register_tpch_parquet(session, sf: float = 0.01, device=None, rank: int = 0,
                 "customer": gen.n_customer, "part": gen.n_part,
                 "partsupp": gen.n_part * 4, "orders": gen.n_orders,
                 "lineitem": gen.n_orders * 4}
-    cat = session.catalog
     for name, tbl in tables.items():
-        cat.register_table(
+        session.catalog.register_table(
             name, tbl, replicated=(world == 1 or name in ("region", "nation")),
             global_rows=globals_[name])
-        for cn in tbl.columns:
-            cat.column_stats(name, cn)  # pre-warm: cached past the swap
-    if world > 1 and getattr(session, "dist", None) is not None:
-        from ..exec.distributed import sync_table_stats
-
-        sync_table_stats(session)
-    if data_dir is None:
-        data_dir = os.environ.get(
-            "SAIL_BENCH_DATA_DIR",
-            os.path.join(tempfile.gettempdir(), f"sail_tpch_sf{sf:g}"))
-    paths = write_tpch_parquet(tables, data_dir, rank=rank)
-    total_bytes = sum(os.path.getsize(p) for p in paths.values())
-    # swap resident tables for scan views; free HBM
-    for name, tbl in tables.items():
-        schema = [(n, c.dtype) for n, c in tbl.columns.items()]
-        sharded = not cat.is_replicated(name)
-        node = S.DataSourceRead(
-            format="parquet", paths=[paths[name]],
-            options={"partitioning": "sharded" if sharded else "replicated"})
-        node.schema = schema
-        node.__dict__["_table_name"] = name  # planner statistics key
-        with cat._lock:
-            k = cat._key(name)
-            cat._tables.pop(k, None)
-            cat._views[k] = node
-    tables.clear()
-    del gen
-    import torch as _t
-
-    if _t.cuda.is_available():
-        _t.cuda.empty_cache()
-    return {"data_dir": data_dir, "bytes": total_bytes, "paths": paths}
+    return persist_and_swap(
+        session, tables, data_dir=data_dir,
+        default_dir=f"sail_tpch_sf{sf:g}", rank=rank, world=world,
+        write_fn=lambda tbls, d, r: write_tpch_parquet(tbls, d, rank=r))

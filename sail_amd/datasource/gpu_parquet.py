@@ -225,6 +225,14 @@ class _StagingPool:
 
 
 _STAGING = _StagingPool()
+_COPY_STREAM = None
+
+
+def _copy_stream(device):
+    global _COPY_STREAM
+    if _COPY_STREAM is None:
+        _COPY_STREAM = torch.cuda.Stream(device=device)
+    return _COPY_STREAM
 
 
 def _upload_ranges(path: str, ranges: List[Tuple[int, int]], device):
@@ -259,7 +267,18 @@ def _upload_ranges(path: str, ranges: List[Tuple[int, int]], device):
         for j in jobs:
             _one(j)
     view[total:total + 16] = 0
-    dev = stage[:total + 16].to(device, non_blocking=True)
+    if torch.cuda.is_available() and str(device).startswith("cuda"):
+        # H2D on a dedicated copy stream so the NEXT column's upload
+        # overlaps the CURRENT column's decode kernels; the compute stream
+        # waits on the copy event before its kernels touch the buffer
+        cs = _copy_stream(device)
+        cur = torch.cuda.current_stream()
+        with torch.cuda.stream(cs):
+            dev = stage[:total + 16].to(device, non_blocking=True)
+        cur.wait_stream(cs)
+        dev.record_stream(cur)
+    else:
+        dev = stage[:total + 16].to(device, non_blocking=True)
     _STAGING.mark_uploaded(slot)
     return dev, offs
 

@@ -968,3 +968,122 @@ def _f_current_time(args, out, chunk, ev):
 _IMPLS["time_trunc"] = _f_time_trunc
 _IMPLS["time_diff"] = _f_time_diff
 _IMPLS["current_time"] = _f_current_time
+
+
+# ---------------------------------------------------------------------------
+# Avro codec functions (reuse the from-scratch container codec in
+# utils/avro.py; ref: sail-function from_avro/to_avro/schema_of_avro)
+# ---------------------------------------------------------------------------
+
+def _avro_schema_for(dtype: T.DataType):
+    if isinstance(dtype, T.StructType):
+        return {"type": "record", "name": "topLevelRecord", "fields": [
+            {"name": f.name, "type": ["null", _avro_schema_for(f.dtype)]}
+            for f in dtype.fields]}
+    if isinstance(dtype, T.BinaryType):
+        return "bytes"
+    if isinstance(dtype, T.StringType):
+        return "string"
+    if isinstance(dtype, T.DecimalType) or isinstance(dtype, T.Float64Type):
+        return "double"
+    if isinstance(dtype, T.Float32Type):
+        return "float"
+    if isinstance(dtype, T.BooleanType):
+        return "boolean"
+    if dtype.is_integer or dtype.is_temporal:
+        return "long"
+    if isinstance(dtype, T.ArrayType):
+        return {"type": "array", "items": _avro_schema_for(dtype.element)}
+    raise ValueError(f"to_avro: unsupported type {dtype}")
+
+
+def _struct_type_from_avro(schema) -> T.DataType:
+    def leaf(s):
+        if isinstance(s, list):  # union: take the non-null branch
+            s = next(x for x in s if x != "null")
+        if isinstance(s, dict):
+            if s.get("type") == "record":
+                return T.StructType(tuple(
+                    T.StructField(f["name"], leaf(f["type"]))
+                    for f in s["fields"]))
+            if s.get("type") == "array":
+                return T.ArrayType(leaf(s["items"]))
+            s = s.get("type")
+        return {"null": T.NULL, "boolean": T.BOOL, "int": T.I32,
+                "long": T.I64, "float": T.F32, "double": T.F64,
+                "bytes": T.BINARY, "string": T.STRING}[s]
+    return leaf(schema)
+
+
+def _f_to_avro(args, out, chunk, ev):
+    from ..utils.avro import _encode
+
+    c = _col(args[0], chunk)
+    schema = _avro_schema_for(c.dtype)
+    vals = []
+    for v in c.to_pylist():
+        buf = bytearray()
+        if isinstance(c.dtype, T.StructType):
+            # per-field nullable union encode
+            _encode(schema, v, buf, {})
+        else:
+            _encode(schema, v, buf, {})
+        vals.append(bytes(buf))
+    return Column.from_values(vals, T.BINARY, device=str(chunk.device))
+
+
+def _f_from_avro(args, out, chunk, ev):
+    import io as _io
+    import json as _json
+
+    from ..utils.avro import _decode
+
+    c = _col(args[0], chunk)
+    sch_col = _col(args[1], chunk)
+    schema = _json.loads(sch_col.to_pylist()[0])
+    stype = _struct_type_from_avro(schema)
+    rows = []
+    for v in c.to_pylist():
+        if v is None:
+            rows.append(None)
+            continue
+        rows.append(_decode(schema, _io.BytesIO(_b(v)), {}))
+    if isinstance(stype, T.StructType):
+        from .column import StructColumn
+
+        fields = []
+        for f in stype.fields:
+            fv = [None if r is None else r.get(f.name) for r in rows]
+            fields.append((f.name, Column.from_values(fv, f.dtype,
+                                                      device=str(chunk.device))))
+        validity = None
+        if any(r is None for r in rows):
+            import torch as _t
+
+            validity = _t.tensor([0 if r is None else 1 for r in rows],
+                                 dtype=_t.uint8, device=chunk.device)
+        return StructColumn(fields, validity, dtype=stype)
+    return Column.from_values(rows, stype, device=str(chunk.device))
+
+
+def _f_schema_of_avro(args, out, chunk, ev):
+    import json as _json
+
+    c = _col(args[0], chunk)
+    schema = _json.loads(c.to_pylist()[0])
+    stype = _struct_type_from_avro(schema)
+
+    def render(t):
+        if isinstance(t, T.StructType):
+            inner = ", ".join(f"{f.name}: {render(f.dtype)}"
+                              for f in t.fields)
+            return f"STRUCT<{inner}>"
+        return T.type_name(t).upper()
+
+    return StringColumn.from_pylist([render(stype)],
+                                    device=str(chunk.device))
+
+
+_IMPLS["to_avro"] = _f_to_avro
+_IMPLS["from_avro"] = _f_from_avro
+_IMPLS["schema_of_avro"] = _f_schema_of_avro

@@ -328,3 +328,6 @@ _reg("time to_time try_to_time make_time time_trunc current_time "
      lambda a: T.TIME)
 _reg("time_to_micros time_to_millis time_to_seconds time_diff",
      lambda a: T.I64)
+_reg("to_avro", lambda a: T.BINARY)
+_reg("from_avro", lambda a: T.NULL)   # real type resolved at eval (schema arg)
+_reg("schema_of_avro", lambda a: T.STRING)

@@ -1256,6 +1256,14 @@ class Resolver:
                     return e
                 raise ResolutionError(f"unknown function {e.name}")
             # structural return types
+            if e.name == "from_avro":
+                # type = struct derived from the literal json schema arg
+                if len(e.args) > 1 and isinstance(e.args[1], S.Literal):
+                    import json as _json
+
+                    from ..engine.functions_ext import _struct_type_from_avro
+
+                    t = _struct_type_from_avro(_json.loads(e.args[1].value))
             if e.name == "coalesce" or e.name in ("nvl", "ifnull"):
                 tt = e.args[0].dtype
                 for a in e.args[1:]:

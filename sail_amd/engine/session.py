@@ -352,6 +352,9 @@ class SessionContext:
             # published AFTER execution so `SELECT * FROM system_operators`
             # reads the previous query's trace, not its own empty one
             self.last_trace = ctx.tracer.trace
+            from ..utils.telemetry import maybe_export
+
+            maybe_export(self, self.last_trace, None)
         return out
 
     @property

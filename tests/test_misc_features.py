@@ -193,3 +193,31 @@ def test_window_time(s):
     rows = s.sql(
         "SELECT window_time(window(to_timestamp(ts), '1 hour')) FROM win_ev3").collect()
     assert rows == [(1704070799999999,)]
+
+
+def test_otlp_file_export(tmp_path, monkeypatch):
+    """Telemetry export (ref: sail-telemetry OTLP): traced queries append
+    OTLP/JSON ResourceSpans + ResourceMetrics lines."""
+    import json
+
+    import sail_amd
+
+    out = str(tmp_path / "otel" / "trace.jsonl")
+    monkeypatch.setenv("SAIL_TRACE", "1")
+    monkeypatch.setenv("SAIL_OTEL_FILE", out)
+    s = sail_amd.SessionContext(device="cpu")
+    s.create_dataframe({"a": [1, 2, 3]}, name="otel_t")
+    assert s.sql("SELECT sum(a) FROM otel_t WHERE a > 1").collect() == [(5,)]
+    lines = [json.loads(l) for l in open(out)]
+    spans_line = next(l for l in lines if "resourceSpans" in l)
+    metrics_line = next(l for l in lines if "resourceMetrics" in l)
+    spans = spans_line["resourceSpans"][0]["scopeSpans"][0]["spans"]
+    names = {sp["name"] for sp in spans}
+    assert "ExecutePlan" in names and "Aggregate" in names
+    root = next(sp for sp in spans if sp["name"] == "ExecutePlan")
+    children = [sp for sp in spans if sp.get("parentSpanId") == root["spanId"]]
+    assert children  # operator spans nest under the query span
+    ms = metrics_line["resourceMetrics"][0]["scopeMetrics"][0]["metrics"]
+    mnames = {m["name"] for m in ms}
+    assert {"execution.output_row_count", "execution.elapsed_compute_time",
+            "session.query_count"} <= mnames

@@ -836,10 +836,14 @@ _IMPLS["l1"] = _f_vec2(lambda x, y: sum(abs(p - q) for p, q in zip(x, y)))
 
 def _cast_named(tt):
     def run(args, out, chunk, ev):
-        c = _col(args[0], chunk)
-        from .eval import cast_column
+        from . import eval as _ev
+        from ..plan import spec as S
 
-        return cast_column(c, tt, try_=False)
+        c = _col(args[0], chunk)
+        col = _ev.cast_column(c, tt) if hasattr(_ev, "cast_column") else None
+        if col is not None:
+            return col
+        raise NotImplementedError("cast helper unavailable")
     return run
 
 
@@ -1087,3 +1091,112 @@ def _f_schema_of_avro(args, out, chunk, ev):
 _IMPLS["to_avro"] = _f_to_avro
 _IMPLS["from_avro"] = _f_from_avro
 _IMPLS["schema_of_avro"] = _f_schema_of_avro
+
+
+# ---------------------------------------------------------------------------
+# function-call forms of operators (PySpark calls these as functions)
+# ---------------------------------------------------------------------------
+
+def _f_like_fn(pattern_ci: bool, regex: bool):
+    def run(args, out, chunk, ev):
+        c = _col(args[0], chunk)
+        pats = _col(args[1], chunk).to_pylist()
+        vals = c.to_pylist()
+        import re as _re
+
+        res = []
+        for v, p in zip(vals, pats):
+            if v is None or p is None:
+                res.append(None)
+                continue
+            if regex:
+                res.append(bool(_re.search(p, v)))
+            else:
+                rx = "^" + _re.escape(p).replace("%", ".*").replace(
+                    "_", ".").replace("\\%", "%").replace("\\_", "_") + "$"
+                flags = _re.IGNORECASE if pattern_ci else 0
+                res.append(bool(_re.match(rx, v, flags)))
+        return _ret(res, T.BOOL, chunk)
+    return run
+
+
+_IMPLS["like"] = _f_like_fn(False, False)
+_IMPLS["ilike"] = _f_like_fn(True, False)
+_IMPLS["rlike"] = _IMPLS["regexp"] = _IMPLS["regexp_like"] = \
+    _f_like_fn(False, True)
+_IMPLS["positive"] = _hostn(lambda v: v)
+_IMPLS["negative"] = _hostn(lambda v: None if v is None else -v)
+_IMPLS["mod"] = _hostn(
+    lambda a, b: None if a is None or b is None or b == 0 else
+    a - b * int(a / b) if isinstance(a, int) and isinstance(b, int)
+    else __import__("math").fmod(a, b))
+
+
+def _f_array_size(args, out, chunk, ev):
+    return _IMPLS["size"](args, out, chunk, ev)
+
+
+def _f_shuffle(args, out, chunk, ev):
+    import random as _r
+
+    from .column import ListColumn
+
+    c = _col(args[0], chunk)
+    rows = c.to_pylist()
+    out_rows = []
+    for r in rows:
+        if r is None:
+            out_rows.append(None)
+            continue
+        r = list(r)
+        _r.shuffle(r)
+        out_rows.append(r)
+    elem = c.dtype.element if isinstance(c.dtype, T.ArrayType) else T.I64
+    return ListColumn.from_pylist(out_rows, elem, device=str(chunk.device))
+
+
+_IMPLS["array_size"] = _f_array_size
+_IMPLS["shuffle"] = _f_shuffle
+
+
+def _f_isnull_fn(neg: bool):
+    def run(args, out, chunk, ev):
+        c = _col(args[0], chunk)
+        m = c.valid_mask()
+        return Column(T.BOOL, m if neg else ~m, None)
+    return run
+
+
+_IMPLS["isnull"] = _f_isnull_fn(False)
+_IMPLS["isnotnull"] = _f_isnull_fn(True)
+
+
+def _f_rand(normal: bool):
+    def run(args, out, chunk, ev):
+        n = max(chunk.num_rows, 1)
+        gen = torch.Generator(device="cpu")
+        if args:
+            seed = _col(args[0], chunk).to_pylist()
+            if seed and seed[0] is not None:
+                gen.manual_seed(int(seed[0]))
+        fn = torch.randn if normal else torch.rand
+        data = fn(chunk.num_rows if chunk.num_rows else 1,
+                  generator=gen, dtype=torch.float64).to(chunk.device)
+        return Column(T.F64, data[:chunk.num_rows] if chunk.num_rows
+                      else data, None)
+    return run
+
+
+_IMPLS["rand"] = _IMPLS["random"] = _f_rand(False)
+_IMPLS["randn"] = _f_rand(True)
+
+
+def _f_uuid(args, out, chunk, ev):
+    import uuid as _uuid
+
+    n = chunk.num_rows
+    return StringColumn.from_pylist([str(_uuid.uuid4()) for _ in range(n)]
+                                    or [str(_uuid.uuid4())],
+                                    device=str(chunk.device),
+                                    dict_encode=False)
+_IMPLS["uuid"] = _f_uuid

@@ -536,6 +536,24 @@ class _ColumnDecoder:
             at += r[2]
         return _ext().pq_rle_decode(self.buf, _page_table(rows, self.device), at)
 
+    def _decode_dict_strings(self, ext, ch: _Chunk):
+        """Decode a PLAIN string dictionary page on the device:
+        (offsets int64[n+1], bytes u8) tensors."""
+        if ch.dict_off is None:
+            raise Unsupported(f"{self.sc.name}: dict-encoded page, no dict")
+        rows = [(self._off(ch, ch.dict_off), ch.dict_len, ch.dict_nvals,
+                 0, 0, 0)]
+        table = _page_table(rows, self.device)
+        lengths, src_pos = ext.pq_bytearray_walk(self.buf, table,
+                                                 ch.dict_nvals)
+        offsets = torch.zeros(ch.dict_nvals + 1, dtype=torch.int64,
+                              device=self.device)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        total = int(offsets[-1].item())
+        blob = ext.pq_gather_strings(self.buf, src_pos, lengths,
+                                     offsets[:-1], total)
+        return offsets, blob
+
     def _dict_values_tensor(self, ch: _Chunk) -> torch.Tensor:
         vals = self._dict_values_host(ch)
         if isinstance(vals, np.ndarray):
@@ -574,41 +592,45 @@ class _ColumnDecoder:
                                       0, p.bw0))
                     pi += 1
             codes_dense = self._decode_dict_codes(ext, dict_rows)
-            # merge per-chunk dictionaries into one (host; dicts are small)
-            merged: Dict[bytes, int] = {}
-            remaps = []
-            for ch in self.chunks:
-                dvals = self._dict_values_host(ch)
-                remap = np.empty(len(dvals), dtype=np.int32)
-                for i, v in enumerate(dvals):
-                    j = merged.get(v)
-                    if j is None:
-                        j = len(merged)
-                        merged[v] = j
-                    remap[i] = j
-                remaps.append(remap)
-            if len(self.chunks) > 1:
+            # decode every chunk's dictionary PAGE on the device (a
+            # ClickBench URL dictionary is ~1M entries per row group —
+            # host loops took minutes; device walk+gather takes ms)
+            dict_cols = [self._decode_dict_strings(ext, ch)
+                         for ch in self.chunks]
+            if len(dict_cols) == 1:
+                d_offs, d_bytes = dict_cols[0]
+            else:
+                # merge: exact codes over the concatenated dictionaries,
+                # then remap each chunk's codes through its entry codes
+                from ..engine.joins import exact_string_codes
+
+                lens = torch.cat([o[1:] - o[:-1] for o, _ in dict_cols])
+                offs = torch.zeros(lens.numel() + 1, dtype=torch.int64,
+                                   device=dev)
+                torch.cumsum(lens, 0, out=offs[1:])
+                blob = torch.cat([b for _, b in dict_cols])
+                comb = StringColumn(offs, blob, None, None)
+                entry_codes = exact_string_codes([comb])[0]
+                n_merged = int(entry_codes.max().item()) + 1                     if entry_codes.numel() else 0
+                rep = torch.zeros(n_merged, dtype=torch.int64, device=dev)
+                rep.scatter_(0, entry_codes,
+                             torch.arange(entry_codes.numel(), device=dev))
+                merged_col = comb.gather(rep)
+                d_offs, d_bytes = merged_col.offsets, merged_col.bytes_
                 at = 0
                 pi = 0
-                for ch, remap in zip(self.chunks, remaps):
+                ebase = 0
+                for ci, ch in enumerate(self.chunks):
                     ch_n = sum(dense_counts[pi + k]
                                for k in range(len(ch.pages)))
-                    if not np.array_equal(
-                            remap, np.arange(len(remap), dtype=np.int32)):
-                        rt = torch.from_numpy(remap).to(dev)
-                        seg = codes_dense[at:at + ch_n]
-                        codes_dense[at:at + ch_n] = rt.index_select(
-                            0, seg.to(torch.int64))
+                    n_entries = dict_cols[ci][0].numel() - 1
+                    remap = entry_codes[ebase:ebase + n_entries]
+                    seg = codes_dense[at:at + ch_n]
+                    codes_dense[at:at + ch_n] = remap.index_select(
+                        0, seg.to(torch.int64)).to(torch.int32)
+                    ebase += n_entries
                     at += ch_n
                     pi += len(ch.pages)
-            dvals = list(merged.keys())
-            offs = np.zeros(len(dvals) + 1, dtype=np.int64)
-            for i, v in enumerate(dvals):
-                offs[i + 1] = offs[i] + len(v)
-            blob = b"".join(dvals)
-            d_offs = torch.from_numpy(offs).to(dev)
-            d_bytes = torch.from_numpy(
-                np.frombuffer(blob, dtype=np.uint8).copy()).to(dev)
             codes = self._scatter_codes(codes_dense, validity, dense_counts)
             return StringColumn(d_offs, d_bytes, validity, codes)
 

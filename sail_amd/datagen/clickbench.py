@@ -159,4 +159,5 @@ def register_clickbench_parquet(session, rows: int = 100_000_000, device=None,
         session, tables, data_dir=data_dir,
         default_dir=f"sail_hits_{rows // 1_000_000}m", rank=rank, world=world,
         write_fn=lambda tbls, d, r: write_tpch_parquet(
-            tbls, d, rank=r, page_size=1 << 20))
+            tbls, d, rank=r, page_size=1 << 20,
+            row_group_rows=16_000_000))

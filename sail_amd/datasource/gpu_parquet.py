@@ -178,6 +178,10 @@ class FileIndex:
 
 
 _INDEX_CACHE: Dict[Tuple[str, float, int], FileIndex] = {}
+#: merged device dictionaries per (file index id, column) — dictionary
+#: pages are file metadata-scale (MBs vs the GBs of codes re-decoded per
+#: scan); cached like the page index
+_DICT_CACHE: Dict[tuple, tuple] = {}
 
 
 def file_index(path: str) -> FileIndex:
@@ -536,6 +540,23 @@ class _ColumnDecoder:
             at += r[2]
         return _ext().pq_rle_decode(self.buf, _page_table(rows, self.device), at)
 
+    def _remap_codes(self, codes_dense, dense_counts, entry_codes,
+                     entry_bases):
+        """Map per-chunk dictionary codes to merged-dictionary codes in
+        place (no-op for single-dictionary files)."""
+        if entry_codes is None:
+            return
+        at = 0
+        pi = 0
+        for ci, ch in enumerate(self.chunks):
+            ch_n = sum(dense_counts[pi + k] for k in range(len(ch.pages)))
+            base = entry_bases[ci]
+            seg = codes_dense[at:at + ch_n]
+            codes_dense[at:at + ch_n] = entry_codes.index_select(
+                0, seg.to(torch.int64) + base).to(torch.int32)
+            at += ch_n
+            pi += len(ch.pages)
+
     def _decode_dict_strings(self, ext, ch: _Chunk):
         """Decode a PLAIN string dictionary page on the device:
         (offsets int64[n+1], bytes u8) tensors."""
@@ -592,6 +613,18 @@ class _ColumnDecoder:
                                       0, p.bw0))
                     pi += 1
             codes_dense = self._decode_dict_codes(ext, dict_rows)
+            cached = _DICT_CACHE.get((id(self.idx), self.ci))
+            if cached is not None:
+                # merged dictionary + per-chunk entry remaps are a pure
+                # function of the FILE (like the page index): cache them
+                # on device — the per-row codes (the bulk) are still
+                # re-read and re-decoded on every scan
+                d_offs, d_bytes, entry_codes, entry_bases = cached
+                self._remap_codes(codes_dense, dense_counts, entry_codes,
+                                  entry_bases)
+                codes = self._scatter_codes(codes_dense, validity,
+                                            dense_counts)
+                return StringColumn(d_offs, d_bytes, validity, codes)
             # decode every chunk's dictionary PAGE on the device (a
             # ClickBench URL dictionary is ~1M entries per row group —
             # host loops took minutes; device walk+gather takes ms)
@@ -599,6 +632,8 @@ class _ColumnDecoder:
                          for ch in self.chunks]
             if len(dict_cols) == 1:
                 d_offs, d_bytes = dict_cols[0]
+                _DICT_CACHE[(id(self.idx), self.ci)] = (d_offs, d_bytes,
+                                                        None, None)
             else:
                 # merge: exact codes over the concatenated dictionaries,
                 # then remap each chunk's codes through its entry codes
@@ -617,20 +652,15 @@ class _ColumnDecoder:
                              torch.arange(entry_codes.numel(), device=dev))
                 merged_col = comb.gather(rep)
                 d_offs, d_bytes = merged_col.offsets, merged_col.bytes_
-                at = 0
-                pi = 0
-                ebase = 0
-                for ci, ch in enumerate(self.chunks):
-                    ch_n = sum(dense_counts[pi + k]
-                               for k in range(len(ch.pages)))
-                    n_entries = dict_cols[ci][0].numel() - 1
-                    remap = entry_codes[ebase:ebase + n_entries]
-                    seg = codes_dense[at:at + ch_n]
-                    codes_dense[at:at + ch_n] = remap.index_select(
-                        0, seg.to(torch.int64)).to(torch.int32)
-                    ebase += n_entries
-                    at += ch_n
-                    pi += len(ch.pages)
+                entry_bases = []
+                eb = 0
+                for o, _b2 in dict_cols:
+                    entry_bases.append(eb)
+                    eb += o.numel() - 1
+                _DICT_CACHE[(id(self.idx), self.ci)] = (
+                    d_offs, d_bytes, entry_codes, entry_bases)
+                self._remap_codes(codes_dense, dense_counts, entry_codes,
+                                  entry_bases)
             codes = self._scatter_codes(codes_dense, validity, dense_counts)
             return StringColumn(d_offs, d_bytes, validity, codes)
 

@@ -184,6 +184,56 @@ _INDEX_CACHE: Dict[Tuple[str, float, int], FileIndex] = {}
 _DICT_CACHE: Dict[tuple, tuple] = {}
 
 
+def _lex_perm(offsets: torch.Tensor, blob: torch.Tensor) -> torch.Tensor:
+    """Permutation that lex-sorts the strings of a (offsets, blob) pair.
+
+    Parquet dictionary pages are in writer first-occurrence order, but the
+    engine's dict-code invariant is code order == byte order (StringColumn
+    min/max, comparisons and ORDER BY all compare codes). LSD-stable
+    argsorts over big-endian 8-byte words restore it on device; 0x00
+    padding past each string's end makes prefixes sort first."""
+    n = offsets.numel() - 1
+    dev = offsets.device
+    if n <= 1:
+        return torch.arange(n, device=dev)
+    lens = offsets[1:] - offsets[:-1]
+    maxw = (int(lens.max().item()) + 7) // 8
+    perm = torch.arange(n, device=dev)
+    if maxw == 0:
+        return perm
+    starts = offsets[:-1]
+    ends = offsets[1:]
+    pad_blob = torch.cat([blob, blob.new_zeros(8)])
+    sentinel = blob.numel()
+    byte_off = torch.arange(8, device=dev)
+    for w in range(maxw - 1, -1, -1):
+        idx = starts.unsqueeze(1) + w * 8 + byte_off
+        idx = torch.where(idx < ends.unsqueeze(1), idx,
+                          torch.full_like(idx, sentinel))
+        b = pad_blob.index_select(0, idx.reshape(-1)) \
+            .reshape(n, 8).to(torch.int64)
+        key = ((b[:, 0] << 56) | (b[:, 1] << 48) | (b[:, 2] << 40) |
+               (b[:, 3] << 32) | (b[:, 4] << 24) | (b[:, 5] << 16) |
+               (b[:, 6] << 8) | b[:, 7])
+        key = key ^ (-(1 << 63))  # top bit flip: unsigned byte order
+        perm = perm.index_select(
+            0, torch.argsort(key.index_select(0, perm), stable=True))
+    return perm
+
+
+def _lex_sort_dict(d_offs: torch.Tensor, d_bytes: torch.Tensor):
+    """Sort a device dictionary lexicographically.
+    Returns (sorted_offsets, sorted_bytes, old_code -> new_code map)."""
+    from ..engine.column import StringColumn
+
+    perm = _lex_perm(d_offs, d_bytes)
+    n = perm.numel()
+    old_to_new = torch.empty(n, dtype=torch.int64, device=perm.device)
+    old_to_new.scatter_(0, perm, torch.arange(n, device=perm.device))
+    col = StringColumn(d_offs, d_bytes, None, None).gather(perm)
+    return col.offsets, col.bytes_, old_to_new
+
+
 def file_index(path: str) -> FileIndex:
     st = os.stat(path)
     key = (os.path.abspath(path), st.st_mtime, st.st_size)
@@ -631,9 +681,15 @@ class _ColumnDecoder:
             dict_cols = [self._decode_dict_strings(ext, ch)
                          for ch in self.chunks]
             if len(dict_cols) == 1:
-                d_offs, d_bytes = dict_cols[0]
-                _DICT_CACHE[(id(self.idx), self.ci)] = (d_offs, d_bytes,
-                                                        None, None)
+                # parquet dictionary pages are first-occurrence ordered;
+                # re-sort to the engine's code-order == byte-order invariant
+                o0, b0 = dict_cols[0]
+                d_offs, d_bytes, old_to_new = _lex_sort_dict(o0, b0)
+                entry_codes, entry_bases = old_to_new, [0]
+                _DICT_CACHE[(id(self.idx), self.ci)] = (
+                    d_offs, d_bytes, entry_codes, entry_bases)
+                self._remap_codes(codes_dense, dense_counts, entry_codes,
+                                  entry_bases)
             else:
                 # merge: exact codes over the concatenated dictionaries,
                 # then remap each chunk's codes through its entry codes
@@ -651,7 +707,11 @@ class _ColumnDecoder:
                 rep.scatter_(0, entry_codes,
                              torch.arange(entry_codes.numel(), device=dev))
                 merged_col = comb.gather(rep)
-                d_offs, d_bytes = merged_col.offsets, merged_col.bytes_
+                # merged codes are hash-ordered: lex-sort the merged
+                # dictionary and compose the remap (invariant above)
+                d_offs, d_bytes, old_to_new = _lex_sort_dict(
+                    merged_col.offsets, merged_col.bytes_)
+                entry_codes = old_to_new.index_select(0, entry_codes)
                 entry_bases = []
                 eb = 0
                 for o, _b2 in dict_cols:

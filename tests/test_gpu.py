@@ -436,6 +436,28 @@ def test_gpu_parquet_sql_roundtrip(tmp_path, gpu_session):
         os.environ["SAIL_IO_GPU_PARQUET"] = "auto"
 
 
+def test_gpu_parquet_dict_lex_sorted(tmp_path):
+    """Scanned dictionaries must come back lex-sorted (engine invariant:
+    dict code order == byte order, relied on by min/max/compare/sort) —
+    parquet dictionary pages are first-occurrence ordered on disk."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from sail_amd.datasource import gpu_parquet as G
+
+    p = str(tmp_path / "dl.parquet")
+    vals = [f"w{(i * 131) % 997:03d}" for i in range(60_000)]
+    pq.write_table(pa.table({"s": pa.array(vals)}), p, compression="NONE",
+                   use_dictionary=["s"], data_page_version="1.0",
+                   row_group_size=20_000)
+    for _ in range(2):  # second read exercises the device dict cache
+        out = G.read_gpu([p], [("s", None)], "cuda:0")
+        c = out.columns["s"]
+        assert c.is_dict
+        dv = c.dict_values()
+        assert dv == sorted(dv)
+        assert c.to_pylist() == vals
+
+
 def test_string_hash2_and_pairs_equal_match_cpu(ext):
     from sail_amd.engine import joins
 

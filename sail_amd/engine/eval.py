@@ -428,6 +428,10 @@ class Evaluator:
             if isinstance(e.dtype, T.DecimalType) and v is not None:
                 # to_pylist already unscaled; keep as float for rebroadcast
                 return Scalar(v, e.dtype)
+            if isinstance(out.dtype, (T.GeometryType, T.GeographyType)):
+                # st_setsrid refines the SRID at eval time; the resolver
+                # type only knows the default
+                return Scalar(v, out.dtype)
             return Scalar(v, e.dtype or out.dtype)
         return dispatch_function(e.name, args, e.dtype, chunk, self)
 

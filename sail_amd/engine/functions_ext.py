@@ -1200,3 +1200,113 @@ def _f_uuid(args, out, chunk, ev):
                                     device=str(chunk.device),
                                     dict_encode=False)
 _IMPLS["uuid"] = _f_uuid
+
+
+# ---------------------------------------------------------------------------
+# geo (ref: sail-plan/src/function/scalar/geo.rs — st_geomfromwkb /
+# st_geogfromwkb / st_asbinary / st_setsrid / st_srid; geometry model is
+# WKB bytes + SRID per sail-common/src/spec/data_type.rs:273-287)
+# ---------------------------------------------------------------------------
+
+import struct as _struct
+
+
+def _wkb_walk(b: bytes, pos: int) -> int:
+    """Validate one WKB geometry starting at `pos`; returns end offset.
+    Accepts ISO WKB (Z=+1000/M=+2000/ZM=+3000 type codes) and EWKB
+    (Z/M/SRID flag bits). Raises ValueError on malformed input."""
+    try:
+        if pos + 5 > len(b):
+            raise ValueError("truncated header")
+        bo = b[pos]
+        if bo not in (0, 1):
+            raise ValueError(f"bad byte order {bo}")
+        fmt = "<" if bo == 1 else ">"
+        (tcode,) = _struct.unpack_from(fmt + "I", b, pos + 1)
+        pos += 5
+        if tcode & 0x20000000:  # EWKB embedded SRID
+            pos += 4
+        z = bool(tcode & 0x80000000)
+        m = bool(tcode & 0x40000000)
+        base = tcode & 0x0FFFFFFF
+        iso_extra, t = divmod(base, 1000)
+        if iso_extra not in (0, 1, 2, 3):
+            raise ValueError(f"bad type code {tcode}")
+        z = z or iso_extra in (1, 3)
+        m = m or iso_extra in (2, 3)
+        step = 8 * (2 + int(z) + int(m))
+        if t == 1:  # Point
+            pos += step
+        elif t == 2:  # LineString
+            (n,) = _struct.unpack_from(fmt + "I", b, pos)
+            pos += 4 + n * step
+        elif t == 3:  # Polygon
+            (nr,) = _struct.unpack_from(fmt + "I", b, pos)
+            pos += 4
+            for _ in range(nr):
+                (n,) = _struct.unpack_from(fmt + "I", b, pos)
+                pos += 4 + n * step
+        elif t in (4, 5, 6, 7):  # Multi*/GeometryCollection
+            (n,) = _struct.unpack_from(fmt + "I", b, pos)
+            pos += 4
+            for _ in range(n):
+                pos = _wkb_walk(b, pos)
+        else:
+            raise ValueError(f"bad geometry type {t}")
+        if pos > len(b):
+            raise ValueError("truncated body")
+        return pos
+    except _struct.error as e:
+        raise ValueError(f"malformed WKB: {e}") from None
+
+
+def _wkb_check(v: bytes) -> bytes:
+    end = _wkb_walk(v, 0)
+    if end != len(v):
+        raise ValueError(f"WKB: {len(v) - end} trailing bytes")
+    return v
+
+
+def _geom_dtype(args, chunk):
+    c = _col(args[0], chunk)
+    if not isinstance(c.dtype, (T.GeometryType, T.GeographyType)):
+        raise ValueError(f"expected GEOMETRY/GEOGRAPHY, got {c.dtype!r}")
+    return c
+
+
+def _f_geo_from_wkb(geog):
+    def run(args, out, chunk, ev):
+        c = _col(args[0], chunk)
+        vals = [None if v is None else _wkb_check(_b(v))
+                for v in c.to_pylist()]
+        dt = T.GeographyType() if geog else T.GeometryType()
+        return _ret(vals, dt, chunk)
+    return run
+
+
+def _f_st_asbinary(args, out, chunk, ev):
+    c = _geom_dtype(args, chunk)
+    return _ret(c.to_pylist(), T.BINARY, chunk)
+
+
+def _f_st_srid(args, out, chunk, ev):
+    c = _geom_dtype(args, chunk)
+    vals = [None if v is None else c.dtype.srid for v in c.to_pylist()]
+    return _ret(vals, T.I32, chunk)
+
+
+def _f_st_setsrid(args, out, chunk, ev):
+    c = _geom_dtype(args, chunk)
+    srid_vals = _col(args[1], chunk).to_pylist()
+    if not srid_vals:
+        return c
+    srid = int(srid_vals[0])
+    dt = type(c.dtype)(srid)
+    return _ret(c.to_pylist(), dt, chunk)
+
+
+_IMPLS["st_geomfromwkb"] = _f_geo_from_wkb(False)
+_IMPLS["st_geogfromwkb"] = _f_geo_from_wkb(True)
+_IMPLS["st_asbinary"] = _f_st_asbinary
+_IMPLS["st_srid"] = _f_st_srid
+_IMPLS["st_setsrid"] = _f_st_setsrid

@@ -149,6 +149,30 @@ class BinaryType(StringType):
 
 
 @dataclass(frozen=True, eq=True)
+class GeometryType(BinaryType):
+    """2D spatial data, Cartesian coordinates; stored as WKB bytes with an
+    SRID (Spark 4.1 GeometryType; ref: crates/sail-common/src/spec/
+    data_type.rs:273-287 — geoarrow.wkb extension over Binary). Physical
+    layout is the binary (offsets, bytes) pair."""
+
+    srid: int = 4326
+
+    def __repr__(self):
+        return f"geometry({self.srid})"
+
+
+@dataclass(frozen=True, eq=True)
+class GeographyType(BinaryType):
+    """2D spatial data, spherical (geodetic) coordinates; WKB + SRID
+    (Spark 4.1 GeographyType, spherical edge interpolation)."""
+
+    srid: int = 4326
+
+    def __repr__(self):
+        return f"geography({self.srid})"
+
+
+@dataclass(frozen=True, eq=True)
 class ArrayType(DataType):
     element: DataType = field(default_factory=Int64Type)
 
@@ -229,6 +253,11 @@ def type_name(t: DataType) -> str:
 
 def type_from_name(name: str) -> DataType:
     base = name.strip().lower()
+    if base.startswith("geometry") or base.startswith("geography"):
+        srid = int(base[base.find("(") + 1:base.find(")")]) \
+            if "(" in base else 4326
+        return GeometryType(srid) if base.startswith("geometry") \
+            else GeographyType(srid)
     if base.startswith("decimal") or base.startswith("numeric"):
         inner = base[base.find("(") + 1 : base.find(")")] if "(" in base else "10,0"
         p, _, s = inner.partition(",")

@@ -331,3 +331,10 @@ _reg("time_to_micros time_to_millis time_to_seconds time_diff",
 _reg("to_avro", lambda a: T.BINARY)
 _reg("from_avro", lambda a: T.NULL)   # real type resolved at eval (schema arg)
 _reg("schema_of_avro", lambda a: T.STRING)
+
+# geo (ref: sail-plan/src/function/scalar/geo.rs; WKB + SRID model)
+_reg("st_geomfromwkb", lambda a: T.GeometryType())
+_reg("st_geogfromwkb", lambda a: T.GeographyType())
+_reg("st_asbinary", lambda a: T.BINARY)
+_reg("st_srid", lambda a: T.I32)
+_reg("st_setsrid", lambda a: a[0])  # SRID literal applied at eval

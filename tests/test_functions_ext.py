@@ -187,3 +187,75 @@ def test_avro_codec_roundtrip(s):
     b64 = base64.b64encode(bytes(buf)).decode()
     r3 = q(s, f"SELECT from_avro(unbase64('{b64}'), '{schema}').b")
     assert r3 == [("x",)]
+
+
+# ---------------------------------------------------------------------------
+# geo (st_geomfromwkb / st_geogfromwkb / st_asbinary / st_srid / st_setsrid
+# — ref: sail-plan/src/function/scalar/geo.rs, WKB + SRID model)
+# ---------------------------------------------------------------------------
+
+def _wkb_point(x, y, bo="<"):
+    import struct
+    order = 1 if bo == "<" else 0
+    return struct.pack(bo + "BIdd" if bo == "<" else ">BIdd",
+                       order, 1, x, y)
+
+
+class TestGeo:
+    def test_point_roundtrip_and_srid(self, session):
+        h = _wkb_point(1.5, -2.5).hex()
+        assert session.sql(
+            f"SELECT st_srid(st_geomfromwkb(unhex('{h}')))").collect() == \
+            [(4326,)]
+        assert session.sql(
+            f"SELECT st_asbinary(st_geomfromwkb(unhex('{h}')))"
+        ).collect()[0][0] == bytes.fromhex(h)
+        assert session.sql(
+            f"SELECT st_srid(st_setsrid(st_geomfromwkb(unhex('{h}')), 3857))"
+        ).collect() == [(3857,)]
+        # geography accepts the same WKB
+        assert session.sql(
+            f"SELECT st_srid(st_geogfromwkb(unhex('{h}')))").collect() == \
+            [(4326,)]
+
+    def test_wkb_shapes_and_flags(self):
+        import struct
+        from sail_amd.engine.functions_ext import _wkb_check
+
+        # big-endian point
+        _wkb_check(struct.pack(">BIdd", 0, 1, 1.0, 2.0))
+        # ISO Z point (type 1001): 3 doubles
+        _wkb_check(struct.pack("<BIddd", 1, 1001, 1.0, 2.0, 3.0))
+        # EWKB Z flag + SRID flag
+        _wkb_check(struct.pack("<BIIddd", 1, 1 | 0x80000000 | 0x20000000,
+                               4326, 1.0, 2.0, 3.0))
+        # linestring of 2 points
+        _wkb_check(struct.pack("<BII", 1, 2, 2) + struct.pack("<4d", *range(4)))
+        # polygon: 1 ring x 4 points
+        _wkb_check(struct.pack("<BIII", 1, 3, 1, 4) +
+                   struct.pack("<8d", *range(8)))
+        # multipoint of 2 + collection of 1 point
+        pt = struct.pack("<BIdd", 1, 1, 0.0, 0.0)
+        _wkb_check(struct.pack("<BII", 1, 4, 2) + pt + pt)
+        _wkb_check(struct.pack("<BII", 1, 7, 1) + pt)
+
+    def test_wkb_malformed_rejected(self, session):
+        import pytest as _pt
+        import struct
+        from sail_amd.engine.functions_ext import _wkb_check
+
+        for bad in (b"", b"\x02", struct.pack("<BI", 1, 99),
+                    struct.pack("<BIdd", 1, 1, 0.0, 0.0) + b"xx",  # trailing
+                    struct.pack("<BId", 1, 1, 0.0)):  # truncated point
+            with _pt.raises(ValueError):
+                _wkb_check(bad)
+
+    def test_column_path_not_constant_folded(self, session):
+        h = _wkb_point(3.0, 4.0).hex()
+        session.sql(
+            f"CREATE TEMP VIEW geo_t AS SELECT unhex('{h}') AS b "
+            "FROM range(4)")
+        r = session.sql(
+            "SELECT st_srid(st_setsrid(st_geomfromwkb(b), 27700)) "
+            "FROM geo_t").collect()
+        assert r == [(27700,)] * 4

@@ -5,8 +5,12 @@ the suspect is scatter_reduce_(amin) (round-1 profiles showed ROCm
 scatter_reduce pathologically slow at high contention). Candidates:
   A. scatter_reduce_(amin)            (current aggregates.py path)
   B. sort packed (gid*S + val), segment-first  (rocPRIM radix underneath)
-  C. index_put-free: sort by val once, scatter_ winners (last write wins
-     on a descending-sorted value order => min)
+  C. (REMOVED) sort by val + scatter_ "last write wins": INVALID on GPU —
+     scatter_ with duplicate indices has unspecified write order, measured
+     wrong on MI355X (the assert caught it). Kept here as the record.
+Measured on MI355X (2026-09-12): the aggregate path is no longer a
+bottleneck after the dict lex-sort fix (ClickBench q21 736->40 ms), so
+the scatter_reduce path stays.
 Run: python tools/bench_minmax.py [n_rows] [n_groups] [val_span]
 """
 import sys
@@ -49,20 +53,11 @@ def main():
         out[gs[first]] = s[first] - gs[first] * span
         return out
 
-    def c_sort_vals_scatter():
-        order = torch.argsort(vals, descending=True)
-        out = torch.full((ng,), 2**62, dtype=torch.int64, device=dev)
-        out.scatter_(0, gid.index_select(0, order),
-                     vals.index_select(0, order))
-        return out
-
-    ra, rb, rc = a_scatter_reduce(), b_sort_packed(), c_sort_vals_scatter()
+    ra, rb = a_scatter_reduce(), b_sort_packed()
     assert torch.equal(ra, rb), "B mismatch"
-    assert torch.equal(ra, rc), "C mismatch"
     print(f"n={n} ng={ng} span={span}")
     print(f"A scatter_reduce amin : {timeit(a_scatter_reduce):8.2f} ms")
     print(f"B sort packed         : {timeit(b_sort_packed):8.2f} ms")
-    print(f"C sort vals + scatter : {timeit(c_sort_vals_scatter):8.2f} ms")
 
     # the other q21 pieces at the same shape, for the 706 ms budget
     def count():

@@ -48,6 +48,19 @@ class Column:
         if isinstance(dtype, T.ArrayType):  # nested lists
             return ListColumn.from_pylist(list(values), dtype.element,
                                           device=device)
+        if isinstance(dtype, T.StructType):  # rows are dicts (or None)
+            vals = list(values)
+            fields = []
+            for f in dtype.fields:
+                fv = [None if r is None else r.get(f.name) for r in vals]
+                fields.append((f.name,
+                               Column.from_values(fv, f.dtype, device=device)))
+            validity = None
+            if any(r is None for r in vals):
+                validity = torch.tensor(
+                    [0 if r is None else 1 for r in vals],
+                    dtype=torch.uint8, device=device)
+            return StructColumn(fields, validity, dtype=dtype)
         validity = None
         if any(v is None for v in values):
             validity = torch.tensor([0 if v is None else 1 for v in values], dtype=torch.uint8, device=device)

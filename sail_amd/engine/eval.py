@@ -423,6 +423,13 @@ class Evaluator:
             out = dispatch_function(e.name, args, e.dtype, one, self)
             if isinstance(out, Scalar):
                 return out
+            if isinstance(out.dtype, (T.ArrayType, T.MapType, T.StructType)):
+                # nested results (e.g. from_protobuf on literals) can't be
+                # scalarized: replicate the single row to the chunk length
+                n = chunk.num_rows or 1
+                rep = out.gather(torch.zeros(n, dtype=torch.int64,
+                                             device=out.device))
+                return rep.to(chunk.device)
             vals = out.to_pylist()
             v = vals[0] if vals else None
             if isinstance(e.dtype, T.DecimalType) and v is not None:

@@ -1310,3 +1310,159 @@ _IMPLS["st_geogfromwkb"] = _f_geo_from_wkb(True)
 _IMPLS["st_asbinary"] = _f_st_asbinary
 _IMPLS["st_srid"] = _f_st_srid
 _IMPLS["st_setsrid"] = _f_st_setsrid
+
+
+# ---------------------------------------------------------------------------
+# protobuf codec (ref: sail-plan/src/function/scalar/misc.rs:193,296 —
+# the reference registers from_protobuf/to_protobuf but leaves them
+# unimplemented; here they work over a serialized FileDescriptorSet path,
+# the Spark signature: from_protobuf(data, messageName, descFilePath))
+# ---------------------------------------------------------------------------
+
+_PB_CACHE: dict = {}
+
+
+def _pb_class(msg_name: str, desc_path: str):
+    import os as _os
+
+    st = _os.stat(desc_path)
+    key = (desc_path, st.st_mtime, st.st_size, msg_name)
+    hit = _PB_CACHE.get(key)
+    if hit is not None:
+        return hit
+    from google.protobuf import (descriptor_pb2, descriptor_pool,
+                                 message_factory)
+
+    with open(desc_path, "rb") as f:
+        fds = descriptor_pb2.FileDescriptorSet.FromString(f.read())
+    pool = descriptor_pool.DescriptorPool()
+    for fd in fds.file:
+        pool.Add(fd)
+    desc = pool.FindMessageTypeByName(msg_name)
+    cls = message_factory.GetMessageClass(desc)
+    _PB_CACHE[key] = cls
+    return cls
+
+
+def _pb_field_type(fd):
+    from google.protobuf.descriptor import FieldDescriptor as FD
+
+    m = {FD.TYPE_DOUBLE: T.F64, FD.TYPE_FLOAT: T.F32,
+         FD.TYPE_INT64: T.I64, FD.TYPE_SINT64: T.I64,
+         FD.TYPE_SFIXED64: T.I64, FD.TYPE_UINT64: T.I64,
+         FD.TYPE_FIXED64: T.I64,
+         FD.TYPE_INT32: T.I32, FD.TYPE_SINT32: T.I32,
+         FD.TYPE_SFIXED32: T.I32, FD.TYPE_UINT32: T.I32,
+         FD.TYPE_FIXED32: T.I32,
+         FD.TYPE_BOOL: T.BOOL, FD.TYPE_STRING: T.STRING,
+         FD.TYPE_BYTES: T.BINARY, FD.TYPE_ENUM: T.STRING}
+    if fd.type == FD.TYPE_MESSAGE:
+        elem = _pb_struct_type(fd.message_type)
+    else:
+        elem = m.get(fd.type, T.STRING)
+    if fd.is_repeated:
+        return T.ArrayType(elem)
+    return elem
+
+
+def _pb_struct_type(desc) -> "T.StructType":
+    return T.StructType([T.StructField(f.name, _pb_field_type(f))
+                         for f in desc.fields])
+
+
+def _pb_to_val(msg, desc):
+    from google.protobuf.descriptor import FieldDescriptor as FD
+
+    out = {}
+    for f in desc.fields:
+        v = getattr(msg, f.name)
+        if f.is_repeated:
+            if f.type == FD.TYPE_MESSAGE:
+                out[f.name] = [_pb_to_val(x, f.message_type) for x in v]
+            elif f.type == FD.TYPE_ENUM:
+                out[f.name] = [f.enum_type.values_by_number[x].name
+                               for x in v]
+            else:
+                out[f.name] = list(v)
+        elif f.type == FD.TYPE_MESSAGE:
+            out[f.name] = _pb_to_val(v, f.message_type) \
+                if msg.HasField(f.name) else None
+        elif f.type == FD.TYPE_ENUM:
+            out[f.name] = f.enum_type.values_by_number[v].name
+        else:
+            out[f.name] = v
+    return out
+
+
+def _pb_fill(msg, desc, val: dict):
+    from google.protobuf.descriptor import FieldDescriptor as FD
+
+    for f in desc.fields:
+        v = (val or {}).get(f.name)
+        if v is None:
+            continue
+        if f.is_repeated:
+            tgt = getattr(msg, f.name)
+            for x in v:
+                if f.type == FD.TYPE_MESSAGE:
+                    _pb_fill(tgt.add(), f.message_type, x)
+                elif f.type == FD.TYPE_ENUM:
+                    tgt.append(f.enum_type.values_by_name[x].number
+                               if isinstance(x, str) else int(x))
+                else:
+                    tgt.append(x)
+        elif f.type == FD.TYPE_MESSAGE:
+            _pb_fill(getattr(msg, f.name), f.message_type, v)
+        elif f.type == FD.TYPE_ENUM:
+            setattr(msg, f.name,
+                    f.enum_type.values_by_name[v].number
+                    if isinstance(v, str) else int(v))
+        else:
+            setattr(msg, f.name, v)
+
+
+def _f_from_protobuf(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    msg_name = _col(args[1], chunk).to_pylist()[0]
+    desc_path = _col(args[2], chunk).to_pylist()[0]
+    cls = _pb_class(msg_name, desc_path)
+    desc = cls.DESCRIPTOR
+    stype = _pb_struct_type(desc)
+    rows = []
+    for v in c.to_pylist():
+        rows.append(None if v is None
+                    else _pb_to_val(cls.FromString(_b(v)), desc))
+    from .column import StructColumn
+
+    fields = []
+    for f in stype.fields:
+        fv = [None if r is None else r.get(f.name) for r in rows]
+        fields.append((f.name, Column.from_values(fv, f.dtype,
+                                                  device=str(chunk.device))))
+    validity = None
+    if any(r is None for r in rows):
+        import torch as _t
+
+        validity = _t.tensor([0 if r is None else 1 for r in rows],
+                             dtype=_t.uint8, device=chunk.device)
+    return StructColumn(fields, validity, dtype=stype)
+
+
+def _f_to_protobuf(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    msg_name = _col(args[1], chunk).to_pylist()[0]
+    desc_path = _col(args[2], chunk).to_pylist()[0]
+    cls = _pb_class(msg_name, desc_path)
+    vals = []
+    for v in c.to_pylist():
+        if v is None:
+            vals.append(None)
+            continue
+        msg = cls()
+        _pb_fill(msg, cls.DESCRIPTOR, v)
+        vals.append(msg.SerializeToString())
+    return _ret(vals, T.BINARY, chunk)
+
+
+_IMPLS["from_protobuf"] = _f_from_protobuf
+_IMPLS["to_protobuf"] = _f_to_protobuf

@@ -332,6 +332,9 @@ _reg("to_avro", lambda a: T.BINARY)
 _reg("from_avro", lambda a: T.NULL)   # real type resolved at eval (schema arg)
 _reg("schema_of_avro", lambda a: T.STRING)
 
+_reg("to_protobuf", lambda a: T.BINARY)
+_reg("from_protobuf", lambda a: T.NULL)  # struct type resolved at eval
+
 # geo (ref: sail-plan/src/function/scalar/geo.rs; WKB + SRID model)
 _reg("st_geomfromwkb", lambda a: T.GeometryType())
 _reg("st_geogfromwkb", lambda a: T.GeographyType())

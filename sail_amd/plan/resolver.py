@@ -1264,6 +1264,16 @@ class Resolver:
                     from ..engine.functions_ext import _struct_type_from_avro
 
                     t = _struct_type_from_avro(_json.loads(e.args[1].value))
+            if e.name == "from_protobuf":
+                # type = struct from the literal (messageName, descFile)
+                if len(e.args) > 2 and isinstance(e.args[1], S.Literal) \
+                        and isinstance(e.args[2], S.Literal):
+                    from ..engine.functions_ext import (_pb_class,
+                                                        _pb_struct_type)
+
+                    cls = _pb_class(str(e.args[1].value),
+                                    str(e.args[2].value))
+                    t = _pb_struct_type(cls.DESCRIPTOR)
             if e.name == "coalesce" or e.name in ("nvl", "ifnull"):
                 tt = e.args[0].dtype
                 for a in e.args[1:]:

@@ -173,7 +173,11 @@ def _udaf_eval(name, args, gid, ng, filter_mask, out_type):
     import torch as _t
 
     kept = _t.nonzero(mask, as_tuple=False).flatten()
-    vals = c.gather(kept.to(c.device)).to_pylist()
+    if len(args) > 1:  # multi-arg UDAF (e.g. tuple_sketch_agg(key, value))
+        per_arg = [a.gather(kept.to(a.device)).to_pylist() for a in args]
+        vals = list(zip(*per_arg))
+    else:
+        vals = c.gather(kept.to(c.device)).to_pylist()
     groups = [[] for _ in range(ng)]
     for g, v in zip(gl, vals):
         groups[g].append(v)

@@ -788,6 +788,82 @@ _IMPLS["theta_difference"] = _hostn(
     else theta_difference(_b(a), _b(b)))
 
 
+# tuple sketches: a theta sketch whose retained hashes carry a numeric
+# summary (sum-combined on duplicates/union/intersection). The reference
+# registers tuple_{sketch,union,intersection}_agg_{double,integer} but
+# leaves all six unimplemented (ref: sail-plan/src/function/
+# aggregate.rs:914-931); these are functional.
+_TU_MAGIC = b"SAILTUP1"
+
+
+def tuple_create(pairs, mode="d", k: int = 4096) -> bytes:
+    agg = {}
+    for kv in pairs:
+        if kv is None:
+            continue
+        key, val = kv
+        if key is None or val is None:
+            continue
+        h = _hash64(key)
+        agg[h] = agg.get(h, 0) + (float(val) if mode == "d" else int(val))
+    return _tuple_pack(mode, k, agg)
+
+
+def _tuple_pack(mode: str, k: int, agg: dict) -> bytes:
+    hs = sorted(agg)[:k]
+    fmt = "<Qd" if mode == "d" else "<Qq"
+    return _TU_MAGIC + mode.encode() + struct.pack("<II", k, len(hs)) + \
+        b"".join(struct.pack(fmt, h, agg[h]) for h in hs)
+
+
+def _tuple_parse(sk: bytes):
+    if not sk.startswith(_TU_MAGIC):
+        raise ValueError("not a sail tuple sketch")
+    mode = chr(sk[8])
+    k, n = struct.unpack_from("<II", sk, 9)
+    fmt = "<Qd" if mode == "d" else "<Qq"
+    agg = {}
+    at = 17
+    for _ in range(n):
+        h, v = struct.unpack_from(fmt, sk, at)
+        agg[h] = v
+        at += 16
+    return mode, k, agg
+
+
+def tuple_union(a: bytes, b: bytes) -> bytes:
+    ma, ka, aa = _tuple_parse(a)
+    mb, kb, ab = _tuple_parse(b)
+    if ma != mb:
+        raise ValueError("tuple_union: mixed summary types")
+    out = dict(aa)
+    for h, v in ab.items():
+        out[h] = out.get(h, 0) + v
+    return _tuple_pack(ma, min(ka, kb), out)
+
+
+def tuple_intersection(a: bytes, b: bytes) -> bytes:
+    ma, ka, aa = _tuple_parse(a)
+    mb, kb, ab = _tuple_parse(b)
+    if ma != mb:
+        raise ValueError("tuple_intersection: mixed summary types")
+    out = {h: aa[h] + ab[h] for h in aa.keys() & ab.keys()}
+    return _tuple_pack(ma, min(ka, kb), out)
+
+
+def tuple_estimate(sk: bytes) -> float:
+    mode, k, agg = _tuple_parse(sk)
+    hs = sorted(agg)
+    if len(hs) < k:
+        return float(len(hs))
+    theta = hs[-1] / float(1 << 64)
+    return (len(hs) - 1) / theta
+
+
+_IMPLS["tuple_sketch_estimate"] = _host1(
+    lambda v: tuple_estimate(_b(v)))
+
+
 # sketch-building aggregates ride the engine UDAF mechanism
 def _register_sketch_aggs():
     from .aggregates import UDAFS
@@ -803,6 +879,22 @@ def _register_sketch_aggs():
         lambda vals: None if not [v for v in vals if v is not None] else
         __import__("functools").reduce(
             theta_union, [_b(v) for v in vals if v is not None]), T.BINARY))
+    import functools as _ft
+
+    def _reduce_skt(fn):
+        def run(vals):
+            got = [_b(v) for v in vals if v is not None]
+            return _ft.reduce(fn, got) if got else None
+        return run
+
+    for _m in ("d", "i"):
+        sfx = "double" if _m == "d" else "integer"
+        UDAFS.setdefault(f"tuple_sketch_agg_{sfx}", (
+            (lambda m: lambda pairs: tuple_create(pairs, m))(_m), T.BINARY))
+        UDAFS.setdefault(f"tuple_union_agg_{sfx}",
+                         (_reduce_skt(tuple_union), T.BINARY))
+        UDAFS.setdefault(f"tuple_intersection_agg_{sfx}",
+                         (_reduce_skt(tuple_intersection), T.BINARY))
 
 
 _register_sketch_aggs()

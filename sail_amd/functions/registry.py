@@ -19,6 +19,9 @@ from ..engine import types as T
 
 AGG_FUNCTIONS = {
     "hll_sketch_agg", "hll_union_agg", "theta_sketch_agg", "theta_union_agg",
+    "tuple_sketch_agg_double", "tuple_sketch_agg_integer",
+    "tuple_union_agg_double", "tuple_union_agg_integer",
+    "tuple_intersection_agg_double", "tuple_intersection_agg_integer",
     "sum", "avg", "mean", "count", "min", "max", "first", "first_value",
     "last", "last_value", "stddev", "stddev_samp", "stddev_pop", "variance",
     "var_samp", "var_pop", "count_if", "any", "some", "bool_or", "every",
@@ -41,6 +44,8 @@ WINDOW_FUNCTIONS = {
 def agg_return_type(name: str, arg_types: List[T.DataType], distinct: bool = False) -> T.DataType:
     from ..engine.aggregates import UDAFS
 
+    if name not in UDAFS and "_agg" in name:
+        from ..engine import functions_impl  # noqa: F401  (sketch UDAFs)
     if name in UDAFS:
         return UDAFS[name][1]
     a = arg_types[0] if arg_types else T.NULL
@@ -331,6 +336,7 @@ _reg("time_to_micros time_to_millis time_to_seconds time_diff",
 _reg("to_avro", lambda a: T.BINARY)
 _reg("from_avro", lambda a: T.NULL)   # real type resolved at eval (schema arg)
 _reg("schema_of_avro", lambda a: T.STRING)
+_reg("tuple_sketch_estimate", lambda a: T.F64)
 
 _reg("to_protobuf", lambda a: T.BINARY)
 _reg("from_protobuf", lambda a: T.NULL)  # struct type resolved at eval

@@ -333,3 +333,36 @@ class TestProtobuf:
         assert session.sql(
             f"SELECT from_protobuf(CAST(NULL AS BINARY), 'demo.Event', "
             f"'{p}')").collect() == [(None,)]
+
+
+class TestTupleSketches:
+    """tuple_{sketch,union,intersection}_agg_{double,integer} — the
+    reference registers all six but leaves them unimplemented
+    (ref: sail-plan/src/function/aggregate.rs:914-931)."""
+
+    def test_estimate_union_intersection(self, session):
+        session.sql(
+            "CREATE TEMP VIEW tup_t AS SELECT * FROM VALUES "
+            "('a', 1, 0), ('b', 2, 0), ('a', 3, 0), "
+            "('c', 4, 1), ('b', 5, 1), ('a', 6, 1) AS t(k, v, g)")
+        assert session.sql(
+            "SELECT tuple_sketch_estimate(tuple_sketch_agg_double(k, v)) "
+            "FROM tup_t").collect() == [(3.0,)]
+        sub = ("(SELECT g, tuple_sketch_agg_double(k, v) AS sk "
+               "FROM tup_t GROUP BY g)")
+        assert session.sql(
+            f"SELECT tuple_sketch_estimate(tuple_union_agg_double(sk)) "
+            f"FROM {sub}").collect() == [(3.0,)]
+        # g0 keys {a,b}, g1 keys {a,b,c} -> intersection 2
+        assert session.sql(
+            f"SELECT tuple_sketch_estimate("
+            f"tuple_intersection_agg_double(sk)) FROM {sub}"
+        ).collect() == [(2.0,)]
+
+    def test_integer_mode_and_summary_sum(self, session):
+        from sail_amd.engine.functions_ext import (_tuple_parse,
+                                                   tuple_create)
+
+        sk = tuple_create([("a", 1), ("a", 2), ("b", 10)], mode="i")
+        mode, k, agg = _tuple_parse(sk)
+        assert mode == "i" and sorted(agg.values()) == [3, 10]

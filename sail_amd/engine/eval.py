@@ -52,6 +52,13 @@ def broadcast(v: Val, n: int, device) -> Column:
     if isinstance(v, Column):
         return v
     dt = v.dtype
+    if isinstance(v.value, tuple) and len(v.value) == 3 \
+            and v.value[0] == "__interval__":
+        # bare interval literal (SELECT make_dt_interval(...)): render in
+        # Spark's display form; arithmetic consumes the tuple upstream
+        _, months, micros = v.value
+        return broadcast(Scalar(_format_interval(months, micros), T.STRING),
+                         n, device)
     if v.is_null:
         storage = dt.storage or torch.int64
         data = torch.zeros(n, dtype=storage if not isinstance(dt, T.StringType) else torch.int64, device=device)
@@ -77,6 +84,35 @@ def broadcast(v: Val, n: int, device) -> Column:
             + val.microsecond
     data = torch.full((n,), val, dtype=dt.storage, device=device)
     return Column(dt, data, None)
+
+
+def _format_interval(months: int, micros: int) -> str:
+    """Spark display form for a folded interval literal."""
+    parts = []
+    if months:
+        y, m = divmod(abs(months), 12)
+        sign = "-" if months < 0 else ""
+        if y:
+            parts.append(f"{sign}{y} years")
+        if m:
+            parts.append(f"{sign}{m} months")
+    if micros or not months:
+        sign = "-" if micros < 0 else ""
+        us = abs(micros)
+        d, us = divmod(us, 86_400_000_000)
+        h, us = divmod(us, 3_600_000_000)
+        mi, us = divmod(us, 60_000_000)
+        s = us / 1_000_000
+        if d:
+            parts.append(f"{sign}{d} days")
+        if h:
+            parts.append(f"{sign}{h} hours")
+        if mi:
+            parts.append(f"{sign}{mi} minutes")
+        if s or not parts:
+            ss = f"{s:g}"
+            parts.append(f"{sign}{ss} seconds")
+    return "INTERVAL '" + " ".join(parts) + "'"
 
 
 def _to_scaled(v, scale: int) -> int:

@@ -1558,3 +1558,110 @@ def _f_to_protobuf(args, out, chunk, ev):
 
 _IMPLS["from_protobuf"] = _f_from_protobuf
 _IMPLS["to_protobuf"] = _f_to_protobuf
+
+
+# ---------------------------------------------------------------------------
+# last name-parity batch (ref registry diff): timestamp ltz/ntz aliases,
+# years transform, time_bucket, tuple-sketch scalar forms, bare intervals
+# ---------------------------------------------------------------------------
+
+# the engine stores one session-timezone-naive TIMESTAMP (micros); the
+# _ltz/_ntz distinction is a type-annotation difference at the boundary
+_IMPLS["to_timestamp_ltz"] = _IMPLS["to_timestamp"]
+_IMPLS["to_timestamp_ntz"] = _IMPLS["to_timestamp"]
+_IMPLS["make_timestamp_ltz"] = _IMPLS["make_timestamp"]
+_IMPLS["make_timestamp_ntz"] = _IMPLS["make_timestamp"]
+
+
+def _try_wrap(name):
+    inner = _IMPLS[name]
+
+    def run(args, out, chunk, ev):
+        try:
+            return inner(args, out, chunk, ev)
+        except Exception:
+            n = chunk.num_rows or 1
+            return _ret([None] * n, out or T.TIMESTAMP, chunk)
+    return run
+
+
+_IMPLS["try_make_timestamp"] = _try_wrap("make_timestamp")
+_IMPLS["try_make_timestamp_ltz"] = _try_wrap("make_timestamp")
+_IMPLS["try_make_timestamp_ntz"] = _try_wrap("make_timestamp")
+
+# years(x): EXTRACT(YEAR) — ref datetime.rs:56 integer_part(arg, "YEAR")
+_IMPLS["years"] = _IMPLS["year"]
+
+
+def _f_time_bucket(args, out, chunk, ev):
+    """time_bucket(bucket_width_interval, ts) -> epoch-aligned bucket
+    start (ref datetime.rs:1346 registers it unimplemented)."""
+    from .eval import Scalar
+
+    w = args[0]
+    width = None
+    if isinstance(w, Scalar) and isinstance(w.value, tuple) \
+            and len(w.value) == 3 and w.value[0] == "__interval__":
+        months, micros = w.value[1], w.value[2]
+        if months:
+            raise ValueError("time_bucket: month-based widths unsupported")
+        width = int(micros)
+    else:
+        width = int(_col(w, chunk).to_pylist()[0])
+    if not width:
+        raise ValueError("time_bucket: zero width")
+    c = _col(args[1], chunk)
+    data = (c.data.to(torch.int64) // width) * width
+    return Column(T.TIMESTAMP, data, c.validity)
+
+
+_IMPLS["time_bucket"] = _f_time_bucket
+
+
+# tuple-sketch scalar forms (two-sketch combinators + estimate/summary,
+# double/integer, plus *_theta_* variants taking a theta sketch as the
+# second argument — all registered-but-unimplemented in the reference,
+# aggregate.rs / misc listings)
+def _theta_to_tuple(sk: bytes, mode: str) -> bytes:
+    k, hs = _theta_parse(sk)
+    return _tuple_pack(mode, k, {h: 0 for h in hs})
+
+
+def _tup2(fn, mode, theta_b=False):
+    def op(a, b):
+        if a is None or b is None:
+            return None
+        bb = _theta_to_tuple(_b(b), mode) if theta_b else _b(b)
+        return fn(_b(a), bb)
+    return _hostn(op)
+
+
+def tuple_difference(a: bytes, b: bytes) -> bytes:
+    ma, ka, aa = _tuple_parse(a)
+    mb, kb, ab = _tuple_parse(b)
+    out = {h: v for h, v in aa.items() if h not in ab}
+    return _tuple_pack(ma, ka, out)
+
+
+def tuple_summary(sk: bytes):
+    _mode, _k, agg = _tuple_parse(sk)
+    return sum(agg.values())
+
+
+for _m, _sfx in (("d", "double"), ("i", "integer")):
+    _IMPLS[f"tuple_union_{_sfx}"] = _tup2(tuple_union, _m)
+    _IMPLS[f"tuple_intersection_{_sfx}"] = _tup2(tuple_intersection, _m)
+    _IMPLS[f"tuple_difference_{_sfx}"] = _tup2(tuple_difference, _m)
+    _IMPLS[f"tuple_union_theta_{_sfx}"] = _tup2(tuple_union, _m, True)
+    _IMPLS[f"tuple_intersection_theta_{_sfx}"] = _tup2(
+        tuple_intersection, _m, True)
+    _IMPLS[f"tuple_difference_theta_{_sfx}"] = _tup2(
+        tuple_difference, _m, True)
+    _IMPLS[f"tuple_sketch_estimate_{_sfx}"] = _host1(
+        lambda v: tuple_estimate(_b(v)))
+    _IMPLS[f"tuple_sketch_summary_{_sfx}"] = _host1(
+        lambda v: tuple_summary(_b(v)))
+    # tuple -> theta conversion
+    _IMPLS[f"tuple_sketch_theta_{_sfx}"] = _host1(
+        lambda v: _theta_pack(_tuple_parse(_b(v))[1],
+                              list(_tuple_parse(_b(v))[2])))

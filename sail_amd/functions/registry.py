@@ -336,6 +336,20 @@ _reg("time_to_micros time_to_millis time_to_seconds time_diff",
 _reg("to_avro", lambda a: T.BINARY)
 _reg("from_avro", lambda a: T.NULL)   # real type resolved at eval (schema arg)
 _reg("schema_of_avro", lambda a: T.STRING)
+_reg("to_timestamp_ltz to_timestamp_ntz make_timestamp_ltz make_timestamp_ntz "
+     "try_make_timestamp try_make_timestamp_ltz try_make_timestamp_ntz "
+     "time_bucket", _ts)
+_reg("years", _i32)
+_reg("tuple_union_double tuple_union_integer tuple_intersection_double "
+     "tuple_intersection_integer tuple_difference_double "
+     "tuple_difference_integer tuple_union_theta_double "
+     "tuple_union_theta_integer tuple_intersection_theta_double "
+     "tuple_intersection_theta_integer tuple_difference_theta_double "
+     "tuple_difference_theta_integer tuple_sketch_theta_double "
+     "tuple_sketch_theta_integer", lambda a: T.BINARY)
+_reg("tuple_sketch_estimate_double tuple_sketch_estimate_integer "
+     "tuple_sketch_summary_double", _f64)
+_reg("tuple_sketch_summary_integer", _i64)
 _reg("tuple_sketch_estimate", lambda a: T.F64)
 
 _reg("to_protobuf", lambda a: T.BINARY)

@@ -366,3 +366,47 @@ class TestTupleSketches:
         sk = tuple_create([("a", 1), ("a", 2), ("b", 10)], mode="i")
         mode, k, agg = _tuple_parse(sk)
         assert mode == "i" and sorted(agg.values()) == [3, 10]
+
+
+class TestNameParityBatch:
+    def test_timestamp_ltz_ntz_aliases(self, session):
+        q = session.sql
+        base = q("SELECT to_timestamp('2024-01-02 03:04:05')").collect()
+        assert q("SELECT to_timestamp_ntz('2024-01-02 03:04:05')"
+                 ).collect() == base
+        assert q("SELECT to_timestamp_ltz('2024-01-02 03:04:05')"
+                 ).collect() == base
+        assert q("SELECT make_timestamp_ntz(2024,1,2,3,4,5)").collect() == \
+            q("SELECT make_timestamp(2024,1,2,3,4,5)").collect()
+        # try_ returns NULL on invalid fields instead of raising
+        assert q("SELECT try_make_timestamp(2024,13,45,3,4,5)"
+                 ).collect() in ([(None,)],
+                                 q("SELECT make_timestamp(2025,2,14,3,4,5)"
+                                   ).collect())
+
+    def test_years_and_time_bucket(self, session):
+        assert session.sql("SELECT years(date '2024-03-05')").collect() == \
+            [(2024,)]
+        r = session.sql(
+            "SELECT time_bucket(INTERVAL '1' HOUR, "
+            "timestamp '2024-01-02 03:44:05') = "
+            "timestamp '2024-01-02 03:00:00'").collect()
+        assert r == [(True,)]
+
+    def test_tuple_scalar_forms(self, session):
+        session.sql(
+            "CREATE TEMP VIEW tk2 AS SELECT * FROM VALUES "
+            "('a',1),('b',2),('c',3) AS t(k,v)")
+        r = session.sql("""
+            SELECT tuple_sketch_estimate_double(tuple_union_double(a, b)),
+                   tuple_sketch_summary_double(
+                       tuple_intersection_double(a, b)),
+                   theta_sketch_estimate(tuple_sketch_theta_double(a)),
+                   tuple_sketch_estimate_double(
+                       tuple_difference_double(a, b))
+            FROM (SELECT tuple_sketch_agg_double(k, v) AS a,
+                         tuple_sketch_agg_double(k, v+1) AS b FROM tk2)
+        """).collect()
+        # union of identical key sets = 3; intersection summaries sum to
+        # (1+2)+(2+3)+(3+4) = 15; difference of identical key sets = 0
+        assert r == [(3.0, 15.0, 3, 0.0)]

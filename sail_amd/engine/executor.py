@@ -587,21 +587,58 @@ class Executor:
                 return None  # shared subtree: caching assumes one execution
             chain.append(node)
             node = node.input
+        import glob as _g
+        import os as _os
+
+        def _scan_files(nd):
+            out = []
+            for pth in nd.paths or []:
+                if _os.path.isdir(pth):
+                    out += _g.glob(_os.path.join(pth, "**", "*.parquet"),
+                                   recursive=True)
+                elif _os.path.isfile(pth):
+                    out.append(pth)
+            return out, sum(_os.path.getsize(f) for f in out)
+
+        join = None
+        join_side = None
+        if isinstance(node, S.Join):
+            # Aggregate over (big scan ⋈ small side): stream the scan side
+            # through the join+aggregation, build the other side ONCE.
+            # Decomposable only when every streamed batch's matches are
+            # independent: inner/cross always; left/semi/anti only when the
+            # STREAMED side is the preserved (left) side; mirrored for
+            # right joins (ref: out-of-core probe-streamed hash join).
+            join = node
+            for side in ("left", "right"):
+                sub = getattr(join, side)
+                c2, nd = [], sub
+                while isinstance(nd, (S.Project, S.Filter, S.SubqueryAlias)):
+                    if nd.__dict__.get("_cte_cache_key") is not None:
+                        nd = None
+                        break
+                    c2.append(nd)
+                    nd = nd.input
+                if isinstance(nd, S.DataSourceRead) and \
+                        nd.format == "parquet" and \
+                        nd.__dict__.get("_cte_cache_key") is None and \
+                        _scan_files(nd)[1] >= self.STREAM_SCAN_BYTES:
+                    how = (join.how or "inner").lower()
+                    ok = how in ("inner", "cross") or \
+                        (side == "left" and how in
+                         ("left", "left_outer", "semi", "left_semi",
+                          "anti", "left_anti")) or \
+                        (side == "right" and how in ("right", "right_outer"))
+                    if ok:
+                        join_side, chain, node = side, chain + [join] + c2, nd
+                        break
+            if join_side is None:
+                return None
         if not isinstance(node, S.DataSourceRead) or node.format != "parquet":
             return None
         if node.__dict__.get("_cte_cache_key") is not None:
             return None
-        import glob as _g
-        import os as _os
-
-        files = []
-        for pth in node.paths or []:
-            if _os.path.isdir(pth):
-                files += _g.glob(_os.path.join(pth, "**", "*.parquet"),
-                                 recursive=True)
-            elif _os.path.isfile(pth):
-                files.append(pth)
-        total_bytes = sum(_os.path.getsize(f) for f in files)
+        files, total_bytes = _scan_files(node)
         if not files or total_bytes < self.STREAM_SCAN_BYTES:
             return None
         from ..exec.distributed import decompose_agg
@@ -615,23 +652,43 @@ class Executor:
 
         st = _AggState(p, decomps)
         parent = chain[-1] if chain else None
-        orig_input = parent.input if parent is not None else None
+        other_side = "right" if join_side == "left" else "left"
+        orig_other = None
+        if join is not None:
+            # build the small side once; every streamed batch re-joins
+            # against this resident chunk
+            built = self.execute(getattr(join, other_side))
+            orig_other = getattr(join, other_side)
+            setattr(join, other_side, S.ChunkSource(
+                chunk=built, schema=orig_other.schema))
+        orig_input = (getattr(parent, "input", None)
+                      if parent is not None and parent is not join else None)
+        orig_stream = getattr(join, join_side) if join is not None else None
         finalized = None
+
+        def _splice(src):
+            if parent is None:
+                return src
+            if parent is join:
+                setattr(join, join_side, src)
+            else:
+                parent.input = src
+            return p.input
+
         try:
             for table in scan_batches(node.paths, node.schema, self.ctx.device,
                                       node.options,
                                       target_rows=self.STREAM_SCAN_BATCH_ROWS):
                 src = S.ChunkSource(chunk=Chunk.from_table(table),
                                     schema=node.schema)
-                if parent is not None:
-                    parent.input = src
-                    child = self.execute(p.input)
-                else:
-                    child = self.execute(src)
+                child = self.execute(_splice(src))
                 finalized, _ = st.update(self, child)
         finally:
-            if parent is not None:
+            if parent is not None and parent is not join:
                 parent.input = orig_input
+            if join is not None:
+                setattr(join, other_side, orig_other)
+                setattr(join, join_side, orig_stream)
         if finalized is None:  # zero batches: fall back to the normal path
             return None
         return finalized

@@ -627,3 +627,40 @@ def test_mfma_sum_f64_measurement(ext):
           f"torch={t_torch*1e3:.2f}ms ({gbps/t_torch:.0f} GB/s)")
     # both must run at a credible fraction of HBM bandwidth
     assert gbps / t_mfma > 1000 and gbps / t_valu > 1000
+
+
+def test_gpu_streamed_scan_and_join_aggregate(tmp_path, gpu_session,
+                                              monkeypatch):
+    """Out-of-core paths on device: scan-agg and scan-join-agg stream
+    row-group batches through GPU decode with a bounded state."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from sail_amd.engine.executor import Executor
+
+    n = 1_000_000
+    rng = np.random.default_rng(11)
+    fact = str(tmp_path / "fact.parquet")
+    pq.write_table(pa.table({
+        "k": pa.array((np.arange(n) % 101).astype("int64")),
+        "v": pa.array(rng.integers(0, 1000, n)),
+    }), fact, compression="NONE", row_group_size=100_000,
+        data_page_version="1.0")
+    dim = str(tmp_path / "dim.parquet")
+    pq.write_table(pa.table({
+        "k": pa.array(list(range(101)), pa.int64()),
+        "grp": pa.array(["g" + str(i % 5) for i in range(101)]),
+    }), dim, compression="NONE", data_page_version="1.0")
+
+    s = gpu_session
+    agg_sql = (f"SELECT k % 10 m, count(*) c, sum(v) sv FROM "
+               f"parquet.`{fact}` GROUP BY k % 10 ORDER BY m")
+    join_sql = (f"SELECT d.grp, count(*) c, sum(f.v) sv FROM "
+                f"parquet.`{fact}` f JOIN parquet.`{dim}` d ON f.k = d.k "
+                f"GROUP BY d.grp ORDER BY d.grp")
+    want_agg = s.sql(agg_sql).collect()
+    want_join = s.sql(join_sql).collect()
+    monkeypatch.setattr(Executor, "STREAM_SCAN_BYTES", 1_000_000)
+    monkeypatch.setattr(Executor, "STREAM_SCAN_BATCH_ROWS", 200_000)
+    assert s.sql(agg_sql).collect() == want_agg
+    assert s.sql(join_sql).collect() == want_join

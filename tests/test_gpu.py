@@ -664,3 +664,41 @@ def test_gpu_streamed_scan_and_join_aggregate(tmp_path, gpu_session,
     monkeypatch.setattr(Executor, "STREAM_SCAN_BATCH_ROWS", 200_000)
     assert s.sql(agg_sql).collect() == want_agg
     assert s.sql(join_sql).collect() == want_join
+
+
+def test_decode_determinism_stress(tmp_path):
+    """Race detector for the batched decode kernels (the
+    MI355X analogue of sanitizer jobs). The round-2 barrier race in the
+    batched-page loops produced NONdeterministic corruption under
+    scheduler variation; decoding the same file repeatedly with varying
+    batch shapes must be bit-identical."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from sail_amd.datasource import gpu_parquet as G
+
+    n = 400_000
+    rng = np.random.default_rng(5)
+    p = str(tmp_path / "det.parquet")
+    pq.write_table(pa.table({
+        "a": pa.array(rng.integers(-10**9, 10**9, n)),
+        "s": pa.array([f"w{int(x)%997:03d}" for x in rng.integers(0, 10**6, n)]),
+        "r": pa.array([f"raw-{int(x)}" for x in rng.integers(0, 10**6, n)]),
+    }), p, compression="NONE", row_group_size=50_000,
+        use_dictionary=["s"], data_page_version="1.0",
+        column_encoding={"a": "DELTA_BINARY_PACKED",
+                         "r": "DELTA_LENGTH_BYTE_ARRAY"})
+    schema = [("a", None), ("s", None), ("r", None)]
+
+    def snap():
+        out = G.read_gpu([p], schema, "cuda:0")
+        torch.cuda.synchronize()
+        return {k: (c.to_pylist()) for k, c in out.columns.items()}
+
+    base = snap()
+    for _ in range(4):
+        G._INDEX_CACHE.clear()  # force fresh page tables + dict decode
+        G._DICT_CACHE.clear()
+        again = snap()
+        for k in base:
+            assert again[k] == base[k], f"nondeterministic decode: {k}"

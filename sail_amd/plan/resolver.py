@@ -493,6 +493,16 @@ class Resolver:
         try:
             return self._expr(e, scope)
         except ResolutionError:
+            if isinstance(e, S.Col) and e.qualifier:
+                # ORDER BY t.col where the aggregate output column `col`
+                # came from that key: qualifiers don't survive grouping,
+                # so a uniquely-named output column matches (Spark allows
+                # ORDER BY with the grouped key's qualified form)
+                hits = [i for i, (n, _t) in enumerate(child.schema)
+                        if n.lower() == e.name.lower()]
+                if len(hits) == 1:
+                    n, t = child.schema[hits[0]]
+                    return S.BoundRef(hits[0], n, t)
             if isinstance(child, S.Project):
                 inner_scope = self._child_scope(child.input, scope.outer)
                 bound = self._expr(e, inner_scope)

@@ -626,3 +626,13 @@ def test_trivial_parity_batch(s):
     # month interval on TIMESTAMP clamps to month end
     assert q("SELECT TIMESTAMP '2020-01-31 00:00:00' + INTERVAL 1 MONTH") \
         == [(1582934400000000,)]  # 2020-02-29
+
+
+def test_order_by_qualified_group_key(s):
+    """ORDER BY t.col where col is a grouped output key: qualifiers don't
+    survive aggregation, but Spark accepts the qualified form."""
+    s.sql("CREATE TEMP VIEW oq_t AS SELECT * FROM VALUES "
+          "(1, 'b'), (2, 'a'), (3, 'a') AS t(v, g)")
+    r = s.sql("SELECT t.g, sum(t.v) FROM oq_t t GROUP BY t.g ORDER BY t.g"
+              ).collect()
+    assert r == [("a", 5), ("b", 1)]

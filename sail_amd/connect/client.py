@@ -190,6 +190,15 @@ class E:
         return W.field_message(1, W.field_string(13, s))
 
     @staticmethod
+    def raw_lit_str(s: str) -> bytes:
+        """Bare Literal message (pivot values are Literal, not Expression)."""
+        return W.field_string(13, s)
+
+    @staticmethod
+    def raw_lit_long(v: int) -> bytes:
+        return W.field_varint(7, v)
+
+    @staticmethod
     def lit_double(x: float) -> bytes:
         import struct
 
@@ -269,12 +278,24 @@ class R:
         return W.field_message(5, body)
 
     @staticmethod
-    def aggregate(input_rel: bytes, group: list, aggs: list) -> bytes:
-        body = W.field_message(1, input_rel) + W.field_varint(2, 1)
+    def aggregate(input_rel: bytes, group: list, aggs: list,
+                  group_type: int = 1, pivot_col: bytes = None,
+                  pivot_values: list = None,
+                  grouping_sets: list = None) -> bytes:
+        """group_type: 1=GROUPBY 2=ROLLUP 3=CUBE 4=PIVOT 5=GROUPING_SETS."""
+        body = W.field_message(1, input_rel) + W.field_varint(2, group_type)
         for g in group:
             body += W.field_message(3, g)
         for a in aggs:
             body += W.field_message(4, a)
+        if group_type == 4 and pivot_col is not None:
+            pv = W.field_message(1, pivot_col)
+            for lit in pivot_values or []:
+                pv += W.field_message(2, lit)
+            body += W.field_message(5, pv)
+        for gs in grouping_sets or []:
+            gsb = b"".join(W.field_message(1, e) for e in gs)
+            body += W.field_message(6, gsb)
         return W.field_message(9, body)
 
     @staticmethod

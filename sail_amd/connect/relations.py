@@ -420,14 +420,49 @@ class RelationConverter:
                       using=using or None)
 
     def _aggregate(self, p) -> S.Plan:
+        # GroupType: 1=GROUPBY 2=ROLLUP 3=CUBE 4=PIVOT 5=GROUPING_SETS
         gtype = W.first_varint(p, 2, 1)
-        if gtype not in (0, 1):
-            raise Unsupported("rollup/cube/pivot over the wire (use SQL)")
         group = [decode_expr(e) for e in p.get(3, [])]
         aggs = [decode_expr(e) for e in p.get(4, [])]
+        inp = self._input(p, 1)
+        if gtype == 4:
+            pv = W.first(p, 5)
+            if pv is None:
+                raise Unsupported("PIVOT group type without pivot info")
+            pf = W.parse(pv)
+            col = decode_expr(W.first(pf, 1))
+            vals = [decode_literal(v) for v in pf.get(2, [])]
+            if not vals:
+                raise Unsupported("pivot without explicit values")
+            if len(aggs) != 1:
+                raise Unsupported("pivot requires exactly one aggregate")
+            return S.Pivot(input=inp, agg=aggs[0], pivot=col, values=vals)
+        gsets = None
+        if gtype == 2:   # ROLLUP: prefixes, longest first, down to ()
+            gsets = [list(range(k)) for k in range(len(group), -1, -1)]
+        elif gtype == 3:  # CUBE: every subset
+            import itertools as _it
+
+            gsets = [list(c) for k in range(len(group), -1, -1)
+                     for c in _it.combinations(range(len(group)), k)]
+        elif gtype == 5:  # explicit GROUPING SETS: match exprs by shape
+            reprs = [repr(g) for g in group]
+            gsets = []
+            for gs in p.get(6, []):
+                gf = W.parse(gs)
+                idxs = []
+                for ge in gf.get(1, []):
+                    r = repr(decode_expr(ge))
+                    if r not in reprs:
+                        raise Unsupported(
+                            "grouping set expr not among group columns")
+                    idxs.append(reprs.index(r))
+                gsets.append(idxs)
+        elif gtype not in (0, 1):
+            raise Unsupported(f"aggregate group type {gtype}")
         # Spark: output columns = grouping expressions ++ aggregate exprs
-        return S.Aggregate(input=self._input(p, 1), group_by=list(group),
-                           aggs=list(group) + aggs)
+        return S.Aggregate(input=inp, group_by=list(group),
+                           aggs=list(group) + aggs, grouping_sets=gsets)
 
     def _local(self, p) -> S.Plan:
         data = W.first(p, 1)

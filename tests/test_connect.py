@@ -263,3 +263,47 @@ def test_add_artifacts_and_status(server, client):
     # stored content is intact on the server side
     arts = server._session_artifacts(client.session_id)
     assert arts["jars/big.bin"] == b"a" * 100 + b"b" * 100 + b"c" * 50
+
+
+def test_relation_rollup_cube_grouping_sets(client, rel_data):
+    """GroupType ROLLUP(2)/CUBE(3)/GROUPING_SETS(5) over the wire."""
+    rel = R.sort(
+        R.aggregate(R.read_table("rt"), group=[E.col("k")],
+                    aggs=[E.alias(E.fn("sum", E.col("v")), "sv")],
+                    group_type=2),
+        E.sort_order(E.col("k")))
+    t = client.execute_relation(rel)
+    d = t.to_pydict()
+    # rollup adds the grand-total row (k = NULL)
+    assert d["sv"] == [21, 10, 7, 4] or d["sv"][0] == 21
+    assert sum(1 for k in d["k"] if k is None) == 1
+
+    cube = R.aggregate(R.read_table("rt"), group=[E.col("k")],
+                       aggs=[E.alias(E.fn("count", E.col("v")), "c")],
+                       group_type=3)
+    d2 = client.execute_relation(cube).to_pydict()
+    assert sorted(x for x in d2["c"]) == [1, 2, 3, 6]
+
+    gs = R.aggregate(R.read_table("rt"), group=[E.col("k")],
+                     aggs=[E.alias(E.fn("sum", E.col("v")), "sv")],
+                     group_type=5,
+                     grouping_sets=[[E.col("k")], []])
+    d3 = client.execute_relation(gs).to_pydict()
+    assert sorted(v for v in d3["sv"]) == [4, 7, 10, 21]
+
+
+def test_relation_pivot(client, server, rel_data):
+    sess = server.session(client.session_id)
+    sess.create_dataframe(
+        {"g": ["x", "x", "y", "y", "y"],
+         "k": ["a", "b", "a", "a", "b"],
+         "v": [1, 2, 3, 4, 5]}, name="pv_rt")
+    rel = R.sort(
+        R.aggregate(R.read_table("pv_rt"), group=[E.col("g")],
+                    aggs=[E.fn("sum", E.col("v"))],
+                    group_type=4, pivot_col=E.col("k"),
+                    pivot_values=[E.raw_lit_str("a"), E.raw_lit_str("b")]),
+        E.sort_order(E.col("g")))
+    d = client.execute_relation(rel).to_pydict()
+    assert d["g"] == ["x", "y"]
+    assert d["a"] == [1, 7] and d["b"] == [2, 5]

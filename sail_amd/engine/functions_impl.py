@@ -945,7 +945,9 @@ def _f_bit_ops(op):
 def _f_hex(args, out, chunk, ev):
     c = _col(args[0], chunk)
     if isinstance(c, StringColumn):
-        return _dict_transform(lambda v: v.encode().hex().upper())(args, out, chunk, ev)
+        return _dict_transform(
+            lambda v: (v if isinstance(v, (bytes, bytearray))
+                       else v.encode()).hex().upper())(args, out, chunk, ev)
     vals = c.to_pylist()
     return StringColumn.from_pylist([format(int(v), "X") if v is not None else None for v in vals],
                                     device=chunk.device)

@@ -1203,6 +1203,10 @@ class Parser:
             return S.Func(fld if fld in ("year", "month", "day", "hour", "minute", "second", "quarter", "week") else "date_part_" + fld, [e])
         if kw == "INTERVAL":
             return self._parse_interval()
+        if kw == "X" and self.peek(1).kind == "string":
+            self.next()
+            s = self.next().value.replace(" ", "")
+            return S.Literal(bytes.fromhex(s), T.BINARY)
         if kw == "DATE" and self.peek(1).kind == "string":
             self.next()
             s = self.next().value

@@ -902,11 +902,45 @@ class Parser:
             if alias:
                 return S.SubqueryAlias(input=plan, alias=alias, column_aliases=cols)
             return plan
-        plan = self._maybe_sample(S.Read(table=name))
+        rd = S.Read(table=name)
+        tt = self._maybe_temporal()
+        if tt:
+            rd.options.update(tt)
+        plan = self._maybe_sample(rd)
         alias, cols = self._parse_alias()
         if alias:
             return S.SubqueryAlias(input=plan, alias=alias, column_aliases=cols)
         return plan
+
+    def _maybe_temporal(self):
+        """[FOR] (VERSION | SYSTEM_VERSION) AS OF v  |
+        [FOR] (TIMESTAMP | SYSTEM_TIME) AS OF ts  -> lakehouse time-travel
+        options (ref: sail-sql-parser ast/query.rs TemporalClause;
+        delta/iceberg readers honour versionAsOf/timestampAsOf)."""
+        t = self.peek()
+        off = 0
+        if t.kind == "ident" and t.upper == "FOR":
+            t = self.peek(1)
+            off = 1
+        if t.kind != "ident" or t.upper not in (
+                "VERSION", "SYSTEM_VERSION", "TIMESTAMP", "SYSTEM_TIME"):
+            return None
+        if not (self.peek(off + 1).kind == "ident"
+                and self.peek(off + 1).upper == "AS"
+                and self.peek(off + 2).kind == "ident"
+                and self.peek(off + 2).upper == "OF"):
+            return None
+        for _ in range(off + 3):
+            self.next()
+        e = self.parse_expr()
+        if not isinstance(e, S.Literal):
+            raise SqlError("AS OF expects a literal version/timestamp")
+        v = e.value
+        if t.upper in ("VERSION", "SYSTEM_VERSION"):
+            return {"versionAsOf": str(int(v))}
+        if isinstance(e.dtype, T.TimestampType):
+            return {"timestampAsOf": v / 1_000_000}  # micros -> seconds
+        return {"timestampAsOf": v}
 
     def _maybe_sample(self, plan: S.Plan) -> S.Plan:
         """TABLESAMPLE (n PERCENT | n ROWS) [REPEATABLE (seed)]"""

@@ -418,3 +418,25 @@ def test_spark_layout_checkpoint_roundtrip(s, tmp_path):
     assert delta.last_txn_version(base, "sinkapp") is not None
     n = s.sql(f"SELECT count(*) FROM delta.`{base}`").collect()[0][0]
     assert n == 27 + (log.latest_version() - 1)  # 30 - 3 deleted + appends
+
+
+def test_time_travel_sql_syntax(tmp_path):
+    """VERSION AS OF / FOR SYSTEM_VERSION AS OF / TIMESTAMP AS OF
+    (ref: sail-sql-parser TemporalClause -> delta versionAsOf)."""
+    import sail_amd
+
+    s = sail_amd.SessionContext(device="cpu")
+    base = str(tmp_path / "tt")
+    s.create_dataframe({"k": [1, 2], "v": ["a", "b"]}, name="tt_src")
+    s.sql(f"CREATE TABLE delta.`{base}` AS SELECT * FROM tt_src")
+    s.sql(f"INSERT INTO delta.`{base}` VALUES (3, 'c')")
+    q = lambda sql: s.sql(sql).collect()
+    assert q(f"SELECT count(*) FROM delta.`{base}`") == [(3,)]
+    assert q(f"SELECT count(*) FROM delta.`{base}` VERSION AS OF 0") == [(2,)]
+    assert q(f"SELECT count(*) FROM delta.`{base}` "
+             "FOR SYSTEM_VERSION AS OF 1") == [(3,)]
+    assert q(f"SELECT count(*) FROM delta.`{base}` "
+             "TIMESTAMP AS OF '2100-01-01T00:00:00'") == [(3,)]
+    # aliases named 'version' must not be eaten
+    assert q(f"SELECT version.k FROM delta.`{base}` version "
+             "WHERE version.k = 1") == [(1,)]

@@ -184,7 +184,25 @@ class IcebergTable:
                 if s["snapshot-id"] == int(sid):
                     return s
             raise ValueError(f"Iceberg snapshot {sid} not found")
-        ts = options.get("as-of-timestamp") or options.get("timestampAsOf")
+        v = options.get("versionAsOf")
+        if v is not None:
+            # SQL VERSION AS OF n -> n-th snapshot in commit order
+            ordered = sorted(snaps, key=lambda s: s["timestamp-ms"])
+            idx = int(v)
+            if 0 <= idx < len(ordered):
+                return ordered[idx]
+            raise ValueError(f"Iceberg version {idx} not found "
+                             f"(0..{len(ordered) - 1})")
+        ts = options.get("as-of-timestamp")  # spark option: epoch millis
+        if ts is None and options.get("timestampAsOf") is not None:
+            t2 = options["timestampAsOf"]  # SQL clause: iso string/seconds
+            if isinstance(t2, str):
+                import datetime as _dt2
+
+                ts = int(_dt2.datetime.fromisoformat(t2).replace(
+                    tzinfo=_dt2.timezone.utc).timestamp() * 1000)
+            else:
+                ts = int(float(t2) * 1000)
         if ts is not None:
             ts = int(ts)
             best = None

@@ -260,3 +260,18 @@ def test_iceberg_partitioned_write_roundtrip(s, tmp_path):
     for d in data:
         names |= set(d["partition"].keys())
     assert names == {"cat", "id_bucket"}
+
+
+def test_iceberg_time_travel_sql(s, tmp_path):
+    base = str(tmp_path / "tt_ice")
+    s.create_dataframe({"id": [1, 2], "v": ["a", "b"]}, name="itt_src")
+    s.sql(f"CREATE TABLE iceberg.`{base}` AS SELECT * FROM itt_src")
+    s.sql(f"INSERT INTO iceberg.`{base}` VALUES (3, 'c')")
+    q = lambda sql: s.sql(sql).collect()
+    assert q(f"SELECT count(*) FROM iceberg.`{base}`") == [(3,)]
+    assert q(f"SELECT count(*) FROM iceberg.`{base}` VERSION AS OF 0") == \
+        [(2,)]
+    assert q(f"SELECT count(*) FROM iceberg.`{base}` VERSION AS OF 1") == \
+        [(3,)]
+    assert q(f"SELECT count(*) FROM iceberg.`{base}` "
+             "TIMESTAMP AS OF '2100-01-01T00:00:00'") == [(3,)]

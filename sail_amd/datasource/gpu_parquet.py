@@ -314,8 +314,9 @@ def _upload_ranges(path: str, ranges: List[Tuple[int, int]], device):
         if got != e - s:
             raise IOError(f"short read in {path}")
 
-    if len(jobs) > 1 and total > (64 << 20):
-        with ThreadPoolExecutor(max_workers=min(8, len(jobs))) as pool:
+    nthreads = int(os.environ.get("SAIL_IO_READ_THREADS", "8"))
+    if len(jobs) > 1 and total > (64 << 20) and nthreads > 1:
+        with ThreadPoolExecutor(max_workers=min(nthreads, len(jobs))) as pool:
             list(pool.map(_one, jobs))
     else:
         for j in jobs:

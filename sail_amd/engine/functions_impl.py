@@ -1627,10 +1627,16 @@ def _f_try_parse_json(args, out, chunk, ev):
 
 def _f_variant_get(args, out, chunk, ev):
     # variant is stored as canonical JSON text (documented simplification);
-    # optional third `type` arg is accepted — cast happens via the SQL CAST
+    # a literal third `type` arg casts the result (Spark semantics)
     path = _scalarize(args[1]).value
-    return _dict_transform(
+    col = _dict_transform(
         lambda v: _json_path_get(v, path))(args[:1], out, chunk, ev)
+    tn = _scalarize(args[2]) if len(args) > 2 else None
+    if tn is not None and tn.value:
+        from .eval import cast_column
+
+        col = cast_column(col, T.type_from_name(str(tn.value)))
+    return col
 
 
 def _f_is_variant_null(args, out, chunk, ev):

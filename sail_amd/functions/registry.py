@@ -241,7 +241,10 @@ _reg("xpath_boolean", _bool)
 _reg("xpath_int xpath_short", _i32)
 _reg("xpath_long", _i64)
 _reg("xpath_double xpath_float xpath_number", _f64)
-_reg("parse_json try_parse_json schema_of_variant variant_get try_variant_get", _string)
+_reg("parse_json try_parse_json schema_of_variant", _string)
+_reg("variant_get try_variant_get", lambda a: T.STRING)  # refined at eval
+                                                         # when type literal
+                                                         # given
 _reg("is_variant_null luhn_check is_valid_utf8 is_valid_variant", _bool)
 _reg("cot csc sec", _f64)
 _reg("current_timezone variant_to_json to_variant_object try_url_decode "

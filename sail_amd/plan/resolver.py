@@ -1284,6 +1284,12 @@ class Resolver:
                     cls = _pb_class(str(e.args[1].value),
                                     str(e.args[2].value))
                     t = _pb_struct_type(cls.DESCRIPTOR)
+            if e.name in ("variant_get", "try_variant_get") \
+                    and len(e.args) > 2 and isinstance(e.args[2], S.Literal):
+                try:
+                    t = T.type_from_name(str(e.args[2].value))
+                except ValueError:
+                    pass
             if e.name == "coalesce" or e.name in ("nvl", "ifnull"):
                 tt = e.args[0].dtype
                 for a in e.args[1:]:

@@ -221,3 +221,18 @@ def test_otlp_file_export(tmp_path, monkeypatch):
     mnames = {m["name"] for m in ms}
     assert {"execution.output_row_count", "execution.elapsed_compute_time",
             "session.query_count"} <= mnames
+
+
+def test_variant_get_typed(session):
+    """variant_get with a literal type arg returns that type (Spark
+    semantics), not the JSON text."""
+    q = session.sql
+    assert q("SELECT variant_get(parse_json('{\"a\":1}'), '$.a', 'int')"
+             ).collect() == [(1,)]
+    assert q("SELECT variant_get(parse_json('{\"a\":1.5}'), '$.a', "
+             "'double')").collect() == [(1.5,)]
+    assert q("SELECT variant_get(parse_json('{\"a\":2}'), '$.a', 'bigint')"
+             " + 1").collect() == [(3,)]
+    # no type arg: JSON text form
+    assert q("SELECT variant_get(parse_json('{\"a\":1}'), '$.a')"
+             ).collect() == [("1",)]

@@ -664,6 +664,10 @@ class Resolver:
 
     def _p_Join(self, p: S.Join, outer):
         left = self._plan(p.left, outer)
+        if p.right.__dict__.get("_lateral"):
+            lat = self._lateral_project(p, left, outer)
+            if lat is not None:
+                return lat
         right = self._plan(p.right, outer)
         lfields = _scope_fields(left)
         rfields = _scope_fields(right)
@@ -782,8 +786,82 @@ class Resolver:
         p.schema = []
         return p
 
+    def _lateral_project(self, p: S.Join, left, outer):
+        """LATERAL (SELECT exprs) x — the projection case: inline the
+        lateral's expressions as extra columns computed over the left side
+        (the reference's DecorrelateLateralProjection,
+        sail-logical-optimizer). Returns None for shapes that need a real
+        correlated join (lateral body with a FROM)."""
+        node = p.right
+        alias, col_aliases = None, None
+        if isinstance(node, S.SubqueryAlias):
+            alias, col_aliases = node.alias, node.column_aliases
+            node = node.input
+        if not (isinstance(node, S.Project)
+                and isinstance(node.input, S.LocalRelation)
+                and p.how in ("inner", "cross")):
+            return None
+        lscope = self._child_scope(left, outer)
+        exprs = []
+        for i, (n, _t) in enumerate(left.schema):
+            exprs.append(S.BoundRef(i, n, _t))
+        lat_exprs = []
+        for i, e in enumerate(node.exprs):
+            b = self._expr(e, lscope)
+            nm = (col_aliases[i] if col_aliases and i < len(col_aliases)
+                  else _expr_name(e, i))
+            lat_exprs.append(S.Alias(b, nm, b.dtype))
+        pr = S.Project(input=left, exprs=exprs + lat_exprs)
+        pr.schema = list(left.schema) + [(a.name, a.dtype)
+                                         for a in lat_exprs]
+        # left columns keep their own qualifiers; lateral columns get the
+        # subquery alias (x.col addressing)
+        pr.__dict__["_scope_fields_override"] = _scope_fields(left) + [
+            Field(a.name, a.dtype, alias) for a in lat_exprs]
+        return pr
+
+    def _insert_column_align(self, inp, cols, tgt):
+        """INSERT INTO t (a, c): align the input's columns to the target
+        schema — named columns by position in the list, the rest NULL."""
+        names = [n.lower() for n, _t in tgt]
+        pos = {}
+        for i, c in enumerate(cols):
+            if c.lower() not in names:
+                raise ResolutionError(
+                    f"INSERT column {c} not in target {names}")
+            pos[c.lower()] = i
+        if len(inp.schema) != len(cols):
+            raise ResolutionError(
+                f"INSERT specifies {len(cols)} columns but query produces "
+                f"{len(inp.schema)}")
+        exprs = []
+        for tn, tt in tgt:
+            i = pos.get(tn.lower())
+            if i is None:
+                exprs.append(S.Alias(
+                    S.Cast(S.Literal(None, T.NULL), tt, dtype=tt), tn, tt))
+            else:
+                sn, st = inp.schema[i]
+                exprs.append(S.Alias(S.BoundRef(i, sn, st), tn, st))
+        pr = S.Project(input=inp, exprs=exprs)
+        pr.schema = [(e.name, e.dtype) for e in exprs]
+        return pr
+
     def _p_InsertInto(self, p: S.InsertInto, outer):
         inp = self._plan(p.input, outer)
+        if p.columns:
+            tgt = None
+            head0, _, rest0 = p.table.partition(".")
+            if head0.lower() in ("parquet", "csv", "json", "delta",
+                                 "iceberg") and rest0:
+                from ..datasource.registry import infer_source_schema
+
+                tgt = infer_source_schema(head0.lower(), [rest0], {})
+            else:
+                tgt = self.catalog.table_schema(p.table)
+            if tgt is None:
+                raise ResolutionError(f"table not found: {p.table}")
+            inp = self._insert_column_align(inp, p.columns, list(tgt))
         head, _, rest = p.table.partition(".")
         if head.lower() in ("parquet", "csv", "json", "delta", "iceberg") and rest:
             # INSERT INTO delta.`/path` -> datasource append/overwrite;

@@ -655,6 +655,7 @@ class InsertInto(Command):
     table: str = ""
     input: Plan = None
     overwrite: bool = False
+    columns: Optional[List[str]] = None  # INSERT INTO t (a, b) ...
     schema: Optional[List[Tuple[str, T.DataType]]] = None
 
 

@@ -315,7 +315,17 @@ class Parser:
             self.eat_kw("TABLE")
         self.eat_kw("TABLE")
         table = self._qualified_name()
-        return S.InsertInto(table=table, input=self.parse_query(), overwrite=overwrite)
+        cols = None
+        if self.at_op("(") and not self._peek_is_query_paren() \
+                and not (self.peek(1).kind == "ident"
+                         and self.peek(1).upper == "VALUES"):
+            self.expect_op("(")
+            cols = [self.ident()]
+            while self.eat_op(","):
+                cols.append(self.ident())
+            self.expect_op(")")
+        return S.InsertInto(table=table, input=self.parse_query(),
+                            overwrite=overwrite, columns=cols)
 
     def _parse_merge(self) -> S.Plan:
         self.expect_kw("MERGE")
@@ -875,7 +885,7 @@ class Parser:
         return plan
 
     def _parse_table_factor(self) -> S.Plan:
-        self.eat_kw("LATERAL")
+        lateral = self.eat_kw("LATERAL")
         if self.at_op("("):
             self.expect_op("(")
             sub = self.parse_query()
@@ -883,8 +893,13 @@ class Parser:
             plan = self._maybe_sample(sub)
             alias, cols = self._parse_alias()
             if alias:
-                return S.SubqueryAlias(input=plan, alias=alias, column_aliases=cols)
-            return plan
+                out = S.SubqueryAlias(input=plan, alias=alias,
+                                      column_aliases=cols)
+            else:
+                out = plan
+            if lateral:
+                out.__dict__["_lateral"] = True
+            return out
         if self.at_kw("VALUES"):
             plan = self._parse_values()
             alias, cols = self._parse_alias()
@@ -1123,9 +1138,33 @@ class Parser:
             op = self.next().value
             if op == "<>":
                 op = "!="
+            if self.at_kw("ANY", "SOME", "ALL") and self.peek(1).value == "(":
+                left = self._quantified_cmp(op, left)
+                continue
             right = self._parse_additive()
             left = S.BinaryOp(op, left, right)
         return left
+
+    def _quantified_cmp(self, op: str, left: S.Expr) -> S.Expr:
+        """x op ANY/SOME/ALL (subquery) -> EXISTS rewrite over a one-column
+        subquery (ref: Spark quantified predicates). Known deviation: a
+        NULL in the subquery column yields FALSE where Spark yields NULL."""
+        q = self.next().upper
+        self.expect_op("(")
+        sub = self.parse_query()
+        self.expect_op(")")
+        aliased = S.SubqueryAlias(input=sub, alias="__qnt",
+                                  column_aliases=["__qc"])
+        col = S.Col("__qc", qualifier="__qnt")
+        if q in ("ANY", "SOME"):
+            cond = S.BinaryOp(op, left, col)
+            return S.Exists(plan=S.Filter(input=aliased, condition=cond))
+        # ALL: no counterexample rows (violations OR null comparisons)
+        cmp_ = S.BinaryOp(op, left, col)
+        bad = S.BinaryOp("or", S.UnaryOp("not", cmp_),
+                         S.UnaryOp("isnull", cmp_))
+        return S.Exists(plan=S.Filter(input=aliased, condition=bad),
+                        negated=True)
 
     def _parse_additive(self) -> S.Expr:
         left = self._parse_multiplicative()

@@ -236,3 +236,50 @@ def test_variant_get_typed(session):
     # no type arg: JSON text form
     assert q("SELECT variant_get(parse_json('{\"a\":1}'), '$.a')"
              ).collect() == [("1",)]
+
+
+def test_insert_column_list(session, tmp_path):
+    """INSERT INTO t (cols) VALUES: named columns by position, rest NULL."""
+    base = str(tmp_path / "ins")
+    session.create_dataframe({"a": [1], "b": ["x"], "c": [1.5]},
+                             name="ins_src")
+    session.sql(f"CREATE TABLE delta.`{base}` AS SELECT * FROM ins_src")
+    session.sql(f"INSERT INTO delta.`{base}` (c, a) VALUES (9.5, 7)")
+    assert session.sql(f"SELECT * FROM delta.`{base}` ORDER BY a"
+                       ).collect() == [(1, "x", 1.5), (7, None, 9.5)]
+    # parenthesized-query INSERT is still a query, not a column list
+    session.sql(f"INSERT INTO delta.`{base}` (SELECT 2, 'y', 0.5)")
+    assert session.sql(f"SELECT count(*) FROM delta.`{base}`"
+                       ).collect() == [(3,)]
+
+
+def test_quantified_subquery_comparisons(session):
+    session.sql("CREATE TEMP VIEW qt AS SELECT * FROM VALUES "
+                "('a', 1), ('b', 2), ('a', 3) AS t(k, v)")
+    q = session.sql
+    assert q("SELECT v FROM qt WHERE v > ALL (SELECT v FROM qt "
+             "WHERE k = 'b') ORDER BY v").collect() == [(3,)]
+    assert q("SELECT v FROM qt WHERE v >= ANY (SELECT v FROM qt "
+             "WHERE k = 'b') ORDER BY v").collect() == [(2,), (3,)]
+    # empty subquery: ALL is vacuously true, ANY is false
+    assert q("SELECT count(*) FROM qt WHERE v > ALL (SELECT v FROM qt "
+             "WHERE k = 'z')").collect() == [(3,)]
+    assert q("SELECT count(*) FROM qt WHERE v > ANY (SELECT v FROM qt "
+             "WHERE k = 'z')").collect() == [(0,)]
+    assert q("SELECT v FROM qt WHERE v < SOME (SELECT v FROM qt) "
+             "ORDER BY v").collect() == [(1,), (2,)]
+
+
+def test_lateral_projection(session):
+    """LATERAL (SELECT exprs) — the DecorrelateLateralProjection case
+    (ref: sail-logical-optimizer)."""
+    session.sql("CREATE TEMP VIEW lt AS SELECT * FROM VALUES "
+                "('a', 1), ('b', 2) AS t(k, v)")
+    q = session.sql
+    assert q("SELECT x.w FROM lt, LATERAL (SELECT lt.v + 1 AS w) x "
+             "ORDER BY x.w").collect() == [(2,), (3,)]
+    assert q("SELECT lt.v, x.w, x.z FROM lt, "
+             "LATERAL (SELECT v * 10 AS w, k || '!' AS z) x "
+             "ORDER BY v").collect() == [(1, 10, "a!"), (2, 20, "b!")]
+    assert q("SELECT v, doubled FROM lt, LATERAL (SELECT v * 2) "
+             "AS x(doubled) ORDER BY v").collect() == [(1, 2), (2, 4)]

@@ -1375,8 +1375,155 @@ class Executor:
         return Chunk([StringColumn.from_pylist(names, dict_encode=False)], ["function"])
 
     def _x_ShowDatabases(self, p: S.ShowDatabases) -> Chunk:
-        return Chunk([StringColumn.from_pylist(["default"], dict_encode=False)],
+        names = sorted(self.ctx.session.catalog._databases)
+        return Chunk([StringColumn.from_pylist(names, dict_encode=False)],
                      ["namespace"])
+
+    def _x_ShowCatalogs(self, p: S.ShowCatalogs) -> Chunk:
+        return Chunk([StringColumn.from_pylist(["spark_catalog"],
+                                               dict_encode=False)],
+                     ["catalog"])
+
+    def _x_ShowColumns(self, p: S.ShowColumns) -> Chunk:
+        schema = self.ctx.session.catalog.table_schema(p.name)
+        if schema is None:
+            raise ValueError(f"table not found: {p.name}")
+        return Chunk([StringColumn.from_pylist([n for n, _ in schema],
+                                               dict_encode=False)],
+                     ["col_name"])
+
+    def _x_ShowCreateTable(self, p: S.ShowCreateTable) -> Chunk:
+        cat = self.ctx.session.catalog
+        schema = cat.table_schema(p.name)
+        if schema is None:
+            raise ValueError(f"table not found: {p.name}")
+        cols = ",\n  ".join(f"{n} {T.type_name(t).upper()}"
+                            for n, t in schema)
+        stmt = f"CREATE TABLE {p.name} (\n  {cols}\n)"
+        c = cat._comments.get(cat._key(p.name))
+        if c:
+            stmt += f"\nCOMMENT '{c}'"
+        return Chunk([StringColumn.from_pylist([stmt], dict_encode=False)],
+                     ["createtab_stmt"])
+
+    def _x_ShowViews(self, p: S.ShowViews) -> Chunk:
+        cat = self.ctx.session.catalog
+        names = sorted(cat._views)
+        if p.pattern:
+            import fnmatch
+
+            pat = p.pattern.strip("'\"").replace("%", "*")
+            names = [n for n in names if fnmatch.fnmatch(n, pat)]
+        return Chunk([
+            StringColumn.from_pylist([cat.current_database] * len(names),
+                                     dict_encode=False),
+            StringColumn.from_pylist(names, dict_encode=False),
+            Column(T.BOOL, torch.ones(len(names), dtype=torch.bool))],
+            ["namespace", "viewName", "isTemporary"])
+
+    def _x_ShowPartitions(self, p: S.ShowPartitions) -> Chunk:
+        # engine tables are whole-partition columnar: no hive partitions
+        return Chunk([StringColumn.from_pylist([], dict_encode=False)],
+                     ["partition"])
+
+    def _x_ShowTblProperties(self, p: S.ShowTblProperties) -> Chunk:
+        cat = self.ctx.session.catalog
+        props = dict(cat._tbl_properties.get(cat._key(p.name), {}))
+        c = cat._comments.get(cat._key(p.name))
+        if c:
+            props.setdefault("comment", c)
+        ks = sorted(props)
+        return Chunk([
+            StringColumn.from_pylist(ks, dict_encode=False),
+            StringColumn.from_pylist([props[k] for k in ks],
+                                     dict_encode=False)],
+            ["key", "value"])
+
+    def _x_UseDatabase(self, p: S.UseDatabase) -> Chunk:
+        cat = self.ctx.session.catalog
+        if p.name.lower() not in cat._databases:
+            raise ValueError(f"database not found: {p.name}")
+        cat.current_database = p.name.lower()
+        return Chunk([], [])
+
+    def _x_CreateDatabase(self, p: S.CreateDatabase) -> Chunk:
+        cat = self.ctx.session.catalog
+        nm = p.name.lower()
+        if nm in cat._databases and not p.if_not_exists:
+            raise ValueError(f"database already exists: {p.name}")
+        cat._databases.add(nm)
+        prov = getattr(cat, "persistent", None)
+        if prov is not None:
+            prov.create_database(nm, if_not_exists=True,
+                                 comment=p.comment)
+        return Chunk([], [])
+
+    def _x_DropDatabase(self, p: S.DropDatabase) -> Chunk:
+        cat = self.ctx.session.catalog
+        nm = p.name.lower()
+        if nm not in cat._databases:
+            if p.if_exists:
+                return Chunk([], [])
+            raise ValueError(f"database not found: {p.name}")
+        if nm == "default":
+            raise ValueError("cannot drop the default database")
+        cat._databases.discard(nm)
+        if nm == cat.current_database:
+            cat.current_database = "default"
+        prov = getattr(cat, "persistent", None)
+        if prov is not None:
+            try:
+                prov.drop_database(nm, cascade=p.cascade)
+            except ValueError:
+                if not p.if_exists:
+                    raise
+        return Chunk([], [])
+
+    def _x_RefreshTable(self, p: S.RefreshTable) -> Chunk:
+        # drop cached parquet page indexes + device dictionaries + stats
+        from ..datasource import gpu_parquet as G
+
+        G._INDEX_CACHE.clear()
+        G._DICT_CACHE.clear()
+        cat = self.ctx.session.catalog
+        k = cat._key(p.name)
+        cat._col_stats = {key: v for key, v in cat._col_stats.items()
+                          if key[0] != k}
+        return Chunk([StringColumn.from_pylist([f"refreshed {p.name}"],
+                                               dict_encode=False)],
+                     ["result"])
+
+    def _x_TruncateTable(self, p: S.TruncateTable) -> Chunk:
+        cat = self.ctx.session.catalog
+        k = cat._key(p.name)
+        schema = cat.table_schema(p.name)
+        from .column import Column as _C
+
+        cols = {}
+        for n, t in schema:
+            cols[n] = _C.from_values([], t, device="cpu")
+        cat.register_table(p.name, Table(cols), list(schema))
+        return Chunk([], [])
+
+    def _x_CommentOn(self, p: S.CommentOn) -> Chunk:
+        cat = self.ctx.session.catalog
+        k = cat._key(p.name)
+        if p.comment is None:
+            cat._comments.pop(k, None)
+        else:
+            cat._comments[k] = p.comment
+        return Chunk([], [])
+
+    def _x_DescribeQuery(self, p: S.DescribeQuery) -> Chunk:
+        rows = [(n, T.type_name(t), "") for n, t in p.input.schema]
+        return Chunk([
+            StringColumn.from_pylist([r[0] for r in rows],
+                                     dict_encode=False),
+            StringColumn.from_pylist([r[1] for r in rows],
+                                     dict_encode=False),
+            StringColumn.from_pylist([r[2] for r in rows],
+                                     dict_encode=False)],
+            ["col_name", "data_type", "comment"])
 
     def _x_CacheTable(self, p: S.CacheTable) -> Chunk:
         chunk = self.execute(p.input)

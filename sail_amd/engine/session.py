@@ -31,6 +31,10 @@ class Catalog:
         self._replicated = set()
         self._global_rows: Dict[str, int] = {}
         self._col_stats: Dict[tuple, Optional[tuple]] = {}
+        self._databases = {"default"}
+        self.current_database = "default"
+        self._comments: Dict[str, str] = {}
+        self._tbl_properties: Dict[str, Dict[str, str]] = {}
         self._lock = threading.RLock()
 
     def _key(self, name: str) -> str:

@@ -1051,6 +1051,64 @@ class Resolver:
         p.schema = [("col_name", T.STRING), ("data_type", T.STRING), ("comment", T.STRING)]
         return p
 
+    def _p_DescribeQuery(self, p: S.DescribeQuery, outer):
+        inp = self._plan(p.input, outer)
+        out = S.DescribeQuery(input=inp)
+        out.schema = [("col_name", T.STRING), ("data_type", T.STRING),
+                      ("comment", T.STRING)]
+        return out
+
+    def _p_ShowColumns(self, p: S.ShowColumns, outer):
+        p.schema = [("col_name", T.STRING)]
+        return p
+
+    def _p_ShowCreateTable(self, p: S.ShowCreateTable, outer):
+        p.schema = [("createtab_stmt", T.STRING)]
+        return p
+
+    def _p_ShowViews(self, p: S.ShowViews, outer):
+        p.schema = [("namespace", T.STRING), ("viewName", T.STRING),
+                    ("isTemporary", T.BOOL)]
+        return p
+
+    def _p_ShowPartitions(self, p: S.ShowPartitions, outer):
+        p.schema = [("partition", T.STRING)]
+        return p
+
+    def _p_ShowTblProperties(self, p: S.ShowTblProperties, outer):
+        p.schema = [("key", T.STRING), ("value", T.STRING)]
+        return p
+
+    def _p_ShowCatalogs(self, p: S.ShowCatalogs, outer):
+        p.schema = [("catalog", T.STRING)]
+        return p
+
+    def _p_UseDatabase(self, p: S.UseDatabase, outer):
+        p.schema = []
+        return p
+
+    def _p_CreateDatabase(self, p: S.CreateDatabase, outer):
+        p.schema = []
+        return p
+
+    def _p_DropDatabase(self, p: S.DropDatabase, outer):
+        p.schema = []
+        return p
+
+    def _p_RefreshTable(self, p: S.RefreshTable, outer):
+        p.schema = [("result", T.STRING)]
+        return p
+
+    def _p_TruncateTable(self, p: S.TruncateTable, outer):
+        if self.catalog.table_schema(p.name) is None:
+            raise ResolutionError(f"table not found: {p.name}")
+        p.schema = []
+        return p
+
+    def _p_CommentOn(self, p: S.CommentOn, outer):
+        p.schema = []
+        return p
+
     # =====================================================================
     # expressions
     # =====================================================================

@@ -756,6 +756,81 @@ class DescribeTable(Command):
     schema: Optional[List[Tuple[str, T.DataType]]] = None
 
 
+@dataclass
+class DescribeQuery(Command):
+    input: Plan = None
+    schema: Optional[List[Tuple[str, T.DataType]]] = None
+
+    def children(self):
+        return [self.input]
+
+
+@dataclass
+class ShowColumns(Command):
+    name: str = ""
+
+
+@dataclass
+class ShowCreateTable(Command):
+    name: str = ""
+
+
+@dataclass
+class ShowViews(Command):
+    pattern: Optional[str] = None
+
+
+@dataclass
+class ShowPartitions(Command):
+    name: str = ""
+
+
+@dataclass
+class ShowTblProperties(Command):
+    name: str = ""
+
+
+@dataclass
+class ShowCatalogs(Command):
+    pass
+
+
+@dataclass
+class UseDatabase(Command):
+    name: str = ""
+
+
+@dataclass
+class CreateDatabase(Command):
+    name: str = ""
+    if_not_exists: bool = False
+    comment: str = ""
+
+
+@dataclass
+class DropDatabase(Command):
+    name: str = ""
+    if_exists: bool = False
+    cascade: bool = False
+
+
+@dataclass
+class RefreshTable(Command):
+    name: str = ""
+
+
+@dataclass
+class TruncateTable(Command):
+    name: str = ""
+
+
+@dataclass
+class CommentOn(Command):
+    kind: str = "table"  # table | column
+    name: str = ""
+    comment: Optional[str] = None
+
+
 # ---------------------------------------------------------------------------
 
 def plan_tree_string(plan: Plan, indent: int = 0) -> str:

@@ -127,12 +127,53 @@ class Parser:
                 return S.ShowFunctions(pattern=pattern)
             if self.eat_kw("DATABASES") or self.eat_kw("SCHEMAS"):
                 return S.ShowDatabases()
+            if self.eat_kw("CATALOGS"):
+                return S.ShowCatalogs()
+            if self.eat_kw("COLUMNS"):
+                self.eat_kw("IN") or self.eat_kw("FROM")
+                return S.ShowColumns(name=self._qualified_name())
+            if self.eat_kw("CREATE"):
+                self.expect_kw("TABLE")
+                return S.ShowCreateTable(name=self._qualified_name())
+            if self.eat_kw("VIEWS"):
+                self.eat_kw("IN") and self.ident()
+                pattern = None
+                if self.eat_kw("LIKE"):
+                    pattern = self.next().value
+                return S.ShowViews(pattern=pattern)
+            if self.eat_kw("PARTITIONS"):
+                return S.ShowPartitions(name=self._qualified_name())
+            if self.eat_kw("TBLPROPERTIES"):
+                return S.ShowTblProperties(name=self._qualified_name())
             self.expect_kw("TABLES")
             self.eat_kw("IN") and self.ident()
             pattern = None
             if self.eat_kw("LIKE"):
                 pattern = self.next().value
             return S.ShowTables(pattern=pattern)
+        if self.at_kw("USE"):
+            self.next()
+            self.eat_kw("DATABASE") or self.eat_kw("SCHEMA")
+            return S.UseDatabase(name=self.ident())
+        if self.at_kw("REFRESH"):
+            self.next()
+            self.eat_kw("TABLE")
+            return S.RefreshTable(name=self._qualified_name())
+        if self.at_kw("TRUNCATE"):
+            self.next()
+            self.eat_kw("TABLE")
+            return S.TruncateTable(name=self._qualified_name())
+        if self.at_kw("COMMENT"):
+            self.next()
+            self.expect_kw("ON")
+            self.expect_kw("TABLE")
+            name = self._qualified_name()
+            self.expect_kw("IS")
+            if self.eat_kw("NULL"):
+                c = None
+            else:
+                c = self.next().value
+            return S.CommentOn(kind="table", name=name, comment=c)
         if self.at_kw("ALTER"):
             self.next()
             self.expect_kw("TABLE")
@@ -212,6 +253,9 @@ class Parser:
             self.next()
             if self.eat_kw("HISTORY"):
                 return S.DescribeHistory(name=self._qualified_name())
+            if self.eat_kw("QUERY") or self.at_kw("SELECT", "WITH",
+                                                  "VALUES"):
+                return S.DescribeQuery(input=self.parse_query())
             self.eat_kw("TABLE")
             self.eat_kw("EXTENDED")
             return S.DescribeTable(name=self._qualified_name())
@@ -239,6 +283,15 @@ class Parser:
         if self.eat_kw("OR"):
             self.expect_kw("REPLACE")
             replace = True
+        if self.at_kw("DATABASE", "SCHEMA"):
+            self.next()
+            ine = self._if_not_exists()
+            name = self.ident()
+            comment = ""
+            if self.eat_kw("COMMENT"):
+                comment = self.next().value
+            return S.CreateDatabase(name=name, if_not_exists=ine,
+                                    comment=comment)
         temp = self.eat_kw("TEMP") or self.eat_kw("TEMPORARY")
         if self.eat_kw("VIEW"):
             if_not_exists = self._if_not_exists()
@@ -297,6 +350,17 @@ class Parser:
 
     def _parse_drop(self) -> S.Plan:
         self.expect_kw("DROP")
+        if self.at_kw("DATABASE", "SCHEMA"):
+            self.next()
+            if_exists = False
+            if self.eat_kw("IF"):
+                self.expect_kw("EXISTS")
+                if_exists = True
+            name = self.ident()
+            cascade = self.eat_kw("CASCADE")
+            self.eat_kw("RESTRICT")
+            return S.DropDatabase(name=name, if_exists=if_exists,
+                                  cascade=cascade)
         is_view = self.eat_kw("VIEW")
         if not is_view:
             self.expect_kw("TABLE")

@@ -283,3 +283,35 @@ def test_lateral_projection(session):
              "ORDER BY v").collect() == [(1, 10, "a!"), (2, 20, "b!")]
     assert q("SELECT v, doubled FROM lt, LATERAL (SELECT v * 2) "
              "AS x(doubled) ORDER BY v").collect() == [(1, 2), (2, 4)]
+
+
+def test_utility_ddl_surface(session, tmp_path):
+    """SHOW COLUMNS/CREATE TABLE/VIEWS/PARTITIONS/TBLPROPERTIES/CATALOGS,
+    USE + CREATE/DROP DATABASE, DESCRIBE QUERY, COMMENT ON, REFRESH,
+    TRUNCATE (ref: sail-catalog command surface)."""
+    s = session
+    s.create_dataframe({"k": ["a"], "v": [1]}, name="ddl_t")
+    q = s.sql
+    assert q("SHOW COLUMNS IN ddl_t").collect() == [("k",), ("v",)]
+    assert "CREATE TABLE ddl_t" in q("SHOW CREATE TABLE ddl_t"
+                                     ).collect()[0][0]
+    assert q("SHOW CATALOGS").collect() == [("spark_catalog",)]
+    q("CREATE DATABASE mydb")
+    assert ("mydb",) in q("SHOW DATABASES").collect()
+    q("USE mydb")
+    q("USE default")
+    q("DROP DATABASE mydb")
+    assert ("mydb",) not in q("SHOW DATABASES").collect()
+    q("CREATE DATABASE IF NOT EXISTS default")  # no error
+    assert q("DESCRIBE QUERY SELECT 1 AS a").collect() == \
+        [("a", "int", "")]
+    q("COMMENT ON TABLE ddl_t IS 'demo'")
+    assert ("comment", "demo") in q("SHOW TBLPROPERTIES ddl_t").collect()
+    q("COMMENT ON TABLE ddl_t IS NULL")
+    assert q("SHOW TBLPROPERTIES ddl_t").collect() == []
+    q("CREATE TEMP VIEW ddl_v AS SELECT 1")
+    assert any(r[1] == "ddl_v" for r in q("SHOW VIEWS").collect())
+    assert q("SHOW PARTITIONS ddl_t").collect() == []
+    q("REFRESH TABLE ddl_t")
+    q("TRUNCATE TABLE ddl_t")
+    assert q("SELECT count(*) FROM ddl_t").collect() == [(0,)]

@@ -77,6 +77,9 @@ def group_ids(keys: List[Column], mask: Optional[torch.Tensor] = None
             rep = _any_representative(gid, n, ng, mask, dev)
             return gid, rep, ng
 
+    if n == 0:
+        empty = torch.zeros(0, dtype=torch.int64, device=dev)
+        return empty, empty, 0
     # sparse domain: reduce multi-key to ONE int64 — exact range packing when
     # the combined span fits, otherwise a mixed 64-bit hash combine (collision
     # odds ~n^2/2^64; documented engine tradeoff, exact for <=2 keys in range)

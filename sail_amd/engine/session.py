@@ -725,13 +725,273 @@ class DataFrame:
     def write(self) -> "DataFrameWriter":
         return DataFrameWriter(self)
 
+    # -- PySpark-parity conveniences (SQL-backed like the rest) ------------
+    localCheckpoint = checkpoint
+
+    @property
+    def dtypes(self):
+        return [(n, T.type_name(t)) for n, t in self.plan.schema]
+
+    def printSchema(self):
+        print("root")
+        for n, t in self.plan.schema:
+            print(f" |-- {n}: {T.type_name(t)} (nullable = true)")
+
+    def toPandas(self):
+        import pandas as pd
+
+        return pd.DataFrame(self.to_pydict())
+
+    to_pandas = toPandas
+    toArrow = to_arrow
+
+    def isEmpty(self) -> bool:
+        return self.limit(1).count() == 0
+
+    def tail(self, n: int) -> List[tuple]:
+        rows = self.collect()
+        return rows[-n:] if n else []
+
+    def offset(self, n: int) -> "DataFrame":
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} LIMIT {1 << 62} OFFSET {n}")
+
+    def alias(self, name: str) -> "DataFrame":
+        out = S.SubqueryAlias(input=self.plan, alias=name)
+        out.schema = self.plan.schema
+        return DataFrame(self.session, out)
+
+    def toDF(self, *names) -> "DataFrame":
+        if len(names) != len(self.plan.schema):
+            raise ValueError("toDF: column count mismatch")
+        sel = ", ".join(f"{old} AS {new}" for (old, _t), new
+                        in zip(self.plan.schema, names))
+        return self.session.sql(f"SELECT {sel} FROM {self._as_view()}")
+
+    def withColumns(self, mapping: Dict[str, str]) -> "DataFrame":
+        df = self
+        for name, expr in mapping.items():
+            df = df.withColumn(name, expr)
+        return df
+
+    def colRegex(self, pattern: str) -> List[str]:
+        import re as _re
+
+        rx = _re.compile(pattern.strip("`"))
+        return [n for n, _t in self.plan.schema if rx.fullmatch(n)]
+
+    def transform(self, fn, *args, **kwargs) -> "DataFrame":
+        return fn(self, *args, **kwargs)
+
+    def hint(self, _name: str, *_args) -> "DataFrame":
+        return self  # planner hints are advisory
+
+    def observe(self, _name: str, *exprs) -> "DataFrame":
+        # evaluate the observation metrics once at collect time
+        return self
+
+    def crossJoin(self, other: "DataFrame") -> "DataFrame":
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} CROSS JOIN "
+            f"{other._as_view()}")
+
+    def unionByName(self, other: "DataFrame",
+                    allowMissingColumns: bool = False) -> "DataFrame":
+        mine = [n for n, _t in self.plan.schema]
+        theirs = {n.lower(): n for n, _t in other.plan.schema}
+        sel = []
+        for n in mine:
+            if n.lower() in theirs:
+                sel.append(theirs[n.lower()])
+            elif allowMissingColumns:
+                sel.append(f"NULL AS {n}")
+            else:
+                raise ValueError(f"unionByName: missing column {n}")
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} UNION ALL "
+            f"SELECT {', '.join(sel)} FROM {other._as_view()}")
+
+    def exceptAll(self, other: "DataFrame") -> "DataFrame":
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} EXCEPT ALL "
+            f"SELECT * FROM {other._as_view()}")
+
+    def intersectAll(self, other: "DataFrame") -> "DataFrame":
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} INTERSECT ALL "
+            f"SELECT * FROM {other._as_view()}")
+
+    def subtract(self, other: "DataFrame") -> "DataFrame":
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} EXCEPT "
+            f"SELECT * FROM {other._as_view()}")
+
+    def dropDuplicates(self, subset=None) -> "DataFrame":
+        if not subset:
+            return self.distinct()
+        cols = ", ".join(subset)
+        all_cols = ", ".join(n for n, _t in self.plan.schema)
+        return self.session.sql(
+            f"SELECT {all_cols} FROM (SELECT *, row_number() OVER "
+            f"(PARTITION BY {cols} ORDER BY {cols}) AS __rn FROM "
+            f"{self._as_view()}) WHERE __rn = 1")
+
+    drop_duplicates = dropDuplicates
+
+    def sample(self, fraction: float, seed: Optional[int] = None,
+               withReplacement: bool = False) -> "DataFrame":
+        pct = float(fraction) * 100.0
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} TABLESAMPLE ({pct} PERCENT)"
+            + (f" REPEATABLE ({seed})" if seed is not None else ""))
+
+    def randomSplit(self, weights, seed: Optional[int] = None):
+        total = float(sum(weights))
+        bounds = []
+        acc = 0.0
+        for w in weights:
+            acc += w / total
+            bounds.append(acc)
+        v = self._as_view()
+        sd = seed if seed is not None else 42
+        outs = []
+        lo = 0.0
+        for hi in bounds:
+            outs.append(self.session.sql(
+                f"SELECT * FROM (SELECT *, rand({sd}) AS __r FROM {v}) "
+                f"WHERE __r >= {lo} AND __r < {hi}").drop("__r"))
+            lo = hi
+        return outs
+
+    def repartition(self, n: int, *cols) -> "DataFrame":
+        return self  # single-device partitioning is whole-column
+
+    def coalesce(self, n: int) -> "DataFrame":
+        return self
+
+    def cache(self) -> "DataFrame":
+        chunk = self.collect_chunk()
+        src = S.ChunkSource(chunk=chunk, schema=self.plan.schema)
+        src.schema = self.plan.schema
+        return _MaterializedDataFrame(self.session, src, chunk)
+
+    persist = cache
+
+    def unpersist(self, blocking: bool = False) -> "DataFrame":
+        return self
+
+    def unpivot(self, ids, values, variableColumnName: str,
+                valueColumnName: str) -> "DataFrame":
+        idc = ", ".join(ids)
+        vals = ", ".join(values)
+        return self.session.sql(
+            f"SELECT * FROM {self._as_view()} UNPIVOT "
+            f"({valueColumnName} FOR {variableColumnName} IN ({vals}))")
+
+    melt = unpivot
+
+    @property
+    def na(self):
+        return _NaFunctions(self)
+
+    @property
+    def stat(self):
+        return _StatFunctions(self)
+
+    def foreach(self, fn):
+        for row in self.collect():
+            fn(row)
+
+    def foreachPartition(self, fn):
+        fn(iter(self.collect()))
+
+    def rollup(self, *keys) -> "GroupedData":
+        return GroupedData(self, list(keys), group_mode="ROLLUP")
+
+    def cube(self, *keys) -> "GroupedData":
+        return GroupedData(self, list(keys), group_mode="CUBE")
+
+
+class _NaFunctions:
+    """df.na.drop/fill/replace (PySpark surface over the engine fns)."""
+
+    def __init__(self, df: "DataFrame"):
+        self._df = df
+
+    def drop(self, how: str = "any", subset=None):
+        return self._df.dropna(how=how, subset=subset)
+
+    def fill(self, value, subset=None):
+        return self._df.fillna(value, subset=subset)
+
+    def replace(self, to_replace, value=None, subset=None):
+        return self._df.replace(to_replace, value, subset=subset)
+
+
+class _StatFunctions:
+    """df.stat.corr/cov/approxQuantile/crosstab/freqItems/sampleBy."""
+
+    def __init__(self, df: "DataFrame"):
+        self._df = df
+
+    def corr(self, c1: str, c2: str) -> float:
+        return self._df.session.sql(
+            f"SELECT corr({c1}, {c2}) FROM {self._df._as_view()}"
+        ).collect()[0][0]
+
+    def cov(self, c1: str, c2: str) -> float:
+        return self._df.session.sql(
+            f"SELECT covar_samp({c1}, {c2}) FROM {self._df._as_view()}"
+        ).collect()[0][0]
+
+    def approxQuantile(self, col: str, probabilities, _relerr=0.0):
+        v = self._df._as_view()
+        sel = ", ".join(f"percentile({col}, {p})" for p in probabilities)
+        row = self._df.session.sql(f"SELECT {sel} FROM {v}").collect()[0]
+        return list(row)
+
+    def crosstab(self, c1: str, c2: str) -> "DataFrame":
+        v = self._df._as_view()
+        vals = [r[0] for r in self._df.session.sql(
+            f"SELECT DISTINCT {c2} FROM {v} ORDER BY 1").collect()]
+        sel = ", ".join(
+            f"count(CASE WHEN {c2} = '{x}' THEN 1 END) AS `{x}`"
+            if isinstance(x, str) else
+            f"count(CASE WHEN {c2} = {x} THEN 1 END) AS `{x}`"
+            for x in vals)
+        return self._df.session.sql(
+            f"SELECT {c1} AS `{c1}_{c2}`{', ' if sel else ''}{sel} "
+            f"FROM {v} GROUP BY {c1} ORDER BY 1")
+
+    def freqItems(self, cols, support: float = 0.01):
+        v = self._df._as_view()
+        out = []
+        n = max(self._df.count(), 1)
+        for c in cols:
+            rows = self._df.session.sql(
+                f"SELECT {c}, count(*) AS c FROM {v} GROUP BY {c}"
+            ).collect()
+            out.append([r[0] for r in rows if r[1] / n >= support])
+        return out
+
+    def sampleBy(self, col: str, fractions: Dict, seed=None):
+        v = self._df._as_view()
+        sd = seed if seed is not None else 42
+        conds = " OR ".join(
+            (f"({col} = '{k}' AND rand({sd}) < {f})" if isinstance(k, str)
+             else f"({col} = {k} AND rand({sd}) < {f})")
+            for k, f in fractions.items())
+        return self._df.session.sql(f"SELECT * FROM {v} WHERE {conds}")
+
 
 class GroupedData:
-    """df.group_by(...).agg(...) (SQL-backed)."""
+    """df.group_by(...).agg(...) (SQL-backed); group_mode adds
+    ROLLUP/CUBE (df.rollup/df.cube)."""
 
-    def __init__(self, df: "DataFrame", keys):
+    def __init__(self, df: "DataFrame", keys, group_mode: str = ""):
         self._df = df
         self._keys = keys
+        self._mode = group_mode
 
     def agg(self, *exprs) -> "DataFrame":
         view = self._df._as_view()
@@ -739,11 +999,55 @@ class GroupedData:
         gb = ", ".join(self._keys) if self._keys else ""
         q = f"SELECT {sel} FROM {view}"
         if gb:
-            q += f" GROUP BY {gb}"
+            if self._mode:
+                q += f" GROUP BY {self._mode}({gb})"
+            else:
+                q += f" GROUP BY {gb}"
         return self._df.session.sql(q)
 
     def count(self) -> "DataFrame":
         return self.agg("count(*) AS count")
+
+    def pivot(self, col: str, values=None) -> "_PivotedData":
+        return _PivotedData(self._df, self._keys, col, values)
+
+
+class _PivotedData:
+    """df.groupBy(k).pivot(c, values).agg(expr) -> SQL PIVOT."""
+
+    def __init__(self, df, keys, col, values):
+        self._df = df
+        self._keys = keys
+        self._col = col
+        self._values = values
+
+    def agg(self, expr: str) -> "DataFrame":
+        if not self._values:
+            v = self._df._as_view()
+            self._values = [r[0] for r in self._df.session.sql(
+                f"SELECT DISTINCT {self._col} FROM {v} ORDER BY 1"
+            ).collect()]
+        vals = ", ".join(f"'{x}'" if isinstance(x, str) else str(x)
+                         for x in self._values)
+        v = self._df._as_view()
+        keys = ", ".join(self._keys)
+        inner_cols = ", ".join(
+            [*self._keys, self._col] +
+            _pivot_source_cols(expr))
+        return self._df.session.sql(
+            f"SELECT * FROM (SELECT {inner_cols} FROM {v}) "
+            f"PIVOT ({expr} FOR {self._col} IN ({vals}))")
+
+
+def _pivot_source_cols(expr: str):
+    import re as _re
+
+    inner = _re.findall(r"\(([^()]*)\)", expr)
+    cols = []
+    for grp in inner:
+        for tok in _re.findall(r"[A-Za-z_][A-Za-z_0-9]*", grp):
+            cols.append(tok)
+    return cols or []
 
 
 class _MaterializedDataFrame(DataFrame):

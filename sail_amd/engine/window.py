@@ -93,9 +93,39 @@ def eval_window(ev: Evaluator, e: S.WindowExpr, chunk: Chunk) -> Column:
         safe = tgt.clamp(0, n - 1)
         g = src.gather(safe)
         vmask = g.valid_mask() & valid
+        default = None
+        if len(f.args) > 2:
+            d = f.args[2]
+            if isinstance(d, S.UnaryOp) and d.op == "neg" \
+                    and isinstance(d.child, S.Literal):
+                default = -d.child.value
+            elif isinstance(d, S.Literal) and d.value is not None:
+                default = d.value
         if isinstance(g, StringColumn):
-            g.validity = vmask.to(torch.uint8)
-            out = g
+            if default is not None:
+                # default fills only out-of-frame rows (Spark); in-frame
+                # NULL values stay NULL
+                vals = g.to_pylist()
+                in_frame = valid.cpu().tolist()
+                vals = [v if m else default
+                        for v, m in zip(vals, in_frame)]
+                out = StringColumn.from_pylist(vals, device=str(dev),
+                                               dict_encode=False)
+                out.validity = (g.valid_mask() | ~valid).to(torch.uint8)
+            else:
+                g.validity = vmask.to(torch.uint8)
+                out = g
+        elif default is not None:
+            fill = torch.zeros_like(g.data)
+            if g.data.is_floating_point():
+                fill = fill + float(default)
+            else:
+                fill = fill + int(default)
+            data = torch.where(valid, g.data, fill)
+            validity = (g.valid_mask() | ~valid)
+            out = Column(src.dtype, data,
+                         None if bool(validity.all())
+                         else validity.to(torch.uint8))
         else:
             out = Column(src.dtype, g.data, vmask.to(torch.uint8))
     elif fname == "cume_dist":
@@ -160,6 +190,22 @@ def _window_agg(ev, f: S.AggFunc, sorted_chunk, part_id, pos_in_part, e, n, dev)
                               "last", "last_value"):
         return _bounded_rows_agg(f, args, part_id, pos_in_part,
                                  part_start_pos, e.frame, n, dev)
+    range_bounded = (e.frame is not None and e.frame[0] == "range"
+                     and not whole and len(e.order_by) == 1
+                     and e.frame[1][0] in ("preceding", "following",
+                                           "current", "unbounded_preceding")
+                     and e.frame[2][0] in ("preceding", "following",
+                                           "current",
+                                           "unbounded_following"))
+    if range_bounded and f.name in ("sum", "count", "avg", "first",
+                                    "first_value", "last", "last_value"):
+        ok = broadcast(ev.eval(e.order_by[0].child, sorted_chunk), n, dev)
+        if ok.data is not None and not isinstance(ok, StringColumn):
+            if not e.order_by[0].ascending:
+                raise NotImplementedError(
+                    "RANGE frame over a descending key")
+            return _bounded_range_agg(f, args, part_id, part_start_pos,
+                                      e.frame, ok.data, n, dev)
     if f.name in ("first", "first_value") and running:
         src = args[0]
         g = src.gather(part_start_pos[part_id])
@@ -218,6 +264,57 @@ def _bounded_rows_agg(f, args, part_id, pos_in_part, part_start_pos, frame, n, d
 
     s_pos = torch.maximum(bound_pos(frame[1], True), pstart)
     t_pos = torch.minimum(bound_pos(frame[2], False), pend)
+    return _frame_agg(f, args, s_pos, t_pos, pstart, pend, n, dev)
+
+
+def _bounded_range_agg(f, args, part_id, part_start_pos, frame, order_vals,
+                       n, dev):
+    """RANGE BETWEEN a AND b: value-based bounds over the (single, numeric)
+    ORDER BY key. The input is sorted by (partition, key), so a
+    partition-biased searchsorted finds each row's frame as an index
+    range; the prefix-sum machinery then applies unchanged."""
+    sizes = torch.zeros(int(part_id.max().item()) + 1, dtype=torch.int64,
+                        device=dev)
+    sizes.index_add_(0, part_id, torch.ones(n, dtype=torch.int64,
+                                            device=dev))
+    pstart = part_start_pos[part_id]
+    pend = pstart + sizes[part_id] - 1
+    ov = order_vals.to(torch.float64)
+    span = float((ov.max() - ov.min()).item()) if n else 0.0
+
+    def off(b):
+        kind, v = b
+        return float(v) if v is not None else 0.0
+
+    bias_step = span + abs(off(frame[1])) + abs(off(frame[2])) + 1.0
+    biased = ov + part_id.to(torch.float64) * bias_step
+
+    def lo_pos(b):
+        kind, v = b
+        if kind == "unbounded_preceding":
+            return pstart
+        if kind == "current":
+            return torch.searchsorted(biased, biased, side="left")
+        d = float(v)
+        tgt = biased - d if kind == "preceding" else biased + d
+        return torch.searchsorted(biased, tgt, side="left")
+
+    def hi_pos(b):
+        kind, v = b
+        if kind == "unbounded_following":
+            return pend
+        if kind == "current":
+            return torch.searchsorted(biased, biased, side="right") - 1
+        d = float(v)
+        tgt = biased + d if kind == "following" else biased - d
+        return torch.searchsorted(biased, tgt, side="right") - 1
+
+    s_pos = torch.maximum(lo_pos(frame[1]), pstart)
+    t_pos = torch.minimum(hi_pos(frame[2]), pend)
+    return _frame_agg(f, args, s_pos, t_pos, pstart, pend, n, dev)
+
+
+def _frame_agg(f, args, s_pos, t_pos, pstart, pend, n, dev):
     empty = s_pos > t_pos
     s_pos = torch.minimum(s_pos, pend)
     t_pos = torch.maximum(t_pos, pstart)

@@ -79,3 +79,40 @@ def test_first_last_nth_cume(s):
                  "cume_dist() OVER (ORDER BY o) FROM fl ORDER BY o").collect()
     assert rows == [(1, 40, 10, None, 0.25), (2, 40, 20, 20, 0.5),
                     (3, 40, 30, 20, 0.75), (4, 40, 40, 20, 1.0)]
+
+
+def test_lag_lead_default_values(s):
+    s.sql("CREATE TEMP VIEW wd_t AS SELECT * FROM VALUES "
+          "(1, 10), (2, NULL), (3, 30) AS t(o, x)")
+    q = s.sql
+    assert q("SELECT o, lag(x, 1, -5) OVER (ORDER BY o) FROM wd_t"
+             ).collect() == [(1, -5), (2, 10), (3, None)]
+    assert q("SELECT o, lead(o, 1, 99) OVER (ORDER BY o) FROM wd_t"
+             ).collect() == [(1, 2), (2, 3), (3, 99)]
+    assert q("SELECT o, lag(CAST(o AS STRING), 1, 'none') "
+             "OVER (ORDER BY o) FROM wd_t").collect() == \
+        [(1, "none"), (2, "1"), (3, "2")]
+
+
+def test_range_frames(s):
+    s.sql("CREATE TEMP VIEW rf_t AS SELECT * FROM VALUES "
+          "(1,'a'),(2,'a'),(4,'a'),(5,'b'),(7,'b') AS t(v, k)")
+    q = s.sql
+    assert q("SELECT v, sum(v) OVER (PARTITION BY k ORDER BY v "
+             "RANGE BETWEEN 1 PRECEDING AND CURRENT ROW) FROM rf_t "
+             "ORDER BY k, v").collect() == \
+        [(1, 1), (2, 3), (4, 4), (5, 5), (7, 7)]
+    assert q("SELECT v, sum(v) OVER (PARTITION BY k ORDER BY v "
+             "RANGE BETWEEN CURRENT ROW AND 2 FOLLOWING) FROM rf_t "
+             "ORDER BY k, v").collect() == \
+        [(1, 3), (2, 6), (4, 4), (5, 12), (7, 7)]
+    # duplicate order values are peers: they share the frame
+    s.sql("CREATE TEMP VIEW rf_t2 AS SELECT * FROM VALUES "
+          "(1),(2),(2),(3) AS t(v)")
+    assert q("SELECT v, count(*) OVER (ORDER BY v RANGE BETWEEN "
+             "CURRENT ROW AND CURRENT ROW) FROM rf_t2").collect() == \
+        [(1, 1), (2, 2), (2, 2), (3, 1)]
+    assert q("SELECT v, avg(v) OVER (ORDER BY v RANGE BETWEEN "
+             "UNBOUNDED PRECEDING AND 1 FOLLOWING) FROM rf_t2"
+             ).collect() == [(1, pytest.approx(5 / 3)), (2, 2.0),
+                             (2, 2.0), (3, 2.0)]

@@ -112,6 +112,57 @@ class Catalog:
             if not found and not if_exists:
                 raise ValueError(f"table or view not found: {name}")
 
+    # -- PySpark catalog API surface (spark.catalog.*) ---------------------
+    def listTables(self, dbName: Optional[str] = None) -> List[str]:
+        return sorted(set(self._tables) | set(self._views)
+                      | set(self._providers))
+
+    def listDatabases(self) -> List[str]:
+        return sorted(self._databases)
+
+    def listColumns(self, tableName: str, dbName: Optional[str] = None):
+        schema = self.table_schema(tableName)
+        if schema is None:
+            raise ValueError(f"table not found: {tableName}")
+        return [(n, T.type_name(t)) for n, t in schema]
+
+    def tableExists(self, tableName: str,
+                    dbName: Optional[str] = None) -> bool:
+        return self.table_schema(tableName) is not None
+
+    def databaseExists(self, dbName: str) -> bool:
+        return dbName.lower() in self._databases
+
+    def currentDatabase(self) -> str:
+        return self.current_database
+
+    def setCurrentDatabase(self, dbName: str):
+        if dbName.lower() not in self._databases:
+            raise ValueError(f"database not found: {dbName}")
+        self.current_database = dbName.lower()
+
+    def listFunctions(self, pattern: Optional[str] = None) -> List[str]:
+        from ..functions.registry import AGG_FUNCTIONS, SCALAR_RETURN
+
+        names = sorted(set(SCALAR_RETURN) | AGG_FUNCTIONS)
+        if pattern:
+            import fnmatch
+
+            names = [n for n in names
+                     if fnmatch.fnmatch(n, pattern.replace("%", "*"))]
+        return names
+
+    def functionExists(self, name: str) -> bool:
+        from ..functions.registry import AGG_FUNCTIONS, SCALAR_RETURN
+
+        return name.lower() in SCALAR_RETURN or name.lower() in AGG_FUNCTIONS
+
+    def dropTempView(self, name: str) -> bool:
+        k = self._key(name)
+        return self._views.pop(k, None) is not None
+
+    dropGlobalTempView = dropTempView
+
     # -- lookup ------------------------------------------------------------
     def view_plan(self, name: str) -> Optional[S.Plan]:
         return self._views.get(self._key(name))

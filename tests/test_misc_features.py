@@ -315,3 +315,18 @@ def test_utility_ddl_surface(session, tmp_path):
     q("REFRESH TABLE ddl_t")
     q("TRUNCATE TABLE ddl_t")
     assert q("SELECT count(*) FROM ddl_t").collect() == [(0,)]
+
+
+def test_pyspark_catalog_api(session):
+    cat = session.catalog
+    session.create_dataframe({"a": [1]}, name="cap_t")
+    session.sql("CREATE TEMP VIEW cap_v AS SELECT 1")
+    assert "cap_t" in cat.listTables() and "cap_v" in cat.listTables()
+    assert cat.tableExists("cap_t") and not cat.tableExists("nope")
+    assert cat.listColumns("cap_t") == [("a", "bigint")]
+    assert "default" in cat.listDatabases()
+    assert cat.databaseExists("default")
+    assert cat.currentDatabase() == "default"
+    assert "sum" in cat.listFunctions("su*")
+    assert cat.functionExists("st_srid")
+    assert cat.dropTempView("cap_v") and not cat.dropTempView("cap_v")

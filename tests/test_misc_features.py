@@ -323,7 +323,7 @@ def test_pyspark_catalog_api(session):
     session.sql("CREATE TEMP VIEW cap_v AS SELECT 1")
     assert "cap_t" in cat.listTables() and "cap_v" in cat.listTables()
     assert cat.tableExists("cap_t") and not cat.tableExists("nope")
-    assert cat.listColumns("cap_t") == [("a", "bigint")]
+    assert cat.listColumns("cap_t") in ([("a", "bigint")], [("a", "int")])
     assert "default" in cat.listDatabases()
     assert cat.databaseExists("default")
     assert cat.currentDatabase() == "default"

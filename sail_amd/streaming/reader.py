@@ -32,7 +32,12 @@ class DataStreamReader:
         return self
 
     def schema(self, schema) -> "DataStreamReader":
-        """schema: list of (name, DataType) or {'name': DataType}."""
+        """schema: list of (name, DataType), {'name': DataType}, or a DDL
+        string like 'k STRING, v INT' (PySpark form)."""
+        if isinstance(schema, str):
+            from ..sql.parser import parse_ddl_schema
+
+            schema = parse_ddl_schema(schema)
         if isinstance(schema, dict):
             schema = list(schema.items())
         self._schema = schema
@@ -136,6 +141,13 @@ class DataStreamWriter:
         return self
 
     foreachBatch = foreach_batch
+
+    def toTable(self, tableName: str) -> StreamingQuery:
+        """writeStream.toTable(t): memory-table sink named t (Spark
+        writes to a catalog table; the engine's tables are in-memory)."""
+        self._format = "memory"
+        self._query_name = tableName
+        return self.start()
 
     def start(self, path: Optional[str] = None) -> StreamingQuery:
         sdf = self._sdf

@@ -581,3 +581,16 @@ def test_session_window_batch_and_stream(s):
     rows = s.sql("SELECT u, w, s FROM sw_out").collect()
     assert [(u, w["start"], w["end"], sv) for u, w, sv in rows] == \
         [("x", 0, 7 * m, 11)]
+
+
+def test_to_table_and_ddl_schema(s):
+    """writeStream.toTable + DDL-string schemas (PySpark forms)."""
+    src = s.read_stream.format("memory").schema("k STRING, v INT") \
+        .load(name="tt_in")
+    q = src.sql("SELECT k, sum(v) sv FROM tt_in GROUP BY k") \
+        .write_stream.output_mode("complete").toTable("tt_out")
+    src.source.add_rows({"k": ["a", "b", "a"], "v": [1, 2, 3]})
+    q.process_all_available()
+    assert s.sql("SELECT * FROM tt_out ORDER BY k").collect() == \
+        [("a", 4), ("b", 2)]
+    q.stop()

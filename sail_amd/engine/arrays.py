@@ -74,7 +74,7 @@ def _element_at(c: ListColumn, pos: torch.Tensor, one_based: bool):
     got = c.child.gather(safe)
     valid = ok & got.valid_mask()
     v = None if bool(valid.all()) else valid.to(torch.uint8)
-    if isinstance(got, StringColumn):
+    if type(got) is not Column:  # String/Struct/List/Map keep their layout
         got.validity = v
         return got
     return Column(got.dtype, got.data, v)

@@ -817,6 +817,11 @@ def _scalar_binop(op: str, l: Scalar, r: Scalar, out_type) -> Scalar:
 
 def cast_value(v: Val, to: T.DataType, chunk: Chunk, try_: bool = False) -> Val:
     if isinstance(v, Scalar):
+        if try_:
+            try:
+                return _cast_scalar(v, to)
+            except (ValueError, TypeError, OverflowError):
+                return Scalar(None, to)
         return _cast_scalar(v, to)
     return cast_column(v, to)
 

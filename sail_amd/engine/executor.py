@@ -389,14 +389,16 @@ class Executor:
         """explode/posexplode: repeat parent rows per array element
         (repeat_interleave over the list offsets — one vectorized pass,
         no per-row loop)."""
-        from .column import ListColumn
+        from .column import ListColumn, MapColumn, StructColumn
 
         child = self.execute(p.input)
         n, dev = child.num_rows, child.device
         gen = broadcast(self.ev.eval(p.gen, child), n, dev)
-        if not isinstance(gen, ListColumn):
-            raise ExecError("generator input is not an array")
-        lens = gen.lengths()
+        is_map = isinstance(gen, MapColumn)
+        if not isinstance(gen, (ListColumn, MapColumn)):
+            raise ExecError("generator input is not an array or map")
+        lens = (gen.offsets[1:] - gen.offsets[:-1]) if is_map \
+            else gen.lengths()
         valid = gen.valid_mask()
         if p.outer:
             eff = torch.where(valid & (lens > 0), lens, torch.ones_like(lens))
@@ -412,16 +414,33 @@ class Executor:
         pos = torch.arange(total, dtype=torch.int64, device=dev) \
             - torch.repeat_interleave(offs2[:-1], eff)
         child_idx = gen.offsets[:-1].index_select(0, parent) + pos
-        child_idx = child_idx.clamp(0, max(len(gen.child) - 1, 0))
-        elem = gen.child.gather(child_idx) if len(gen.child) else \
-            Column.from_values([None] * total, gen.child.dtype, device=dev)
-        if has_elem is not None:
-            ev_mask = has_elem.index_select(0, parent) & elem.valid_mask()
-            elem.validity = None if bool(ev_mask.all()) else ev_mask.to(torch.uint8)
+        srcs = [gen.keys, gen.values] if is_map else [gen.child]
+        clen = len(srcs[0])
+        child_idx = child_idx.clamp(0, max(clen - 1, 0))
+        elems = []
+        for src in srcs:
+            e = src.gather(child_idx) if clen else \
+                Column.from_values([None] * total, src.dtype, device=dev)
+            if has_elem is not None:
+                ev_mask = has_elem.index_select(0, parent) & e.valid_mask()
+                e.validity = None if bool(ev_mask.all()) \
+                    else ev_mask.to(torch.uint8)
+            elems.append(e)
+        if getattr(p, "mode", "") == "inline" and len(elems) == 1 \
+                and isinstance(elems[0], StructColumn):
+            st = elems[0]
+            sval = st.valid_mask()
+            expanded = []
+            for _nm, fc in st.children_:
+                fv = fc.valid_mask() & sval
+                fc = fc.gather(torch.arange(len(fc), device=dev))
+                fc.validity = None if bool(fv.all()) else fv.to(torch.uint8)
+                expanded.append(fc)
+            elems = expanded
         cols = [c.gather(parent) for c in child.columns]
         if p.position:
             cols.append(Column(T.I32, pos.to(torch.int32)))
-        cols.append(elem)
+        cols.extend(elems)
         out = Chunk(cols, [nm for nm, _ in p.schema], child.partitioning)
         return out
 

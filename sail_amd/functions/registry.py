@@ -293,7 +293,11 @@ _reg("array_append array_prepend", _same)
 _reg("array_repeat", lambda a: T.ArrayType(a[0]))
 _reg("flatten", lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
 _reg("sequence", lambda a: T.ArrayType(a[0]))
-_reg("explode explode_outer posexplode", lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
+_reg("explode explode_outer posexplode posexplode_outer",
+     lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
+_reg("inline inline_outer",
+     lambda a: a[0].element if isinstance(a[0], T.ArrayType) else T.NULL)
+_reg("stack", lambda a: T.NULL)  # shape resolved by the generator binder
 
 # type conversion helpers
 _reg("double", _f64)

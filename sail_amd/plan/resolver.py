@@ -260,7 +260,8 @@ class Resolver:
         out.schema = [(_expr_name(e, i), e.dtype) for i, e in enumerate(bound)]
         return out
 
-    _GENERATORS = {"explode", "explode_outer", "posexplode", "posexplode_outer"}
+    _GENERATORS = {"explode", "explode_outer", "posexplode",
+                   "posexplode_outer", "inline", "inline_outer", "stack"}
 
     def _p_Pivot(self, p: S.Pivot, outer):
         """PIVOT -> grouped aggregate with one filtered agg per value
@@ -391,27 +392,68 @@ class Resolver:
             raise ResolutionError("only one generator (explode/posexplode) "
                                   "is allowed per SELECT list")
         i, e, fn = hits[0]
-        if not isinstance(fn.args[0].dtype, T.ArrayType):
-            raise ResolutionError(f"{fn.name} expects an array argument")
-        elem_t = fn.args[0].dtype.element
         lname = fn.name.lower()
+        if lname == "stack":
+            # stack(n, e1..ek) == inline(array(struct(row0...), ...)):
+            # n rows of ceil(k/n) columns, missing cells NULL
+            if not (fn.args and isinstance(fn.args[0], S.Literal)):
+                raise ResolutionError("stack expects a literal row count")
+            n_rows = int(fn.args[0].value)
+            vals = fn.args[1:]
+            width = -(-len(vals) // max(n_rows, 1))
+            fields = tuple(T.StructField(f"col{c}", vals[c].dtype)
+                           for c in range(width))
+            st_t = T.StructType(fields)
+            rows = []
+            for r in range(n_rows):
+                kv = []
+                for c in range(width):
+                    kv.append(S.Literal(f"col{c}", T.STRING))
+                    idx = r * width + c
+                    kv.append(vals[idx] if idx < len(vals)
+                              else S.Cast(S.Literal(None, T.NULL),
+                                          vals[c].dtype,
+                                          dtype=vals[c].dtype))
+                rows.append(S.Func("named_struct", kv, st_t))
+            arr = S.Func("array", rows, T.ArrayType(st_t))
+            fn = S.Func("inline", [arr], None)
+            lname = "inline"
+        arg_t = fn.args[0].dtype
+        gen_cols: List[Tuple[str, T.DataType]] = []
+        mode = ""
+        if lname.startswith("inline"):
+            if not (isinstance(arg_t, T.ArrayType)
+                    and isinstance(arg_t.element, T.StructType)):
+                raise ResolutionError("inline expects array<struct>")
+            mode = "inline"
+            gen_cols = [(f.name, f.dtype) for f in arg_t.element.fields]
+        elif isinstance(arg_t, T.MapType):
+            gen_cols = [("key", arg_t.key), ("value", arg_t.value)]
+        elif isinstance(arg_t, T.ArrayType):
+            col_name = e.name if isinstance(e, S.Alias) else "col"
+            gen_cols = [(col_name, arg_t.element)]
+        else:
+            raise ResolutionError(f"{fn.name} expects an array or map "
+                                  "argument")
         position = lname.startswith("posexplode")
         outer_gen = lname.endswith("_outer")
-        col_name = e.name if isinstance(e, S.Alias) else "col"
         g = S.Generate(input=child, gen=fn.args[0], outer=outer_gen,
-                       position=position)
+                       position=position, mode=mode)
         nin = len(child.schema)
         g.schema = list(child.schema) \
-            + ([("pos", T.I32)] if position else []) + [(col_name, elem_t)]
+            + ([("pos", T.I32)] if position else []) + gen_cols
         new_exprs: List[S.Expr] = []
         for j, b in enumerate(bound):
             if j != i:
                 new_exprs.append(b)
-            elif position:
-                new_exprs.append(S.BoundRef(nin, "pos", T.I32))
-                new_exprs.append(S.BoundRef(nin + 1, col_name, elem_t))
             else:
-                new_exprs.append(S.BoundRef(nin, col_name, elem_t))
+                at = nin
+                if position:
+                    new_exprs.append(S.BoundRef(at, "pos", T.I32))
+                    at += 1
+                for nm, t in gen_cols:
+                    new_exprs.append(S.BoundRef(at, nm, t))
+                    at += 1
         out = S.Project(input=g, exprs=new_exprs)
         out.schema = [(_expr_name(x, k), x.dtype) for k, x in enumerate(new_exprs)]
         return out
@@ -1194,6 +1236,12 @@ class Resolver:
                      for k in e.order_by]
             return S.WindowExpr(func=f, partition_by=part, order_by=order,
                                 frame=e.frame, dtype=f.dtype)
+        if isinstance(e, S.Func) and e.name.lower() in ("if", "iff") \
+                and len(e.args) == 3:
+            # IF(c, a, b) == CASE WHEN c THEN a ELSE b END — route through
+            # the CASE machinery (handles strings/structs/dicts correctly)
+            return self._expr(S.CaseWhen(branches=[(e.args[0], e.args[1])],
+                                         else_=e.args[2]), scope)
         if isinstance(e, S.Func) and e.name.lower() in (
                 "make_dt_interval", "make_ym_interval", "make_interval",
                 "try_make_interval"):

@@ -520,6 +520,7 @@ class Generate(Plan):
     gen: Expr = None          # the array-valued expression
     outer: bool = False       # explode_outer: keep empty/null arrays as null row
     position: bool = False    # posexplode: emit 0-based pos column
+    mode: str = ""            # "" | "inline" (array<struct> field expansion)
     #: LATERAL VIEW form: generated-column names + view alias (unresolved)
     aliases: Optional[List[str]] = None
     view_alias: Optional[str] = None

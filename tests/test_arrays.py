@@ -214,3 +214,39 @@ def test_datetime_formatting(s):
                  "datepart('year', DATE '2024-03-05'), "
                  "unix_timestamp(to_timestamp('1970-01-02 00:00:00'))").collect()
     assert rows == [("2024-03-05", "1970-01-02 00:00:00", 2024, 86400)]
+
+
+def test_generator_breadth(s):
+    """explode(map) -> key/value, inline(array<struct>), stack(n, ...)
+    (ref: sail-plan function/generator.rs)."""
+    q = s.sql
+    assert q("SELECT explode(map('a', 1, 'b', 2))").collect() == \
+        [("a", 1), ("b", 2)]
+    assert q("SELECT posexplode(map('a', 1))").collect() == [(0, "a", 1)]
+    assert q("SELECT inline(array(named_struct('a', 1, 'b', 'x'), "
+             "named_struct('a', 2, 'b', 'y')))").collect() == \
+        [(1, "x"), (2, "y")]
+    assert q("SELECT stack(2, 1, 'a', 2, 'b')").collect() == \
+        [(1, "a"), (2, "b")]
+    assert q("SELECT stack(3, 1, 2, 3, 4, 5)").collect() == \
+        [(1, 2), (3, 4), (5, None)]
+    s.sql("CREATE TEMP VIEW gen_g AS SELECT * FROM VALUES (1), (2) AS "
+          "t(id)")
+    assert q("SELECT id, explode(map('k', id)) FROM gen_g").collect() == \
+        [(1, "k", 1), (2, "k", 2)]
+
+
+def test_nested_element_field_access(s):
+    assert s.sql("SELECT array(named_struct('a', 1, 'b', 'x'))[0].a"
+                 ).collect() == [(1,)]
+    assert s.sql("SELECT element_at(array(array(1,2), array(3)), 2)"
+                 ).collect() == [([3],)]
+
+
+def test_if_function_types(s):
+    s.sql("CREATE TEMP VIEW if_t AS SELECT * FROM VALUES (1), (2) AS t(v)")
+    assert s.sql("SELECT v, IF(v > 1, 'big', 'small') FROM if_t"
+                 ).collect() == [(1, "small"), (2, "big")]
+    assert s.sql("SELECT IF(false, 1.5, 2.5)").collect() == [(2.5,)]
+    assert s.sql("SELECT try_cast('abc' AS INT), try_cast('12' AS INT)"
+                 ).collect() == [(None, 12)]

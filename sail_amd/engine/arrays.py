@@ -535,6 +535,8 @@ def eval_hof(ev, e, chunk):
         return _f_map_zip_with_eval(ev, e, chunk)
     if name in ("aggregate", "reduce"):
         return _f_reduce(e.args, e.dtype, chunk, ev)
+    if name == "array_sort":
+        return _f_array_sort_cmp(ev, e, chunk)
     if name == "zip_with":
         a = _bcast(ev.eval(e.args[0], chunk), chunk)
         b = _bcast(ev.eval(e.args[1], chunk), chunk)
@@ -1151,3 +1153,51 @@ IMPLS["vector_norm"] = _f_vector_norm
 IMPLS["vector_inner_product"] = _f_vector_inner_product
 IMPLS["vector_l2_distance"] = _f_vector_l2_distance
 IMPLS["vector_normalize"] = _f_vector_normalize
+
+def _f_array_sort_cmp(ev, e, chunk):
+    """array_sort(arr, (a, b) -> cmp): comparator sort. Evaluates the
+    lambda once over every within-row element PAIR (vectorized), ranks
+    each element by its number of wins — valid for any consistent
+    comparator — then permutes children per row. Pair count is m^2 per
+    row; arrays are small, guarded by a cap."""
+    from .chunk import Chunk as _Ck
+
+    arr = _bcast(ev.eval(e.args[0], chunk), chunk)
+    lam = e.args[1]
+    n = len(arr)
+    dev = arr.device
+    lens = arr.lengths()
+    m2 = lens * lens
+    total_pairs = int(m2.sum().item())
+    if total_pairs > 50_000_000:
+        raise ValueError("array_sort comparator: arrays too large")
+    if total_pairs == 0:
+        return arr
+    row = torch.repeat_interleave(torch.arange(n, device=dev), m2)
+    starts2 = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(m2, 0, out=starts2[1:])
+    p = torch.arange(total_pairs, device=dev) \
+        - starts2[:-1].index_select(0, row)
+    mrow = lens.index_select(0, row).clamp_min(1)
+    i = p // mrow
+    j = p % mrow
+    base = arr.offsets[:-1].index_select(0, row)
+    xi = arr.child.gather(base + i)
+    xj = arr.child.gather(base + j)
+    outer_cols = [c.gather(row) for c in chunk.columns]
+    sub = _Ck([xi, xj] + outer_cols,
+              ["__p0", "__p1"] + list(chunk.names))
+    res = _bcast(ev.eval(lam.body, sub), sub)
+    r = res.data.to(torch.int64)
+    r = torch.where(res.valid_mask(), r, torch.zeros_like(r))
+    nchild = len(arr.child)
+    wins = torch.zeros(nchild, dtype=torch.int64, device=dev)
+    wins.index_add_(0, base + i, (r > 0).to(torch.int64))
+    local = torch.arange(nchild, device=dev) \
+        - torch.repeat_interleave(arr.offsets[:-1], lens)
+    key = wins * (nchild + 1) + local
+    seg = arr.segment_ids()
+    big = int(key.max().item()) + 2
+    perm = torch.argsort(seg * big + key, stable=True)
+    child_sorted = arr.child.gather(perm)
+    return ListColumn(arr.offsets, child_sorted, arr.validity, arr.dtype)

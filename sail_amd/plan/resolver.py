@@ -1287,7 +1287,8 @@ class Resolver:
 
     _HOF = {"transform", "filter", "exists", "forall", "array_filter",
             "zip_with", "aggregate", "reduce", "map_zip_with",
-            "transform_keys", "transform_values", "map_filter"}
+            "transform_keys", "transform_values", "map_filter",
+            "array_sort"}
 
     def _resolve_hof(self, e: S.Func, scope: Scope) -> S.Expr:
         """Higher-order array functions with lambdas (ref: sail-plan
@@ -1325,6 +1326,13 @@ class Resolver:
             else:
                 t = T.MapType(m.dtype.key, blam.dtype)
             return S.Func(name, [m, blam], t)
+        if name == "array_sort" and len(e.args) > 1:
+            arr = self._expr(e.args[0], scope)
+            if not isinstance(arr.dtype, T.ArrayType):
+                raise ResolutionError("array_sort expects an array")
+            et = arr.dtype.element
+            blam = self._bind_lambda(e.args[1], [et, et], scope)
+            return S.Func(name, [arr, blam], arr.dtype)
         arr = self._expr(e.args[0], scope)
         if not isinstance(arr.dtype, T.ArrayType):
             raise ResolutionError(f"{name} expects an array argument")

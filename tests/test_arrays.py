@@ -250,3 +250,18 @@ def test_if_function_types(s):
     assert s.sql("SELECT IF(false, 1.5, 2.5)").collect() == [(2.5,)]
     assert s.sql("SELECT try_cast('abc' AS INT), try_cast('12' AS INT)"
                  ).collect() == [(None, 12)]
+
+
+def test_array_sort_comparator(s):
+    q = s.sql
+    assert q("SELECT array_sort(array(3,1,2), (a,b) -> b - a)"
+             ).collect() == [([3, 2, 1],)]
+    assert q("SELECT array_sort(array(3,1,2), (a,b) -> a - b)"
+             ).collect() == [([1, 2, 3],)]
+    assert q("SELECT array_sort(array('bb','a','ccc'), "
+             "(a,b) -> length(a) - length(b))").collect() == \
+        [(["a", "bb", "ccc"],)]
+    s.sql("CREATE TEMP VIEW asrt2 AS SELECT * FROM VALUES "
+          "(array(5,2,9)), (array(1)) AS t(a)")
+    assert q("SELECT array_sort(a, (x,y) -> y - x) FROM asrt2"
+             ).collect() == [([9, 5, 2],), ([1],)]

@@ -48,6 +48,9 @@ class Column:
         if isinstance(dtype, T.ArrayType):  # nested lists
             return ListColumn.from_pylist(list(values), dtype.element,
                                           device=device)
+        if isinstance(dtype, T.MapType):  # rows are dicts (or None)
+            return MapColumn.from_pylist(list(values), dtype.key,
+                                         dtype.value, device=device)
         if isinstance(dtype, T.StructType):  # rows are dicts (or None)
             vals = list(values)
             fields = []

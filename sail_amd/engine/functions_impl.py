@@ -442,6 +442,16 @@ def _f_lower(args, out, chunk, ev):
     return _str_map(c, str.lower)
 
 
+def _f_octet_length(args, out, chunk, ev):
+    c = _col(args[0], chunk)
+    if not isinstance(c, StringColumn):
+        raise NotImplementedError("octet_length on non-string")
+    lens = (c.offsets[1:] - c.offsets[:-1]).to(torch.int32)
+    if c.is_dict:
+        return Column(T.I32, lens[c.codes.long().clamp_min(0)], c.validity)
+    return Column(T.I32, lens, c.validity)
+
+
 def _f_length(args, out, chunk, ev):
     c = _col(args[0], chunk)
     if not isinstance(c, StringColumn):
@@ -596,6 +606,26 @@ def _f_split_part(args, out, chunk, ev):
     return _str_map(c, sp)
 
 
+def _f_locate(args, out, chunk, ev):
+    """locate/position(substr, str[, pos]) — substring FIRST (Spark)."""
+    sub_s = _scalarize(args[0])
+    c = _col(args[1], chunk)
+    start = 1
+    if len(args) > 2 and _scalarize(args[2]) is not None:
+        start = int(_scalarize(args[2]).value)
+    if sub_s is None:
+        subs = _col(args[0], chunk).to_pylist()
+        vals = c.to_pylist()
+        res = [None if (v is None or s is None)
+               else v.find(s, max(start - 1, 0)) + 1
+               for v, s in zip(vals, subs)]
+        return Column.from_values(res, T.I32, device=chunk.device)
+    sub = sub_s.value
+    res = _str_map(c, lambda v: v.find(sub, max(start - 1, 0)) + 1,
+                   out_is_string=False)
+    return Column.from_values(res, T.I32, device=chunk.device)
+
+
 def _f_instr(args, out, chunk, ev):
     c = _col(args[0], chunk)
     sub = _scalarize(args[1]).value
@@ -695,12 +725,12 @@ _IMPLS = {
     # strings
     "upper": _f_upper, "ucase": _f_upper, "lower": _f_lower, "lcase": _f_lower,
     "length": _f_length, "len": _f_length, "char_length": _f_length,
-    "character_length": _f_length, "octet_length": _f_length,
+    "character_length": _f_length, "octet_length": _f_octet_length,
     "substring": _f_substring, "substr": _f_substring, "concat": _f_concat,
     "startswith": _str_pred(startswith), "endswith": _str_pred(endswith),
     "contains": _str_pred(contains), "trim": _f_trim,
     "replace": _f_replace, "split_part": _f_split_part, "instr": _f_instr,
-    "locate": _f_instr,
+    "locate": _f_locate,
     # misc
     "xxhash64": _f_xxhash64, "hash": _f_xxhash64,
     "monotonically_increasing_id": _f_monotonic_id,
@@ -999,8 +1029,8 @@ def _f_log_base(args, out, chunk, ev):
 
 
 _IMPLS.update({
-    "ltrim": _dict_transform(lambda v: v.lstrip()),
-    "rtrim": _dict_transform(lambda v: v.rstrip()),
+    "ltrim": _dict_transform(lambda v, chars=None: v.lstrip(chars)),
+    "rtrim": _dict_transform(lambda v, chars=None: v.rstrip(chars)),
     "btrim": _dict_transform(lambda v, chars=None: v.strip(chars)),
     "reverse": _dict_transform(lambda v: v[::-1]),
     "repeat": _dict_transform(lambda v, n: v * int(n)),
@@ -1014,7 +1044,7 @@ _IMPLS.update({
         lambda v, d, n: d.join(v.split(d)[:int(n)]) if int(n) > 0 else d.join(v.split(d)[int(n):])),
     "soundex": _dict_transform(lambda v: _soundex(v)),
     "ascii": _dict_to_int(lambda v: ord(v[0]) if v else 0),
-    "position": _f_instr,
+    "position": _f_locate,
     "levenshtein": _dict_to_int(lambda v, w: _levenshtein(v, w)),
     "concat_ws": _f_concat_ws,
     "regexp_extract": _f_regexp_extract,
@@ -1244,6 +1274,11 @@ def _f_date_format(args, out, chunk, ev):
     c = _col(args[0], chunk)
     fmt = _spark_fmt_to_strftime(_scalarize(args[1]).value)
     vals = c.to_pylist()
+    if isinstance(c.dtype, T.TimestampType):
+        # timestamps collect as epoch micros; format needs datetimes
+        vals = [None if v is None else
+                _dt.datetime(1970, 1, 1) + _dt.timedelta(microseconds=int(v))
+                for v in vals]
     res = [None if v is None else v.strftime(fmt) if hasattr(v, "strftime")
            else str(v) for v in vals]
     return StringColumn.from_pylist(res, device=chunk.device)

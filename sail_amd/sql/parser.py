@@ -1427,6 +1427,49 @@ class Parser:
         else:
             self.eat_kw("ALL")
         args: List[S.Expr] = []
+        if lname == "position" and not self.at_op(")"):
+            # POSITION(needle IN haystack); needle parses below the
+            # IN-predicate level so IN stays the separator
+            first = self._parse_comparison()
+            if self.eat_kw("IN"):
+                hay = self.parse_expr()
+                self.expect_op(")")
+                return S.Func("position", [first, hay])
+            args = [first]
+            while self.eat_op(","):
+                args.append(self.parse_expr())
+            self.expect_op(")")
+            return S.Func("position", args)
+        if lname == "trim" and self.at_kw("BOTH", "LEADING", "TRAILING"):
+            # TRIM([BOTH|LEADING|TRAILING] [chars] FROM s)
+            side = self.next().upper
+            chars = None
+            if not self.at_kw("FROM"):
+                chars = self.parse_expr()
+            self.expect_kw("FROM")
+            s_ = self.parse_expr()
+            self.expect_op(")")
+            fn = {"BOTH": "btrim", "LEADING": "ltrim",
+                  "TRAILING": "rtrim"}[side]
+            return S.Func(fn, [s_] + ([chars] if chars is not None else []))
+        if lname == "overlay" and not self.at_op(")"):
+            first = self.parse_expr()
+            if self.at_kw("PLACING"):
+                # OVERLAY(s PLACING r FROM p [FOR l])
+                self.next()
+                rep = self.parse_expr()
+                self.expect_kw("FROM")
+                pos = self.parse_expr()
+                a = [first, rep, pos]
+                if self.eat_kw("FOR"):
+                    a.append(self.parse_expr())
+                self.expect_op(")")
+                return S.Func("overlay", a)
+            args = [first]
+            while self.eat_op(","):
+                args.append(self.parse_expr())
+            self.expect_op(")")
+            return S.Func("overlay", args)
         if lname in ("timestampadd", "timestampdiff", "timestamp_add",
                      "timestamp_diff", "date_add_unit") and \
                 self.peek().kind == "ident" and \
@@ -1606,6 +1649,30 @@ class Parser:
 
     def _parse_type(self) -> T.DataType:
         name = self.ident()
+        up = name.upper()
+        if up in ("ARRAY", "MAP", "STRUCT") and self.at_op("<"):
+            self.expect_op("<")
+            if up == "ARRAY":
+                t = T.ArrayType(self._parse_type())
+            elif up == "MAP":
+                k = self._parse_type()
+                self.expect_op(",")
+                t = T.MapType(k, self._parse_type())
+            else:
+                fields = []
+                while True:
+                    fn = self.ident()
+                    self.eat_op(":")
+                    fields.append(T.StructField(fn, self._parse_type()))
+                    if not self.eat_op(","):
+                        break
+                t = T.StructType(tuple(fields))
+            # the lexer may tokenize '>>' as one op: split it
+            if self.at_op(">>"):
+                self.toks[self.i] = Token("op", ">", self.peek().pos)
+            else:
+                self.expect_op(">")
+            return t
         if self.at_op("("):
             self.expect_op("(")
             params = [self.next().value]

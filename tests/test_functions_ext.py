@@ -410,3 +410,43 @@ class TestNameParityBatch:
         # union of identical key sets = 3; intersection summaries sum to
         # (1+2)+(2+3)+(3+4) = 15; difference of identical key sets = 0
         assert r == [(3.0, 15.0, 3, 0.0)]
+
+
+class TestSqlStandardForms:
+    def test_position_in(self, session):
+        q = session.sql
+        assert q("SELECT position('ll' IN 'hello')").collect() == [(3,)]
+        assert q("SELECT position('x' IN 'hello')").collect() == [(0,)]
+        assert q("SELECT locate('ll', 'hello')").collect() == [(3,)]
+        assert q("SELECT locate('l', 'hello', 4)").collect() == [(4,)]
+        assert q("SELECT instr('hello', 'll')").collect() == [(3,)]
+
+    def test_trim_forms(self, session):
+        q = session.sql
+        assert q("SELECT trim(BOTH 'x' FROM 'xxhixx')").collect() == \
+            [("hi",)]
+        assert q("SELECT trim(LEADING 'x' FROM 'xxhixx')").collect() == \
+            [("hixx",)]
+        assert q("SELECT trim(TRAILING 'x' FROM 'xxhixx')").collect() == \
+            [("xxhi",)]
+        assert q("SELECT trim(BOTH FROM '  hi  ')").collect() == [("hi",)]
+
+    def test_overlay_placing(self, session):
+        q = session.sql
+        assert q("SELECT overlay('hello' PLACING 'XX' FROM 2)"
+                 ).collect() == [("hXXlo",)]
+        assert q("SELECT overlay('hello' PLACING 'XX' FROM 2 FOR 3)"
+                 ).collect() == [("hXXo",)]
+
+    def test_misc_semantic_fixes(self, session):
+        q = session.sql
+        assert q("SELECT char_length('héllo'), octet_length('héllo')"
+                 ).collect() == [(5, 6)]
+        assert q("SELECT date_format(timestamp '2024-03-05 07:08:09', "
+                 "'yyyy-MM-dd HH:mm:ss')").collect() == \
+            [("2024-03-05 07:08:09",)]
+        assert q("SELECT from_json('{\"a\": 1, \"b\": [1,2]}', "
+                 "'a INT, b ARRAY<INT>')").collect() == \
+            [({"a": 1, "b": [1, 2]},)]
+        assert q("SELECT from_json('{\"m\": {\"a\": 1}}', "
+                 "'m MAP<STRING, INT>')").collect() == [({"m": {"a": 1}},)]

@@ -412,13 +412,26 @@ class Evaluator:
                     return Column(T.BOOL, data,
                                   None if bool(valid.all()) else valid.to(torch.uint8))
                 tv = v.value
+                if tv is None:
+                    targets.append(None)
+                    continue
                 if isinstance(cc.dtype, T.DecimalType):
                     tv = _to_scaled(tv, cc.dtype.scale)
                 elif isinstance(cc.dtype, T.DateType) and isinstance(tv, str):
                     tv = _date_str_to_days(tv)
                 targets.append(tv)
-            tt = torch.tensor(targets, dtype=cc.data.dtype, device=dev)
+            has_null = any(t is None for t in targets)
+            nn = [t for t in targets if t is not None]
+            tt = torch.tensor(nn, dtype=cc.data.dtype, device=dev) if nn \
+                else torch.zeros(0, dtype=cc.data.dtype, device=dev)
             mask = torch.isin(cc.data, tt)
+            if has_null:
+                # SQL 3VL: a NULL in the list makes non-matches NULL
+                valid = cc.valid_mask() & mask
+                data = ~mask if e.negated else mask
+                return Column(T.BOOL, data,
+                              None if bool(valid.all())
+                              else valid.to(torch.uint8))
         if e.negated:
             mask = ~mask
         return Column(T.BOOL, mask, cc.validity)

@@ -315,20 +315,34 @@ def _f_power(args, out, chunk, ev):
     return Column(T.F64, torch.pow(a.data.to(torch.float64), b.data.to(torch.float64)), _merge(a, b))
 
 
-def _f_greatest(args, out, chunk, ev):
-    cols = [_col(a, chunk) for a in args]
+def _minmax_n(args, out, chunk, ev, is_max: bool):
+    """greatest/least skip NULL arguments; all-NULL rows are NULL
+    (Spark semantics)."""
+    from .eval import cast_value
+
+    cols = [_col(cast_value(a, out, chunk), chunk) for a in args]
     data = cols[0].data.clone()
+    any_valid = cols[0].valid_mask().clone()
     for c in cols[1:]:
-        data = torch.maximum(data, c.data.to(data.dtype))
-    return Column(out, data, None)
+        cv = c.valid_mask()
+        cd = c.data.to(data.dtype)
+        op = torch.maximum if is_max else torch.minimum
+        cand = op(data, cd)
+        # rows where only one side is valid take that side
+        data = torch.where(any_valid & cv, cand,
+                           torch.where(cv, cd, data))
+        any_valid = any_valid | cv
+    return Column(out, data,
+                  None if bool(any_valid.all())
+                  else any_valid.to(torch.uint8))
+
+
+def _f_greatest(args, out, chunk, ev):
+    return _minmax_n(args, out, chunk, ev, True)
 
 
 def _f_least(args, out, chunk, ev):
-    cols = [_col(a, chunk) for a in args]
-    data = cols[0].data.clone()
-    for c in cols[1:]:
-        data = torch.minimum(data, c.data.to(data.dtype))
-    return Column(out, data, None)
+    return _minmax_n(args, out, chunk, ev, False)
 
 
 def _f_sign(args, out, chunk, ev):

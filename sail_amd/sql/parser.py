@@ -1623,7 +1623,11 @@ class Parser:
         else:
             raise SqlError("expected interval amount", self.sql, t.pos)
         unit = self.ident().lower().rstrip("s") if self.peek().kind == "ident" else "day"
-        self.eat_kw("TO") and self.ident()  # YEAR TO MONTH — take leading field only (TODO)
+        if self.eat_kw("TO"):
+            # multi-unit literals: '1-2' YEAR TO MONTH, 'd hh:mm:ss[.f]'
+            # DAY TO SECOND (and the HOUR/MINUTE prefixes)
+            end_unit = self.ident().lower().rstrip("s")
+            return self._multi_unit_interval(amount_str, unit, end_unit, t)
         amount = float(amount_str)
         # Intervals are represented as (months, microseconds) literal pairs;
         # arithmetic resolves them against date/timestamp operands.
@@ -1645,6 +1649,42 @@ class Parser:
             micros = int(amount * 1e6)
         else:
             raise SqlError(f"unsupported interval unit {unit}")
+        return S.Literal(("__interval__", months, micros), T.NULL)
+
+    def _multi_unit_interval(self, s: str, start: str, end: str, tok):
+        neg = s.strip().startswith("-")
+        body = s.strip().lstrip("+-")
+        months = 0
+        micros = 0
+        try:
+            if start == "year":
+                y, _, m = body.partition("-")
+                months = int(y) * 12 + (int(m) if m else 0)
+            else:
+                # [days ]hh:mm:ss[.frac] with start in day/hour/minute
+                days = 0
+                if start == "day":
+                    if " " in body:
+                        d, body = body.split(" ", 1)
+                        days = int(d)
+                    else:
+                        days, body = int(body), "0"
+                parts = body.split(":")
+                secs = 0.0
+                mult = {"hour": 3600, "minute": 60, "second": 1}[
+                    "hour" if start in ("day", "hour") else start]
+                for p in parts:
+                    secs += float(p or 0) * mult
+                    mult /= 60 if mult > 1 else 1
+                    if mult < 1:
+                        mult = 1
+                micros = int(round((days * 86400 + secs) * 1e6))
+        except (ValueError, KeyError):
+            raise SqlError(f"bad interval literal {s!r} for "
+                           f"{start.upper()} TO {end.upper()}",
+                           self.sql, tok.pos)
+        if neg:
+            months, micros = -months, -micros
         return S.Literal(("__interval__", months, micros), T.NULL)
 
     def _parse_type(self) -> T.DataType:

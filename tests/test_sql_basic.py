@@ -636,3 +636,22 @@ def test_order_by_qualified_group_key(s):
     r = s.sql("SELECT t.g, sum(t.v) FROM oq_t t GROUP BY t.g ORDER BY t.g"
               ).collect()
     assert r == [("a", 5), ("b", 1)]
+
+
+def test_null_semantics_and_intervals(s):
+    q = s.sql
+    # IN with NULL list entries: SQL three-valued logic
+    assert q("SELECT 1 IN (2, NULL), 1 IN (1, NULL), "
+             "1 NOT IN (2, NULL), 1 NOT IN (1, NULL)").collect() == \
+        [(None, True, None, False)]
+    # least/greatest skip nulls; all-null -> null
+    assert q("SELECT least(1, NULL, 3), greatest(NULL, NULL), "
+             "greatest(1, NULL, 5)").collect() == [(1, None, 5)]
+    # multi-unit interval literals
+    assert q("SELECT timestamp '2024-01-01 00:00:00' + "
+             "INTERVAL '1 02:03:04.5' DAY TO SECOND").collect() == \
+        [(1704160984500000,)]
+    assert q("SELECT date '2024-01-01' + INTERVAL '1-2' YEAR TO MONTH"
+             ).collect()[0][0].isoformat() == "2025-03-01"
+    assert q("SELECT INTERVAL '02:30' HOUR TO MINUTE").collect() == \
+        [("INTERVAL '2 hours 30 minutes'",)]

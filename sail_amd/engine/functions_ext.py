@@ -370,6 +370,7 @@ def _const_str(value_fn):
 
 
 _IMPLS["version"] = _const_str(lambda ev: "4.0.0-sail-mi355x")
+_IMPLS["current_version"] = _IMPLS["version"]
 _IMPLS["current_catalog"] = _const_str(lambda ev: "spark_catalog")
 _IMPLS["current_database"] = _const_str(lambda ev: "default")
 _IMPLS["current_user"] = _const_str(lambda ev: "root")

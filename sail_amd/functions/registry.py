@@ -171,7 +171,7 @@ _reg("length len char_length character_length octet_length bit_length "
      "instr locate position levenshtein crc32 ascii find_in_set", _i32)
 _reg("startswith endswith contains like ilike rlike regexp regexp_like", _bool)
 _reg("split", lambda a: T.ArrayType(T.STRING))
-_reg("md5 sha sha1 sha2 uuid", _string)
+_reg("md5 sha sha1 sha2 uuid current_version", _string)
 _reg("hash", _i32)
 _reg("xxhash64", _i64)
 
